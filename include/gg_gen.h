@@ -1,0 +1,157 @@
+/*
+ * Deterministic TPC-H-shaped synthetic data generator — the single
+ * definition used by BOTH the CPU oracle and the HIP device kernels
+ * (there is no network for real datasets; BASELINE.md §"Workloads"
+ * prescribes deterministic seeded dbgen-shaped synthetic inputs).
+ *
+ * Every column value is a pure function of (seed, table, row index) via
+ * splitmix64, so any subrange of any table can be (re)generated on any
+ * device without materializing anything else — the CPU oracle streams
+ * rows through the same functions the GPU generator kernel uses.
+ *
+ * Shapes (per TPC-H spec sizes):
+ *   customers(SF) = 150,000 × SF     c_custkey = i+1 (dense)
+ *   orders(SF)    = 1,500,000 × SF   o_orderkey = i+1 (dense)
+ *   lineitem(SF)  = 4 × orders(SF)   l_orderkey = i/4 + 1 (exactly 4
+ *                                    lines/order; dbgen draws 1–7, avg 4 —
+ *                                    documented deviation)
+ *
+ * Distributions follow dbgen's shapes (TPC-H spec 4.2.2/4.2.3):
+ *   o_orderdate   uniform [1992-01-01, 1998-08-02]
+ *   o_custkey     uniform 1..customers(SF)
+ *   l_shipdate    = o_orderdate + 1..121
+ *   l_receiptdate = l_shipdate + 1..30
+ *   l_returnflag  'R' or 'A' (50/50) if receiptdate <= 1995-06-17 else 'N'
+ *   l_linestatus  'O' if shipdate > 1995-06-17 else 'F'
+ *   l_quantity    1..50 (stored as cents: ×100, numeric(15,2))
+ *   l_extendedprice cents, 90,000..10,000,000 (~$900..$100k, dbgen range)
+ *   l_discount    0.00..0.10 (stored ×100: 0..10)
+ *   l_tax         0.00..0.08 (stored ×100: 0..8)
+ *   c_mktsegment  uniform over 5 segments; code 2 = 'MACHINERY'
+ *
+ * Decimal columns are scaled int64 (cents, scale 2) exactly as the engine
+ * stores numeric(15,2) — see DESIGN.md §data layout.
+ */
+#ifndef GG_GEN_H
+#define GG_GEN_H
+
+#include <stdint.h>
+#include "gg_pgdate.h"
+
+#if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
+#define GG_GEN_HOSTDEV __host__ __device__ static inline
+#else
+#define GG_GEN_HOSTDEV static inline
+#endif
+
+#define GG_SEED_DEFAULT 42ull
+
+/* table tags mixed into the stream so tables are independent */
+#define GG_TAB_LINEITEM 0x4c49ull	/* 'LI' */
+#define GG_TAB_ORDERS   0x4f52ull	/* 'OR' */
+#define GG_TAB_CUSTOMER 0x4355ull	/* 'CU' */
+
+#define GG_MKTSEG_MACHINERY 2
+
+/* dict codes for the char(1) columns: the actual ASCII byte is stored */
+#define GG_RF_A 'A'
+#define GG_RF_N 'N'
+#define GG_RF_R 'R'
+#define GG_LS_F 'F'
+#define GG_LS_O 'O'
+
+GG_GEN_HOSTDEV uint64_t gg_splitmix64(uint64_t x)
+{
+	x += 0x9e3779b97f4a7c15ull;
+	x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ull;
+	x = (x ^ (x >> 27)) * 0x94d049bb133111ebull;
+	return x ^ (x >> 31);
+}
+
+/* one 64-bit stream value for (seed, table, row, column-slot) */
+GG_GEN_HOSTDEV uint64_t gg_rnd(uint64_t seed, uint64_t tab, uint64_t row,
+			       uint64_t slot)
+{
+	return gg_splitmix64(seed ^ (tab << 48) ^ (slot << 40) ^ row);
+}
+
+/* ---- table sizes; sf100 = 100 etc. (sf as integer ×1; SF1 => sf=1) ---- */
+GG_GEN_HOSTDEV int64_t gg_n_customers(int64_t sf) { return 150000 * sf; }
+GG_GEN_HOSTDEV int64_t gg_n_orders(int64_t sf)    { return 1500000 * sf; }
+GG_GEN_HOSTDEV int64_t gg_n_lineitem(int64_t sf)  { return 6000000 * sf; }
+
+/* date-range constants are generated, never hardcoded */
+GG_GEN_HOSTDEV int32_t gg_orderdate_lo(void) { return gg_pgdate(1992, 1, 1); }
+GG_GEN_HOSTDEV int32_t gg_orderdate_hi(void) { return gg_pgdate(1998, 8, 2); }
+
+/* ---- orders ---- */
+GG_GEN_HOSTDEV int32_t gg_o_orderdate(uint64_t seed, int64_t orderkey)
+{
+	const int32_t lo = gg_orderdate_lo();
+	const int32_t n = gg_orderdate_hi() - lo + 1;
+
+	return lo + (int32_t) (gg_rnd(seed, GG_TAB_ORDERS,
+				      (uint64_t) orderkey, 0) % (uint64_t) n);
+}
+
+GG_GEN_HOSTDEV int64_t gg_o_custkey(uint64_t seed, int64_t orderkey,
+				    int64_t sf)
+{
+	return 1 + (int64_t) (gg_rnd(seed, GG_TAB_ORDERS,
+				     (uint64_t) orderkey, 1) %
+			      (uint64_t) gg_n_customers(sf));
+}
+
+GG_GEN_HOSTDEV int32_t gg_o_shippriority(uint64_t seed, int64_t orderkey)
+{
+	(void) seed; (void) orderkey;
+	return 0;		/* dbgen: constant 0 */
+}
+
+/* ---- customer ---- */
+GG_GEN_HOSTDEV uint8_t gg_c_mktsegment(uint64_t seed, int64_t custkey)
+{
+	return (uint8_t) (gg_rnd(seed, GG_TAB_CUSTOMER,
+				 (uint64_t) custkey, 0) % 5);
+}
+
+/* ---- lineitem (row i, 0-based; orderkey = i/4 + 1) ---- */
+typedef struct gg_lineitem_row
+{
+	int64_t l_orderkey;
+	int64_t l_quantity_c;	/* cents (scale 2) */
+	int64_t l_extendedprice_c;
+	int64_t l_discount_c;	/* 0..10 */
+	int64_t l_tax_c;	/* 0..8 */
+	int32_t l_shipdate;	/* DateADT */
+	uint8_t l_returnflag;	/* 'A'|'N'|'R' */
+	uint8_t l_linestatus;	/* 'F'|'O' */
+} gg_lineitem_row;
+
+GG_GEN_HOSTDEV void gg_gen_lineitem(uint64_t seed, int64_t i,
+				    gg_lineitem_row *r)
+{
+	const int64_t orderkey = (i >> 2) + 1;
+	const uint64_t u = (uint64_t) i;
+	const int32_t odate = gg_o_orderdate(seed, orderkey);
+	const int32_t cur = gg_pgdate(1995, 6, 17);	/* dbgen CURRENTDATE */
+	int32_t ship, receipt;
+
+	r->l_orderkey = orderkey;
+	r->l_quantity_c = (int64_t) (1 + gg_rnd(seed, GG_TAB_LINEITEM, u, 0) % 50) * 100;
+	r->l_extendedprice_c = (int64_t) (90000 +
+		gg_rnd(seed, GG_TAB_LINEITEM, u, 1) % 9910001ull);
+	r->l_discount_c = (int64_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 2) % 11);
+	r->l_tax_c = (int64_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 3) % 9);
+	ship = odate + 1 + (int32_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 4) % 121);
+	receipt = ship + 1 + (int32_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 5) % 30);
+	r->l_shipdate = ship;
+	r->l_linestatus = (ship > cur) ? GG_LS_O : GG_LS_F;
+	if (receipt <= cur)
+		r->l_returnflag = (gg_rnd(seed, GG_TAB_LINEITEM, u, 6) & 1)
+			? GG_RF_R : GG_RF_A;
+	else
+		r->l_returnflag = GG_RF_N;
+}
+
+#endif /* GG_GEN_H */

@@ -1,0 +1,333 @@
+"""ctypes bindings for the CPU oracle (oracle/liboracle.so) and the
+reference-pinned hash library (oracle/_ref/libpg_hashref.so).
+
+TEST INFRASTRUCTURE ONLY: imported by tests/, __graft_entry__.smoke()'s
+checker and bench.py's cpu_baseline leg — never by the product package.
+"""
+import ctypes
+import os
+import subprocess
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+
+
+class Q1Group(ctypes.Structure):
+    _fields_ = [
+        ("count", ctypes.c_int64),
+        ("sum_qty_c", ctypes.c_int64),
+        ("sum_base_c", ctypes.c_int64),
+        ("sum_dcol_c", ctypes.c_int64),
+        ("disc_lo", ctypes.c_uint64),
+        ("disc_hi", ctypes.c_int64),
+        ("charge_lo", ctypes.c_uint64),
+        ("charge_hi", ctypes.c_int64),
+    ]
+
+    def as_dict(self):
+        return {
+            "count": self.count,
+            "sum_qty_c": self.sum_qty_c,
+            "sum_base_c": self.sum_base_c,
+            "sum_dcol_c": self.sum_dcol_c,
+            "sum_disc4": (self.disc_hi << 64) | self.disc_lo,
+            "sum_charge6": (self.charge_hi << 64) | self.charge_lo,
+        }
+
+
+class Q3Row(ctypes.Structure):
+    _fields_ = [
+        ("orderkey", ctypes.c_int64),
+        ("rev_lo", ctypes.c_uint64),
+        ("rev_hi", ctypes.c_int64),
+        ("orderdate", ctypes.c_int32),
+        ("shippriority", ctypes.c_int32),
+    ]
+
+    def as_dict(self):
+        return {
+            "orderkey": self.orderkey,
+            "revenue4": (self.rev_hi << 64) | self.rev_lo,
+            "orderdate": self.orderdate,
+            "shippriority": self.shippriority,
+        }
+
+
+class Q3Result(ctypes.Structure):
+    _fields_ = [
+        ("n_out", ctypes.c_int64),
+        ("n_groups", ctypes.c_int64),
+        ("rev_sum_lo", ctypes.c_uint64),
+        ("rev_sum_hi", ctypes.c_int64),
+        ("group_checksum", ctypes.c_uint64),
+        ("n_join_rows", ctypes.c_int64),
+    ]
+
+    def as_dict(self):
+        return {
+            "n_groups": self.n_groups,
+            "rev_sum4": (self.rev_sum_hi << 64) | self.rev_sum_lo,
+            "group_checksum": self.group_checksum,
+            "n_join_rows": self.n_join_rows,
+        }
+
+
+def _build():
+    subprocess.run(["make", "-s", "-C", _DIR], check=True)
+
+
+def _load(name):
+    path = os.path.join(_DIR, name)
+    if not os.path.exists(path):
+        _build()
+    return ctypes.CDLL(path)
+
+
+_lib = None
+_ref = None
+
+I64 = ctypes.c_int64
+I32 = ctypes.c_int32
+U64 = ctypes.c_uint64
+U32 = ctypes.c_uint32
+
+_P_I64 = np.ctypeslib.ndpointer(np.int64, flags="C_CONTIGUOUS")
+_P_I32 = np.ctypeslib.ndpointer(np.int32, flags="C_CONTIGUOUS")
+_P_U8 = np.ctypeslib.ndpointer(np.uint8, flags="C_CONTIGUOUS")
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        _lib = _load("liboracle.so")
+        L = _lib
+        L.gg_oracle_q1_arrays.restype = ctypes.c_int
+        L.gg_oracle_q1_arrays.argtypes = [
+            _P_I32, _P_U8, _P_U8, _P_I64, _P_I64, _P_I64, _P_I64,
+            I64, I32, ctypes.POINTER(Q1Group), ctypes.c_int]
+        L.gg_oracle_q1_synth.restype = ctypes.c_int
+        L.gg_oracle_q1_synth.argtypes = [
+            U64, I64, I64, I64, I32, ctypes.POINTER(Q1Group), ctypes.c_int]
+        L.gg_oracle_q1_synth_segment.restype = ctypes.c_int
+        L.gg_oracle_q1_synth_segment.argtypes = [
+            U64, I64, I32, I32, I32, ctypes.POINTER(Q1Group), ctypes.c_int]
+        L.gg_numeric_to_str.argtypes = [U64, I64, ctypes.c_int, ctypes.c_char_p]
+        L.gg_avg_rscale.restype = ctypes.c_int
+        L.gg_avg_rscale.argtypes = [U64, I64, ctypes.c_int, I64]
+        L.gg_numeric_avg_to_str.argtypes = [U64, I64, ctypes.c_int, I64,
+                                            ctypes.c_char_p]
+        L.gg_oracle_q3_arrays.restype = ctypes.c_int
+        L.gg_oracle_q3_arrays.argtypes = [
+            _P_I64, _P_U8, I64, ctypes.c_uint8,
+            _P_I64, _P_I64, _P_I32, _P_I32, I64,
+            _P_I64, _P_I32, _P_I64, _P_I64, I64,
+            I32, I64, ctypes.POINTER(Q3Row), ctypes.POINTER(Q3Result),
+            ctypes.c_int]
+        L.gg_oracle_q3_synth.restype = ctypes.c_int
+        L.gg_oracle_q3_synth.argtypes = [
+            U64, I64, I32, I64, ctypes.POINTER(Q3Row),
+            ctypes.POINTER(Q3Result), ctypes.c_int]
+        L.gg_oracle_sumprice_arrays.restype = ctypes.c_int
+        L.gg_oracle_sumprice_arrays.argtypes = [
+            _P_I32, _P_I64, I64, I32, ctypes.POINTER(I64),
+            ctypes.POINTER(I64), ctypes.c_int]
+        L.gg_oracle_sumprice_synth.restype = ctypes.c_int
+        L.gg_oracle_sumprice_synth.argtypes = [
+            U64, I64, I32, ctypes.POINTER(I64), ctypes.POINTER(I64),
+            ctypes.c_int]
+        L.gg_oracle_hash_any.restype = U32
+        L.gg_oracle_hash_any.argtypes = [ctypes.c_char_p, ctypes.c_int]
+        L.gg_oracle_hash_uint32.restype = U32
+        L.gg_oracle_hash_uint32.argtypes = [U32]
+        L.gg_oracle_hashint4.restype = U32
+        L.gg_oracle_hashint4.argtypes = [I32]
+        L.gg_oracle_hashint8.restype = U32
+        L.gg_oracle_hashint8.argtypes = [I64]
+        L.gg_oracle_hashchar.restype = U32
+        L.gg_oracle_hashchar.argtypes = [ctypes.c_char]
+        L.gg_oracle_segment_int8.restype = I32
+        L.gg_oracle_segment_int8.argtypes = [I64, I32]
+        L.gg_oracle_segment_int4.restype = I32
+        L.gg_oracle_segment_int4.argtypes = [I32, I32]
+        L.gg_oracle_jump_hash.restype = I32
+        L.gg_oracle_jump_hash.argtypes = [U64, I32]
+        L.gg_oracle_pgdate.restype = I32
+        L.gg_oracle_pgdate.argtypes = [ctypes.c_int] * 3
+        L.gg_oracle_gen_lineitem.argtypes = [
+            U64, I64, I64, _P_I64, _P_I64, _P_I64, _P_I64, _P_I64,
+            _P_I32, _P_U8, _P_U8]
+        L.gg_oracle_gen_orders.argtypes = [
+            U64, I64, I64, I64, _P_I64, _P_I64, _P_I32, _P_I32]
+        L.gg_oracle_gen_customer.argtypes = [U64, I64, I64, _P_I64, _P_U8]
+    return _lib
+
+
+def ref():
+    """The reference's own hashfunc.c, compiled in place (None if the
+    prebuilt .so is missing and /root/reference is absent)."""
+    global _ref
+    if _ref is None:
+        path = os.path.join(_DIR, "_ref", "libpg_hashref.so")
+        if not os.path.exists(path) and os.path.isdir("/root/reference"):
+            _build()
+        if not os.path.exists(path):
+            return None
+        _ref = ctypes.CDLL(path)
+        R = _ref
+        R.ref_hash_any.restype = U32
+        R.ref_hash_any.argtypes = [ctypes.c_char_p, ctypes.c_int]
+        R.ref_hash_uint32.restype = U32
+        R.ref_hash_uint32.argtypes = [U32]
+        R.ref_hashint4.restype = U32
+        R.ref_hashint4.argtypes = [I32]
+        R.ref_hashint8.restype = U32
+        R.ref_hashint8.argtypes = [I64]
+        R.ref_hashchar.restype = U32
+        R.ref_hashchar.argtypes = [ctypes.c_char]
+    return _ref
+
+
+def pgdate(y, m, d):
+    return lib().gg_oracle_pgdate(y, m, d)
+
+
+def numeric_str(val128, scale):
+    buf = ctypes.create_string_buffer(80)
+    lo = val128 & ((1 << 64) - 1)
+    hi = val128 >> 64
+    lib().gg_numeric_to_str(U64(lo), I64(hi), scale, buf)
+    return buf.value.decode()
+
+
+def avg_str(sum128, sum_scale, count):
+    buf = ctypes.create_string_buffer(80)
+    lo = sum128 & ((1 << 64) - 1)
+    hi = sum128 >> 64
+    lib().gg_numeric_avg_to_str(U64(lo), I64(hi), sum_scale, I64(count), buf)
+    return buf.value.decode()
+
+
+def q1_arrays(shipdate, rflag, lstatus, qty, price, disc, tax, cutoff,
+              nthreads=0):
+    out = (Q1Group * 6)()
+    rc = lib().gg_oracle_q1_arrays(
+        shipdate, rflag, lstatus, qty, price, disc, tax,
+        len(shipdate), cutoff, out, nthreads)
+    assert rc == 0, rc
+    return [g.as_dict() for g in out]
+
+
+def q1_synth(seed, sf, cutoff, row_lo=0, row_hi=-1, nthreads=0):
+    out = (Q1Group * 6)()
+    rc = lib().gg_oracle_q1_synth(seed, sf, row_lo, row_hi, cutoff, out,
+                                  nthreads)
+    assert rc == 0, rc
+    return [g.as_dict() for g in out]
+
+
+def q1_synth_segment(seed, sf, nseg, seg, cutoff, nthreads=0):
+    out = (Q1Group * 6)()
+    rc = lib().gg_oracle_q1_synth_segment(seed, sf, nseg, seg, cutoff, out,
+                                          nthreads)
+    assert rc == 0, rc
+    return [g.as_dict() for g in out]
+
+
+def q1_finalize(groups):
+    """Occupied groups in (returnflag, linestatus) order, with the
+    reference's output strings (bb_mpph mpph1 column order)."""
+    names = [("A", "F"), ("A", "O"), ("N", "F"), ("N", "O"), ("R", "F"),
+             ("R", "O")]
+    rows = []
+    for g, (rf, ls) in zip(groups, names):
+        if g["count"] == 0:
+            continue
+        rows.append({
+            "l_returnflag": rf,
+            "l_linestatus": ls,
+            "sum_qty": numeric_str(g["sum_qty_c"], 2),
+            "sum_base_price": numeric_str(g["sum_base_c"], 2),
+            "sum_disc_price": numeric_str(g["sum_disc4"], 4),
+            "sum_charge": numeric_str(g["sum_charge6"], 6),
+            "avg_qty": avg_str(g["sum_qty_c"], 2, g["count"]),
+            "avg_price": avg_str(g["sum_base_c"], 2, g["count"]),
+            "avg_disc": avg_str(g["sum_dcol_c"], 2, g["count"]),
+            "count_order": g["count"],
+        })
+    return rows
+
+
+def q3_arrays(c_custkey, c_mktseg, seg_code, o_orderkey, o_custkey,
+              o_orderdate, o_shippriority, l_orderkey, l_shipdate, l_price,
+              l_disc, cutoff, k=10, nthreads=0):
+    topk = (Q3Row * 64)()
+    res = Q3Result()
+    rc = lib().gg_oracle_q3_arrays(
+        c_custkey, c_mktseg, len(c_custkey), seg_code,
+        o_orderkey, o_custkey, o_orderdate, o_shippriority, len(o_orderkey),
+        l_orderkey, l_shipdate, l_price, l_disc, len(l_orderkey),
+        cutoff, k, topk, ctypes.byref(res), nthreads)
+    assert rc == 0, rc
+    return ([topk[i].as_dict() for i in range(res.n_out)], res.as_dict())
+
+
+def q3_synth(seed, sf, cutoff, k=10, nthreads=0):
+    topk = (Q3Row * 64)()
+    res = Q3Result()
+    rc = lib().gg_oracle_q3_synth(seed, sf, cutoff, k, topk,
+                                  ctypes.byref(res), nthreads)
+    assert rc == 0, rc
+    return ([topk[i].as_dict() for i in range(res.n_out)], res.as_dict())
+
+
+def sumprice_synth(seed, sf, cutoff, nthreads=0):
+    s = I64()
+    c = I64()
+    rc = lib().gg_oracle_sumprice_synth(seed, sf, cutoff, ctypes.byref(s),
+                                        ctypes.byref(c), nthreads)
+    assert rc == 0, rc
+    return s.value, c.value
+
+
+def sumprice_arrays(shipdate, price, cutoff, nthreads=0):
+    s = I64()
+    c = I64()
+    rc = lib().gg_oracle_sumprice_arrays(shipdate, price, len(shipdate),
+                                         cutoff, ctypes.byref(s),
+                                         ctypes.byref(c), nthreads)
+    assert rc == 0, rc
+    return s.value, c.value
+
+
+def gen_lineitem(seed, row_lo, row_hi):
+    n = row_hi - row_lo
+    cols = dict(
+        orderkey=np.empty(n, np.int64), qty=np.empty(n, np.int64),
+        price=np.empty(n, np.int64), disc=np.empty(n, np.int64),
+        tax=np.empty(n, np.int64), shipdate=np.empty(n, np.int32),
+        rflag=np.empty(n, np.uint8), lstatus=np.empty(n, np.uint8))
+    lib().gg_oracle_gen_lineitem(
+        seed, row_lo, row_hi, cols["orderkey"], cols["qty"], cols["price"],
+        cols["disc"], cols["tax"], cols["shipdate"], cols["rflag"],
+        cols["lstatus"])
+    return cols
+
+
+def gen_orders(seed, sf, row_lo, row_hi):
+    n = row_hi - row_lo
+    cols = dict(
+        orderkey=np.empty(n, np.int64), custkey=np.empty(n, np.int64),
+        orderdate=np.empty(n, np.int32), shippriority=np.empty(n, np.int32))
+    lib().gg_oracle_gen_orders(seed, sf, row_lo, row_hi, cols["orderkey"],
+                               cols["custkey"], cols["orderdate"],
+                               cols["shippriority"])
+    return cols
+
+
+def gen_customer(seed, row_lo, row_hi):
+    n = row_hi - row_lo
+    cols = dict(custkey=np.empty(n, np.int64), mktseg=np.empty(n, np.uint8))
+    lib().gg_oracle_gen_customer(seed, row_lo, row_hi, cols["custkey"],
+                                 cols["mktseg"])
+    return cols
