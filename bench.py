@@ -1,0 +1,240 @@
+#!/usr/bin/env python3
+"""Benchmark driver (contract: see repo prompt / BASELINE.md).
+
+Workload at N=1: BASELINE.json configs[1] — TPC-H Q1 (scan + filter +
+group-by on lineitem) at SF100 on one MI355X, metric rows/s (lineitem
+rows scanned per second, whole-job).  At N>1 the same SF100 tables are
+hash-sharded across N GPUs (1 segment = 1 GPU, bit-exact cdbhash), so
+total work is fixed: "scaling": "strong".
+
+A step = one full Q1 pipeline pass over the HBM-resident shard
+(generation/upload excluded — inputs resident before the timed region).
+The line also carries an untimed "extra" with Q3 (configs[2]/[3])
+measured the same way outside the main timed region.
+
+Run:  python bench.py --gpus N --steps K --warmup W
+(N>1 via torch.distributed.run; RANK/LOCAL_RANK/WORLD_SIZE honored.)
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+SEED = 42
+SF = 100
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def cpu_baseline_leg(cutoff):
+    """Oracle ('port' of the reference executor semantics) timed on host
+    cores over a bounded sample of the same SF100 Q1 workload."""
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import pyoracle
+
+    cores = os.cpu_count() or 1
+    probe = 2_000_000
+    t0 = time.perf_counter()
+    pyoracle.q1_synth(SEED, SF, cutoff, 0, probe)
+    dt = time.perf_counter() - t0
+    rate = probe / dt
+    target_s = 12.0
+    sample = int(min(rate * target_s, 6_000_000 * SF))
+    t0 = time.perf_counter()
+    pyoracle.q1_synth(SEED, SF, cutoff, 0, sample)
+    dt = time.perf_counter() - t0
+    return {
+        "value": sample / dt,
+        "unit": "rows/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"Q1 over lineitem rows [0, {sample}) of SF{SF} "
+                  f"(streamed oracle, OpenMP {cores} threads, {dt:.1f}s)",
+    }
+
+
+def load_traffic_calibration():
+    """rocprofv3 PMC-measured HBM bytes per q1_agg launch, committed by a
+    profiling run (profiles/traffic_q1.json); null when absent."""
+    path = os.path.join(REPO, "profiles", "traffic_q1.json")
+    try:
+        with open(path) as f:
+            d = json.load(f)
+        if d.get("sf") == SF:
+            return d.get("bytes_per_launch")
+    except Exception:
+        pass
+    return None
+
+
+def main():
+    global SF
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int,
+                    default=int(os.environ.get("WORLD_SIZE", "1")))
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--sf", type=int, default=SF)
+    ap.add_argument("--skip-q3", action="store_true")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+    SF = args.sf
+
+    import torch
+    import torch.distributed as dist
+    from greengage_amd import Engine, PGDate
+    from greengage_amd.engine import PIPE_Q1, PIPE_Q3
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n = args.gpus
+    assert world == n or world == 1, (world, n)
+
+    if n > 1:
+        dist.init_process_group("gloo")
+
+    assert torch.cuda.is_available(), "bench requires an MI355X"
+    torch.cuda.set_device(local_rank)
+
+    eng = Engine(device=local_rank, n_segments=n, segment_id=rank)
+    if n > 1:
+        obj = [eng.comm_id() if rank == 0 else None]
+        dist.broadcast_object_list(obj, src=0)
+        eng.comm_init(obj[0])
+
+    cutoff_q1 = PGDate("1998-08-15")   # mpph1: 1998-12-01 - 108 days
+    cutoff_q3 = PGDate("1995-03-15")
+
+    log(f"[bench] generating SF{SF} shard on GPU (rank {rank}/{n})...")
+    t0 = time.perf_counter()
+    li = eng.register_synth("lineitem", seed=SEED, sf=SF)
+    li_rows = eng.table_nrows(li)
+    log(f"[bench] lineitem shard: {li_rows} rows "
+        f"({time.perf_counter() - t0:.1f}s)")
+
+    p_q1 = eng.compile(PIPE_Q1, lineitem=li, cutoff_date=cutoff_q1)
+
+    total_rows = 6_000_000 * SF
+
+    def barrier_sync():
+        torch.cuda.synchronize()
+        if n > 1:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        eng.execute_q1(p_q1)
+
+    stats_before = {s["name"]: dict(s) for s in eng.stats(p_q1)}
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        res_q1 = eng.execute_q1(p_q1)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    if n > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    value = total_rows * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    # roofline of the dominant kernel (q1_agg), rank-0 shard, HIP events
+    stats = {s["name"]: s for s in eng.stats(p_q1)}
+    q1s = stats["q1_agg"]
+    dl = q1s["launches"] - stats_before["q1_agg"]["launches"]
+    dms = q1s["total_ms"] - stats_before["q1_agg"]["total_ms"]
+    drows = q1s["rows_in"] - stats_before["q1_agg"]["rows_in"]
+    per_launch_s = (dms / dl) / 1000.0
+    rows_per_launch = drows / dl
+    algo_bytes = rows_per_launch * 38          # 38 B/row
+    achieved = algo_bytes / per_launch_s / 1e9
+    peak = 8000.0                               # HBM3E spec GB/s
+    roofline = {
+        "bound": "hbm",
+        "achieved": achieved,
+        "peak": peak,
+        "unit": "GB/s",
+        "frac": achieved / peak,
+        "traffic": load_traffic_calibration(),
+    }
+
+    extra = {}
+    if not args.skip_q3:
+        log("[bench] Q3 extra (untimed region)...")
+        od = eng.register_synth("orders", seed=SEED, sf=SF)
+        cu = eng.register_synth("customer", seed=SEED, sf=SF)
+        p_q3 = eng.compile(PIPE_Q3, lineitem=li, orders=od, customer=cu,
+                           cutoff_date=cutoff_q3, mktsegment=2, limit_k=10)
+        eng.execute_q3(p_q3)       # warmup
+        barrier_sync()
+        t0 = time.perf_counter()
+        q3_steps = 3
+        for _ in range(q3_steps):
+            rows_q3, hdr_q3 = eng.execute_q3(p_q3)
+        barrier_sync()
+        q3_el = time.perf_counter() - t0
+        if n > 1:
+            t = torch.tensor([q3_el], dtype=torch.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            q3_el = float(t.item())
+        # Q3 scans lineitem + orders + customer once per pass
+        q3_total = (6_000_000 + 1_500_000 + 150_000) * SF
+        extra = {
+            "q3_sf100_rows_per_s": q3_total * q3_steps / q3_el,
+            "q3_ms_per_step": q3_el / q3_steps * 1000.0,
+            "q3_n_groups": hdr_q3["n_groups"],
+            "q3_n_join_rows": hdr_q3["n_join_rows"],
+            "q3_kernel_stats": eng.stats(p_q3),
+        }
+
+    cpu_baseline = None
+    if rank == 0 and n == 1 and not args.skip_cpu_baseline:
+        log("[bench] CPU baseline (oracle, bounded sample)...")
+        cpu_baseline = cpu_baseline_leg(cutoff_q1)
+
+    if rank == 0:
+        line = {
+            "metric": "rows/s",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": "tpch_q1_sf100",
+                "sf": SF,
+                "query": "mpph1 (Q1, date - 108 days)",
+                "rows": total_rows,
+                "parallelism": f"mpp{n} (1 segment = 1 GPU, cdbhash "
+                               f"sharded, RCCL combine)",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+            "result_q1_count_total": sum(g["count"] for g in res_q1),
+            "extra": extra,
+        }
+        print(json.dumps(line), flush=True)
+
+    if n > 1:
+        dist.barrier()
+    eng.shutdown()
+
+
+if __name__ == "__main__":
+    main()

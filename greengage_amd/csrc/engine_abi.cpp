@@ -1,0 +1,1358 @@
+/*
+ * C-ABI engine implementation (see include/engine_abi.h for the
+ * boundary contract and reference citations).  Host orchestration for
+ * the HIP kernels in kernels.hip + RCCL exchanges in comm.cpp.
+ *
+ * PRODUCT code: self-contained (HIP runtime + RCCL only), no torch, no
+ * oracle/.  Loaded via ctypes by greengage_amd/ and — in a live
+ * Greengage deployment — by the segment shim.
+ */
+#include <algorithm>
+#include <cstdarg>
+#include <cstdio>
+#include <cstring>
+#include <vector>
+
+#include "engine_internal.h"
+#include "../../include/gg_gen.h"
+
+namespace gg
+{
+
+static thread_local char g_err[512];
+
+gg_status fail(gg_status st, const char *fmt, ...)
+{
+	va_list ap;
+
+	va_start(ap, fmt);
+	vsnprintf(g_err, sizeof(g_err), fmt, ap);
+	va_end(ap);
+	return st;
+}
+
+Engine &engine()
+{
+	static Engine e;
+
+	return e;
+}
+
+/* comm.cpp */
+gg_status comm_make_id(void *out_id128);
+gg_status comm_init(const void *id128);
+gg_status comm_destroy();
+bool comm_ready();
+gg_status comm_allgather_u64(const void *dev_send, void *dev_recv,
+			     size_t count);
+gg_status comm_alltoallv_i64(const int64_t *send_base,
+			     const unsigned long long *send_offs,
+			     const unsigned long long *send_cnts, int64_t *recv_base,
+			     const unsigned long long *recv_offs,
+			     const unsigned long long *recv_cnts);
+
+static uint64_t next_pow2(uint64_t v)
+{
+	uint64_t p = 1024;
+
+	while (p < v)
+		p <<= 1;
+	return p;
+}
+
+static size_t coltype_size(gg_coltype t)
+{
+	switch (t)
+	{
+		case GG_COL_INT64:
+		case GG_COL_DEC64_S2:
+			return 8;
+		case GG_COL_INT32:
+			return 4;
+		case GG_COL_CHAR1:
+			return 1;
+	}
+	return 0;
+}
+
+#define GG_TRY(expr) \
+	do { gg_status _s = (expr); if (_s != GG_OK) return _s; } while (0)
+
+/* device scratch counter helpers */
+static gg_status dev_counter(unsigned long long **p)
+{
+	GG_HIP(hipMalloc((void **) p, sizeof(unsigned long long)));
+	GG_HIP(hipMemset(*p, 0, sizeof(unsigned long long)));
+	return GG_OK;
+}
+
+static gg_status read_counter(unsigned long long *p, unsigned long long *out)
+{
+	GG_HIP(hipMemcpy(out, p, sizeof(*out), hipMemcpyDeviceToHost));
+	return GG_OK;
+}
+
+/* ---------------- int128 host helpers (combine/finalize) ---------------- */
+
+typedef __int128 i128;
+typedef unsigned __int128 u128;
+
+static i128 mk128(uint64_t lo, int64_t hi)
+{
+	return ((i128) hi << 64) | (i128) lo;
+}
+
+static void split128(i128 v, uint64_t *lo, int64_t *hi)
+{
+	*lo = (uint64_t) (u128) v;
+	*hi = (int64_t) (v >> 64);
+}
+
+/* ---------------- lifecycle ---------------- */
+
+extern "C" const char *gg_engine_last_error(void)
+{
+	return g_err;
+}
+
+extern "C" const char *gg_engine_build_info(void)
+{
+	return "greengage_amd engine: HIP gfx950 (CDNA4) + RCCL; built "
+		__DATE__ " " __TIME__;
+}
+
+extern "C" gg_status gg_engine_init(const gg_engine_config *cfg)
+{
+	Engine &e = engine();
+
+	if (e.inited)
+		return fail(GG_ESTATE, "engine already initialized");
+	if (!cfg || cfg->n_segments < 1 || cfg->segment_id < 0 ||
+	    cfg->segment_id >= cfg->n_segments)
+		return fail(GG_EINVAL, "bad engine config");
+	e.cfg = *cfg;
+	GG_HIP(hipSetDevice(cfg->device));
+	GG_HIP(hipStreamCreate(&e.stream));
+	e.inited = true;
+	return GG_OK;
+}
+
+extern "C" gg_status gg_engine_shutdown(void)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return GG_OK;
+	comm_destroy();
+	for (auto *t : e.tables)
+	{
+		if (t)
+			for (auto &c : t->cols)
+				(void) hipFree(c.dev);
+		delete t;
+	}
+	e.tables.clear();
+	for (auto *p : e.pipelines)
+		delete p;
+	e.pipelines.clear();
+	(void) hipStreamDestroy(e.stream);
+	e.stream = nullptr;
+	e.inited = false;
+	return GG_OK;
+}
+
+/* ---------------- tables ---------------- */
+
+static gg_status add_col(Table *t, const char *name, gg_coltype type,
+			 int64_t nrows, const void *host_src,
+			 void **out_dev)
+{
+	Table::Col c;
+
+	c.name = name;
+	c.type = type;
+	c.bytes = (size_t) nrows * coltype_size(type);
+	GG_HIP(hipMalloc(&c.dev, c.bytes ? c.bytes : 1));
+	if (host_src)
+		GG_HIP(hipMemcpy(c.dev, host_src, c.bytes,
+				 hipMemcpyHostToDevice));
+	t->cols.push_back(c);
+	if (out_dev)
+		*out_dev = c.dev;
+	return GG_OK;
+}
+
+extern "C" gg_status
+gg_engine_register_table(const char *name, const gg_column_desc *cols,
+			 int ncols, int64_t nrows, gg_table *out)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!name || !cols || ncols <= 0 || nrows < 0 || !out)
+		return fail(GG_EINVAL, "bad register_table args");
+	Table *t = new Table();
+
+	t->name = name;
+	t->nrows = nrows;
+	for (int i = 0; i < ncols; i++)
+	{
+		if (cols[i].device_data)
+		{
+			Table::Col c;
+
+			c.name = cols[i].name;
+			c.type = cols[i].type;
+			c.dev = const_cast<void *>(cols[i].device_data);
+			c.bytes = 0;	/* not owned */
+			t->cols.push_back(c);
+		}
+		else
+		{
+			gg_status s = add_col(t, cols[i].name, cols[i].type,
+					      nrows, cols[i].host_data,
+					      nullptr);
+			if (s != GG_OK)
+			{
+				delete t;
+				return s;
+			}
+		}
+	}
+	e.tables.push_back(t);
+	*out = (gg_table) (e.tables.size() - 1);
+	return GG_OK;
+}
+
+static Table *get_table(gg_table h)
+{
+	Engine &e = engine();
+
+	if (h < 0 || (size_t) h >= e.tables.size())
+		return nullptr;
+	return e.tables[h];
+}
+
+extern "C" gg_status gg_engine_table_nrows(gg_table h, int64_t *out)
+{
+	Table *t = get_table(h);
+
+	if (!t)
+		return fail(GG_EINVAL, "bad table handle %d", h);
+	*out = t->nrows;
+	return GG_OK;
+}
+
+extern "C" gg_status
+gg_engine_fetch_column(gg_table h, const char *col_name, void *host_buf,
+		       size_t buf_bytes)
+{
+	Table *t = get_table(h);
+
+	if (!t)
+		return fail(GG_EINVAL, "bad table handle %d", h);
+	for (auto &c : t->cols)
+		if (c.name == col_name)
+		{
+			size_t want = (size_t) t->nrows * coltype_size(c.type);
+
+			if (buf_bytes != want)
+				return fail(GG_EINVAL,
+					    "fetch_column size %zu != %zu",
+					    buf_bytes, want);
+			if (want)
+				GG_HIP(hipMemcpy(host_buf, c.dev, want,
+						 hipMemcpyDeviceToHost));
+			return GG_OK;
+		}
+	return fail(GG_EINVAL, "no column '%s' in table '%s'", col_name,
+		    t->name.c_str());
+}
+
+extern "C" gg_status gg_engine_drop_table(gg_table h)
+{
+	Engine &e = engine();
+	Table *t = get_table(h);
+
+	if (!t)
+		return fail(GG_EINVAL, "bad table handle %d", h);
+	for (auto &c : t->cols)
+		if (c.bytes)
+			(void) hipFree(c.dev);
+	delete t;
+	e.tables[h] = nullptr;
+	return GG_OK;
+}
+
+/*
+ * Synthetic shard registration: generates this segment's shard of the
+ * named TPC-H table directly in HBM (include/gg_gen.h; sharding by the
+ * table's DISTRIBUTED BY key via bit-exact cdbhash —
+ * input/bb_mpph.source:11–92 DDL, cdbhash.c:190/:549).
+ */
+extern "C" gg_status
+gg_engine_register_synth(const char *table_name, uint64_t seed, int64_t sf,
+			 gg_table *out)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	int nseg = e.cfg.n_segments;
+	int seg = e.cfg.segment_id;
+	std::string nm = table_name ? table_name : "";
+	int which;
+	int64_t total;
+
+	if (nm == "lineitem")
+	{
+		which = 0;
+		total = gg_n_lineitem(sf);
+	}
+	else if (nm == "orders")
+	{
+		which = 1;
+		total = gg_n_orders(sf);
+	}
+	else if (nm == "customer")
+	{
+		which = 2;
+		total = gg_n_customers(sf);
+	}
+	else
+		return fail(GG_EINVAL, "unknown synth table '%s'",
+			    table_name);
+
+	int64_t shard_rows = total;
+	unsigned long long *ctr = nullptr;
+
+	GG_TRY(dev_counter(&ctr));
+	if (nseg > 1)
+	{
+		unsigned long long c = 0;
+
+		GG_HIP(launch_count_shard(e.stream, 0, total, nseg, seg,
+					  which, seed, sf, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &c));
+		shard_rows = (int64_t) c;
+		GG_HIP(hipMemset(ctr, 0, 8));
+	}
+
+	Table *t = new Table();
+
+	t->name = nm;
+	t->nrows = shard_rows;
+
+	gg_status s = GG_OK;
+
+	if (nm == "lineitem")
+	{
+		void *ok, *q, *p, *d, *tx, *sd, *rf, *ls;
+
+		if ((s = add_col(t, "orderkey", GG_COL_INT64, shard_rows, nullptr, &ok)) == GG_OK &&
+		    (s = add_col(t, "qty", GG_COL_DEC64_S2, shard_rows, nullptr, &q)) == GG_OK &&
+		    (s = add_col(t, "price", GG_COL_DEC64_S2, shard_rows, nullptr, &p)) == GG_OK &&
+		    (s = add_col(t, "disc", GG_COL_DEC64_S2, shard_rows, nullptr, &d)) == GG_OK &&
+		    (s = add_col(t, "tax", GG_COL_DEC64_S2, shard_rows, nullptr, &tx)) == GG_OK &&
+		    (s = add_col(t, "shipdate", GG_COL_INT32, shard_rows, nullptr, &sd)) == GG_OK &&
+		    (s = add_col(t, "rflag", GG_COL_CHAR1, shard_rows, nullptr, &rf)) == GG_OK &&
+		    (s = add_col(t, "lstatus", GG_COL_CHAR1, shard_rows, nullptr, &ls)) == GG_OK)
+		{
+			hipError_t he = launch_gen_lineitem(
+				e.stream, seed, 0, total, nseg, seg,
+				(int64_t *) ok, (int64_t *) q, (int64_t *) p,
+				(int64_t *) d, (int64_t *) tx, (int32_t *) sd,
+				(uint8_t *) rf, (uint8_t *) ls, ctr);
+			if (he != hipSuccess)
+				s = fail(GG_EGPU, "gen_lineitem: %s",
+					 hipGetErrorString(he));
+		}
+	}
+	else if (nm == "orders")
+	{
+		void *ok, *ck, *od, *pr;
+
+		if ((s = add_col(t, "orderkey", GG_COL_INT64, shard_rows, nullptr, &ok)) == GG_OK &&
+		    (s = add_col(t, "custkey", GG_COL_INT64, shard_rows, nullptr, &ck)) == GG_OK &&
+		    (s = add_col(t, "orderdate", GG_COL_INT32, shard_rows, nullptr, &od)) == GG_OK &&
+		    (s = add_col(t, "shippriority", GG_COL_INT32, shard_rows, nullptr, &pr)) == GG_OK)
+		{
+			hipError_t he = launch_gen_orders(
+				e.stream, seed, sf, 0, total, nseg, seg,
+				(int64_t *) ok, (int64_t *) ck,
+				(int32_t *) od, (int32_t *) pr, ctr);
+			if (he != hipSuccess)
+				s = fail(GG_EGPU, "gen_orders: %s",
+					 hipGetErrorString(he));
+		}
+	}
+	else
+	{
+		void *ck, *ms;
+
+		if ((s = add_col(t, "custkey", GG_COL_INT64, shard_rows, nullptr, &ck)) == GG_OK &&
+		    (s = add_col(t, "mktseg", GG_COL_CHAR1, shard_rows, nullptr, &ms)) == GG_OK)
+		{
+			hipError_t he = launch_gen_customer(
+				e.stream, seed, 0, total, nseg, seg,
+				(int64_t *) ck, (uint8_t *) ms, ctr);
+			if (he != hipSuccess)
+				s = fail(GG_EGPU, "gen_customer: %s",
+					 hipGetErrorString(he));
+		}
+	}
+	GG_HIP(hipStreamSynchronize(e.stream));
+	(void) hipFree(ctr);
+	if (s != GG_OK)
+	{
+		for (auto &c : t->cols)
+			(void) hipFree(c.dev);
+		delete t;
+		return s;
+	}
+	e.tables.push_back(t);
+	*out = (gg_table) (e.tables.size() - 1);
+	return GG_OK;
+}
+
+/* ---------------- pipelines ---------------- */
+
+extern "C" gg_status
+gg_engine_compile_pipeline(const gg_pipeline_desc *desc, gg_pipeline *out)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!desc || !out)
+		return fail(GG_EINVAL, "null pipeline args");
+	switch (desc->kind)
+	{
+		case GG_PIPE_Q1:
+		case GG_PIPE_SUMPRICE:
+			if (!get_table(desc->lineitem))
+				return fail(GG_EINVAL, "Q1: bad lineitem table");
+			break;
+		case GG_PIPE_Q3:
+			if (!get_table(desc->lineitem) ||
+			    !get_table(desc->orders) ||
+			    !get_table(desc->customer))
+				return fail(GG_EINVAL, "Q3: bad table handle");
+			break;
+		default:
+			return fail(GG_ENOTSUP,
+				    "pipeline kind %d not supported (fall "
+				    "back to standard_ExecutorRun)",
+				    (int) desc->kind);
+	}
+	Pipeline *p = new Pipeline();
+
+	p->desc = *desc;
+	e.pipelines.push_back(p);
+	*out = (gg_pipeline) (e.pipelines.size() - 1);
+	return GG_OK;
+}
+
+extern "C" gg_status gg_engine_drop_pipeline(gg_pipeline h)
+{
+	Engine &e = engine();
+
+	if (h < 0 || (size_t) h >= e.pipelines.size() || !e.pipelines[h])
+		return fail(GG_EINVAL, "bad pipeline handle");
+	delete e.pipelines[h];
+	e.pipelines[h] = nullptr;
+	return GG_OK;
+}
+
+/* ---------------- execution: SUMPRICE ---------------- */
+
+static gg_status exec_sumprice(Pipeline *p, void *arena, size_t bytes,
+			       size_t *written)
+{
+	Engine &e = engine();
+	Table *li = get_table(p->desc.lineitem);
+
+	if (bytes < sizeof(gg_sumprice_result))
+		return fail(GG_EINVAL, "arena too small");
+	const int32_t *sd = (const int32_t *) li->col("shipdate");
+	const int64_t *pr = (const int64_t *) li->col("price");
+
+	if (!sd || !pr)
+		return fail(GG_EINVAL, "lineitem lacks shipdate/price");
+	SumPriceAcc *acc;
+
+	GG_HIP(hipMalloc((void **) &acc, sizeof(*acc)));
+	GG_HIP(hipMemset(acc, 0, sizeof(*acc)));
+	{
+		Timed tm(e.stream);
+
+		GG_HIP(launch_sumprice(e.stream, sd, pr, li->nrows,
+				       p->desc.cutoff_date, acc));
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("sumprice");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += li->nrows;
+		st.hbm_bytes += li->nrows * 12;
+	}
+	SumPriceAcc h;
+
+	GG_HIP(hipMemcpy(&h, acc, sizeof(h), hipMemcpyDeviceToHost));
+	(void) hipFree(acc);
+
+	u128 sum = h.sum_c;
+	u128 cnt = h.count;
+
+	if (e.cfg.n_segments > 1)
+	{
+		if (!comm_ready())
+			return fail(GG_ESTATE, "multi-segment without comm");
+		unsigned long long *dbuf;
+		std::vector<unsigned long long> all(2 * e.cfg.n_segments);
+
+		GG_HIP(hipMalloc((void **) &dbuf,
+				 (2 + 2 * e.cfg.n_segments) * 8));
+		GG_HIP(hipMemcpy(dbuf, &h, 16, hipMemcpyHostToDevice));
+		GG_TRY(comm_allgather_u64(dbuf, dbuf + 2, 2));
+		GG_HIP(hipMemcpy(all.data(), dbuf + 2, all.size() * 8,
+				 hipMemcpyDeviceToHost));
+		(void) hipFree(dbuf);
+		sum = 0;
+		cnt = 0;
+		for (int r = 0; r < e.cfg.n_segments; r++)
+		{
+			sum += all[2 * r];
+			cnt += all[2 * r + 1];
+		}
+	}
+	gg_sumprice_result *res = (gg_sumprice_result *) arena;
+
+	res->sum_c = (int64_t) sum;
+	res->count = (int64_t) cnt;
+	*written = sizeof(*res);
+	return GG_OK;
+}
+
+/* ---------------- execution: Q1 ---------------- */
+
+static gg_status exec_q1(Pipeline *p, void *arena, size_t bytes,
+			 size_t *written)
+{
+	Engine &e = engine();
+	Table *li = get_table(p->desc.lineitem);
+
+	if (bytes < sizeof(gg_q1_result))
+		return fail(GG_EINVAL, "arena too small");
+	const int32_t *sd = (const int32_t *) li->col("shipdate");
+	const uint8_t *rf = (const uint8_t *) li->col("rflag");
+	const uint8_t *ls = (const uint8_t *) li->col("lstatus");
+	const int64_t *q = (const int64_t *) li->col("qty");
+	const int64_t *pr = (const int64_t *) li->col("price");
+	const int64_t *d = (const int64_t *) li->col("disc");
+	const int64_t *tx = (const int64_t *) li->col("tax");
+
+	if (!sd || !rf || !ls || !q || !pr || !d || !tx)
+		return fail(GG_EINVAL, "lineitem lacks a Q1 column");
+
+	Q1DeviceAcc *acc;
+
+	GG_HIP(hipMalloc((void **) &acc, sizeof(*acc)));
+	GG_HIP(hipMemset(acc, 0, sizeof(*acc)));
+	{
+		Timed tm(e.stream);
+
+		GG_HIP(launch_q1(e.stream, sd, rf, ls, q, pr, d, tx,
+				 li->nrows, p->desc.cutoff_date, acc));
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("q1_agg");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += li->nrows;
+		st.hbm_bytes += li->nrows * 38;	/* SURVEY §8(d) */
+	}
+
+	/* combine partial states (2-stage agg: cdbgroup.c:1245 /
+	 * nodeAgg.c:2130–2142 — exact partials + one combine) */
+	static_assert(sizeof(Q1DeviceAcc) == 49 * 8, "acc layout");
+	std::vector<unsigned long long> parts;
+	int nseg = e.cfg.n_segments;
+
+	if (nseg > 1)
+	{
+		if (!comm_ready())
+			return fail(GG_ESTATE, "multi-segment without comm");
+		unsigned long long *gbuf;
+
+		GG_HIP(hipMalloc((void **) &gbuf, 49 * 8 * (size_t) (nseg + 1)));
+		GG_HIP(hipMemcpy(gbuf, acc, 49 * 8, hipMemcpyDeviceToDevice));
+		GG_TRY(comm_allgather_u64(gbuf, gbuf + 49, 49));
+		parts.resize(49 * (size_t) nseg);
+		GG_HIP(hipMemcpy(parts.data(), gbuf + 49, parts.size() * 8,
+				 hipMemcpyDeviceToHost));
+		(void) hipFree(gbuf);
+	}
+	else
+	{
+		parts.resize(49);
+		GG_HIP(hipMemcpy(parts.data(), acc, 49 * 8,
+				 hipMemcpyDeviceToHost));
+	}
+	(void) hipFree(acc);
+
+	gg_q1_result *res = (gg_q1_result *) arena;
+
+	std::memset(res, 0, sizeof(*res));
+	{
+		static const char RF[6] = { 'A', 'A', 'N', 'N', 'R', 'R' };
+		static const char LS[6] = { 'F', 'O', 'F', 'O', 'F', 'O' };
+		int occupied = 0;
+
+		for (int g = 0; g < 6; g++)
+		{
+			u128 cnt = 0, qty = 0, base = 0, dcol = 0;
+			i128 disc = 0, charge = 0;
+
+			for (int r = 0; r < nseg; r++)
+			{
+				const unsigned long long *v =
+					&parts[49 * (size_t) r + 8 * (size_t) g];
+
+				cnt += v[0];
+				qty += v[1];
+				base += v[2];
+				dcol += v[3];
+				disc += mk128(v[4], (int64_t) v[5]);
+				charge += mk128(v[6], (int64_t) v[7]);
+			}
+			for (int r = 0; r < nseg; r++)
+				if (parts[49 * (size_t) r + 48])
+					return fail(GG_EINVAL,
+						    "Q1: unexpected returnflag/"
+						    "linestatus byte in input");
+			gg_q1_result_group *o = &res->groups[g];
+
+			o->count = (int64_t) cnt;
+			o->sum_qty_c = (int64_t) qty;
+			o->sum_base_c = (int64_t) base;
+			o->sum_dcol_c = (int64_t) dcol;
+			split128(disc, &o->disc_lo, &o->disc_hi);
+			split128(charge, &o->charge_lo, &o->charge_hi);
+			o->returnflag = RF[g];
+			o->linestatus = LS[g];
+			if (o->count)
+				occupied++;
+		}
+		res->n_groups = occupied;
+	}
+	*written = sizeof(*res);
+	return GG_OK;
+}
+
+/* ---------------- execution: Q3 ---------------- */
+
+struct Q3TopkCmp
+{
+	bool operator()(const gg_q3_result_row &a,
+			const gg_q3_result_row &b) const
+	{
+		u128 ra = ((u128) (uint64_t) a.rev_hi << 64) | a.rev_lo;
+		u128 rb = ((u128) (uint64_t) b.rev_hi << 64) | b.rev_lo;
+
+		if (ra != rb)
+			return ra > rb;
+		if (a.orderdate != b.orderdate)
+			return a.orderdate < b.orderdate;
+		return a.orderkey < b.orderkey;
+	}
+};
+
+static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
+			 size_t *written)
+{
+	Engine &e = engine();
+	Table *li = get_table(p->desc.lineitem);
+	Table *od = get_table(p->desc.orders);
+	Table *cu = get_table(p->desc.customer);
+	int nseg = e.cfg.n_segments;
+	int64_t k = p->desc.limit_k > 0 ? p->desc.limit_k : 10;
+
+	if (bytes < sizeof(gg_q3_result_hdr) +
+	    (size_t) k * sizeof(gg_q3_result_row))
+		return fail(GG_EINVAL, "arena too small");
+	if (k > 1000)
+		return fail(GG_ENOTSUP, "LIMIT > 1000 not supported in v1");
+
+	const int64_t *c_ck = (const int64_t *) cu->col("custkey");
+	const uint8_t *c_ms = (const uint8_t *) cu->col("mktseg");
+	const int64_t *o_ok = (const int64_t *) od->col("orderkey");
+	const int64_t *o_ck = (const int64_t *) od->col("custkey");
+	const int32_t *o_dt = (const int32_t *) od->col("orderdate");
+	const int32_t *o_pr = (const int32_t *) od->col("shippriority");
+	const int64_t *l_ok = (const int64_t *) li->col("orderkey");
+	const int32_t *l_sd = (const int32_t *) li->col("shipdate");
+	const int64_t *l_pc = (const int64_t *) li->col("price");
+	const int64_t *l_dc = (const int64_t *) li->col("disc");
+
+	if (!c_ck || !c_ms || !o_ok || !o_ck || !o_dt || !o_pr || !l_ok ||
+	    !l_sd || !l_pc || !l_dc)
+		return fail(GG_EINVAL, "Q3: missing column");
+
+	int32_t cutoff = p->desc.cutoff_date;
+	uint8_t segcode = p->desc.mktsegment;
+	unsigned long long *ctr;
+
+	GG_TRY(dev_counter(&ctr));
+
+	/* 1. customer build side: size then build (nodeHash.c:450/:905) */
+	DeviceHashTable cust{};
+	{
+		unsigned long long nfil = 0;
+		Timed tm(e.stream);
+
+		GG_HIP(launch_count_filter_u8(e.stream, c_ms, segcode,
+					      cu->nrows, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nfil));
+		cust.nslots = next_pow2(2 * (nfil + 1));
+		GG_HIP(hipMalloc((void **) &cust.keys, cust.nslots * 8));
+		GG_HIP(hipMemset(cust.keys, 0, cust.nslots * 8));
+		GG_HIP(launch_build_set(e.stream, c_ck, c_ms, segcode,
+					cu->nrows, cust));
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("build_customer");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += cu->nrows;
+		st.rows_out += (int64_t) nfil;
+		st.hbm_bytes += cu->nrows * 9 * 2;
+	}
+
+	/* 2. orders side → orders hash table (single-seg: direct fuse;
+	 * multi-seg: two Motion redistributes over RCCL, §8(e)) */
+	DeviceHashTable ord{};
+	unsigned long long nmatch = 0;
+
+	if (nseg == 1)
+	{
+		GG_HIP(hipMemset(ctr, 0, 8));
+		{
+			Timed tm(e.stream);
+
+			GG_HIP(launch_count_orders_match(e.stream, o_ck, o_dt,
+							 od->nrows, cutoff,
+							 cust, ctr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr, &nmatch));
+			ord.nslots = next_pow2(2 * (nmatch + 1));
+			GG_HIP(hipMalloc((void **) &ord.keys, ord.nslots * 8));
+			GG_HIP(hipMalloc((void **) &ord.payload, ord.nslots * 8));
+			GG_HIP(hipMalloc((void **) &ord.rev, ord.nslots * 8));
+			GG_HIP(hipMemset(ord.keys, 0, ord.nslots * 8));
+			GG_HIP(hipMemset(ord.rev, 0, ord.nslots * 8));
+			GG_HIP(hipMemset(ctr, 0, 8));
+			GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt,
+						   o_pr, od->nrows, cutoff,
+						   cust, ord, ctr));
+			double ms = tm.stop();
+			KernelStatAcc &st = p->stat("build_orders");
+
+			st.launches++;
+			st.total_ms += ms;
+			st.rows_in += od->nrows;
+			st.rows_out += (int64_t) nmatch;
+			st.hbm_bytes += od->nrows * 24 * 2;
+		}
+	}
+	else
+	{
+		if (!comm_ready())
+			return fail(GG_ESTATE, "multi-segment without comm");
+		Timed tm(e.stream);
+
+		/* leg 1: filtered orders → owner of o_custkey */
+		int64_t *f_ck, *f_ok, *f_pay;
+		unsigned long long nfil = 0;
+
+		GG_HIP(hipMalloc((void **) &f_ck, (od->nrows + 1) * 8));
+		GG_HIP(hipMalloc((void **) &f_ok, (od->nrows + 1) * 8));
+		GG_HIP(hipMalloc((void **) &f_pay, (od->nrows + 1) * 8));
+		GG_HIP(hipMemset(ctr, 0, 8));
+		GG_HIP(launch_orders_filter_compact(e.stream, o_ok, o_ck,
+						    o_dt, o_pr, od->nrows,
+						    cutoff, f_ck, f_ok, f_pay,
+						    ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nfil));
+
+		/* partition by custkey */
+		unsigned long long *dcnt;
+		std::vector<unsigned long long> cnts(nseg), offs(nseg + 1, 0);
+
+		GG_HIP(hipMalloc((void **) &dcnt, nseg * 8));
+		GG_HIP(hipMemset(dcnt, 0, nseg * 8));
+		GG_HIP(launch_part_count(e.stream, f_ck, (int64_t) nfil, nseg,
+					 dcnt));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(cnts.data(), dcnt, nseg * 8,
+				 hipMemcpyDeviceToHost));
+		for (int i = 0; i < nseg; i++)
+			offs[i + 1] = offs[i] + cnts[i];
+
+		int64_t *s_ck, *s_ok, *s_pay;
+
+		GG_HIP(hipMalloc((void **) &s_ck, (nfil + 1) * 8));
+		GG_HIP(hipMalloc((void **) &s_ok, (nfil + 1) * 8));
+		GG_HIP(hipMalloc((void **) &s_pay, (nfil + 1) * 8));
+		{
+			unsigned long long *doffs;
+
+			GG_HIP(hipMalloc((void **) &doffs, nseg * 8));
+			GG_HIP(hipMemcpy(doffs, offs.data(), nseg * 8,
+					 hipMemcpyHostToDevice));
+			GG_HIP(launch_part_scatter3(e.stream, f_ck,
+						    (int64_t) nfil, nseg,
+						    f_ck, f_ok, f_pay, doffs,
+						    s_ck, s_ok, s_pay));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			(void) hipFree(doffs);
+		}
+		(void) hipFree(f_ck);
+		(void) hipFree(f_ok);
+		(void) hipFree(f_pay);
+
+		/* exchange counts: allgather the per-dest count vector */
+		std::vector<unsigned long long> allcnt((size_t) nseg * nseg);
+		{
+			unsigned long long *g;
+
+			GG_HIP(hipMalloc((void **) &g,
+					 (size_t) (nseg + (size_t) nseg * nseg) * 8));
+			GG_HIP(hipMemcpy(g, cnts.data(), nseg * 8,
+					 hipMemcpyHostToDevice));
+			GG_TRY(comm_allgather_u64(g, g + nseg, nseg));
+			GG_HIP(hipMemcpy(allcnt.data(), g + nseg,
+					 allcnt.size() * 8,
+					 hipMemcpyDeviceToHost));
+			(void) hipFree(g);
+		}
+		std::vector<unsigned long long> rcnts(nseg), roffs(nseg + 1, 0);
+
+		for (int s = 0; s < nseg; s++)
+			rcnts[s] = allcnt[(size_t) s * nseg + e.cfg.segment_id];
+		for (int i = 0; i < nseg; i++)
+			roffs[i + 1] = roffs[i] + rcnts[i];
+		uint64_t rtotal = roffs[nseg];
+
+		int64_t *r_ck, *r_ok, *r_pay;
+
+		GG_HIP(hipMalloc((void **) &r_ck, (rtotal + 1) * 8));
+		GG_HIP(hipMalloc((void **) &r_ok, (rtotal + 1) * 8));
+		GG_HIP(hipMalloc((void **) &r_pay, (rtotal + 1) * 8));
+		GG_TRY(comm_alltoallv_i64(s_ck, offs.data(), cnts.data(),
+					  r_ck, roffs.data(), rcnts.data()));
+		GG_TRY(comm_alltoallv_i64(s_ok, offs.data(), cnts.data(),
+					  r_ok, roffs.data(), rcnts.data()));
+		GG_TRY(comm_alltoallv_i64(s_pay, offs.data(), cnts.data(),
+					  r_pay, roffs.data(), rcnts.data()));
+		(void) hipFree(s_ck);
+		(void) hipFree(s_ok);
+		(void) hipFree(s_pay);
+
+		/* probe local customers → matched (okey, pay) */
+		int64_t *m_ok, *m_pay;
+		unsigned long long nm = 0;
+
+		GG_HIP(hipMalloc((void **) &m_ok, (rtotal + 1) * 8));
+		GG_HIP(hipMalloc((void **) &m_pay, (rtotal + 1) * 8));
+		GG_HIP(hipMemset(ctr, 0, 8));
+		GG_HIP(launch_probe_cust_compact(e.stream, r_ck, r_ok, r_pay,
+						 (int64_t) rtotal, cust, m_ok,
+						 m_pay, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nm));
+		(void) hipFree(r_ck);
+		(void) hipFree(r_ok);
+		(void) hipFree(r_pay);
+
+		/* leg 2: matched orders → owner of o_orderkey */
+		std::vector<unsigned long long> cnts2(nseg), offs2(nseg + 1, 0);
+
+		GG_HIP(hipMemset(dcnt, 0, nseg * 8));
+		GG_HIP(launch_part_count(e.stream, m_ok, (int64_t) nm, nseg,
+					 dcnt));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(cnts2.data(), dcnt, nseg * 8,
+				 hipMemcpyDeviceToHost));
+		(void) hipFree(dcnt);
+		for (int i = 0; i < nseg; i++)
+			offs2[i + 1] = offs2[i] + cnts2[i];
+
+		int64_t *s2_ok, *s2_pay;
+
+		GG_HIP(hipMalloc((void **) &s2_ok, (nm + 1) * 8));
+		GG_HIP(hipMalloc((void **) &s2_pay, (nm + 1) * 8));
+		{
+			unsigned long long *doffs;
+
+			GG_HIP(hipMalloc((void **) &doffs, nseg * 8));
+			GG_HIP(hipMemcpy(doffs, offs2.data(), nseg * 8,
+					 hipMemcpyHostToDevice));
+			GG_HIP(launch_part_scatter3(e.stream, m_ok,
+						    (int64_t) nm, nseg, m_ok,
+						    m_pay, nullptr, doffs,
+						    s2_ok, s2_pay, nullptr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			(void) hipFree(doffs);
+		}
+		(void) hipFree(m_ok);
+		(void) hipFree(m_pay);
+
+		std::vector<unsigned long long> allcnt2((size_t) nseg * nseg);
+		{
+			unsigned long long *g;
+
+			GG_HIP(hipMalloc((void **) &g,
+					 (size_t) (nseg + (size_t) nseg * nseg) * 8));
+			GG_HIP(hipMemcpy(g, cnts2.data(), nseg * 8,
+					 hipMemcpyHostToDevice));
+			GG_TRY(comm_allgather_u64(g, g + nseg, nseg));
+			GG_HIP(hipMemcpy(allcnt2.data(), g + nseg,
+					 allcnt2.size() * 8,
+					 hipMemcpyDeviceToHost));
+			(void) hipFree(g);
+		}
+		std::vector<unsigned long long> rcnts2(nseg), roffs2(nseg + 1, 0);
+
+		for (int s = 0; s < nseg; s++)
+			rcnts2[s] = allcnt2[(size_t) s * nseg + e.cfg.segment_id];
+		for (int i = 0; i < nseg; i++)
+			roffs2[i + 1] = roffs2[i] + rcnts2[i];
+		uint64_t rtotal2 = roffs2[nseg];
+
+		int64_t *r2_ok, *r2_pay;
+
+		GG_HIP(hipMalloc((void **) &r2_ok, (rtotal2 + 1) * 8));
+		GG_HIP(hipMalloc((void **) &r2_pay, (rtotal2 + 1) * 8));
+		GG_TRY(comm_alltoallv_i64(s2_ok, offs2.data(), cnts2.data(),
+					  r2_ok, roffs2.data(), rcnts2.data()));
+		GG_TRY(comm_alltoallv_i64(s2_pay, offs2.data(), cnts2.data(),
+					  r2_pay, roffs2.data(),
+					  rcnts2.data()));
+		(void) hipFree(s2_ok);
+		(void) hipFree(s2_pay);
+
+		nmatch = rtotal2;
+		ord.nslots = next_pow2(2 * (rtotal2 + 1));
+		GG_HIP(hipMalloc((void **) &ord.keys, ord.nslots * 8));
+		GG_HIP(hipMalloc((void **) &ord.payload, ord.nslots * 8));
+		GG_HIP(hipMalloc((void **) &ord.rev, ord.nslots * 8));
+		GG_HIP(hipMemset(ord.keys, 0, ord.nslots * 8));
+		GG_HIP(hipMemset(ord.rev, 0, ord.nslots * 8));
+		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_pay,
+					    (int64_t) rtotal2, ord));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		(void) hipFree(r2_ok);
+		(void) hipFree(r2_pay);
+
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("orders_exchange");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += od->nrows;
+		st.rows_out += (int64_t) nmatch;
+	}
+
+	/* 3. lineitem probe + group aggregation */
+	GG_HIP(hipMemset(ctr, 0, 8));
+	unsigned long long njoin = 0;
+	{
+		Timed tm(e.stream);
+
+		GG_HIP(launch_probe_lineitem(e.stream, l_ok, l_sd, l_pc, l_dc,
+					     li->nrows, cutoff, ord, ctr));
+		double ms = tm.stop();
+		GG_TRY(read_counter(ctr, &njoin));
+		KernelStatAcc &st = p->stat("probe_lineitem");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += li->nrows;
+		st.rows_out += (int64_t) njoin;
+		st.hbm_bytes += li->nrows * 28;	/* SURVEY §8(d) */
+	}
+
+	/* 4. stats + top-k */
+	unsigned long long *stats4;
+	unsigned long long hstats[4] = {0, 0, 0, 0};
+
+	GG_HIP(hipMalloc((void **) &stats4, 4 * 8));
+	GG_HIP(hipMemset(stats4, 0, 4 * 8));
+	{
+		Timed tm(e.stream);
+
+		GG_HIP(launch_q3_stats(e.stream, ord, stats4));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(hstats, stats4, 32, hipMemcpyDeviceToHost));
+
+		/* local top-k: histogram select then exact host sort */
+		unsigned long long maxrev = 0;
+
+		GG_HIP(hipMemset(stats4, 0, 8));
+		GG_HIP(launch_q3_maxrev(e.stream, ord, stats4));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(&maxrev, stats4, 8, hipMemcpyDeviceToHost));
+
+		std::vector<gg_q3_result_row> cand;
+
+		if (maxrev > 0)
+		{
+			int shift = 0;
+
+			while ((maxrev >> shift) > 65535)
+				shift++;
+			unsigned int *dhist;
+			std::vector<unsigned int> hist(65536);
+
+			GG_HIP(hipMalloc((void **) &dhist, 65536 * 4));
+			GG_HIP(hipMemset(dhist, 0, 65536 * 4));
+			GG_HIP(launch_q3_hist(e.stream, ord, shift, 0, dhist));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_HIP(hipMemcpy(hist.data(), dhist, 65536 * 4,
+					 hipMemcpyDeviceToHost));
+			(void) hipFree(dhist);
+
+			uint64_t cum = 0, thr_bin = 0;
+
+			for (int64_t b = 65535; b >= 0; b--)
+			{
+				cum += hist[b];
+				if (cum >= (uint64_t) k || b == 0)
+				{
+					thr_bin = (uint64_t) b;
+					break;
+				}
+			}
+			uint64_t threshold = thr_bin << shift;
+			uint64_t cap = cum + 65536;	/* bin-granule slack */
+			gg_q3_result_row *dout;
+			unsigned long long ncand = 0;
+
+			GG_HIP(hipMalloc((void **) &dout,
+					 cap * sizeof(gg_q3_result_row)));
+			GG_HIP(hipMemset(ctr, 0, 8));
+			GG_HIP(launch_q3_collect(e.stream, ord,
+						 threshold ? threshold : 1,
+						 dout, ctr, cap));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr, &ncand));
+			if (ncand > cap)
+				return fail(GG_EINVAL,
+					    "top-k candidate overflow (%llu)",
+					    ncand);
+			cand.resize(ncand);
+			if (ncand)
+				GG_HIP(hipMemcpy(cand.data(), dout,
+						 ncand * sizeof(gg_q3_result_row),
+						 hipMemcpyDeviceToHost));
+			(void) hipFree(dout);
+			std::sort(cand.begin(), cand.end(), Q3TopkCmp());
+			if ((int64_t) cand.size() > k)
+				cand.resize(k);
+		}
+
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("q3_topk");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += (int64_t) hstats[0];
+		st.rows_out += (int64_t) cand.size();
+
+		/* 5. global combine (gather of partial states + candidate
+		 * rows; the reference's Gather Motion to the QD) */
+		u128 revsum = ((u128) hstats[1]) | ((u128) hstats[2] << 64);
+		uint64_t ngroups = hstats[0];
+		uint64_t checksum = hstats[3];
+		uint64_t njoin_g = njoin;
+
+		if (nseg > 1)
+		{
+			/* stats: 4 words + k rows (32B each) per rank */
+			size_t words = 4 + (size_t) k * 4;
+			std::vector<unsigned long long> lbuf(words, 0);
+
+			lbuf[0] = ngroups;
+			lbuf[1] = (uint64_t) revsum;
+			lbuf[2] = (uint64_t) (revsum >> 64);
+			lbuf[3] = njoin_g;
+			/* checksum folded into rows? keep separate word:
+			 * extend buffer */
+			lbuf.resize(words + 1);
+			lbuf[words] = checksum;
+			for (size_t i = 0; i < cand.size(); i++)
+				std::memcpy(&lbuf[4 + i * 4], &cand[i], 32);
+
+			size_t per = lbuf.size();
+			unsigned long long *g;
+
+			GG_HIP(hipMalloc((void **) &g,
+					 (per + per * (size_t) nseg) * 8));
+			GG_HIP(hipMemcpy(g, lbuf.data(), per * 8,
+					 hipMemcpyHostToDevice));
+			GG_TRY(comm_allgather_u64(g, g + per, per));
+			std::vector<unsigned long long> all(per * (size_t) nseg);
+
+			GG_HIP(hipMemcpy(all.data(), g + per, all.size() * 8,
+					 hipMemcpyDeviceToHost));
+			(void) hipFree(g);
+
+			ngroups = 0;
+			revsum = 0;
+			checksum = 0;
+			njoin_g = 0;
+			cand.clear();
+			for (int r = 0; r < nseg; r++)
+			{
+				const unsigned long long *v = &all[per * (size_t) r];
+
+				ngroups += v[0];
+				revsum += ((u128) v[1]) | ((u128) v[2] << 64);
+				njoin_g += v[3];
+				checksum += v[per - 1];
+				for (int64_t i = 0; i < k; i++)
+				{
+					gg_q3_result_row row;
+
+					std::memcpy(&row, &v[4 + i * 4], 32);
+					if (row.rev_lo || row.rev_hi)
+						cand.push_back(row);
+				}
+			}
+			std::sort(cand.begin(), cand.end(), Q3TopkCmp());
+			if ((int64_t) cand.size() > k)
+				cand.resize(k);
+		}
+
+		gg_q3_result_hdr *hdr = (gg_q3_result_hdr *) arena;
+
+		hdr->n_out = (int64_t) cand.size();
+		hdr->n_groups = (int64_t) ngroups;
+		hdr->rev_sum_lo = (uint64_t) revsum;
+		hdr->rev_sum_hi = (int64_t) (revsum >> 64);
+		hdr->group_checksum = checksum;
+		hdr->n_join_rows = (int64_t) njoin_g;
+		std::memcpy(hdr + 1, cand.data(),
+			    cand.size() * sizeof(gg_q3_result_row));
+		*written = sizeof(*hdr) + cand.size() * sizeof(gg_q3_result_row);
+	}
+
+	(void) hipFree(stats4);
+	(void) hipFree(ctr);
+	(void) hipFree(cust.keys);
+	(void) hipFree(ord.keys);
+	(void) hipFree(ord.payload);
+	(void) hipFree(ord.rev);
+	return GG_OK;
+}
+
+extern "C" gg_status
+gg_engine_execute(gg_pipeline h, void *arena, size_t bytes, size_t *written)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (h < 0 || (size_t) h >= e.pipelines.size() || !e.pipelines[h])
+		return fail(GG_EINVAL, "bad pipeline handle");
+	if (!arena || !written)
+		return fail(GG_EINVAL, "null arena");
+	Pipeline *p = e.pipelines[h];
+
+	switch (p->desc.kind)
+	{
+		case GG_PIPE_Q1:
+			return exec_q1(p, arena, bytes, written);
+		case GG_PIPE_Q3:
+			return exec_q3(p, arena, bytes, written);
+		case GG_PIPE_SUMPRICE:
+			return exec_sumprice(p, arena, bytes, written);
+	}
+	return fail(GG_ENOTSUP, "unsupported pipeline");
+}
+
+/* ---------------- comm ---------------- */
+
+extern "C" gg_status gg_engine_comm_id(void *out_id128)
+{
+	if (!out_id128)
+		return fail(GG_EINVAL, "null id");
+	return comm_make_id(out_id128);
+}
+
+extern "C" gg_status gg_engine_comm_init(const void *id128)
+{
+	if (!engine().inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!id128)
+		return fail(GG_EINVAL, "null id");
+	return comm_init(id128);
+}
+
+extern "C" gg_status gg_engine_comm_destroy(void)
+{
+	return comm_destroy();
+}
+
+/* ---------------- stats ---------------- */
+
+extern "C" gg_status
+gg_engine_stats(gg_pipeline h, gg_kernel_stat *out, int cap, int *out_n)
+{
+	Engine &e = engine();
+
+	if (h < 0 || (size_t) h >= e.pipelines.size() || !e.pipelines[h])
+		return fail(GG_EINVAL, "bad pipeline handle");
+	Pipeline *p = e.pipelines[h];
+	int n = 0;
+
+	for (auto &s : p->stats)
+	{
+		if (n >= cap)
+			break;
+		std::snprintf(out[n].name, sizeof(out[n].name), "%s",
+			      s.name.c_str());
+		out[n].launches = s.launches;
+		out[n].total_ms = s.total_ms;
+		out[n].rows_in = s.rows_in;
+		out[n].rows_out = s.rows_out;
+		out[n].hbm_bytes_algorithmic = s.hbm_bytes;
+		n++;
+	}
+	*out_n = n;
+	return GG_OK;
+}
+
+/* ---------------- numeric display (product restatement of
+ * numeric.c display + select_div_scale:7144 + round_var) ---------------- */
+
+static int u128_digits_p(u128 v, int digs[48])
+{
+	int nd = 0;
+
+	if (v == 0)
+	{
+		digs[0] = 0;
+		return 1;
+	}
+	while (v)
+	{
+		digs[nd++] = (int) (v % 10);
+		v /= 10;
+	}
+	return nd;
+}
+
+extern "C" void
+gg_engine_numeric_str(uint64_t lo, int64_t hi, int scale, char *buf)
+{
+	i128 v = mk128(lo, hi);
+	u128 mag = v < 0 ? (u128) (-v) : (u128) v;
+	int digs[48];
+	int nd = u128_digits_p(mag, digs);
+	int p = 0;
+
+	if (v < 0)
+		buf[p++] = '-';
+	if (nd - scale <= 0)
+		buf[p++] = '0';
+	else
+		for (int i = nd - 1; i >= scale; i--)
+			buf[p++] = (char) ('0' + digs[i]);
+	if (scale > 0)
+	{
+		buf[p++] = '.';
+		for (int i = scale - 1; i >= 0; i--)
+			buf[p++] = (char) ('0' + (i < nd ? digs[i] : 0));
+	}
+	buf[p] = '\0';
+}
+
+static void nbase_norm_p(u128 mag, int scale, int *weight, int *firstdigit)
+{
+	int digs[48], nd, pmax, g;
+
+	if (mag == 0)
+	{
+		*weight = 0;
+		*firstdigit = 0;
+		return;
+	}
+	nd = u128_digits_p(mag, digs);
+	pmax = nd - 1 - scale;
+	g = (pmax >= 0) ? pmax / 4 : -((-pmax + 3) / 4);
+	{
+		int fd = 0;
+
+		for (int off = 3; off >= 0; off--)
+		{
+			int pos = 4 * g + off;
+			int i = pos + scale;
+
+			fd = fd * 10 + ((i >= 0 && i < nd) ? digs[i] : 0);
+		}
+		*weight = g;
+		*firstdigit = fd;
+	}
+}
+
+extern "C" void
+gg_engine_avg_str(uint64_t sum_lo, int64_t sum_hi, int sum_scale,
+		  int64_t count, char *buf)
+{
+	i128 sum = mk128(sum_lo, sum_hi);
+
+	if (count == 0)
+	{
+		std::strcpy(buf, "NULL");
+		return;
+	}
+	int w1, fd1, w2, fd2;
+	u128 smag = sum < 0 ? (u128) (-sum) : (u128) sum;
+	u128 cmag = count < 0 ? (u128) (-(i128) count) : (u128) count;
+
+	nbase_norm_p(smag, sum_scale, &w1, &fd1);
+	nbase_norm_p(cmag, 0, &w2, &fd2);
+	int qweight = w1 - w2;
+
+	if (fd1 <= fd2)
+		qweight--;
+	int rscale = 16 - qweight * 4;
+
+	if (rscale < sum_scale)
+		rscale = sum_scale;
+	if (rscale < 0)
+		rscale = 0;
+	if (rscale > 1000)
+		rscale = 1000;
+
+	u128 n = smag;
+
+	for (int i = 0; i < rscale - sum_scale; i++)
+		n *= 10;
+	u128 q = (2 * n + cmag) / (2 * cmag);
+	int neg = (sum < 0) ^ (count < 0);
+	i128 sq = neg ? -(i128) q : (i128) q;
+	uint64_t qlo;
+	int64_t qhi;
+
+	split128(sq, &qlo, &qhi);
+	gg_engine_numeric_str(qlo, qhi, rscale, buf);
+}
+
+}				/* namespace gg */

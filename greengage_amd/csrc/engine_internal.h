@@ -1,0 +1,234 @@
+/*
+ * Internal engine declarations — host side (engine_abi.cpp, comm.cpp)
+ * and the kernel-launcher entry points implemented in kernels.hip.
+ * Product code: never includes anything from oracle/.
+ */
+#ifndef GG_ENGINE_INTERNAL_H
+#define GG_ENGINE_INTERNAL_H
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <string>
+#include <vector>
+
+#include "../../include/engine_abi.h"
+#include "../../include/gg_result.h"
+
+namespace gg
+{
+
+/* error plumbing: set message, return status */
+gg_status fail(gg_status st, const char *fmt, ...);
+
+#define GG_HIP(call) \
+	do { \
+		hipError_t _e = (call); \
+		if (_e != hipSuccess) \
+			return ::gg::fail(GG_EGPU, "%s:%d HIP error %s in %s", \
+					  __FILE__, __LINE__, \
+					  hipGetErrorString(_e), #call); \
+	} while (0)
+
+/* device accumulator block for Q1 (matches kernels.hip layout) */
+struct Q1DeviceAcc
+{
+	/* [group][field]: count, qty, base, dcol, disc_lo, disc_hi,
+	 * charge_lo, charge_hi — 6 groups × 8 u64 (the two int128 sums
+	 * carry into _hi on _lo wraparound) */
+	unsigned long long v[6][8];
+	unsigned long long err;	/* nonzero: unexpected flag/status byte */
+};
+
+struct SumPriceAcc
+{
+	unsigned long long sum_c;
+	unsigned long long count;
+};
+
+/* open-addressing hash table (SoA), power-of-two slots, key 0 = empty */
+struct DeviceHashTable
+{
+	unsigned long long *keys = nullptr;	/* build keys (int64 > 0) */
+	unsigned long long *payload = nullptr;	/* (date u32) | (prio u64<<32) */
+	unsigned long long *rev = nullptr;	/* Q3 group revenue, scale 4 */
+	uint64_t nslots = 0;			/* power of two */
+};
+
+struct Table
+{
+	std::string name;
+	int64_t nrows = 0;
+	/* named device columns (all 8-byte or 4/1-byte arrays) */
+	struct Col
+	{
+		std::string name;
+		gg_coltype type;
+		void *dev = nullptr;
+		size_t bytes = 0;
+	};
+	std::vector<Col> cols;
+
+	void *col(const char *n) const
+	{
+		for (auto &c : cols)
+			if (c.name == n)
+				return c.dev;
+		return nullptr;
+	}
+};
+
+struct KernelStatAcc
+{
+	std::string name;
+	int64_t launches = 0;
+	double total_ms = 0;
+	int64_t rows_in = 0;
+	int64_t rows_out = 0;
+	int64_t hbm_bytes = 0;
+};
+
+struct Pipeline
+{
+	gg_pipeline_desc desc;
+	std::vector<KernelStatAcc> stats;
+
+	KernelStatAcc &stat(const char *name)
+	{
+		for (auto &s : stats)
+			if (s.name == name)
+				return s;
+		stats.push_back({});
+		stats.back().name = name;
+		return stats.back();
+	}
+};
+
+struct Engine
+{
+	gg_engine_config cfg{};
+	bool inited = false;
+	hipStream_t stream = nullptr;
+	std::vector<Table *> tables;
+	std::vector<Pipeline *> pipelines;
+	void *comm = nullptr;	/* ncclComm_t when comm_init'ed */
+};
+
+Engine &engine();
+
+/* timed launch helper: records HIP events around fn(stream) */
+struct Timed
+{
+	hipEvent_t a = nullptr, b = nullptr;
+	hipStream_t s;
+	explicit Timed(hipStream_t st) : s(st)
+	{
+		(void) hipEventCreate(&a);
+		(void) hipEventCreate(&b);
+		(void) hipEventRecord(a, s);
+	}
+	/* returns elapsed ms (synchronizes the events) */
+	double stop()
+	{
+		float ms = 0;
+		(void) hipEventRecord(b, s);
+		(void) hipEventSynchronize(b);
+		(void) hipEventElapsedTime(&ms, a, b);
+		(void) hipEventDestroy(a);
+		(void) hipEventDestroy(b);
+		return ms;
+	}
+};
+
+/* ---- kernel launchers (kernels.hip) — return hipError_t ---- */
+
+hipError_t launch_gen_lineitem(hipStream_t s, uint64_t seed, int64_t row_lo,
+			       int64_t n, int nseg, int seg,
+			       int64_t *orderkey, int64_t *qty, int64_t *price,
+			       int64_t *disc, int64_t *tax, int32_t *shipdate,
+			       uint8_t *rflag, uint8_t *lstatus,
+			       unsigned long long *out_count);
+hipError_t launch_gen_orders(hipStream_t s, uint64_t seed, int64_t sf,
+			     int64_t row_lo, int64_t n, int nseg, int seg,
+			     int64_t *orderkey, int64_t *custkey,
+			     int32_t *orderdate, int32_t *prio,
+			     unsigned long long *out_count);
+hipError_t launch_gen_customer(hipStream_t s, uint64_t seed, int64_t row_lo,
+			       int64_t n, int nseg, int seg, int64_t *custkey,
+			       uint8_t *mktseg, unsigned long long *out_count);
+
+hipError_t launch_q1(hipStream_t s, const int32_t *shipdate,
+		     const uint8_t *rflag, const uint8_t *lstatus,
+		     const int64_t *qty, const int64_t *price,
+		     const int64_t *disc, const int64_t *tax, int64_t n,
+		     int32_t cutoff, Q1DeviceAcc *acc);
+
+hipError_t launch_count_shard(hipStream_t s, int64_t row_lo, int64_t n,
+			      int nseg, int seg, int which_table,
+			      uint64_t seed, int64_t sf,
+			      unsigned long long *out_count);
+
+hipError_t launch_sumprice(hipStream_t s, const int32_t *shipdate,
+			   const int64_t *price, int64_t n, int32_t cutoff,
+			   SumPriceAcc *acc);
+
+hipError_t launch_build_set(hipStream_t s, const int64_t *keys,
+			    const uint8_t *filter_col, uint8_t filter_val,
+			    int64_t n, DeviceHashTable t);
+hipError_t launch_build_orders(hipStream_t s, const int64_t *okey,
+			       const int64_t *ckey, const int32_t *odate,
+			       const int32_t *prio, int64_t n, int32_t cutoff,
+			       DeviceHashTable cust, DeviceHashTable ord,
+			       unsigned long long *match_count);
+hipError_t launch_probe_lineitem(hipStream_t s, const int64_t *okey,
+				 const int32_t *shipdate,
+				 const int64_t *price, const int64_t *disc,
+				 int64_t n, int32_t cutoff,
+				 DeviceHashTable ord,
+				 unsigned long long *join_rows);
+hipError_t launch_q3_stats(hipStream_t s, DeviceHashTable ord,
+			   unsigned long long *out4 /* ngroups, revlo,
+						     * revhi, checksum */);
+hipError_t launch_q3_maxrev(hipStream_t s, DeviceHashTable ord,
+			    unsigned long long *out_max);
+hipError_t launch_q3_hist(hipStream_t s, DeviceHashTable ord, int shift,
+			  uint64_t lo_bound, unsigned int *hist64k);
+hipError_t launch_q3_collect(hipStream_t s, DeviceHashTable ord,
+			     uint64_t threshold, gg_q3_result_row *out,
+			     unsigned long long *out_count, uint64_t cap);
+
+hipError_t launch_count_filter_u8(hipStream_t s, const uint8_t *col,
+				  uint8_t val, int64_t n,
+				  unsigned long long *out);
+hipError_t launch_count_orders_match(hipStream_t s, const int64_t *ckey,
+				     const int32_t *odate, int64_t n,
+				     int32_t cutoff, DeviceHashTable cust,
+				     unsigned long long *out);
+hipError_t launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
+					const int64_t *ckey,
+					const int32_t *odate,
+					const int32_t *prio, int64_t n,
+					int32_t cutoff, int64_t *out_ckey,
+					int64_t *out_okey, int64_t *out_pay,
+					unsigned long long *out_count);
+hipError_t launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,
+				     const int64_t *okey, const int64_t *pay,
+				     int64_t n, DeviceHashTable cust,
+				     int64_t *out_okey, int64_t *out_pay,
+				     unsigned long long *out_count);
+hipError_t launch_insert_orders(hipStream_t s, const int64_t *okey,
+				const int64_t *pay, int64_t n,
+				DeviceHashTable ord);
+
+/* Motion partition: count then scatter rows to per-dest buffers */
+hipError_t launch_part_count(hipStream_t s, const int64_t *key, int64_t n,
+			     int nseg, unsigned long long *counts);
+hipError_t launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n,
+				int nseg,
+				const int64_t *a, const int64_t *b,
+				const int64_t *c,
+				unsigned long long *offsets,
+				int64_t *oa, int64_t *ob, int64_t *oc);
+
+}				/* namespace gg */
+
+#endif

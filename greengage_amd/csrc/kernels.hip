@@ -1,0 +1,1067 @@
+/*
+ * MI355X (gfx950, CDNA4) kernels for the hot path — written for 64-wide
+ * wavefronts, LDS-privatized accumulators, coalesced 8-byte column
+ * streams from HBM.  No MFMA: every kernel here is HBM-bandwidth-bound
+ * gather/scan work (SURVEY §8(d) roofline).
+ *
+ * Reference semantics implemented (cited per kernel):
+ *   k_q1_agg          execScan.c:110–214 + execHHashagg.c:905–1081 +
+ *                     nodeAgg.c:393–860 (scan+filter+group agg, exact
+ *                     scaled-int arithmetic)
+ *   k_build_* / k_probe_*  nodeHash.c:88–176/:905/:1002, nodeHashjoin.c:
+ *                     78–510 (hash build/probe; open addressing instead
+ *                     of chained buckets — GPU-native layout, same join
+ *                     row-set semantics; rows a5–a7)
+ *   k_sumprice        BASELINE config 1
+ *   k_gen_*           include/gg_gen.h (deterministic synthetic data)
+ *   k_part_*          nodeMotion.c:1574 doSendTuple hash routing via
+ *                     bit-exact cdbhash (include/gg_pg_hash.h; row a11)
+ */
+#include <hip/hip_runtime.h>
+
+#include "../../include/gg_pg_hash.h"
+#include "../../include/gg_gen.h"
+#include "../../include/gg_checksum.h"
+#include "engine_internal.h"
+
+namespace gg
+{
+
+static constexpr int THREADS = 256;
+/* 256 CUs × 8 blocks of 256 threads per CU (Guideline 11) */
+static constexpr int MAX_BLOCKS = 2048;
+
+static inline int grid_for(int64_t n)
+{
+	int64_t b = (n + THREADS - 1) / THREADS;
+
+	return (int) (b < 1 ? 1 : (b > MAX_BLOCKS ? MAX_BLOCKS : b));
+}
+
+/* ------------------------------------------------------------------ */
+/* Q1: fused scan + predicate + 6-group aggregation                    */
+/* ------------------------------------------------------------------ */
+
+/* LDS accumulator: 32 replicas (lane % 32) × 6 groups × 8 fields,
+ * replica stride padded to 49 u64 so consecutive replicas land on
+ * different LDS banks for the b64 atomics. */
+static constexpr int Q1_REPL = 32;
+static constexpr int Q1_FIELDS = 8;	/* cnt,qty,base,dcol,disc,+pad.. */
+static constexpr int Q1_SLOTS = 6 * Q1_FIELDS;	/* 48 */
+static constexpr int Q1_STRIDE = Q1_SLOTS + 1;	/* 49: odd → bank spread */
+
+__global__ __launch_bounds__(THREADS, 2)
+void k_q1_agg(const int32_t *__restrict__ shipdate,
+	      const uint8_t *__restrict__ rflag,
+	      const uint8_t *__restrict__ lstatus,
+	      const int64_t *__restrict__ qty,
+	      const int64_t *__restrict__ price,
+	      const int64_t *__restrict__ disc,
+	      const int64_t *__restrict__ tax,
+	      int64_t n, int32_t cutoff, Q1DeviceAcc *acc)
+{
+	__shared__ unsigned long long lds[Q1_REPL * Q1_STRIDE];
+
+	for (int i = threadIdx.x; i < Q1_REPL * Q1_STRIDE; i += blockDim.x)
+		lds[i] = 0;
+	__syncthreads();
+
+	const int rep = threadIdx.x & (Q1_REPL - 1);
+	unsigned long long *mine = &lds[rep * Q1_STRIDE];
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long bad = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int32_t sd = shipdate[i];
+
+		if (sd > cutoff)	/* qual: l_shipdate <= cutoff */
+			continue;
+		{
+			uint8_t rf = rflag[i];
+			uint8_t ls = lstatus[i];
+			int fi = (rf == 'N') ? 1 : ((rf == 'R') ? 2 : 0);
+			int si = (ls == 'O') ? 1 : 0;
+
+			bad |= (rf != 'A' && rf != 'N' && rf != 'R');
+			bad |= (ls != 'F' && ls != 'O');
+			{
+				int g = (fi * 2 + si) * Q1_FIELDS;
+				int64_t q = qty[i];
+				int64_t p = price[i];
+				int64_t d = disc[i];
+				int64_t t = tax[i];
+				/* mul_var exact: scale 2+2 then 4+2 */
+				unsigned long long disc4 =
+					(unsigned long long) (p * (100 - d));
+				unsigned long long charge6 =
+					disc4 * (unsigned long long) (100 + t);
+
+				atomicAdd(&mine[g + 0], 1ull);
+				atomicAdd(&mine[g + 1], (unsigned long long) q);
+				atomicAdd(&mine[g + 2], (unsigned long long) p);
+				atomicAdd(&mine[g + 3], (unsigned long long) d);
+				atomicAdd(&mine[g + 4], disc4);
+				atomicAdd(&mine[g + 5], charge6);
+			}
+		}
+	}
+	if (bad)
+		atomicOr(&acc->err, 1ull);
+	__syncthreads();
+
+	/* flush: per-block sums fit u64 (DESIGN.md §overflow budget);
+	 * 128-bit carry handled at the global accumulator */
+	for (int slot = threadIdx.x; slot < Q1_SLOTS; slot += blockDim.x)
+	{
+		unsigned long long sum = 0;
+
+		for (int r = 0; r < Q1_REPL; r++)
+			sum += lds[r * Q1_STRIDE + slot];
+		if (!sum)
+			continue;
+		{
+			int g = slot / Q1_FIELDS;
+			int f = slot % Q1_FIELDS;
+
+			switch (f)
+			{
+				case 0: atomicAdd(&acc->v[g][0], sum); break;
+				case 1: atomicAdd(&acc->v[g][1], sum); break;
+				case 2: atomicAdd(&acc->v[g][2], sum); break;
+				case 3: atomicAdd(&acc->v[g][3], sum); break;
+				case 4:	/* disc4 int128: lo + carry */
+				{
+					unsigned long long old =
+						atomicAdd(&acc->v[g][4], sum);
+					if (old + sum < old)
+						atomicAdd(&acc->v[g][5], 1ull);
+					break;
+				}
+				case 5:	/* charge6 int128 */
+				{
+					unsigned long long old =
+						atomicAdd(&acc->v[g][6], sum);
+					if (old + sum < old)
+						atomicAdd(&acc->v[g][7], 1ull);
+					break;
+				}
+				default: break;
+			}
+		}
+	}
+}
+
+hipError_t
+launch_q1(hipStream_t s, const int32_t *shipdate, const uint8_t *rflag,
+	  const uint8_t *lstatus, const int64_t *qty, const int64_t *price,
+	  const int64_t *disc, const int64_t *tax, int64_t n, int32_t cutoff,
+	  Q1DeviceAcc *acc)
+{
+	hipLaunchKernelGGL(k_q1_agg, dim3(grid_for(n)), dim3(THREADS), 0, s,
+			   shipdate, rflag, lstatus, qty, price, disc, tax,
+			   n, cutoff, acc);
+	return hipGetLastError();
+}
+
+/* ------------------------------------------------------------------ */
+/* BASELINE config 1: sum(l_extendedprice) where l_shipdate < cutoff   */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(THREADS, 2)
+void k_sumprice(const int32_t *__restrict__ shipdate,
+		const int64_t *__restrict__ price, int64_t n, int32_t cutoff,
+		SumPriceAcc *acc)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long sum = 0, cnt = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		if (shipdate[i] < cutoff)
+		{
+			sum += (unsigned long long) price[i];
+			cnt++;
+		}
+	/* wave reduction then one atomic per wave */
+	for (int off = 32; off; off >>= 1)
+	{
+		sum += __shfl_down(sum, off, 64);
+		cnt += __shfl_down(cnt, off, 64);
+	}
+	if ((threadIdx.x & 63) == 0)
+	{
+		atomicAdd(&acc->sum_c, sum);
+		atomicAdd(&acc->count, cnt);
+	}
+}
+
+hipError_t
+launch_sumprice(hipStream_t s, const int32_t *shipdate, const int64_t *price,
+		int64_t n, int32_t cutoff, SumPriceAcc *acc)
+{
+	hipLaunchKernelGGL(k_sumprice, dim3(grid_for(n)), dim3(THREADS), 0, s,
+			   shipdate, price, n, cutoff, acc);
+	return hipGetLastError();
+}
+
+/* ------------------------------------------------------------------ */
+/* synthetic generation (gg_gen.h, shared with the CPU oracle)         */
+/* ------------------------------------------------------------------ */
+
+/* which_table for shard counting */
+enum { TAB_LI = 0, TAB_ORD = 1, TAB_CUST = 2 };
+
+__device__ inline int
+li_seg(uint64_t /*seed*/, int64_t row, int nseg)
+{
+	int64_t orderkey = (row >> 2) + 1;
+
+	return gg_cdbhash_segment_int8(orderkey, nseg);
+}
+
+__global__ void
+k_count_shard(int64_t row_lo, int64_t n, int nseg, int seg, int which,
+	      uint64_t seed, int64_t sf, unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long c = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t row = row_lo + i;
+		int s;
+
+		if (which == TAB_LI)
+			s = li_seg(seed, row, nseg);
+		else	/* orders / customer: key = row+1 (dense) */
+			s = gg_cdbhash_segment_int8(row + 1, nseg);
+		if (s == seg)
+			c++;
+	}
+	for (int off = 32; off; off >>= 1)
+		c += __shfl_down(c, off, 64);
+	if ((threadIdx.x & 63) == 0)
+		atomicAdd(out_count, c);
+}
+
+hipError_t
+launch_count_shard(hipStream_t s, int64_t row_lo, int64_t n, int nseg,
+		   int seg, int which_table, uint64_t seed, int64_t sf,
+		   unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_count_shard, dim3(grid_for(n)), dim3(THREADS), 0,
+			   s, row_lo, n, nseg, seg, which_table, seed, sf,
+			   out_count);
+	return hipGetLastError();
+}
+
+__global__ void
+k_gen_lineitem(uint64_t seed, int64_t row_lo, int64_t n, int nseg, int seg,
+	       int64_t *__restrict__ orderkey, int64_t *__restrict__ qty,
+	       int64_t *__restrict__ price, int64_t *__restrict__ disc,
+	       int64_t *__restrict__ tax, int32_t *__restrict__ shipdate,
+	       uint8_t *__restrict__ rflag, uint8_t *__restrict__ lstatus,
+	       unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t row = row_lo + i;
+		bool own = (nseg <= 1) || (li_seg(seed, row, nseg) == seg);
+		int64_t idx;
+
+		if (nseg <= 1)
+			idx = i;
+		else
+		{
+			/* wave-aggregated append (Guideline 12) */
+			unsigned long long mask = __ballot(own);
+			int lane = threadIdx.x & 63;
+			unsigned long long before =
+				mask & ((lane == 63) ? ~0ull >> 1
+					: ((1ull << lane) - 1));
+			int nbefore = __popcll(mask & ((1ull << lane) - 1));
+			int total = __popcll(mask);
+			unsigned long long base = 0;
+
+			(void) before;
+			if (lane == __ffsll((long long) mask) - 1 && total)
+				base = atomicAdd(out_count,
+						 (unsigned long long) total);
+			base = __shfl(base, __ffsll((long long) mask) - 1, 64);
+			idx = (int64_t) (base + (unsigned) nbefore);
+		}
+		if (!own)
+			continue;
+		{
+			gg_lineitem_row r;
+
+			gg_gen_lineitem(seed, row, &r);
+			orderkey[idx] = r.l_orderkey;
+			qty[idx] = r.l_quantity_c;
+			price[idx] = r.l_extendedprice_c;
+			disc[idx] = r.l_discount_c;
+			tax[idx] = r.l_tax_c;
+			shipdate[idx] = r.l_shipdate;
+			rflag[idx] = r.l_returnflag;
+			lstatus[idx] = r.l_linestatus;
+		}
+	}
+}
+
+hipError_t
+launch_gen_lineitem(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
+		    int nseg, int seg, int64_t *orderkey, int64_t *qty,
+		    int64_t *price, int64_t *disc, int64_t *tax,
+		    int32_t *shipdate, uint8_t *rflag, uint8_t *lstatus,
+		    unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_gen_lineitem, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, seed, row_lo, n, nseg, seg, orderkey, qty,
+			   price, disc, tax, shipdate, rflag, lstatus,
+			   out_count);
+	return hipGetLastError();
+}
+
+__global__ void
+k_gen_orders(uint64_t seed, int64_t sf, int64_t row_lo, int64_t n, int nseg,
+	     int seg, int64_t *__restrict__ orderkey,
+	     int64_t *__restrict__ custkey, int32_t *__restrict__ orderdate,
+	     int32_t *__restrict__ prio, unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t okey = row_lo + i + 1;
+		bool own = (nseg <= 1) ||
+			(gg_cdbhash_segment_int8(okey, nseg) == seg);
+		int64_t idx = i;
+
+		if (nseg > 1)
+		{
+			unsigned long long mask = __ballot(own);
+			int lane = threadIdx.x & 63;
+			int nbefore = __popcll(mask & ((1ull << lane) - 1));
+			int total = __popcll(mask);
+			int leader = __ffsll((long long) mask) - 1;
+			unsigned long long base = 0;
+
+			if (total && lane == leader)
+				base = atomicAdd(out_count,
+						 (unsigned long long) total);
+			if (total)
+				base = __shfl(base, leader, 64);
+			idx = (int64_t) (base + (unsigned) nbefore);
+		}
+		if (!own)
+			continue;
+		orderkey[idx] = okey;
+		custkey[idx] = gg_o_custkey(seed, okey, sf);
+		orderdate[idx] = gg_o_orderdate(seed, okey);
+		prio[idx] = gg_o_shippriority(seed, okey);
+	}
+}
+
+hipError_t
+launch_gen_orders(hipStream_t s, uint64_t seed, int64_t sf, int64_t row_lo,
+		  int64_t n, int nseg, int seg, int64_t *orderkey,
+		  int64_t *custkey, int32_t *orderdate, int32_t *prio,
+		  unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_gen_orders, dim3(grid_for(n)), dim3(THREADS), 0,
+			   s, seed, sf, row_lo, n, nseg, seg, orderkey,
+			   custkey, orderdate, prio, out_count);
+	return hipGetLastError();
+}
+
+__global__ void
+k_gen_customer(uint64_t seed, int64_t row_lo, int64_t n, int nseg, int seg,
+	       int64_t *__restrict__ custkey, uint8_t *__restrict__ mktseg,
+	       unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t ckey = row_lo + i + 1;
+		bool own = (nseg <= 1) ||
+			(gg_cdbhash_segment_int8(ckey, nseg) == seg);
+		int64_t idx = i;
+
+		if (nseg > 1)
+		{
+			unsigned long long mask = __ballot(own);
+			int lane = threadIdx.x & 63;
+			int nbefore = __popcll(mask & ((1ull << lane) - 1));
+			int total = __popcll(mask);
+			int leader = __ffsll((long long) mask) - 1;
+			unsigned long long base = 0;
+
+			if (total && lane == leader)
+				base = atomicAdd(out_count,
+						 (unsigned long long) total);
+			if (total)
+				base = __shfl(base, leader, 64);
+			idx = (int64_t) (base + (unsigned) nbefore);
+		}
+		if (!own)
+			continue;
+		custkey[idx] = ckey;
+		mktseg[idx] = gg_c_mktsegment(seed, ckey);
+	}
+}
+
+hipError_t
+launch_gen_customer(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
+		    int nseg, int seg, int64_t *custkey, uint8_t *mktseg,
+		    unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_gen_customer, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, seed, row_lo, n, nseg, seg, custkey, mktseg,
+			   out_count);
+	return hipGetLastError();
+}
+
+/* ------------------------------------------------------------------ */
+/* hash join build/probe (open addressing, key 0 = empty)              */
+/* ------------------------------------------------------------------ */
+
+__device__ inline uint64_t
+ht_start(int64_t key, uint64_t nslots)
+{
+	return (uint64_t) gg_hashint8(key) & (nslots - 1);
+}
+
+/* insert key into a set (customer build side, nodeHash.c:905) */
+__global__ void
+k_build_set(const int64_t *__restrict__ keys,
+	    const uint8_t *__restrict__ filter_col, uint8_t filter_val,
+	    int64_t n, unsigned long long *__restrict__ tkeys,
+	    uint64_t nslots)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k;
+		uint64_t pos;
+
+		if (filter_col && filter_col[i] != filter_val)
+			continue;
+		k = keys[i];
+		pos = ht_start(k, nslots);
+		for (;;)
+		{
+			unsigned long long prev =
+				atomicCAS(&tkeys[pos], 0ull,
+					  (unsigned long long) k);
+			if (prev == 0 || prev == (unsigned long long) k)
+				break;
+			pos = (pos + 1) & (nslots - 1);
+		}
+	}
+}
+
+hipError_t
+launch_build_set(hipStream_t s, const int64_t *keys,
+		 const uint8_t *filter_col, uint8_t filter_val, int64_t n,
+		 DeviceHashTable t)
+{
+	hipLaunchKernelGGL(k_build_set, dim3(grid_for(n)), dim3(THREADS), 0,
+			   s, keys, filter_col, filter_val, n, t.keys,
+			   t.nslots);
+	return hipGetLastError();
+}
+
+__device__ inline bool
+ht_contains(const unsigned long long *__restrict__ tkeys, uint64_t nslots,
+	    int64_t key)
+{
+	uint64_t pos = ht_start(key, nslots);
+
+	for (;;)
+	{
+		unsigned long long v = tkeys[pos];
+
+		if (v == (unsigned long long) key)
+			return true;
+		if (v == 0)
+			return false;
+		pos = (pos + 1) & (nslots - 1);
+	}
+}
+
+/* orders: filter date, probe customer set, insert (okey → date|prio)
+ * — the build side of the o⋈l join (nodeHashjoin state machine
+ * HJ_BUILD_HASHTABLE, nodeHash.c:88–176) fused with the c⋈o probe */
+__global__ void
+k_build_orders(const int64_t *__restrict__ okey,
+	       const int64_t *__restrict__ ckey,
+	       const int32_t *__restrict__ odate,
+	       const int32_t *__restrict__ prio, int64_t n, int32_t cutoff,
+	       const unsigned long long *__restrict__ cust_keys,
+	       uint64_t cust_slots, unsigned long long *__restrict__ tkeys,
+	       unsigned long long *__restrict__ tpayload, uint64_t nslots,
+	       unsigned long long *match_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long matches = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		if (odate[i] >= cutoff)	/* qual: o_orderdate < cutoff */
+			continue;
+		if (!ht_contains(cust_keys, cust_slots, ckey[i]))
+			continue;
+		matches++;
+		{
+			int64_t k = okey[i];
+			uint64_t pos = ht_start(k, nslots);
+			unsigned long long pay =
+				(unsigned long long) (uint32_t) odate[i] |
+				((unsigned long long) (uint32_t) prio[i] << 32);
+
+			for (;;)
+			{
+				unsigned long long prev =
+					atomicCAS(&tkeys[pos], 0ull,
+						  (unsigned long long) k);
+				if (prev == 0)
+				{
+					tpayload[pos] = pay;
+					break;
+				}
+				if (prev == (unsigned long long) k)
+					break;	/* PK: cannot happen */
+				pos = (pos + 1) & (nslots - 1);
+			}
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+		matches += __shfl_down(matches, off, 64);
+	if ((threadIdx.x & 63) == 0 && matches)
+		atomicAdd(match_count, matches);
+}
+
+hipError_t
+launch_build_orders(hipStream_t s, const int64_t *okey, const int64_t *ckey,
+		    const int32_t *odate, const int32_t *prio, int64_t n,
+		    int32_t cutoff, DeviceHashTable cust, DeviceHashTable ord,
+		    unsigned long long *match_count)
+{
+	hipLaunchKernelGGL(k_build_orders, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, okey, ckey, odate, prio, n, cutoff,
+			   cust.keys, cust.nslots, ord.keys, ord.payload,
+			   ord.nslots, match_count);
+	return hipGetLastError();
+}
+
+/* count-only variants used to size the tables exactly (two-phase;
+ * ExecChooseHashTableSize analog, nodeHash.c:450) */
+__global__ void
+k_count_filter_u8(const uint8_t *__restrict__ col, uint8_t val, int64_t n,
+		  unsigned long long *out)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long c = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		c += (col[i] == val);
+	for (int off = 32; off; off >>= 1)
+		c += __shfl_down(c, off, 64);
+	if ((threadIdx.x & 63) == 0 && c)
+		atomicAdd(out, c);
+}
+
+__global__ void
+k_count_orders_match(const int64_t *__restrict__ ckey,
+		     const int32_t *__restrict__ odate, int64_t n,
+		     int32_t cutoff,
+		     const unsigned long long *__restrict__ cust_keys,
+		     uint64_t cust_slots, unsigned long long *out)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long c = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		if (odate[i] < cutoff &&
+		    ht_contains(cust_keys, cust_slots, ckey[i]))
+			c++;
+	for (int off = 32; off; off >>= 1)
+		c += __shfl_down(c, off, 64);
+	if ((threadIdx.x & 63) == 0 && c)
+		atomicAdd(out, c);
+}
+
+/* lineitem probe + revenue aggregation into the order slot — the probe
+ * loop (nodeHash.c:1163 ExecScanHashBucket) fused with the group-by
+ * transition (group key l_orderkey ≡ join key, so the matched order's
+ * slot IS the group slot; execHHashagg.c:456 lookup + nodeAgg advance) */
+__global__ __launch_bounds__(THREADS, 2)
+void k_probe_lineitem(const int64_t *__restrict__ okey,
+		      const int32_t *__restrict__ shipdate,
+		      const int64_t *__restrict__ price,
+		      const int64_t *__restrict__ disc, int64_t n,
+		      int32_t cutoff,
+		      const unsigned long long *__restrict__ tkeys,
+		      unsigned long long *__restrict__ trev, uint64_t nslots,
+		      unsigned long long *join_rows)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long joined = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		if (shipdate[i] <= cutoff)	/* qual: l_shipdate > cutoff */
+			continue;
+		{
+			int64_t k = okey[i];
+			uint64_t pos = ht_start(k, nslots);
+
+			for (;;)
+			{
+				unsigned long long v = tkeys[pos];
+
+				if (v == 0)
+					break;	/* miss */
+				if (v == (unsigned long long) k)
+				{
+					unsigned long long rev4 =
+						(unsigned long long)
+						(price[i] * (100 - disc[i]));
+					atomicAdd(&trev[pos], rev4);
+					joined++;
+					break;
+				}
+				pos = (pos + 1) & (nslots - 1);
+			}
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+		joined += __shfl_down(joined, off, 64);
+	if ((threadIdx.x & 63) == 0 && joined)
+		atomicAdd(join_rows, joined);
+}
+
+hipError_t
+launch_probe_lineitem(hipStream_t s, const int64_t *okey,
+		      const int32_t *shipdate, const int64_t *price,
+		      const int64_t *disc, int64_t n, int32_t cutoff,
+		      DeviceHashTable ord, unsigned long long *join_rows)
+{
+	hipLaunchKernelGGL(k_probe_lineitem, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, okey, shipdate, price, disc, n, cutoff,
+			   ord.keys, ord.rev, ord.nslots, join_rows);
+	return hipGetLastError();
+}
+
+/* ------------------------------------------------------------------ */
+/* Q3 result extraction: stats, max, histogram-select, collect         */
+/* ------------------------------------------------------------------ */
+
+__global__ void
+k_q3_stats(const unsigned long long *__restrict__ tkeys,
+	   const unsigned long long *__restrict__ tpayload,
+	   const unsigned long long *__restrict__ trev, uint64_t nslots,
+	   unsigned long long *out4)
+{
+	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+	unsigned long long ng = 0, rev = 0, carry = 0, ck = 0;
+
+	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nslots; i += stride)
+	{
+		unsigned long long k = tkeys[i];
+		unsigned long long r;
+
+		if (!k)
+			continue;
+		r = trev[i];
+		if (!r)
+			continue;	/* order matched but no lines joined */
+		ng++;
+		{
+			unsigned long long old = rev;
+
+			rev += r;
+			carry += (rev < old);
+		}
+		{
+			unsigned long long pay = tpayload[i];
+			int32_t date = (int32_t) (uint32_t) pay;
+			int32_t prio = (int32_t) (uint32_t) (pay >> 32);
+
+			ck += gg_group_hash((int64_t) k, r, 0, date, prio);
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+	{
+		unsigned long long orev = rev;
+
+		ng += __shfl_down(ng, off, 64);
+		rev += __shfl_down(rev, off, 64);
+		carry += __shfl_down(carry, off, 64) + (rev < orev);
+		ck += __shfl_down(ck, off, 64);
+	}
+	if ((threadIdx.x & 63) == 0 && ng)
+	{
+		atomicAdd(&out4[0], ng);
+		{
+			unsigned long long old = atomicAdd(&out4[1], rev);
+
+			if (old + rev < old)
+				atomicAdd(&out4[2], 1ull);
+			atomicAdd(&out4[2], carry);
+		}
+		atomicAdd(&out4[3], ck);
+	}
+}
+
+hipError_t
+launch_q3_stats(hipStream_t s, DeviceHashTable ord, unsigned long long *out4)
+{
+	hipLaunchKernelGGL(k_q3_stats, dim3(grid_for((int64_t) ord.nslots)),
+			   dim3(THREADS), 0, s, ord.keys, ord.payload,
+			   ord.rev, ord.nslots, out4);
+	return hipGetLastError();
+}
+
+__global__ void
+k_q3_maxrev(const unsigned long long *__restrict__ tkeys,
+	    const unsigned long long *__restrict__ trev, uint64_t nslots,
+	    unsigned long long *out_max)
+{
+	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+	unsigned long long m = 0;
+
+	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nslots; i += stride)
+		if (tkeys[i])
+			m = max(m, trev[i]);
+	for (int off = 32; off; off >>= 1)
+		m = max(m, __shfl_down(m, off, 64));
+	if ((threadIdx.x & 63) == 0 && m)
+		atomicMax(out_max, m);
+}
+
+hipError_t
+launch_q3_maxrev(hipStream_t s, DeviceHashTable ord,
+		 unsigned long long *out_max)
+{
+	hipLaunchKernelGGL(k_q3_maxrev, dim3(grid_for((int64_t) ord.nslots)),
+			   dim3(THREADS), 0, s, ord.keys, ord.rev, ord.nslots,
+			   out_max);
+	return hipGetLastError();
+}
+
+__global__ void
+k_q3_hist(const unsigned long long *__restrict__ tkeys,
+	  const unsigned long long *__restrict__ trev, uint64_t nslots,
+	  int shift, unsigned int *__restrict__ hist64k)
+{
+	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+
+	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nslots; i += stride)
+	{
+		unsigned long long r;
+
+		if (!tkeys[i])
+			continue;
+		r = trev[i];
+		if (!r)
+			continue;
+		{
+			unsigned long long bin = r >> shift;
+
+			if (bin > 65535)
+				bin = 65535;
+			atomicAdd(&hist64k[bin], 1u);
+		}
+	}
+}
+
+hipError_t
+launch_q3_hist(hipStream_t s, DeviceHashTable ord, int shift,
+	       uint64_t /*lo_bound*/, unsigned int *hist64k)
+{
+	hipLaunchKernelGGL(k_q3_hist, dim3(grid_for((int64_t) ord.nslots)),
+			   dim3(THREADS), 0, s, ord.keys, ord.rev, ord.nslots,
+			   shift, hist64k);
+	return hipGetLastError();
+}
+
+__global__ void
+k_q3_collect(const unsigned long long *__restrict__ tkeys,
+	     const unsigned long long *__restrict__ tpayload,
+	     const unsigned long long *__restrict__ trev, uint64_t nslots,
+	     unsigned long long threshold, gg_q3_result_row *__restrict__ out,
+	     unsigned long long *out_count, uint64_t cap)
+{
+	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+
+	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nslots; i += stride)
+	{
+		unsigned long long k = tkeys[i], r;
+
+		if (!k)
+			continue;
+		r = trev[i];
+		if (r < threshold || !r)
+			continue;
+		{
+			unsigned long long idx = atomicAdd(out_count, 1ull);
+
+			if (idx < cap)
+			{
+				unsigned long long pay = tpayload[i];
+
+				out[idx].orderkey = (int64_t) k;
+				out[idx].rev_lo = r;
+				out[idx].rev_hi = 0;
+				out[idx].orderdate =
+					(int32_t) (uint32_t) pay;
+				out[idx].shippriority =
+					(int32_t) (uint32_t) (pay >> 32);
+			}
+		}
+	}
+}
+
+hipError_t
+launch_q3_collect(hipStream_t s, DeviceHashTable ord, uint64_t threshold,
+		  gg_q3_result_row *out, unsigned long long *out_count,
+		  uint64_t cap)
+{
+	hipLaunchKernelGGL(k_q3_collect, dim3(grid_for((int64_t) ord.nslots)),
+			   dim3(THREADS), 0, s, ord.keys, ord.payload,
+			   ord.rev, ord.nslots, threshold, out, out_count,
+			   cap);
+	return hipGetLastError();
+}
+
+/* ------------------------------------------------------------------ */
+/* Motion partition (redistribute routing, nodeMotion.c:1600–1636)     */
+/* ------------------------------------------------------------------ */
+
+__global__ void
+k_part_count(const int64_t *__restrict__ key, int64_t n, int nseg,
+	     unsigned long long *__restrict__ counts)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		atomicAdd(&counts[gg_cdbhash_segment_int8(key[i], nseg)],
+			  1ull);
+}
+
+hipError_t
+launch_part_count(hipStream_t s, const int64_t *key, int64_t n, int nseg,
+		  unsigned long long *counts)
+{
+	hipLaunchKernelGGL(k_part_count, dim3(grid_for(n)), dim3(THREADS), 0,
+			   s, key, n, nseg, counts);
+	return hipGetLastError();
+}
+
+__global__ void
+k_part_scatter3(const int64_t *__restrict__ key, int64_t n, int nseg,
+		const int64_t *__restrict__ a, const int64_t *__restrict__ b,
+		const int64_t *__restrict__ c,
+		unsigned long long *__restrict__ offsets,
+		int64_t *__restrict__ oa, int64_t *__restrict__ ob,
+		int64_t *__restrict__ oc)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int d = gg_cdbhash_segment_int8(key[i], nseg);
+		unsigned long long idx = atomicAdd(&offsets[d], 1ull);
+
+		oa[idx] = a[i];
+		if (b)
+			ob[idx] = b[i];
+		if (c)
+			oc[idx] = c[i];
+	}
+}
+
+hipError_t
+launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n, int nseg,
+		     const int64_t *a, const int64_t *b, const int64_t *c,
+		     unsigned long long *offsets, int64_t *oa, int64_t *ob,
+		     int64_t *oc)
+{
+	hipLaunchKernelGGL(k_part_scatter3, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, key, n, nseg, a, b, c, offsets, oa, ob, oc);
+	return hipGetLastError();
+}
+
+/* ------------------------------------------------------------------ */
+/* Q3 multi-GPU exchange helpers (Motion redistribute legs, §8(e))     */
+/* ------------------------------------------------------------------ */
+
+/* orders shard: filter date → compact (ckey, okey, date|prio) triplets */
+__global__ void
+k_orders_filter_compact(const int64_t *__restrict__ okey,
+			const int64_t *__restrict__ ckey,
+			const int32_t *__restrict__ odate,
+			const int32_t *__restrict__ prio, int64_t n,
+			int32_t cutoff, int64_t *__restrict__ out_ckey,
+			int64_t *__restrict__ out_okey,
+			int64_t *__restrict__ out_pay,
+			unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		if (odate[i] >= cutoff)
+			continue;
+		{
+			unsigned long long idx = atomicAdd(out_count, 1ull);
+
+			out_ckey[idx] = ckey[i];
+			out_okey[idx] = okey[i];
+			out_pay[idx] = (int64_t)
+				((unsigned long long) (uint32_t) odate[i] |
+				 ((unsigned long long) (uint32_t) prio[i]
+				  << 32));
+		}
+	}
+}
+
+hipError_t
+launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
+			     const int64_t *ckey, const int32_t *odate,
+			     const int32_t *prio, int64_t n, int32_t cutoff,
+			     int64_t *out_ckey, int64_t *out_okey,
+			     int64_t *out_pay, unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_orders_filter_compact, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, okey, ckey, odate, prio, n,
+			   cutoff, out_ckey, out_okey, out_pay, out_count);
+	return hipGetLastError();
+}
+
+/* received (ckey, okey, pay) rows: probe local customer set → compact
+ * matched (okey, pay) */
+__global__ void
+k_probe_cust_compact(const int64_t *__restrict__ ckey,
+		     const int64_t *__restrict__ okey,
+		     const int64_t *__restrict__ pay, int64_t n,
+		     const unsigned long long *__restrict__ cust_keys,
+		     uint64_t cust_slots, int64_t *__restrict__ out_okey,
+		     int64_t *__restrict__ out_pay,
+		     unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		if (!ht_contains(cust_keys, cust_slots, ckey[i]))
+			continue;
+		{
+			unsigned long long idx = atomicAdd(out_count, 1ull);
+
+			out_okey[idx] = okey[i];
+			out_pay[idx] = pay[i];
+		}
+	}
+}
+
+hipError_t
+launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,
+			  const int64_t *okey, const int64_t *pay, int64_t n,
+			  DeviceHashTable cust, int64_t *out_okey,
+			  int64_t *out_pay, unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_probe_cust_compact, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, ckey, okey, pay, n, cust.keys,
+			   cust.nslots, out_okey, out_pay, out_count);
+	return hipGetLastError();
+}
+
+/* insert (okey, pay) rows into the orders table */
+__global__ void
+k_insert_orders(const int64_t *__restrict__ okey,
+		const int64_t *__restrict__ pay, int64_t n,
+		unsigned long long *__restrict__ tkeys,
+		unsigned long long *__restrict__ tpayload, uint64_t nslots)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k = okey[i];
+		uint64_t pos = ht_start(k, nslots);
+
+		for (;;)
+		{
+			unsigned long long prev =
+				atomicCAS(&tkeys[pos], 0ull,
+					  (unsigned long long) k);
+			if (prev == 0)
+			{
+				tpayload[pos] = (unsigned long long) pay[i];
+				break;
+			}
+			if (prev == (unsigned long long) k)
+				break;
+			pos = (pos + 1) & (nslots - 1);
+		}
+	}
+}
+
+hipError_t
+launch_insert_orders(hipStream_t s, const int64_t *okey, const int64_t *pay,
+		     int64_t n, DeviceHashTable ord)
+{
+	hipLaunchKernelGGL(k_insert_orders, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, okey, pay, n, ord.keys, ord.payload,
+			   ord.nslots);
+	return hipGetLastError();
+}
+
+/* count-helper launchers used by engine_abi.cpp */
+hipError_t
+launch_count_filter_u8(hipStream_t s, const uint8_t *col, uint8_t val,
+		       int64_t n, unsigned long long *out)
+{
+	hipLaunchKernelGGL(k_count_filter_u8, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, col, val, n, out);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_count_orders_match(hipStream_t s, const int64_t *ckey,
+			  const int32_t *odate, int64_t n, int32_t cutoff,
+			  DeviceHashTable cust, unsigned long long *out)
+{
+	hipLaunchKernelGGL(k_count_orders_match, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, ckey, odate, n, cutoff,
+			   cust.keys, cust.nslots, out);
+	return hipGetLastError();
+}
+
+}				/* namespace gg */

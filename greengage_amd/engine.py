@@ -1,0 +1,296 @@
+"""ctypes bindings for the C-ABI engine (include/engine_abi.h).
+
+The engine library is product native code (HIP gfx950 + RCCL); this
+module is plumbing only.  It fails loudly if the library is missing —
+no silent CPU fallback (project rule: GPU tests must run the native
+path).
+"""
+import ctypes
+import datetime
+import os
+import subprocess
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_DIR, "libgreengage_engine.so")
+
+I8 = ctypes.c_uint8
+I32 = ctypes.c_int32
+I64 = ctypes.c_int64
+U64 = ctypes.c_uint64
+
+
+class EngineError(RuntimeError):
+    pass
+
+
+class _Config(ctypes.Structure):
+    _fields_ = [("device", ctypes.c_int), ("n_segments", ctypes.c_int),
+                ("segment_id", ctypes.c_int), ("hbm_limit_bytes", U64)]
+
+
+class _ColumnDesc(ctypes.Structure):
+    _fields_ = [("name", ctypes.c_char_p), ("type", ctypes.c_int),
+                ("host_data", ctypes.c_void_p),
+                ("device_data", ctypes.c_void_p)]
+
+
+class _PipelineDesc(ctypes.Structure):
+    _fields_ = [("kind", ctypes.c_int), ("lineitem", I32), ("orders", I32),
+                ("customer", I32), ("cutoff_date", I32), ("mktsegment", I8),
+                ("limit_k", I64)]
+
+
+class KernelStat(ctypes.Structure):
+    _fields_ = [("name", ctypes.c_char * 48), ("launches", I64),
+                ("total_ms", ctypes.c_double), ("rows_in", I64),
+                ("rows_out", I64), ("hbm_bytes_algorithmic", I64)]
+
+
+COLTYPE = {"int64": 0, "int32": 1, "dec64": 2, "char1": 3}
+PIPE_Q1, PIPE_Q3, PIPE_SUMPRICE = 1, 2, 3
+
+_EPOCH = datetime.date(2000, 1, 1)
+
+
+def pgdate(y, m, d):
+    """PG DateADT encoding (int32 days since 2000-01-01)."""
+    return (datetime.date(y, m, d) - _EPOCH).days
+
+
+def PGDate(iso):
+    y, m, d = map(int, iso.split("-"))
+    return pgdate(y, m, d)
+
+
+def build_library():
+    subprocess.run(["make", "-s", "-C", os.path.join(_DIR, "csrc")],
+                   check=True)
+
+
+def _load():
+    if not os.path.exists(_LIB_PATH):
+        raise EngineError(
+            f"engine library missing: {_LIB_PATH}. Build it with "
+            "`make -C greengage_amd/csrc` (or __graft_entry__.build()). "
+            "There is no CPU fallback.")
+    lib = ctypes.CDLL(_LIB_PATH)
+    lib.gg_engine_last_error.restype = ctypes.c_char_p
+    lib.gg_engine_build_info.restype = ctypes.c_char_p
+    lib.gg_engine_init.argtypes = [ctypes.POINTER(_Config)]
+    lib.gg_engine_register_table.argtypes = [
+        ctypes.c_char_p, ctypes.POINTER(_ColumnDesc), ctypes.c_int, I64,
+        ctypes.POINTER(I32)]
+    lib.gg_engine_register_synth.argtypes = [ctypes.c_char_p, U64, I64,
+                                             ctypes.POINTER(I32)]
+    lib.gg_engine_table_nrows.argtypes = [I32, ctypes.POINTER(I64)]
+    lib.gg_engine_compile_pipeline.argtypes = [
+        ctypes.POINTER(_PipelineDesc), ctypes.POINTER(I32)]
+    lib.gg_engine_execute.argtypes = [I32, ctypes.c_void_p, ctypes.c_size_t,
+                                      ctypes.POINTER(ctypes.c_size_t)]
+    lib.gg_engine_stats.argtypes = [I32, ctypes.POINTER(KernelStat),
+                                    ctypes.c_int, ctypes.POINTER(ctypes.c_int)]
+    lib.gg_engine_comm_id.argtypes = [ctypes.c_void_p]
+    lib.gg_engine_comm_init.argtypes = [ctypes.c_void_p]
+    lib.gg_engine_numeric_str.argtypes = [U64, I64, ctypes.c_int,
+                                          ctypes.c_char_p]
+    lib.gg_engine_avg_str.argtypes = [U64, I64, ctypes.c_int, I64,
+                                      ctypes.c_char_p]
+    return lib
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        _lib = _load()
+    return _lib
+
+
+def _check(rc, what):
+    if rc != 0:
+        raise EngineError(
+            f"{what}: status {rc}: {lib().gg_engine_last_error().decode()}")
+
+
+def _i128(lo, hi):
+    return (hi << 64) | lo
+
+
+class Engine:
+    """One engine instance per process (= per GPU = per segment)."""
+
+    def __init__(self, device=0, n_segments=1, segment_id=0):
+        cfg = _Config(device=device, n_segments=n_segments,
+                      segment_id=segment_id, hbm_limit_bytes=0)
+        _check(lib().gg_engine_init(ctypes.byref(cfg)), "engine_init")
+        self.n_segments = n_segments
+        self.segment_id = segment_id
+
+    def shutdown(self):
+        _check(lib().gg_engine_shutdown(), "engine_shutdown")
+
+    # ---- comm (Motion-equivalent RCCL bootstrap) ----
+    def comm_id(self):
+        buf = ctypes.create_string_buffer(128)
+        _check(lib().gg_engine_comm_id(buf), "comm_id")
+        return buf.raw
+
+    def comm_init(self, id_bytes):
+        buf = ctypes.create_string_buffer(bytes(id_bytes), 128)
+        _check(lib().gg_engine_comm_init(buf), "comm_init")
+
+    # ---- tables ----
+    def register_synth(self, name, seed=42, sf=1):
+        h = I32()
+        _check(lib().gg_engine_register_synth(name.encode(), seed, sf,
+                                              ctypes.byref(h)),
+               f"register_synth({name})")
+        return h.value
+
+    def register_table(self, name, cols, nrows):
+        """cols: list of (name, typename, numpy_array)."""
+        descs = (_ColumnDesc * len(cols))()
+        keep = []
+        for i, (cname, ctype, arr) in enumerate(cols):
+            arr = _ascontig(arr, ctype)
+            keep.append(arr)
+            descs[i].name = cname.encode()
+            descs[i].type = COLTYPE[ctype]
+            descs[i].host_data = arr.ctypes.data_as(ctypes.c_void_p)
+            descs[i].device_data = None
+        h = I32()
+        _check(lib().gg_engine_register_table(name.encode(), descs,
+                                              len(cols), nrows,
+                                              ctypes.byref(h)),
+               f"register_table({name})")
+        return h.value
+
+    def table_nrows(self, h):
+        n = I64()
+        _check(lib().gg_engine_table_nrows(h, ctypes.byref(n)), "nrows")
+        return n.value
+
+    def fetch_column(self, h, col_name, dtype):
+        import numpy as np
+        n = self.table_nrows(h)
+        arr = np.empty(n, dtype=dtype)
+        _check(lib().gg_engine_fetch_column(
+            h, col_name.encode(), arr.ctypes.data_as(ctypes.c_void_p),
+            arr.nbytes), f"fetch_column({col_name})")
+        return arr
+
+    # ---- pipelines ----
+    def compile(self, kind, lineitem=-1, orders=-1, customer=-1,
+                cutoff_date=0, mktsegment=0, limit_k=10):
+        d = _PipelineDesc(kind=kind, lineitem=lineitem, orders=orders,
+                          customer=customer, cutoff_date=cutoff_date,
+                          mktsegment=mktsegment, limit_k=limit_k)
+        h = I32()
+        _check(lib().gg_engine_compile_pipeline(ctypes.byref(d),
+                                                ctypes.byref(h)), "compile")
+        return h.value
+
+    def execute_raw(self, p, arena_bytes):
+        arena = ctypes.create_string_buffer(arena_bytes)
+        written = ctypes.c_size_t()
+        _check(lib().gg_engine_execute(p, arena, arena_bytes,
+                                       ctypes.byref(written)), "execute")
+        return arena.raw[:written.value]
+
+    def stats(self, p):
+        out = (KernelStat * 32)()
+        n = ctypes.c_int()
+        _check(lib().gg_engine_stats(p, out, 32, ctypes.byref(n)), "stats")
+        return [{
+            "name": out[i].name.decode(),
+            "launches": out[i].launches,
+            "total_ms": out[i].total_ms,
+            "rows_in": out[i].rows_in,
+            "rows_out": out[i].rows_out,
+            "hbm_bytes_algorithmic": out[i].hbm_bytes_algorithmic,
+        } for i in range(n.value)]
+
+    # ---- typed result decoding (gg_result.h layouts) ----
+    def execute_q1(self, p):
+        raw = self.execute_raw(p, 8 + 6 * 72)
+        n_groups = int.from_bytes(raw[0:4], "little", signed=True)
+        groups = []
+        off = 8
+        for _ in range(6):
+            f = _unpack_q1_group(raw[off:off + 72])
+            off += 72
+            if f["count"]:
+                groups.append(f)
+        assert len(groups) == n_groups
+        return groups
+
+    def execute_q3(self, p, k=10):
+        raw = self.execute_raw(p, 48 + 32 * max(k, 1))
+        hdr = {
+            "n_out": int.from_bytes(raw[0:8], "little", signed=True),
+            "n_groups": int.from_bytes(raw[8:16], "little", signed=True),
+            "rev_sum4": _i128(int.from_bytes(raw[16:24], "little"),
+                              int.from_bytes(raw[24:32], "little",
+                                             signed=True)),
+            "group_checksum": int.from_bytes(raw[32:40], "little"),
+            "n_join_rows": int.from_bytes(raw[40:48], "little", signed=True),
+        }
+        rows = []
+        off = 48
+        for _ in range(hdr["n_out"]):
+            b = raw[off:off + 32]
+            off += 32
+            rows.append({
+                "orderkey": int.from_bytes(b[0:8], "little", signed=True),
+                "revenue4": _i128(int.from_bytes(b[8:16], "little"),
+                                  int.from_bytes(b[16:24], "little",
+                                                 signed=True)),
+                "orderdate": int.from_bytes(b[24:28], "little", signed=True),
+                "shippriority": int.from_bytes(b[28:32], "little",
+                                               signed=True),
+            })
+        return rows, hdr
+
+    def execute_sumprice(self, p):
+        raw = self.execute_raw(p, 16)
+        return (int.from_bytes(raw[0:8], "little", signed=True),
+                int.from_bytes(raw[8:16], "little", signed=True))
+
+    # ---- numeric display (product-side finalize) ----
+    @staticmethod
+    def numeric_str(val128, scale):
+        buf = ctypes.create_string_buffer(80)
+        lib().gg_engine_numeric_str(val128 & ((1 << 64) - 1), val128 >> 64,
+                                    scale, buf)
+        return buf.value.decode()
+
+    @staticmethod
+    def avg_str(sum128, sum_scale, count):
+        buf = ctypes.create_string_buffer(80)
+        lib().gg_engine_avg_str(sum128 & ((1 << 64) - 1), sum128 >> 64,
+                                sum_scale, count, buf)
+        return buf.value.decode()
+
+
+def _unpack_q1_group(b):
+    return {
+        "count": int.from_bytes(b[0:8], "little", signed=True),
+        "sum_qty_c": int.from_bytes(b[8:16], "little", signed=True),
+        "sum_base_c": int.from_bytes(b[16:24], "little", signed=True),
+        "sum_dcol_c": int.from_bytes(b[24:32], "little", signed=True),
+        "sum_disc4": _i128(int.from_bytes(b[32:40], "little"),
+                           int.from_bytes(b[40:48], "little", signed=True)),
+        "sum_charge6": _i128(int.from_bytes(b[48:56], "little"),
+                             int.from_bytes(b[56:64], "little", signed=True)),
+        "returnflag": chr(b[64]),
+        "linestatus": chr(b[65]),
+    }
+
+
+def _ascontig(arr, ctype):
+    import numpy as np
+    want = {"int64": np.int64, "dec64": np.int64, "int32": np.int32,
+            "char1": np.uint8}[ctype]
+    return np.ascontiguousarray(arr, dtype=want)

@@ -1,0 +1,203 @@
+/*
+ * engine_abi.h — the C-ABI drop-in boundary of the MI355X columnar
+ * query-operator engine (libgreengage_engine.so).
+ *
+ * This is the surface a Greengage segment-side shim binds: the segment
+ * postgres process loads a .so via shared_preload_libraries
+ * (reference: utils/misc/guc.c:3001) whose _PG_init
+ * (utils/fmgr/dfmgr.c:176–280) installs
+ * ExecutorStart_hook/ExecutorRun_hook/ExecutorEnd_hook
+ * (executor/execMain.c:129–132, call sites :334/:1036/:1309; typedefs
+ * include/executor/executor.h:90–105).  The shim walks the PlanState
+ * tree of the QueryDesc, claims recognized
+ * SeqScan→HashJoin→Agg→Sort/Motion-bounded subtrees, and calls this
+ * engine; everything else falls back to standard_ExecutorRun.
+ * INTEGRATION.md shows the shim-side binding.
+ *
+ * Per-entry-point reference interface each call replaces:
+ *   gg_engine_init/shutdown   — _PG_init-time setup (dfmgr.c:279)
+ *   gg_engine_register_table  — the table-AM scan source the claimed
+ *                               SeqScan would read (nodeSeqscan.c:141
+ *                               InitScanRelation / aocs_beginscan
+ *                               aocsam.c:426): caller hands columnar
+ *                               buffers instead of a Relation
+ *   gg_engine_compile_pipeline— ExecutorStart of the claimed subtree
+ *                               (execMain.c:334 standard_ExecutorStart →
+ *                               InitPlan)
+ *   gg_engine_execute         — ExecutorRun/ExecutePlan of the subtree
+ *                               (execMain.c:988/:3218): runs the whole
+ *                               pipeline to completion and materializes
+ *                               result batches into caller-owned memory
+ *                               (slot ownership rule: execScan.c:142,
+ *                               SURVEY §8(b))
+ *   gg_engine_comm_*          — the Motion exchange the subtree's
+ *                               Motion node would perform
+ *                               (nodeMotion.c:1574 doSendTuple /
+ *                               cdbmotion.c:434 SendTuple over
+ *                               ic_udpifc.c), carried on RCCL over xGMI
+ *   gg_engine_stats           — per-node Instrumentation the shim feeds
+ *                               into EXPLAIN ANALYZE (instrument.c:423,
+ *                               explain_gp.c cdbexplain_recvExecStats)
+ *
+ * Error model: status-code returns + gg_engine_last_error() — never
+ * longjmp across HIP frames; the shim converts to ereport(ERROR)
+ * (SURVEY §8(b) "Errors").  Threading: entry points are called from the
+ * backend thread; the engine may use HIP streams/threads internally.
+ */
+#ifndef GG_ENGINE_ABI_H
+#define GG_ENGINE_ABI_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- status codes ---- */
+typedef enum gg_status
+{
+	GG_OK = 0,
+	GG_EINVAL = 1,		/* bad argument / unknown handle */
+	GG_EGPU = 2,		/* HIP runtime failure */
+	GG_ENOMEM = 3,		/* device or host allocation failure */
+	GG_ECOMM = 4,		/* RCCL failure */
+	GG_ENOTSUP = 5,		/* plan shape not supported (shim must
+				 * fall back to standard_ExecutorRun) */
+	GG_ESTATE = 6		/* call sequence violation */
+} gg_status;
+
+/* thread-local message for the last non-OK return */
+const char *gg_engine_last_error(void);
+
+/* ---- engine lifecycle ---- */
+typedef struct gg_engine_config
+{
+	int device;		/* HIP device ordinal (segment's GPU) */
+	int n_segments;		/* gp segment count = world size */
+	int segment_id;		/* this segment's id = rank */
+	uint64_t hbm_limit_bytes;	/* 0 = no cap */
+} gg_engine_config;
+
+gg_status gg_engine_init(const gg_engine_config *cfg);
+gg_status gg_engine_shutdown(void);
+
+/* ---- column/table registry ---- */
+typedef enum gg_coltype
+{
+	GG_COL_INT64 = 0,	/* int8/bigint */
+	GG_COL_INT32 = 1,	/* int4/date (DateADT int32 days) */
+	GG_COL_DEC64_S2 = 2,	/* numeric(15,2) as scaled int64 (cents) */
+	GG_COL_CHAR1 = 3	/* char(1)/dict code as uint8 */
+} gg_coltype;
+
+typedef struct gg_column_desc
+{
+	const char *name;
+	gg_coltype type;
+	const void *host_data;	/* host pointer (engine uploads), or NULL
+				 * if device_data is given */
+	const void *device_data;/* optional: already-resident HBM buffer */
+} gg_column_desc;
+
+typedef int32_t gg_table;
+
+/* Register a columnar table (the claimed SeqScan's source relation).
+ * All rows are visible (read-only analytics path; visimap honored by
+ * the shim at staging time — SURVEY §2 note). */
+gg_status gg_engine_register_table(const char *name,
+				   const gg_column_desc *cols, int ncols,
+				   int64_t nrows, gg_table *out);
+
+/* Engine-side synthetic registration for benchmarks: generates the
+ * named TPC-H table's shard for (segment_id, n_segments) directly in
+ * HBM per include/gg_gen.h (no host staging). */
+gg_status gg_engine_register_synth(const char *table_name, uint64_t seed,
+				   int64_t sf, gg_table *out);
+
+gg_status gg_engine_drop_table(gg_table t);
+gg_status gg_engine_table_nrows(gg_table t, int64_t *out_nrows);
+
+/* Copy a registered column back to host (test/debug surface; the shim
+ * uses result arenas, not this). buf_bytes must equal nrows × elsize. */
+gg_status gg_engine_fetch_column(gg_table t, const char *col_name,
+				 void *host_buf, size_t buf_bytes);
+
+/* ---- pipeline descriptor ----
+ * The shim extracts the claimed subtree into this descriptor.  v1
+ * supports the §8 hot-path shapes; anything else: GG_ENOTSUP.
+ * (PG 9.4 has no CustomScan — SURVEY §8(b) — so the descriptor mirrors
+ * the PlanState subtree the ExecutorRun hook claims.) */
+typedef enum gg_pipeline_kind
+{
+	/* scan(lineitem) → filter(shipdate<=c) → hashagg(rf,ls; Q1 aggs) */
+	GG_PIPE_Q1 = 1,
+	/* scan(customer)⋈scan(orders)⋈scan(lineitem) → hashagg(okey)
+	 * → sort(rev desc, odate) limit k  [+ Motion redistributes] */
+	GG_PIPE_Q3 = 2,
+	/* scan(lineitem) → filter(shipdate<c) → agg(sum(price),count) */
+	GG_PIPE_SUMPRICE = 3
+} gg_pipeline_kind;
+
+typedef struct gg_pipeline_desc
+{
+	gg_pipeline_kind kind;
+	gg_table lineitem;	/* tables by role; -1 if unused */
+	gg_table orders;
+	gg_table customer;
+	int32_t cutoff_date;	/* the qual's date constant (DateADT) */
+	uint8_t mktsegment;	/* Q3: dict code of c_mktsegment literal */
+	int64_t limit_k;	/* Q3: LIMIT bound (nodeSort.c:143) */
+} gg_pipeline_desc;
+
+typedef int32_t gg_pipeline;
+
+gg_status gg_engine_compile_pipeline(const gg_pipeline_desc *desc,
+				     gg_pipeline *out);
+gg_status gg_engine_drop_pipeline(gg_pipeline p);
+
+/* ---- results ----
+ * gg_engine_execute runs the pipeline on this segment's shard(s) and
+ * materializes the (partial or final) result into the caller-owned
+ * arena.  Layout per pipeline kind is defined in gg_result.h (fixed
+ * little-endian structs).  If the pipeline contains the final Motion,
+ * rank 0's arena receives the globally-combined result. */
+gg_status gg_engine_execute(gg_pipeline p, void *out_arena,
+			    size_t arena_bytes, size_t *out_written);
+
+/* ---- Motion-equivalent collectives (RCCL over xGMI) ---- */
+/* Bootstrap: rank 0 creates an opaque 128-byte id the caller
+ * distributes (the shim would ship it via the QD dispatch;
+ * bench/tests ship it via torch TCPStore). */
+gg_status gg_engine_comm_id(void *out_id128);
+gg_status gg_engine_comm_init(const void *id128);
+gg_status gg_engine_comm_destroy(void);
+
+/* ---- instrumentation (EXPLAIN ANALYZE-equivalent) ---- */
+typedef struct gg_kernel_stat
+{
+	char name[48];
+	int64_t launches;
+	double total_ms;	/* HIP-event measured */
+	int64_t rows_in;
+	int64_t rows_out;
+	int64_t hbm_bytes_algorithmic;
+} gg_kernel_stat;
+
+gg_status gg_engine_stats(gg_pipeline p, gg_kernel_stat *out, int cap,
+			  int *out_n);
+
+/* ---- numeric finalization (reference numeric.c display semantics;
+ * product-side restatement, independent of oracle/) ---- */
+void gg_engine_numeric_str(uint64_t lo, int64_t hi, int scale, char *buf64);
+void gg_engine_avg_str(uint64_t sum_lo, int64_t sum_hi, int sum_scale,
+		       int64_t count, char *buf64);
+
+/* build info: "gfx950" etc. — lets callers assert the native path */
+const char *gg_engine_build_info(void);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* GG_ENGINE_ABI_H */
