@@ -75,6 +75,23 @@ GG_GEN_HOSTDEV uint64_t gg_rnd(uint64_t seed, uint64_t tab, uint64_t row,
 	return gg_splitmix64(seed ^ (tab << 48) ^ (slot << 40) ^ row);
 }
 
+/*
+ * Uniform value in [0, n) by multiply-shift on the hash's high word.
+ * Deliberately avoids 64-bit modulo: hipcc (ROCm 7.2, gfx950) miscompiled
+ * `u64 % const` inside the large generator kernel (the magic-multiply
+ * quotient reconstruction dropped high-word bits → garbage in bits 29–31
+ * of the remainder; isolated kernels were fine).  Multiply-shift uses
+ * only 32×32→64 multiplies, is bias-free enough for benchmark data
+ * (n ≤ 2^32, bias < 2^-32), and is identical on host and device.
+ */
+GG_GEN_HOSTDEV uint32_t gg_rnd_range(uint64_t seed, uint64_t tab,
+				     uint64_t row, uint64_t slot, uint32_t n)
+{
+	uint32_t hi = (uint32_t) (gg_rnd(seed, tab, row, slot) >> 32);
+
+	return (uint32_t) (((uint64_t) hi * n) >> 32);
+}
+
 /* ---- table sizes; sf100 = 100 etc. (sf as integer ×1; SF1 => sf=1) ---- */
 GG_GEN_HOSTDEV int64_t gg_n_customers(int64_t sf) { return 150000 * sf; }
 GG_GEN_HOSTDEV int64_t gg_n_orders(int64_t sf)    { return 1500000 * sf; }
@@ -88,18 +105,18 @@ GG_GEN_HOSTDEV int32_t gg_orderdate_hi(void) { return gg_pgdate(1998, 8, 2); }
 GG_GEN_HOSTDEV int32_t gg_o_orderdate(uint64_t seed, int64_t orderkey)
 {
 	const int32_t lo = gg_orderdate_lo();
-	const int32_t n = gg_orderdate_hi() - lo + 1;
+	const uint32_t n = (uint32_t) (gg_orderdate_hi() - lo + 1);
 
-	return lo + (int32_t) (gg_rnd(seed, GG_TAB_ORDERS,
-				      (uint64_t) orderkey, 0) % (uint64_t) n);
+	return lo + (int32_t) gg_rnd_range(seed, GG_TAB_ORDERS,
+					   (uint64_t) orderkey, 0, n);
 }
 
 GG_GEN_HOSTDEV int64_t gg_o_custkey(uint64_t seed, int64_t orderkey,
 				    int64_t sf)
 {
-	return 1 + (int64_t) (gg_rnd(seed, GG_TAB_ORDERS,
-				     (uint64_t) orderkey, 1) %
-			      (uint64_t) gg_n_customers(sf));
+	return 1 + (int64_t) gg_rnd_range(seed, GG_TAB_ORDERS,
+					  (uint64_t) orderkey, 1,
+					  (uint32_t) gg_n_customers(sf));
 }
 
 GG_GEN_HOSTDEV int32_t gg_o_shippriority(uint64_t seed, int64_t orderkey)
@@ -111,8 +128,8 @@ GG_GEN_HOSTDEV int32_t gg_o_shippriority(uint64_t seed, int64_t orderkey)
 /* ---- customer ---- */
 GG_GEN_HOSTDEV uint8_t gg_c_mktsegment(uint64_t seed, int64_t custkey)
 {
-	return (uint8_t) (gg_rnd(seed, GG_TAB_CUSTOMER,
-				 (uint64_t) custkey, 0) % 5);
+	return (uint8_t) gg_rnd_range(seed, GG_TAB_CUSTOMER,
+				      (uint64_t) custkey, 0, 5);
 }
 
 /* ---- lineitem (row i, 0-based; orderkey = i/4 + 1) ---- */
@@ -137,14 +154,23 @@ GG_GEN_HOSTDEV void gg_gen_lineitem(uint64_t seed, int64_t i,
 	const int32_t cur = gg_pgdate(1995, 6, 17);	/* dbgen CURRENTDATE */
 	int32_t ship, receipt;
 
+	/* range mapping via gg_rnd_range (no 64-bit modulo — see its
+	 * comment; the generator freeze vectors in tests/golden pin the
+	 * exact values) */
+	const int32_t m_qty = (int32_t) gg_rnd_range(seed, GG_TAB_LINEITEM, u, 0, 50);
+	const int32_t m_price = (int32_t) gg_rnd_range(seed, GG_TAB_LINEITEM, u, 1, 9910001u);
+	const int32_t m_disc = (int32_t) gg_rnd_range(seed, GG_TAB_LINEITEM, u, 2, 11);
+	const int32_t m_tax = (int32_t) gg_rnd_range(seed, GG_TAB_LINEITEM, u, 3, 9);
+	const int32_t m_ship = (int32_t) gg_rnd_range(seed, GG_TAB_LINEITEM, u, 4, 121);
+	const int32_t m_rcpt = (int32_t) gg_rnd_range(seed, GG_TAB_LINEITEM, u, 5, 30);
+
 	r->l_orderkey = orderkey;
-	r->l_quantity_c = (int64_t) (1 + gg_rnd(seed, GG_TAB_LINEITEM, u, 0) % 50) * 100;
-	r->l_extendedprice_c = (int64_t) (90000 +
-		gg_rnd(seed, GG_TAB_LINEITEM, u, 1) % 9910001ull);
-	r->l_discount_c = (int64_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 2) % 11);
-	r->l_tax_c = (int64_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 3) % 9);
-	ship = odate + 1 + (int32_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 4) % 121);
-	receipt = ship + 1 + (int32_t) (gg_rnd(seed, GG_TAB_LINEITEM, u, 5) % 30);
+	r->l_quantity_c = (int64_t) ((1 + m_qty) * 100);
+	r->l_extendedprice_c = (int64_t) (90000 + m_price);
+	r->l_discount_c = (int64_t) m_disc;
+	r->l_tax_c = (int64_t) m_tax;
+	ship = odate + 1 + m_ship;
+	receipt = ship + 1 + m_rcpt;
 	r->l_shipdate = ship;
 	r->l_linestatus = (ship > cur) ? GG_LS_O : GG_LS_F;
 	if (receipt <= cur)
