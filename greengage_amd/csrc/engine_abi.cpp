@@ -678,7 +678,12 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	Table *od = get_table(p->desc.orders);
 	Table *cu = get_table(p->desc.customer);
 	int nseg = e.cfg.n_segments;
+	int me = e.cfg.segment_id;
 	int64_t k = p->desc.limit_k > 0 ? p->desc.limit_k : 10;
+	/* the Motion-exchange path normally runs for nseg > 1; the env
+	 * knob forces it at nseg == 1 (self-loopback over RCCL) so the
+	 * whole redistribute plumbing is testable on one GPU */
+	bool exch = nseg > 1 || getenv("GG_FORCE_EXCHANGE") != nullptr;
 
 	if (bytes < sizeof(gg_q3_result_hdr) +
 	    (size_t) k * sizeof(gg_q3_result_row))
@@ -697,29 +702,41 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	const int64_t *l_pc = (const int64_t *) li->col("price");
 	const int64_t *l_dc = (const int64_t *) li->col("disc");
 
-	if (!c_ck || !c_ms || !o_ok || !o_ck || !o_dt || !o_pr || !l_ok ||
-	    !l_sd || !l_pc || !l_dc)
+	if (!c_ck || !c_ms || !o_ok || !o_dt || !o_pr || !l_ok || !l_sd ||
+	    !l_pc || !l_dc || (!o_ck && exch) || (!o_ck && !exch))
 		return fail(GG_EINVAL, "Q3: missing column");
 
 	int32_t cutoff = p->desc.cutoff_date;
 	uint8_t segcode = p->desc.mktsegment;
-	unsigned long long *ctr;
+	unsigned long long *ctr =
+		(unsigned long long *) p->sget("ctr", 8);
 
-	GG_TRY(dev_counter(&ctr));
+	if (!ctr)
+		return fail(GG_ENOMEM, "scratch");
 
-	/* 1. customer build side: size then build (nodeHash.c:450/:905) */
+	/* 1. customer build side (nodeHash.c:450 sizing / :905 insert) */
 	DeviceHashTable cust{};
 	{
-		unsigned long long nfil = 0;
 		Timed tm(e.stream);
 
-		GG_HIP(launch_count_filter_u8(e.stream, c_ms, segcode,
-					      cu->nrows, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_TRY(read_counter(ctr, &nfil));
-		cust.nslots = next_pow2(2 * (nfil + 1));
-		GG_HIP(hipMalloc((void **) &cust.keys, cust.nslots * 8));
-		GG_HIP(hipMemset(cust.keys, 0, cust.nslots * 8));
+		if (!p->cust_slots)
+		{
+			unsigned long long nfil = 0;
+
+			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+			GG_HIP(launch_count_filter_u8(e.stream, c_ms, segcode,
+						      cu->nrows, ctr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr, &nfil));
+			p->cust_slots = next_pow2(2 * (nfil + 1));
+		}
+		cust.nslots = p->cust_slots;
+		cust.keys = (unsigned long long *)
+			p->sget("cust.keys", cust.nslots * 8);
+		if (!cust.keys)
+			return fail(GG_ENOMEM, "cust table");
+		GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8,
+				      e.stream));
 		GG_HIP(launch_build_set(e.stream, c_ck, c_ms, segcode,
 					cu->nrows, cust));
 		double ms = tm.stop();
@@ -728,236 +745,230 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.launches++;
 		st.total_ms += ms;
 		st.rows_in += cu->nrows;
-		st.rows_out += (int64_t) nfil;
-		st.hbm_bytes += cu->nrows * 9 * 2;
+		st.hbm_bytes += cu->nrows * 9;
 	}
 
-	/* 2. orders side → orders hash table (single-seg: direct fuse;
-	 * multi-seg: two Motion redistributes over RCCL, §8(e)) */
+	/* 2. orders side → orders hash table */
 	DeviceHashTable ord{};
 	unsigned long long nmatch = 0;
 
-	if (nseg == 1)
+	if (!exch)
 	{
-		GG_HIP(hipMemset(ctr, 0, 8));
-		{
-			Timed tm(e.stream);
+		Timed tm(e.stream);
 
+		if (!p->ord_slots)
+		{
+			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 			GG_HIP(launch_count_orders_match(e.stream, o_ck, o_dt,
 							 od->nrows, cutoff,
 							 cust, ctr));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
-			ord.nslots = next_pow2(2 * (nmatch + 1));
-			GG_HIP(hipMalloc((void **) &ord.keys, ord.nslots * 8));
-			GG_HIP(hipMalloc((void **) &ord.payload, ord.nslots * 8));
-			GG_HIP(hipMalloc((void **) &ord.rev, ord.nslots * 8));
-			GG_HIP(hipMemset(ord.keys, 0, ord.nslots * 8));
-			GG_HIP(hipMemset(ord.rev, 0, ord.nslots * 8));
-			GG_HIP(hipMemset(ctr, 0, 8));
-			GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt,
-						   o_pr, od->nrows, cutoff,
-						   cust, ord, ctr));
-			double ms = tm.stop();
-			KernelStatAcc &st = p->stat("build_orders");
-
-			st.launches++;
-			st.total_ms += ms;
-			st.rows_in += od->nrows;
-			st.rows_out += (int64_t) nmatch;
-			st.hbm_bytes += od->nrows * 24 * 2;
+			p->ord_slots = next_pow2(2 * (nmatch + 1));
 		}
+		ord.nslots = p->ord_slots;
+		ord.keys = (unsigned long long *)
+			p->sget("ord.keys", ord.nslots * 8);
+		ord.payload = (unsigned long long *)
+			p->sget("ord.payload", ord.nslots * 8);
+		ord.rev = (unsigned long long *)
+			p->sget("ord.rev", ord.nslots * 8);
+		if (!ord.keys || !ord.payload || !ord.rev)
+			return fail(GG_ENOMEM, "ord table");
+		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt, o_pr,
+					   od->nrows, cutoff, cust, ord, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nmatch));
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("build_orders");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += od->nrows;
+		st.rows_out += (int64_t) nmatch;
+		st.hbm_bytes += od->nrows * 24;
 	}
 	else
 	{
+		/* Motion redistribute legs:
+		 * leg 1: filtered orders → owner of o_custkey;
+		 * leg 2: customer-joined orders → owner of o_orderkey. */
 		if (!comm_ready())
-			return fail(GG_ESTATE, "multi-segment without comm");
+			return fail(GG_ESTATE,
+				    "exchange path requires gg_engine_comm_init");
 		Timed tm(e.stream);
 
-		/* leg 1: filtered orders → owner of o_custkey */
-		int64_t *f_ck, *f_ok, *f_pay;
+		int64_t *f_ck = (int64_t *) p->sget("f_ck", (od->nrows + 1) * 8);
+		int64_t *f_ok = (int64_t *) p->sget("f_ok", (od->nrows + 1) * 8);
+		int64_t *f_pay = (int64_t *) p->sget("f_pay", (od->nrows + 1) * 8);
+		unsigned long long *dcnt =
+			(unsigned long long *) p->sget("dcnt", (size_t) nseg * 8);
+		unsigned long long *doffs =
+			(unsigned long long *) p->sget("doffs", (size_t) nseg * 8);
+
+		if (!f_ck || !f_ok || !f_pay || !dcnt || !doffs)
+			return fail(GG_ENOMEM, "exchange scratch");
+
 		unsigned long long nfil = 0;
 
-		GG_HIP(hipMalloc((void **) &f_ck, (od->nrows + 1) * 8));
-		GG_HIP(hipMalloc((void **) &f_ok, (od->nrows + 1) * 8));
-		GG_HIP(hipMalloc((void **) &f_pay, (od->nrows + 1) * 8));
-		GG_HIP(hipMemset(ctr, 0, 8));
-		GG_HIP(launch_orders_filter_compact(e.stream, o_ok, o_ck,
-						    o_dt, o_pr, od->nrows,
-						    cutoff, f_ck, f_ok, f_pay,
-						    ctr));
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_orders_filter_compact(e.stream, o_ok, o_ck, o_dt,
+						    o_pr, od->nrows, cutoff,
+						    f_ck, f_ok, f_pay, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nfil));
 
-		/* partition by custkey */
-		unsigned long long *dcnt;
+		/* partition by custkey (doSendTuple routing,
+		 * nodeMotion.c:1600–1636, bit-exact cdbhash) */
 		std::vector<unsigned long long> cnts(nseg), offs(nseg + 1, 0);
 
-		GG_HIP(hipMalloc((void **) &dcnt, nseg * 8));
-		GG_HIP(hipMemset(dcnt, 0, nseg * 8));
+		GG_HIP(hipMemsetAsync(dcnt, 0, (size_t) nseg * 8, e.stream));
 		GG_HIP(launch_part_count(e.stream, f_ck, (int64_t) nfil, nseg,
 					 dcnt));
 		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_HIP(hipMemcpy(cnts.data(), dcnt, nseg * 8,
+		GG_HIP(hipMemcpy(cnts.data(), dcnt, (size_t) nseg * 8,
 				 hipMemcpyDeviceToHost));
 		for (int i = 0; i < nseg; i++)
 			offs[i + 1] = offs[i] + cnts[i];
 
-		int64_t *s_ck, *s_ok, *s_pay;
+		int64_t *s_ck = (int64_t *) p->sget("s_ck", (nfil + 1) * 8);
+		int64_t *s_ok = (int64_t *) p->sget("s_ok", (nfil + 1) * 8);
+		int64_t *s_pay = (int64_t *) p->sget("s_pay", (nfil + 1) * 8);
 
-		GG_HIP(hipMalloc((void **) &s_ck, (nfil + 1) * 8));
-		GG_HIP(hipMalloc((void **) &s_ok, (nfil + 1) * 8));
-		GG_HIP(hipMalloc((void **) &s_pay, (nfil + 1) * 8));
-		{
-			unsigned long long *doffs;
+		if (!s_ck || !s_ok || !s_pay)
+			return fail(GG_ENOMEM, "send scratch");
+		GG_HIP(hipMemcpy(doffs, offs.data(), (size_t) nseg * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(launch_part_scatter3(e.stream, f_ck, (int64_t) nfil,
+					    nseg, f_ck, f_ok, f_pay, doffs,
+					    s_ck, s_ok, s_pay));
+		GG_HIP(hipStreamSynchronize(e.stream));
 
-			GG_HIP(hipMalloc((void **) &doffs, nseg * 8));
-			GG_HIP(hipMemcpy(doffs, offs.data(), nseg * 8,
-					 hipMemcpyHostToDevice));
-			GG_HIP(launch_part_scatter3(e.stream, f_ck,
-						    (int64_t) nfil, nseg,
-						    f_ck, f_ok, f_pay, doffs,
-						    s_ck, s_ok, s_pay));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			(void) hipFree(doffs);
-		}
-		(void) hipFree(f_ck);
-		(void) hipFree(f_ok);
-		(void) hipFree(f_pay);
-
-		/* exchange counts: allgather the per-dest count vector */
+		/* exchange the count vector, derive receive layout */
 		std::vector<unsigned long long> allcnt((size_t) nseg * nseg);
 		{
-			unsigned long long *g;
+			unsigned long long *g = (unsigned long long *)
+				p->sget("cntg", (size_t) (nseg + nseg * nseg) * 8);
 
-			GG_HIP(hipMalloc((void **) &g,
-					 (size_t) (nseg + (size_t) nseg * nseg) * 8));
-			GG_HIP(hipMemcpy(g, cnts.data(), nseg * 8,
+			if (!g)
+				return fail(GG_ENOMEM, "cnt allgather");
+			GG_HIP(hipMemcpy(g, cnts.data(), (size_t) nseg * 8,
 					 hipMemcpyHostToDevice));
 			GG_TRY(comm_allgather_u64(g, g + nseg, nseg));
 			GG_HIP(hipMemcpy(allcnt.data(), g + nseg,
 					 allcnt.size() * 8,
 					 hipMemcpyDeviceToHost));
-			(void) hipFree(g);
 		}
 		std::vector<unsigned long long> rcnts(nseg), roffs(nseg + 1, 0);
 
 		for (int s = 0; s < nseg; s++)
-			rcnts[s] = allcnt[(size_t) s * nseg + e.cfg.segment_id];
+			rcnts[s] = allcnt[(size_t) s * nseg + me];
 		for (int i = 0; i < nseg; i++)
 			roffs[i + 1] = roffs[i] + rcnts[i];
 		uint64_t rtotal = roffs[nseg];
 
-		int64_t *r_ck, *r_ok, *r_pay;
+		int64_t *r_ck = (int64_t *) p->sget("r_ck", (rtotal + 1) * 8);
+		int64_t *r_ok = (int64_t *) p->sget("r_ok", (rtotal + 1) * 8);
+		int64_t *r_pay = (int64_t *) p->sget("r_pay", (rtotal + 1) * 8);
 
-		GG_HIP(hipMalloc((void **) &r_ck, (rtotal + 1) * 8));
-		GG_HIP(hipMalloc((void **) &r_ok, (rtotal + 1) * 8));
-		GG_HIP(hipMalloc((void **) &r_pay, (rtotal + 1) * 8));
+		if (!r_ck || !r_ok || !r_pay)
+			return fail(GG_ENOMEM, "recv scratch");
 		GG_TRY(comm_alltoallv_i64(s_ck, offs.data(), cnts.data(),
 					  r_ck, roffs.data(), rcnts.data()));
 		GG_TRY(comm_alltoallv_i64(s_ok, offs.data(), cnts.data(),
 					  r_ok, roffs.data(), rcnts.data()));
 		GG_TRY(comm_alltoallv_i64(s_pay, offs.data(), cnts.data(),
 					  r_pay, roffs.data(), rcnts.data()));
-		(void) hipFree(s_ck);
-		(void) hipFree(s_ok);
-		(void) hipFree(s_pay);
 
-		/* probe local customers → matched (okey, pay) */
-		int64_t *m_ok, *m_pay;
+		/* probe local customer set → matched (okey, pay) */
 		unsigned long long nm = 0;
+		int64_t *m_ok = (int64_t *) p->sget("m_ok", (rtotal + 1) * 8);
+		int64_t *m_pay = (int64_t *) p->sget("m_pay", (rtotal + 1) * 8);
 
-		GG_HIP(hipMalloc((void **) &m_ok, (rtotal + 1) * 8));
-		GG_HIP(hipMalloc((void **) &m_pay, (rtotal + 1) * 8));
-		GG_HIP(hipMemset(ctr, 0, 8));
+		if (!m_ok || !m_pay)
+			return fail(GG_ENOMEM, "match scratch");
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_probe_cust_compact(e.stream, r_ck, r_ok, r_pay,
 						 (int64_t) rtotal, cust, m_ok,
 						 m_pay, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nm));
-		(void) hipFree(r_ck);
-		(void) hipFree(r_ok);
-		(void) hipFree(r_pay);
 
 		/* leg 2: matched orders → owner of o_orderkey */
 		std::vector<unsigned long long> cnts2(nseg), offs2(nseg + 1, 0);
 
-		GG_HIP(hipMemset(dcnt, 0, nseg * 8));
+		GG_HIP(hipMemsetAsync(dcnt, 0, (size_t) nseg * 8, e.stream));
 		GG_HIP(launch_part_count(e.stream, m_ok, (int64_t) nm, nseg,
 					 dcnt));
 		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_HIP(hipMemcpy(cnts2.data(), dcnt, nseg * 8,
+		GG_HIP(hipMemcpy(cnts2.data(), dcnt, (size_t) nseg * 8,
 				 hipMemcpyDeviceToHost));
-		(void) hipFree(dcnt);
 		for (int i = 0; i < nseg; i++)
 			offs2[i + 1] = offs2[i] + cnts2[i];
 
-		int64_t *s2_ok, *s2_pay;
+		int64_t *s2_ok = (int64_t *) p->sget("s2_ok", (nm + 1) * 8);
+		int64_t *s2_pay = (int64_t *) p->sget("s2_pay", (nm + 1) * 8);
 
-		GG_HIP(hipMalloc((void **) &s2_ok, (nm + 1) * 8));
-		GG_HIP(hipMalloc((void **) &s2_pay, (nm + 1) * 8));
-		{
-			unsigned long long *doffs;
-
-			GG_HIP(hipMalloc((void **) &doffs, nseg * 8));
-			GG_HIP(hipMemcpy(doffs, offs2.data(), nseg * 8,
-					 hipMemcpyHostToDevice));
-			GG_HIP(launch_part_scatter3(e.stream, m_ok,
-						    (int64_t) nm, nseg, m_ok,
-						    m_pay, nullptr, doffs,
-						    s2_ok, s2_pay, nullptr));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			(void) hipFree(doffs);
-		}
-		(void) hipFree(m_ok);
-		(void) hipFree(m_pay);
+		if (!s2_ok || !s2_pay)
+			return fail(GG_ENOMEM, "send2 scratch");
+		GG_HIP(hipMemcpy(doffs, offs2.data(), (size_t) nseg * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(launch_part_scatter3(e.stream, m_ok, (int64_t) nm, nseg,
+					    m_ok, m_pay, nullptr, doffs,
+					    s2_ok, s2_pay, nullptr));
+		GG_HIP(hipStreamSynchronize(e.stream));
 
 		std::vector<unsigned long long> allcnt2((size_t) nseg * nseg);
 		{
-			unsigned long long *g;
+			unsigned long long *g = (unsigned long long *)
+				p->sget("cntg", (size_t) (nseg + nseg * nseg) * 8);
 
-			GG_HIP(hipMalloc((void **) &g,
-					 (size_t) (nseg + (size_t) nseg * nseg) * 8));
-			GG_HIP(hipMemcpy(g, cnts2.data(), nseg * 8,
+			GG_HIP(hipMemcpy(g, cnts2.data(), (size_t) nseg * 8,
 					 hipMemcpyHostToDevice));
 			GG_TRY(comm_allgather_u64(g, g + nseg, nseg));
 			GG_HIP(hipMemcpy(allcnt2.data(), g + nseg,
 					 allcnt2.size() * 8,
 					 hipMemcpyDeviceToHost));
-			(void) hipFree(g);
 		}
 		std::vector<unsigned long long> rcnts2(nseg), roffs2(nseg + 1, 0);
 
 		for (int s = 0; s < nseg; s++)
-			rcnts2[s] = allcnt2[(size_t) s * nseg + e.cfg.segment_id];
+			rcnts2[s] = allcnt2[(size_t) s * nseg + me];
 		for (int i = 0; i < nseg; i++)
 			roffs2[i + 1] = roffs2[i] + rcnts2[i];
 		uint64_t rtotal2 = roffs2[nseg];
 
-		int64_t *r2_ok, *r2_pay;
+		int64_t *r2_ok = (int64_t *) p->sget("r2_ok", (rtotal2 + 1) * 8);
+		int64_t *r2_pay = (int64_t *) p->sget("r2_pay", (rtotal2 + 1) * 8);
 
-		GG_HIP(hipMalloc((void **) &r2_ok, (rtotal2 + 1) * 8));
-		GG_HIP(hipMalloc((void **) &r2_pay, (rtotal2 + 1) * 8));
+		if (!r2_ok || !r2_pay)
+			return fail(GG_ENOMEM, "recv2 scratch");
 		GG_TRY(comm_alltoallv_i64(s2_ok, offs2.data(), cnts2.data(),
 					  r2_ok, roffs2.data(), rcnts2.data()));
 		GG_TRY(comm_alltoallv_i64(s2_pay, offs2.data(), cnts2.data(),
 					  r2_pay, roffs2.data(),
 					  rcnts2.data()));
-		(void) hipFree(s2_ok);
-		(void) hipFree(s2_pay);
 
 		nmatch = rtotal2;
-		ord.nslots = next_pow2(2 * (rtotal2 + 1));
-		GG_HIP(hipMalloc((void **) &ord.keys, ord.nslots * 8));
-		GG_HIP(hipMalloc((void **) &ord.payload, ord.nslots * 8));
-		GG_HIP(hipMalloc((void **) &ord.rev, ord.nslots * 8));
-		GG_HIP(hipMemset(ord.keys, 0, ord.nslots * 8));
-		GG_HIP(hipMemset(ord.rev, 0, ord.nslots * 8));
+		if (!p->ord_slots)
+			p->ord_slots = next_pow2(2 * (rtotal2 + 1));
+		ord.nslots = p->ord_slots;
+		ord.keys = (unsigned long long *)
+			p->sget("ord.keys", ord.nslots * 8);
+		ord.payload = (unsigned long long *)
+			p->sget("ord.payload", ord.nslots * 8);
+		ord.rev = (unsigned long long *)
+			p->sget("ord.rev", ord.nslots * 8);
+		if (!ord.keys || !ord.payload || !ord.rev)
+			return fail(GG_ENOMEM, "ord table");
+		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8, e.stream));
 		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_pay,
 					    (int64_t) rtotal2, ord));
 		GG_HIP(hipStreamSynchronize(e.stream));
-		(void) hipFree(r2_ok);
-		(void) hipFree(r2_pay);
 
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("orders_exchange");
@@ -968,9 +979,11 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.rows_out += (int64_t) nmatch;
 	}
 
-	/* 3. lineitem probe + group aggregation */
-	GG_HIP(hipMemset(ctr, 0, 8));
+	/* 3. lineitem probe + group aggregation (probe fused with the
+	 * group-by transition; group slot ≡ matched-order slot) */
 	unsigned long long njoin = 0;
+
+	GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 	{
 		Timed tm(e.stream);
 
@@ -987,27 +1000,22 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.hbm_bytes += li->nrows * 28;	/* SURVEY §8(d) */
 	}
 
-	/* 4. stats + top-k */
-	unsigned long long *stats4;
-	unsigned long long hstats[4] = {0, 0, 0, 0};
+	/* 4. stats (+max) in one pass, then histogram-select top-k */
+	unsigned long long hstats[5] = {0, 0, 0, 0, 0};
+	unsigned long long *stats5 =
+		(unsigned long long *) p->sget("stats5", 5 * 8);
 
-	GG_HIP(hipMalloc((void **) &stats4, 4 * 8));
-	GG_HIP(hipMemset(stats4, 0, 4 * 8));
+	if (!stats5)
+		return fail(GG_ENOMEM, "stats scratch");
 	{
 		Timed tm(e.stream);
 
-		GG_HIP(launch_q3_stats(e.stream, ord, stats4));
+		GG_HIP(hipMemsetAsync(stats5, 0, 5 * 8, e.stream));
+		GG_HIP(launch_q3_stats(e.stream, ord, stats5));
 		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_HIP(hipMemcpy(hstats, stats4, 32, hipMemcpyDeviceToHost));
+		GG_HIP(hipMemcpy(hstats, stats5, 40, hipMemcpyDeviceToHost));
 
-		/* local top-k: histogram select then exact host sort */
-		unsigned long long maxrev = 0;
-
-		GG_HIP(hipMemset(stats4, 0, 8));
-		GG_HIP(launch_q3_maxrev(e.stream, ord, stats4));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_HIP(hipMemcpy(&maxrev, stats4, 8, hipMemcpyDeviceToHost));
-
+		unsigned long long maxrev = hstats[4];
 		std::vector<gg_q3_result_row> cand;
 
 		if (maxrev > 0)
@@ -1016,16 +1024,17 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 			while ((maxrev >> shift) > 65535)
 				shift++;
-			unsigned int *dhist;
+			unsigned int *dhist =
+				(unsigned int *) p->sget("hist", 65536 * 4);
 			std::vector<unsigned int> hist(65536);
 
-			GG_HIP(hipMalloc((void **) &dhist, 65536 * 4));
-			GG_HIP(hipMemset(dhist, 0, 65536 * 4));
+			if (!dhist)
+				return fail(GG_ENOMEM, "hist scratch");
+			GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4, e.stream));
 			GG_HIP(launch_q3_hist(e.stream, ord, shift, 0, dhist));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_HIP(hipMemcpy(hist.data(), dhist, 65536 * 4,
 					 hipMemcpyDeviceToHost));
-			(void) hipFree(dhist);
 
 			uint64_t cum = 0, thr_bin = 0;
 
@@ -1040,12 +1049,13 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			}
 			uint64_t threshold = thr_bin << shift;
 			uint64_t cap = cum + 65536;	/* bin-granule slack */
-			gg_q3_result_row *dout;
+			gg_q3_result_row *dout = (gg_q3_result_row *)
+				p->sget("cand", cap * sizeof(gg_q3_result_row));
 			unsigned long long ncand = 0;
 
-			GG_HIP(hipMalloc((void **) &dout,
-					 cap * sizeof(gg_q3_result_row)));
-			GG_HIP(hipMemset(ctr, 0, 8));
+			if (!dout)
+				return fail(GG_ENOMEM, "cand scratch");
+			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 			GG_HIP(launch_q3_collect(e.stream, ord,
 						 threshold ? threshold : 1,
 						 dout, ctr, cap));
@@ -1060,7 +1070,6 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				GG_HIP(hipMemcpy(cand.data(), dout,
 						 ncand * sizeof(gg_q3_result_row),
 						 hipMemcpyDeviceToHost));
-			(void) hipFree(dout);
 			std::sort(cand.begin(), cand.end(), Q3TopkCmp());
 			if ((int64_t) cand.size() > k)
 				cand.resize(k);
@@ -1074,8 +1083,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.rows_in += (int64_t) hstats[0];
 		st.rows_out += (int64_t) cand.size();
 
-		/* 5. global combine (gather of partial states + candidate
-		 * rows; the reference's Gather Motion to the QD) */
+		/* 5. global combine (the Gather Motion to the QD:
+		 * nodeMotion gather + final-stage combine) */
 		u128 revsum = ((u128) hstats[1]) | ((u128) hstats[2] << 64);
 		uint64_t ngroups = hstats[0];
 		uint64_t checksum = hstats[3];
@@ -1083,26 +1092,22 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 		if (nseg > 1)
 		{
-			/* stats: 4 words + k rows (32B each) per rank */
-			size_t words = 4 + (size_t) k * 4;
-			std::vector<unsigned long long> lbuf(words, 0);
+			size_t per = 5 + (size_t) k * 4;
+			std::vector<unsigned long long> lbuf(per, 0);
 
 			lbuf[0] = ngroups;
 			lbuf[1] = (uint64_t) revsum;
 			lbuf[2] = (uint64_t) (revsum >> 64);
 			lbuf[3] = njoin_g;
-			/* checksum folded into rows? keep separate word:
-			 * extend buffer */
-			lbuf.resize(words + 1);
-			lbuf[words] = checksum;
+			lbuf[4] = checksum;
 			for (size_t i = 0; i < cand.size(); i++)
-				std::memcpy(&lbuf[4 + i * 4], &cand[i], 32);
+				std::memcpy(&lbuf[5 + i * 4], &cand[i], 32);
 
-			size_t per = lbuf.size();
-			unsigned long long *g;
+			unsigned long long *g = (unsigned long long *)
+				p->sget("gatherg", (per + per * (size_t) nseg) * 8);
 
-			GG_HIP(hipMalloc((void **) &g,
-					 (per + per * (size_t) nseg) * 8));
+			if (!g)
+				return fail(GG_ENOMEM, "gather scratch");
 			GG_HIP(hipMemcpy(g, lbuf.data(), per * 8,
 					 hipMemcpyHostToDevice));
 			GG_TRY(comm_allgather_u64(g, g + per, per));
@@ -1110,7 +1115,6 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 			GG_HIP(hipMemcpy(all.data(), g + per, all.size() * 8,
 					 hipMemcpyDeviceToHost));
-			(void) hipFree(g);
 
 			ngroups = 0;
 			revsum = 0;
@@ -1119,17 +1123,18 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			cand.clear();
 			for (int r = 0; r < nseg; r++)
 			{
-				const unsigned long long *v = &all[per * (size_t) r];
+				const unsigned long long *v =
+					&all[per * (size_t) r];
 
 				ngroups += v[0];
 				revsum += ((u128) v[1]) | ((u128) v[2] << 64);
 				njoin_g += v[3];
-				checksum += v[per - 1];
+				checksum += v[4];
 				for (int64_t i = 0; i < k; i++)
 				{
 					gg_q3_result_row row;
 
-					std::memcpy(&row, &v[4 + i * 4], 32);
+					std::memcpy(&row, &v[5 + i * 4], 32);
 					if (row.rev_lo || row.rev_hi)
 						cand.push_back(row);
 				}
@@ -1149,15 +1154,9 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		hdr->n_join_rows = (int64_t) njoin_g;
 		std::memcpy(hdr + 1, cand.data(),
 			    cand.size() * sizeof(gg_q3_result_row));
-		*written = sizeof(*hdr) + cand.size() * sizeof(gg_q3_result_row);
+		*written = sizeof(*hdr) +
+			cand.size() * sizeof(gg_q3_result_row);
 	}
-
-	(void) hipFree(stats4);
-	(void) hipFree(ctr);
-	(void) hipFree(cust.keys);
-	(void) hipFree(ord.keys);
-	(void) hipFree(ord.payload);
-	(void) hipFree(ord.rev);
 	return GG_OK;
 }
 

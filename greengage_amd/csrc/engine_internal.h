@@ -92,6 +92,16 @@ struct Pipeline
 	gg_pipeline_desc desc;
 	std::vector<KernelStatAcc> stats;
 
+	/* named device scratch reused across executes (hash tables,
+	 * exchange buffers): avoids per-execute hipMalloc/hipFree of
+	 * hundreds of MB (ExecHashTableCreate is likewise once per
+	 * rescan in the reference, nodeHash.c:270) */
+	std::vector<std::pair<std::string, std::pair<void *, size_t>>> scratch;
+	/* table sizings resolved on first execute (data is immutable
+	 * per pipeline), skipping the count passes afterwards */
+	uint64_t cust_slots = 0;
+	uint64_t ord_slots = 0;
+
 	KernelStatAcc &stat(const char *name)
 	{
 		for (auto &s : stats)
@@ -100,6 +110,34 @@ struct Pipeline
 		stats.push_back({});
 		stats.back().name = name;
 		return stats.back();
+	}
+
+	void *sget(const char *name, size_t bytes)
+	{
+		for (auto &kv : scratch)
+			if (kv.first == name)
+			{
+				if (kv.second.second >= bytes)
+					return kv.second.first;
+				(void) hipFree(kv.second.first);
+				kv.second.first = nullptr;
+				if (hipMalloc(&kv.second.first, bytes) != hipSuccess)
+					return nullptr;
+				kv.second.second = bytes;
+				return kv.second.first;
+			}
+		void *p = nullptr;
+
+		if (hipMalloc(&p, bytes ? bytes : 1) != hipSuccess)
+			return nullptr;
+		scratch.push_back({name, {p, bytes}});
+		return p;
+	}
+
+	~Pipeline()
+	{
+		for (auto &kv : scratch)
+			(void) hipFree(kv.second.first);
 	}
 };
 

@@ -672,14 +672,16 @@ launch_probe_lineitem(hipStream_t s, const int64_t *okey,
 /* Q3 result extraction: stats, max, histogram-select, collect         */
 /* ------------------------------------------------------------------ */
 
+/* one pass: group count, int128 revenue sum, checksum AND max revenue
+ * (out5 = {ng, rev_lo, rev_carry, checksum, max_rev}) */
 __global__ void
 k_q3_stats(const unsigned long long *__restrict__ tkeys,
 	   const unsigned long long *__restrict__ tpayload,
 	   const unsigned long long *__restrict__ trev, uint64_t nslots,
-	   unsigned long long *out4)
+	   unsigned long long *out5)
 {
 	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
-	unsigned long long ng = 0, rev = 0, carry = 0, ck = 0;
+	unsigned long long ng = 0, rev = 0, carry = 0, ck = 0, mx = 0;
 
 	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < nslots; i += stride)
@@ -693,6 +695,7 @@ k_q3_stats(const unsigned long long *__restrict__ tkeys,
 		if (!r)
 			continue;	/* order matched but no lines joined */
 		ng++;
+		mx = max(mx, r);
 		{
 			unsigned long long old = rev;
 
@@ -715,27 +718,29 @@ k_q3_stats(const unsigned long long *__restrict__ tkeys,
 		rev += __shfl_down(rev, off, 64);
 		carry += __shfl_down(carry, off, 64) + (rev < orev);
 		ck += __shfl_down(ck, off, 64);
+		mx = max(mx, __shfl_down(mx, off, 64));
 	}
 	if ((threadIdx.x & 63) == 0 && ng)
 	{
-		atomicAdd(&out4[0], ng);
+		atomicAdd(&out5[0], ng);
 		{
-			unsigned long long old = atomicAdd(&out4[1], rev);
+			unsigned long long old = atomicAdd(&out5[1], rev);
 
 			if (old + rev < old)
-				atomicAdd(&out4[2], 1ull);
-			atomicAdd(&out4[2], carry);
+				atomicAdd(&out5[2], 1ull);
+			atomicAdd(&out5[2], carry);
 		}
-		atomicAdd(&out4[3], ck);
+		atomicAdd(&out5[3], ck);
+		atomicMax(&out5[4], mx);
 	}
 }
 
 hipError_t
-launch_q3_stats(hipStream_t s, DeviceHashTable ord, unsigned long long *out4)
+launch_q3_stats(hipStream_t s, DeviceHashTable ord, unsigned long long *out5)
 {
 	hipLaunchKernelGGL(k_q3_stats, dim3(grid_for((int64_t) ord.nslots)),
 			   dim3(THREADS), 0, s, ord.keys, ord.payload,
-			   ord.rev, ord.nslots, out4);
+			   ord.rev, ord.nslots, out5);
 	return hipGetLastError();
 }
 
