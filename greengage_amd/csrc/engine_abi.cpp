@@ -731,11 +731,17 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			p->cust_slots = next_pow2(2 * (nfil + 1));
 		}
 		cust.nslots = p->cust_slots;
+		cust.bloom_words = cust.nslots / 8 < 1024
+			? 1024 : cust.nslots / 8;
 		cust.keys = (unsigned long long *)
 			p->sget("cust.keys", cust.nslots * 8);
-		if (!cust.keys)
+		cust.bloom = (unsigned long long *)
+			p->sget("cust.bloom", cust.bloom_words * 8);
+		if (!cust.keys || !cust.bloom)
 			return fail(GG_ENOMEM, "cust table");
 		GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8,
+				      e.stream));
+		GG_HIP(hipMemsetAsync(cust.bloom, 0, cust.bloom_words * 8,
 				      e.stream));
 		GG_HIP(launch_build_set(e.stream, c_ck, c_ms, segcode,
 					cu->nrows, cust));
@@ -767,16 +773,22 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			p->ord_slots = next_pow2(2 * (nmatch + 1));
 		}
 		ord.nslots = p->ord_slots;
+		ord.bloom_words = ord.nslots / 8 < 1024
+			? 1024 : ord.nslots / 8;
 		ord.keys = (unsigned long long *)
 			p->sget("ord.keys", ord.nslots * 8);
 		ord.payload = (unsigned long long *)
 			p->sget("ord.payload", ord.nslots * 8);
 		ord.rev = (unsigned long long *)
 			p->sget("ord.rev", ord.nslots * 8);
-		if (!ord.keys || !ord.payload || !ord.rev)
+		ord.bloom = (unsigned long long *)
+			p->sget("ord.bloom", ord.bloom_words * 8);
+		if (!ord.keys || !ord.payload || !ord.rev || !ord.bloom)
 			return fail(GG_ENOMEM, "ord table");
 		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
 		GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
+				      e.stream));
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt, o_pr,
 					   od->nrows, cutoff, cust, ord, ctr));
@@ -956,16 +968,22 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		if (!p->ord_slots)
 			p->ord_slots = next_pow2(2 * (rtotal2 + 1));
 		ord.nslots = p->ord_slots;
+		ord.bloom_words = ord.nslots / 8 < 1024
+			? 1024 : ord.nslots / 8;
 		ord.keys = (unsigned long long *)
 			p->sget("ord.keys", ord.nslots * 8);
 		ord.payload = (unsigned long long *)
 			p->sget("ord.payload", ord.nslots * 8);
 		ord.rev = (unsigned long long *)
 			p->sget("ord.rev", ord.nslots * 8);
-		if (!ord.keys || !ord.payload || !ord.rev)
+		ord.bloom = (unsigned long long *)
+			p->sget("ord.bloom", ord.bloom_words * 8);
+		if (!ord.keys || !ord.payload || !ord.rev || !ord.bloom)
 			return fail(GG_ENOMEM, "ord table");
 		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
 		GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
+				      e.stream));
 		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_pay,
 					    (int64_t) rtotal2, ord));
 		GG_HIP(hipStreamSynchronize(e.stream));

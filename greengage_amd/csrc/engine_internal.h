@@ -45,13 +45,18 @@ struct SumPriceAcc
 	unsigned long long count;
 };
 
-/* open-addressing hash table (SoA), power-of-two slots, key 0 = empty */
+/* open-addressing hash table (SoA), power-of-two slots, key 0 = empty.
+ * bloom: blocked Bloom filter (one u64 word per key, 2 bits set) probed
+ * before the slot array — the GPU analog of the reference's
+ * bloom-tagged hash buckets (execHHashagg.c:481 BLOOMVAL). */
 struct DeviceHashTable
 {
 	unsigned long long *keys = nullptr;	/* build keys (int64 > 0) */
 	unsigned long long *payload = nullptr;	/* (date u32) | (prio u64<<32) */
 	unsigned long long *rev = nullptr;	/* Q3 group revenue, scale 4 */
 	uint64_t nslots = 0;			/* power of two */
+	unsigned long long *bloom = nullptr;	/* pow2 words; may be null */
+	uint64_t bloom_words = 0;
 };
 
 struct Table

@@ -440,12 +440,43 @@ ht_start(int64_t key, uint64_t nslots)
 	return (uint64_t) gg_hashint8(key) & (nslots - 1);
 }
 
+/* blocked Bloom: one word per key, two bits derived from a remix of the
+ * same 32-bit hash; one memory access per membership test */
+__device__ inline unsigned long long
+bloom_mask(uint32_t h, uint64_t words, uint64_t *word_idx)
+{
+	uint32_t h2 = h * 0x9E3779B1u;
+
+	*word_idx = (uint64_t) (h >> 6) & (words - 1);
+	return (1ull << (h2 & 63)) | (1ull << ((h2 >> 6) & 63));
+}
+
+__device__ inline void
+bloom_insert(unsigned long long *bloom, uint64_t words, int64_t key)
+{
+	uint64_t w;
+	unsigned long long m = bloom_mask(gg_hashint8(key), words, &w);
+
+	atomicOr(&bloom[w], m);
+}
+
+__device__ inline bool
+bloom_maybe(const unsigned long long *__restrict__ bloom, uint64_t words,
+	    uint32_t h)
+{
+	uint64_t w;
+	unsigned long long m = bloom_mask(h, words, &w);
+
+	return (bloom[w] & m) == m;
+}
+
 /* insert key into a set (customer build side, nodeHash.c:905) */
 __global__ void
 k_build_set(const int64_t *__restrict__ keys,
 	    const uint8_t *__restrict__ filter_col, uint8_t filter_val,
 	    int64_t n, unsigned long long *__restrict__ tkeys,
-	    uint64_t nslots)
+	    uint64_t nslots, unsigned long long *__restrict__ bloom,
+	    uint64_t bwords)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
@@ -458,6 +489,8 @@ k_build_set(const int64_t *__restrict__ keys,
 		if (filter_col && filter_col[i] != filter_val)
 			continue;
 		k = keys[i];
+		if (bloom)
+			bloom_insert(bloom, bwords, k);
 		pos = ht_start(k, nslots);
 		for (;;)
 		{
@@ -478,8 +511,33 @@ launch_build_set(hipStream_t s, const int64_t *keys,
 {
 	hipLaunchKernelGGL(k_build_set, dim3(grid_for(n)), dim3(THREADS), 0,
 			   s, keys, filter_col, filter_val, n, t.keys,
-			   t.nslots);
+			   t.nslots, t.bloom, t.bloom_words);
 	return hipGetLastError();
+}
+
+__device__ inline bool
+ht_contains_b(const unsigned long long *__restrict__ tkeys, uint64_t nslots,
+	      const unsigned long long *__restrict__ bloom, uint64_t bwords,
+	      int64_t key)
+{
+	uint32_t h = gg_hashint8(key);
+
+	if (bloom && !bloom_maybe(bloom, bwords, h))
+		return false;
+	{
+		uint64_t pos = (uint64_t) h & (nslots - 1);
+
+		for (;;)
+		{
+			unsigned long long v = tkeys[pos];
+
+			if (v == (unsigned long long) key)
+				return true;
+			if (v == 0)
+				return false;
+			pos = (pos + 1) & (nslots - 1);
+		}
+	}
 }
 
 __device__ inline bool
@@ -509,8 +567,11 @@ k_build_orders(const int64_t *__restrict__ okey,
 	       const int32_t *__restrict__ odate,
 	       const int32_t *__restrict__ prio, int64_t n, int32_t cutoff,
 	       const unsigned long long *__restrict__ cust_keys,
-	       uint64_t cust_slots, unsigned long long *__restrict__ tkeys,
+	       uint64_t cust_slots,
+	       const unsigned long long *__restrict__ cust_bloom,
+	       uint64_t cust_bwords, unsigned long long *__restrict__ tkeys,
 	       unsigned long long *__restrict__ tpayload, uint64_t nslots,
+	       unsigned long long *__restrict__ bloom, uint64_t bwords,
 	       unsigned long long *match_count)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
@@ -521,12 +582,16 @@ k_build_orders(const int64_t *__restrict__ okey,
 	{
 		if (odate[i] >= cutoff)	/* qual: o_orderdate < cutoff */
 			continue;
-		if (!ht_contains(cust_keys, cust_slots, ckey[i]))
+		if (!ht_contains_b(cust_keys, cust_slots, cust_bloom,
+				   cust_bwords, ckey[i]))
 			continue;
 		matches++;
 		{
 			int64_t k = okey[i];
 			uint64_t pos = ht_start(k, nslots);
+
+			if (bloom)
+				bloom_insert(bloom, bwords, k);
 			unsigned long long pay =
 				(unsigned long long) (uint32_t) odate[i] |
 				((unsigned long long) (uint32_t) prio[i] << 32);
@@ -561,8 +626,10 @@ launch_build_orders(hipStream_t s, const int64_t *okey, const int64_t *ckey,
 {
 	hipLaunchKernelGGL(k_build_orders, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, okey, ckey, odate, prio, n, cutoff,
-			   cust.keys, cust.nslots, ord.keys, ord.payload,
-			   ord.nslots, match_count);
+			   cust.keys, cust.nslots, cust.bloom,
+			   cust.bloom_words, ord.keys, ord.payload,
+			   ord.nslots, ord.bloom, ord.bloom_words,
+			   match_count);
 	return hipGetLastError();
 }
 
@@ -589,7 +656,9 @@ k_count_orders_match(const int64_t *__restrict__ ckey,
 		     const int32_t *__restrict__ odate, int64_t n,
 		     int32_t cutoff,
 		     const unsigned long long *__restrict__ cust_keys,
-		     uint64_t cust_slots, unsigned long long *out)
+		     uint64_t cust_slots,
+		     const unsigned long long *__restrict__ cust_bloom,
+		     uint64_t cust_bwords, unsigned long long *out)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long c = 0;
@@ -597,7 +666,8 @@ k_count_orders_match(const int64_t *__restrict__ ckey,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 		if (odate[i] < cutoff &&
-		    ht_contains(cust_keys, cust_slots, ckey[i]))
+		    ht_contains_b(cust_keys, cust_slots, cust_bloom,
+				  cust_bwords, ckey[i]))
 			c++;
 	for (int off = 32; off; off >>= 1)
 		c += __shfl_down(c, off, 64);
@@ -617,7 +687,8 @@ void k_probe_lineitem(const int64_t *__restrict__ okey,
 		      int32_t cutoff,
 		      const unsigned long long *__restrict__ tkeys,
 		      unsigned long long *__restrict__ trev, uint64_t nslots,
-		      unsigned long long *join_rows)
+		      const unsigned long long *__restrict__ bloom,
+		      uint64_t bwords, unsigned long long *join_rows)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long joined = 0;
@@ -629,8 +700,12 @@ void k_probe_lineitem(const int64_t *__restrict__ okey,
 			continue;
 		{
 			int64_t k = okey[i];
-			uint64_t pos = ht_start(k, nslots);
+			uint32_t h = gg_hashint8(k);
+			uint64_t pos;
 
+			if (bloom && !bloom_maybe(bloom, bwords, h))
+				continue;
+			pos = (uint64_t) h & (nslots - 1);
 			for (;;)
 			{
 				unsigned long long v = tkeys[pos];
@@ -664,7 +739,8 @@ launch_probe_lineitem(hipStream_t s, const int64_t *okey,
 {
 	hipLaunchKernelGGL(k_probe_lineitem, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, okey, shipdate, price, disc, n, cutoff,
-			   ord.keys, ord.rev, ord.nslots, join_rows);
+			   ord.keys, ord.rev, ord.nslots, ord.bloom,
+			   ord.bloom_words, join_rows);
 	return hipGetLastError();
 }
 
@@ -974,7 +1050,9 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 		     const int64_t *__restrict__ okey,
 		     const int64_t *__restrict__ pay, int64_t n,
 		     const unsigned long long *__restrict__ cust_keys,
-		     uint64_t cust_slots, int64_t *__restrict__ out_okey,
+		     uint64_t cust_slots,
+		     const unsigned long long *__restrict__ cust_bloom,
+		     uint64_t cust_bwords, int64_t *__restrict__ out_okey,
 		     int64_t *__restrict__ out_pay,
 		     unsigned long long *out_count)
 {
@@ -983,7 +1061,8 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		if (!ht_contains(cust_keys, cust_slots, ckey[i]))
+		if (!ht_contains_b(cust_keys, cust_slots, cust_bloom,
+				   cust_bwords, ckey[i]))
 			continue;
 		{
 			unsigned long long idx = atomicAdd(out_count, 1ull);
@@ -1002,7 +1081,8 @@ launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,
 {
 	hipLaunchKernelGGL(k_probe_cust_compact, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, ckey, okey, pay, n, cust.keys,
-			   cust.nslots, out_okey, out_pay, out_count);
+			   cust.nslots, cust.bloom, cust.bloom_words,
+			   out_okey, out_pay, out_count);
 	return hipGetLastError();
 }
 
@@ -1011,7 +1091,8 @@ __global__ void
 k_insert_orders(const int64_t *__restrict__ okey,
 		const int64_t *__restrict__ pay, int64_t n,
 		unsigned long long *__restrict__ tkeys,
-		unsigned long long *__restrict__ tpayload, uint64_t nslots)
+		unsigned long long *__restrict__ tpayload, uint64_t nslots,
+		unsigned long long *__restrict__ bloom, uint64_t bwords)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
@@ -1021,6 +1102,8 @@ k_insert_orders(const int64_t *__restrict__ okey,
 		int64_t k = okey[i];
 		uint64_t pos = ht_start(k, nslots);
 
+		if (bloom)
+			bloom_insert(bloom, bwords, k);
 		for (;;)
 		{
 			unsigned long long prev =
@@ -1044,7 +1127,7 @@ launch_insert_orders(hipStream_t s, const int64_t *okey, const int64_t *pay,
 {
 	hipLaunchKernelGGL(k_insert_orders, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, okey, pay, n, ord.keys, ord.payload,
-			   ord.nslots);
+			   ord.nslots, ord.bloom, ord.bloom_words);
 	return hipGetLastError();
 }
 
@@ -1065,7 +1148,8 @@ launch_count_orders_match(hipStream_t s, const int64_t *ckey,
 {
 	hipLaunchKernelGGL(k_count_orders_match, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, ckey, odate, n, cutoff,
-			   cust.keys, cust.nslots, out);
+			   cust.keys, cust.nslots, cust.bloom,
+			   cust.bloom_words, out);
 	return hipGetLastError();
 }
 
