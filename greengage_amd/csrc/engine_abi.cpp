@@ -320,6 +320,41 @@ gg_engine_register_synth(const char *table_name, uint64_t seed, int64_t sf,
 		which = 2;
 		total = gg_n_customers(sf);
 	}
+	else if (nm == "supplier")
+	{
+		which = 3;	/* keys dense: shard by suppkey like cust */
+		total = gg_n_suppliers(sf);
+	}
+	else if (nm == "nation")
+	{
+		/* tiny dim (25 rows), replicated on every segment — the
+		 * reference broadcasts/replicates small dims */
+		Table *t = new Table();
+		int32_t natk[GG_NNATIONS], regk[GG_NNATIONS];
+
+		for (int i = 0; i < GG_NNATIONS; i++)
+		{
+			natk[i] = i;
+			regk[i] = gg_nation_region(i);
+		}
+		t->name = nm;
+		t->nrows = GG_NNATIONS;
+		{
+			gg_status s2;
+
+			if ((s2 = add_col(t, "nationkey", GG_COL_INT32,
+					  GG_NNATIONS, natk, nullptr)) != GG_OK ||
+			    (s2 = add_col(t, "regionkey", GG_COL_INT32,
+					  GG_NNATIONS, regk, nullptr)) != GG_OK)
+			{
+				delete t;
+				return s2;
+			}
+		}
+		e.tables.push_back(t);
+		*out = (gg_table) (e.tables.size() - 1);
+		return GG_OK;
+	}
 	else
 		return fail(GG_EINVAL, "unknown synth table '%s'",
 			    table_name);
@@ -349,7 +384,7 @@ gg_engine_register_synth(const char *table_name, uint64_t seed, int64_t sf,
 
 	if (nm == "lineitem")
 	{
-		void *ok, *q, *p, *d, *tx, *sd, *rf, *ls;
+		void *ok, *q, *p, *d, *tx, *sd, *rf, *ls, *sk;
 
 		if ((s = add_col(t, "orderkey", GG_COL_INT64, shard_rows, nullptr, &ok)) == GG_OK &&
 		    (s = add_col(t, "qty", GG_COL_DEC64_S2, shard_rows, nullptr, &q)) == GG_OK &&
@@ -358,13 +393,15 @@ gg_engine_register_synth(const char *table_name, uint64_t seed, int64_t sf,
 		    (s = add_col(t, "tax", GG_COL_DEC64_S2, shard_rows, nullptr, &tx)) == GG_OK &&
 		    (s = add_col(t, "shipdate", GG_COL_INT32, shard_rows, nullptr, &sd)) == GG_OK &&
 		    (s = add_col(t, "rflag", GG_COL_CHAR1, shard_rows, nullptr, &rf)) == GG_OK &&
-		    (s = add_col(t, "lstatus", GG_COL_CHAR1, shard_rows, nullptr, &ls)) == GG_OK)
+		    (s = add_col(t, "lstatus", GG_COL_CHAR1, shard_rows, nullptr, &ls)) == GG_OK &&
+		    (s = add_col(t, "suppkey", GG_COL_INT64, shard_rows, nullptr, &sk)) == GG_OK)
 		{
 			hipError_t he = launch_gen_lineitem(
-				e.stream, seed, 0, total, nseg, seg,
+				e.stream, seed, sf, 0, total, nseg, seg,
 				(int64_t *) ok, (int64_t *) q, (int64_t *) p,
 				(int64_t *) d, (int64_t *) tx, (int32_t *) sd,
-				(uint8_t *) rf, (uint8_t *) ls, ctr);
+				(uint8_t *) rf, (uint8_t *) ls,
+				(int64_t *) sk, ctr);
 			if (he != hipSuccess)
 				s = fail(GG_EGPU, "gen_lineitem: %s",
 					 hipGetErrorString(he));
@@ -388,18 +425,35 @@ gg_engine_register_synth(const char *table_name, uint64_t seed, int64_t sf,
 					 hipGetErrorString(he));
 		}
 	}
-	else
+	else if (nm == "customer")
 	{
-		void *ck, *ms;
+		void *ck, *ms, *nk;
 
 		if ((s = add_col(t, "custkey", GG_COL_INT64, shard_rows, nullptr, &ck)) == GG_OK &&
-		    (s = add_col(t, "mktseg", GG_COL_CHAR1, shard_rows, nullptr, &ms)) == GG_OK)
+		    (s = add_col(t, "mktseg", GG_COL_CHAR1, shard_rows, nullptr, &ms)) == GG_OK &&
+		    (s = add_col(t, "nationkey", GG_COL_CHAR1, shard_rows, nullptr, &nk)) == GG_OK)
 		{
 			hipError_t he = launch_gen_customer(
 				e.stream, seed, 0, total, nseg, seg,
-				(int64_t *) ck, (uint8_t *) ms, ctr);
+				(int64_t *) ck, (uint8_t *) ms,
+				(uint8_t *) nk, ctr);
 			if (he != hipSuccess)
 				s = fail(GG_EGPU, "gen_customer: %s",
+					 hipGetErrorString(he));
+		}
+	}
+	else	/* supplier */
+	{
+		void *sk, *nk;
+
+		if ((s = add_col(t, "suppkey", GG_COL_INT64, shard_rows, nullptr, &sk)) == GG_OK &&
+		    (s = add_col(t, "nationkey", GG_COL_CHAR1, shard_rows, nullptr, &nk)) == GG_OK)
+		{
+			hipError_t he = launch_gen_supplier(
+				e.stream, seed, 0, total, nseg, seg,
+				(int64_t *) sk, (uint8_t *) nk, ctr);
+			if (he != hipSuccess)
+				s = fail(GG_EGPU, "gen_supplier: %s",
 					 hipGetErrorString(he));
 		}
 	}
@@ -440,6 +494,14 @@ gg_engine_compile_pipeline(const gg_pipeline_desc *desc, gg_pipeline *out)
 			    !get_table(desc->orders) ||
 			    !get_table(desc->customer))
 				return fail(GG_EINVAL, "Q3: bad table handle");
+			break;
+		case GG_PIPE_Q5:
+			if (!get_table(desc->lineitem) ||
+			    !get_table(desc->orders) ||
+			    !get_table(desc->customer) ||
+			    !get_table(desc->supplier) ||
+			    !get_table(desc->nation))
+				return fail(GG_EINVAL, "Q5: bad table handle");
 			break;
 		default:
 			return fail(GG_ENOTSUP,
@@ -828,7 +890,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_orders_filter_compact(e.stream, o_ok, o_ck, o_dt,
-						    o_pr, od->nrows, cutoff,
+						    o_pr, od->nrows,
+						    INT32_MIN, cutoff,
 						    f_ck, f_ok, f_pay, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nfil));
@@ -1178,6 +1241,577 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	return GG_OK;
 }
 
+/* ---------------- execution: Q5 (mpph5) ---------------- */
+
+/* n_name strings in nationkey order (reference fixture nation.csv;
+ * needed only for the ORDER BY tiebreak and display) */
+static const char *const GG_NATION_NAMES[GG_NNATIONS] = {
+	"ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+	"FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ", "JAPAN",
+	"JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU", "CHINA",
+	"ROMANIA", "SAUDI ARABIA", "VIETNAM", "RUSSIA", "UNITED KINGDOM",
+	"UNITED STATES"
+};
+
+struct Q5RowCmp
+{
+	bool operator()(const gg_q5_result_row &a,
+			const gg_q5_result_row &b) const
+	{
+		u128 ra = ((u128) (uint64_t) a.rev_hi << 64) | a.rev_lo;
+		u128 rb = ((u128) (uint64_t) b.rev_hi << 64) | b.rev_lo;
+
+		if (ra != rb)
+			return ra > rb;	/* ORDER BY revenue DESC */
+		return std::strcmp(GG_NATION_NAMES[a.nationkey],
+				   GG_NATION_NAMES[b.nationkey]) < 0;
+	}
+};
+
+static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
+			 size_t *written)
+{
+	Engine &e = engine();
+	Table *li = get_table(p->desc.lineitem);
+	Table *od = get_table(p->desc.orders);
+	Table *cu = get_table(p->desc.customer);
+	Table *su = get_table(p->desc.supplier);
+	Table *na = get_table(p->desc.nation);
+	int nseg = e.cfg.n_segments;
+	int me = e.cfg.segment_id;
+	bool exch = nseg > 1 || getenv("GG_FORCE_EXCHANGE") != nullptr;
+
+	if (bytes < sizeof(gg_q5_result_hdr) +
+	    GG_NNATIONS * sizeof(gg_q5_result_row))
+		return fail(GG_EINVAL, "arena too small");
+
+	const int64_t *l_ok = (const int64_t *) li->col("orderkey");
+	const int64_t *l_sk = (const int64_t *) li->col("suppkey");
+	const int64_t *l_pc = (const int64_t *) li->col("price");
+	const int64_t *l_dc = (const int64_t *) li->col("disc");
+	const int64_t *o_ok = (const int64_t *) od->col("orderkey");
+	const int64_t *o_ck = (const int64_t *) od->col("custkey");
+	const int32_t *o_dt = (const int32_t *) od->col("orderdate");
+	const int64_t *c_ck = (const int64_t *) cu->col("custkey");
+	const uint8_t *c_nk = (const uint8_t *) cu->col("nationkey");
+	const int64_t *s_sk = (const int64_t *) su->col("suppkey");
+	const uint8_t *s_nk = (const uint8_t *) su->col("nationkey");
+	const int32_t *n_nk = (const int32_t *) na->col("nationkey");
+	const int32_t *n_rk = (const int32_t *) na->col("regionkey");
+
+	if (!l_ok || !l_sk || !l_pc || !l_dc || !o_ok || !o_ck || !o_dt ||
+	    !c_ck || !c_nk || !s_sk || !s_nk || !n_nk || !n_rk)
+		return fail(GG_EINVAL, "Q5: missing column");
+
+	int32_t date_lo = p->desc.cutoff_date;
+	int32_t date_hi = p->desc.cutoff_hi;
+	uint8_t regionkey = p->desc.regionkey;
+	unsigned long long *ctr =
+		(unsigned long long *) p->sget("ctr", 8);
+
+	if (!ctr)
+		return fail(GG_ENOMEM, "scratch");
+
+	/* nation dim (25 rows) → host → device u8 region_of[25]; the
+	 * region⋈nation semi-join resolved here, the broadcast-Motion
+	 * analog for the tiny dims */
+	uint8_t region_of_h[GG_NNATIONS];
+	{
+		std::vector<int32_t> nk(na->nrows), rk(na->nrows);
+
+		GG_HIP(hipMemcpy(nk.data(), n_nk, na->nrows * 4,
+				 hipMemcpyDeviceToHost));
+		GG_HIP(hipMemcpy(rk.data(), n_rk, na->nrows * 4,
+				 hipMemcpyDeviceToHost));
+		std::memset(region_of_h, 0xff, sizeof(region_of_h));
+		for (int64_t i = 0; i < na->nrows; i++)
+			if (nk[i] >= 0 && nk[i] < GG_NNATIONS)
+				region_of_h[nk[i]] = (uint8_t) rk[i];
+	}
+	uint8_t *region_of = (uint8_t *) p->sget("region_of", 32);
+
+	if (!region_of)
+		return fail(GG_ENOMEM, "region_of");
+	GG_HIP(hipMemcpy(region_of, region_of_h, GG_NNATIONS,
+			 hipMemcpyHostToDevice));
+
+	/* 1. customer map: c_custkey → c_nationkey (no filter, no bloom) */
+	DeviceHashTable cust{};
+	{
+		Timed tm(e.stream);
+
+		if (!p->cust_slots)
+			p->cust_slots = next_pow2(2 * (uint64_t) (cu->nrows + 1));
+		cust.nslots = p->cust_slots;
+		cust.keys = (unsigned long long *)
+			p->sget("cust.keys", cust.nslots * 8);
+		cust.payload = (unsigned long long *)
+			p->sget("cust.pay", cust.nslots * 8);
+		if (!cust.keys || !cust.payload)
+			return fail(GG_ENOMEM, "cust map");
+		GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8, e.stream));
+		GG_HIP(launch_build_kv(e.stream, c_ck, c_nk, cu->nrows, cust));
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("build_customer_map");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += cu->nrows;
+		st.hbm_bytes += cu->nrows * 9;
+	}
+
+	/* 2. supplier table: s_suppkey → s_nationkey, in-region only */
+	DeviceHashTable supp{};
+	{
+		Timed tm(e.stream);
+
+		if (!exch)
+		{
+			if (!p->supp_slots)
+				p->supp_slots =
+					next_pow2(2 * (uint64_t) (su->nrows + 1));
+			supp.nslots = p->supp_slots;
+			supp.bloom_words = supp.nslots / 8 < 1024
+				? 1024 : supp.nslots / 8;
+			supp.keys = (unsigned long long *)
+				p->sget("supp.keys", supp.nslots * 8);
+			supp.payload = (unsigned long long *)
+				p->sget("supp.pay", supp.nslots * 8);
+			supp.bloom = (unsigned long long *)
+				p->sget("supp.bloom", supp.bloom_words * 8);
+			if (!supp.keys || !supp.payload || !supp.bloom)
+				return fail(GG_ENOMEM, "supp table");
+			GG_HIP(hipMemsetAsync(supp.keys, 0, supp.nslots * 8,
+					      e.stream));
+			GG_HIP(hipMemsetAsync(supp.bloom, 0,
+					      supp.bloom_words * 8, e.stream));
+			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+			GG_HIP(launch_build_supp(e.stream, s_sk, s_nk,
+						 su->nrows, region_of,
+						 regionkey, supp, ctr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+		}
+		else
+		{
+			/* broadcast Motion of the small dim: compact the
+			 * local in-region shard, allgather (padded), insert
+			 * every rank's rows locally */
+			if (!comm_ready())
+				return fail(GG_ESTATE,
+					    "exchange path requires comm");
+			unsigned long long nkept = 0;
+			int64_t *csk = (int64_t *) p->sget("supp.csk",
+							   (su->nrows + 1) * 8);
+			int64_t *csn = (int64_t *) p->sget("supp.csn",
+							   (su->nrows + 1) * 8);
+
+			if (!csk || !csn)
+				return fail(GG_ENOMEM, "supp compact");
+			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+			GG_HIP(launch_supp_filter_compact(e.stream, s_sk, s_nk,
+							  su->nrows, region_of,
+							  regionkey, csk, csn,
+							  ctr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr, &nkept));
+
+			/* exchange counts */
+			std::vector<unsigned long long> counts(nseg);
+			{
+				unsigned long long *g = (unsigned long long *)
+					p->sget("supp.cnt", (1 + (size_t) nseg) * 8);
+
+				if (!g)
+					return fail(GG_ENOMEM, "supp cnt");
+				GG_HIP(hipMemcpy(g, &nkept, 8,
+						 hipMemcpyHostToDevice));
+				GG_TRY(comm_allgather_u64(g, g + 1, 1));
+				GG_HIP(hipMemcpy(counts.data(), g + 1,
+						 (size_t) nseg * 8,
+						 hipMemcpyDeviceToHost));
+			}
+			unsigned long long maxc = 0, total = 0;
+
+			for (int r = 0; r < nseg; r++)
+			{
+				total += counts[r];
+				if (counts[r] > maxc)
+					maxc = counts[r];
+			}
+			/* padded allgather of (sk, sn) pairs */
+			uint64_t per = 2 * (maxc + 1);
+			int64_t *gs = (int64_t *)
+				p->sget("supp.gather",
+					(per + per * (size_t) nseg) * 8);
+
+			if (!gs)
+				return fail(GG_ENOMEM, "supp gather");
+			GG_HIP(hipMemcpyAsync(gs, csk, (nkept + 1) * 8,
+					      hipMemcpyDeviceToDevice, e.stream));
+			GG_HIP(hipMemcpyAsync(gs + (maxc + 1), csn,
+					      (nkept + 1) * 8,
+					      hipMemcpyDeviceToDevice, e.stream));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(comm_allgather_u64(gs, gs + per, per));
+
+			if (!p->supp_slots)
+				p->supp_slots = next_pow2(2 * (total + 1));
+			supp.nslots = p->supp_slots;
+			supp.bloom_words = supp.nslots / 8 < 1024
+				? 1024 : supp.nslots / 8;
+			supp.keys = (unsigned long long *)
+				p->sget("supp.keys", supp.nslots * 8);
+			supp.payload = (unsigned long long *)
+				p->sget("supp.pay", supp.nslots * 8);
+			supp.bloom = (unsigned long long *)
+				p->sget("supp.bloom", supp.bloom_words * 8);
+			if (!supp.keys || !supp.payload || !supp.bloom)
+				return fail(GG_ENOMEM, "supp table");
+			GG_HIP(hipMemsetAsync(supp.keys, 0, supp.nslots * 8,
+					      e.stream));
+			GG_HIP(hipMemsetAsync(supp.bloom, 0,
+					      supp.bloom_words * 8, e.stream));
+			for (int r = 0; r < nseg; r++)
+			{
+				int64_t *base = gs + per + (size_t) r * per;
+
+				if (counts[r])
+					GG_HIP(launch_insert_supp(
+						e.stream, base,
+						base + (maxc + 1),
+						(int64_t) counts[r], supp));
+			}
+			GG_HIP(hipStreamSynchronize(e.stream));
+		}
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("build_supplier");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += su->nrows;
+	}
+
+	/* 3. orders → okey → c_nationkey map */
+	DeviceHashTable ord{};
+	unsigned long long nmatch = 0;
+
+	if (!exch)
+	{
+		Timed tm(e.stream);
+
+		if (!p->ord_slots)
+		{
+			unsigned long long nfil = 0;
+
+			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+			GG_HIP(launch_count_date_range(e.stream, o_dt,
+						       od->nrows, date_lo,
+						       date_hi, ctr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr, &nfil));
+			p->ord_slots = next_pow2(2 * (nfil + 1));
+		}
+		ord.nslots = p->ord_slots;
+		ord.bloom_words = ord.nslots / 8 < 1024 ? 1024 : ord.nslots / 8;
+		ord.keys = (unsigned long long *)
+			p->sget("ord.keys", ord.nslots * 8);
+		ord.payload = (unsigned long long *)
+			p->sget("ord.pay", ord.nslots * 8);
+		ord.bloom = (unsigned long long *)
+			p->sget("ord.bloom", ord.bloom_words * 8);
+		if (!ord.keys || !ord.payload || !ord.bloom)
+			return fail(GG_ENOMEM, "ord table");
+		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
+				      e.stream));
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_build_orders_q5(e.stream, o_ok, o_ck, o_dt,
+					      od->nrows, date_lo, date_hi,
+					      cust, ord, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nmatch));
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("build_orders");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += od->nrows;
+		st.rows_out += (int64_t) nmatch;
+	}
+	else
+	{
+		/* two Motion redistribute legs, as in Q3 but with the
+		 * customer MAP payload (c_nationkey) */
+		if (!comm_ready())
+			return fail(GG_ESTATE, "exchange path requires comm");
+		Timed tm(e.stream);
+		int64_t *f_ck = (int64_t *) p->sget("f_ck", (od->nrows + 1) * 8);
+		int64_t *f_ok = (int64_t *) p->sget("f_ok", (od->nrows + 1) * 8);
+		int64_t *f_pay = (int64_t *) p->sget("f_pay", (od->nrows + 1) * 8);
+		unsigned long long *dcnt =
+			(unsigned long long *) p->sget("dcnt", (size_t) nseg * 8);
+		unsigned long long *doffs =
+			(unsigned long long *) p->sget("doffs", (size_t) nseg * 8);
+		unsigned long long nfil = 0;
+
+		if (!f_ck || !f_ok || !f_pay || !dcnt || !doffs)
+			return fail(GG_ENOMEM, "exchange scratch");
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_orders_filter_compact(e.stream, o_ok, o_ck, o_dt,
+						    o_dt /* prio unused */,
+						    od->nrows, date_lo,
+						    date_hi, f_ck, f_ok,
+						    f_pay, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nfil));
+
+		std::vector<unsigned long long> cnts(nseg), offs(nseg + 1, 0);
+
+		GG_HIP(hipMemsetAsync(dcnt, 0, (size_t) nseg * 8, e.stream));
+		GG_HIP(launch_part_count(e.stream, f_ck, (int64_t) nfil, nseg,
+					 dcnt));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(cnts.data(), dcnt, (size_t) nseg * 8,
+				 hipMemcpyDeviceToHost));
+		for (int i = 0; i < nseg; i++)
+			offs[i + 1] = offs[i] + cnts[i];
+
+		int64_t *s_ckb = (int64_t *) p->sget("s_ck", (nfil + 1) * 8);
+		int64_t *s_okb = (int64_t *) p->sget("s_ok", (nfil + 1) * 8);
+
+		if (!s_ckb || !s_okb)
+			return fail(GG_ENOMEM, "send scratch");
+		GG_HIP(hipMemcpy(doffs, offs.data(), (size_t) nseg * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(launch_part_scatter3(e.stream, f_ck, (int64_t) nfil,
+					    nseg, f_ck, f_ok, nullptr, doffs,
+					    s_ckb, s_okb, nullptr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+
+		std::vector<unsigned long long> allcnt((size_t) nseg * nseg);
+		{
+			unsigned long long *g = (unsigned long long *)
+				p->sget("cntg", (size_t) (nseg + nseg * nseg) * 8);
+
+			if (!g)
+				return fail(GG_ENOMEM, "cnt allgather");
+			GG_HIP(hipMemcpy(g, cnts.data(), (size_t) nseg * 8,
+					 hipMemcpyHostToDevice));
+			GG_TRY(comm_allgather_u64(g, g + nseg, nseg));
+			GG_HIP(hipMemcpy(allcnt.data(), g + nseg,
+					 allcnt.size() * 8,
+					 hipMemcpyDeviceToHost));
+		}
+		std::vector<unsigned long long> rcnts(nseg), roffs(nseg + 1, 0);
+
+		for (int s2 = 0; s2 < nseg; s2++)
+			rcnts[s2] = allcnt[(size_t) s2 * nseg + me];
+		for (int i = 0; i < nseg; i++)
+			roffs[i + 1] = roffs[i] + rcnts[i];
+		uint64_t rtotal = roffs[nseg];
+
+		int64_t *r_ck = (int64_t *) p->sget("r_ck", (rtotal + 1) * 8);
+		int64_t *r_ok = (int64_t *) p->sget("r_ok", (rtotal + 1) * 8);
+
+		if (!r_ck || !r_ok)
+			return fail(GG_ENOMEM, "recv scratch");
+		GG_TRY(comm_alltoallv_i64(s_ckb, offs.data(), cnts.data(),
+					  r_ck, roffs.data(), rcnts.data()));
+		GG_TRY(comm_alltoallv_i64(s_okb, offs.data(), cnts.data(),
+					  r_ok, roffs.data(), rcnts.data()));
+
+		unsigned long long nm = 0;
+		int64_t *m_ok = (int64_t *) p->sget("m_ok", (rtotal + 1) * 8);
+		int64_t *m_nat = (int64_t *) p->sget("m_pay", (rtotal + 1) * 8);
+
+		if (!m_ok || !m_nat)
+			return fail(GG_ENOMEM, "match scratch");
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_probe_cust_map_compact(e.stream, r_ck, r_ok,
+						     (int64_t) rtotal, cust,
+						     m_ok, m_nat, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nm));
+
+		std::vector<unsigned long long> cnts2(nseg), offs2(nseg + 1, 0);
+
+		GG_HIP(hipMemsetAsync(dcnt, 0, (size_t) nseg * 8, e.stream));
+		GG_HIP(launch_part_count(e.stream, m_ok, (int64_t) nm, nseg,
+					 dcnt));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(cnts2.data(), dcnt, (size_t) nseg * 8,
+				 hipMemcpyDeviceToHost));
+		for (int i = 0; i < nseg; i++)
+			offs2[i + 1] = offs2[i] + cnts2[i];
+
+		int64_t *s2_ok = (int64_t *) p->sget("s2_ok", (nm + 1) * 8);
+		int64_t *s2_nat = (int64_t *) p->sget("s2_pay", (nm + 1) * 8);
+
+		if (!s2_ok || !s2_nat)
+			return fail(GG_ENOMEM, "send2 scratch");
+		GG_HIP(hipMemcpy(doffs, offs2.data(), (size_t) nseg * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(launch_part_scatter3(e.stream, m_ok, (int64_t) nm, nseg,
+					    m_ok, m_nat, nullptr, doffs,
+					    s2_ok, s2_nat, nullptr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+
+		std::vector<unsigned long long> allcnt2((size_t) nseg * nseg);
+		{
+			unsigned long long *g = (unsigned long long *)
+				p->sget("cntg", (size_t) (nseg + nseg * nseg) * 8);
+
+			GG_HIP(hipMemcpy(g, cnts2.data(), (size_t) nseg * 8,
+					 hipMemcpyHostToDevice));
+			GG_TRY(comm_allgather_u64(g, g + nseg, nseg));
+			GG_HIP(hipMemcpy(allcnt2.data(), g + nseg,
+					 allcnt2.size() * 8,
+					 hipMemcpyDeviceToHost));
+		}
+		std::vector<unsigned long long> rcnts2(nseg), roffs2(nseg + 1, 0);
+
+		for (int s2 = 0; s2 < nseg; s2++)
+			rcnts2[s2] = allcnt2[(size_t) s2 * nseg + me];
+		for (int i = 0; i < nseg; i++)
+			roffs2[i + 1] = roffs2[i] + rcnts2[i];
+		uint64_t rtotal2 = roffs2[nseg];
+
+		int64_t *r2_ok = (int64_t *) p->sget("r2_ok", (rtotal2 + 1) * 8);
+		int64_t *r2_nat = (int64_t *) p->sget("r2_pay", (rtotal2 + 1) * 8);
+
+		if (!r2_ok || !r2_nat)
+			return fail(GG_ENOMEM, "recv2 scratch");
+		GG_TRY(comm_alltoallv_i64(s2_ok, offs2.data(), cnts2.data(),
+					  r2_ok, roffs2.data(), rcnts2.data()));
+		GG_TRY(comm_alltoallv_i64(s2_nat, offs2.data(), cnts2.data(),
+					  r2_nat, roffs2.data(),
+					  rcnts2.data()));
+
+		nmatch = rtotal2;
+		if (!p->ord_slots)
+			p->ord_slots = next_pow2(2 * (rtotal2 + 1));
+		ord.nslots = p->ord_slots;
+		ord.bloom_words = ord.nslots / 8 < 1024 ? 1024 : ord.nslots / 8;
+		ord.keys = (unsigned long long *)
+			p->sget("ord.keys", ord.nslots * 8);
+		ord.payload = (unsigned long long *)
+			p->sget("ord.pay", ord.nslots * 8);
+		ord.bloom = (unsigned long long *)
+			p->sget("ord.bloom", ord.bloom_words * 8);
+		if (!ord.keys || !ord.payload || !ord.bloom)
+			return fail(GG_ENOMEM, "ord table");
+		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
+				      e.stream));
+		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_nat,
+					    (int64_t) rtotal2, ord));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		double ms = tm.stop();
+		KernelStatAcc &st = p->stat("orders_exchange");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += od->nrows;
+		st.rows_out += (int64_t) nmatch;
+	}
+
+	/* 4. lineitem probe + 25-nation aggregation */
+	unsigned long long *acc = (unsigned long long *)
+		p->sget("q5acc", GG_NNATIONS * 3 * 8);
+	unsigned long long njoin = 0;
+
+	if (!acc)
+		return fail(GG_ENOMEM, "acc");
+	GG_HIP(hipMemsetAsync(acc, 0, GG_NNATIONS * 3 * 8, e.stream));
+	GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+	{
+		Timed tm(e.stream);
+
+		GG_HIP(launch_probe_lineitem_q5(e.stream, l_ok, l_sk, l_pc,
+						l_dc, li->nrows, ord, supp,
+						acc, ctr));
+		double ms = tm.stop();
+		GG_TRY(read_counter(ctr, &njoin));
+		KernelStatAcc &st = p->stat("probe_lineitem_q5");
+
+		st.launches++;
+		st.total_ms += ms;
+		st.rows_in += li->nrows;
+		st.rows_out += (int64_t) njoin;
+		st.hbm_bytes += li->nrows * 32;	/* okey,skey,price,disc */
+	}
+
+	/* 5. combine + finalize */
+	std::vector<unsigned long long> hacc(GG_NNATIONS * 3);
+
+	GG_HIP(hipMemcpy(hacc.data(), acc, hacc.size() * 8,
+			 hipMemcpyDeviceToHost));
+
+	u128 rev[GG_NNATIONS];
+	uint64_t cnt[GG_NNATIONS];
+
+	for (int n = 0; n < GG_NNATIONS; n++)
+	{
+		cnt[n] = hacc[n * 3];
+		rev[n] = ((u128) hacc[n * 3 + 1]) |
+			((u128) hacc[n * 3 + 2] << 64);
+	}
+	if (nseg > 1)
+	{
+		size_t per = GG_NNATIONS * 3;
+		unsigned long long *g = (unsigned long long *)
+			p->sget("q5gather", (per + per * (size_t) nseg) * 8);
+
+		if (!g)
+			return fail(GG_ENOMEM, "q5 gather");
+		GG_HIP(hipMemcpy(g, acc, per * 8, hipMemcpyDeviceToDevice));
+		GG_TRY(comm_allgather_u64(g, g + per, per));
+		std::vector<unsigned long long> all(per * (size_t) nseg);
+
+		GG_HIP(hipMemcpy(all.data(), g + per, all.size() * 8,
+				 hipMemcpyDeviceToHost));
+		for (int n = 0; n < GG_NNATIONS; n++)
+		{
+			cnt[n] = 0;
+			rev[n] = 0;
+		}
+		for (int r = 0; r < nseg; r++)
+			for (int n = 0; n < GG_NNATIONS; n++)
+			{
+				const unsigned long long *v =
+					&all[per * (size_t) r + (size_t) n * 3];
+
+				cnt[n] += v[0];
+				rev[n] += ((u128) v[1]) | ((u128) v[2] << 64);
+			}
+	}
+
+	std::vector<gg_q5_result_row> rows;
+
+	for (int n = 0; n < GG_NNATIONS; n++)
+	{
+		if (!cnt[n])
+			continue;
+		gg_q5_result_row r;
+
+		r.nationkey = n;
+		r._pad = 0;
+		r.count = (int64_t) cnt[n];
+		r.rev_lo = (uint64_t) rev[n];
+		r.rev_hi = (int64_t) (rev[n] >> 64);
+		rows.push_back(r);
+	}
+	std::sort(rows.begin(), rows.end(), Q5RowCmp());
+
+	gg_q5_result_hdr *hdr = (gg_q5_result_hdr *) arena;
+
+	hdr->n_out = (int64_t) rows.size();
+	std::memcpy(hdr + 1, rows.data(),
+		    rows.size() * sizeof(gg_q5_result_row));
+	*written = sizeof(*hdr) + rows.size() * sizeof(gg_q5_result_row);
+	return GG_OK;
+}
+
 extern "C" gg_status
 gg_engine_execute(gg_pipeline h, void *arena, size_t bytes, size_t *written)
 {
@@ -1199,6 +1833,8 @@ gg_engine_execute(gg_pipeline h, void *arena, size_t bytes, size_t *written)
 			return exec_q3(p, arena, bytes, written);
 		case GG_PIPE_SUMPRICE:
 			return exec_sumprice(p, arena, bytes, written);
+		case GG_PIPE_Q5:
+			return exec_q5(p, arena, bytes, written);
 	}
 	return fail(GG_ENOTSUP, "unsupported pipeline");
 }

@@ -106,6 +106,7 @@ struct Pipeline
 	 * per pipeline), skipping the count passes afterwards */
 	uint64_t cust_slots = 0;
 	uint64_t ord_slots = 0;
+	uint64_t supp_slots = 0;
 
 	KernelStatAcc &stat(const char *name)
 	{
@@ -184,11 +185,12 @@ struct Timed
 
 /* ---- kernel launchers (kernels.hip) — return hipError_t ---- */
 
-hipError_t launch_gen_lineitem(hipStream_t s, uint64_t seed, int64_t row_lo,
-			       int64_t n, int nseg, int seg,
+hipError_t launch_gen_lineitem(hipStream_t s, uint64_t seed, int64_t sf,
+			       int64_t row_lo, int64_t n, int nseg, int seg,
 			       int64_t *orderkey, int64_t *qty, int64_t *price,
 			       int64_t *disc, int64_t *tax, int32_t *shipdate,
 			       uint8_t *rflag, uint8_t *lstatus,
+			       int64_t *suppkey,
 			       unsigned long long *out_count);
 hipError_t launch_gen_orders(hipStream_t s, uint64_t seed, int64_t sf,
 			     int64_t row_lo, int64_t n, int nseg, int seg,
@@ -197,7 +199,45 @@ hipError_t launch_gen_orders(hipStream_t s, uint64_t seed, int64_t sf,
 			     unsigned long long *out_count);
 hipError_t launch_gen_customer(hipStream_t s, uint64_t seed, int64_t row_lo,
 			       int64_t n, int nseg, int seg, int64_t *custkey,
-			       uint8_t *mktseg, unsigned long long *out_count);
+			       uint8_t *mktseg, uint8_t *nationkey,
+			       unsigned long long *out_count);
+hipError_t launch_gen_supplier(hipStream_t s, uint64_t seed, int64_t row_lo,
+			       int64_t n, int nseg, int seg, int64_t *suppkey,
+			       uint8_t *nation, unsigned long long *out_count);
+hipError_t launch_build_kv(hipStream_t s, const int64_t *keys,
+			   const uint8_t *vals, int64_t n, DeviceHashTable t);
+hipError_t launch_build_supp(hipStream_t s, const int64_t *suppkey,
+			     const uint8_t *snation, int64_t n,
+			     const uint8_t *region_of, uint8_t regionkey,
+			     DeviceHashTable t, unsigned long long *out_count);
+hipError_t launch_supp_filter_compact(hipStream_t s, const int64_t *suppkey,
+				      const uint8_t *snation, int64_t n,
+				      const uint8_t *region_of,
+				      uint8_t regionkey, int64_t *out_sk,
+				      int64_t *out_sn,
+				      unsigned long long *out_count);
+hipError_t launch_insert_supp(hipStream_t s, const int64_t *sk,
+			      const int64_t *sn, int64_t n,
+			      DeviceHashTable t);
+hipError_t launch_build_orders_q5(hipStream_t s, const int64_t *okey,
+				  const int64_t *ckey, const int32_t *odate,
+				  int64_t n, int32_t date_lo, int32_t date_hi,
+				  DeviceHashTable cust, DeviceHashTable ord,
+				  unsigned long long *match_count);
+hipError_t launch_count_date_range(hipStream_t s, const int32_t *odate,
+				   int64_t n, int32_t date_lo,
+				   int32_t date_hi, unsigned long long *out);
+hipError_t launch_probe_cust_map_compact(hipStream_t s, const int64_t *ckey,
+					 const int64_t *okey, int64_t n,
+					 DeviceHashTable cust,
+					 int64_t *out_okey, int64_t *out_nat,
+					 unsigned long long *out_count);
+hipError_t launch_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
+				    const int64_t *skey, const int64_t *price,
+				    const int64_t *disc, int64_t n,
+				    DeviceHashTable ord, DeviceHashTable supp,
+				    unsigned long long *acc,
+				    unsigned long long *join_rows);
 
 hipError_t launch_q1(hipStream_t s, const int32_t *shipdate,
 		     const uint8_t *rflag, const uint8_t *lstatus,
@@ -250,7 +290,8 @@ hipError_t launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
 					const int64_t *ckey,
 					const int32_t *odate,
 					const int32_t *prio, int64_t n,
-					int32_t cutoff, int64_t *out_ckey,
+					int32_t date_lo, int32_t date_hi,
+					int64_t *out_ckey,
 					int64_t *out_okey, int64_t *out_pay,
 					unsigned long long *out_count);
 hipError_t launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,

@@ -259,12 +259,13 @@ launch_count_shard(hipStream_t s, int64_t row_lo, int64_t n, int nseg,
 }
 
 __global__ void
-k_gen_lineitem(uint64_t seed, int64_t row_lo, int64_t n, int nseg, int seg,
+k_gen_lineitem(uint64_t seed, int64_t sf, int64_t row_lo, int64_t n,
+	       int nseg, int seg,
 	       int64_t *__restrict__ orderkey, int64_t *__restrict__ qty,
 	       int64_t *__restrict__ price, int64_t *__restrict__ disc,
 	       int64_t *__restrict__ tax, int32_t *__restrict__ shipdate,
 	       uint8_t *__restrict__ rflag, uint8_t *__restrict__ lstatus,
-	       unsigned long long *out_count)
+	       int64_t *__restrict__ suppkey, unsigned long long *out_count)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
@@ -310,21 +311,22 @@ k_gen_lineitem(uint64_t seed, int64_t row_lo, int64_t n, int nseg, int seg,
 			shipdate[idx] = r.l_shipdate;
 			rflag[idx] = r.l_returnflag;
 			lstatus[idx] = r.l_linestatus;
+			suppkey[idx] = gg_l_suppkey(seed, row, sf);
 		}
 	}
 }
 
 hipError_t
-launch_gen_lineitem(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
-		    int nseg, int seg, int64_t *orderkey, int64_t *qty,
-		    int64_t *price, int64_t *disc, int64_t *tax,
+launch_gen_lineitem(hipStream_t s, uint64_t seed, int64_t sf, int64_t row_lo,
+		    int64_t n, int nseg, int seg, int64_t *orderkey,
+		    int64_t *qty, int64_t *price, int64_t *disc, int64_t *tax,
 		    int32_t *shipdate, uint8_t *rflag, uint8_t *lstatus,
-		    unsigned long long *out_count)
+		    int64_t *suppkey, unsigned long long *out_count)
 {
 	hipLaunchKernelGGL(k_gen_lineitem, dim3(grid_for(n)), dim3(THREADS),
-			   0, s, seed, row_lo, n, nseg, seg, orderkey, qty,
-			   price, disc, tax, shipdate, rflag, lstatus,
-			   out_count);
+			   0, s, seed, sf, row_lo, n, nseg, seg, orderkey,
+			   qty, price, disc, tax, shipdate, rflag, lstatus,
+			   suppkey, out_count);
 	return hipGetLastError();
 }
 
@@ -384,7 +386,7 @@ launch_gen_orders(hipStream_t s, uint64_t seed, int64_t sf, int64_t row_lo,
 __global__ void
 k_gen_customer(uint64_t seed, int64_t row_lo, int64_t n, int nseg, int seg,
 	       int64_t *__restrict__ custkey, uint8_t *__restrict__ mktseg,
-	       unsigned long long *out_count)
+	       uint8_t *__restrict__ nationkey, unsigned long long *out_count)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
@@ -416,17 +418,18 @@ k_gen_customer(uint64_t seed, int64_t row_lo, int64_t n, int nseg, int seg,
 			continue;
 		custkey[idx] = ckey;
 		mktseg[idx] = gg_c_mktsegment(seed, ckey);
+		nationkey[idx] = gg_c_nationkey(seed, ckey);
 	}
 }
 
 hipError_t
 launch_gen_customer(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
 		    int nseg, int seg, int64_t *custkey, uint8_t *mktseg,
-		    unsigned long long *out_count)
+		    uint8_t *nationkey, unsigned long long *out_count)
 {
 	hipLaunchKernelGGL(k_gen_customer, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, seed, row_lo, n, nseg, seg, custkey, mktseg,
-			   out_count);
+			   nationkey, out_count);
 	return hipGetLastError();
 }
 
@@ -1005,7 +1008,8 @@ k_orders_filter_compact(const int64_t *__restrict__ okey,
 			const int64_t *__restrict__ ckey,
 			const int32_t *__restrict__ odate,
 			const int32_t *__restrict__ prio, int64_t n,
-			int32_t cutoff, int64_t *__restrict__ out_ckey,
+			int32_t date_lo, int32_t date_hi,
+			int64_t *__restrict__ out_ckey,
 			int64_t *__restrict__ out_okey,
 			int64_t *__restrict__ out_pay,
 			unsigned long long *out_count)
@@ -1015,7 +1019,7 @@ k_orders_filter_compact(const int64_t *__restrict__ okey,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		if (odate[i] >= cutoff)
+		if (odate[i] < date_lo || odate[i] >= date_hi)
 			continue;
 		{
 			unsigned long long idx = atomicAdd(out_count, 1ull);
@@ -1033,13 +1037,15 @@ k_orders_filter_compact(const int64_t *__restrict__ okey,
 hipError_t
 launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
 			     const int64_t *ckey, const int32_t *odate,
-			     const int32_t *prio, int64_t n, int32_t cutoff,
-			     int64_t *out_ckey, int64_t *out_okey,
-			     int64_t *out_pay, unsigned long long *out_count)
+			     const int32_t *prio, int64_t n, int32_t date_lo,
+			     int32_t date_hi, int64_t *out_ckey,
+			     int64_t *out_okey, int64_t *out_pay,
+			     unsigned long long *out_count)
 {
 	hipLaunchKernelGGL(k_orders_filter_compact, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, okey, ckey, odate, prio, n,
-			   cutoff, out_ckey, out_okey, out_pay, out_count);
+			   date_lo, date_hi, out_ckey, out_okey, out_pay,
+			   out_count);
 	return hipGetLastError();
 }
 
@@ -1128,6 +1134,512 @@ launch_insert_orders(hipStream_t s, const int64_t *okey, const int64_t *pay,
 	hipLaunchKernelGGL(k_insert_orders, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, okey, pay, n, ord.keys, ord.payload,
 			   ord.nslots, ord.bloom, ord.bloom_words);
+	return hipGetLastError();
+}
+
+
+/* ------------------------------------------------------------------ */
+/* Q5 (mpph5): supplier/customer dims, orders map, 25-nation agg       */
+/* ------------------------------------------------------------------ */
+
+/* generic key→value build: insert every (key, val) pair (the customer
+ * c_custkey → c_nationkey map; no filter, so no bloom) */
+__global__ void
+k_build_kv(const int64_t *__restrict__ keys, const uint8_t *__restrict__ vals,
+	   int64_t n, unsigned long long *__restrict__ tkeys,
+	   unsigned long long *__restrict__ tpayload, uint64_t nslots)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k = keys[i];
+		uint64_t pos = ht_start(k, nslots);
+
+		for (;;)
+		{
+			unsigned long long prev =
+				atomicCAS(&tkeys[pos], 0ull,
+					  (unsigned long long) k);
+			if (prev == 0)
+			{
+				tpayload[pos] = vals[i];
+				break;
+			}
+			if (prev == (unsigned long long) k)
+				break;
+			pos = (pos + 1) & (nslots - 1);
+		}
+	}
+}
+
+hipError_t
+launch_build_kv(hipStream_t s, const int64_t *keys, const uint8_t *vals,
+		int64_t n, DeviceHashTable t)
+{
+	hipLaunchKernelGGL(k_build_kv, dim3(grid_for(n)), dim3(THREADS), 0, s,
+			   keys, vals, n, t.keys, t.payload, t.nslots);
+	return hipGetLastError();
+}
+
+/* probe a kv map; return payload or miss */
+__device__ inline bool
+ht_lookup_b(const unsigned long long *__restrict__ tkeys,
+	    const unsigned long long *__restrict__ tpayload, uint64_t nslots,
+	    const unsigned long long *__restrict__ bloom, uint64_t bwords,
+	    int64_t key, unsigned long long *out_payload)
+{
+	uint32_t h = gg_hashint8(key);
+
+	if (bloom && !bloom_maybe(bloom, bwords, h))
+		return false;
+	{
+		uint64_t pos = (uint64_t) h & (nslots - 1);
+
+		for (;;)
+		{
+			unsigned long long v = tkeys[pos];
+
+			if (v == (unsigned long long) key)
+			{
+				*out_payload = tpayload[pos];
+				return true;
+			}
+			if (v == 0)
+				return false;
+			pos = (pos + 1) & (nslots - 1);
+		}
+	}
+}
+
+/* supplier build: keep s_suppkey → s_nationkey for suppliers whose
+ * nation is in the target region (the nation⋈region semi-filter pushed
+ * into the build, cdbpath-style) */
+__global__ void
+k_build_supp(const int64_t *__restrict__ suppkey,
+	     const uint8_t *__restrict__ snation, int64_t n,
+	     const uint8_t *__restrict__ region_of /* [25] device */,
+	     uint8_t regionkey, unsigned long long *__restrict__ tkeys,
+	     unsigned long long *__restrict__ tpayload, uint64_t nslots,
+	     unsigned long long *__restrict__ bloom, uint64_t bwords,
+	     unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long kept = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		uint8_t sn = snation[i];
+
+		if (sn >= 25 || region_of[sn] != regionkey)
+			continue;
+		kept++;
+		{
+			int64_t k = suppkey[i];
+			uint64_t pos = ht_start(k, nslots);
+
+			if (bloom)
+				bloom_insert(bloom, bwords, k);
+			for (;;)
+			{
+				unsigned long long prev =
+					atomicCAS(&tkeys[pos], 0ull,
+						  (unsigned long long) k);
+				if (prev == 0)
+				{
+					tpayload[pos] = sn;
+					break;
+				}
+				if (prev == (unsigned long long) k)
+					break;
+				pos = (pos + 1) & (nslots - 1);
+			}
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+		kept += __shfl_down(kept, off, 64);
+	if ((threadIdx.x & 63) == 0 && kept)
+		atomicAdd(out_count, kept);
+}
+
+hipError_t
+launch_build_supp(hipStream_t s, const int64_t *suppkey,
+		  const uint8_t *snation, int64_t n, const uint8_t *region_of,
+		  uint8_t regionkey, DeviceHashTable t,
+		  unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_build_supp, dim3(grid_for(n)), dim3(THREADS), 0,
+			   s, suppkey, snation, n, region_of, regionkey,
+			   t.keys, t.payload, t.nslots, t.bloom,
+			   t.bloom_words, out_count);
+	return hipGetLastError();
+}
+
+/* supplier shard → compacted (suppkey, nation) pairs for the broadcast
+ * Motion (in-region only) */
+__global__ void
+k_supp_filter_compact(const int64_t *__restrict__ suppkey,
+		      const uint8_t *__restrict__ snation, int64_t n,
+		      const uint8_t *__restrict__ region_of, uint8_t regionkey,
+		      int64_t *__restrict__ out_sk,
+		      int64_t *__restrict__ out_sn,
+		      unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		uint8_t sn = snation[i];
+
+		if (sn >= 25 || region_of[sn] != regionkey)
+			continue;
+		{
+			unsigned long long idx = atomicAdd(out_count, 1ull);
+
+			out_sk[idx] = suppkey[i];
+			out_sn[idx] = sn;
+		}
+	}
+}
+
+hipError_t
+launch_supp_filter_compact(hipStream_t s, const int64_t *suppkey,
+			   const uint8_t *snation, int64_t n,
+			   const uint8_t *region_of, uint8_t regionkey,
+			   int64_t *out_sk, int64_t *out_sn,
+			   unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_supp_filter_compact, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, suppkey, snation, n,
+			   region_of, regionkey, out_sk, out_sn, out_count);
+	return hipGetLastError();
+}
+
+/* insert broadcast (suppkey, nation-as-i64) pairs into the supp table */
+__global__ void
+k_insert_supp(const int64_t *__restrict__ sk, const int64_t *__restrict__ sn,
+	      int64_t n, unsigned long long *__restrict__ tkeys,
+	      unsigned long long *__restrict__ tpayload, uint64_t nslots,
+	      unsigned long long *__restrict__ bloom, uint64_t bwords)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k = sk[i];
+		uint64_t pos = ht_start(k, nslots);
+
+		if (bloom)
+			bloom_insert(bloom, bwords, k);
+		for (;;)
+		{
+			unsigned long long prev =
+				atomicCAS(&tkeys[pos], 0ull,
+					  (unsigned long long) k);
+			if (prev == 0)
+			{
+				tpayload[pos] = (unsigned long long) sn[i];
+				break;
+			}
+			if (prev == (unsigned long long) k)
+				break;
+			pos = (pos + 1) & (nslots - 1);
+		}
+	}
+}
+
+hipError_t
+launch_insert_supp(hipStream_t s, const int64_t *sk, const int64_t *sn,
+		   int64_t n, DeviceHashTable t)
+{
+	hipLaunchKernelGGL(k_insert_supp, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, sk, sn, n, t.keys, t.payload, t.nslots,
+			   t.bloom, t.bloom_words);
+	return hipGetLastError();
+}
+
+/* orders (fused, single-segment): date range filter, probe customer
+ * map, insert okey → c_nationkey */
+__global__ void
+k_build_orders_q5(const int64_t *__restrict__ okey,
+		  const int64_t *__restrict__ ckey,
+		  const int32_t *__restrict__ odate, int64_t n,
+		  int32_t date_lo, int32_t date_hi,
+		  const unsigned long long *__restrict__ cust_keys,
+		  const unsigned long long *__restrict__ cust_pay,
+		  uint64_t cust_slots, unsigned long long *__restrict__ tkeys,
+		  unsigned long long *__restrict__ tpayload, uint64_t nslots,
+		  unsigned long long *__restrict__ bloom, uint64_t bwords,
+		  unsigned long long *match_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long matches = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int32_t d = odate[i];
+		unsigned long long nat;
+
+		if (d < date_lo || d >= date_hi)
+			continue;
+		if (!ht_lookup_b(cust_keys, cust_pay, cust_slots, nullptr, 0,
+				 ckey[i], &nat))
+			continue;	/* inner join: unmatched drops */
+		matches++;
+		{
+			int64_t k = okey[i];
+			uint64_t pos = ht_start(k, nslots);
+
+			if (bloom)
+				bloom_insert(bloom, bwords, k);
+			for (;;)
+			{
+				unsigned long long prev =
+					atomicCAS(&tkeys[pos], 0ull,
+						  (unsigned long long) k);
+				if (prev == 0)
+				{
+					tpayload[pos] = nat;
+					break;
+				}
+				if (prev == (unsigned long long) k)
+					break;
+				pos = (pos + 1) & (nslots - 1);
+			}
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+		matches += __shfl_down(matches, off, 64);
+	if ((threadIdx.x & 63) == 0 && matches)
+		atomicAdd(match_count, matches);
+}
+
+hipError_t
+launch_build_orders_q5(hipStream_t s, const int64_t *okey,
+		       const int64_t *ckey, const int32_t *odate, int64_t n,
+		       int32_t date_lo, int32_t date_hi, DeviceHashTable cust,
+		       DeviceHashTable ord, unsigned long long *match_count)
+{
+	hipLaunchKernelGGL(k_build_orders_q5, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, okey, ckey, odate, n,
+			   date_lo, date_hi, cust.keys, cust.payload,
+			   cust.nslots, ord.keys, ord.payload, ord.nslots,
+			   ord.bloom, ord.bloom_words, match_count);
+	return hipGetLastError();
+}
+
+/* count orders in [lo, hi) — exact-sizing pass */
+__global__ void
+k_count_date_range(const int32_t *__restrict__ odate, int64_t n,
+		   int32_t date_lo, int32_t date_hi,
+		   unsigned long long *out)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long c = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		c += (odate[i] >= date_lo && odate[i] < date_hi);
+	for (int off = 32; off; off >>= 1)
+		c += __shfl_down(c, off, 64);
+	if ((threadIdx.x & 63) == 0 && c)
+		atomicAdd(out, c);
+}
+
+hipError_t
+launch_count_date_range(hipStream_t s, const int32_t *odate, int64_t n,
+			int32_t date_lo, int32_t date_hi,
+			unsigned long long *out)
+{
+	hipLaunchKernelGGL(k_count_date_range, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, odate, n, date_lo, date_hi,
+			   out);
+	return hipGetLastError();
+}
+
+/* received (ckey, okey) rows: probe customer MAP → (okey, nation) */
+__global__ void
+k_probe_cust_map_compact(const int64_t *__restrict__ ckey,
+			 const int64_t *__restrict__ okey, int64_t n,
+			 const unsigned long long *__restrict__ cust_keys,
+			 const unsigned long long *__restrict__ cust_pay,
+			 uint64_t cust_slots, int64_t *__restrict__ out_okey,
+			 int64_t *__restrict__ out_nat,
+			 unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		unsigned long long nat;
+
+		if (!ht_lookup_b(cust_keys, cust_pay, cust_slots, nullptr, 0,
+				 ckey[i], &nat))
+			continue;
+		{
+			unsigned long long idx = atomicAdd(out_count, 1ull);
+
+			out_okey[idx] = okey[i];
+			out_nat[idx] = (int64_t) nat;
+		}
+	}
+}
+
+hipError_t
+launch_probe_cust_map_compact(hipStream_t s, const int64_t *ckey,
+			      const int64_t *okey, int64_t n,
+			      DeviceHashTable cust, int64_t *out_okey,
+			      int64_t *out_nat, unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_probe_cust_map_compact, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, ckey, okey, n, cust.keys,
+			   cust.payload, cust.nslots, out_okey, out_nat,
+			   out_count);
+	return hipGetLastError();
+}
+
+/* lineitem probe: supplier (small, bloom) then orders (bloom), nation
+ * equality, 25-slot LDS-privatized aggregation.
+ * acc layout: [nation][0]=count, [nation][1]=rev_lo, [nation][2]=rev_hi */
+__global__ __launch_bounds__(THREADS, 2)
+void k_probe_lineitem_q5(const int64_t *__restrict__ okey,
+			 const int64_t *__restrict__ skey,
+			 const int64_t *__restrict__ price,
+			 const int64_t *__restrict__ disc, int64_t n,
+			 const unsigned long long *__restrict__ ord_keys,
+			 const unsigned long long *__restrict__ ord_pay,
+			 uint64_t ord_slots,
+			 const unsigned long long *__restrict__ ord_bloom,
+			 uint64_t ord_bwords,
+			 const unsigned long long *__restrict__ supp_keys,
+			 const unsigned long long *__restrict__ supp_pay,
+			 uint64_t supp_slots,
+			 const unsigned long long *__restrict__ supp_bloom,
+			 uint64_t supp_bwords,
+			 unsigned long long *__restrict__ acc /* [25][3] */,
+			 unsigned long long *join_rows)
+{
+	__shared__ unsigned long long lds[25][2];	/* cnt, rev */
+
+	for (int i = threadIdx.x; i < 50; i += blockDim.x)
+		((unsigned long long *) lds)[i] = 0;
+	__syncthreads();
+
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long joined = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		unsigned long long snat, onat;
+
+		if (!ht_lookup_b(supp_keys, supp_pay, supp_slots, supp_bloom,
+				 supp_bwords, skey[i], &snat))
+			continue;
+		if (!ht_lookup_b(ord_keys, ord_pay, ord_slots, ord_bloom,
+				 ord_bwords, okey[i], &onat))
+			continue;
+		if (snat != onat)	/* c_nationkey = s_nationkey */
+			continue;
+		joined++;
+		{
+			unsigned long long rev4 = (unsigned long long)
+				(price[i] * (100 - disc[i]));
+
+			atomicAdd(&lds[snat][0], 1ull);
+			atomicAdd(&lds[snat][1], rev4);
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+		joined += __shfl_down(joined, off, 64);
+	if ((threadIdx.x & 63) == 0 && joined)
+		atomicAdd(join_rows, joined);
+	__syncthreads();
+	for (int nat = threadIdx.x; nat < 25; nat += blockDim.x)
+	{
+		unsigned long long c = lds[nat][0];
+		unsigned long long r = lds[nat][1];
+
+		if (!c)
+			continue;
+		atomicAdd(&acc[nat * 3 + 0], c);
+		{
+			unsigned long long old =
+				atomicAdd(&acc[nat * 3 + 1], r);
+			if (old + r < old)
+				atomicAdd(&acc[nat * 3 + 2], 1ull);
+		}
+	}
+}
+
+hipError_t
+launch_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
+			 const int64_t *skey, const int64_t *price,
+			 const int64_t *disc, int64_t n, DeviceHashTable ord,
+			 DeviceHashTable supp, unsigned long long *acc,
+			 unsigned long long *join_rows)
+{
+	hipLaunchKernelGGL(k_probe_lineitem_q5, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, okey, skey, price, disc, n,
+			   ord.keys, ord.payload, ord.nslots, ord.bloom,
+			   ord.bloom_words, supp.keys, supp.payload,
+			   supp.nslots, supp.bloom, supp.bloom_words, acc,
+			   join_rows);
+	return hipGetLastError();
+}
+
+__global__ void
+k_gen_supplier(uint64_t seed, int64_t row_lo, int64_t n, int nseg, int seg,
+	       int64_t *__restrict__ suppkey, uint8_t *__restrict__ nation,
+	       unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t sk = row_lo + i + 1;
+		bool own = (nseg <= 1) ||
+			(gg_cdbhash_segment_int8(sk, nseg) == seg);
+		int64_t idx = i;
+
+		if (nseg > 1)
+		{
+			unsigned long long mask = __ballot(own);
+			int lane = threadIdx.x & 63;
+			int nbefore = __popcll(mask & ((1ull << lane) - 1));
+			int total = __popcll(mask);
+			int leader = __ffsll((long long) mask) - 1;
+			unsigned long long base = 0;
+
+			if (total && lane == leader)
+				base = atomicAdd(out_count,
+						 (unsigned long long) total);
+			if (total)
+				base = __shfl(base, leader, 64);
+			idx = (int64_t) (base + (unsigned) nbefore);
+		}
+		if (!own)
+			continue;
+		suppkey[idx] = sk;
+		nation[idx] = gg_s_nationkey(seed, sk);
+	}
+}
+
+hipError_t
+launch_gen_supplier(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
+		    int nseg, int seg, int64_t *suppkey, uint8_t *nation,
+		    unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_gen_supplier, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, seed, row_lo, n, nseg, seg, suppkey, nation,
+			   out_count);
 	return hipGetLastError();
 }
 

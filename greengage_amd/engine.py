@@ -36,8 +36,9 @@ class _ColumnDesc(ctypes.Structure):
 
 class _PipelineDesc(ctypes.Structure):
     _fields_ = [("kind", ctypes.c_int), ("lineitem", I32), ("orders", I32),
-                ("customer", I32), ("cutoff_date", I32), ("mktsegment", I8),
-                ("limit_k", I64)]
+                ("customer", I32), ("supplier", I32), ("nation", I32),
+                ("cutoff_date", I32), ("cutoff_hi", I32), ("mktsegment", I8),
+                ("regionkey", I8), ("limit_k", I64)]
 
 
 class KernelStat(ctypes.Structure):
@@ -47,7 +48,14 @@ class KernelStat(ctypes.Structure):
 
 
 COLTYPE = {"int64": 0, "int32": 1, "dec64": 2, "char1": 3}
-PIPE_Q1, PIPE_Q3, PIPE_SUMPRICE = 1, 2, 3
+PIPE_Q1, PIPE_Q3, PIPE_SUMPRICE, PIPE_Q5 = 1, 2, 3, 4
+
+# n_name strings in nationkey order (reference fixture nation.csv)
+NATION_NAMES = [
+    "ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+    "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ", "JAPAN",
+    "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU", "CHINA", "ROMANIA",
+    "SAUDI ARABIA", "VIETNAM", "RUSSIA", "UNITED KINGDOM", "UNITED STATES"]
 
 _EPOCH = datetime.date(2000, 1, 1)
 
@@ -83,6 +91,8 @@ def _load():
     lib.gg_engine_register_synth.argtypes = [ctypes.c_char_p, U64, I64,
                                              ctypes.POINTER(I32)]
     lib.gg_engine_table_nrows.argtypes = [I32, ctypes.POINTER(I64)]
+    lib.gg_engine_fetch_column.argtypes = [I32, ctypes.c_char_p,
+                                           ctypes.c_void_p, ctypes.c_size_t]
     lib.gg_engine_compile_pipeline.argtypes = [
         ctypes.POINTER(_PipelineDesc), ctypes.POINTER(I32)]
     lib.gg_engine_execute.argtypes = [I32, ctypes.c_void_p, ctypes.c_size_t,
@@ -183,10 +193,13 @@ class Engine:
 
     # ---- pipelines ----
     def compile(self, kind, lineitem=-1, orders=-1, customer=-1,
-                cutoff_date=0, mktsegment=0, limit_k=10):
+                supplier=-1, nation=-1, cutoff_date=0, cutoff_hi=0,
+                mktsegment=0, regionkey=0, limit_k=10):
         d = _PipelineDesc(kind=kind, lineitem=lineitem, orders=orders,
-                          customer=customer, cutoff_date=cutoff_date,
-                          mktsegment=mktsegment, limit_k=limit_k)
+                          customer=customer, supplier=supplier,
+                          nation=nation, cutoff_date=cutoff_date,
+                          cutoff_hi=cutoff_hi, mktsegment=mktsegment,
+                          regionkey=regionkey, limit_k=limit_k)
         h = I32()
         _check(lib().gg_engine_compile_pipeline(ctypes.byref(d),
                                                 ctypes.byref(h)), "compile")
@@ -252,6 +265,25 @@ class Engine:
                                                signed=True),
             })
         return rows, hdr
+
+    def execute_q5(self, p):
+        raw = self.execute_raw(p, 8 + 25 * 32)
+        n_out = int.from_bytes(raw[0:8], "little", signed=True)
+        rows = []
+        off = 8
+        for _ in range(n_out):
+            b = raw[off:off + 32]
+            off += 32
+            nk = int.from_bytes(b[0:4], "little", signed=True)
+            rows.append({
+                "nationkey": nk,
+                "n_name": NATION_NAMES[nk],
+                "count": int.from_bytes(b[8:16], "little", signed=True),
+                "revenue4": _i128(int.from_bytes(b[16:24], "little"),
+                                  int.from_bytes(b[24:32], "little",
+                                                 signed=True)),
+            })
+        return rows
 
     def execute_sumprice(self, p):
         raw = self.execute_raw(p, 16)

@@ -136,7 +136,11 @@ typedef enum gg_pipeline_kind
 	 * → sort(rev desc, odate) limit k  [+ Motion redistributes] */
 	GG_PIPE_Q3 = 2,
 	/* scan(lineitem) → filter(shipdate<c) → agg(sum(price),count) */
-	GG_PIPE_SUMPRICE = 3
+	GG_PIPE_SUMPRICE = 3,
+	/* Q5 (mpph5): customer⋈orders⋈lineitem⋈supplier⋈nation⋈region →
+	 * hashagg(n_name) → sort(revenue desc); region/nation are the
+	 * broadcast-Motion dims */
+	GG_PIPE_Q5 = 4
 } gg_pipeline_kind;
 
 typedef struct gg_pipeline_desc
@@ -145,8 +149,12 @@ typedef struct gg_pipeline_desc
 	gg_table lineitem;	/* tables by role; -1 if unused */
 	gg_table orders;
 	gg_table customer;
-	int32_t cutoff_date;	/* the qual's date constant (DateADT) */
+	gg_table supplier;	/* Q5 */
+	gg_table nation;	/* Q5: (nationkey, regionkey) dim */
+	int32_t cutoff_date;	/* date qual lower constant (DateADT) */
+	int32_t cutoff_hi;	/* Q5: o_orderdate < cutoff_hi */
 	uint8_t mktsegment;	/* Q3: dict code of c_mktsegment literal */
+	uint8_t regionkey;	/* Q5: r_name literal resolved to its key */
 	int64_t limit_k;	/* Q3: LIMIT bound (nodeSort.c:143) */
 } gg_pipeline_desc;
 

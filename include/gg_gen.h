@@ -50,6 +50,7 @@
 #define GG_TAB_LINEITEM 0x4c49ull	/* 'LI' */
 #define GG_TAB_ORDERS   0x4f52ull	/* 'OR' */
 #define GG_TAB_CUSTOMER 0x4355ull	/* 'CU' */
+#define GG_TAB_SUPPLIER 0x5355ull	/* 'SU' */
 
 #define GG_MKTSEG_MACHINERY 2
 
@@ -96,6 +97,24 @@ GG_GEN_HOSTDEV uint32_t gg_rnd_range(uint64_t seed, uint64_t tab,
 GG_GEN_HOSTDEV int64_t gg_n_customers(int64_t sf) { return 150000 * sf; }
 GG_GEN_HOSTDEV int64_t gg_n_orders(int64_t sf)    { return 1500000 * sf; }
 GG_GEN_HOSTDEV int64_t gg_n_lineitem(int64_t sf)  { return 6000000 * sf; }
+GG_GEN_HOSTDEV int64_t gg_n_suppliers(int64_t sf) { return 10000 * sf; }
+
+/* ---- nation/region: the 25 TPC-H nations (n_nationkey → n_regionkey),
+ * verified against the reference fixture
+ * src/test/regress/data/nation.csv / region.csv ---- */
+#define GG_NNATIONS 25
+#define GG_NREGIONS 5
+GG_GEN_HOSTDEV int32_t gg_nation_region(int32_t nationkey)
+{
+	/* ALGERIA..UNITED STATES in nationkey order */
+	const uint8_t reg[GG_NNATIONS] = {
+		0, 1, 1, 1, 4, 0, 3, 3, 2, 2, 4, 4, 2, 4, 0,
+		0, 0, 1, 2, 3, 4, 2, 3, 3, 1
+	};
+
+	return (nationkey >= 0 && nationkey < GG_NNATIONS)
+		? (int32_t) reg[nationkey] : -1;
+}
 
 /* date-range constants are generated, never hardcoded */
 GG_GEN_HOSTDEV int32_t gg_orderdate_lo(void) { return gg_pgdate(1992, 1, 1); }
@@ -130,6 +149,27 @@ GG_GEN_HOSTDEV uint8_t gg_c_mktsegment(uint64_t seed, int64_t custkey)
 {
 	return (uint8_t) gg_rnd_range(seed, GG_TAB_CUSTOMER,
 				      (uint64_t) custkey, 0, 5);
+}
+
+GG_GEN_HOSTDEV uint8_t gg_c_nationkey(uint64_t seed, int64_t custkey)
+{
+	return (uint8_t) gg_rnd_range(seed, GG_TAB_CUSTOMER,
+				      (uint64_t) custkey, 1, GG_NNATIONS);
+}
+
+/* ---- supplier ---- */
+GG_GEN_HOSTDEV uint8_t gg_s_nationkey(uint64_t seed, int64_t suppkey)
+{
+	return (uint8_t) gg_rnd_range(seed, GG_TAB_SUPPLIER,
+				      (uint64_t) suppkey, 0, GG_NNATIONS);
+}
+
+/* lineitem's supplier (slot 7; dbgen spreads uniformly over suppliers) */
+GG_GEN_HOSTDEV int64_t gg_l_suppkey(uint64_t seed, int64_t row, int64_t sf)
+{
+	return 1 + (int64_t) gg_rnd_range(seed, GG_TAB_LINEITEM,
+					  (uint64_t) row, 7,
+					  (uint32_t) gg_n_suppliers(sf));
 }
 
 /* ---- lineitem (row i, 0-based; orderkey = i/4 + 1) ---- */

@@ -59,6 +59,23 @@ typedef struct gg_q3_result_hdr
 } gg_q3_result_hdr;
 /* arena: gg_q3_result_hdr then n_out × gg_q3_result_row */
 
+/* GG_PIPE_Q5: rows ordered by revenue DESC (ties: n_name ASC, the
+ * shared refinement); only nations with count > 0 are emitted */
+typedef struct gg_q5_result_row
+{
+	int32_t nationkey;
+	int32_t _pad;
+	int64_t count;
+	uint64_t rev_lo;	/* scale 4 int128 */
+	int64_t rev_hi;
+} gg_q5_result_row;
+
+typedef struct gg_q5_result_hdr
+{
+	int64_t n_out;
+} gg_q5_result_hdr;
+/* arena: gg_q5_result_hdr then n_out × gg_q5_result_row */
+
 /* GG_PIPE_SUMPRICE */
 typedef struct gg_sumprice_result
 {

@@ -167,6 +167,7 @@ def load_lineitem_small():
         f = line.rstrip("\n").split("|")
         rows.append({
             "orderkey": int(f[0]),
+            "suppkey": int(f[2]),
             "quantity": Decimal(f[4]),
             "extendedprice": Decimal(f[5]),
             "discount": Decimal(f[6]),
@@ -290,6 +291,64 @@ def make_q3_small():
     }
 
 
+# ---------- 4b. Q5 over the small fixtures via Decimal ----------
+
+def load_nation_region():
+    nations = {}
+    for line in open(os.path.join(REF, "src/test/regress/data/nation.csv")):
+        f = line.rstrip("\n").split("|")
+        nations[int(f[0])] = (f[1], int(f[2]))
+    regions = {}
+    for line in open(os.path.join(REF, "src/test/regress/data/region.csv")):
+        f = line.rstrip("\n").split("|")
+        regions[f[1]] = int(f[0])
+    return nations, regions
+
+
+def make_q5_small():
+    # mpph5: r_name='AMERICA', o_orderdate in [1997-01-01, 1998-01-01)
+    nations, regions = load_nation_region()
+    regionkey = regions["AMERICA"]
+    date_lo, date_hi = pgdate("1997-01-01"), pgdate("1998-01-01")
+    cust_nation = {}
+    for line in open(os.path.join(REF, "src/test/regress/data/customer.csv")):
+        f = line.rstrip("\n").split("|")
+        cust_nation[int(f[0])] = int(f[3])
+    supp_nation = {}
+    for line in open(os.path.join(REF, "src/test/regress/data/supplier.csv")):
+        f = line.rstrip("\n").split("|")
+        supp_nation[int(f[0])] = int(f[3])
+    onation = {}
+    for name in ("order_small.csv", "order.csv"):
+        for line in open(os.path.join(REF, "src/test/regress/data", name)):
+            f = line.rstrip("\n").split("|")
+            odate = pgdate(f[4])
+            if date_lo <= odate < date_hi:
+                onation[int(f[0])] = cust_nation[int(f[1])]
+    rev = {}
+    cnt = {}
+    for r in load_lineitem_small():
+        on = onation.get(r["orderkey"])
+        if on is None:
+            continue
+        sn = supp_nation.get(r["suppkey"])
+        if sn != on:
+            continue
+        if nations[sn][1] != regionkey:
+            continue
+        rev[sn] = rev.get(sn, Decimal(0)) + \
+            r["extendedprice"] * (1 - r["discount"])
+        cnt[sn] = cnt.get(sn, 0) + 1
+    rows = [{"nationkey": n, "n_name": nations[n][0],
+             "revenue4": int(v.scaleb(4)),
+             "revenue": num_string(int(v.scaleb(4)), 4),
+             "count": cnt[n]} for n, v in rev.items()]
+    rows.sort(key=lambda r: (-r["revenue4"], r["n_name"]))
+    return {"regionkey": regionkey, "date_lo": date_lo, "date_hi": date_hi,
+            "nation_region": [nations[n][1] for n in range(25)],
+            "rows": rows}
+
+
 # ---------- 5. generator freeze vectors ----------
 
 def make_gen_vectors():
@@ -297,11 +356,14 @@ def make_gen_vectors():
     li2 = pyoracle.gen_lineitem(42, 5999936, 6000000)   # tail of SF1
     od = pyoracle.gen_orders(42, 1, 0, 32)
     cu = pyoracle.gen_customer(42, 0, 32)
+    su = pyoracle.gen_supplier(42, 0, 32)
+    lsk = pyoracle.gen_l_suppkey(42, 1, 0, 64)
 
     def cols(d):
         return {k: v.tolist() for k, v in d.items()}
     return {"seed": 42, "lineitem_head": cols(li), "lineitem_sf1_tail": cols(li2),
-            "orders_head_sf1": cols(od), "customer_head": cols(cu)}
+            "orders_head_sf1": cols(od), "customer_head": cols(cu),
+            "supplier_head": cols(su), "l_suppkey_head_sf1": lsk.tolist()}
 
 
 # ---------- 6. encoded small-fixture input columns (travel to GPU box) ----------
@@ -319,14 +381,24 @@ def make_small_inputs():
         "li_rflag": np.array([ord(r["returnflag"]) for r in li], np.uint8),
         "li_lstatus": np.array([ord(r["linestatus"]) for r in li], np.uint8),
     }
-    ck, seg = [], []
+    arr["li_suppkey"] = np.array([r["suppkey"] for r in li], np.int64)
+    ck, seg, cnat = [], [], []
     for line in open(os.path.join(REF, "src/test/regress/data/customer.csv")):
         f = line.rstrip("\n").split("|")
         ck.append(int(f[0]))
         seg.append(1 if f[6] == "MACHINERY" else 0)
+        cnat.append(int(f[3]))
     arr["c_custkey"] = np.array(ck, np.int64)
     # encode segment as the engine's dict code: MACHINERY = GG_MKTSEG_MACHINERY
     arr["c_mktseg"] = np.array([2 if s else 0 for s in seg], np.uint8)
+    arr["c_nationkey"] = np.array(cnat, np.uint8)
+    sk, snat = [], []
+    for line in open(os.path.join(REF, "src/test/regress/data/supplier.csv")):
+        f = line.rstrip("\n").split("|")
+        sk.append(int(f[0]))
+        snat.append(int(f[3]))
+    arr["s_suppkey"] = np.array(sk, np.int64)
+    arr["s_nationkey"] = np.array(snat, np.uint8)
     ok, oc, od, op = [], [], [], []
     for name in ("order_small.csv", "order.csv"):
         for line in open(os.path.join(REF, "src/test/regress/data", name)):
@@ -353,6 +425,7 @@ def main():
         "bb_mpph_pins.json": parse_bb_mpph(),
         "q1_small.json": make_q1_small(),
         "q3_small.json": make_q3_small(),
+        "q5_small.json": make_q5_small(),
         "gen_vectors.json": make_gen_vectors(),
     }
     for name, data in fixtures.items():

@@ -673,6 +673,191 @@ gg_oracle_q3_synth(uint64_t seed, int64_t sf, int32_t cutoff_date, int64_t k,
 	return 0;
 }
 
+/* ---------------- Q5 (mpph5 semantics; see oracle.h) ---------------- */
+
+#define NATION_ABSENT 255
+
+static void
+q5_lineitem_update(int64_t price_c, int64_t disc_c, uint8_t onation,
+		   uint8_t snation, const int32_t *nation_region,
+		   int32_t regionkey, i128 rev[25], int64_t cnt[25])
+{
+	if (onation == NATION_ABSENT)
+		return;
+	if (snation != onation)	/* c_nationkey = s_nationkey */
+		return;
+	if (snation >= 25 || nation_region[snation] != regionkey)
+		return;
+	rev[snation] += (i128) (price_c * (100 - disc_c));
+	cnt[snation] += 1;
+}
+
+static void
+q5_out(const i128 rev[25], const int64_t cnt[25], gg_q5_group out[25])
+{
+	for (int n = 0; n < 25; n++)
+	{
+		out[n].count = cnt[n];
+		split128(rev[n], &out[n].rev_lo, &out[n].rev_hi);
+	}
+}
+
+int
+gg_oracle_q5_synth(uint64_t seed, int64_t sf, int32_t regionkey,
+		   int32_t date_lo, int32_t date_hi, gg_q5_group out[25],
+		   int nthreads)
+{
+	int64_t nord = gg_n_orders(sf);
+	int64_t nli = gg_n_lineitem(sf);
+	uint8_t *onation;
+	int32_t nation_region[25];
+	i128 rev[25] = {0};
+	int64_t cnt[25] = {0};
+
+	for (int n = 0; n < 25; n++)
+		nation_region[n] = gg_nation_region(n);
+	set_threads(nthreads);
+	onation = malloc((size_t) nord + 1);
+	if (!onation)
+		return -1;
+	memset(onation, NATION_ABSENT, (size_t) nord + 1);
+
+	/* orders ⋈ customer: FK always matches; date range filter */
+#pragma omp parallel for schedule(static)
+	for (int64_t okey = 1; okey <= nord; okey++)
+	{
+		int32_t odate = gg_o_orderdate(seed, okey);
+
+		if (odate < date_lo || odate >= date_hi)
+			continue;
+		onation[okey] =
+			gg_c_nationkey(seed, gg_o_custkey(seed, okey, sf));
+	}
+
+#pragma omp parallel
+	{
+		i128 lrev[25] = {0};
+		int64_t lcnt[25] = {0};
+
+#pragma omp for schedule(static)
+		for (int64_t i = 0; i < nli; i++)
+		{
+			gg_lineitem_row r;
+			int64_t sk;
+
+			gg_gen_lineitem(seed, i, &r);
+			if (onation[r.l_orderkey] == NATION_ABSENT)
+				continue;
+			sk = gg_l_suppkey(seed, i, sf);
+			q5_lineitem_update(r.l_extendedprice_c, r.l_discount_c,
+					   onation[r.l_orderkey],
+					   gg_s_nationkey(seed, sk),
+					   nation_region, regionkey, lrev,
+					   lcnt);
+		}
+#pragma omp critical
+		for (int n = 0; n < 25; n++)
+		{
+			rev[n] += lrev[n];
+			cnt[n] += lcnt[n];
+		}
+	}
+
+	free(onation);
+	q5_out(rev, cnt, out);
+	return 0;
+}
+
+int
+gg_oracle_q5_arrays(const int64_t *c_custkey, const uint8_t *c_nation,
+		    int64_t ncust, const int64_t *o_orderkey,
+		    const int64_t *o_custkey, const int32_t *o_orderdate,
+		    int64_t nord, const int64_t *l_orderkey,
+		    const int64_t *l_suppkey, const int64_t *l_price_c,
+		    const int64_t *l_disc_c, int64_t nli,
+		    const int64_t *s_suppkey, const uint8_t *s_nation,
+		    int64_t nsupp, const int32_t *nation_region,
+		    int32_t regionkey, int32_t date_lo, int32_t date_hi,
+		    gg_q5_group out[25], int nthreads)
+{
+	int64_t max_ck = 0, max_ok = 0, max_sk = 0;
+	uint8_t *ck_nation, *onation, *sk_nation;
+	i128 rev[25] = {0};
+	int64_t cnt[25] = {0};
+
+	set_threads(nthreads);
+	for (int64_t i = 0; i < ncust; i++)
+		if (c_custkey[i] > max_ck)
+			max_ck = c_custkey[i];
+	for (int64_t i = 0; i < nord; i++)
+		if (o_orderkey[i] > max_ok)
+			max_ok = o_orderkey[i];
+	for (int64_t i = 0; i < nsupp; i++)
+		if (s_suppkey[i] > max_sk)
+			max_sk = s_suppkey[i];
+	if (max_ck > 8 * ncust + 16 || max_ok > 8 * nord + 16 ||
+	    max_sk > 8 * nsupp + 16)
+		return -2;
+	ck_nation = malloc((size_t) max_ck + 1);
+	onation = malloc((size_t) max_ok + 1);
+	sk_nation = malloc((size_t) max_sk + 1);
+	if (!ck_nation || !onation || !sk_nation)
+		return -1;
+	memset(ck_nation, NATION_ABSENT, (size_t) max_ck + 1);
+	memset(onation, NATION_ABSENT, (size_t) max_ok + 1);
+	memset(sk_nation, NATION_ABSENT, (size_t) max_sk + 1);
+
+#pragma omp parallel for schedule(static)
+	for (int64_t i = 0; i < ncust; i++)
+		ck_nation[c_custkey[i]] = c_nation[i];
+#pragma omp parallel for schedule(static)
+	for (int64_t i = 0; i < nsupp; i++)
+		sk_nation[s_suppkey[i]] = s_nation[i];
+#pragma omp parallel for schedule(static)
+	for (int64_t i = 0; i < nord; i++)
+	{
+		if (o_orderdate[i] < date_lo || o_orderdate[i] >= date_hi)
+			continue;
+		if (o_custkey[i] > max_ck)
+			continue;
+		onation[o_orderkey[i]] = ck_nation[o_custkey[i]];
+	}
+
+#pragma omp parallel
+	{
+		i128 lrev[25] = {0};
+		int64_t lcnt[25] = {0};
+
+#pragma omp for schedule(static)
+		for (int64_t i = 0; i < nli; i++)
+		{
+			int64_t ok = l_orderkey[i];
+			int64_t sk = l_suppkey[i];
+
+			if (ok > max_ok || sk > max_sk)
+				continue;
+			if (sk_nation[sk] == NATION_ABSENT)
+				continue;
+			q5_lineitem_update(l_price_c[i], l_disc_c[i],
+					   onation[ok], sk_nation[sk],
+					   nation_region, regionkey, lrev,
+					   lcnt);
+		}
+#pragma omp critical
+		for (int n = 0; n < 25; n++)
+		{
+			rev[n] += lrev[n];
+			cnt[n] += lcnt[n];
+		}
+	}
+
+	free(ck_nation);
+	free(onation);
+	free(sk_nation);
+	q5_out(rev, cnt, out);
+	return 0;
+}
+
 /* ---------------- config 1: sum(price) where shipdate < cutoff ---------------- */
 
 int
@@ -781,7 +966,7 @@ gg_oracle_gen_orders(uint64_t seed, int64_t sf, int64_t row_lo,
 
 void
 gg_oracle_gen_customer(uint64_t seed, int64_t row_lo, int64_t row_hi,
-		       int64_t *custkey, uint8_t *mktseg)
+		       int64_t *custkey, uint8_t *mktseg, uint8_t *nationkey)
 {
 #pragma omp parallel for schedule(static)
 	for (int64_t i = row_lo; i < row_hi; i++)
@@ -791,5 +976,31 @@ gg_oracle_gen_customer(uint64_t seed, int64_t row_lo, int64_t row_hi,
 
 		custkey[j] = ck;
 		mktseg[j] = gg_c_mktsegment(seed, ck);
+		if (nationkey)
+			nationkey[j] = gg_c_nationkey(seed, ck);
 	}
+}
+
+void
+gg_oracle_gen_supplier(uint64_t seed, int64_t row_lo, int64_t row_hi,
+		       int64_t *suppkey, uint8_t *nationkey)
+{
+#pragma omp parallel for schedule(static)
+	for (int64_t i = row_lo; i < row_hi; i++)
+	{
+		int64_t sk = i + 1;
+		int64_t j = i - row_lo;
+
+		suppkey[j] = sk;
+		nationkey[j] = gg_s_nationkey(seed, sk);
+	}
+}
+
+void
+gg_oracle_gen_l_suppkey(uint64_t seed, int64_t sf, int64_t row_lo,
+			int64_t row_hi, int64_t *suppkey)
+{
+#pragma omp parallel for schedule(static)
+	for (int64_t i = row_lo; i < row_hi; i++)
+		suppkey[i - row_lo] = gg_l_suppkey(seed, i, sf);
 }

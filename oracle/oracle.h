@@ -130,6 +130,37 @@ int gg_oracle_q3_synth(uint64_t seed, int64_t sf, int32_t cutoff_date,
 		       int64_t k, gg_q3_row *out_topk, gg_q3_result *res,
 		       int nthreads);
 
+/* ---- Q5 (mpph5): 6-way join + group-by nation + order by revenue ----
+ * select n_name, sum(l_extendedprice*(1-l_discount)) from customer,
+ * orders, lineitem, supplier, nation, region where c_custkey=o_custkey
+ * and l_orderkey=o_orderkey and l_suppkey=s_suppkey and
+ * c_nationkey=s_nationkey and s_nationkey=n_nationkey and
+ * n_regionkey=r_regionkey and r_name=<region> and o_orderdate in
+ * [date_lo, date_hi) group by n_name order by revenue desc
+ * (input/bb_mpph.source mpph5; dates 1997-01-01 + 1 year, AMERICA).
+ * out[nationkey]: count + exact scale-4 revenue (int128). */
+typedef struct gg_q5_group
+{
+	int64_t count;
+	uint64_t rev_lo;
+	int64_t rev_hi;
+} gg_q5_group;
+
+int gg_oracle_q5_synth(uint64_t seed, int64_t sf, int32_t regionkey,
+		       int32_t date_lo, int32_t date_hi,
+		       gg_q5_group out[25], int nthreads);
+
+int gg_oracle_q5_arrays(
+	const int64_t *c_custkey, const uint8_t *c_nation, int64_t ncust,
+	const int64_t *o_orderkey, const int64_t *o_custkey,
+	const int32_t *o_orderdate, int64_t nord,
+	const int64_t *l_orderkey, const int64_t *l_suppkey,
+	const int64_t *l_price_c, const int64_t *l_disc_c, int64_t nli,
+	const int64_t *s_suppkey, const uint8_t *s_nation, int64_t nsupp,
+	const int32_t *nation_region /* [25] */, int32_t regionkey,
+	int32_t date_lo, int32_t date_hi, gg_q5_group out[25],
+	int nthreads);
+
 /* ---- BASELINE config 1: sum(l_extendedprice) where l_shipdate < cutoff ---- */
 int gg_oracle_sumprice_arrays(const int32_t *shipdate, const int64_t *price_c,
 			      int64_t nrows, int32_t cutoff,
@@ -160,7 +191,12 @@ void gg_oracle_gen_orders(uint64_t seed, int64_t sf, int64_t row_lo,
 			  int64_t row_hi, int64_t *orderkey, int64_t *custkey,
 			  int32_t *orderdate, int32_t *shippriority);
 void gg_oracle_gen_customer(uint64_t seed, int64_t row_lo, int64_t row_hi,
-			    int64_t *custkey, uint8_t *mktseg);
+			    int64_t *custkey, uint8_t *mktseg,
+			    uint8_t *nationkey);
+void gg_oracle_gen_supplier(uint64_t seed, int64_t row_lo, int64_t row_hi,
+			    int64_t *suppkey, uint8_t *nationkey);
+void gg_oracle_gen_l_suppkey(uint64_t seed, int64_t sf, int64_t row_lo,
+			     int64_t row_hi, int64_t *suppkey);
 
 #ifdef __cplusplus
 }

@@ -54,6 +54,18 @@ class Q3Row(ctypes.Structure):
         }
 
 
+class Q5Group(ctypes.Structure):
+    _fields_ = [
+        ("count", ctypes.c_int64),
+        ("rev_lo", ctypes.c_uint64),
+        ("rev_hi", ctypes.c_int64),
+    ]
+
+    def as_dict(self):
+        return {"count": self.count,
+                "revenue4": (self.rev_hi << 64) | self.rev_lo}
+
+
 class Q3Result(ctypes.Structure):
     _fields_ = [
         ("n_out", ctypes.c_int64),
@@ -159,7 +171,21 @@ def lib():
             _P_I32, _P_U8, _P_U8]
         L.gg_oracle_gen_orders.argtypes = [
             U64, I64, I64, I64, _P_I64, _P_I64, _P_I32, _P_I32]
-        L.gg_oracle_gen_customer.argtypes = [U64, I64, I64, _P_I64, _P_U8]
+        L.gg_oracle_gen_customer.argtypes = [U64, I64, I64, _P_I64, _P_U8,
+                                             _P_U8]
+        L.gg_oracle_gen_supplier.argtypes = [U64, I64, I64, _P_I64, _P_U8]
+        L.gg_oracle_gen_l_suppkey.argtypes = [U64, I64, I64, I64, _P_I64]
+        L.gg_oracle_q5_synth.restype = ctypes.c_int
+        L.gg_oracle_q5_synth.argtypes = [U64, I64, I32, I32, I32,
+                                         ctypes.POINTER(Q5Group),
+                                         ctypes.c_int]
+        L.gg_oracle_q5_arrays.restype = ctypes.c_int
+        L.gg_oracle_q5_arrays.argtypes = [
+            _P_I64, _P_U8, I64,
+            _P_I64, _P_I64, _P_I32, I64,
+            _P_I64, _P_I64, _P_I64, _P_I64, I64,
+            _P_I64, _P_U8, I64,
+            _P_I32, I32, I32, I32, ctypes.POINTER(Q5Group), ctypes.c_int]
     return _lib
 
 
@@ -327,7 +353,68 @@ def gen_orders(seed, sf, row_lo, row_hi):
 
 def gen_customer(seed, row_lo, row_hi):
     n = row_hi - row_lo
-    cols = dict(custkey=np.empty(n, np.int64), mktseg=np.empty(n, np.uint8))
+    cols = dict(custkey=np.empty(n, np.int64), mktseg=np.empty(n, np.uint8),
+                nationkey=np.empty(n, np.uint8))
     lib().gg_oracle_gen_customer(seed, row_lo, row_hi, cols["custkey"],
-                                 cols["mktseg"])
+                                 cols["mktseg"], cols["nationkey"])
     return cols
+
+
+def gen_supplier(seed, row_lo, row_hi):
+    n = row_hi - row_lo
+    cols = dict(suppkey=np.empty(n, np.int64),
+                nationkey=np.empty(n, np.uint8))
+    lib().gg_oracle_gen_supplier(seed, row_lo, row_hi, cols["suppkey"],
+                                 cols["nationkey"])
+    return cols
+
+
+def gen_l_suppkey(seed, sf, row_lo, row_hi):
+    out = np.empty(row_hi - row_lo, np.int64)
+    lib().gg_oracle_gen_l_suppkey(seed, sf, row_lo, row_hi, out)
+    return out
+
+
+# nation names in nationkey order (reference fixture nation.csv)
+NATION_NAMES = [
+    "ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+    "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ", "JAPAN",
+    "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU", "CHINA", "ROMANIA",
+    "SAUDI ARABIA", "VIETNAM", "RUSSIA", "UNITED KINGDOM", "UNITED STATES"]
+NATION_REGION = [0, 1, 1, 1, 4, 0, 3, 3, 2, 2, 4, 4, 2, 4, 0,
+                 0, 0, 1, 2, 3, 4, 2, 3, 3, 1]
+REGION_NAMES = ["AFRICA", "AMERICA", "ASIA", "EUROPE", "MIDDLE EAST"]
+
+
+def q5_synth(seed, sf, regionkey, date_lo, date_hi, nthreads=0):
+    out = (Q5Group * 25)()
+    rc = lib().gg_oracle_q5_synth(seed, sf, regionkey, date_lo, date_hi,
+                                  out, nthreads)
+    assert rc == 0, rc
+    return [g.as_dict() for g in out]
+
+
+def q5_arrays(c_custkey, c_nation, o_orderkey, o_custkey, o_orderdate,
+              l_orderkey, l_suppkey, l_price, l_disc, s_suppkey, s_nation,
+              regionkey, date_lo, date_hi, nation_region=None, nthreads=0):
+    out = (Q5Group * 25)()
+    nr = np.array(nation_region if nation_region is not None
+                  else NATION_REGION, np.int32)
+    rc = lib().gg_oracle_q5_arrays(
+        c_custkey, c_nation, len(c_custkey),
+        o_orderkey, o_custkey, o_orderdate, len(o_orderkey),
+        l_orderkey, l_suppkey, l_price, l_disc, len(l_orderkey),
+        s_suppkey, s_nation, len(s_suppkey),
+        nr, regionkey, date_lo, date_hi, out, nthreads)
+    assert rc == 0, rc
+    return [g.as_dict() for g in out]
+
+
+def q5_rows(groups):
+    """Occupied groups as output rows ordered by revenue desc (the
+    query's ORDER BY; ties refined by n_name asc — both sides use it)."""
+    rows = [{"nationkey": n, "n_name": NATION_NAMES[n],
+             "revenue4": g["revenue4"], "count": g["count"]}
+            for n, g in enumerate(groups) if g["count"]]
+    rows.sort(key=lambda r: (-r["revenue4"], r["n_name"]))
+    return rows

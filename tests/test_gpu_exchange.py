@@ -73,3 +73,43 @@ def test_q1_steady_state_reuse():
             assert eng.execute_q3(p3) == r0
     finally:
         eng.shutdown()
+
+
+def test_q5_exchange_loopback_bitexact():
+    """Q5's broadcast-Motion (supplier) + two redistribute legs as a
+    world-1 self-loopback must equal the fused path."""
+    import os
+    from greengage_amd import Engine, PGDate
+    from greengage_amd.engine import PIPE_Q5
+
+    eng = Engine(device=0, n_segments=1, segment_id=0)
+    try:
+        eng.comm_init(eng.comm_id())
+        li = eng.register_synth("lineitem", seed=42, sf=1)
+        od = eng.register_synth("orders", seed=42, sf=1)
+        cu = eng.register_synth("customer", seed=42, sf=1)
+        su = eng.register_synth("supplier", seed=42, sf=1)
+        na = eng.register_synth("nation", seed=42, sf=1)
+        lo, hi = PGDate("1997-01-01"), PGDate("1998-01-01")
+        p = eng.compile(PIPE_Q5, lineitem=li, orders=od, customer=cu,
+                        supplier=su, nation=na, cutoff_date=lo,
+                        cutoff_hi=hi, regionkey=1)
+        rows_f = eng.execute_q5(p)
+        os.environ["GG_FORCE_EXCHANGE"] = "1"
+        try:
+            p_x = eng.compile(PIPE_Q5, lineitem=li, orders=od, customer=cu,
+                              supplier=su, nation=na, cutoff_date=lo,
+                              cutoff_hi=hi, regionkey=1)
+            rows_x = eng.execute_q5(p_x)
+            rows_x2 = eng.execute_q5(p_x)
+        finally:
+            del os.environ["GG_FORCE_EXCHANGE"]
+        assert rows_x == rows_f
+        assert rows_x2 == rows_f
+        expect = __import__("pyoracle").q5_rows(
+            __import__("pyoracle").q5_synth(42, 1, 1, lo, hi))
+        assert [(r["nationkey"], r["revenue4"], r["count"])
+                for r in rows_f] == \
+            [(r["nationkey"], r["revenue4"], r["count"]) for r in expect]
+    finally:
+        eng.shutdown()

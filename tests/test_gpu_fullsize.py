@@ -69,3 +69,21 @@ def test_sumprice_fullsize(eng):
     p = eng.compile(PIPE_SUMPRICE, lineitem=li, cutoff_date=cutoff)
     assert eng.execute_sumprice(p) == \
         pyoracle.sumprice_synth(42, SF, cutoff)
+
+
+def test_q5_fullsize_bitexact(eng):
+    from greengage_amd import PGDate
+    from greengage_amd.engine import PIPE_Q5
+    li = eng.register_synth("lineitem", seed=42, sf=SF)
+    od = eng.register_synth("orders", seed=42, sf=SF)
+    cu = eng.register_synth("customer", seed=42, sf=SF)
+    su = eng.register_synth("supplier", seed=42, sf=SF)
+    na = eng.register_synth("nation", seed=42, sf=SF)
+    lo, hi = PGDate("1997-01-01"), PGDate("1998-01-01")
+    p = eng.compile(PIPE_Q5, lineitem=li, orders=od, customer=cu,
+                    supplier=su, nation=na, cutoff_date=lo, cutoff_hi=hi,
+                    regionkey=1)
+    rows = eng.execute_q5(p)
+    expect = pyoracle.q5_rows(pyoracle.q5_synth(42, SF, 1, lo, hi))
+    assert [(r["nationkey"], r["revenue4"], r["count"]) for r in rows] == \
+        [(r["nationkey"], r["revenue4"], r["count"]) for r in expect]
