@@ -170,7 +170,7 @@ def main():
 
     extra = {}
     if not args.skip_q3:
-        log("[bench] Q3 extra (untimed region)...")
+        log("[bench] Q3/Q5 extra (untimed region)...")
         od = eng.register_synth("orders", seed=SEED, sf=SF)
         cu = eng.register_synth("customer", seed=SEED, sf=SF)
         p_q3 = eng.compile(PIPE_Q3, lineitem=li, orders=od, customer=cu,
@@ -196,6 +196,35 @@ def main():
             "q3_n_join_rows": hdr_q3["n_join_rows"],
             "q3_kernel_stats": eng.stats(p_q3),
         }
+
+        # Q5 (configs[4]-shaped; SF given by --sf): broadcast Motion +
+        # 6-way join + nation group-by
+        from greengage_amd.engine import PIPE_Q5
+        su = eng.register_synth("supplier", seed=SEED, sf=SF)
+        na = eng.register_synth("nation", seed=SEED, sf=SF)
+        p_q5 = eng.compile(PIPE_Q5, lineitem=li, orders=od, customer=cu,
+                           supplier=su, nation=na,
+                           cutoff_date=PGDate("1997-01-01"),
+                           cutoff_hi=PGDate("1998-01-01"), regionkey=1)
+        eng.execute_q5(p_q5)
+        barrier_sync()
+        t0 = time.perf_counter()
+        q5_steps = 3
+        for _ in range(q5_steps):
+            rows_q5 = eng.execute_q5(p_q5)
+        barrier_sync()
+        q5_el = time.perf_counter() - t0
+        if n > 1:
+            t = torch.tensor([q5_el], dtype=torch.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            q5_el = float(t.item())
+        q5_total = (6_000_000 + 1_500_000 + 150_000 + 10_000) * SF
+        extra.update({
+            "q5_rows_per_s": q5_total * q5_steps / q5_el,
+            "q5_ms_per_step": q5_el / q5_steps * 1000.0,
+            "q5_n_out": len(rows_q5),
+            "q5_kernel_stats": eng.stats(p_q5),
+        })
 
     cpu_baseline = None
     if rank == 0 and n == 1 and not args.skip_cpu_baseline:
