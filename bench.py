@@ -102,9 +102,12 @@ def main():
         dist.init_process_group("gloo")
 
     assert torch.cuda.is_available(), "bench requires an MI355X"
-    torch.cuda.set_device(local_rank)
+    # normally 1 rank per GPU; modulo lets oversubscribed debug runs
+    # (2 ranks on a 1-GPU box) exercise the multi-rank RCCL paths
+    device = local_rank % torch.cuda.device_count()
+    torch.cuda.set_device(device)
 
-    eng = Engine(device=local_rank, n_segments=n, segment_id=rank)
+    eng = Engine(device=device, n_segments=n, segment_id=rank)
     if n > 1:
         obj = [eng.comm_id() if rank == 0 else None]
         dist.broadcast_object_list(obj, src=0)

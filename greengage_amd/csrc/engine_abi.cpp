@@ -1891,6 +1891,91 @@ gg_engine_stats(gg_pipeline h, gg_kernel_stat *out, int cap, int *out_n)
 	return GG_OK;
 }
 
+
+/* ---------------- general radix sort (ABI surface) ---------------- */
+
+extern "C" gg_status
+gg_engine_radix_sort_u64(uint64_t *keys, uint64_t *payload_or_null,
+			 int64_t n, int key_bytes, int descending)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!keys || n < 0 || n >= (int64_t) 1 << 32 || key_bytes < 1 ||
+	    key_bytes > 8)
+		return fail(GG_EINVAL, "bad radix_sort args");
+	if (n == 0)
+		return GG_OK;
+
+	/* stable descending = stable ascending on bit-inverted keys
+	 * (inverted only within the sorted byte range so higher bytes
+	 * stay untouched) */
+	uint64_t inv = 0;
+
+	if (descending)
+	{
+		inv = (key_bytes == 8) ? ~0ull
+			: ((1ull << (8 * key_bytes)) - 1);
+		for (int64_t i = 0; i < n; i++)
+			keys[i] ^= inv;
+	}
+
+	unsigned long long *d[2] = {nullptr, nullptr};
+	unsigned long long *dp[2] = {nullptr, nullptr};
+	unsigned int *hist = nullptr;
+	int nblocks = radix_sort_nblocks(n);
+	bool pay = payload_or_null != nullptr;
+	gg_status st = GG_OK;
+
+	GG_HIP(hipMalloc((void **) &d[0], (size_t) n * 8));
+	GG_HIP(hipMalloc((void **) &d[1], (size_t) n * 8));
+	if (pay)
+	{
+		GG_HIP(hipMalloc((void **) &dp[0], (size_t) n * 8));
+		GG_HIP(hipMalloc((void **) &dp[1], (size_t) n * 8));
+	}
+	GG_HIP(hipMalloc((void **) &hist, (size_t) 256 * nblocks * 4));
+	GG_HIP(hipMemcpy(d[0], keys, (size_t) n * 8, hipMemcpyHostToDevice));
+	if (pay)
+		GG_HIP(hipMemcpy(dp[0], payload_or_null, (size_t) n * 8,
+				 hipMemcpyHostToDevice));
+
+	int cur = 0;
+
+	for (int pass = 0; pass < key_bytes && st == GG_OK; pass++)
+	{
+		hipError_t he = launch_radix_sort_pass(
+			e.stream, d[cur], pay ? dp[cur] : nullptr, n,
+			8 * pass, hist, nblocks, d[1 - cur],
+			pay ? dp[1 - cur] : nullptr);
+
+		if (he != hipSuccess)
+			st = fail(GG_EGPU, "radix pass: %s",
+				  hipGetErrorString(he));
+		cur = 1 - cur;
+	}
+	if (st == GG_OK)
+	{
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(keys, d[cur], (size_t) n * 8,
+				 hipMemcpyDeviceToHost));
+		if (pay)
+			GG_HIP(hipMemcpy(payload_or_null, dp[cur],
+					 (size_t) n * 8,
+					 hipMemcpyDeviceToHost));
+		if (descending)
+			for (int64_t i = 0; i < n; i++)
+				keys[i] ^= inv;
+	}
+	(void) hipFree(d[0]);
+	(void) hipFree(d[1]);
+	(void) hipFree(dp[0]);
+	(void) hipFree(dp[1]);
+	(void) hipFree(hist);
+	return st;
+}
+
 /* ---------------- numeric display (product restatement of
  * numeric.c display + select_div_scale:7144 + round_var) ---------------- */
 

@@ -313,6 +313,14 @@ hipError_t launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n,
 				unsigned long long *offsets,
 				int64_t *oa, int64_t *ob, int64_t *oc);
 
+hipError_t launch_radix_sort_pass(hipStream_t s,
+				  const unsigned long long *keys,
+				  const unsigned long long *pay, int64_t n,
+				  int shift, unsigned int *block_hist,
+				  int nblocks, unsigned long long *out_keys,
+				  unsigned long long *out_pay);
+int radix_sort_nblocks(int64_t n);
+
 }				/* namespace gg */
 
 #endif

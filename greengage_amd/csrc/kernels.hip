@@ -1537,13 +1537,25 @@ void k_probe_lineitem_q5(const int64_t *__restrict__ okey,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
+		/* both keys loaded densely (coalesced), both Bloom filters
+		 * tested before any slot-array probe: sparse exec-masked
+		 * loads mid-kernel cost a full cache line per surviving
+		 * lane, so the cheap rejections come first */
+		int64_t sk = skey[i];
+		int64_t ok = okey[i];
+		uint32_t hsk = gg_hashint8(sk);
+		uint32_t hok = gg_hashint8(ok);
 		unsigned long long snat, onat;
 
-		if (!ht_lookup_b(supp_keys, supp_pay, supp_slots, supp_bloom,
-				 supp_bwords, skey[i], &snat))
+		if (supp_bloom && !bloom_maybe(supp_bloom, supp_bwords, hsk))
 			continue;
-		if (!ht_lookup_b(ord_keys, ord_pay, ord_slots, ord_bloom,
-				 ord_bwords, okey[i], &onat))
+		if (ord_bloom && !bloom_maybe(ord_bloom, ord_bwords, hok))
+			continue;
+		if (!ht_lookup_b(supp_keys, supp_pay, supp_slots, nullptr, 0,
+				 sk, &snat))
+			continue;
+		if (!ht_lookup_b(ord_keys, ord_pay, ord_slots, nullptr, 0,
+				 ok, &onat))
 			continue;
 		if (snat != onat)	/* c_nationkey = s_nationkey */
 			continue;

@@ -103,6 +103,9 @@ def _load():
     lib.gg_engine_comm_init.argtypes = [ctypes.c_void_p]
     lib.gg_engine_numeric_str.argtypes = [U64, I64, ctypes.c_int,
                                           ctypes.c_char_p]
+    lib.gg_engine_radix_sort_u64.argtypes = [ctypes.c_void_p,
+                                             ctypes.c_void_p, I64,
+                                             ctypes.c_int, ctypes.c_int]
     lib.gg_engine_avg_str.argtypes = [U64, I64, ctypes.c_int, I64,
                                       ctypes.c_char_p]
     return lib
@@ -289,6 +292,21 @@ class Engine:
         raw = self.execute_raw(p, 16)
         return (int.from_bytes(raw[0:8], "little", signed=True),
                 int.from_bytes(raw[8:16], "little", signed=True))
+
+    # ---- general sort (ORDER BY operator; LSB radix on GPU) ----
+    @staticmethod
+    def radix_sort(keys, payload=None, key_bytes=8, descending=False):
+        import numpy as np
+        keys = np.ascontiguousarray(keys, dtype=np.uint64)
+        pp = None
+        if payload is not None:
+            payload = np.ascontiguousarray(payload, dtype=np.uint64)
+            assert len(payload) == len(keys)
+            pp = payload.ctypes.data_as(ctypes.c_void_p)
+        _check(lib().gg_engine_radix_sort_u64(
+            keys.ctypes.data_as(ctypes.c_void_p), pp, len(keys),
+            key_bytes, 1 if descending else 0), "radix_sort")
+        return keys, payload
 
     # ---- numeric display (product-side finalize) ----
     @staticmethod

@@ -201,6 +201,15 @@ void gg_engine_numeric_str(uint64_t lo, int64_t hi, int scale, char *buf64);
 void gg_engine_avg_str(uint64_t sum_lo, int64_t sum_hi, int sum_scale,
 		       int64_t count, char *buf64);
 
+/* General ORDER BY operator: stable LSB radix sort of (u64 key, u64
+ * payload) pairs on the GPU (nodeSort.c:48 / tuplesort.c semantics for
+ * unbounded sorts; the LIMIT-k case uses histogram select instead,
+ * mirroring tuplesort.c:1360's bounded-heap switch).  key_bytes limits
+ * the passes (e.g. 4 for int32-derived keys); descending inverts the
+ * key bits.  Host buffers; n < 2^32. */
+gg_status gg_engine_radix_sort_u64(uint64_t *keys, uint64_t *payload_or_null,
+				   int64_t n, int key_bytes, int descending);
+
 /* build info: "gfx950" etc. — lets callers assert the native path */
 const char *gg_engine_build_info(void);
 
