@@ -764,8 +764,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	const int64_t *l_pc = (const int64_t *) li->col("price");
 	const int64_t *l_dc = (const int64_t *) li->col("disc");
 
-	if (!c_ck || !c_ms || !o_ok || !o_dt || !o_pr || !l_ok || !l_sd ||
-	    !l_pc || !l_dc || (!o_ck && exch) || (!o_ck && !exch))
+	if (!c_ck || !c_ms || !o_ok || !o_ck || !o_dt || !o_pr || !l_ok ||
+	    !l_sd || !l_pc || !l_dc)
 		return fail(GG_EINVAL, "Q3: missing column");
 
 	int32_t cutoff = p->desc.cutoff_date;

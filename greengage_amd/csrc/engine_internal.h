@@ -271,8 +271,6 @@ hipError_t launch_probe_lineitem(hipStream_t s, const int64_t *okey,
 hipError_t launch_q3_stats(hipStream_t s, DeviceHashTable ord,
 			   unsigned long long *out4 /* ngroups, revlo,
 						     * revhi, checksum */);
-hipError_t launch_q3_maxrev(hipStream_t s, DeviceHashTable ord,
-			    unsigned long long *out_max);
 hipError_t launch_q3_hist(hipStream_t s, DeviceHashTable ord, int shift,
 			  uint64_t lo_bound, unsigned int *hist64k);
 hipError_t launch_q3_collect(hipStream_t s, DeviceHashTable ord,

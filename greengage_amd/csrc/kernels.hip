@@ -283,18 +283,16 @@ k_gen_lineitem(uint64_t seed, int64_t sf, int64_t row_lo, int64_t n,
 			/* wave-aggregated append (Guideline 12) */
 			unsigned long long mask = __ballot(own);
 			int lane = threadIdx.x & 63;
-			unsigned long long before =
-				mask & ((lane == 63) ? ~0ull >> 1
-					: ((1ull << lane) - 1));
 			int nbefore = __popcll(mask & ((1ull << lane) - 1));
 			int total = __popcll(mask);
+			int leader = __ffsll((long long) mask) - 1;
 			unsigned long long base = 0;
 
-			(void) before;
-			if (lane == __ffsll((long long) mask) - 1 && total)
+			if (total && lane == leader)
 				base = atomicAdd(out_count,
 						 (unsigned long long) total);
-			base = __shfl(base, __ffsll((long long) mask) - 1, 64);
+			if (total)
+				base = __shfl(base, leader, 64);
 			idx = (int64_t) (base + (unsigned) nbefore);
 		}
 		if (!own)
@@ -820,34 +818,6 @@ launch_q3_stats(hipStream_t s, DeviceHashTable ord, unsigned long long *out5)
 	hipLaunchKernelGGL(k_q3_stats, dim3(grid_for((int64_t) ord.nslots)),
 			   dim3(THREADS), 0, s, ord.keys, ord.payload,
 			   ord.rev, ord.nslots, out5);
-	return hipGetLastError();
-}
-
-__global__ void
-k_q3_maxrev(const unsigned long long *__restrict__ tkeys,
-	    const unsigned long long *__restrict__ trev, uint64_t nslots,
-	    unsigned long long *out_max)
-{
-	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
-	unsigned long long m = 0;
-
-	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < nslots; i += stride)
-		if (tkeys[i])
-			m = max(m, trev[i]);
-	for (int off = 32; off; off >>= 1)
-		m = max(m, __shfl_down(m, off, 64));
-	if ((threadIdx.x & 63) == 0 && m)
-		atomicMax(out_max, m);
-}
-
-hipError_t
-launch_q3_maxrev(hipStream_t s, DeviceHashTable ord,
-		 unsigned long long *out_max)
-{
-	hipLaunchKernelGGL(k_q3_maxrev, dim3(grid_for((int64_t) ord.nslots)),
-			   dim3(THREADS), 0, s, ord.keys, ord.rev, ord.nslots,
-			   out_max);
 	return hipGetLastError();
 }
 
