@@ -52,43 +52,63 @@ void k_rs_hist(const unsigned long long *__restrict__ keys, int64_t n,
 		block_hist[(size_t) d * gridDim.x + blockIdx.x] = h[d];
 }
 
-/* single-block exclusive scan of the (digit-major) histogram table */
+/* single-block exclusive scan of the (digit-major) histogram table:
+ * chunks of 1024 swept in order; inside a chunk, wave-level shuffle
+ * scan + cross-wave LDS partials (16 waves) */
 __global__ __launch_bounds__(1024)
 void k_rs_scan(unsigned int *__restrict__ hist, int64_t len)
 {
 	__shared__ unsigned long long carry;
+	__shared__ unsigned int wavesum[16];
 
 	if (threadIdx.x == 0)
 		carry = 0;
 	__syncthreads();
-	/* serial-chunk scan: 1024 threads sweep chunks in order */
 	for (int64_t base = 0; base < len; base += blockDim.x)
 	{
 		int64_t i = base + threadIdx.x;
 		unsigned int v = (i < len) ? hist[i] : 0;
-		/* block-wide exclusive scan of v via LDS */
-		__shared__ unsigned int tmp[1024];
+		int lane = (int) (threadIdx.x & 63);
+		int wave = (int) (threadIdx.x >> 6);
 
-		tmp[threadIdx.x] = v;
-		__syncthreads();
-		unsigned int acc = 0;
+		/* wave-inclusive scan */
+		unsigned int inc = v;
 
-		for (int t = 0; t < (int) threadIdx.x; t++)
-			acc += tmp[t];
-		/* (O(T^2) inside LDS but len is only 256×blocks; this
-		 * scan is <1% of sort time at any realistic size) */
-		if (i < len)
-			hist[i] = (unsigned int) (carry + acc);
-		__syncthreads();
-		if (threadIdx.x == 0)
+		for (int off = 1; off < 64; off <<= 1)
 		{
-			unsigned long long s = 0;
+			unsigned int up = __shfl_up(inc, off, 64);
 
-			for (int t = 0; t < (int) blockDim.x &&
-			     base + t < len; t++)
-				s += tmp[t];
-			carry += s;
+			if (lane >= off)
+				inc += up;
 		}
+		if (lane == 63)
+			wavesum[wave] = inc;
+		__syncthreads();
+		/* wave 0 scans the 16 wave sums */
+		if (threadIdx.x < 16)
+		{
+			unsigned int w = wavesum[threadIdx.x];
+			unsigned int winc = w;
+
+			for (int off = 1; off < 16; off <<= 1)
+			{
+				unsigned int up = __shfl_up(winc, off, 64);
+
+				if ((int) threadIdx.x >= off)
+					winc += up;
+			}
+			wavesum[threadIdx.x] = winc - w;	/* exclusive */
+		}
+		__syncthreads();
+		{
+			unsigned int excl = inc - v + wavesum[wave];
+
+			if (i < len)
+				hist[i] = (unsigned int) (carry + excl);
+		}
+		__syncthreads();
+		if (threadIdx.x == 1023)
+			carry += (unsigned long long) (inc + wavesum[wave]);
 		__syncthreads();
 	}
 }
