@@ -581,14 +581,20 @@ k_build_orders(const int64_t *__restrict__ okey,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		if (odate[i] >= cutoff)	/* qual: o_orderdate < cutoff */
+		/* dense loads: at ~46% date selectivity a masked load
+		 * touches nearly every line anyway, so load all three
+		 * streams coalesced and filter afterwards */
+		int32_t d = odate[i];
+		int64_t ck = ckey[i];
+		int64_t k = okey[i];
+
+		if (d >= cutoff)	/* qual: o_orderdate < cutoff */
 			continue;
 		if (!ht_contains_b(cust_keys, cust_slots, cust_bloom,
-				   cust_bwords, ckey[i]))
+				   cust_bwords, ck))
 			continue;
 		matches++;
 		{
-			int64_t k = okey[i];
 			uint64_t pos = ht_start(k, nslots);
 
 			if (bloom)
@@ -697,10 +703,14 @@ void k_probe_lineitem(const int64_t *__restrict__ okey,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		if (shipdate[i] <= cutoff)	/* qual: l_shipdate > cutoff */
+		/* dense loads (54% date selectivity): shipdate + okey
+		 * coalesced, then Bloom before any table access */
+		int32_t sd = shipdate[i];
+		int64_t k = okey[i];
+
+		if (sd <= cutoff)	/* qual: l_shipdate > cutoff */
 			continue;
 		{
-			int64_t k = okey[i];
 			uint32_t h = gg_hashint8(k);
 			uint64_t pos;
 
