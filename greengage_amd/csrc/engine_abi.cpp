@@ -1101,45 +1101,26 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 		if (maxrev > 0)
 		{
-			int shift = 0;
-
-			while ((maxrev >> shift) > 65535)
-				shift++;
+			/* device-side threshold select: histogram + suffix
+			 * walk + collect without host round trips */
 			unsigned int *dhist =
 				(unsigned int *) p->sget("hist", 65536 * 4);
-			std::vector<unsigned int> hist(65536);
-
-			if (!dhist)
-				return fail(GG_ENOMEM, "hist scratch");
-			GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4, e.stream));
-			GG_HIP(launch_q3_hist(e.stream, ord, shift, 0, dhist));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			GG_HIP(hipMemcpy(hist.data(), dhist, 65536 * 4,
-					 hipMemcpyDeviceToHost));
-
-			uint64_t cum = 0, thr_bin = 0;
-
-			for (int64_t b = 65535; b >= 0; b--)
-			{
-				cum += hist[b];
-				if (cum >= (uint64_t) k || b == 0)
-				{
-					thr_bin = (uint64_t) b;
-					break;
-				}
-			}
-			uint64_t threshold = thr_bin << shift;
-			uint64_t cap = cum + 65536;	/* bin-granule slack */
+			unsigned long long *dthr =
+				(unsigned long long *) p->sget("thr", 8);
+			uint64_t cap = hstats[0] + 1;	/* <= n_groups */
 			gg_q3_result_row *dout = (gg_q3_result_row *)
 				p->sget("cand", cap * sizeof(gg_q3_result_row));
 			unsigned long long ncand = 0;
 
-			if (!dout)
-				return fail(GG_ENOMEM, "cand scratch");
+			if (!dhist || !dthr || !dout)
+				return fail(GG_ENOMEM, "topk scratch");
+			GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4, e.stream));
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-			GG_HIP(launch_q3_collect(e.stream, ord,
-						 threshold ? threshold : 1,
-						 dout, ctr, cap));
+			GG_HIP(launch_q3_hist(e.stream, ord, stats5, dhist));
+			GG_HIP(launch_q3_threshold(e.stream, dhist, stats5, k,
+						   dthr));
+			GG_HIP(launch_q3_collect(e.stream, ord, dthr, dout,
+						 ctr, cap));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &ncand));
 			if (ncand > cap)
@@ -1156,7 +1137,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				cand.resize(k);
 		}
 
-		double ms = tm.stop();
+				double ms = tm.stop();
 		KernelStatAcc &st = p->stat("q3_topk");
 
 		st.launches++;

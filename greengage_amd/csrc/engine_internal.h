@@ -271,10 +271,15 @@ hipError_t launch_probe_lineitem(hipStream_t s, const int64_t *okey,
 hipError_t launch_q3_stats(hipStream_t s, DeviceHashTable ord,
 			   unsigned long long *out4 /* ngroups, revlo,
 						     * revhi, checksum */);
-hipError_t launch_q3_hist(hipStream_t s, DeviceHashTable ord, int shift,
-			  uint64_t lo_bound, unsigned int *hist64k);
+hipError_t launch_q3_hist(hipStream_t s, DeviceHashTable ord,
+			  const unsigned long long *stats5,
+			  unsigned int *hist64k);
+hipError_t launch_q3_threshold(hipStream_t s, const unsigned int *hist64k,
+			       const unsigned long long *stats5, int64_t k,
+			       unsigned long long *out_thr);
 hipError_t launch_q3_collect(hipStream_t s, DeviceHashTable ord,
-			     uint64_t threshold, gg_q3_result_row *out,
+			     const unsigned long long *thr_ptr,
+			     gg_q3_result_row *out,
 			     unsigned long long *out_count, uint64_t cap);
 
 hipError_t launch_count_filter_u8(hipStream_t s, const uint8_t *col,

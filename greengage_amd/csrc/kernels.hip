@@ -831,12 +831,24 @@ launch_q3_stats(hipStream_t s, DeviceHashTable ord, unsigned long long *out5)
 	return hipGetLastError();
 }
 
+__device__ inline int
+q3_shift_for(unsigned long long maxrev)
+{
+	int shift = 0;
+
+	while ((maxrev >> shift) > 65535)
+		shift++;
+	return shift;
+}
+
 __global__ void
 k_q3_hist(const unsigned long long *__restrict__ tkeys,
 	  const unsigned long long *__restrict__ trev, uint64_t nslots,
-	  int shift, unsigned int *__restrict__ hist64k)
+	  const unsigned long long *__restrict__ stats5,
+	  unsigned int *__restrict__ hist64k)
 {
 	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+	const int shift = q3_shift_for(stats5[4]);
 
 	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < nslots; i += stride)
@@ -859,12 +871,73 @@ k_q3_hist(const unsigned long long *__restrict__ tkeys,
 }
 
 hipError_t
-launch_q3_hist(hipStream_t s, DeviceHashTable ord, int shift,
-	       uint64_t /*lo_bound*/, unsigned int *hist64k)
+launch_q3_hist(hipStream_t s, DeviceHashTable ord,
+	       const unsigned long long *stats5, unsigned int *hist64k)
 {
 	hipLaunchKernelGGL(k_q3_hist, dim3(grid_for((int64_t) ord.nslots)),
 			   dim3(THREADS), 0, s, ord.keys, ord.rev, ord.nslots,
-			   shift, hist64k);
+			   stats5, hist64k);
+	return hipGetLastError();
+}
+
+/* one block: walk the histogram from the top bin until >= k candidates;
+ * emit the revenue threshold (device-side — no host round trip) */
+__global__ __launch_bounds__(1024)
+void k_q3_threshold(const unsigned int *__restrict__ hist64k,
+		    const unsigned long long *__restrict__ stats5, int64_t k,
+		    unsigned long long *__restrict__ out_thr)
+{
+	__shared__ unsigned long long partial[1024];
+
+	/* each thread sums its 64-bin chunk (chunk 1023 = highest bins) */
+	unsigned long long s = 0;
+
+	for (int b = 0; b < 64; b++)
+		s += hist64k[threadIdx.x * 64 + b];
+	partial[threadIdx.x] = s;
+	__syncthreads();
+	if (threadIdx.x == 0)
+	{
+		unsigned long long cum = 0;
+		int chunk = 1023;
+
+		for (; chunk > 0; chunk--)
+		{
+			if (cum + partial[chunk] >= (unsigned long long) k)
+				break;
+			cum += partial[chunk];
+		}
+		{
+			unsigned long long bin = (unsigned long long) chunk * 64 + 63;
+			int b = 63;
+
+			for (; b > 0; b--)
+			{
+				unsigned long long c =
+					hist64k[chunk * 64 + b];
+
+				if (cum + c >= (unsigned long long) k)
+					break;
+				cum += c;
+			}
+			bin = (unsigned long long) chunk * 64 + b;
+			{
+				int shift = q3_shift_for(stats5[4]);
+				unsigned long long thr = bin << shift;
+
+				*out_thr = thr ? thr : 1;
+			}
+		}
+	}
+}
+
+hipError_t
+launch_q3_threshold(hipStream_t s, const unsigned int *hist64k,
+		    const unsigned long long *stats5, int64_t k,
+		    unsigned long long *out_thr)
+{
+	hipLaunchKernelGGL(k_q3_threshold, dim3(1), dim3(1024), 0, s,
+			   hist64k, stats5, k, out_thr);
 	return hipGetLastError();
 }
 
@@ -872,10 +945,12 @@ __global__ void
 k_q3_collect(const unsigned long long *__restrict__ tkeys,
 	     const unsigned long long *__restrict__ tpayload,
 	     const unsigned long long *__restrict__ trev, uint64_t nslots,
-	     unsigned long long threshold, gg_q3_result_row *__restrict__ out,
+	     const unsigned long long *__restrict__ thr_ptr,
+	     gg_q3_result_row *__restrict__ out,
 	     unsigned long long *out_count, uint64_t cap)
 {
 	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+	const unsigned long long threshold = *thr_ptr;
 
 	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < nslots; i += stride)
@@ -907,13 +982,13 @@ k_q3_collect(const unsigned long long *__restrict__ tkeys,
 }
 
 hipError_t
-launch_q3_collect(hipStream_t s, DeviceHashTable ord, uint64_t threshold,
-		  gg_q3_result_row *out, unsigned long long *out_count,
-		  uint64_t cap)
+launch_q3_collect(hipStream_t s, DeviceHashTable ord,
+		  const unsigned long long *thr_ptr, gg_q3_result_row *out,
+		  unsigned long long *out_count, uint64_t cap)
 {
 	hipLaunchKernelGGL(k_q3_collect, dim3(grid_for((int64_t) ord.nslots)),
 			   dim3(THREADS), 0, s, ord.keys, ord.payload,
-			   ord.rev, ord.nslots, threshold, out, out_count,
+			   ord.rev, ord.nslots, thr_ptr, out, out_count,
 			   cap);
 	return hipGetLastError();
 }
