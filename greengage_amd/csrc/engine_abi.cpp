@@ -642,8 +642,10 @@ static gg_status exec_q1(Pipeline *p, void *arena, size_t bytes,
 	static_assert(sizeof(Q1DeviceAcc) == 49 * 8, "acc layout");
 	std::vector<unsigned long long> parts;
 	int nseg = e.cfg.n_segments;
+	bool exch = nseg > 1 || (getenv("GG_FORCE_EXCHANGE") != nullptr &&
+				 comm_ready());
 
-	if (nseg > 1)
+	if (exch)
 	{
 		if (!comm_ready())
 			return fail(GG_ESTATE, "multi-segment without comm");

@@ -50,6 +50,28 @@ def test_q3_exchange_loopback_bitexact():
         eng.shutdown()
 
 
+def test_q1_allgather_loopback():
+    """Q1's 2-stage-agg combine over RCCL allgather, world-1 loopback."""
+    import os
+    from greengage_amd import Engine, PGDate
+    from greengage_amd.engine import PIPE_Q1
+
+    eng = Engine(device=0, n_segments=1, segment_id=0)
+    try:
+        eng.comm_init(eng.comm_id())
+        li = eng.register_synth("lineitem", seed=42, sf=1)
+        p = eng.compile(PIPE_Q1, lineitem=li,
+                        cutoff_date=PGDate("1998-08-15"))
+        base = eng.execute_q1(p)
+        os.environ["GG_FORCE_EXCHANGE"] = "1"
+        try:
+            assert eng.execute_q1(p) == base
+        finally:
+            del os.environ["GG_FORCE_EXCHANGE"]
+    finally:
+        eng.shutdown()
+
+
 def test_q1_steady_state_reuse():
     """Repeated executes on one pipeline (scratch reuse) stay
     bit-identical — the bench measures exactly this steady state."""
