@@ -18,6 +18,7 @@
  *                     bit-exact cdbhash (include/gg_pg_hash.h; row a11)
  */
 #include <hip/hip_runtime.h>
+#include <cstdlib>
 
 #include "../../include/gg_pg_hash.h"
 #include "../../include/gg_gen.h"
@@ -50,6 +51,13 @@ static constexpr int Q1_FIELDS = 8;	/* cnt,qty,base,dcol,disc,+pad.. */
 static constexpr int Q1_SLOTS = 6 * Q1_FIELDS;	/* 48 */
 static constexpr int Q1_STRIDE = Q1_SLOTS + 1;	/* 49: odd → bank spread */
 
+template <bool NT>
+__device__ inline int64_t q1_ld64(const int64_t *p)
+{
+	return NT ? __builtin_nontemporal_load(p) : *p;
+}
+
+template <bool NT>
 __global__ __launch_bounds__(THREADS, 2)
 void k_q1_agg(const int32_t *__restrict__ shipdate,
 	      const uint8_t *__restrict__ rflag,
@@ -74,7 +82,8 @@ void k_q1_agg(const int32_t *__restrict__ shipdate,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		int32_t sd = shipdate[i];
+		int32_t sd = NT ? __builtin_nontemporal_load(&shipdate[i])
+			: shipdate[i];
 
 		if (sd > cutoff)	/* qual: l_shipdate <= cutoff */
 			continue;
@@ -88,10 +97,10 @@ void k_q1_agg(const int32_t *__restrict__ shipdate,
 			bad |= (ls != 'F' && ls != 'O');
 			{
 				int g = (fi * 2 + si) * Q1_FIELDS;
-				int64_t q = qty[i];
-				int64_t p = price[i];
-				int64_t d = disc[i];
-				int64_t t = tax[i];
+				int64_t q = q1_ld64<NT>(&qty[i]);
+				int64_t p = q1_ld64<NT>(&price[i]);
+				int64_t d = q1_ld64<NT>(&disc[i]);
+				int64_t t = q1_ld64<NT>(&tax[i]);
 				/* mul_var exact: scale 2+2 then 4+2 */
 				unsigned long long disc4 =
 					(unsigned long long) (p * (100 - d));
@@ -159,9 +168,27 @@ launch_q1(hipStream_t s, const int32_t *shipdate, const uint8_t *rflag,
 	  const int64_t *disc, const int64_t *tax, int64_t n, int32_t cutoff,
 	  Q1DeviceAcc *acc)
 {
-	hipLaunchKernelGGL(k_q1_agg, dim3(grid_for(n)), dim3(THREADS), 0, s,
-			   shipdate, rflag, lstatus, qty, price, disc, tax,
-			   n, cutoff, acc);
+	/* tuning knobs (benchmark sweeps; defaults are the shipped config) */
+	int grid = grid_for(n);
+	const char *gs = getenv("GG_Q1_GRID");
+	bool nt = getenv("GG_Q1_NT") != nullptr;
+
+	if (gs)
+	{
+		int64_t cap = (n + THREADS - 1) / THREADS;
+		int g = atoi(gs);
+
+		if (g > 0)
+			grid = (int) (g < cap ? g : cap);
+	}
+	if (nt)
+		hipLaunchKernelGGL(k_q1_agg<true>, dim3(grid), dim3(THREADS),
+				   0, s, shipdate, rflag, lstatus, qty, price,
+				   disc, tax, n, cutoff, acc);
+	else
+		hipLaunchKernelGGL(k_q1_agg<false>, dim3(grid), dim3(THREADS),
+				   0, s, shipdate, rflag, lstatus, qty, price,
+				   disc, tax, n, cutoff, acc);
 	return hipGetLastError();
 }
 
