@@ -32,6 +32,17 @@ static constexpr int THREADS = 256;
 /* 256 CUs × 8 blocks of 256 threads per CU (Guideline 11) */
 static constexpr int MAX_BLOCKS = 2048;
 
+/* nontemporal loads for read-once streams (bypass L1, early L2 evict;
+ * never used on hash-table / bloom traffic, which wants caching) */
+__device__ inline int64_t nt_ld64(const int64_t *p)
+{
+	return __builtin_nontemporal_load(p);
+}
+__device__ inline int32_t nt_ld32(const int32_t *p)
+{
+	return __builtin_nontemporal_load(p);
+}
+
 static inline int grid_for(int64_t n)
 {
 	int64_t b = (n + THREADS - 1) / THREADS;
@@ -171,7 +182,10 @@ launch_q1(hipStream_t s, const int32_t *shipdate, const uint8_t *rflag,
 	/* tuning knobs (benchmark sweeps; defaults are the shipped config) */
 	int grid = grid_for(n);
 	const char *gs = getenv("GG_Q1_GRID");
-	bool nt = getenv("GG_Q1_NT") != nullptr;
+	/* nontemporal streaming loads measured +11%% on this kernel
+	 * (6.38 vs 5.55 TB/s — tools/q1_sweep.py); default ON */
+	const char *nts = getenv("GG_Q1_NT");
+	bool nt = !(nts && nts[0] == '0');
 
 	if (gs)
 	{
@@ -206,9 +220,9 @@ void k_sumprice(const int32_t *__restrict__ shipdate,
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
-		if (shipdate[i] < cutoff)
+		if (nt_ld32(&shipdate[i]) < cutoff)
 		{
-			sum += (unsigned long long) price[i];
+			sum += (unsigned long long) nt_ld64(&price[i]);
 			cnt++;
 		}
 	/* wave reduction then one atomic per wave */
@@ -611,9 +625,9 @@ k_build_orders(const int64_t *__restrict__ okey,
 		/* dense loads: at ~46% date selectivity a masked load
 		 * touches nearly every line anyway, so load all three
 		 * streams coalesced and filter afterwards */
-		int32_t d = odate[i];
-		int64_t ck = ckey[i];
-		int64_t k = okey[i];
+		int32_t d = nt_ld32(&odate[i]);
+		int64_t ck = nt_ld64(&ckey[i]);
+		int64_t k = nt_ld64(&okey[i]);
 
 		if (d >= cutoff)	/* qual: o_orderdate < cutoff */
 			continue;
@@ -732,8 +746,8 @@ void k_probe_lineitem(const int64_t *__restrict__ okey,
 	{
 		/* dense loads (54% date selectivity): shipdate + okey
 		 * coalesced, then Bloom before any table access */
-		int32_t sd = shipdate[i];
-		int64_t k = okey[i];
+		int32_t sd = nt_ld32(&shipdate[i]);
+		int64_t k = nt_ld64(&okey[i]);
 
 		if (sd <= cutoff)	/* qual: l_shipdate > cutoff */
 			continue;
@@ -1623,8 +1637,8 @@ void k_probe_lineitem_q5(const int64_t *__restrict__ okey,
 		 * tested before any slot-array probe: sparse exec-masked
 		 * loads mid-kernel cost a full cache line per surviving
 		 * lane, so the cheap rejections come first */
-		int64_t sk = skey[i];
-		int64_t ok = okey[i];
+		int64_t sk = nt_ld64(&skey[i]);
+		int64_t ok = nt_ld64(&okey[i]);
 		uint32_t hsk = gg_hashint8(sk);
 		uint32_t hok = gg_hashint8(ok);
 		unsigned long long snat, onat;
