@@ -1875,6 +1875,95 @@ gg_engine_stats(gg_pipeline h, gg_kernel_stat *out, int cap, int *out_n)
 }
 
 
+
+/* ---------------- general hash group-by (ABI surface) ---------------- */
+
+extern "C" gg_status
+gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,
+			   int64_t n, int64_t *out_keys, int64_t *out_sums,
+			   int64_t *out_counts, int64_t cap,
+			   int64_t *out_ngroups)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!keys || !vals || n < 0 || !out_keys || !out_sums ||
+	    !out_counts || !out_ngroups)
+		return fail(GG_EINVAL, "bad groupby args");
+	*out_ngroups = 0;
+	if (n == 0)
+		return GG_OK;
+
+	uint64_t nslots = next_pow2(2 * (uint64_t) n);
+	int64_t *dk = nullptr, *dv = nullptr;
+	unsigned long long *tk = nullptr, *ts = nullptr, *tc = nullptr;
+	int64_t *ok = nullptr, *os = nullptr, *oc = nullptr;
+	unsigned long long *ctr = nullptr;
+
+	GG_HIP(hipMalloc((void **) &dk, (size_t) n * 8));
+	GG_HIP(hipMalloc((void **) &dv, (size_t) n * 8));
+	GG_HIP(hipMalloc((void **) &tk, nslots * 8));
+	GG_HIP(hipMalloc((void **) &ts, nslots * 8));
+	GG_HIP(hipMalloc((void **) &tc, nslots * 8));
+	GG_HIP(hipMalloc((void **) &ok, (size_t) n * 8));
+	GG_HIP(hipMalloc((void **) &os, (size_t) n * 8));
+	GG_HIP(hipMalloc((void **) &oc, (size_t) n * 8));
+	GG_HIP(hipMalloc((void **) &ctr, 8));
+	GG_HIP(hipMemcpy(dk, keys, (size_t) n * 8, hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(dv, vals, (size_t) n * 8, hipMemcpyHostToDevice));
+	/* the empty sentinel (INT64_MIN) is not a memset byte pattern —
+	 * fill with a kernel */
+	GG_HIP(launch_fill_u64(e.stream, tk, nslots, 0x8000000000000000ull));
+	GG_HIP(hipMemsetAsync(ts, 0, nslots * 8, e.stream));
+	GG_HIP(hipMemsetAsync(tc, 0, nslots * 8, e.stream));
+	GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+	GG_HIP(launch_groupby_build(e.stream, dk, dv, n, tk, ts, tc, nslots));
+	GG_HIP(launch_groupby_compact(e.stream, tk, ts, tc, nslots, ok, os,
+				      oc, ctr, (uint64_t) n));
+	GG_HIP(hipStreamSynchronize(e.stream));
+
+	unsigned long long ng = 0;
+
+	GG_HIP(hipMemcpy(&ng, ctr, 8, hipMemcpyDeviceToHost));
+	gg_status st = GG_OK;
+
+	if ((int64_t) ng > cap)
+		st = fail(GG_EINVAL, "groupby cap %lld < %llu groups",
+			  (long long) cap, ng);
+	if (st == GG_OK && ng)
+	{
+		std::vector<int64_t> hk(ng), hs(ng), hc(ng);
+
+		GG_HIP(hipMemcpy(hk.data(), ok, ng * 8, hipMemcpyDeviceToHost));
+		GG_HIP(hipMemcpy(hs.data(), os, ng * 8, hipMemcpyDeviceToHost));
+		GG_HIP(hipMemcpy(hc.data(), oc, ng * 8, hipMemcpyDeviceToHost));
+		std::vector<size_t> idx(ng);
+
+		for (size_t i = 0; i < ng; i++)
+			idx[i] = i;
+		std::sort(idx.begin(), idx.end(),
+			  [&](size_t a2, size_t b2) { return hk[a2] < hk[b2]; });
+		for (size_t i = 0; i < ng; i++)
+		{
+			out_keys[i] = hk[idx[i]];
+			out_sums[i] = hs[idx[i]];
+			out_counts[i] = hc[idx[i]];
+		}
+		*out_ngroups = (int64_t) ng;
+	}
+	(void) hipFree(dk);
+	(void) hipFree(dv);
+	(void) hipFree(tk);
+	(void) hipFree(ts);
+	(void) hipFree(tc);
+	(void) hipFree(ok);
+	(void) hipFree(os);
+	(void) hipFree(oc);
+	(void) hipFree(ctr);
+	return st;
+}
+
 /* ---------------- general radix sort (ABI surface) ---------------- */
 
 extern "C" gg_status

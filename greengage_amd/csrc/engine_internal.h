@@ -316,6 +316,20 @@ hipError_t launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n,
 				unsigned long long *offsets,
 				int64_t *oa, int64_t *ob, int64_t *oc);
 
+hipError_t launch_fill_u64(hipStream_t s, unsigned long long *p,
+			   uint64_t n, unsigned long long v);
+hipError_t launch_groupby_build(hipStream_t s, const int64_t *keys,
+				const int64_t *vals, int64_t n,
+				unsigned long long *tkeys,
+				unsigned long long *tsum,
+				unsigned long long *tcnt, uint64_t nslots);
+hipError_t launch_groupby_compact(hipStream_t s,
+				  const unsigned long long *tkeys,
+				  const unsigned long long *tsum,
+				  const unsigned long long *tcnt,
+				  uint64_t nslots, int64_t *out_keys,
+				  int64_t *out_sums, int64_t *out_cnts,
+				  unsigned long long *out_n, uint64_t cap);
 hipError_t launch_radix_sort_pass(hipStream_t s,
 				  const unsigned long long *keys,
 				  const unsigned long long *pay, int64_t n,

@@ -1751,6 +1751,129 @@ launch_gen_supplier(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
 	return hipGetLastError();
 }
 
+
+/* ------------------------------------------------------------------ */
+/* general hash group-by: arbitrary int64 keys, many groups            */
+/* (execHHashagg.c:456 lookup_agg_hash_entry semantics: find-or-create */
+/* group, transition per row; open addressing, EMPTY = INT64_MIN)      */
+/* ------------------------------------------------------------------ */
+
+#define GB_EMPTY 0x8000000000000000ull	/* INT64_MIN as the empty slot */
+
+__global__ void
+k_groupby_build(const int64_t *__restrict__ keys,
+		const int64_t *__restrict__ vals, int64_t n,
+		unsigned long long *__restrict__ tkeys,
+		unsigned long long *__restrict__ tsum,
+		unsigned long long *__restrict__ tcnt, uint64_t nslots)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k = nt_ld64(&keys[i]);
+		int64_t v = nt_ld64(&vals[i]);
+		uint64_t pos = (uint64_t) gg_hashint8(k) & (nslots - 1);
+
+		for (;;)
+		{
+			unsigned long long cur = tkeys[pos];
+
+			if (cur == (unsigned long long) k)
+				break;
+			if (cur == GB_EMPTY)
+			{
+				unsigned long long prev =
+					atomicCAS(&tkeys[pos], GB_EMPTY,
+						  (unsigned long long) k);
+				if (prev == GB_EMPTY ||
+				    prev == (unsigned long long) k)
+					break;
+			}
+			pos = (pos + 1) & (nslots - 1);
+		}
+		/* wrapping u64 add == two's-complement int64 sum */
+		atomicAdd(&tsum[pos], (unsigned long long) v);
+		atomicAdd(&tcnt[pos], 1ull);
+	}
+}
+
+__global__ void
+k_groupby_compact(const unsigned long long *__restrict__ tkeys,
+		  const unsigned long long *__restrict__ tsum,
+		  const unsigned long long *__restrict__ tcnt,
+		  uint64_t nslots, int64_t *__restrict__ out_keys,
+		  int64_t *__restrict__ out_sums,
+		  int64_t *__restrict__ out_cnts,
+		  unsigned long long *out_n, uint64_t cap)
+{
+	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+
+	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nslots; i += stride)
+	{
+		if (tkeys[i] == GB_EMPTY)
+			continue;
+		{
+			unsigned long long idx = atomicAdd(out_n, 1ull);
+
+			if (idx < cap)
+			{
+				out_keys[idx] = (int64_t) tkeys[i];
+				out_sums[idx] = (int64_t) tsum[i];
+				out_cnts[idx] = (int64_t) tcnt[i];
+			}
+		}
+	}
+}
+
+hipError_t
+launch_groupby_build(hipStream_t s, const int64_t *keys, const int64_t *vals,
+		     int64_t n, unsigned long long *tkeys,
+		     unsigned long long *tsum, unsigned long long *tcnt,
+		     uint64_t nslots)
+{
+	hipLaunchKernelGGL(k_groupby_build, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, keys, vals, n, tkeys, tsum, tcnt, nslots);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_groupby_compact(hipStream_t s, const unsigned long long *tkeys,
+		       const unsigned long long *tsum,
+		       const unsigned long long *tcnt, uint64_t nslots,
+		       int64_t *out_keys, int64_t *out_sums,
+		       int64_t *out_cnts, unsigned long long *out_n,
+		       uint64_t cap)
+{
+	hipLaunchKernelGGL(k_groupby_compact,
+			   dim3(grid_for((int64_t) nslots)), dim3(THREADS),
+			   0, s, tkeys, tsum, tcnt, nslots, out_keys,
+			   out_sums, out_cnts, out_n, cap);
+	return hipGetLastError();
+}
+
+__global__ void
+k_fill_u64(unsigned long long *__restrict__ p, uint64_t n,
+	   unsigned long long v)
+{
+	const uint64_t stride = (uint64_t) gridDim.x * blockDim.x;
+
+	for (uint64_t i = (uint64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		p[i] = v;
+}
+
+hipError_t
+launch_fill_u64(hipStream_t s, unsigned long long *p, uint64_t n,
+		unsigned long long v)
+{
+	hipLaunchKernelGGL(k_fill_u64, dim3(grid_for((int64_t) n)),
+			   dim3(THREADS), 0, s, p, n, v);
+	return hipGetLastError();
+}
+
 /* count-helper launchers used by engine_abi.cpp */
 hipError_t
 launch_count_filter_u8(hipStream_t s, const uint8_t *col, uint8_t val,

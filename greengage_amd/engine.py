@@ -106,6 +106,9 @@ def _load():
     lib.gg_engine_radix_sort_u64.argtypes = [ctypes.c_void_p,
                                              ctypes.c_void_p, I64,
                                              ctypes.c_int, ctypes.c_int]
+    lib.gg_engine_hash_groupby_i64.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, I64, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_void_p, I64, ctypes.POINTER(I64)]
     lib.gg_engine_avg_str.argtypes = [U64, I64, ctypes.c_int, I64,
                                       ctypes.c_char_p]
     return lib
@@ -292,6 +295,28 @@ class Engine:
         raw = self.execute_raw(p, 16)
         return (int.from_bytes(raw[0:8], "little", signed=True),
                 int.from_bytes(raw[8:16], "little", signed=True))
+
+    # ---- general hash group-by (arbitrary int64 keys, SUM+COUNT) ----
+    @staticmethod
+    def hash_groupby(keys, vals):
+        import numpy as np
+        keys = np.ascontiguousarray(keys, dtype=np.int64)
+        vals = np.ascontiguousarray(vals, dtype=np.int64)
+        assert len(keys) == len(vals)
+        cap = len(keys) if len(keys) else 1
+        ok = np.empty(cap, np.int64)
+        os_ = np.empty(cap, np.int64)
+        oc = np.empty(cap, np.int64)
+        ng = I64()
+        _check(lib().gg_engine_hash_groupby_i64(
+            keys.ctypes.data_as(ctypes.c_void_p),
+            vals.ctypes.data_as(ctypes.c_void_p), len(keys),
+            ok.ctypes.data_as(ctypes.c_void_p),
+            os_.ctypes.data_as(ctypes.c_void_p),
+            oc.ctypes.data_as(ctypes.c_void_p), cap,
+            ctypes.byref(ng)), "hash_groupby")
+        n = ng.value
+        return ok[:n], os_[:n], oc[:n]
 
     # ---- general sort (ORDER BY operator; LSB radix on GPU) ----
     @staticmethod

@@ -201,6 +201,17 @@ void gg_engine_numeric_str(uint64_t lo, int64_t hi, int scale, char *buf64);
 void gg_engine_avg_str(uint64_t sum_lo, int64_t sum_hi, int sum_scale,
 		       int64_t count, char *buf64);
 
+/* General hash group-by (execHHashagg.c find-or-create semantics on
+ * arbitrary int64 keys, SUM+COUNT transitions): host buffers in,
+ * groups out sorted by key ascending.  Keys may be any int64 except
+ * INT64_MIN (the open-addressing empty sentinel).  Returns the group
+ * count; fails with GG_EINVAL if cap is too small. */
+gg_status gg_engine_hash_groupby_i64(const int64_t *keys,
+				     const int64_t *vals, int64_t n,
+				     int64_t *out_keys, int64_t *out_sums,
+				     int64_t *out_counts, int64_t cap,
+				     int64_t *out_ngroups);
+
 /* General ORDER BY operator: stable LSB radix sort of (u64 key, u64
  * payload) pairs on the GPU (nodeSort.c:48 / tuplesort.c semantics for
  * unbounded sorts; the LIMIT-k case uses histogram select instead,
