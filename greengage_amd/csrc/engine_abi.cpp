@@ -716,61 +716,6 @@ static gg_status exec_q1(Pipeline *p, void *arena, size_t bytes,
 	return GG_OK;
 }
 
-
-/* insert (key, pay) pairs slot-ordered: bucket histogram → prefix →
- * scatter → ordered inserts into 512 KB windows (bucket-ordered CAS
- * measured 3.4× faster than random on gfx950 — tools/insert_bench,
- * SURVEY §7 step 5 radix-partitioned build) */
-static gg_status partitioned_insert(Pipeline *p, const char *tag,
-				    const int64_t *keys, const int64_t *pay,
-				    uint64_t m, DeviceHashTable t)
-{
-	Engine &e = engine();
-
-	if (m == 0)
-		return GG_OK;
-	const int WIN_SHIFT = 16;	/* 2^16 slots = 512 KB key window */
-	uint64_t nb = t.nslots >> WIN_SHIFT;
-
-	if (nb < 2)
-	{
-		GG_HIP(launch_insert_orders(e.stream, keys, pay, (int64_t) m,
-					    t));
-		return GG_OK;
-	}
-	std::string base(tag);
-	unsigned long long *cnts = (unsigned long long *)
-		p->sget((base + ".bcnt").c_str(), nb * 8);
-	int64_t *sk = (int64_t *) p->sget((base + ".bk").c_str(), (m + 1) * 8);
-	int64_t *sp = (int64_t *) p->sget((base + ".bp").c_str(), (m + 1) * 8);
-
-	if (!cnts || !sk || !sp)
-		return fail(GG_ENOMEM, "partition scratch");
-	GG_HIP(hipMemsetAsync(cnts, 0, nb * 8, e.stream));
-	GG_HIP(launch_bucket_count(e.stream, keys, (int64_t) m, t.nslots,
-				   WIN_SHIFT, cnts));
-	GG_HIP(hipStreamSynchronize(e.stream));
-	{
-		std::vector<unsigned long long> h(nb), offs(nb);
-
-		GG_HIP(hipMemcpy(h.data(), cnts, nb * 8,
-				 hipMemcpyDeviceToHost));
-		unsigned long long run = 0;
-
-		for (uint64_t b = 0; b < nb; b++)
-		{
-			offs[b] = run;
-			run += h[b];
-		}
-		GG_HIP(hipMemcpy(cnts, offs.data(), nb * 8,
-				 hipMemcpyHostToDevice));
-	}
-	GG_HIP(launch_bucket_scatter2(e.stream, keys, pay, (int64_t) m,
-				      t.nslots, WIN_SHIFT, cnts, sk, sp));
-	GG_HIP(launch_insert_orders(e.stream, sk, sp, (int64_t) m, t));
-	return GG_OK;
-}
-
 /* ---------------- execution: Q3 ---------------- */
 
 struct Q3TopkCmp
@@ -909,27 +854,10 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
 				      e.stream));
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		{
-			/* compact matched (okey, pay), then slot-ordered
-			 * partitioned insert (cap: nslots >= 2x matches) */
-			int64_t *mk = (int64_t *)
-				p->sget("ord.mk", (ord.nslots / 2 + 1) * 8);
-			int64_t *mp = (int64_t *)
-				p->sget("ord.mp", (ord.nslots / 2 + 1) * 8);
-
-			if (!mk || !mp)
-				return fail(GG_ENOMEM, "match scratch");
-			GG_HIP(launch_orders_match_compact(
-				e.stream, o_ok, o_ck, o_dt, o_pr, od->nrows,
-				cutoff, cust, mk, mp, ctr));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			GG_TRY(read_counter(ctr, &nmatch));
-			if (nmatch > ord.nslots / 2)
-				return fail(GG_EINVAL, "ord table undersized");
-			GG_TRY(partitioned_insert(p, "ord", mk, mp, nmatch,
-						  ord));
-			GG_HIP(hipStreamSynchronize(e.stream));
-		}
+		GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt, o_pr,
+					   od->nrows, cutoff, cust, ord, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nmatch));
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("build_orders");
 
@@ -1121,8 +1049,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8, e.stream));
 		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
 				      e.stream));
-		GG_TRY(partitioned_insert(p, "ord", r2_ok, r2_pay, rtotal2,
-					  ord));
+		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_pay,
+					    (int64_t) rtotal2, ord));
 		GG_HIP(hipStreamSynchronize(e.stream));
 
 		double ms = tm.stop();
@@ -1405,20 +1333,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		if (!cust.keys || !cust.payload)
 			return fail(GG_ENOMEM, "cust map");
 		GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8, e.stream));
-		{
-			int64_t *wk = (int64_t *)
-				p->sget("cust.wk", (uint64_t) (cu->nrows + 1) * 8);
-			int64_t *wp = (int64_t *)
-				p->sget("cust.wp", (uint64_t) (cu->nrows + 1) * 8);
-
-			if (!wk || !wp)
-				return fail(GG_ENOMEM, "cust widen scratch");
-			GG_HIP(launch_kv_widen(e.stream, c_ck, c_nk,
-					       cu->nrows, wk, wp));
-			GG_TRY(partitioned_insert(p, "cust", wk, wp,
-						  (uint64_t) cu->nrows, cust));
-			GG_HIP(hipStreamSynchronize(e.stream));
-		}
+		GG_HIP(launch_build_kv(e.stream, c_ck, c_nk, cu->nrows, cust));
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("build_customer_map");
 
@@ -1593,25 +1508,11 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
 				      e.stream));
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		{
-			int64_t *mk = (int64_t *)
-				p->sget("ord.mk", (ord.nslots / 2 + 1) * 8);
-			int64_t *mp = (int64_t *)
-				p->sget("ord.mp", (ord.nslots / 2 + 1) * 8);
-
-			if (!mk || !mp)
-				return fail(GG_ENOMEM, "match scratch");
-			GG_HIP(launch_orders_match_compact_q5(
-				e.stream, o_ok, o_ck, o_dt, od->nrows,
-				date_lo, date_hi, cust, mk, mp, ctr));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			GG_TRY(read_counter(ctr, &nmatch));
-			if (nmatch > ord.nslots / 2)
-				return fail(GG_EINVAL, "ord table undersized");
-			GG_TRY(partitioned_insert(p, "ord", mk, mp, nmatch,
-						  ord));
-			GG_HIP(hipStreamSynchronize(e.stream));
-		}
+		GG_HIP(launch_build_orders_q5(e.stream, o_ok, o_ck, o_dt,
+					      od->nrows, date_lo, date_hi,
+					      cust, ord, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &nmatch));
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("build_orders");
 
@@ -1785,8 +1686,8 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
 		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
 				      e.stream));
-		GG_TRY(partitioned_insert(p, "ord", r2_ok, r2_nat, rtotal2,
-					  ord));
+		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_nat,
+					    (int64_t) rtotal2, ord));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("orders_exchange");

@@ -316,31 +316,6 @@ hipError_t launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n,
 				unsigned long long *offsets,
 				int64_t *oa, int64_t *ob, int64_t *oc);
 
-hipError_t launch_bucket_count(hipStream_t s, const int64_t *keys,
-			       int64_t n, uint64_t nslots, int win_shift,
-			       unsigned long long *counts);
-hipError_t launch_bucket_scatter2(hipStream_t s, const int64_t *keys,
-				  const int64_t *pay, int64_t n,
-				  uint64_t nslots, int win_shift,
-				  unsigned long long *offsets, int64_t *out_k,
-				  int64_t *out_p);
-hipError_t launch_orders_match_compact(hipStream_t s, const int64_t *okey,
-				       const int64_t *ckey,
-				       const int32_t *odate,
-				       const int32_t *prio, int64_t n,
-				       int32_t cutoff, DeviceHashTable cust,
-				       int64_t *out_k, int64_t *out_p,
-				       unsigned long long *out_count);
-hipError_t launch_orders_match_compact_q5(hipStream_t s, const int64_t *okey,
-					  const int64_t *ckey,
-					  const int32_t *odate, int64_t n,
-					  int32_t date_lo, int32_t date_hi,
-					  DeviceHashTable cust,
-					  int64_t *out_k, int64_t *out_p,
-					  unsigned long long *out_count);
-hipError_t launch_kv_widen(hipStream_t s, const int64_t *keys,
-			   const uint8_t *v8, int64_t n, int64_t *out_k,
-			   int64_t *out_p);
 hipError_t launch_fill_u64(hipStream_t s, unsigned long long *p,
 			   uint64_t n, unsigned long long v);
 hipError_t launch_groupby_build(hipStream_t s, const int64_t *keys,

@@ -1874,203 +1874,6 @@ launch_fill_u64(hipStream_t s, unsigned long long *p, uint64_t n,
 	return hipGetLastError();
 }
 
-
-/* ------------------------------------------------------------------ */
-/* radix-partitioned inserts (histogram + prefix +   */
-/* scatter so inserts arrive in slot order — bucket-ordered CAS        */
-/* measured 3.4x faster than random on gfx950, tools/insert_bench)     */
-/* ------------------------------------------------------------------ */
-
-/* bucket of a key = its target slot's high bits (window = 2^shift
- * slots; ordered inserts then walk the table almost sequentially) */
-__global__ void
-k_bucket_count(const int64_t *__restrict__ keys, int64_t n, uint64_t nslots,
-	       int win_shift, unsigned long long *__restrict__ counts)
-{
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
-	{
-		uint64_t slot = (uint64_t) gg_hashint8(nt_ld64(&keys[i])) &
-			(nslots - 1);
-
-		atomicAdd(&counts[slot >> win_shift], 1ull);
-	}
-}
-
-__global__ void
-k_bucket_scatter2(const int64_t *__restrict__ keys,
-		  const int64_t *__restrict__ pay, int64_t n, uint64_t nslots,
-		  int win_shift, unsigned long long *__restrict__ offsets,
-		  int64_t *__restrict__ out_k, int64_t *__restrict__ out_p)
-{
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
-	{
-		int64_t k = nt_ld64(&keys[i]);
-		uint64_t slot = (uint64_t) gg_hashint8(k) & (nslots - 1);
-		unsigned long long idx =
-			atomicAdd(&offsets[slot >> win_shift], 1ull);
-
-		out_k[idx] = k;
-		out_p[idx] = pay ? nt_ld64(&pay[i]) : 0;
-	}
-}
-
-hipError_t
-launch_bucket_count(hipStream_t s, const int64_t *keys, int64_t n,
-		    uint64_t nslots, int win_shift,
-		    unsigned long long *counts)
-{
-	hipLaunchKernelGGL(k_bucket_count, dim3(grid_for(n)), dim3(THREADS),
-			   0, s, keys, n, nslots, win_shift, counts);
-	return hipGetLastError();
-}
-
-hipError_t
-launch_bucket_scatter2(hipStream_t s, const int64_t *keys,
-		       const int64_t *pay, int64_t n, uint64_t nslots,
-		       int win_shift, unsigned long long *offsets,
-		       int64_t *out_k, int64_t *out_p)
-{
-	hipLaunchKernelGGL(k_bucket_scatter2, dim3(grid_for(n)),
-			   dim3(THREADS), 0, s, keys, pay, n, nslots,
-			   win_shift, offsets, out_k, out_p);
-	return hipGetLastError();
-}
-
-/* Q3 orders: filter + customer probe → compacted (okey, date|prio) */
-__global__ void
-k_orders_match_compact(const int64_t *__restrict__ okey,
-		       const int64_t *__restrict__ ckey,
-		       const int32_t *__restrict__ odate,
-		       const int32_t *__restrict__ prio, int64_t n,
-		       int32_t cutoff,
-		       const unsigned long long *__restrict__ cust_keys,
-		       uint64_t cust_slots,
-		       const unsigned long long *__restrict__ cust_bloom,
-		       uint64_t cust_bwords, int64_t *__restrict__ out_k,
-		       int64_t *__restrict__ out_p,
-		       unsigned long long *out_count)
-{
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
-	{
-		int32_t d = nt_ld32(&odate[i]);
-		int64_t ck = nt_ld64(&ckey[i]);
-		int64_t k = nt_ld64(&okey[i]);
-
-		if (d >= cutoff)
-			continue;
-		if (!ht_contains_b(cust_keys, cust_slots, cust_bloom,
-				   cust_bwords, ck))
-			continue;
-		{
-			unsigned long long idx = atomicAdd(out_count, 1ull);
-
-			out_k[idx] = k;
-			out_p[idx] = (int64_t)
-				((unsigned long long) (uint32_t) d |
-				 ((unsigned long long) (uint32_t) prio[i]
-				  << 32));
-		}
-	}
-}
-
-hipError_t
-launch_orders_match_compact(hipStream_t s, const int64_t *okey,
-			    const int64_t *ckey, const int32_t *odate,
-			    const int32_t *prio, int64_t n, int32_t cutoff,
-			    DeviceHashTable cust, int64_t *out_k,
-			    int64_t *out_p, unsigned long long *out_count)
-{
-	hipLaunchKernelGGL(k_orders_match_compact, dim3(grid_for(n)),
-			   dim3(THREADS), 0, s, okey, ckey, odate, prio, n,
-			   cutoff, cust.keys, cust.nslots, cust.bloom,
-			   cust.bloom_words, out_k, out_p, out_count);
-	return hipGetLastError();
-}
-
-/* Q5 orders: date-range filter + customer-map probe → (okey, nation) */
-__global__ void
-k_orders_match_compact_q5(const int64_t *__restrict__ okey,
-			  const int64_t *__restrict__ ckey,
-			  const int32_t *__restrict__ odate, int64_t n,
-			  int32_t date_lo, int32_t date_hi,
-			  const unsigned long long *__restrict__ cust_keys,
-			  const unsigned long long *__restrict__ cust_pay,
-			  uint64_t cust_slots, int64_t *__restrict__ out_k,
-			  int64_t *__restrict__ out_p,
-			  unsigned long long *out_count)
-{
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
-	{
-		int32_t d = nt_ld32(&odate[i]);
-		int64_t ck = nt_ld64(&ckey[i]);
-		int64_t k = nt_ld64(&okey[i]);
-		unsigned long long nat;
-
-		if (d < date_lo || d >= date_hi)
-			continue;
-		if (!ht_lookup_b(cust_keys, cust_pay, cust_slots, nullptr, 0,
-				 ck, &nat))
-			continue;
-		{
-			unsigned long long idx = atomicAdd(out_count, 1ull);
-
-			out_k[idx] = k;
-			out_p[idx] = (int64_t) nat;
-		}
-	}
-}
-
-hipError_t
-launch_orders_match_compact_q5(hipStream_t s, const int64_t *okey,
-			       const int64_t *ckey, const int32_t *odate,
-			       int64_t n, int32_t date_lo, int32_t date_hi,
-			       DeviceHashTable cust, int64_t *out_k,
-			       int64_t *out_p, unsigned long long *out_count)
-{
-	hipLaunchKernelGGL(k_orders_match_compact_q5, dim3(grid_for(n)),
-			   dim3(THREADS), 0, s, okey, ckey, odate, n,
-			   date_lo, date_hi, cust.keys, cust.payload,
-			   cust.nslots, out_k, out_p, out_count);
-	return hipGetLastError();
-}
-
-/* widen u8 vals to i64 pay alongside keys (customer map staging) */
-__global__ void
-k_kv_widen(const int64_t *__restrict__ keys, const uint8_t *__restrict__ v8,
-	   int64_t n, int64_t *__restrict__ out_k,
-	   int64_t *__restrict__ out_p)
-{
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
-	{
-		out_k[i] = nt_ld64(&keys[i]);
-		out_p[i] = v8[i];
-	}
-}
-
-hipError_t
-launch_kv_widen(hipStream_t s, const int64_t *keys, const uint8_t *v8,
-		int64_t n, int64_t *out_k, int64_t *out_p)
-{
-	hipLaunchKernelGGL(k_kv_widen, dim3(grid_for(n)), dim3(THREADS), 0,
-			   s, keys, v8, n, out_k, out_p);
-	return hipGetLastError();
-}
-
 /* count-helper launchers used by engine_abi.cpp */
 hipError_t
 launch_count_filter_u8(hipStream_t s, const uint8_t *col, uint8_t val,
