@@ -1343,12 +1343,43 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		st.hbm_bytes += cu->nrows * 9;
 	}
 
-	/* 2. supplier table: s_suppkey → s_nationkey, in-region only */
+	/* 2. supplier table: s_suppkey → s_nationkey, in-region only.
+	 * Dense keys (max <= 8x rows) use a u8 direct-map array in L2 —
+	 * one byte load replaces bloom+table probes; hash fallback kept
+	 * for sparse keys. */
 	DeviceHashTable supp{};
+	uint8_t *supp_dense = nullptr;
+	int64_t supp_dense_len = 0;
+	{
+		unsigned long long maxk = 0;
+
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_max_i64(e.stream, s_sk, su->nrows, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &maxk));
+		if (su->nrows > 0 && maxk > 0 &&
+		    (int64_t) maxk <= 8 * su->nrows + 16)
+			supp_dense_len = (int64_t) maxk + 1;
+	}
 	{
 		Timed tm(e.stream);
 
-		if (!exch)
+		if (supp_dense_len && !exch)
+		{
+			supp_dense = (uint8_t *)
+				p->sget("supp.dense", (size_t) supp_dense_len);
+			if (!supp_dense)
+				return fail(GG_ENOMEM, "supp dense");
+			GG_HIP(hipMemsetAsync(supp_dense, 0xff,
+					      (size_t) supp_dense_len,
+					      e.stream));
+			GG_HIP(launch_supp_dense_fill(e.stream, s_sk, s_nk,
+						      su->nrows, region_of,
+						      regionkey, supp_dense,
+						      supp_dense_len));
+			GG_HIP(hipStreamSynchronize(e.stream));
+		}
+		else if (!exch)
 		{
 			if (!p->supp_slots)
 				p->supp_slots =
@@ -1463,6 +1494,56 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 						e.stream, base,
 						base + (maxc + 1),
 						(int64_t) counts[r], supp));
+			}
+			/* dense direct-map over the gathered pairs too,
+			 * but only after checking the GLOBAL max key stays
+			 * within the density bound (the pre-exchange check
+			 * saw only the local shard) */
+			supp_dense_len = 0;
+			{
+				unsigned long long gmax = 0;
+
+				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+				for (int r = 0; r < nseg; r++)
+				{
+					int64_t *base =
+						gs + per + (size_t) r * per;
+
+					if (counts[r])
+						GG_HIP(launch_max_i64(
+							e.stream, base,
+							(int64_t) counts[r],
+							ctr));
+				}
+				GG_HIP(hipStreamSynchronize(e.stream));
+				GG_TRY(read_counter(ctr, &gmax));
+				if (gmax > 0 && (int64_t) gmax <=
+				    8 * su->nrows * (int64_t) nseg + 16)
+					supp_dense_len = (int64_t) gmax + 1;
+			}
+			if (supp_dense_len)
+			{
+				supp_dense = (uint8_t *)
+					p->sget("supp.dense",
+						(size_t) supp_dense_len);
+				if (!supp_dense)
+					return fail(GG_ENOMEM, "supp dense");
+				GG_HIP(hipMemsetAsync(supp_dense, 0xff,
+						      (size_t) supp_dense_len,
+						      e.stream));
+				for (int r = 0; r < nseg; r++)
+				{
+					int64_t *base =
+						gs + per + (size_t) r * per;
+
+					if (counts[r])
+						GG_HIP(launch_supp_dense_fill_pairs(
+							e.stream, base,
+							base + (maxc + 1),
+							(int64_t) counts[r],
+							supp_dense,
+							supp_dense_len));
+				}
 			}
 			GG_HIP(hipStreamSynchronize(e.stream));
 		}
@@ -1712,6 +1793,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 
 		GG_HIP(launch_probe_lineitem_q5(e.stream, l_ok, l_sk, l_pc,
 						l_dc, li->nrows, ord, supp,
+						supp_dense, supp_dense_len,
 						acc, ctr));
 		double ms = tm.stop();
 		GG_TRY(read_counter(ctr, &njoin));

@@ -236,8 +236,19 @@ hipError_t launch_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
 				    const int64_t *skey, const int64_t *price,
 				    const int64_t *disc, int64_t n,
 				    DeviceHashTable ord, DeviceHashTable supp,
+				    const uint8_t *supp_dense,
+				    int64_t supp_dense_len,
 				    unsigned long long *acc,
 				    unsigned long long *join_rows);
+hipError_t launch_max_i64(hipStream_t s, const int64_t *v, int64_t n,
+			  unsigned long long *out_max);
+hipError_t launch_supp_dense_fill(hipStream_t s, const int64_t *suppkey,
+				  const uint8_t *snation, int64_t n,
+				  const uint8_t *region_of, uint8_t regionkey,
+				  uint8_t *dense, int64_t dense_len);
+hipError_t launch_supp_dense_fill_pairs(hipStream_t s, const int64_t *sk,
+					const int64_t *sn, int64_t n,
+					uint8_t *dense, int64_t dense_len);
 
 hipError_t launch_q1(hipStream_t s, const int32_t *shipdate,
 		     const uint8_t *rflag, const uint8_t *lstatus,

@@ -1631,6 +1631,8 @@ void k_probe_lineitem_q5(const int64_t *__restrict__ okey,
 			 uint64_t supp_slots,
 			 const unsigned long long *__restrict__ supp_bloom,
 			 uint64_t supp_bwords,
+			 const uint8_t *__restrict__ supp_dense,
+			 int64_t supp_dense_len,
 			 unsigned long long *__restrict__ acc /* [25][3] */,
 			 unsigned long long *join_rows)
 {
@@ -1652,20 +1654,38 @@ void k_probe_lineitem_q5(const int64_t *__restrict__ okey,
 		 * lane, so the cheap rejections come first */
 		int64_t sk = nt_ld64(&skey[i]);
 		int64_t ok = nt_ld64(&okey[i]);
-		uint32_t hsk = gg_hashint8(sk);
-		uint32_t hok = gg_hashint8(ok);
 		unsigned long long snat, onat;
 
-		if (supp_bloom && !bloom_maybe(supp_bloom, supp_bwords, hsk))
-			continue;
-		if (ord_bloom && !bloom_maybe(ord_bloom, ord_bwords, hok))
-			continue;
-		if (!ht_lookup_b(supp_keys, supp_pay, supp_slots, nullptr, 0,
-				 sk, &snat))
-			continue;
-		if (!ht_lookup_b(ord_keys, ord_pay, ord_slots, nullptr, 0,
-				 ok, &onat))
-			continue;
+		if (supp_dense)
+		{
+			/* dense path: one L2-resident byte lookup */
+			if (sk < 0 || sk >= supp_dense_len)
+				continue;
+			snat = supp_dense[sk];
+			if (snat == 255)
+				continue;
+		}
+		else
+		{
+			uint32_t hsk = gg_hashint8(sk);
+
+			if (supp_bloom &&
+			    !bloom_maybe(supp_bloom, supp_bwords, hsk))
+				continue;
+			if (!ht_lookup_b(supp_keys, supp_pay, supp_slots,
+					 nullptr, 0, sk, &snat))
+				continue;
+		}
+		{
+			uint32_t hok = gg_hashint8(ok);
+
+			if (ord_bloom &&
+			    !bloom_maybe(ord_bloom, ord_bwords, hok))
+				continue;
+			if (!ht_lookup_b(ord_keys, ord_pay, ord_slots,
+					 nullptr, 0, ok, &onat))
+				continue;
+		}
 		if (snat != onat)	/* c_nationkey = s_nationkey */
 			continue;
 		joined++;
@@ -1703,15 +1723,16 @@ hipError_t
 launch_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
 			 const int64_t *skey, const int64_t *price,
 			 const int64_t *disc, int64_t n, DeviceHashTable ord,
-			 DeviceHashTable supp, unsigned long long *acc,
+			 DeviceHashTable supp, const uint8_t *supp_dense,
+			 int64_t supp_dense_len, unsigned long long *acc,
 			 unsigned long long *join_rows)
 {
 	hipLaunchKernelGGL(k_probe_lineitem_q5, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, okey, skey, price, disc, n,
 			   ord.keys, ord.payload, ord.nslots, ord.bloom,
 			   ord.bloom_words, supp.keys, supp.payload,
-			   supp.nslots, supp.bloom, supp.bloom_words, acc,
-			   join_rows);
+			   supp.nslots, supp.bloom, supp.bloom_words,
+			   supp_dense, supp_dense_len, acc, join_rows);
 	return hipGetLastError();
 }
 
@@ -1880,6 +1901,108 @@ launch_fill_u64(hipStream_t s, unsigned long long *p, uint64_t n,
 {
 	hipLaunchKernelGGL(k_fill_u64, dim3(grid_for((int64_t) n)),
 			   dim3(THREADS), 0, s, p, n, v);
+	return hipGetLastError();
+}
+
+
+/* dense supplier side: when s_suppkey is (near-)dense the whole
+ * in-region supplier map collapses to a u8 array (suppkey -> nation,
+ * 255 = absent) that lives in L2 — one byte load replaces bloom+table
+ * probes.  Guarded by max_key <= 8*nrows at build; hash fallback
+ * otherwise. */
+__global__ void
+k_max_i64(const int64_t *__restrict__ v, int64_t n,
+	  unsigned long long *out_max)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long m = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t x = nt_ld64(&v[i]);
+
+		if (x > 0 && (unsigned long long) x > m)
+			m = (unsigned long long) x;
+	}
+	for (int off = 32; off; off >>= 1)
+	{
+		unsigned long long o = __shfl_down(m, off, 64);
+
+		if (o > m)
+			m = o;
+	}
+	if ((threadIdx.x & 63) == 0 && m)
+		atomicMax(out_max, m);
+}
+
+hipError_t
+launch_max_i64(hipStream_t s, const int64_t *v, int64_t n,
+	       unsigned long long *out_max)
+{
+	hipLaunchKernelGGL(k_max_i64, dim3(grid_for(n)), dim3(THREADS), 0, s,
+			   v, n, out_max);
+	return hipGetLastError();
+}
+
+__global__ void
+k_supp_dense_fill(const int64_t *__restrict__ suppkey,
+		  const uint8_t *__restrict__ snation, int64_t n,
+		  const uint8_t *__restrict__ region_of, uint8_t regionkey,
+		  uint8_t *__restrict__ dense, int64_t dense_len)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		uint8_t sn = snation[i];
+		int64_t k = suppkey[i];
+
+		if (k < 0 || k >= dense_len)
+			continue;
+		if (sn < 25 && region_of[sn] == regionkey)
+			dense[k] = sn;
+	}
+}
+
+hipError_t
+launch_supp_dense_fill(hipStream_t s, const int64_t *suppkey,
+		       const uint8_t *snation, int64_t n,
+		       const uint8_t *region_of, uint8_t regionkey,
+		       uint8_t *dense, int64_t dense_len)
+{
+	hipLaunchKernelGGL(k_supp_dense_fill, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, suppkey, snation, n,
+			   region_of, regionkey, dense, dense_len);
+	return hipGetLastError();
+}
+
+/* dense fill from broadcast (suppkey, nation-as-i64) pairs */
+__global__ void
+k_supp_dense_fill_pairs(const int64_t *__restrict__ sk,
+			const int64_t *__restrict__ sn, int64_t n,
+			uint8_t *__restrict__ dense, int64_t dense_len)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k = sk[i];
+
+		if (k >= 0 && k < dense_len)
+			dense[k] = (uint8_t) sn[i];
+	}
+}
+
+hipError_t
+launch_supp_dense_fill_pairs(hipStream_t s, const int64_t *sk,
+			     const int64_t *sn, int64_t n, uint8_t *dense,
+			     int64_t dense_len)
+{
+	hipLaunchKernelGGL(k_supp_dense_fill_pairs, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, sk, sn, n, dense, dense_len);
 	return hipGetLastError();
 }
 
