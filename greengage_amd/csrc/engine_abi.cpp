@@ -1958,6 +1958,120 @@ gg_engine_stats(gg_pipeline h, gg_kernel_stat *out, int cap, int *out_n)
 
 
 
+
+/* ---------------- AOCS datum-stream decode (ABI surface) ---------------- */
+
+extern "C" gg_status
+gg_engine_aocs_decode(const uint8_t *stream, int64_t stream_len, int version,
+		      int datumlen, void *out_vals, int out_width,
+		      uint8_t *out_nulls, int64_t cap, int64_t *out_nrows)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!stream || stream_len < 0 || (version < 0 || version > 2) ||
+	    (datumlen != 4 && datumlen != 8) ||
+	    (out_width != 4 && out_width != 8) || !out_vals || !out_nulls ||
+	    !out_nrows)
+		return fail(GG_EINVAL, "bad aocs_decode args");
+
+	/* host parse of the frame stream into per-block descriptors */
+	std::vector<int64_t> offs, out_offs;
+	std::vector<int32_t> sizes, rows;
+	int64_t pos = 0, total_rows = 0;
+
+	while (pos < stream_len)
+	{
+		int32_t sz, rc;
+
+		if (pos + 8 > stream_len)
+			return fail(GG_EINVAL, "truncated frame header");
+		std::memcpy(&sz, stream + pos, 4);
+		std::memcpy(&rc, stream + pos + 4, 4);
+		if (sz < 16 || pos + 8 + sz > stream_len || rc < 0)
+			return fail(GG_EINVAL, "bad frame at %lld",
+				    (long long) pos);
+		offs.push_back(pos + 8);
+		sizes.push_back(sz);
+		rows.push_back(rc);
+		out_offs.push_back(total_rows);
+		total_rows += rc;
+		pos += 8 + sz;
+	}
+	if (total_rows > cap)
+		return fail(GG_EINVAL, "cap %lld < rows %lld",
+			    (long long) cap, (long long) total_rows);
+	*out_nrows = total_rows;
+	if (total_rows == 0)
+		return GG_OK;
+
+	uint8_t *d_stream = nullptr;
+	int64_t *d_offs = nullptr, *d_oo = nullptr;
+	int32_t *d_sizes = nullptr, *d_rows = nullptr;
+	void *d_vals = nullptr;
+	uint8_t *d_nulls = nullptr;
+	unsigned long long *d_err = nullptr;
+	size_t nb = offs.size();
+	gg_status st = GG_OK;
+
+	GG_HIP(hipMalloc((void **) &d_stream, (size_t) stream_len));
+	GG_HIP(hipMalloc((void **) &d_offs, nb * 8));
+	GG_HIP(hipMalloc((void **) &d_oo, nb * 8));
+	GG_HIP(hipMalloc((void **) &d_sizes, nb * 4));
+	GG_HIP(hipMalloc((void **) &d_rows, nb * 4));
+	GG_HIP(hipMalloc(&d_vals, (size_t) total_rows * out_width));
+	GG_HIP(hipMalloc((void **) &d_nulls, (size_t) total_rows));
+	GG_HIP(hipMalloc((void **) &d_err, 8));
+	GG_HIP(hipMemcpy(d_stream, stream, (size_t) stream_len,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_offs, offs.data(), nb * 8, hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_oo, out_offs.data(), nb * 8,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_sizes, sizes.data(), nb * 4,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_rows, rows.data(), nb * 4, hipMemcpyHostToDevice));
+	GG_HIP(hipMemset(d_err, 0, 8));
+	{
+		hipError_t he = launch_dsb_decode(
+			e.stream, d_stream, d_offs, d_sizes, d_rows, d_oo,
+			(int32_t) nb, version, datumlen, d_vals, d_nulls,
+			out_width, d_err);
+
+		if (he != hipSuccess)
+			st = fail(GG_EGPU, "dsb_decode: %s",
+				  hipGetErrorString(he));
+	}
+	if (st == GG_OK)
+	{
+		unsigned long long herr = 0;
+
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(&herr, d_err, 8, hipMemcpyDeviceToHost));
+		if (herr)
+			st = fail(GG_EINVAL,
+				  "block decode error mask 0x%llx", herr);
+		else
+		{
+			GG_HIP(hipMemcpy(out_vals, d_vals,
+					 (size_t) total_rows * out_width,
+					 hipMemcpyDeviceToHost));
+			GG_HIP(hipMemcpy(out_nulls, d_nulls,
+					 (size_t) total_rows,
+					 hipMemcpyDeviceToHost));
+		}
+	}
+	(void) hipFree(d_stream);
+	(void) hipFree(d_offs);
+	(void) hipFree(d_oo);
+	(void) hipFree(d_sizes);
+	(void) hipFree(d_rows);
+	(void) hipFree(d_vals);
+	(void) hipFree(d_nulls);
+	(void) hipFree(d_err);
+	return st;
+}
+
 /* ---------------- general hash group-by (ABI surface) ---------------- */
 
 extern "C" gg_status

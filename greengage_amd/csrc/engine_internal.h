@@ -341,6 +341,13 @@ hipError_t launch_groupby_compact(hipStream_t s,
 				  uint64_t nslots, int64_t *out_keys,
 				  int64_t *out_sums, int64_t *out_cnts,
 				  unsigned long long *out_n, uint64_t cap);
+hipError_t launch_dsb_decode(hipStream_t s, const uint8_t *stream,
+			     const int64_t *offsets, const int32_t *sizes,
+			     const int32_t *rowcounts,
+			     const int64_t *out_offsets, int32_t nblocks,
+			     int version, int datumlen, void *out_vals,
+			     uint8_t *out_nulls, int out_width,
+			     unsigned long long *err);
 hipError_t launch_radix_sort_pass(hipStream_t s,
 				  const unsigned long long *keys,
 				  const unsigned long long *pay, int64_t n,

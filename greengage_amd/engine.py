@@ -109,6 +109,9 @@ def _load():
     lib.gg_engine_hash_groupby_i64.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, I64, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_void_p, I64, ctypes.POINTER(I64)]
+    lib.gg_engine_aocs_decode.argtypes = [
+        ctypes.c_void_p, I64, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_void_p, I64, ctypes.POINTER(I64)]
     lib.gg_engine_avg_str.argtypes = [U64, I64, ctypes.c_int, I64,
                                       ctypes.c_char_p]
     return lib
@@ -295,6 +298,21 @@ class Engine:
         raw = self.execute_raw(p, 16)
         return (int.from_bytes(raw[0:8], "little", signed=True),
                 int.from_bytes(raw[8:16], "little", signed=True))
+
+    # ---- AOCS datum-stream decode (on-disk columnar blocks) ----
+    @staticmethod
+    def aocs_decode(stream, version, datumlen, nmax, out_width=8):
+        import numpy as np
+        stream = np.ascontiguousarray(stream, dtype=np.uint8)
+        vals = np.empty(nmax, np.int32 if out_width == 4 else np.int64)
+        nulls = np.empty(nmax, np.uint8)
+        n = I64()
+        _check(lib().gg_engine_aocs_decode(
+            stream.ctypes.data_as(ctypes.c_void_p), len(stream), version,
+            datumlen, vals.ctypes.data_as(ctypes.c_void_p), out_width,
+            nulls.ctypes.data_as(ctypes.c_void_p), nmax,
+            ctypes.byref(n)), "aocs_decode")
+        return vals[:n.value], nulls[:n.value]
 
     # ---- general hash group-by (arbitrary int64 keys, SUM+COUNT) ----
     @staticmethod

@@ -201,6 +201,19 @@ void gg_engine_numeric_str(uint64_t lo, int64_t hi, int scale, char *buf64);
 void gg_engine_avg_str(uint64_t sum_lo, int64_t sum_hi, int sum_scale,
 		       int64_t count, char *buf64);
 
+/* AOCS datum-stream block decode on GPU: decodes a
+ * stream of reference-format blocks (datumstreamblock.c content layer;
+ * versions Orig=0 / Dense=1 / Dense_Enhanced=2 with null bitmap, RLE
+ * and delta compression) for fixed-length int32/int64 columns into
+ * host value/null arrays.  Stream framing: repeat
+ * [int32 block_size][int32 row_count][block bytes]; the Append-Only
+ * Storage block-header layer and zlib/zstd codecs are the remaining
+ * sub-layers.  out_width 4 or 8. */
+gg_status gg_engine_aocs_decode(const uint8_t *stream, int64_t stream_len,
+				int version, int datumlen, void *out_vals,
+				int out_width, uint8_t *out_nulls,
+				int64_t cap, int64_t *out_nrows);
+
 /* General hash group-by (execHHashagg.c find-or-create semantics on
  * arbitrary int64 keys, SUM+COUNT transitions): host buffers in,
  * groups out sorted by key ascending.  Keys may be any int64 except

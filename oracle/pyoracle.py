@@ -214,6 +214,65 @@ def ref():
     return _ref
 
 
+_dsb = None
+
+
+def dsb_ref():
+    """The reference's own AOCS datum-stream codec (datumstreamblock.c
+    compiled in place); None when the prebuilt .so is missing and the
+    reference tree is absent."""
+    global _dsb
+    if _dsb is None:
+        path = os.path.join(_DIR, "_ref", "libpg_dsbref.so")
+        if not os.path.exists(path) and os.path.isdir("/root/reference"):
+            _build()
+        if not os.path.exists(path):
+            return None
+        _dsb = ctypes.CDLL(path)
+        D = _dsb
+        D.ref_dsb_encode.restype = ctypes.c_int
+        D.ref_dsb_encode.argtypes = [
+            _P_I64, _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int, I32, _P_U8, I64, ctypes.POINTER(I64),
+            ctypes.POINTER(ctypes.c_int32)]
+        D.ref_dsb_decode.restype = ctypes.c_int
+        D.ref_dsb_decode.argtypes = [
+            _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            _P_I64, _P_U8, I64, ctypes.POINTER(I64)]
+    return _dsb
+
+
+def dsb_encode(vals, nulls, datumlen, version, rle, delta, blocksz=32768):
+    """Encode with the REFERENCE writer; returns the framed stream."""
+    D = dsb_ref()
+    assert D is not None, "reference dsb codec missing"
+    v = np.ascontiguousarray(vals, np.int64)
+    nl = np.ascontiguousarray(nulls, np.uint8)
+    cap = max(1 << 21, len(v) * 16)
+    out = np.zeros(cap, np.uint8)
+    olen = I64()
+    nb = ctypes.c_int32()
+    rc = D.ref_dsb_encode(v, nl, len(v), datumlen, version, rle, delta,
+                          blocksz, out, cap, ctypes.byref(olen),
+                          ctypes.byref(nb))
+    assert rc == 0, rc
+    return out[:olen.value].copy(), nb.value
+
+
+def dsb_decode(stream, datumlen, version, rle, nmax):
+    """Decode with the REFERENCE reader."""
+    D = dsb_ref()
+    assert D is not None, "reference dsb codec missing"
+    stream = np.ascontiguousarray(stream, np.uint8)
+    dv = np.zeros(nmax, np.int64)
+    dn = np.zeros(nmax, np.uint8)
+    n = I64()
+    rc = D.ref_dsb_decode(stream, len(stream), datumlen, version, rle,
+                          dv, dn, nmax, ctypes.byref(n))
+    assert rc == 0, rc
+    return dv[:n.value], dn[:n.value]
+
+
 def pgdate(y, m, d):
     return lib().gg_oracle_pgdate(y, m, d)
 

@@ -1,0 +1,219 @@
+/*
+ * Plain-C export wrapper around the REFERENCE's own datumstreamblock.c
+ * (compiled in place — see Makefile): the genuine AOCS columnar block
+ * encoder/decoder, used as the parity oracle for the engine's GPU
+ * datumstream decoder.
+ *
+ * Streams are a simple framing of reference-encoded blocks:
+ *   repeat: [int32 block_size][int32 row_count][block bytes...]
+ * (the framing replaces the Append-Only Storage block headers, which
+ * belong to cdbappendonlystorageformat.c — a separate layer; the datum
+ * stream CONTENT bytes here are exactly what the reference writes).
+ *
+ * Fixed-length by-value types only (int32/int64 — the hot path's
+ * column types); datumlen 4 or 8.
+ */
+#include "postgres.h"
+#include "access/tupmacs.h"
+#include "utils/datumstreamblock.h"
+
+/* GUC definitions the compiled reference code references */
+bool		Debug_appendonly_print_insert = false;
+bool		Debug_appendonly_print_insert_tuple = false;
+bool		Debug_appendonly_print_scan = false;
+bool		Debug_appendonly_print_scan_tuple = false;
+bool		Debug_datumstream_write_print_small_varlena_info = false;
+bool		Debug_datumstream_write_print_large_varlena_info = false;
+bool		Debug_datumstream_read_print_varlena_info = false;
+bool		Debug_datumstream_write_use_small_initial_buffers = false;
+bool		Debug_datumstream_block_read_check_integrity = true;
+bool		Debug_datumstream_block_write_check_integrity = true;
+bool		Debug_datumstream_read_check_large_varlena_integrity = false;
+bool		Debug_datumstream_write_check_large_varlena_integrity = false;
+
+MemoryContext CurrentMemoryContext = NULL;
+
+int
+stub_errmsg(const char *fmt,...)
+{
+	(void) fmt;
+	return 0;
+}
+
+int
+stub_errfinish(int level)
+{
+	if (level >= ERROR)
+	{
+		fprintf(stderr, "reference dsb code raised ERROR (level %d)\n",
+				level);
+		abort();
+	}
+	return 0;
+}
+
+/* varlena helpers: never reached for by-value fixed-length columns */
+void
+varattrib_untoast_ptr_len(Datum d, char **datastart, int *len, void **tofree)
+{
+	(void) d; (void) datastart; (void) len; (void) tofree;
+	fprintf(stderr, "varattrib_untoast_ptr_len: varlena unsupported in wrapper\n");
+	abort();
+}
+
+bool
+value_type_could_short(Pointer ptr, Oid typid)
+{
+	(void) ptr; (void) typid;
+	return false;
+}
+
+#define MAXDATUM_ORIG 0x3FFF	/* AOSmallContentHeader_MaxRowCount */
+#define MAXDATUM_DENSE 0x3FFFFFFF	/* AONonBulkDense..MaxLargeRowCount */
+
+static void
+typeinfo_for(DatumStreamTypeInfo * ti, int datumlen)
+{
+	ti->datumlen = datumlen;
+	ti->typid = (datumlen == 4) ? 23 : 20;	/* int4 / int8 oids */
+	ti->align = (datumlen == 4) ? 'i' : 'd';
+	ti->byval = true;
+}
+
+/*
+ * Encode n values (vals[i] significant low datumlen bytes; nulls[i]
+ * nonzero = NULL) into the framed stream.  version: 0 Orig, 1 Dense,
+ * 2 Dense_Enhanced; rle/delta per DatumStreamVersion capabilities.
+ * Returns 0, or -1 if out_cap too small.
+ */
+int
+ref_dsb_encode(const int64 *vals, const uint8 *nulls, int64 n, int datumlen,
+			   int version, int rle, int delta, int32 max_block_size,
+			   uint8 *out, int64 out_cap, int64 *out_len,
+			   int32 *out_nblocks)
+{
+	DatumStreamTypeInfo ti;
+	DatumStreamBlockWrite dsw;
+	int64		pos = 0;
+	int32		nblocks = 0;
+	int64		i = 0;
+
+	memset(&dsw, 0, sizeof(dsw));
+	typeinfo_for(&ti, datumlen);
+	DatumStreamBlockWrite_Init(&dsw, &ti, (DatumStreamVersion) version,
+							   rle != 0, delta != 0,
+							   (version == 0) ? MAXDATUM_ORIG : MAXDATUM_ORIG,
+							   (version == 0) ? MAXDATUM_ORIG : MAXDATUM_DENSE,
+							   max_block_size,
+							   NULL, NULL, NULL, NULL);
+	DatumStreamBlockWrite_GetReady(&dsw);
+
+	while (i < n)
+	{
+		Datum		d;
+		bool		isnull = nulls && nulls[i];
+		void	   *tofree = NULL;
+		int			res;
+
+		d = (datumlen == 4) ? Int32GetDatum((int32) vals[i])
+			: Int64GetDatum(vals[i]);
+		res = DatumStreamBlockWrite_Put(&dsw, isnull ? 0 : d, isnull,
+										&tofree);
+		if (res >= 0)
+		{
+			i++;
+			continue;
+		}
+		/* block full: emit it */
+		{
+			int32		rowcount = DatumStreamBlockWrite_Nth(&dsw);
+			int64		sz;
+
+			if (pos + 8 + max_block_size > out_cap)
+				return -1;
+			sz = DatumStreamBlockWrite_Block(&dsw, out + pos + 8);
+			memcpy(out + pos, &sz, 4);
+			memcpy(out + pos + 4, &rowcount, 4);
+			pos += 8 + sz;
+			nblocks++;
+			DatumStreamBlockWrite_GetReady(&dsw);
+		}
+	}
+	if (DatumStreamBlockWrite_Nth(&dsw) > 0)
+	{
+		int32		rowcount = DatumStreamBlockWrite_Nth(&dsw);
+		int64		sz;
+
+		if (pos + 8 + max_block_size > out_cap)
+			return -1;
+		sz = DatumStreamBlockWrite_Block(&dsw, out + pos + 8);
+		memcpy(out + pos, &sz, 4);
+		memcpy(out + pos + 4, &rowcount, 4);
+		pos += 8 + sz;
+		nblocks++;
+	}
+	DatumStreamBlockWrite_Finish(&dsw);
+	*out_len = pos;
+	*out_nblocks = nblocks;
+	return 0;
+}
+
+/*
+ * Decode the framed stream back with the REFERENCE reader.
+ * Returns 0, or -1 on capacity, -2 on row-count mismatch.
+ */
+int
+ref_dsb_decode(const uint8 *stream, int64 stream_len, int datumlen,
+			   int version, int rle, int64 *out_vals, uint8 *out_nulls,
+			   int64 cap, int64 *out_n)
+{
+	DatumStreamTypeInfo ti;
+	DatumStreamBlockRead dsr;
+	int64		pos = 0;
+	int64		nout = 0;
+
+	memset(&dsr, 0, sizeof(dsr));
+	typeinfo_for(&ti, datumlen);
+	DatumStreamBlockRead_Init(&dsr, &ti, (DatumStreamVersion) version,
+							  rle != 0, NULL, NULL, NULL, NULL);
+
+	while (pos < stream_len)
+	{
+		int32		sz,
+					rowcount;
+		bool		adjusted = false;
+		int32		adjustedRowCount = 0;
+		int			have;
+
+		memcpy(&sz, stream + pos, 4);
+		memcpy(&rowcount, stream + pos + 4, 4);
+		pos += 8;
+		DatumStreamBlockRead_Reset(&dsr);	/* nth = -1 before GetReady */
+		DatumStreamBlockRead_GetReady(&dsr, (uint8 *) stream + pos, sz,
+									  1 /* firstRowNum */ , rowcount,
+									  &adjusted, &adjustedRowCount);
+		pos += sz;
+		for (have = 0; have < rowcount; have++)
+		{
+			Datum		d = 0;
+			bool		isnull = false;
+
+			if (DatumStreamBlockRead_Advance(&dsr) == 0)
+				return -2;
+			DatumStreamBlockRead_Get(&dsr, &d, &isnull);
+			if (nout >= cap)
+				return -1;
+			out_nulls[nout] = isnull ? 1 : 0;
+			if (isnull)
+				out_vals[nout] = 0;
+			else
+				out_vals[nout] = (datumlen == 4)
+					? (int64) DatumGetInt32(d)
+					: DatumGetInt64(d);
+			nout++;
+		}
+	}
+	DatumStreamBlockRead_Finish(&dsr);
+	*out_n = nout;
+	return 0;
+}

@@ -1,0 +1,201 @@
+/*
+ * Stand-in "postgres.h" environment for compiling the REFERENCE's own
+ * src/backend/utils/datumstream/datumstreamblock.c in place (see
+ * ../Makefile) — the genuine AOCS block encoder/decoder becomes the
+ * parity oracle for the engine's GPU datumstream decoder.
+ * Only the ABI surface that file touches is re-declared; semantics of
+ * each macro follow the PostgreSQL 9.4 originals the reference builds
+ * against.  Little-endian hosts only.
+ */
+#ifndef ORACLE_STUB_DSB_POSTGRES_H
+#define ORACLE_STUB_DSB_POSTGRES_H
+
+#include <stdint.h>
+#include <stddef.h>
+#include <string.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <stdbool.h>
+
+typedef uintptr_t Datum;
+typedef uint8_t uint8;
+typedef uint16_t uint16;
+typedef uint32_t uint32;
+typedef uint64_t uint64;
+typedef int8_t int8;
+typedef int16_t int16;
+typedef int32_t int32;
+typedef int64_t int64;
+typedef float float4;
+typedef double float8;
+typedef uint32 Oid;
+typedef char *Pointer;
+
+#define InvalidOid ((Oid) 0)
+#define Assert(x) do { if (!(x)) { fprintf(stderr, "Assert failed: %s (%s:%d)\n", #x, __FILE__, __LINE__); abort(); } } while (0)
+#define AssertImply(a, b) Assert(!(a) || (b))
+#define StaticAssertStmt(cond, msg) ((void) 0)
+#define MemSet(start, val, len) memset(start, val, len)
+#define lengthof(array) (sizeof(array) / sizeof((array)[0]))
+#define pg_attribute_always_inline inline
+#define INT64_FORMAT "%ld"
+
+/* ---- error reporting: print + abort on >= ERROR ---- */
+#define DEBUG5 10
+#define DEBUG4 11
+#define DEBUG3 12
+#define DEBUG2 13
+#define DEBUG1 14
+#define LOG 15
+#define INFO 17
+#define NOTICE 18
+#define WARNING 19
+#define ERROR 20
+#define FATAL 21
+#define PANIC 22
+
+extern int stub_errmsg(const char *fmt, ...);
+extern int stub_errfinish(int level);
+#define elog(level, ...) \
+	do { \
+		if ((level) >= WARNING) \
+			{ fprintf(stderr, "elog(%d): ", level); fprintf(stderr, __VA_ARGS__); fprintf(stderr, "\n"); } \
+		if ((level) >= ERROR) abort(); \
+	} while (0)
+#define ereport(level, rest) \
+	do { \
+		(void) (rest); \
+		(void) stub_errfinish(level); \
+	} while (0)
+#define errmsg stub_errmsg
+#define errmsg_internal stub_errmsg
+#define errdetail stub_errmsg
+#define errdetail_internal stub_errmsg
+#define errcontext stub_errmsg
+#define errhint stub_errmsg
+#define errcode(c) (c)
+#define errOmitLocation(b) 0
+#define ERRCODE_INTERNAL_ERROR 0
+#define ERRCODE_GP_INTERNAL_ERROR 0
+#define ERRCODE_APPENDONLY_INTERNAL_ERROR 0
+#define ERRCODE_DATA_CORRUPTED 0
+
+/* ---- memory: malloc-backed palloc; contexts are inert tokens ---- */
+typedef struct MemoryContextData *MemoryContext;
+extern MemoryContext CurrentMemoryContext;
+static inline MemoryContext MemoryContextSwitchTo(MemoryContext ctx)
+{
+	MemoryContext old = CurrentMemoryContext;
+	CurrentMemoryContext = ctx;
+	return old;
+}
+static inline void *palloc(size_t sz) { return malloc(sz); }
+static inline void *palloc0(size_t sz) { return calloc(1, sz); }
+static inline void pfree(void *p) { free(p); }
+
+/* ---- Datum conversions (64-bit, by-value; postgres.h originals) ---- */
+#define DatumGetPointer(X) ((Pointer) (X))
+#define PointerGetDatum(X) ((Datum) (X))
+#define DatumGetUInt8(X) ((uint8) (X))
+#define DatumGetUInt16(X) ((uint16) (X))
+#define DatumGetUInt32(X) ((uint32) (X))
+#define DatumGetInt32(X) ((int32) (X))
+#define DatumGetInt64(X) ((int64) (X))
+#define DatumGetChar(X) ((char) (X))
+#define DatumGetCString(X) ((char *) (X))
+#define Int32GetDatum(X) ((Datum) (uint32) (X))
+#define Int64GetDatum(X) ((Datum) (X))
+#define UInt32GetDatum(X) ((Datum) (X))
+
+/* ---- alignment (c.h originals) ---- */
+#define TYPEALIGN(ALIGNVAL, LEN) \
+	(((uintptr_t) (LEN) + ((ALIGNVAL) - 1)) & ~((uintptr_t) ((ALIGNVAL) - 1)))
+#define SHORTALIGN(LEN) TYPEALIGN(2, (LEN))
+#define INTALIGN(LEN) TYPEALIGN(4, (LEN))
+#define LONGALIGN(LEN) TYPEALIGN(8, (LEN))
+#define DOUBLEALIGN(LEN) TYPEALIGN(8, (LEN))
+#define MAXIMUM_ALIGNOF 8
+#define MAXALIGN(LEN) TYPEALIGN(MAXIMUM_ALIGNOF, (LEN))
+
+/* ---- varlena (postgres.h 9.4, little-endian 1-byte-header forms) ---- */
+typedef struct varlena
+{
+	char		vl_len_[4];
+	char		vl_dat[1];
+} varlena;
+#define VARHDRSZ ((int32) sizeof(int32))
+#define VARHDRSZ_SHORT 1
+
+/* postgres.h 9.4 varattrib_4b is a union with a compressed form */
+typedef union
+{
+	struct
+	{
+		uint32		va_header;
+		char		va_data[1];
+	}			va_4byte;
+	struct
+	{
+		uint32		va_header;
+		uint32		va_rawsize;
+		char		va_data[1];
+	}			va_compressed;
+} varattrib_4b;
+typedef varattrib_4b varattrib_4b_stub;
+typedef struct
+{
+	uint8		va_header;
+	char		va_data[1];
+} varattrib_1b_stub;
+typedef varattrib_1b_stub varattrib_1b;
+
+#define VARATT_IS_4B(PTR)  ((((varattrib_1b_stub *) (PTR))->va_header & 0x03) == 0x00)
+#define VARATT_IS_4B_U(PTR) ((((varattrib_1b_stub *) (PTR))->va_header & 0x03) == 0x00)
+#define VARATT_IS_4B_C(PTR) ((((varattrib_1b_stub *) (PTR))->va_header & 0x03) == 0x02)
+#define VARATT_IS_1B(PTR)  ((((varattrib_1b_stub *) (PTR))->va_header & 0x01) == 0x01)
+#define VARATT_IS_1B_E(PTR) ((((varattrib_1b_stub *) (PTR))->va_header) == 0x01)
+#define VARATT_IS_COMPRESSED(PTR) VARATT_IS_4B_C(PTR)
+#define VARATT_IS_EXTERNAL(PTR) VARATT_IS_1B_E(PTR)
+#define VARATT_IS_SHORT(PTR) VARATT_IS_1B(PTR)
+#define VARATT_IS_EXTENDED(PTR) (!VARATT_IS_4B_U(PTR))
+
+#define VARSIZE_4B(PTR) ((((varattrib_4b_stub *) (PTR))->va_4byte.va_header >> 2) & 0x3FFFFFFF)
+#define VARSIZE_1B(PTR) ((((varattrib_1b_stub *) (PTR))->va_header >> 1) & 0x7F)
+#define VARSIZE(PTR) VARSIZE_4B(PTR)
+#define VARSIZE_SHORT(PTR) VARSIZE_1B(PTR)
+#define VARSIZE_ANY(PTR) \
+	(VARATT_IS_1B(PTR) ? VARSIZE_1B(PTR) : VARSIZE_4B(PTR))
+#define VARDATA_4B(PTR) (((varattrib_4b_stub *) (PTR))->va_4byte.va_data)
+#define VARDATA_1B(PTR) (((varattrib_1b_stub *) (PTR))->va_data)
+#define VARDATA(PTR) VARDATA_4B(PTR)
+#define VARDATA_SHORT(PTR) VARDATA_1B(PTR)
+#define VARDATA_ANY(PTR) \
+	(VARATT_IS_1B(PTR) ? VARDATA_1B(PTR) : VARDATA_4B(PTR))
+#define SET_VARSIZE(PTR, len) \
+	(((varattrib_4b_stub *) (PTR))->va_4byte.va_header = (((uint32) (len)) << 2))
+#define SET_VARSIZE_SHORT(PTR, len) \
+	(((varattrib_1b_stub *) (PTR))->va_header = (((uint8) (len)) << 1) | 0x01)
+#define VARSIZE_TO_SHORT(len) ((len) - VARHDRSZ + VARHDRSZ_SHORT)
+#define VARSIZE_TO_SHORT_D(D) VARSIZE_TO_SHORT(VARSIZE(DatumGetPointer(D)))
+#define VARATT_CAN_MAKE_SHORT(PTR) \
+	(VARATT_IS_4B_U(PTR) && \
+	 (VARSIZE(PTR) - VARHDRSZ + VARHDRSZ_SHORT) <= 0x7F)
+
+/* ---- extras datumstreamblock.c references (real decls:
+ * access/tupmacs.h:202 value_type_could_short,
+ * access/tuptoaster.h:182 varattrib_untoast_ptr_len,
+ * cdbappendonlystorage.h IsAligned) ---- */
+#define IsAligned(ptr, alignment) \
+	(((uintptr_t) (ptr)) % (alignment) == 0)
+typedef struct varatt_external
+{
+	int32		va_rawsize;
+	int32		va_extsize;
+	Oid			va_valueid;
+	Oid			va_toastrelid;
+} varatt_external;
+extern void varattrib_untoast_ptr_len(Datum d, char **datastart, int *len,
+				      void **tofree);
+extern bool value_type_could_short(Pointer ptr, Oid typid);
+
+#endif
