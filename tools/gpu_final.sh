@@ -1,0 +1,6 @@
+#!/bin/bash
+{ timeout 300 python __graft_entry__.py smoke; echo "SMOKE_RC=$?";
+  GG_BIG=1 timeout 600 python -m pytest tests/test_gpu_fullsize.py -m gpu -q -x; echo "BIG_RC=$?";
+  timeout 500 python bench.py --steps 5 --warmup 2; echo "BENCH_RC=$?";
+} > gpurun_out/final.log 2>&1
+grep -E "RC=|passed|failed|smoke ok" gpurun_out/final.log
