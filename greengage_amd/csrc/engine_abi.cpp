@@ -831,7 +831,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 			GG_HIP(launch_count_orders_match(e.stream, o_ck, o_dt,
 							 od->nrows, cutoff,
-							 cust, ctr));
+							 cust, cust_dense,
+							 cust_dlen, ctr));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
 			p->ord_slots = next_pow2(2 * (nmatch + 1));
@@ -855,7 +856,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				      e.stream));
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt, o_pr,
-					   od->nrows, cutoff, cust, ord, ctr));
+					   od->nrows, cutoff, cust, cust_dense,
+					   cust_dlen, ord, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nmatch));
 		double ms = tm.stop();
@@ -969,7 +971,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			return fail(GG_ENOMEM, "match scratch");
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_probe_cust_compact(e.stream, r_ck, r_ok, r_pay,
-						 (int64_t) rtotal, cust, m_ok,
+						 (int64_t) rtotal, cust,
+						 cust_dense, cust_dlen, m_ok,
 						 m_pay, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nm));
@@ -1318,22 +1321,55 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 	GG_HIP(hipMemcpy(region_of, region_of_h, GG_NNATIONS,
 			 hipMemcpyHostToDevice));
 
-	/* 1. customer map: c_custkey → c_nationkey (no filter, no bloom) */
+	/* 1. customer map: c_custkey → c_nationkey.  Dense custkeys use a
+	 * u8 nation array (255 = absent) in L2/L3; hash map fallback. */
 	DeviceHashTable cust{};
+	uint8_t *cust_dense = nullptr;
+	int64_t cust_dlen = 0;
+	{
+		unsigned long long maxk = 0;
+
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_max_i64(e.stream, c_ck, cu->nrows, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &maxk));
+		if (cu->nrows > 0 && maxk > 0 &&
+		    (int64_t) maxk <= 8 * cu->nrows + 16)
+			cust_dlen = (int64_t) maxk + 1;
+	}
 	{
 		Timed tm(e.stream);
 
-		if (!p->cust_slots)
-			p->cust_slots = next_pow2(2 * (uint64_t) (cu->nrows + 1));
-		cust.nslots = p->cust_slots;
-		cust.keys = (unsigned long long *)
-			p->sget("cust.keys", cust.nslots * 8);
-		cust.payload = (unsigned long long *)
-			p->sget("cust.pay", cust.nslots * 8);
-		if (!cust.keys || !cust.payload)
-			return fail(GG_ENOMEM, "cust map");
-		GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8, e.stream));
-		GG_HIP(launch_build_kv(e.stream, c_ck, c_nk, cu->nrows, cust));
+		if (cust_dlen)
+		{
+			cust_dense = (uint8_t *)
+				p->sget("cust.dense", (size_t) cust_dlen);
+			if (!cust_dense)
+				return fail(GG_ENOMEM, "cust dense");
+			GG_HIP(hipMemsetAsync(cust_dense, 0xff,
+					      (size_t) cust_dlen, e.stream));
+			GG_HIP(launch_cust_dense_fill_nat(e.stream, c_ck,
+							  c_nk, cu->nrows,
+							  cust_dense,
+							  cust_dlen));
+		}
+		else
+		{
+			if (!p->cust_slots)
+				p->cust_slots =
+					next_pow2(2 * (uint64_t) (cu->nrows + 1));
+			cust.nslots = p->cust_slots;
+			cust.keys = (unsigned long long *)
+				p->sget("cust.keys", cust.nslots * 8);
+			cust.payload = (unsigned long long *)
+				p->sget("cust.pay", cust.nslots * 8);
+			if (!cust.keys || !cust.payload)
+				return fail(GG_ENOMEM, "cust map");
+			GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8,
+					      e.stream));
+			GG_HIP(launch_build_kv(e.stream, c_ck, c_nk,
+					       cu->nrows, cust));
+		}
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("build_customer_map");
 
@@ -1591,7 +1627,8 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_build_orders_q5(e.stream, o_ok, o_ck, o_dt,
 					      od->nrows, date_lo, date_hi,
-					      cust, ord, ctr));
+					      cust, cust_dense, cust_dlen,
+					      ord, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nmatch));
 		double ms = tm.stop();
@@ -1693,6 +1730,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_probe_cust_map_compact(e.stream, r_ck, r_ok,
 						     (int64_t) rtotal, cust,
+						     cust_dense, cust_dlen,
 						     m_ok, m_nat, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nm));

@@ -222,7 +222,9 @@ hipError_t launch_insert_supp(hipStream_t s, const int64_t *sk,
 hipError_t launch_build_orders_q5(hipStream_t s, const int64_t *okey,
 				  const int64_t *ckey, const int32_t *odate,
 				  int64_t n, int32_t date_lo, int32_t date_hi,
-				  DeviceHashTable cust, DeviceHashTable ord,
+				  DeviceHashTable cust,
+				  const uint8_t *cust_dense,
+				  int64_t cust_dlen, DeviceHashTable ord,
 				  unsigned long long *match_count);
 hipError_t launch_count_date_range(hipStream_t s, const int32_t *odate,
 				   int64_t n, int32_t date_lo,
@@ -230,6 +232,8 @@ hipError_t launch_count_date_range(hipStream_t s, const int32_t *odate,
 hipError_t launch_probe_cust_map_compact(hipStream_t s, const int64_t *ckey,
 					 const int64_t *okey, int64_t n,
 					 DeviceHashTable cust,
+					 const uint8_t *cust_dense,
+					 int64_t cust_dlen,
 					 int64_t *out_okey, int64_t *out_nat,
 					 unsigned long long *out_count);
 hipError_t launch_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
@@ -271,8 +275,16 @@ hipError_t launch_build_set(hipStream_t s, const int64_t *keys,
 hipError_t launch_build_orders(hipStream_t s, const int64_t *okey,
 			       const int64_t *ckey, const int32_t *odate,
 			       const int32_t *prio, int64_t n, int32_t cutoff,
-			       DeviceHashTable cust, DeviceHashTable ord,
+			       DeviceHashTable cust, const uint8_t *cust_dense,
+			       int64_t cust_dlen, DeviceHashTable ord,
 			       unsigned long long *match_count);
+hipError_t launch_cust_dense_fill_seg(hipStream_t s, const int64_t *custkey,
+				      const uint8_t *mktseg, int64_t n,
+				      uint8_t segcode, uint8_t *dense,
+				      int64_t dense_len);
+hipError_t launch_cust_dense_fill_nat(hipStream_t s, const int64_t *custkey,
+				      const uint8_t *nation, int64_t n,
+				      uint8_t *dense, int64_t dense_len);
 hipError_t launch_probe_lineitem(hipStream_t s, const int64_t *okey,
 				 const int32_t *shipdate,
 				 const int64_t *price, const int64_t *disc,
@@ -299,6 +311,8 @@ hipError_t launch_count_filter_u8(hipStream_t s, const uint8_t *col,
 hipError_t launch_count_orders_match(hipStream_t s, const int64_t *ckey,
 				     const int32_t *odate, int64_t n,
 				     int32_t cutoff, DeviceHashTable cust,
+				     const uint8_t *cust_dense,
+				     int64_t cust_dlen,
 				     unsigned long long *out);
 hipError_t launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
 					const int64_t *ckey,
@@ -311,6 +325,8 @@ hipError_t launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
 hipError_t launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,
 				     const int64_t *okey, const int64_t *pay,
 				     int64_t n, DeviceHashTable cust,
+				     const uint8_t *cust_dense,
+				     int64_t cust_dlen,
 				     int64_t *out_okey, int64_t *out_pay,
 				     unsigned long long *out_count);
 hipError_t launch_insert_orders(hipStream_t s, const int64_t *okey,

@@ -636,7 +636,9 @@ k_build_orders(const int64_t *__restrict__ okey,
 	       const unsigned long long *__restrict__ cust_keys,
 	       uint64_t cust_slots,
 	       const unsigned long long *__restrict__ cust_bloom,
-	       uint64_t cust_bwords, unsigned long long *__restrict__ tkeys,
+	       uint64_t cust_bwords,
+	       const uint8_t *__restrict__ cust_dense, int64_t cust_dlen,
+	       unsigned long long *__restrict__ tkeys,
 	       unsigned long long *__restrict__ tpayload, uint64_t nslots,
 	       unsigned long long *__restrict__ bloom, uint64_t bwords,
 	       unsigned long long *match_count)
@@ -656,8 +658,13 @@ k_build_orders(const int64_t *__restrict__ okey,
 
 		if (d >= cutoff)	/* qual: o_orderdate < cutoff */
 			continue;
-		if (!ht_contains_b(cust_keys, cust_slots, cust_bloom,
-				   cust_bwords, ck))
+		if (cust_dense)
+		{
+			if (ck < 0 || ck >= cust_dlen || !cust_dense[ck])
+				continue;
+		}
+		else if (!ht_contains_b(cust_keys, cust_slots, cust_bloom,
+					cust_bwords, ck))
 			continue;
 		matches++;
 		{
@@ -694,13 +701,15 @@ k_build_orders(const int64_t *__restrict__ okey,
 hipError_t
 launch_build_orders(hipStream_t s, const int64_t *okey, const int64_t *ckey,
 		    const int32_t *odate, const int32_t *prio, int64_t n,
-		    int32_t cutoff, DeviceHashTable cust, DeviceHashTable ord,
-		    unsigned long long *match_count)
+		    int32_t cutoff, DeviceHashTable cust,
+		    const uint8_t *cust_dense, int64_t cust_dlen,
+		    DeviceHashTable ord, unsigned long long *match_count)
 {
 	hipLaunchKernelGGL(k_build_orders, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, okey, ckey, odate, prio, n, cutoff,
 			   cust.keys, cust.nslots, cust.bloom,
-			   cust.bloom_words, ord.keys, ord.payload,
+			   cust.bloom_words, cust_dense, cust_dlen,
+			   ord.keys, ord.payload,
 			   ord.nslots, ord.bloom, ord.bloom_words,
 			   match_count);
 	return hipGetLastError();
@@ -731,17 +740,26 @@ k_count_orders_match(const int64_t *__restrict__ ckey,
 		     const unsigned long long *__restrict__ cust_keys,
 		     uint64_t cust_slots,
 		     const unsigned long long *__restrict__ cust_bloom,
-		     uint64_t cust_bwords, unsigned long long *out)
+		     uint64_t cust_bwords,
+		     const uint8_t *__restrict__ cust_dense,
+		     int64_t cust_dlen, unsigned long long *out)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long c = 0;
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
-		if (odate[i] < cutoff &&
-		    ht_contains_b(cust_keys, cust_slots, cust_bloom,
-				  cust_bwords, ckey[i]))
-			c++;
+		if (odate[i] >= cutoff)
+			continue;
+		if (cust_dense)
+		{
+			int64_t ck = ckey[i];
+
+			c += (ck >= 0 && ck < cust_dlen && cust_dense[ck]);
+		}
+		else
+			c += ht_contains_b(cust_keys, cust_slots, cust_bloom,
+					   cust_bwords, ckey[i]);
 	for (int off = 32; off; off >>= 1)
 		c += __shfl_down(c, off, 64);
 	if ((threadIdx.x & 63) == 0 && c)
@@ -1172,7 +1190,9 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 		     const unsigned long long *__restrict__ cust_keys,
 		     uint64_t cust_slots,
 		     const unsigned long long *__restrict__ cust_bloom,
-		     uint64_t cust_bwords, int64_t *__restrict__ out_okey,
+		     uint64_t cust_bwords,
+		     const uint8_t *__restrict__ cust_dense,
+		     int64_t cust_dlen, int64_t *__restrict__ out_okey,
 		     int64_t *__restrict__ out_pay,
 		     unsigned long long *out_count)
 {
@@ -1181,8 +1201,11 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		bool take = ht_contains_b(cust_keys, cust_slots, cust_bloom,
-					  cust_bwords, ckey[i]);
+		int64_t ck = ckey[i];
+		bool take = cust_dense
+			? (ck >= 0 && ck < cust_dlen && cust_dense[ck] != 0)
+			: ht_contains_b(cust_keys, cust_slots, cust_bloom,
+					cust_bwords, ck);
 		unsigned long long idx = wave_append(out_count, take);
 
 		if (!take)
@@ -1195,12 +1218,14 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 hipError_t
 launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,
 			  const int64_t *okey, const int64_t *pay, int64_t n,
-			  DeviceHashTable cust, int64_t *out_okey,
+			  DeviceHashTable cust, const uint8_t *cust_dense,
+			  int64_t cust_dlen, int64_t *out_okey,
 			  int64_t *out_pay, unsigned long long *out_count)
 {
 	hipLaunchKernelGGL(k_probe_cust_compact, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, ckey, okey, pay, n, cust.keys,
 			   cust.nslots, cust.bloom, cust.bloom_words,
+			   cust_dense, cust_dlen,
 			   out_okey, out_pay, out_count);
 	return hipGetLastError();
 }
@@ -1482,7 +1507,9 @@ k_build_orders_q5(const int64_t *__restrict__ okey,
 		  int32_t date_lo, int32_t date_hi,
 		  const unsigned long long *__restrict__ cust_keys,
 		  const unsigned long long *__restrict__ cust_pay,
-		  uint64_t cust_slots, unsigned long long *__restrict__ tkeys,
+		  uint64_t cust_slots,
+		  const uint8_t *__restrict__ cust_dense, int64_t cust_dlen,
+		  unsigned long long *__restrict__ tkeys,
 		  unsigned long long *__restrict__ tpayload, uint64_t nslots,
 		  unsigned long long *__restrict__ bloom, uint64_t bwords,
 		  unsigned long long *match_count)
@@ -1498,8 +1525,16 @@ k_build_orders_q5(const int64_t *__restrict__ okey,
 
 		if (d < date_lo || d >= date_hi)
 			continue;
-		if (!ht_lookup_b(cust_keys, cust_pay, cust_slots, nullptr, 0,
-				 ckey[i], &nat))
+		if (cust_dense)
+		{
+			int64_t ck = ckey[i];
+
+			if (ck < 0 || ck >= cust_dlen || cust_dense[ck] == 255)
+				continue;
+			nat = cust_dense[ck];
+		}
+		else if (!ht_lookup_b(cust_keys, cust_pay, cust_slots, nullptr,
+				      0, ckey[i], &nat))
 			continue;	/* inner join: unmatched drops */
 		matches++;
 		{
@@ -1534,12 +1569,14 @@ hipError_t
 launch_build_orders_q5(hipStream_t s, const int64_t *okey,
 		       const int64_t *ckey, const int32_t *odate, int64_t n,
 		       int32_t date_lo, int32_t date_hi, DeviceHashTable cust,
+		       const uint8_t *cust_dense, int64_t cust_dlen,
 		       DeviceHashTable ord, unsigned long long *match_count)
 {
 	hipLaunchKernelGGL(k_build_orders_q5, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, okey, ckey, odate, n,
 			   date_lo, date_hi, cust.keys, cust.payload,
-			   cust.nslots, ord.keys, ord.payload, ord.nslots,
+			   cust.nslots, cust_dense, cust_dlen,
+			   ord.keys, ord.payload, ord.nslots,
 			   ord.bloom, ord.bloom_words, match_count);
 	return hipGetLastError();
 }
@@ -1579,7 +1616,9 @@ k_probe_cust_map_compact(const int64_t *__restrict__ ckey,
 			 const int64_t *__restrict__ okey, int64_t n,
 			 const unsigned long long *__restrict__ cust_keys,
 			 const unsigned long long *__restrict__ cust_pay,
-			 uint64_t cust_slots, int64_t *__restrict__ out_okey,
+			 uint64_t cust_slots,
+			 const uint8_t *__restrict__ cust_dense,
+			 int64_t cust_dlen, int64_t *__restrict__ out_okey,
 			 int64_t *__restrict__ out_nat,
 			 unsigned long long *out_count)
 {
@@ -1589,8 +1628,20 @@ k_probe_cust_map_compact(const int64_t *__restrict__ ckey,
 	     i < n; i += stride)
 	{
 		unsigned long long nat = 0;
-		bool take = ht_lookup_b(cust_keys, cust_pay, cust_slots,
-					nullptr, 0, ckey[i], &nat);
+		bool take;
+
+		if (cust_dense)
+		{
+			int64_t ck = ckey[i];
+
+			take = (ck >= 0 && ck < cust_dlen &&
+				cust_dense[ck] != 255);
+			if (take)
+				nat = cust_dense[ck];
+		}
+		else
+			take = ht_lookup_b(cust_keys, cust_pay, cust_slots,
+					   nullptr, 0, ckey[i], &nat);
 		unsigned long long idx = wave_append(out_count, take);
 
 		if (!take)
@@ -1603,13 +1654,14 @@ k_probe_cust_map_compact(const int64_t *__restrict__ ckey,
 hipError_t
 launch_probe_cust_map_compact(hipStream_t s, const int64_t *ckey,
 			      const int64_t *okey, int64_t n,
-			      DeviceHashTable cust, int64_t *out_okey,
+			      DeviceHashTable cust, const uint8_t *cust_dense,
+			      int64_t cust_dlen, int64_t *out_okey,
 			      int64_t *out_nat, unsigned long long *out_count)
 {
 	hipLaunchKernelGGL(k_probe_cust_map_compact, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, ckey, okey, n, cust.keys,
-			   cust.payload, cust.nslots, out_okey, out_nat,
-			   out_count);
+			   cust.payload, cust.nslots, cust_dense, cust_dlen,
+			   out_okey, out_nat, out_count);
 	return hipGetLastError();
 }
 
@@ -2006,6 +2058,68 @@ launch_supp_dense_fill_pairs(hipStream_t s, const int64_t *sk,
 	return hipGetLastError();
 }
 
+
+/* dense customer side (guarded like the supplier one): when c_custkey
+ * is (near-)dense, the customer build side collapses to a u8 array —
+ * Q3: 1 = in the mktsegment, 0 = not; Q5: nationkey, 255 = absent.
+ * One L2/L3-resident byte load replaces bloom + 512 MB-table probes. */
+__global__ void
+k_cust_dense_fill_seg(const int64_t *__restrict__ custkey,
+		      const uint8_t *__restrict__ mktseg, int64_t n,
+		      uint8_t segcode, uint8_t *__restrict__ dense,
+		      int64_t dense_len)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k = custkey[i];
+
+		if (k >= 0 && k < dense_len && mktseg[i] == segcode)
+			dense[k] = 1;
+	}
+}
+
+hipError_t
+launch_cust_dense_fill_seg(hipStream_t s, const int64_t *custkey,
+			   const uint8_t *mktseg, int64_t n, uint8_t segcode,
+			   uint8_t *dense, int64_t dense_len)
+{
+	hipLaunchKernelGGL(k_cust_dense_fill_seg, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, custkey, mktseg, n, segcode,
+			   dense, dense_len);
+	return hipGetLastError();
+}
+
+__global__ void
+k_cust_dense_fill_nat(const int64_t *__restrict__ custkey,
+		      const uint8_t *__restrict__ nation, int64_t n,
+		      uint8_t *__restrict__ dense, int64_t dense_len)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t k = custkey[i];
+
+		if (k >= 0 && k < dense_len)
+			dense[k] = nation[i];
+	}
+}
+
+hipError_t
+launch_cust_dense_fill_nat(hipStream_t s, const int64_t *custkey,
+			   const uint8_t *nation, int64_t n, uint8_t *dense,
+			   int64_t dense_len)
+{
+	hipLaunchKernelGGL(k_cust_dense_fill_nat, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, custkey, nation, n, dense,
+			   dense_len);
+	return hipGetLastError();
+}
+
 /* count-helper launchers used by engine_abi.cpp */
 hipError_t
 launch_count_filter_u8(hipStream_t s, const uint8_t *col, uint8_t val,
@@ -2019,12 +2133,13 @@ launch_count_filter_u8(hipStream_t s, const uint8_t *col, uint8_t val,
 hipError_t
 launch_count_orders_match(hipStream_t s, const int64_t *ckey,
 			  const int32_t *odate, int64_t n, int32_t cutoff,
-			  DeviceHashTable cust, unsigned long long *out)
+			  DeviceHashTable cust, const uint8_t *cust_dense,
+			  int64_t cust_dlen, unsigned long long *out)
 {
 	hipLaunchKernelGGL(k_count_orders_match, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, ckey, odate, n, cutoff,
 			   cust.keys, cust.nslots, cust.bloom,
-			   cust.bloom_words, out);
+			   cust.bloom_words, cust_dense, cust_dlen, out);
 	return hipGetLastError();
 }
 
