@@ -749,6 +749,7 @@ k_count_orders_match(const int64_t *__restrict__ ckey,
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
+	{
 		if (odate[i] >= cutoff)
 			continue;
 		if (cust_dense)
@@ -760,6 +761,7 @@ k_count_orders_match(const int64_t *__restrict__ ckey,
 		else
 			c += ht_contains_b(cust_keys, cust_slots, cust_bloom,
 					   cust_bwords, ckey[i]);
+	}
 	for (int off = 32; off; off >>= 1)
 		c += __shfl_down(c, off, 64);
 	if ((threadIdx.x & 63) == 0 && c)
