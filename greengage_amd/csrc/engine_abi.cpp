@@ -778,37 +778,69 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	if (!ctr)
 		return fail(GG_ENOMEM, "scratch");
 
-	/* 1. customer build side (nodeHash.c:450 sizing / :905 insert) */
+	/* 1. customer build side (nodeHash.c:450 sizing / :905 insert).
+	 * Dense custkeys (max <= 8x rows) collapse to a u8 membership
+	 * array in L2/L3; hash set fallback for sparse keys. */
 	DeviceHashTable cust{};
+	uint8_t *cust_dense = nullptr;
+	int64_t cust_dlen = 0;
+	{
+		unsigned long long maxk = 0;
+
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_max_i64(e.stream, c_ck, cu->nrows, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &maxk));
+		if (cu->nrows > 0 && maxk > 0 &&
+		    (int64_t) maxk <= 8 * cu->nrows + 16)
+			cust_dlen = (int64_t) maxk + 1;
+	}
 	{
 		Timed tm(e.stream);
 
-		if (!p->cust_slots)
+		if (cust_dlen)
 		{
-			unsigned long long nfil = 0;
-
-			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-			GG_HIP(launch_count_filter_u8(e.stream, c_ms, segcode,
-						      cu->nrows, ctr));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			GG_TRY(read_counter(ctr, &nfil));
-			p->cust_slots = next_pow2(2 * (nfil + 1));
+			cust_dense = (uint8_t *)
+				p->sget("cust.dense", (size_t) cust_dlen);
+			if (!cust_dense)
+				return fail(GG_ENOMEM, "cust dense");
+			GG_HIP(hipMemsetAsync(cust_dense, 0,
+					      (size_t) cust_dlen, e.stream));
+			GG_HIP(launch_cust_dense_fill_seg(e.stream, c_ck,
+							  c_ms, cu->nrows,
+							  segcode, cust_dense,
+							  cust_dlen));
 		}
-		cust.nslots = p->cust_slots;
-		cust.bloom_words = cust.nslots / 8 < 1024
-			? 1024 : cust.nslots / 8;
-		cust.keys = (unsigned long long *)
-			p->sget("cust.keys", cust.nslots * 8);
-		cust.bloom = (unsigned long long *)
-			p->sget("cust.bloom", cust.bloom_words * 8);
-		if (!cust.keys || !cust.bloom)
-			return fail(GG_ENOMEM, "cust table");
-		GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8,
-				      e.stream));
-		GG_HIP(hipMemsetAsync(cust.bloom, 0, cust.bloom_words * 8,
-				      e.stream));
-		GG_HIP(launch_build_set(e.stream, c_ck, c_ms, segcode,
-					cu->nrows, cust));
+		else
+		{
+			if (!p->cust_slots)
+			{
+				unsigned long long nfil = 0;
+
+				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+				GG_HIP(launch_count_filter_u8(e.stream, c_ms,
+							      segcode,
+							      cu->nrows, ctr));
+				GG_HIP(hipStreamSynchronize(e.stream));
+				GG_TRY(read_counter(ctr, &nfil));
+				p->cust_slots = next_pow2(2 * (nfil + 1));
+			}
+			cust.nslots = p->cust_slots;
+			cust.bloom_words = cust.nslots / 8 < 1024
+				? 1024 : cust.nslots / 8;
+			cust.keys = (unsigned long long *)
+				p->sget("cust.keys", cust.nslots * 8);
+			cust.bloom = (unsigned long long *)
+				p->sget("cust.bloom", cust.bloom_words * 8);
+			if (!cust.keys || !cust.bloom)
+				return fail(GG_ENOMEM, "cust table");
+			GG_HIP(hipMemsetAsync(cust.keys, 0, cust.nslots * 8,
+					      e.stream));
+			GG_HIP(hipMemsetAsync(cust.bloom, 0,
+					      cust.bloom_words * 8, e.stream));
+			GG_HIP(launch_build_set(e.stream, c_ck, c_ms, segcode,
+						cu->nrows, cust));
+		}
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("build_customer");
 
