@@ -850,48 +850,100 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.hbm_bytes += cu->nrows * 9;
 	}
 
-	/* 2. orders side → orders hash table */
+	/* 2. orders side.  Dense o_orderkey (the common case: TPC-H PK)
+	 * collapses the orders hash table to direct-map pay/rev arrays
+	 * (q3_dense.hip); hash table fallback otherwise. */
 	DeviceHashTable ord{};
+	unsigned long long *ordd_pay = nullptr;
+	unsigned long long *ordd_rev = nullptr;
+	unsigned long long *ordd_bloom = nullptr;
+	uint64_t ordd_bwords = 0;
+	int64_t ord_dlen = 0;
 	unsigned long long nmatch = 0;
+
+	{
+		unsigned long long maxk = 0;
+
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_max_i64(e.stream, o_ok, od->nrows, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &maxk));
+		if (od->nrows > 0 && maxk > 0 &&
+		    (int64_t) maxk <= 8 * od->nrows + 16)
+			ord_dlen = (int64_t) maxk + 1;
+	}
+	if (ord_dlen)
+	{
+		ordd_bwords = next_pow2((uint64_t) (od->nrows / 32 + 1024));
+		ordd_pay = (unsigned long long *)
+			p->sget("ordd.pay", (size_t) ord_dlen * 8);
+		ordd_rev = (unsigned long long *)
+			p->sget("ordd.rev", (size_t) ord_dlen * 8);
+		ordd_bloom = (unsigned long long *)
+			p->sget("ordd.bloom", ordd_bwords * 8);
+		if (!ordd_pay || !ordd_rev || !ordd_bloom)
+			return fail(GG_ENOMEM, "ord dense");
+		GG_HIP(hipMemsetAsync(ordd_pay, 0xff, (size_t) ord_dlen * 8,
+				      e.stream));
+		GG_HIP(hipMemsetAsync(ordd_rev, 0, (size_t) ord_dlen * 8,
+				      e.stream));
+		GG_HIP(hipMemsetAsync(ordd_bloom, 0, ordd_bwords * 8,
+				      e.stream));
+	}
 
 	if (!exch)
 	{
 		Timed tm(e.stream);
 
-		if (!p->ord_slots)
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		if (ord_dlen)
 		{
-			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-			GG_HIP(launch_count_orders_match(e.stream, o_ck, o_dt,
-							 od->nrows, cutoff,
-							 cust, cust_dense,
-							 cust_dlen, ctr));
+			GG_HIP(launch_dn_build_orders(
+				e.stream, o_ok, o_ck, o_dt, o_pr, od->nrows,
+				cutoff, cust, cust_dense, cust_dlen, ordd_pay,
+				ord_dlen, ordd_bloom, ordd_bwords, ctr));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
-			p->ord_slots = next_pow2(2 * (nmatch + 1));
 		}
-		ord.nslots = p->ord_slots;
-		ord.bloom_words = ord.nslots / 8 < 1024
-			? 1024 : ord.nslots / 8;
-		ord.keys = (unsigned long long *)
-			p->sget("ord.keys", ord.nslots * 8);
-		ord.payload = (unsigned long long *)
-			p->sget("ord.payload", ord.nslots * 8);
-		ord.rev = (unsigned long long *)
-			p->sget("ord.rev", ord.nslots * 8);
-		ord.bloom = (unsigned long long *)
-			p->sget("ord.bloom", ord.bloom_words * 8);
-		if (!ord.keys || !ord.payload || !ord.rev || !ord.bloom)
-			return fail(GG_ENOMEM, "ord table");
-		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
-		GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8, e.stream));
-		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
-				      e.stream));
-		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt, o_pr,
-					   od->nrows, cutoff, cust, cust_dense,
-					   cust_dlen, ord, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_TRY(read_counter(ctr, &nmatch));
+		else
+		{
+			if (!p->ord_slots)
+			{
+				GG_HIP(launch_count_orders_match(
+					e.stream, o_ck, o_dt, od->nrows,
+					cutoff, cust, cust_dense, cust_dlen,
+					ctr));
+				GG_HIP(hipStreamSynchronize(e.stream));
+				GG_TRY(read_counter(ctr, &nmatch));
+				p->ord_slots = next_pow2(2 * (nmatch + 1));
+				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+			}
+			ord.nslots = p->ord_slots;
+			ord.bloom_words = ord.nslots / 8 < 1024
+				? 1024 : ord.nslots / 8;
+			ord.keys = (unsigned long long *)
+				p->sget("ord.keys", ord.nslots * 8);
+			ord.payload = (unsigned long long *)
+				p->sget("ord.payload", ord.nslots * 8);
+			ord.rev = (unsigned long long *)
+				p->sget("ord.rev", ord.nslots * 8);
+			ord.bloom = (unsigned long long *)
+				p->sget("ord.bloom", ord.bloom_words * 8);
+			if (!ord.keys || !ord.payload || !ord.rev || !ord.bloom)
+				return fail(GG_ENOMEM, "ord table");
+			GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8,
+					      e.stream));
+			GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8,
+					      e.stream));
+			GG_HIP(hipMemsetAsync(ord.bloom, 0,
+					      ord.bloom_words * 8, e.stream));
+			GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt,
+						   o_pr, od->nrows, cutoff,
+						   cust, cust_dense, cust_dlen,
+						   ord, ctr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr, &nmatch));
+		}
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("build_orders");
 
@@ -975,8 +1027,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		}
 		std::vector<unsigned long long> rcnts(nseg), roffs(nseg + 1, 0);
 
-		for (int s = 0; s < nseg; s++)
-			rcnts[s] = allcnt[(size_t) s * nseg + me];
+		for (int s2 = 0; s2 < nseg; s2++)
+			rcnts[s2] = allcnt[(size_t) s2 * nseg + me];
 		for (int i = 0; i < nseg; i++)
 			roffs[i + 1] = roffs[i] + rcnts[i];
 		uint64_t rtotal = roffs[nseg];
@@ -1047,8 +1099,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		}
 		std::vector<unsigned long long> rcnts2(nseg), roffs2(nseg + 1, 0);
 
-		for (int s = 0; s < nseg; s++)
-			rcnts2[s] = allcnt2[(size_t) s * nseg + me];
+		for (int s2 = 0; s2 < nseg; s2++)
+			rcnts2[s2] = allcnt2[(size_t) s2 * nseg + me];
 		for (int i = 0; i < nseg; i++)
 			roffs2[i + 1] = roffs2[i] + rcnts2[i];
 		uint64_t rtotal2 = roffs2[nseg];
@@ -1065,28 +1117,42 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 					  rcnts2.data()));
 
 		nmatch = rtotal2;
-		if (!p->ord_slots)
-			p->ord_slots = next_pow2(2 * (rtotal2 + 1));
-		ord.nslots = p->ord_slots;
-		ord.bloom_words = ord.nslots / 8 < 1024
-			? 1024 : ord.nslots / 8;
-		ord.keys = (unsigned long long *)
-			p->sget("ord.keys", ord.nslots * 8);
-		ord.payload = (unsigned long long *)
-			p->sget("ord.payload", ord.nslots * 8);
-		ord.rev = (unsigned long long *)
-			p->sget("ord.rev", ord.nslots * 8);
-		ord.bloom = (unsigned long long *)
-			p->sget("ord.bloom", ord.bloom_words * 8);
-		if (!ord.keys || !ord.payload || !ord.rev || !ord.bloom)
-			return fail(GG_ENOMEM, "ord table");
-		GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8, e.stream));
-		GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8, e.stream));
-		GG_HIP(hipMemsetAsync(ord.bloom, 0, ord.bloom_words * 8,
-				      e.stream));
-		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_pay,
-					    (int64_t) rtotal2, ord));
-		GG_HIP(hipStreamSynchronize(e.stream));
+		if (ord_dlen)
+		{
+			GG_HIP(launch_dn_insert_orders(e.stream, r2_ok, r2_pay,
+						       (int64_t) rtotal2,
+						       ordd_pay, ord_dlen,
+						       ordd_bloom,
+						       ordd_bwords));
+			GG_HIP(hipStreamSynchronize(e.stream));
+		}
+		else
+		{
+			if (!p->ord_slots)
+				p->ord_slots = next_pow2(2 * (rtotal2 + 1));
+			ord.nslots = p->ord_slots;
+			ord.bloom_words = ord.nslots / 8 < 1024
+				? 1024 : ord.nslots / 8;
+			ord.keys = (unsigned long long *)
+				p->sget("ord.keys", ord.nslots * 8);
+			ord.payload = (unsigned long long *)
+				p->sget("ord.payload", ord.nslots * 8);
+			ord.rev = (unsigned long long *)
+				p->sget("ord.rev", ord.nslots * 8);
+			ord.bloom = (unsigned long long *)
+				p->sget("ord.bloom", ord.bloom_words * 8);
+			if (!ord.keys || !ord.payload || !ord.rev || !ord.bloom)
+				return fail(GG_ENOMEM, "ord table");
+			GG_HIP(hipMemsetAsync(ord.keys, 0, ord.nslots * 8,
+					      e.stream));
+			GG_HIP(hipMemsetAsync(ord.rev, 0, ord.nslots * 8,
+					      e.stream));
+			GG_HIP(hipMemsetAsync(ord.bloom, 0,
+					      ord.bloom_words * 8, e.stream));
+			GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_pay,
+						    (int64_t) rtotal2, ord));
+			GG_HIP(hipStreamSynchronize(e.stream));
+		}
 
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("orders_exchange");
@@ -1105,8 +1171,15 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	{
 		Timed tm(e.stream);
 
-		GG_HIP(launch_probe_lineitem(e.stream, l_ok, l_sd, l_pc, l_dc,
-					     li->nrows, cutoff, ord, ctr));
+		if (ord_dlen)
+			GG_HIP(launch_dn_probe_lineitem(
+				e.stream, l_ok, l_sd, l_pc, l_dc, li->nrows,
+				cutoff, ordd_pay, ordd_rev, ord_dlen,
+				ordd_bloom, ordd_bwords, ctr));
+		else
+			GG_HIP(launch_probe_lineitem(e.stream, l_ok, l_sd,
+						     l_pc, l_dc, li->nrows,
+						     cutoff, ord, ctr));
 		double ms = tm.stop();
 		GG_TRY(read_counter(ctr, &njoin));
 		KernelStatAcc &st = p->stat("probe_lineitem");
@@ -1129,7 +1202,11 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		Timed tm(e.stream);
 
 		GG_HIP(hipMemsetAsync(stats5, 0, 5 * 8, e.stream));
-		GG_HIP(launch_q3_stats(e.stream, ord, stats5));
+		if (ord_dlen)
+			GG_HIP(launch_dn_q3_stats(e.stream, ordd_pay,
+						  ordd_rev, ord_dlen, stats5));
+		else
+			GG_HIP(launch_q3_stats(e.stream, ord, stats5));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_HIP(hipMemcpy(hstats, stats5, 40, hipMemcpyDeviceToHost));
 
@@ -1153,11 +1230,27 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				return fail(GG_ENOMEM, "topk scratch");
 			GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4, e.stream));
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-			GG_HIP(launch_q3_hist(e.stream, ord, stats5, dhist));
-			GG_HIP(launch_q3_threshold(e.stream, dhist, stats5, k,
-						   dthr));
-			GG_HIP(launch_q3_collect(e.stream, ord, dthr, dout,
-						 ctr, cap));
+			if (ord_dlen)
+			{
+				GG_HIP(launch_dn_q3_hist(e.stream, ordd_rev,
+							 ord_dlen, stats5,
+							 dhist));
+				GG_HIP(launch_q3_threshold(e.stream, dhist,
+							   stats5, k, dthr));
+				GG_HIP(launch_dn_q3_collect(e.stream, ordd_pay,
+							    ordd_rev, ord_dlen,
+							    dthr, dout, ctr,
+							    cap));
+			}
+			else
+			{
+				GG_HIP(launch_q3_hist(e.stream, ord, stats5,
+						      dhist));
+				GG_HIP(launch_q3_threshold(e.stream, dhist,
+							   stats5, k, dthr));
+				GG_HIP(launch_q3_collect(e.stream, ord, dthr,
+							 dout, ctr, cap));
+			}
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &ncand));
 			if (ncand > cap)
@@ -1174,7 +1267,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				cand.resize(k);
 		}
 
-				double ms = tm.stop();
+		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("q3_topk");
 
 		st.launches++;
@@ -1623,14 +1716,66 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		st.rows_in += su->nrows;
 	}
 
-	/* 3. orders → okey → c_nationkey map */
+	/* 3. orders → okey → c_nationkey map.  Dense o_orderkey (with a
+	 * dense supplier side) uses the direct-map arrays (q3_dense.hip);
+	 * hash table fallback otherwise. */
 	DeviceHashTable ord{};
+	unsigned long long *ordd_pay = nullptr;
+	unsigned long long *ordd_bloom = nullptr;
+	uint64_t ordd_bwords = 0;
+	int64_t ord_dlen = 0;
 	unsigned long long nmatch = 0;
+
+	if (supp_dense)
+	{
+		unsigned long long maxk = 0;
+
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_max_i64(e.stream, o_ok, od->nrows, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_TRY(read_counter(ctr, &maxk));
+		if (od->nrows > 0 && maxk > 0 &&
+		    (int64_t) maxk <= 8 * od->nrows + 16)
+			ord_dlen = (int64_t) maxk + 1;
+	}
+	if (ord_dlen)
+	{
+		ordd_bwords = next_pow2((uint64_t) (od->nrows / 32 + 1024));
+		ordd_pay = (unsigned long long *)
+			p->sget("ordd.pay", (size_t) ord_dlen * 8);
+		ordd_bloom = (unsigned long long *)
+			p->sget("ordd.bloom", ordd_bwords * 8);
+		if (!ordd_pay || !ordd_bloom)
+			return fail(GG_ENOMEM, "ord dense");
+		GG_HIP(hipMemsetAsync(ordd_pay, 0xff, (size_t) ord_dlen * 8,
+				      e.stream));
+		GG_HIP(hipMemsetAsync(ordd_bloom, 0, ordd_bwords * 8,
+				      e.stream));
+	}
 
 	if (!exch)
 	{
 		Timed tm(e.stream);
 
+		if (ord_dlen)
+		{
+			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+			GG_HIP(launch_dn_build_orders_q5(
+				e.stream, o_ok, o_ck, o_dt, od->nrows,
+				date_lo, date_hi, cust, cust_dense, cust_dlen,
+				ordd_pay, ord_dlen, ordd_bloom, ordd_bwords,
+				ctr));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr, &nmatch));
+			double ms0 = tm.stop();
+			KernelStatAcc &st0 = p->stat("build_orders");
+
+			st0.launches++;
+			st0.total_ms += ms0;
+			st0.rows_in += od->nrows;
+			st0.rows_out += (int64_t) nmatch;
+			goto orders_done;
+		}
 		if (!p->ord_slots)
 		{
 			unsigned long long nfil = 0;
@@ -1663,13 +1808,16 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 					      ord, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nmatch));
-		double ms = tm.stop();
-		KernelStatAcc &st = p->stat("build_orders");
+		{
+			double ms = tm.stop();
+			KernelStatAcc &st = p->stat("build_orders");
 
-		st.launches++;
-		st.total_ms += ms;
-		st.rows_in += od->nrows;
-		st.rows_out += (int64_t) nmatch;
+			st.launches++;
+			st.total_ms += ms;
+			st.rows_in += od->nrows;
+			st.rows_out += (int64_t) nmatch;
+		}
+orders_done:;
 	}
 	else
 	{
@@ -1822,6 +1970,17 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 					  rcnts2.data()));
 
 		nmatch = rtotal2;
+		if (ord_dlen)
+		{
+			GG_HIP(launch_dn_insert_orders(e.stream, r2_ok, r2_nat,
+						       (int64_t) rtotal2,
+						       ordd_pay, ord_dlen,
+						       ordd_bloom,
+						       ordd_bwords));
+			GG_HIP(hipStreamSynchronize(e.stream));
+		}
+		else
+		{
 		if (!p->ord_slots)
 			p->ord_slots = next_pow2(2 * (rtotal2 + 1));
 		ord.nslots = p->ord_slots;
@@ -1840,6 +1999,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(launch_insert_orders(e.stream, r2_ok, r2_nat,
 					    (int64_t) rtotal2, ord));
 		GG_HIP(hipStreamSynchronize(e.stream));
+		}
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("orders_exchange");
 
@@ -1861,10 +2021,16 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 	{
 		Timed tm(e.stream);
 
-		GG_HIP(launch_probe_lineitem_q5(e.stream, l_ok, l_sk, l_pc,
-						l_dc, li->nrows, ord, supp,
-						supp_dense, supp_dense_len,
-						acc, ctr));
+		if (ord_dlen)
+			GG_HIP(launch_dn_probe_lineitem_q5(
+				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,
+				ordd_pay, ord_dlen, ordd_bloom, ordd_bwords,
+				supp_dense, supp_dense_len, acc, ctr));
+		else
+			GG_HIP(launch_probe_lineitem_q5(
+				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,
+				ord, supp, supp_dense, supp_dense_len, acc,
+				ctr));
 		double ms = tm.stop();
 		GG_TRY(read_counter(ctr, &njoin));
 		KernelStatAcc &st = p->stat("probe_lineitem_q5");

@@ -357,6 +357,71 @@ hipError_t launch_groupby_compact(hipStream_t s,
 				  uint64_t nslots, int64_t *out_keys,
 				  int64_t *out_sums, int64_t *out_cnts,
 				  unsigned long long *out_n, uint64_t cap);
+/* dense-orderkey fast path (q3_dense.hip) */
+hipError_t launch_dn_build_orders(hipStream_t s, const int64_t *okey,
+				  const int64_t *ckey, const int32_t *odate,
+				  const int32_t *prio, int64_t n,
+				  int32_t cutoff, DeviceHashTable cust,
+				  const uint8_t *cust_dense,
+				  int64_t cust_dlen, unsigned long long *pay,
+				  int64_t dense_len, unsigned long long *bloom,
+				  uint64_t bwords,
+				  unsigned long long *match_count);
+hipError_t launch_dn_build_orders_q5(hipStream_t s, const int64_t *okey,
+				     const int64_t *ckey,
+				     const int32_t *odate, int64_t n,
+				     int32_t date_lo, int32_t date_hi,
+				     DeviceHashTable cust,
+				     const uint8_t *cust_dense,
+				     int64_t cust_dlen,
+				     unsigned long long *pay,
+				     int64_t dense_len,
+				     unsigned long long *bloom,
+				     uint64_t bwords,
+				     unsigned long long *match_count);
+hipError_t launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
+				   const int64_t *rowpay, int64_t n,
+				   unsigned long long *pay,
+				   int64_t dense_len,
+				   unsigned long long *bloom,
+				   uint64_t bwords);
+hipError_t launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
+				    const int32_t *shipdate,
+				    const int64_t *price,
+				    const int64_t *disc, int64_t n,
+				    int32_t cutoff, unsigned long long *pay,
+				    unsigned long long *rev,
+				    int64_t dense_len,
+				    unsigned long long *bloom,
+				    uint64_t bwords,
+				    unsigned long long *join_rows);
+hipError_t launch_dn_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
+				       const int64_t *skey,
+				       const int64_t *price,
+				       const int64_t *disc, int64_t n,
+				       unsigned long long *pay,
+				       int64_t dense_len,
+				       unsigned long long *bloom,
+				       uint64_t bwords,
+				       const uint8_t *supp_dense,
+				       int64_t supp_dlen,
+				       unsigned long long *acc,
+				       unsigned long long *join_rows);
+hipError_t launch_dn_q3_stats(hipStream_t s, const unsigned long long *pay,
+			      const unsigned long long *rev,
+			      int64_t dense_len, unsigned long long *out5);
+hipError_t launch_dn_q3_hist(hipStream_t s, const unsigned long long *rev,
+			     int64_t dense_len,
+			     const unsigned long long *stats5,
+			     unsigned int *hist64k);
+hipError_t launch_dn_q3_collect(hipStream_t s,
+				const unsigned long long *pay,
+				const unsigned long long *rev,
+				int64_t dense_len,
+				const unsigned long long *thr_ptr,
+				gg_q3_result_row *out,
+				unsigned long long *out_count, uint64_t cap);
+
 hipError_t launch_dsb_decode(hipStream_t s, const uint8_t *stream,
 			     const int64_t *offsets, const int32_t *sizes,
 			     const int32_t *rowcounts,
