@@ -112,6 +112,12 @@ def _load():
     lib.gg_engine_aocs_decode.argtypes = [
         ctypes.c_void_p, I64, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
         ctypes.c_int, ctypes.c_void_p, I64, ctypes.POINTER(I64)]
+    lib.gg_engine_aocs_decode_ao.restype = ctypes.c_int
+    lib.gg_engine_aocs_decode_ao.argtypes = [
+        ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64, ctypes.c_int,
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_int, ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64)]
     lib.gg_engine_avg_str.argtypes = [U64, I64, ctypes.c_int, I64,
                                       ctypes.c_char_p]
     return lib
@@ -312,6 +318,24 @@ class Engine:
             datumlen, vals.ctypes.data_as(ctypes.c_void_p), out_width,
             nulls.ctypes.data_as(ctypes.c_void_p), nmax,
             ctypes.byref(n)), "aocs_decode")
+        return vals[:n.value], nulls[:n.value]
+
+    @staticmethod
+    def aocs_decode_ao(stream, checksums, ao_version, dsb_version,
+                       datumlen, nmax, out_width=8):
+        """Decode REAL AO storage blocks: restated header parse +
+        CRC32C verify on the host, datum-stream content on the GPU."""
+        import numpy as np
+        stream = np.ascontiguousarray(stream, dtype=np.uint8)
+        vals = np.empty(nmax, np.int32 if out_width == 4 else np.int64)
+        nulls = np.empty(nmax, np.uint8)
+        n = I64()
+        _check(lib().gg_engine_aocs_decode_ao(
+            stream.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            len(stream), checksums, ao_version, dsb_version, datumlen,
+            vals.ctypes.data_as(ctypes.c_void_p), out_width,
+            nulls.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)), nmax,
+            ctypes.byref(n)), "aocs_decode_ao")
         return vals[:n.value], nulls[:n.value]
 
     # ---- general hash group-by (arbitrary int64 keys, SUM+COUNT) ----

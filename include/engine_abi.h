@@ -214,6 +214,20 @@ gg_status gg_engine_aocs_decode(const uint8_t *stream, int64_t stream_len,
 				int out_width, uint8_t *out_nulls,
 				int64_t cap, int64_t *out_nrows);
 
+/* Decode REAL Append-Only storage blocks (headers + CRC32C verify +
+ * datum-stream content on the GPU).  Replaces the AO read path's
+ * header/checksum layer (cdbappendonlystorageformat.c:125,1661 and
+ * GetSmallContentHeaderInfo/GetNonBulkDenseContentHeaderInfo) for
+ * uncompressed int32/int64 columns; ao_version is the
+ * AORelationVersion (>=2), dsb_version the datum-stream block version
+ * (0=Orig,1=Dense,2=Dense_Enhanced). */
+gg_status gg_engine_aocs_decode_ao(const uint8_t *stream,
+				   int64_t stream_len, int checksums,
+				   int ao_version, int dsb_version,
+				   int datumlen, void *out_vals,
+				   int out_width, uint8_t *out_nulls,
+				   int64_t cap, int64_t *out_nrows);
+
 /* General hash group-by (execHHashagg.c find-or-create semantics on
  * arbitrary int64 keys, SUM+COUNT transitions): host buffers in,
  * groups out sorted by key ascending.  Keys may be any int64 except

@@ -239,6 +239,13 @@ def dsb_ref():
         D.ref_dsb_decode.argtypes = [
             _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
             _P_I64, _P_U8, I64, ctypes.POINTER(I64)]
+        D.ref_ao_wrap_stream.restype = ctypes.c_int
+        D.ref_ao_wrap_stream.argtypes = [
+            _P_U8, I64, ctypes.c_int, ctypes.c_int, _P_U8, I64,
+            ctypes.POINTER(I64)]
+        D.ref_ao_probe_block.restype = ctypes.c_int
+        D.ref_ao_probe_block.argtypes = [
+            _P_U8, ctypes.c_int] + [ctypes.POINTER(ctypes.c_int32)] * 6
     return _dsb
 
 
@@ -271,6 +278,35 @@ def dsb_decode(stream, datumlen, version, rle, nmax):
                           dv, dn, nmax, ctypes.byref(n))
     assert rc == 0, rc
     return dv[:n.value], dn[:n.value]
+
+
+def ao_wrap(framed, checksums=1, firstrownum=1):
+    """Wrap a framed datum-stream into REAL AO storage blocks with the
+    REFERENCE's header/checksum writer (cdbappendonlystorageformat.c
+    compiled in place)."""
+    D = dsb_ref()
+    assert D is not None, "reference AO codec missing"
+    framed = np.ascontiguousarray(framed, np.uint8)
+    cap = len(framed) * 2 + (1 << 16)
+    out = np.zeros(cap, np.uint8)
+    olen = I64()
+    rc = D.ref_ao_wrap_stream(framed, len(framed), checksums, firstrownum,
+                              out, cap, ctypes.byref(olen))
+    assert rc == 0, rc
+    return out[:olen.value].copy()
+
+
+def ao_probe(block, checksums=1):
+    """Parse ONE AO block header with the REFERENCE parser; returns
+    (kind, rowcount, datalen, content_off, overall, cksum_ok)."""
+    D = dsb_ref()
+    assert D is not None, "reference AO codec missing"
+    block = np.ascontiguousarray(block, np.uint8)
+    outs = [ctypes.c_int32() for _ in range(6)]
+    rc = D.ref_ao_probe_block(block, checksums,
+                              *[ctypes.byref(o) for o in outs])
+    assert rc == 0, rc
+    return tuple(o.value for o in outs)
 
 
 def pgdate(y, m, d):

@@ -217,3 +217,139 @@ ref_dsb_decode(const uint8 *stream, int64 stream_len, int datumlen,
 	*out_n = nout;
 	return 0;
 }
+
+/* ------------------------------------------------------------------ */
+/* Append-Only storage block layer (cdbappendonlystorageformat.c,      */
+/* compiled in place): wrap datum-stream content into REAL AO blocks   */
+/* with block/header CRC32C checksums.                                 */
+/* ------------------------------------------------------------------ */
+bool		Debug_appendonly_print_storage_headers = false;
+
+#include "cdb/cdbappendonlystorage_int.h"
+#include "cdb/cdbappendonlystorage.h"
+#include "cdb/cdbappendonlystorageformat.h"
+
+#define REF_AO_VERSION 2	/* AORelationVersion_Aligned64bit */
+#define REF_AO_EXECKIND 1	/* AOCSBK_BLOCK */
+
+static int32 ao_roundup8(int32 l) { return ((l + 7) / 8) * 8; }
+
+/*
+ * Convert a [int32 size][int32 rowcount][content] framed stream (from
+ * ref_dsb_encode) into a stream of real AO storage blocks:
+ *   [8B header][block crc32c][header crc32c][firstRowNum(8B, optional)]
+ *   [content, zero-padded to 8B]
+ * Small-content headers when rowcount fits 14 bits, NonBulkDense
+ * otherwise — the same choice datumstream.c:944 makes.
+ */
+int
+ref_ao_wrap_stream(const uint8 *framed, int64 framed_len, int checksums,
+		   int has_firstrownum, uint8 *out, int64 out_cap,
+		   int64 *out_len)
+{
+	int64		pos = 0,
+				opos = 0;
+	int64		firstRowNum = 1;
+
+	while (pos < framed_len)
+	{
+		int32		sz,
+					rowcount;
+		int32		hdrlen,
+					padded,
+					overall;
+		uint8	   *hdr;
+
+		memcpy(&sz, framed + pos, 4);
+		memcpy(&rowcount, framed + pos + 4, 4);
+		pos += 8;
+		hdrlen = (checksums ? 16 : 8) + (has_firstrownum ? 8 : 0);
+		padded = ao_roundup8(sz);
+		overall = hdrlen + padded;
+		if (opos + overall > out_cap)
+			return -1;
+		hdr = out + opos;
+		memset(hdr, 0, overall);
+		memcpy(hdr + hdrlen, framed + pos, sz);
+		if (rowcount <= 0x3FFF)
+			AppendOnlyStorageFormat_MakeSmallContentHeader(
+				hdr, checksums != 0, has_firstrownum != 0,
+				REF_AO_VERSION, firstRowNum, REF_AO_EXECKIND,
+				rowcount, sz, /* compressedLength */ 0);
+		else
+			AppendOnlyStorageFormat_MakeNonBulkDenseContentHeader(
+				hdr, checksums != 0, has_firstrownum != 0,
+				REF_AO_VERSION, firstRowNum, REF_AO_EXECKIND,
+				rowcount, sz);
+		firstRowNum += rowcount;
+		pos += sz;
+		opos += overall;
+	}
+	*out_len = opos;
+	return 0;
+}
+
+/* reference-side parse of one AO block (for cross-checking the
+ * engine's restated parser in tests) */
+int
+ref_ao_probe_block(uint8 *block, int checksums, int32 *kind,
+		   int32 *rowcount, int32 *datalen, int32 *content_off,
+		   int32 *overall, int32 *cksum_ok)
+{
+	AoHeaderKind hk;
+	int32		actualHeaderLen;
+	AOHeaderCheckError err;
+
+	err = AppendOnlyStorageFormat_GetHeaderInfo(block, checksums != 0,
+						    &hk, &actualHeaderLen);
+	if (err != AOHeaderCheckOk)
+		return -1;
+	*kind = (int32) hk;
+	*cksum_ok = 1;
+	if (checksums)
+	{
+		pg_crc32	stored,
+					computed;
+
+		if (!AppendOnlyStorageFormat_VerifyHeaderChecksum(
+			block, &stored, &computed))
+			*cksum_ok = 0;
+	}
+	{
+		int32		offset = 0,
+					uncompressedLen = 0,
+					compressedLen = 0;
+		int			execKind = 0,
+					rc = 0;
+		bool		hasFRN = false,
+					isCompressed = false;
+		int64		frn = 0;
+
+		if (hk == AoHeaderKind_SmallContent)
+			err = AppendOnlyStorageFormat_GetSmallContentHeaderInfo(
+				block, actualHeaderLen, checksums != 0,
+				1 << 30, overall, &offset, &uncompressedLen,
+				&execKind, &hasFRN, REF_AO_VERSION, &frn, &rc,
+				&isCompressed, &compressedLen);
+		else
+			err = AppendOnlyStorageFormat_GetNonBulkDenseContentHeaderInfo(
+				block, actualHeaderLen, checksums != 0,
+				1 << 30, overall, &offset, &uncompressedLen,
+				&execKind, &hasFRN, REF_AO_VERSION, &frn, &rc);
+		if (err != AOHeaderCheckOk)
+			return -2;
+		*rowcount = rc;
+		*datalen = uncompressedLen;
+		*content_off = offset;
+	}
+	if (checksums && *cksum_ok)
+	{
+		pg_crc32	stored,
+					computed;
+
+		if (!AppendOnlyStorageFormat_VerifyBlockChecksum(
+			block, *overall, &stored, &computed))
+			*cksum_ok = 0;
+	}
+	return 0;
+}

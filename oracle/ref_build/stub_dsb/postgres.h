@@ -14,6 +14,8 @@
 #include <stddef.h>
 #include <string.h>
 #include <stdio.h>
+#include <stdarg.h>
+#include <stdlib.h>
 #include <stdlib.h>
 #include <stdbool.h>
 
@@ -38,6 +40,7 @@ typedef char *Pointer;
 #define MemSet(start, val, len) memset(start, val, len)
 #define lengthof(array) (sizeof(array) / sizeof((array)[0]))
 #define pg_attribute_always_inline inline
+#define PGDLLIMPORT
 #define INT64_FORMAT "%ld"
 
 /* ---- error reporting: print + abort on >= ERROR ---- */
@@ -197,5 +200,25 @@ typedef struct varatt_external
 extern void varattrib_untoast_ptr_len(Datum d, char **datastart, int *len,
 				      void **tofree);
 extern bool value_type_could_short(Pointer ptr, Oid typid);
+
+
+/* extras for cdbappendonlystorageformat.c */
+#define elogif(p, level, ...) do { if (p) elog(level, __VA_ARGS__); } while (0)
+#ifndef INT64CONST
+#define INT64CONST(x) ((int64) (x##LL))
+#endif
+/* pg_appendonly.h:113 — version > AORelationVersion_Original(=1) */
+#define IsAOBlockAndMemtupleAlignmentFixed(version) ((version) > 1)
+static inline char *
+psprintf(const char *fmt, ...)
+{
+	char	   *buf = (char *) malloc(1024);
+	va_list		ap;
+
+	va_start(ap, fmt);
+	vsnprintf(buf, 1024, fmt, ap);
+	va_end(ap);
+	return buf;
+}
 
 #endif
