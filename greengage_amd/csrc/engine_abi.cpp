@@ -2365,11 +2365,19 @@ ao_crc32c(const uint8_t *p, int64_t len)
  * via gg_engine_aocs_decode.  Compressed blocks (compressedLength!=0)
  * and ao_version<2 are out of scope this round.
  */
+/* zlib (pg_compression.c:253 binds compress2/uncompress) */
+#include <zlib.h>
+/* zstd public API — libzstd.so.1 ships without dev headers here */
+extern "C" size_t ZSTD_decompress(void *dst, size_t dstCap,
+				  const void *src, size_t srcSize);
+extern "C" unsigned ZSTD_isError(size_t code);
+
 extern "C" gg_status
 gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 			 int checksums, int ao_version, int dsb_version,
-			 int datumlen, void *out_vals, int out_width,
-			 uint8_t *out_nulls, int64_t cap, int64_t *out_nrows)
+			 int comptype, int datumlen, void *out_vals,
+			 int out_width, uint8_t *out_nulls, int64_t cap,
+			 int64_t *out_nrows)
 {
 	if (!stream || stream_len < 0)
 		return fail(GG_EINVAL, "bad ao stream");
@@ -2395,19 +2403,18 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 		std::memcpy(&w1, stream + pos + 4, 4);
 		int kind = (int) ((w0 >> 28) & 7);
 		int has_frn = (int) ((w0 >> 27) & 1);
-		int32_t rowcount, datalen;
+		int32_t rowcount, datalen, complen = 0;
 
 		if (kind == 1)
 		{		/* AoHeaderKind_SmallContent */
 			rowcount = (int32_t) ((w0 >> 10) & 0x3FFF);
 			datalen = (int32_t) (((w0 & 0x3FF) << 11) |
 					     ((w1 >> 21) & 0x7FF));
-			int32_t complen = (int32_t) (w1 & 0x1FFFFF);
-
-			if (complen != 0)
+			complen = (int32_t) (w1 & 0x1FFFFF);
+			if (complen != 0 && comptype == 0)
 				return fail(GG_EINVAL,
-					    "compressed AO block at %lld "
-					    "(out of scope)",
+					    "compressed AO block at %lld but "
+					    "comptype none",
 					    (long long) pos);
 		}
 		else if (kind == 3)
@@ -2420,9 +2427,13 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 				    "unsupported AoHeaderKind %d at %lld",
 				    kind, (long long) pos);
 
+		/* stored bytes = compressed when compressedLength != 0
+		 * (CompressAppend rule, cdbappendonlystoragewrite.c:1207) */
+		int32_t stored = complen ? complen : datalen;
+
 		hdr_end = pos + 8 + (checksums ? 8 : 0) +
 			(has_frn ? 8 : 0);
-		int64_t padded = ((int64_t) datalen + 7) & ~(int64_t) 7;
+		int64_t padded = ((int64_t) stored + 7) & ~(int64_t) 7;
 		int64_t overall = hdr_end + padded;
 
 		if (overall > stream_len || datalen < 0 || rowcount < 0)
@@ -2471,8 +2482,38 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 		framed.resize(fpos + 8 + (size_t) datalen);
 		std::memcpy(framed.data() + fpos, &datalen, 4);
 		std::memcpy(framed.data() + fpos + 4, &rowcount, 4);
-		std::memcpy(framed.data() + fpos + 8, stream + hdr_end,
-			    (size_t) datalen);
+		if (complen == 0)
+			std::memcpy(framed.data() + fpos + 8,
+				    stream + hdr_end, (size_t) datalen);
+		else if (comptype == 1)
+		{
+			uLongf dl = (uLongf) datalen;
+			int zrc = uncompress(framed.data() + fpos + 8, &dl,
+					     stream + hdr_end,
+					     (uLong) complen);
+
+			if (zrc != Z_OK || dl != (uLongf) datalen)
+				return fail(GG_EINVAL,
+					    "zlib decompress failed at %lld "
+					    "(rc %d, %lu of %d bytes)",
+					    (long long) pos, zrc,
+					    (unsigned long) dl, datalen);
+		}
+		else if (comptype == 2)
+		{
+			size_t dl = ZSTD_decompress(framed.data() + fpos + 8,
+						    (size_t) datalen,
+						    stream + hdr_end,
+						    (size_t) complen);
+
+			if (ZSTD_isError(dl) || dl != (size_t) datalen)
+				return fail(GG_EINVAL,
+					    "zstd decompress failed at %lld",
+					    (long long) pos);
+		}
+		else
+			return fail(GG_EINVAL, "unknown comptype %d",
+				    comptype);
 		pos = overall;
 	}
 	return gg_engine_aocs_decode(framed.data(),

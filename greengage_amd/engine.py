@@ -115,8 +115,9 @@ def _load():
     lib.gg_engine_aocs_decode_ao.restype = ctypes.c_int
     lib.gg_engine_aocs_decode_ao.argtypes = [
         ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64, ctypes.c_int,
-        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
-        ctypes.c_int, ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64,
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64,
         ctypes.POINTER(ctypes.c_int64)]
     lib.gg_engine_avg_str.argtypes = [U64, I64, ctypes.c_int, I64,
                                       ctypes.c_char_p]
@@ -322,7 +323,7 @@ class Engine:
 
     @staticmethod
     def aocs_decode_ao(stream, checksums, ao_version, dsb_version,
-                       datumlen, nmax, out_width=8):
+                       datumlen, nmax, out_width=8, comptype=0):
         """Decode REAL AO storage blocks: restated header parse +
         CRC32C verify on the host, datum-stream content on the GPU."""
         import numpy as np
@@ -332,7 +333,8 @@ class Engine:
         n = I64()
         _check(lib().gg_engine_aocs_decode_ao(
             stream.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
-            len(stream), checksums, ao_version, dsb_version, datumlen,
+            len(stream), checksums, ao_version, dsb_version, comptype,
+            datumlen,
             vals.ctypes.data_as(ctypes.c_void_p), out_width,
             nulls.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)), nmax,
             ctypes.byref(n)), "aocs_decode_ao")

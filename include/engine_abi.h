@@ -218,15 +218,19 @@ gg_status gg_engine_aocs_decode(const uint8_t *stream, int64_t stream_len,
  * datum-stream content on the GPU).  Replaces the AO read path's
  * header/checksum layer (cdbappendonlystorageformat.c:125,1661 and
  * GetSmallContentHeaderInfo/GetNonBulkDenseContentHeaderInfo) for
- * uncompressed int32/int64 columns; ao_version is the
- * AORelationVersion (>=2), dsb_version the datum-stream block version
- * (0=Orig,1=Dense,2=Dense_Enhanced). */
+ * int32/int64 columns; ao_version is the AORelationVersion (>=2),
+ * dsb_version the datum-stream block version
+ * (0=Orig,1=Dense,2=Dense_Enhanced), comptype the catalog's
+ * compresstype (0=none, 1=zlib, 2=zstd — pg_compression.c registry);
+ * compressed SmallContent blocks (compressedLength!=0) are
+ * decompressed host-side before GPU decode. */
 gg_status gg_engine_aocs_decode_ao(const uint8_t *stream,
 				   int64_t stream_len, int checksums,
 				   int ao_version, int dsb_version,
-				   int datumlen, void *out_vals,
-				   int out_width, uint8_t *out_nulls,
-				   int64_t cap, int64_t *out_nrows);
+				   int comptype, int datumlen,
+				   void *out_vals, int out_width,
+				   uint8_t *out_nulls, int64_t cap,
+				   int64_t *out_nrows);
 
 /* General hash group-by (execHHashagg.c find-or-create semantics on
  * arbitrary int64 keys, SUM+COUNT transitions): host buffers in,

@@ -243,6 +243,10 @@ def dsb_ref():
         D.ref_ao_wrap_stream.argtypes = [
             _P_U8, I64, ctypes.c_int, ctypes.c_int, _P_U8, I64,
             ctypes.POINTER(I64)]
+        D.ref_ao_wrap_stream_c.restype = ctypes.c_int
+        D.ref_ao_wrap_stream_c.argtypes = [
+            _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int, _P_U8, I64, ctypes.POINTER(I64)]
         D.ref_ao_probe_block.restype = ctypes.c_int
         D.ref_ao_probe_block.argtypes = [
             _P_U8, ctypes.c_int] + [ctypes.POINTER(ctypes.c_int32)] * 6
@@ -292,6 +296,24 @@ def ao_wrap(framed, checksums=1, firstrownum=1):
     olen = I64()
     rc = D.ref_ao_wrap_stream(framed, len(framed), checksums, firstrownum,
                               out, cap, ctypes.byref(olen))
+    assert rc == 0, rc
+    return out[:olen.value].copy()
+
+
+def ao_wrap_compressed(framed, comptype, complevel, checksums=1,
+                       firstrownum=1):
+    """Wrap with bulk compression (1=zlib, 2=zstd) via the REFERENCE's
+    header writer + the same codec libraries/parameters the reference
+    binds (pg_compression.c:253, zstd_compression.c:117)."""
+    D = dsb_ref()
+    assert D is not None, "reference AO codec missing"
+    framed = np.ascontiguousarray(framed, np.uint8)
+    cap = len(framed) * 2 + (1 << 16)
+    out = np.zeros(cap, np.uint8)
+    olen = I64()
+    rc = D.ref_ao_wrap_stream_c(framed, len(framed), checksums,
+                                firstrownum, comptype, complevel, out,
+                                cap, ctypes.byref(olen))
     assert rc == 0, rc
     return out[:olen.value].copy()
 

@@ -289,6 +289,105 @@ ref_ao_wrap_stream(const uint8 *framed, int64 framed_len, int checksums,
 	return 0;
 }
 
+/*
+ * Same, with bulk compression: the reference's compressed-block write
+ * rule (AppendOnlyStorageWrite_CompressAppend,
+ * cdbappendonlystoragewrite.c:1134–1230): content is stored compressed
+ * only when strictly smaller, dataLength = uncompressed size,
+ * compressedLength = stored size (0 when kept uncompressed), pad to
+ * RoundUp8 of what is stored.  Codecs are the ones the reference binds
+ * (pg_compression.c:253 — zlib compress2; gpcontrib zstd_compression.c
+ * :117 — ZSTD plain).  Compression rides the SmallContent header only
+ * (NonBulkDense has no compressedLength field).
+ */
+#include <zlib.h>
+/* libzstd.so.1 is present without dev headers; public stable API: */
+extern size_t ZSTD_compress(void *dst, size_t dstCap, const void *src,
+			    size_t srcSize, int level);
+extern unsigned ZSTD_isError(size_t code);
+
+int
+ref_ao_wrap_stream_c(const uint8 *framed, int64 framed_len, int checksums,
+		     int has_firstrownum, int comptype, int complevel,
+		     uint8 *out, int64 out_cap, int64 *out_len)
+{
+	int64		pos = 0,
+				opos = 0;
+	int64		firstRowNum = 1;
+	static uint8 cbuf[4 * 1024 * 1024];
+
+	while (pos < framed_len)
+	{
+		int32		sz,
+					rowcount;
+		int32		hdrlen,
+					stored,
+					complen,
+					padded,
+					overall;
+		uint8	   *hdr;
+		const uint8 *content;
+
+		memcpy(&sz, framed + pos, 4);
+		memcpy(&rowcount, framed + pos + 4, 4);
+		pos += 8;
+		if (rowcount > 0x3FFF)
+			return -2;	/* compression => SmallContent only */
+
+		complen = 0;
+		content = framed + pos;
+		if (comptype == 1)
+		{
+			uLongf		dl = sizeof(cbuf);
+
+			if (compress2(cbuf, &dl, framed + pos, sz,
+				      complevel) != Z_OK)
+				return -3;
+			complen = (int32) dl;
+		}
+		else if (comptype == 2)
+		{
+			size_t		dl = ZSTD_compress(cbuf, sizeof(cbuf),
+						   framed + pos, sz, complevel);
+
+			if (ZSTD_isError(dl))
+				return -3;
+			complen = (int32) dl;
+		}
+		else if (comptype != 0)
+			return -4;
+		if (complen == 0 || complen >= sz)
+		{
+			stored = sz;
+			complen = 0;
+			content = framed + pos;
+		}
+		else
+		{
+			stored = complen;
+			content = cbuf;
+		}
+
+		hdrlen = (checksums ? 16 : 8) + (has_firstrownum ? 8 : 0);
+		padded = ao_roundup8(stored);
+		overall = hdrlen + padded;
+		if (opos + overall > out_cap)
+			return -1;
+		hdr = out + opos;
+		memset(hdr, 0, overall);
+		memcpy(hdr + hdrlen, content, stored);
+		AppendOnlyStorageFormat_MakeSmallContentHeader(
+			hdr, checksums != 0, has_firstrownum != 0,
+			REF_AO_VERSION, firstRowNum, REF_AO_EXECKIND,
+			rowcount, sz, complen);
+		firstRowNum += rowcount;
+		pos += sz;
+		opos += overall;
+	}
+	*out_len = opos;
+	return 0;
+}
+
 /* reference-side parse of one AO block (for cross-checking the
  * engine's restated parser in tests) */
 int
