@@ -1720,9 +1720,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 	 * dense supplier side) uses the direct-map arrays (q3_dense.hip);
 	 * hash table fallback otherwise. */
 	DeviceHashTable ord{};
-	unsigned long long *ordd_pay = nullptr;
-	unsigned long long *ordd_bloom = nullptr;
-	uint64_t ordd_bwords = 0;
+	uint8_t *ordd_pay8 = nullptr;
 	int64_t ord_dlen = 0;
 	unsigned long long nmatch = 0;
 
@@ -1740,16 +1738,13 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 	}
 	if (ord_dlen)
 	{
-		ordd_bwords = next_pow2((uint64_t) (od->nrows / 32 + 1024));
-		ordd_pay = (unsigned long long *)
-			p->sget("ordd.pay", (size_t) ord_dlen * 8);
-		ordd_bloom = (unsigned long long *)
-			p->sget("ordd.bloom", ordd_bwords * 8);
-		if (!ordd_pay || !ordd_bloom)
+		/* nation fits a byte; 255 = no matching order.  The u8 map
+		 * is its own membership filter — no Bloom reads on probe. */
+		ordd_pay8 = (uint8_t *)
+			p->sget("ordd.pay8", (size_t) ord_dlen);
+		if (!ordd_pay8)
 			return fail(GG_ENOMEM, "ord dense");
-		GG_HIP(hipMemsetAsync(ordd_pay, 0xff, (size_t) ord_dlen * 8,
-				      e.stream));
-		GG_HIP(hipMemsetAsync(ordd_bloom, 0, ordd_bwords * 8,
+		GG_HIP(hipMemsetAsync(ordd_pay8, 0xff, (size_t) ord_dlen,
 				      e.stream));
 	}
 
@@ -1760,11 +1755,10 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		if (ord_dlen)
 		{
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-			GG_HIP(launch_dn_build_orders_q5(
+			GG_HIP(launch_dn_build_orders_q5_u8(
 				e.stream, o_ok, o_ck, o_dt, od->nrows,
 				date_lo, date_hi, cust, cust_dense, cust_dlen,
-				ordd_pay, ord_dlen, ordd_bloom, ordd_bwords,
-				ctr));
+				ordd_pay8, ord_dlen, ctr));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
 			double ms0 = tm.stop();
@@ -1972,11 +1966,9 @@ orders_done:;
 		nmatch = rtotal2;
 		if (ord_dlen)
 		{
-			GG_HIP(launch_dn_insert_orders(e.stream, r2_ok, r2_nat,
-						       (int64_t) rtotal2,
-						       ordd_pay, ord_dlen,
-						       ordd_bloom,
-						       ordd_bwords));
+			GG_HIP(launch_dn_insert_orders_q5_u8(
+				e.stream, r2_ok, r2_nat, (int64_t) rtotal2,
+				ordd_pay8, ord_dlen));
 			GG_HIP(hipStreamSynchronize(e.stream));
 		}
 		else
@@ -2022,10 +2014,10 @@ orders_done:;
 		Timed tm(e.stream);
 
 		if (ord_dlen)
-			GG_HIP(launch_dn_probe_lineitem_q5(
+			GG_HIP(launch_dn_probe_lineitem_q5_u8(
 				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,
-				ordd_pay, ord_dlen, ordd_bloom, ordd_bwords,
-				supp_dense, supp_dense_len, acc, ctr));
+				ordd_pay8, ord_dlen, supp_dense,
+				supp_dense_len, acc, ctr));
 		else
 			GG_HIP(launch_probe_lineitem_q5(
 				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,

@@ -395,6 +395,29 @@ hipError_t launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 				    unsigned long long *bloom,
 				    uint64_t bwords,
 				    unsigned long long *join_rows);
+hipError_t launch_dn_build_orders_q5_u8(hipStream_t s, const int64_t *okey,
+					 const int64_t *ckey,
+					 const int32_t *odate, int64_t n,
+					 int32_t date_lo, int32_t date_hi,
+					 DeviceHashTable cust,
+					 const uint8_t *cust_dense,
+					 int64_t cust_dlen, uint8_t *pay8,
+					 int64_t dense_len,
+					 unsigned long long *match_count);
+hipError_t launch_dn_insert_orders_q5_u8(hipStream_t s, const int64_t *okey,
+					 const int64_t *rownat, int64_t n,
+					 uint8_t *pay8, int64_t dense_len);
+hipError_t launch_dn_probe_lineitem_q5_u8(hipStream_t s,
+					  const int64_t *okey,
+					  const int64_t *skey,
+					  const int64_t *price,
+					  const int64_t *disc, int64_t n,
+					  const uint8_t *pay8,
+					  int64_t dense_len,
+					  const uint8_t *supp_dense,
+					  int64_t supp_dlen,
+					  unsigned long long *acc,
+					  unsigned long long *join_rows);
 hipError_t launch_dn_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
 				       const int64_t *skey,
 				       const int64_t *price,
