@@ -874,7 +874,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	}
 	if (ord_dlen)
 	{
-		ordd_bwords = next_pow2((uint64_t) (od->nrows / 32 + 1024));
+		/* exact membership bitmap over dense keys */
+		ordd_bwords = (uint64_t) (ord_dlen / 64 + 2);
 		ordd_pay = (unsigned long long *)
 			p->sget("ordd.pay", (size_t) ord_dlen * 8);
 		ordd_rev = (unsigned long long *)
