@@ -2515,6 +2515,286 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 				     out_nulls, cap, out_nrows);
 }
 
+/* -------------- MemTuple codec (ABI surface, §8(f)3) -------------- */
+
+/*
+ * Binding introspection (host-only — CPU-testable): writes
+ * [column_align, null_bitmap_extra, var_start] then per attr
+ * [offset, len, len_aligned, null_byte, null_mask] into out
+ * (3 + 5*natts int32s).  Pinned against the reference's
+ * create_memtuple_binding in tests.
+ */
+extern "C" gg_status
+gg_engine_memtuple_binding(int natts, const int32_t *attlen,
+			   const char *attalign, int32_t *out)
+{
+	if (!attlen || !attalign || !out)
+		return fail(GG_EINVAL, "bad memtuple_binding args");
+
+	MtBind b;
+	int rc = mt_compute_binding(natts, attlen, attalign, &b);
+
+	if (rc)
+		return fail(GG_EINVAL, "unsupported memtuple schema (%d)",
+			    rc);
+	out[0] = b.column_align;
+	out[1] = b.null_bitmap_extra;
+	out[2] = b.var_start;
+	for (int i = 0; i < natts; i++)
+	{
+		out[3 + i * 5 + 0] = b.offset[i];
+		out[3 + i * 5 + 1] = b.len[i];
+		out[3 + i * 5 + 2] = b.len_aligned[i];
+		out[3 + i * 5 + 3] = b.null_byte[i];
+		out[3 + i * 5 + 4] = b.null_mask[i];
+	}
+	return GG_OK;
+}
+
+/*
+ * Bulk-encode host column arrays into a MemTuple byte stream (GPU does
+ * the per-tuple layout).  cols[i] points at nrows elements of width
+ * attlen[i]; nulls[i] is a byte-per-row flag array or NULL for a
+ * NOT NULL column.  See memtuple.hip for the restated format rules.
+ */
+extern "C" gg_status
+gg_engine_memtuple_encode(int natts, const int32_t *attlen,
+			  const char *attalign, const void *const *cols,
+			  const uint8_t *const *nulls, int64_t nrows,
+			  uint8_t *out, int64_t cap, int64_t *out_len)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!attlen || !attalign || !cols || nrows < 0 || !out || !out_len)
+		return fail(GG_EINVAL, "bad memtuple_encode args");
+
+	MtBind b;
+	int rc = mt_compute_binding(natts, attlen, attalign, &b);
+
+	if (rc)
+		return fail(GG_EINVAL, "unsupported memtuple schema (%d)",
+			    rc);
+
+	/* per-row lengths + offsets on the host (it owns the null flags) */
+	std::vector<int64_t> offs(nrows + 1, 0);
+
+	for (int64_t r = 0; r < nrows; r++)
+	{
+		int64_t len = b.var_start;
+		bool hasnull = false;
+
+		for (int i = 0; i < natts; i++)
+			if (nulls && nulls[i] && nulls[i][r])
+			{
+				hasnull = true;
+				len -= b.len_aligned[i];
+			}
+		if (hasnull)
+			len += b.null_bitmap_extra;
+		offs[r + 1] = offs[r] + ((len + 7) & ~(int64_t) 7);
+	}
+	int64_t total = offs[nrows];
+
+	if (total > cap)
+		return fail(GG_EINVAL, "cap %lld < stream %lld",
+			    (long long) cap, (long long) total);
+	*out_len = total;
+	if (nrows == 0)
+		return GG_OK;
+
+	gg_status st = GG_OK;
+	uint8_t *d_out = nullptr;
+	int64_t *d_offs = nullptr;
+	std::vector<void *> d_cols(natts, nullptr);
+	std::vector<uint8_t *> d_nulls(natts, nullptr);
+	void **d_colp = nullptr;
+	uint8_t **d_nullp = nullptr;
+
+	GG_HIP(hipMalloc((void **) &d_out, (size_t) total));
+	GG_HIP(hipMalloc((void **) &d_offs, (size_t) (nrows + 1) * 8));
+	GG_HIP(hipMemcpy(d_offs, offs.data(), (size_t) (nrows + 1) * 8,
+			 hipMemcpyHostToDevice));
+	for (int i = 0; i < natts; i++)
+	{
+		size_t nb = (size_t) nrows * attlen[i];
+
+		GG_HIP(hipMalloc(&d_cols[i], nb));
+		GG_HIP(hipMemcpy(d_cols[i], cols[i], nb,
+				 hipMemcpyHostToDevice));
+		if (nulls && nulls[i])
+		{
+			GG_HIP(hipMalloc((void **) &d_nulls[i],
+					 (size_t) nrows));
+			GG_HIP(hipMemcpy(d_nulls[i], nulls[i],
+					 (size_t) nrows,
+					 hipMemcpyHostToDevice));
+		}
+	}
+	GG_HIP(hipMalloc((void **) &d_colp, natts * sizeof(void *)));
+	GG_HIP(hipMalloc((void **) &d_nullp, natts * sizeof(void *)));
+	GG_HIP(hipMemcpy(d_colp, d_cols.data(), natts * sizeof(void *),
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_nullp, d_nulls.data(), natts * sizeof(void *),
+			 hipMemcpyHostToDevice));
+	{
+		hipError_t he = launch_mt_encode(
+			e.stream, &b, (const void *const *) d_colp,
+			(const uint8_t *const *) d_nullp, nrows, d_offs,
+			d_out);
+
+		if (he != hipSuccess)
+			st = fail(GG_EGPU, "mt_encode: %s",
+				  hipGetErrorString(he));
+	}
+	if (st == GG_OK)
+	{
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(out, d_out, (size_t) total,
+				 hipMemcpyDeviceToHost));
+	}
+	(void) hipFree(d_out);
+	(void) hipFree(d_offs);
+	for (int i = 0; i < natts; i++)
+	{
+		(void) hipFree(d_cols[i]);
+		(void) hipFree(d_nulls[i]);
+	}
+	(void) hipFree(d_colp);
+	(void) hipFree(d_nullp);
+	return st;
+}
+
+extern "C" gg_status
+gg_engine_memtuple_decode(int natts, const int32_t *attlen,
+			  const char *attalign, const uint8_t *stream,
+			  int64_t stream_len, void *const *cols,
+			  uint8_t *const *nulls, int64_t cap_rows,
+			  int64_t *out_nrows)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!attlen || !attalign || !stream || stream_len < 0 || !cols ||
+	    !out_nrows)
+		return fail(GG_EINVAL, "bad memtuple_decode args");
+
+	MtBind b;
+	int rc = mt_compute_binding(natts, attlen, attalign, &b);
+
+	if (rc)
+		return fail(GG_EINVAL, "unsupported memtuple schema (%d)",
+			    rc);
+
+	/* walk tuple headers to find offsets (MEMTUP_LEN_MASK) */
+	std::vector<int64_t> offs;
+	int64_t pos = 0;
+
+	while (pos < stream_len)
+	{
+		uint32_t hdr;
+
+		if (pos + 8 > stream_len)
+			return fail(GG_EINVAL, "truncated memtuple at %lld",
+				    (long long) pos);
+		std::memcpy(&hdr, stream + pos, 4);
+		if (!(hdr & 0x80000000u))
+			return fail(GG_EINVAL,
+				    "lead bit clear at %lld (heaptuple?)",
+				    (long long) pos);
+		int64_t len = (int64_t) (hdr & 0x3FFFFFF8u);
+
+		if (len < 8 || pos + len > stream_len)
+			return fail(GG_EINVAL, "bad memtuple len %lld at "
+				    "%lld", (long long) len,
+				    (long long) pos);
+		offs.push_back(pos);
+		pos += len;
+	}
+	int64_t nrows = (int64_t) offs.size();
+
+	if (nrows > cap_rows)
+		return fail(GG_EINVAL, "cap_rows %lld < rows %lld",
+			    (long long) cap_rows, (long long) nrows);
+	*out_nrows = nrows;
+	if (nrows == 0)
+		return GG_OK;
+
+	gg_status st = GG_OK;
+	uint8_t *d_in = nullptr;
+	int64_t *d_offs = nullptr;
+	unsigned long long *d_err = nullptr;
+	std::vector<void *> d_cols(natts, nullptr);
+	std::vector<uint8_t *> d_nulls(natts, nullptr);
+	void **d_colp = nullptr;
+	uint8_t **d_nullp = nullptr;
+
+	GG_HIP(hipMalloc((void **) &d_in, (size_t) stream_len));
+	GG_HIP(hipMalloc((void **) &d_offs, (size_t) nrows * 8));
+	GG_HIP(hipMalloc((void **) &d_err, 8));
+	GG_HIP(hipMemcpy(d_in, stream, (size_t) stream_len,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_offs, offs.data(), (size_t) nrows * 8,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemset(d_err, 0, 8));
+	for (int i = 0; i < natts; i++)
+	{
+		GG_HIP(hipMalloc(&d_cols[i], (size_t) nrows * attlen[i]));
+		GG_HIP(hipMalloc((void **) &d_nulls[i], (size_t) nrows));
+	}
+	GG_HIP(hipMalloc((void **) &d_colp, natts * sizeof(void *)));
+	GG_HIP(hipMalloc((void **) &d_nullp, natts * sizeof(void *)));
+	GG_HIP(hipMemcpy(d_colp, d_cols.data(), natts * sizeof(void *),
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_nullp, d_nulls.data(), natts * sizeof(void *),
+			 hipMemcpyHostToDevice));
+	{
+		hipError_t he = launch_mt_decode(
+			e.stream, &b, d_offs, nrows, d_in,
+			(void *const *) d_colp,
+			(uint8_t *const *) d_nullp, d_err);
+
+		if (he != hipSuccess)
+			st = fail(GG_EGPU, "mt_decode: %s",
+				  hipGetErrorString(he));
+	}
+	if (st == GG_OK)
+	{
+		unsigned long long herr = 0;
+
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(&herr, d_err, 8, hipMemcpyDeviceToHost));
+		if (herr)
+			st = fail(GG_EINVAL, "memtuple decode error 0x%llx",
+				  herr);
+		else
+			for (int i = 0; i < natts; i++)
+			{
+				GG_HIP(hipMemcpy(cols[i], d_cols[i],
+						 (size_t) nrows * attlen[i],
+						 hipMemcpyDeviceToHost));
+				if (nulls && nulls[i])
+					GG_HIP(hipMemcpy(nulls[i],
+							 d_nulls[i],
+							 (size_t) nrows,
+							 hipMemcpyDeviceToHost));
+			}
+	}
+	(void) hipFree(d_in);
+	(void) hipFree(d_offs);
+	(void) hipFree(d_err);
+	for (int i = 0; i < natts; i++)
+	{
+		(void) hipFree(d_cols[i]);
+		(void) hipFree(d_nulls[i]);
+	}
+	(void) hipFree(d_colp);
+	(void) hipFree(d_nullp);
+	return st;
+}
+
 /* ---------------- general hash group-by (ABI surface) ---------------- */
 
 extern "C" gg_status

@@ -395,6 +395,33 @@ hipError_t launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 				    unsigned long long *bloom,
 				    uint64_t bwords,
 				    unsigned long long *join_rows);
+#define GG_MT_MAX_ATTS 32
+
+/* memtuple.hip — MemTuple codec */
+struct MtBind
+{
+	int32_t natts;
+	int32_t column_align;
+	int32_t null_bitmap_extra;
+	int32_t var_start;
+	int32_t offset[GG_MT_MAX_ATTS];
+	int32_t len[GG_MT_MAX_ATTS];
+	int32_t len_aligned[GG_MT_MAX_ATTS];
+	int32_t null_byte[GG_MT_MAX_ATTS];
+	uint8_t null_mask[GG_MT_MAX_ATTS];
+	int32_t phys[GG_MT_MAX_ATTS];
+};
+int mt_compute_binding(int natts, const int32_t *attlen,
+		       const char *attalign, MtBind *out);
+hipError_t launch_mt_encode(hipStream_t s, const MtBind *b,
+			    const void *const *cols,
+			    const uint8_t *const *nulls, int64_t nrows,
+			    const int64_t *offs, uint8_t *out);
+hipError_t launch_mt_decode(hipStream_t s, const MtBind *b,
+			    const int64_t *offs, int64_t nrows,
+			    const uint8_t *in, void *const *cols,
+			    uint8_t *const *nulls, unsigned long long *err);
+
 hipError_t launch_dn_build_orders_q5_u8(hipStream_t s, const int64_t *okey,
 					 const int64_t *ckey,
 					 const int32_t *odate, int64_t n,

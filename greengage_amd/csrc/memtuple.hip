@@ -1,0 +1,340 @@
+/*
+ * MemTuple codec — GPDB's compact tuple format, used in executor hash
+ * tables and on the Motion wire (reference
+ * access/common/memtuple.c, format comment :24–35).
+ *
+ * The engine's native exchange is columnar (SoA over RCCL) — strictly
+ * better on a GPU — so this codec is the BOUNDARY piece: bulk
+ * conversion between device column arrays and MemTuple byte streams,
+ * for interoperating with CPU segments in a mixed gang.
+ *
+ * Restated layout rules (each cited to the reference):
+ *  - binding: attrs are laid out physically by descending alignment
+ *    pass (8-aligned, then 4, 2, 1 — create_col_bind,
+ *    memtuple.c:200–416); offset starts at 8 when any 'd'-aligned attr
+ *    exists else 4 (create_memtuple_binding :420–440,
+ *    cur_offset init :206);
+ *  - header: uint32 [lead bit 0x80000000 | len | HASNULL=1]
+ *    (memtup.h MEMTUP_* macros); len is MEMTUP_ALIGN(…)=8-aligned;
+ *  - null bitmap: one bit per PHYSICAL column (null_byte/null_mask,
+ *    :262–264), stored at tuple+4; when it needs more than the 4
+ *    header-pad bytes (col_align 8) or any bytes at all (col_align 4),
+ *    attrs shift by null_bitmap_extra_size
+ *    (compute_null_bitmap_extra_size :58–71, form_to :685–691);
+ *  - null attrs occupy no space: each later attr moves back by the sum
+ *    of len_aligned of preceding-physical null attrs
+ *    (add_null_save_aligned :130, memtuple_get_attr_ptr :524 with
+ *    null_saves_aligned, form_to :717);
+ *  - tuples over 0xFFF0 bytes use the large binding (form_to :622) —
+ *    identical to the small one for fixed-width attrs, which is all
+ *    this round supports (varlena is the remaining (f)3 sub-row).
+ *
+ * Parity is pinned against memtuple.c itself compiled in place
+ * (oracle/ref_build: ref_mt_create_binding/ref_mt_form/ref_mt_getattr).
+ */
+#include <hip/hip_runtime.h>
+#include <cstring>
+
+#include "engine_internal.h"
+
+namespace gg
+{
+
+static inline uint32_t
+mt_align(uint32_t off, int align)
+{
+	return (off + (uint32_t) align - 1) & ~((uint32_t) align - 1);
+}
+
+/*
+ * Restatement of create_memtuple_binding + create_col_bind for
+ * fixed-width attrs (attlen 1/2/4/8, attalign c/s/i/d, byval).
+ * Returns 0 on success, nonzero for unsupported schemas.
+ */
+int
+mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
+		   MtBind *out)
+{
+	if (natts < 1 || natts > GG_MT_MAX_ATTS)
+		return 1;
+	for (int i = 0; i < natts; i++)
+	{
+		if (attlen[i] != 1 && attlen[i] != 2 && attlen[i] != 4 &&
+		    attlen[i] != 8)
+			return 2;	/* varlena: next round */
+		char a = attalign[i];
+
+		if (a != 'c' && a != 's' && a != 'i' && a != 'd')
+			return 3;
+	}
+	out->natts = natts;
+	out->column_align = 4;
+	for (int i = 0; i < natts; i++)
+		if (attlen[i] > 0 && attalign[i] == 'd')
+			out->column_align = 8;
+	{
+		/* compute_null_bitmap_extra_size (memtuple.c:58) */
+		int nbytes = (natts + 7) >> 3;
+		int avail = (out->column_align == 4) ? 0 : 4;
+
+		out->null_bitmap_extra = nbytes <= avail ? 0 :
+			(int32_t) mt_align((uint32_t) (nbytes - avail),
+					   out->column_align);
+	}
+	/* four passes by alignment, as create_col_bind does */
+	uint32_t cur = (out->column_align == 8) ? 8 : 4;
+	int physical = 0;
+	int prev = -1;
+	static const char pass_align[4] = {'d', 'i', 's', 'c'};
+
+	for (int pass = 0; pass < 4; pass++)
+	{
+		for (int i = 0; i < natts; i++)
+		{
+			if (attalign[i] != pass_align[pass])
+				continue;
+			char a = attalign[i];
+			int al = (a == 'd') ? 8 : (a == 'i') ? 4 :
+				(a == 's') ? 2 : 1;
+
+			cur = mt_align(cur, al);
+			out->offset[i] = (int32_t) cur;
+			out->len[i] = attlen[i];
+			out->null_byte[i] = physical >> 3;
+			out->null_mask[i] =
+				(uint8_t) (1u << (physical & 7));
+			out->phys[i] = physical;
+			if (prev >= 0)	/* len_aligned of the PREVIOUS
+					 * physical attr is its len aligned
+					 * to THIS attr's alignment
+					 * (add_null_save_aligned) */
+				out->len_aligned[prev] = (int32_t)
+					mt_align((uint32_t) out->len[prev],
+						 al);
+			prev = i;
+			physical++;
+			cur += (uint32_t) attlen[i];
+		}
+	}
+	if (prev >= 0)		/* last attr: no extra alignment ('c') */
+		out->len_aligned[prev] = out->len[prev];
+	out->var_start = (int32_t) cur;
+	return 0;
+}
+
+namespace
+{
+
+__device__ inline uint32_t
+d_mt_align8(uint32_t x)
+{
+	return (x + 7u) & ~7u;
+}
+
+/* tuple length for one row (compute_memtuple_size_using_bind :462) */
+__device__ uint32_t
+d_mt_len(const MtBind &b, const uint8_t *const *nulls, int64_t row,
+	 bool *hasnull_out)
+{
+	uint32_t len = (uint32_t) b.var_start;
+	bool hasnull = false;
+
+	for (int i = 0; i < b.natts; i++)
+		if (nulls[i] && nulls[i][row])
+		{
+			hasnull = true;
+			len -= (uint32_t) b.len_aligned[i];
+		}
+	if (hasnull)
+		len += (uint32_t) b.null_bitmap_extra;
+	*hasnull_out = hasnull;
+	return d_mt_align8(len);
+}
+
+__global__ void
+k_mt_encode(MtBind b, const void *const *__restrict__ cols,
+	    const uint8_t *const *__restrict__ nulls, int64_t nrows,
+	    const int64_t *__restrict__ offs, uint8_t *__restrict__ out)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t r = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     r < nrows; r += stride)
+	{
+		bool hasnull;
+		uint32_t len = d_mt_len(b, nulls, r, &hasnull);
+		uint8_t *tup = out + offs[r];
+
+		for (uint32_t z = 0; z < len; z++)
+			tup[z] = 0;
+		/* header: lead bit | len | hasnull (memtup.h) */
+		{
+			uint32_t hdr = 0x80000000u | len |
+				(hasnull ? 1u : 0u);
+
+			memcpy(tup, &hdr, 4);
+		}
+		uint8_t *start = tup +
+			(hasnull ? b.null_bitmap_extra : 0);
+		uint8_t *nullp = tup + 4;
+
+		if (hasnull)
+			for (int i = 0; i < b.natts; i++)
+				if (nulls[i] && nulls[i][r])
+					nullp[b.null_byte[i]] |=
+						b.null_mask[i];
+		for (int i = 0; i < b.natts; i++)
+		{
+			if (nulls[i] && nulls[i][r])
+				continue;
+			/* shift back by len_aligned of preceding-physical
+			 * null attrs (memtuple_get_attr_ptr) */
+			uint32_t save = 0;
+
+			if (hasnull)
+				for (int j = 0; j < b.natts; j++)
+					if (b.phys[j] < b.phys[i] &&
+					    nulls[j] && nulls[j][r])
+						save += (uint32_t)
+							b.len_aligned[j];
+			uint8_t *dst = start + b.offset[i] - save;
+
+			switch (b.len[i])
+			{
+				case 1:
+					*dst = ((const uint8_t *)
+						cols[i])[r];
+					break;
+				case 2:
+					memcpy(dst, (const int16_t *)
+					       cols[i] + r, 2);
+					break;
+				case 4:
+					memcpy(dst, (const int32_t *)
+					       cols[i] + r, 4);
+					break;
+				default:
+					memcpy(dst, (const int64_t *)
+					       cols[i] + r, 8);
+			}
+		}
+	}
+}
+
+__global__ void
+k_mt_decode(MtBind b, const int64_t *__restrict__ offs, int64_t nrows,
+	    const uint8_t *__restrict__ in, void *const *__restrict__ cols,
+	    uint8_t *const *__restrict__ nulls,
+	    unsigned long long *__restrict__ err)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t r = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     r < nrows; r += stride)
+	{
+		const uint8_t *tup = in + offs[r];
+		uint32_t hdr;
+
+		memcpy(&hdr, tup, 4);
+		if (!(hdr & 0x80000000u))
+		{
+			atomicOr(err, 1ull);	/* not a memtuple */
+			continue;
+		}
+		bool hasnull = (hdr & 1u) != 0;
+		const uint8_t *start = tup +
+			(hasnull ? b.null_bitmap_extra : 0);
+		const uint8_t *nullp = tup + 4;
+
+		for (int i = 0; i < b.natts; i++)
+		{
+			bool isnull = hasnull &&
+				(nullp[b.null_byte[i]] & b.null_mask[i]);
+
+			if (nulls[i])
+				nulls[i][r] = isnull ? 1 : 0;
+			if (isnull)
+			{
+				switch (b.len[i])
+				{
+					case 1:
+						((uint8_t *) cols[i])[r] = 0;
+						break;
+					case 2:
+						((int16_t *) cols[i])[r] = 0;
+						break;
+					case 4:
+						((int32_t *) cols[i])[r] = 0;
+						break;
+					default:
+						((int64_t *) cols[i])[r] = 0;
+				}
+				continue;
+			}
+			uint32_t save = 0;
+
+			if (hasnull)
+				for (int j = 0; j < b.natts; j++)
+					if (b.phys[j] < b.phys[i] &&
+					    (nullp[b.null_byte[j]] &
+					     b.null_mask[j]))
+						save += (uint32_t)
+							b.len_aligned[j];
+			const uint8_t *src = start + b.offset[i] - save;
+
+			switch (b.len[i])
+			{
+				case 1:
+					((uint8_t *) cols[i])[r] = *src;
+					break;
+				case 2:
+					memcpy((int16_t *) cols[i] + r, src,
+					       2);
+					break;
+				case 4:
+					memcpy((int32_t *) cols[i] + r, src,
+					       4);
+					break;
+				default:
+					memcpy((int64_t *) cols[i] + r, src,
+					       8);
+			}
+		}
+	}
+}
+
+int
+mt_grid(int64_t n)
+{
+	int64_t blk = (n + 255) / 256;
+
+	if (blk > 2048)
+		blk = 2048;
+	if (blk < 1)
+		blk = 1;
+	return (int) blk;
+}
+
+}				/* anonymous namespace */
+
+hipError_t
+launch_mt_encode(hipStream_t s, const MtBind *b, const void *const *cols,
+		 const uint8_t *const *nulls, int64_t nrows,
+		 const int64_t *offs, uint8_t *out)
+{
+	hipLaunchKernelGGL(k_mt_encode, dim3(mt_grid(nrows)), dim3(256), 0,
+			   s, *b, cols, nulls, nrows, offs, out);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_mt_decode(hipStream_t s, const MtBind *b, const int64_t *offs,
+		 int64_t nrows, const uint8_t *in, void *const *cols,
+		 uint8_t *const *nulls, unsigned long long *err)
+{
+	hipLaunchKernelGGL(k_mt_decode, dim3(mt_grid(nrows)), dim3(256), 0,
+			   s, *b, offs, nrows, in, cols, nulls, err);
+	return hipGetLastError();
+}
+
+}				/* namespace gg */

@@ -176,8 +176,12 @@ k_dn_build_orders(const int64_t *__restrict__ okey,
 		if (!ok || k < 0 || k >= dense_len)
 			continue;
 		matches++;
-		pay[k] = (unsigned long long) (uint32_t) d |
-			((unsigned long long) (uint32_t) prio[i] << 32);
+		/* NT store: skip the read-for-ownership on lines we only
+		 * partially fill (~39% of keys pass the date filter) */
+		__builtin_nontemporal_store(
+			(unsigned long long) (uint32_t) d |
+			((unsigned long long) (uint32_t) prio[i] << 32),
+			&pay[k]);
 		dn_bit_set(bloom, k);
 	}
 	for (int off = 32; off; off >>= 1)
@@ -528,7 +532,7 @@ k_dn_build_orders_q5_u8(const int64_t *__restrict__ okey,
 		if (!ok || k < 0 || k >= dense_len)
 			continue;
 		matches++;
-		pay8[k] = (uint8_t) nat;
+		__builtin_nontemporal_store((uint8_t) nat, &pay8[k]);
 	}
 	for (int off = 32; off; off >>= 1)
 		matches += __shfl_down(matches, off, 64);

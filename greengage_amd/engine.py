@@ -112,6 +112,22 @@ def _load():
     lib.gg_engine_aocs_decode.argtypes = [
         ctypes.c_void_p, I64, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
         ctypes.c_int, ctypes.c_void_p, I64, ctypes.POINTER(I64)]
+    lib.gg_engine_memtuple_binding.restype = ctypes.c_int
+    lib.gg_engine_memtuple_binding.argtypes = [
+        ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
+        ctypes.POINTER(ctypes.c_int32)]
+    lib.gg_engine_memtuple_encode.restype = ctypes.c_int
+    lib.gg_engine_memtuple_encode.argtypes = [
+        ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_void_p),
+        ctypes.c_int64, ctypes.c_void_p, ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64)]
+    lib.gg_engine_memtuple_decode.restype = ctypes.c_int
+    lib.gg_engine_memtuple_decode.argtypes = [
+        ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.POINTER(ctypes.c_void_p),
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64)]
     lib.gg_engine_aocs_decode_ao.restype = ctypes.c_int
     lib.gg_engine_aocs_decode_ao.argtypes = [
         ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64, ctypes.c_int,
@@ -339,6 +355,66 @@ class Engine:
             nulls.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)), nmax,
             ctypes.byref(n)), "aocs_decode_ao")
         return vals[:n.value], nulls[:n.value]
+
+    @staticmethod
+    def memtuple_binding(attlen, attalign):
+        """Host-only restated binding layout (CPU-testable)."""
+        natts = len(attlen)
+        al = (ctypes.c_int32 * natts)(*attlen)
+        out = (ctypes.c_int32 * (3 + 5 * natts))()
+        _check(lib().gg_engine_memtuple_binding(
+            natts, al, "".join(attalign).encode(), out),
+            "memtuple_binding")
+        meta = (out[0], out[1], out[2])
+        per = [tuple(out[3 + i * 5: 8 + i * 5]) for i in range(natts)]
+        return meta, per
+
+    @staticmethod
+    def memtuple_encode(attlen, attalign, cols, nulls):
+        """Bulk GPU encode of column arrays into a MemTuple stream.
+        cols[i]: numpy array with itemsize attlen[i]; nulls[i]: uint8
+        array or None."""
+        import numpy as np
+        natts = len(attlen)
+        al = (ctypes.c_int32 * natts)(*attlen)
+        nrows = len(cols[0])
+        carr = [np.ascontiguousarray(c) for c in cols]
+        colp = (ctypes.c_void_p * natts)(
+            *[c.ctypes.data_as(ctypes.c_void_p).value for c in carr])
+        narr = [None if n is None else
+                np.ascontiguousarray(n, np.uint8) for n in nulls]
+        nullp = (ctypes.c_void_p * natts)(
+            *[0 if n is None else
+              n.ctypes.data_as(ctypes.c_void_p).value for n in narr])
+        cap = nrows * (8 + sum(8 + x for x in attlen)) + 64
+        out = np.zeros(cap, np.uint8)
+        olen = I64()
+        _check(lib().gg_engine_memtuple_encode(
+            natts, al, "".join(attalign).encode(), colp, nullp, nrows,
+            out.ctypes.data_as(ctypes.c_void_p), cap,
+            ctypes.byref(olen)), "memtuple_encode")
+        return out[:olen.value].copy()
+
+    @staticmethod
+    def memtuple_decode(attlen, attalign, stream, cap_rows):
+        import numpy as np
+        natts = len(attlen)
+        al = (ctypes.c_int32 * natts)(*attlen)
+        stream = np.ascontiguousarray(stream, np.uint8)
+        dt = {1: np.uint8, 2: np.int16, 4: np.int32, 8: np.int64}
+        cols = [np.zeros(cap_rows, dt[l]) for l in attlen]
+        nulls = [np.zeros(cap_rows, np.uint8) for _ in attlen]
+        colp = (ctypes.c_void_p * natts)(
+            *[c.ctypes.data_as(ctypes.c_void_p).value for c in cols])
+        nullp = (ctypes.c_void_p * natts)(
+            *[n.ctypes.data_as(ctypes.c_void_p).value for n in nulls])
+        n = I64()
+        _check(lib().gg_engine_memtuple_decode(
+            natts, al, "".join(attalign).encode(),
+            stream.ctypes.data_as(ctypes.c_void_p), len(stream), colp,
+            nullp, cap_rows, ctypes.byref(n)), "memtuple_decode")
+        return ([c[:n.value] for c in cols],
+                [nl[:n.value] for nl in nulls])
 
     # ---- general hash group-by (arbitrary int64 keys, SUM+COUNT) ----
     @staticmethod

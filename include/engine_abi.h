@@ -214,6 +214,27 @@ gg_status gg_engine_aocs_decode(const uint8_t *stream, int64_t stream_len,
 				int out_width, uint8_t *out_nulls,
 				int64_t cap, int64_t *out_nrows);
 
+/* MemTuple codec (access/common/memtuple.c format, the tuple layout
+ * used in executor hash tables and on the Motion wire): bulk GPU
+ * conversion between column arrays and MemTuple byte streams.
+ * Fixed-width by-value attrs (attlen 1/2/4/8, attalign c/s/i/d),
+ * nullable; cols[i] holds nrows elements of width attlen[i], nulls[i]
+ * is a byte-per-row flag array or NULL for NOT NULL columns. */
+gg_status gg_engine_memtuple_binding(int natts, const int32_t *attlen,
+				     const char *attalign, int32_t *out);
+gg_status gg_engine_memtuple_encode(int natts, const int32_t *attlen,
+				    const char *attalign,
+				    const void *const *cols,
+				    const uint8_t *const *nulls,
+				    int64_t nrows, uint8_t *out,
+				    int64_t cap, int64_t *out_len);
+gg_status gg_engine_memtuple_decode(int natts, const int32_t *attlen,
+				    const char *attalign,
+				    const uint8_t *stream,
+				    int64_t stream_len, void *const *cols,
+				    uint8_t *const *nulls,
+				    int64_t cap_rows, int64_t *out_nrows);
+
 /* Decode REAL Append-Only storage blocks (headers + CRC32C verify +
  * datum-stream content on the GPU).  Replaces the AO read path's
  * header/checksum layer (cdbappendonlystorageformat.c:125,1661 and

@@ -247,6 +247,24 @@ def dsb_ref():
         D.ref_ao_wrap_stream_c.argtypes = [
             _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.c_int, _P_U8, I64, ctypes.POINTER(I64)]
+        D.ref_mt_create_binding.restype = ctypes.c_void_p
+        D.ref_mt_create_binding.argtypes = [
+            ctypes.c_int, ctypes.POINTER(ctypes.c_int32), _P_U8,
+            ctypes.c_char_p, ctypes.POINTER(ctypes.c_uint32)]
+        D.ref_mt_form.restype = ctypes.c_int32
+        D.ref_mt_form.argtypes = [
+            ctypes.c_void_p, _P_I64, _P_U8, _P_U8, ctypes.c_int32]
+        D.ref_mt_getattr.restype = ctypes.c_int32
+        D.ref_mt_getattr.argtypes = [
+            ctypes.c_void_p, _P_U8, ctypes.c_int, ctypes.POINTER(I64),
+            ctypes.POINTER(ctypes.c_uint8)]
+        D.ref_mt_get_colbind.restype = ctypes.c_int32
+        D.ref_mt_get_colbind.argtypes = [
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_int] + \
+            [ctypes.POINTER(ctypes.c_int32)] * 6
+        D.ref_mt_binding_info.restype = ctypes.c_int32
+        D.ref_mt_binding_info.argtypes = [
+            ctypes.c_void_p] + [ctypes.POINTER(ctypes.c_int32)] * 3
         D.ref_ao_probe_block.restype = ctypes.c_int
         D.ref_ao_probe_block.argtypes = [
             _P_U8, ctypes.c_int] + [ctypes.POINTER(ctypes.c_int32)] * 6
@@ -329,6 +347,55 @@ def ao_probe(block, checksums=1):
                               *[ctypes.byref(o) for o in outs])
     assert rc == 0, rc
     return tuple(o.value for o in outs)
+
+
+class MtSchema:
+    """A fixed-width memtuple schema bound through the REFERENCE's
+    create_memtuple_binding (memtuple.c compiled in place)."""
+
+    def __init__(self, attlen, attalign, atttypid=None):
+        D = dsb_ref()
+        assert D is not None, "reference memtuple codec missing"
+        self.D = D
+        self.natts = len(attlen)
+        self.attlen = list(attlen)
+        if atttypid is None:
+            atttypid = [{8: 20, 4: 23, 2: 21, 1: 18}[l] for l in attlen]
+        al = (ctypes.c_int32 * self.natts)(*attlen)
+        bv = np.ones(self.natts, np.uint8)
+        ti = (ctypes.c_uint32 * self.natts)(*atttypid)
+        self.bind = D.ref_mt_create_binding(
+            self.natts, al, bv, "".join(attalign).encode(), ti)
+        assert self.bind
+
+    def form(self, values, isnull):
+        v = np.array([int(x) for x in values], np.int64)
+        nl = np.array([int(x) for x in isnull], np.uint8)
+        out = np.zeros(1 << 16, np.uint8)
+        ln = self.D.ref_mt_form(self.bind, v, nl, out, 1 << 16)
+        assert ln > 0, ln
+        return out[:ln].copy()
+
+    def getattr(self, tup, attnum):
+        tup = np.ascontiguousarray(tup, np.uint8)
+        v, isn = I64(), ctypes.c_uint8()
+        rc = self.D.ref_mt_getattr(self.bind, tup, attnum,
+                                   ctypes.byref(v), ctypes.byref(isn))
+        assert rc == 0
+        return v.value, isn.value
+
+    def colbind(self, attnum, large=False):
+        outs = [ctypes.c_int32() for _ in range(6)]
+        rc = self.D.ref_mt_get_colbind(self.bind, attnum, int(large),
+                                       *[ctypes.byref(o) for o in outs])
+        assert rc == 0
+        return tuple(o.value for o in outs)
+
+    def info(self):
+        outs = [ctypes.c_int32() for _ in range(3)]
+        self.D.ref_mt_binding_info(self.bind,
+                                   *[ctypes.byref(o) for o in outs])
+        return tuple(o.value for o in outs)
 
 
 def pgdate(y, m, d):

@@ -388,6 +388,131 @@ ref_ao_wrap_stream_c(const uint8 *framed, int64 framed_len, int checksums,
 	return 0;
 }
 
+/* ------------------------------------------------------------------ */
+/* MemTuple layer (access/common/memtuple.c compiled in place): GPDB's */
+/* compact tuple format used in executor hash tables and on the Motion */
+/* wire (format comment memtuple.c:24-35).  The wrapper exposes the    */
+/* REAL binding/form/getattr so the engine's restated codec can be     */
+/* pinned bit-exactly.                                                 */
+/* ------------------------------------------------------------------ */
+#include "access/memtup.h"
+
+/* only reached for external toast pointers — never from the wrapper */
+struct varlena *
+heap_tuple_fetch_attr(struct varlena *attr)
+{
+	fprintf(stderr, "heap_tuple_fetch_attr: unsupported in wrapper\n");
+	abort();
+}
+
+void *
+ref_mt_create_binding(int natts, const int32 *attlen,
+		      const uint8 *attbyval, const char *attalign,
+		      const uint32 *atttypid)
+{
+	TupleDesc	desc = (TupleDesc) calloc(1, sizeof(*desc));
+	int			i;
+
+	desc->natts = natts;
+	desc->tdhasoid = false;
+	desc->attrs = (Form_pg_attribute *)
+		calloc(natts, sizeof(Form_pg_attribute));
+	for (i = 0; i < natts; i++)
+	{
+		Form_pg_attribute a = (Form_pg_attribute)
+			calloc(1, sizeof(*a));
+
+		a->atttypid = atttypid[i];
+		a->attlen = (int16) attlen[i];
+		a->attbyval = attbyval[i] != 0;
+		a->attalign = attalign[i];
+		a->attisdropped = false;
+		a->attstorage = attlen[i] > 0 ? 'p' : 'x';
+		a->atttypmod = -1;
+		desc->attrs[i] = a;
+	}
+	return create_memtuple_binding(desc);
+}
+
+int32
+ref_mt_get_colbind(void *pbind, int attnum_1based, int use_large,
+		   int32 *offset, int32 *len, int32 *len_aligned,
+		   int32 *flag, int32 *null_byte, int32 *null_mask)
+{
+	MemTupleBinding *b = (MemTupleBinding *) pbind;
+	MemTupleBindingCols *cols = use_large ? &b->large_bind : &b->bind;
+	MemTupleAttrBinding *ab;
+
+	if (attnum_1based < 1 || attnum_1based > b->tupdesc->natts)
+		return -1;
+	ab = &cols->bindings[attnum_1based - 1];
+	*offset = ab->offset;
+	*len = ab->len;
+	*len_aligned = ab->len_aligned;
+	*flag = (int32) ab->flag;
+	*null_byte = ab->null_byte;
+	*null_mask = ab->null_mask;
+	return 0;
+}
+
+int32
+ref_mt_binding_info(void *pbind, int32 *column_align,
+		    int32 *null_bitmap_extra, int32 *var_start)
+{
+	MemTupleBinding *b = (MemTupleBinding *) pbind;
+
+	*column_align = b->column_align;
+	*null_bitmap_extra = b->null_bitmap_extra_size;
+	*var_start = (int32) b->bind.var_start;
+	return 0;
+}
+
+/* form one tuple from int64 datums; returns total length, or -1 */
+int32
+ref_mt_form(void *pbind, const int64 *values, const uint8 *isnull,
+	    uint8 *out, int32 cap)
+{
+	MemTupleBinding *b = (MemTupleBinding *) pbind;
+	int			n = b->tupdesc->natts;
+	Datum		dvals[64];
+	bool		dnull[64];
+	uint32		destlen = 0;
+	int			i;
+	MemTuple	res;
+
+	if (n > 64)
+		return -1;
+	for (i = 0; i < n; i++)
+	{
+		dvals[i] = (Datum) values[i];
+		dnull[i] = isnull[i] != 0;
+	}
+	/* first call computes required length */
+	memtuple_form_to(b, dvals, dnull, NULL, &destlen, false);
+	if ((int32) destlen > cap)
+		return -2;
+	memset(out, 0, destlen);
+	res = memtuple_form_to(b, dvals, dnull, (MemTuple) out, &destlen,
+			       false);
+	if (res == NULL)
+		return -3;
+	return (int32) destlen;
+}
+
+int32
+ref_mt_getattr(void *pbind, uint8 *tup, int attnum_1based, int64 *val,
+	       uint8 *out_isnull)
+{
+	MemTupleBinding *b = (MemTupleBinding *) pbind;
+	bool		isnull = false;
+	Datum		d;
+
+	d = memtuple_getattr((MemTuple) tup, b, attnum_1based, &isnull);
+	*val = (int64) d;
+	*out_isnull = isnull ? 1 : 0;
+	return 0;
+}
+
 /* reference-side parse of one AO block (for cross-checking the
  * engine's restated parser in tests) */
 int

@@ -8,6 +8,9 @@ typedef struct FormData_pg_attribute
 	int16		attlen;
 	bool		attbyval;
 	char		attalign;
+	bool		attisdropped;
+	char		attstorage;
+	int32		atttypmod;
 } FormData_pg_attribute;
 typedef FormData_pg_attribute *Form_pg_attribute;
 #endif

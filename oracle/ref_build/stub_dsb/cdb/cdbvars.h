@@ -1,0 +1,4 @@
+/* Stub: memtuple.c includes this but uses nothing from it. */
+#ifndef ORACLE_STUB_DSB_CDBVARS_H
+#define ORACLE_STUB_DSB_CDBVARS_H
+#endif

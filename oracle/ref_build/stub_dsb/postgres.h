@@ -107,6 +107,9 @@ static inline void pfree(void *p) { free(p); }
 #define DatumGetChar(X) ((char) (X))
 #define DatumGetCString(X) ((char *) (X))
 #define Int32GetDatum(X) ((Datum) (uint32) (X))
+#define Int16GetDatum(X) ((Datum) (uint16) (X))
+#define DatumGetInt16(X) ((int16) (X))
+#define CharGetDatum(X) ((Datum) (uint8) (X))
 #define Int64GetDatum(X) ((Datum) (X))
 #define UInt32GetDatum(X) ((Datum) (X))
 
@@ -220,5 +223,15 @@ psprintf(const char *fmt, ...)
 	va_end(ap);
 	return buf;
 }
+
+
+/* external (toast-pointer) sizes: the wrapper never feeds external
+ * datums; keep the macros compilable, trap if ever reached */
+#define VARHDRSZ_EXTERNAL 2
+#define VARSIZE_EXTERNAL(PTR) (abort(), 0)
+#define VARSIZE_ANY_EXHDR(PTR) \
+	(VARATT_IS_1B_E(PTR) ? (abort(), 0) : \
+	 (VARATT_IS_1B(PTR) ? VARSIZE_1B(PTR) - VARHDRSZ_SHORT : \
+	  VARSIZE_4B(PTR) - VARHDRSZ))
 
 #endif
