@@ -779,10 +779,12 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		return fail(GG_ENOMEM, "scratch");
 
 	/* 1. customer build side (nodeHash.c:450 sizing / :905 insert).
-	 * Dense custkeys (max <= 8x rows) collapse to a u8 membership
-	 * array in L2/L3; hash set fallback for sparse keys. */
+	 * Dense custkeys (max <= 8x rows) collapse to a 1-bit-per-key
+	 * membership bitmap (SF100: 15M keys = 1.9 MB — resident in
+	 * every XCD's L2, unlike a 15 MB byte array); hash set fallback
+	 * for sparse keys. */
 	DeviceHashTable cust{};
-	uint8_t *cust_dense = nullptr;
+	unsigned long long *cust_bits = nullptr;
 	int64_t cust_dlen = 0;
 	{
 		unsigned long long maxk = 0;
@@ -800,15 +802,17 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 		if (cust_dlen)
 		{
-			cust_dense = (uint8_t *)
-				p->sget("cust.dense", (size_t) cust_dlen);
-			if (!cust_dense)
-				return fail(GG_ENOMEM, "cust dense");
-			GG_HIP(hipMemsetAsync(cust_dense, 0,
-					      (size_t) cust_dlen, e.stream));
+			size_t words = (size_t) (cust_dlen / 64 + 2);
+
+			cust_bits = (unsigned long long *)
+				p->sget("cust.bits", words * 8);
+			if (!cust_bits)
+				return fail(GG_ENOMEM, "cust bits");
+			GG_HIP(hipMemsetAsync(cust_bits, 0, words * 8,
+					      e.stream));
 			GG_HIP(launch_cust_dense_fill_seg(e.stream, c_ck,
 							  c_ms, cu->nrows,
-							  segcode, cust_dense,
+							  segcode, cust_bits,
 							  cust_dlen));
 		}
 		else
@@ -901,7 +905,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		{
 			GG_HIP(launch_dn_build_orders(
 				e.stream, o_ok, o_ck, o_dt, o_pr, od->nrows,
-				cutoff, cust, cust_dense, cust_dlen, ordd_pay,
+				cutoff, cust, cust_bits, cust_dlen, ordd_pay,
 				ord_dlen, ordd_bloom, ordd_bwords, ctr));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
@@ -912,7 +916,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			{
 				GG_HIP(launch_count_orders_match(
 					e.stream, o_ck, o_dt, od->nrows,
-					cutoff, cust, cust_dense, cust_dlen,
+					cutoff, cust, cust_bits, cust_dlen,
 					ctr));
 				GG_HIP(hipStreamSynchronize(e.stream));
 				GG_TRY(read_counter(ctr, &nmatch));
@@ -940,7 +944,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 					      ord.bloom_words * 8, e.stream));
 			GG_HIP(launch_build_orders(e.stream, o_ok, o_ck, o_dt,
 						   o_pr, od->nrows, cutoff,
-						   cust, cust_dense, cust_dlen,
+						   cust, cust_bits, cust_dlen,
 						   ord, ctr));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
@@ -1057,7 +1061,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 		GG_HIP(launch_probe_cust_compact(e.stream, r_ck, r_ok, r_pay,
 						 (int64_t) rtotal, cust,
-						 cust_dense, cust_dlen, m_ok,
+						 cust_bits, cust_dlen, m_ok,
 						 m_pay, ctr));
 		GG_HIP(hipStreamSynchronize(e.stream));
 		GG_TRY(read_counter(ctr, &nm));

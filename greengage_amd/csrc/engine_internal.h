@@ -275,12 +275,12 @@ hipError_t launch_build_set(hipStream_t s, const int64_t *keys,
 hipError_t launch_build_orders(hipStream_t s, const int64_t *okey,
 			       const int64_t *ckey, const int32_t *odate,
 			       const int32_t *prio, int64_t n, int32_t cutoff,
-			       DeviceHashTable cust, const uint8_t *cust_dense,
+			       DeviceHashTable cust, const unsigned long long *cust_bits,
 			       int64_t cust_dlen, DeviceHashTable ord,
 			       unsigned long long *match_count);
 hipError_t launch_cust_dense_fill_seg(hipStream_t s, const int64_t *custkey,
 				      const uint8_t *mktseg, int64_t n,
-				      uint8_t segcode, uint8_t *dense,
+				      uint8_t segcode, unsigned long long *bits,
 				      int64_t dense_len);
 hipError_t launch_cust_dense_fill_nat(hipStream_t s, const int64_t *custkey,
 				      const uint8_t *nation, int64_t n,
@@ -311,7 +311,7 @@ hipError_t launch_count_filter_u8(hipStream_t s, const uint8_t *col,
 hipError_t launch_count_orders_match(hipStream_t s, const int64_t *ckey,
 				     const int32_t *odate, int64_t n,
 				     int32_t cutoff, DeviceHashTable cust,
-				     const uint8_t *cust_dense,
+				     const unsigned long long *cust_bits,
 				     int64_t cust_dlen,
 				     unsigned long long *out);
 hipError_t launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
@@ -325,7 +325,7 @@ hipError_t launch_orders_filter_compact(hipStream_t s, const int64_t *okey,
 hipError_t launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,
 				     const int64_t *okey, const int64_t *pay,
 				     int64_t n, DeviceHashTable cust,
-				     const uint8_t *cust_dense,
+				     const unsigned long long *cust_bits,
 				     int64_t cust_dlen,
 				     int64_t *out_okey, int64_t *out_pay,
 				     unsigned long long *out_count);
@@ -362,7 +362,7 @@ hipError_t launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 				  const int64_t *ckey, const int32_t *odate,
 				  const int32_t *prio, int64_t n,
 				  int32_t cutoff, DeviceHashTable cust,
-				  const uint8_t *cust_dense,
+				  const unsigned long long *cust_bits,
 				  int64_t cust_dlen, unsigned long long *pay,
 				  int64_t dense_len, unsigned long long *bloom,
 				  uint64_t bwords,

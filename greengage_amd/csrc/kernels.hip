@@ -637,7 +637,8 @@ k_build_orders(const int64_t *__restrict__ okey,
 	       uint64_t cust_slots,
 	       const unsigned long long *__restrict__ cust_bloom,
 	       uint64_t cust_bwords,
-	       const uint8_t *__restrict__ cust_dense, int64_t cust_dlen,
+	       const unsigned long long *__restrict__ cust_bits,
+	       int64_t cust_dlen,
 	       unsigned long long *__restrict__ tkeys,
 	       unsigned long long *__restrict__ tpayload, uint64_t nslots,
 	       unsigned long long *__restrict__ bloom, uint64_t bwords,
@@ -658,9 +659,10 @@ k_build_orders(const int64_t *__restrict__ okey,
 
 		if (d >= cutoff)	/* qual: o_orderdate < cutoff */
 			continue;
-		if (cust_dense)
+		if (cust_bits)
 		{
-			if (ck < 0 || ck >= cust_dlen || !cust_dense[ck])
+			if (ck < 0 || ck >= cust_dlen ||
+			    !((cust_bits[ck >> 6] >> (ck & 63)) & 1))
 				continue;
 		}
 		else if (!ht_contains_b(cust_keys, cust_slots, cust_bloom,
@@ -702,13 +704,13 @@ hipError_t
 launch_build_orders(hipStream_t s, const int64_t *okey, const int64_t *ckey,
 		    const int32_t *odate, const int32_t *prio, int64_t n,
 		    int32_t cutoff, DeviceHashTable cust,
-		    const uint8_t *cust_dense, int64_t cust_dlen,
+		    const unsigned long long *cust_bits, int64_t cust_dlen,
 		    DeviceHashTable ord, unsigned long long *match_count)
 {
 	hipLaunchKernelGGL(k_build_orders, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, okey, ckey, odate, prio, n, cutoff,
 			   cust.keys, cust.nslots, cust.bloom,
-			   cust.bloom_words, cust_dense, cust_dlen,
+			   cust.bloom_words, cust_bits, cust_dlen,
 			   ord.keys, ord.payload,
 			   ord.nslots, ord.bloom, ord.bloom_words,
 			   match_count);
@@ -741,7 +743,7 @@ k_count_orders_match(const int64_t *__restrict__ ckey,
 		     uint64_t cust_slots,
 		     const unsigned long long *__restrict__ cust_bloom,
 		     uint64_t cust_bwords,
-		     const uint8_t *__restrict__ cust_dense,
+		     const unsigned long long *__restrict__ cust_bits,
 		     int64_t cust_dlen, unsigned long long *out)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
@@ -752,11 +754,12 @@ k_count_orders_match(const int64_t *__restrict__ ckey,
 	{
 		if (odate[i] >= cutoff)
 			continue;
-		if (cust_dense)
+		if (cust_bits)
 		{
 			int64_t ck = ckey[i];
 
-			c += (ck >= 0 && ck < cust_dlen && cust_dense[ck]);
+			c += (ck >= 0 && ck < cust_dlen &&
+			      ((cust_bits[ck >> 6] >> (ck & 63)) & 1));
 		}
 		else
 			c += ht_contains_b(cust_keys, cust_slots, cust_bloom,
@@ -1193,7 +1196,7 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 		     uint64_t cust_slots,
 		     const unsigned long long *__restrict__ cust_bloom,
 		     uint64_t cust_bwords,
-		     const uint8_t *__restrict__ cust_dense,
+		     const unsigned long long *__restrict__ cust_bits,
 		     int64_t cust_dlen, int64_t *__restrict__ out_okey,
 		     int64_t *__restrict__ out_pay,
 		     unsigned long long *out_count)
@@ -1204,8 +1207,9 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 	     i < n; i += stride)
 	{
 		int64_t ck = ckey[i];
-		bool take = cust_dense
-			? (ck >= 0 && ck < cust_dlen && cust_dense[ck] != 0)
+		bool take = cust_bits
+			? (ck >= 0 && ck < cust_dlen &&
+			   ((cust_bits[ck >> 6] >> (ck & 63)) & 1))
 			: ht_contains_b(cust_keys, cust_slots, cust_bloom,
 					cust_bwords, ck);
 		unsigned long long idx = wave_append(out_count, take);
@@ -1220,14 +1224,15 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 hipError_t
 launch_probe_cust_compact(hipStream_t s, const int64_t *ckey,
 			  const int64_t *okey, const int64_t *pay, int64_t n,
-			  DeviceHashTable cust, const uint8_t *cust_dense,
+			  DeviceHashTable cust,
+			  const unsigned long long *cust_bits,
 			  int64_t cust_dlen, int64_t *out_okey,
 			  int64_t *out_pay, unsigned long long *out_count)
 {
 	hipLaunchKernelGGL(k_probe_cust_compact, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, ckey, okey, pay, n, cust.keys,
 			   cust.nslots, cust.bloom, cust.bloom_words,
-			   cust_dense, cust_dlen,
+			   cust_bits, cust_dlen,
 			   out_okey, out_pay, out_count);
 	return hipGetLastError();
 }
@@ -2068,7 +2073,7 @@ launch_supp_dense_fill_pairs(hipStream_t s, const int64_t *sk,
 __global__ void
 k_cust_dense_fill_seg(const int64_t *__restrict__ custkey,
 		      const uint8_t *__restrict__ mktseg, int64_t n,
-		      uint8_t segcode, uint8_t *__restrict__ dense,
+		      uint8_t segcode, unsigned long long *__restrict__ bits,
 		      int64_t dense_len)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
@@ -2079,18 +2084,18 @@ k_cust_dense_fill_seg(const int64_t *__restrict__ custkey,
 		int64_t k = custkey[i];
 
 		if (k >= 0 && k < dense_len && mktseg[i] == segcode)
-			dense[k] = 1;
+			atomicOr(&bits[k >> 6], 1ull << (k & 63));
 	}
 }
 
 hipError_t
 launch_cust_dense_fill_seg(hipStream_t s, const int64_t *custkey,
 			   const uint8_t *mktseg, int64_t n, uint8_t segcode,
-			   uint8_t *dense, int64_t dense_len)
+			   unsigned long long *bits, int64_t dense_len)
 {
 	hipLaunchKernelGGL(k_cust_dense_fill_seg, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, custkey, mktseg, n, segcode,
-			   dense, dense_len);
+			   bits, dense_len);
 	return hipGetLastError();
 }
 
@@ -2135,13 +2140,13 @@ launch_count_filter_u8(hipStream_t s, const uint8_t *col, uint8_t val,
 hipError_t
 launch_count_orders_match(hipStream_t s, const int64_t *ckey,
 			  const int32_t *odate, int64_t n, int32_t cutoff,
-			  DeviceHashTable cust, const uint8_t *cust_dense,
+			  DeviceHashTable cust, const unsigned long long *cust_bits,
 			  int64_t cust_dlen, unsigned long long *out)
 {
 	hipLaunchKernelGGL(k_count_orders_match, dim3(grid_for(n)),
 			   dim3(THREADS), 0, s, ckey, odate, n, cutoff,
 			   cust.keys, cust.nslots, cust.bloom,
-			   cust.bloom_words, cust_dense, cust_dlen, out);
+			   cust.bloom_words, cust_bits, cust_dlen, out);
 	return hipGetLastError();
 }
 

@@ -118,7 +118,8 @@ k_dn_build_orders(const int64_t *__restrict__ okey,
 		  uint64_t cust_slots,
 		  const unsigned long long *__restrict__ cust_bloom,
 		  uint64_t cust_bwords,
-		  const uint8_t *__restrict__ cust_dense, int64_t cust_dlen,
+		  const unsigned long long *__restrict__ cust_bits,
+		  int64_t cust_dlen,
 		  unsigned long long *__restrict__ pay, int64_t dense_len,
 		  unsigned long long *__restrict__ bloom, uint64_t bwords,
 		  unsigned long long *match_count)
@@ -136,8 +137,9 @@ k_dn_build_orders(const int64_t *__restrict__ okey,
 
 		if (d >= cutoff)
 			continue;
-		if (cust_dense)
-			ok = (ck >= 0 && ck < cust_dlen && cust_dense[ck]);
+		if (cust_bits)
+			ok = (ck >= 0 && ck < cust_dlen &&
+			      ((cust_bits[ck >> 6] >> (ck & 63)) & 1));
 		else
 		{
 			uint32_t h = gg_hashint8(ck);
@@ -194,7 +196,8 @@ hipError_t
 launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 		       const int64_t *ckey, const int32_t *odate,
 		       const int32_t *prio, int64_t n, int32_t cutoff,
-		       DeviceHashTable cust, const uint8_t *cust_dense,
+		       DeviceHashTable cust,
+		       const unsigned long long *cust_bits,
 		       int64_t cust_dlen, unsigned long long *pay,
 		       int64_t dense_len, unsigned long long *bloom,
 		       uint64_t bwords, unsigned long long *match_count)
@@ -202,7 +205,7 @@ launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 	hipLaunchKernelGGL(k_dn_build_orders, dim3(dn_grid(n)),
 			   dim3(DN_THREADS), 0, s, okey, ckey, odate, prio, n,
 			   cutoff, cust.keys, cust.nslots, cust.bloom,
-			   cust.bloom_words, cust_dense, cust_dlen, pay,
+			   cust.bloom_words, cust_bits, cust_dlen, pay,
 			   dense_len, bloom, bwords, match_count);
 	return hipGetLastError();
 }
