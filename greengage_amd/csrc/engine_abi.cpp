@@ -2230,7 +2230,8 @@ gg_engine_aocs_decode(const uint8_t *stream, int64_t stream_len, int version,
 		rows.push_back(rc);
 		out_offs.push_back(total_rows);
 		total_rows += rc;
-		pos += 8 + sz;
+		/* frames are 8-aligned, mirroring AO storage block alignment */
+		pos = (pos + 8 + sz + 7) & ~(int64_t) 7;
 	}
 	if (total_rows > cap)
 		return fail(GG_EINVAL, "cap %lld < rows %lld",
@@ -2476,7 +2477,7 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 		}
 		size_t fpos = framed.size();
 
-		framed.resize(fpos + 8 + (size_t) datalen);
+		framed.resize((fpos + 8 + (size_t) datalen + 7) & ~(size_t) 7);
 		std::memcpy(framed.data() + fpos, &datalen, 4);
 		std::memcpy(framed.data() + fpos + 4, &rowcount, 4);
 		if (complen == 0)

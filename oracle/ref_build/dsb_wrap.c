@@ -36,7 +36,13 @@ MemoryContext CurrentMemoryContext = NULL;
 int
 stub_errmsg(const char *fmt,...)
 {
-	(void) fmt;
+	va_list		ap;
+
+	fprintf(stderr, "[refmsg] ");
+	va_start(ap, fmt);
+	vfprintf(stderr, fmt, ap);
+	va_end(ap);
+	fprintf(stderr, "\n");
 	return 0;
 }
 
@@ -52,20 +58,43 @@ stub_errfinish(int level)
 	return 0;
 }
 
-/* varlena helpers: never reached for by-value fixed-length columns */
+/* varlena helpers — restated from the reference for the in-memory
+ * (non-toasted, non-compressed) forms the wrapper produces:
+ * varattrib_untoast_ptr_len (tuptoaster.c:179) and
+ * value_type_could_short (tupmacs.h:202). */
 void
 varattrib_untoast_ptr_len(Datum d, char **datastart, int *len, void **tofree)
 {
-	(void) d; (void) datastart; (void) len; (void) tofree;
-	fprintf(stderr, "varattrib_untoast_ptr_len: varlena unsupported in wrapper\n");
-	abort();
+	char	   *p = (char *) DatumGetPointer(d);
+
+	*tofree = NULL;
+	if (VARATT_IS_EXTERNAL(p) || VARATT_IS_COMPRESSED(p))
+	{
+		fprintf(stderr, "varattrib_untoast_ptr_len: toasted datum "
+			"unsupported in wrapper\n");
+		abort();
+	}
+	if (VARATT_IS_SHORT(p))
+	{
+		*len = VARSIZE_SHORT(p) - VARHDRSZ_SHORT;
+		*datastart = p + VARHDRSZ_SHORT;
+	}
+	else
+	{
+		*len = VARSIZE(p) - VARHDRSZ;
+		*datastart = p + VARHDRSZ;
+	}
 }
 
 bool
 value_type_could_short(Pointer ptr, Oid typid)
 {
-	(void) ptr; (void) typid;
-	return false;
+	return !VARATT_IS_EXTERNAL(ptr) &&
+		(VARATT_IS_SHORT(ptr) ||
+		 (VARATT_CAN_MAKE_SHORT(ptr) &&
+		  typid != 22 /* INT2VECTOROID */ &&
+		  typid != 30 /* OIDVECTOROID */ &&
+		  typid < 16384 /* FirstNormalObjectId */ ));
 }
 
 #define MAXDATUM_ORIG 0x3FFF	/* AOSmallContentHeader_MaxRowCount */
@@ -135,6 +164,11 @@ ref_dsb_encode(const int64 *vals, const uint8 *nulls, int64 n, int datumlen,
 			memcpy(out + pos, &sz, 4);
 			memcpy(out + pos + 4, &rowcount, 4);
 			pos += 8 + sz;
+			/* keep every block 8-aligned, as AO storage does
+			 * (AOStorage_RoundUp8) — the reference's varlena walk
+			 * is absolute-pointer-aligned */
+			while (pos & 7)
+				out[pos++] = 0;
 			nblocks++;
 			DatumStreamBlockWrite_GetReady(&dsw);
 		}
@@ -150,6 +184,8 @@ ref_dsb_encode(const int64 *vals, const uint8 *nulls, int64 n, int datumlen,
 		memcpy(out + pos, &sz, 4);
 		memcpy(out + pos + 4, &rowcount, 4);
 		pos += 8 + sz;
+		while (pos & 7)
+			out[pos++] = 0;
 		nblocks++;
 	}
 	DatumStreamBlockWrite_Finish(&dsw);
@@ -193,6 +229,7 @@ ref_dsb_decode(const uint8 *stream, int64 stream_len, int datumlen,
 									  1 /* firstRowNum */ , rowcount,
 									  &adjusted, &adjustedRowCount);
 		pos += sz;
+		pos = (pos + 7) & ~(int64) 7;
 		for (have = 0; have < rowcount; have++)
 		{
 			Datum		d = 0;
@@ -211,6 +248,184 @@ ref_dsb_decode(const uint8 *stream, int64 stream_len, int datumlen,
 					? (int64) DatumGetInt32(d)
 					: DatumGetInt64(d);
 			nout++;
+		}
+	}
+	DatumStreamBlockRead_Finish(&dsr);
+	*out_n = nout;
+	return 0;
+}
+
+/* ---- varlena (text) variants: typid 25, align 'i', byval false ---- */
+
+#define REF_VARDATA_ANY(PTR) \
+	(VARATT_IS_1B(PTR) ? ((char *) (PTR) + 1) : ((char *) (PTR) + 4))
+#define REF_VARSIZE_ANY_EXHDR(PTR) \
+	(VARATT_IS_1B(PTR) ? VARSIZE_1B(PTR) - 1 : VARSIZE_4B(PTR) - 4)
+
+static void
+typeinfo_text(DatumStreamTypeInfo * ti)
+{
+	ti->datumlen = -1;
+	ti->typid = 25;		/* TEXTOID */
+	ti->align = 'i';
+	ti->byval = false;
+}
+
+/*
+ * Encode n text values (bytes[offs[i]..offs[i+1]) is row i's payload;
+ * nulls[i] nonzero = NULL) with the REFERENCE writer.  Same framing as
+ * ref_dsb_encode.
+ */
+int
+ref_dsb_encode_text(const uint8 *bytes, const int64 *offs,
+		    const uint8 *nulls, int64 n, int version, int rle,
+		    int32 max_block_size, uint8 *out, int64 out_cap,
+		    int64 *out_len, int32 *out_nblocks)
+{
+	DatumStreamTypeInfo ti;
+	DatumStreamBlockWrite dsw;
+	int64		pos = 0;
+	int32		nblocks = 0;
+	int64		i = 0;
+	static uint8 vbuf[1 << 20];
+
+	memset(&dsw, 0, sizeof(dsw));
+	typeinfo_text(&ti);
+	DatumStreamBlockWrite_Init(&dsw, &ti, (DatumStreamVersion) version,
+				   rle != 0, false /* no delta for text */ ,
+				   MAXDATUM_ORIG,
+				   (version == 0) ? MAXDATUM_ORIG
+				   : MAXDATUM_DENSE,
+				   max_block_size, NULL, NULL, NULL, NULL);
+	DatumStreamBlockWrite_GetReady(&dsw);
+
+	while (i < n)
+	{
+		Datum		d = 0;
+		bool		isnull = nulls && nulls[i];
+		void	   *tofree = NULL;
+		int			res;
+		int64		len = offs[i + 1] - offs[i];
+
+		if (!isnull)
+		{
+			if (len + 4 > (int64) sizeof(vbuf))
+				return -3;
+			/* standard 4-byte-header varlena; the writer itself
+			 * converts short candidates (datumstreamblock.c:1646) */
+			SET_VARSIZE(vbuf, len + 4);
+			memcpy(vbuf + 4, bytes + offs[i], len);
+			d = PointerGetDatum(vbuf);
+		}
+		res = DatumStreamBlockWrite_Put(&dsw, d, isnull, &tofree);
+		if (res >= 0)
+		{
+			i++;
+			continue;
+		}
+		{
+			int32		rowcount = DatumStreamBlockWrite_Nth(&dsw);
+			int64		sz;
+
+			if (rowcount == 0)
+				return -4;	/* datum larger than a block */
+			if (pos + 8 + max_block_size > out_cap)
+				return -1;
+			sz = DatumStreamBlockWrite_Block(&dsw, out + pos + 8);
+			memcpy(out + pos, &sz, 4);
+			memcpy(out + pos + 4, &rowcount, 4);
+			pos += 8 + sz;
+			/* keep every block 8-aligned, as AO storage does
+			 * (AOStorage_RoundUp8) — the reference's varlena walk
+			 * is absolute-pointer-aligned */
+			while (pos & 7)
+				out[pos++] = 0;
+			nblocks++;
+			DatumStreamBlockWrite_GetReady(&dsw);
+		}
+	}
+	if (DatumStreamBlockWrite_Nth(&dsw) > 0)
+	{
+		int32		rowcount = DatumStreamBlockWrite_Nth(&dsw);
+		int64		sz;
+
+		if (pos + 8 + max_block_size > out_cap)
+			return -1;
+		sz = DatumStreamBlockWrite_Block(&dsw, out + pos + 8);
+		memcpy(out + pos, &sz, 4);
+		memcpy(out + pos + 4, &rowcount, 4);
+		pos += 8 + sz;
+		while (pos & 7)
+			out[pos++] = 0;
+		nblocks++;
+	}
+	DatumStreamBlockWrite_Finish(&dsw);
+	*out_len = pos;
+	*out_nblocks = nblocks;
+	return 0;
+}
+
+/* Decode text back with the REFERENCE reader: writes payload bytes to
+ * out_bytes, row offsets to out_offs[0..n], null flags to out_nulls. */
+int
+ref_dsb_decode_text(const uint8 *stream, int64 stream_len, int version,
+		    int rle, uint8 *out_bytes, int64 bytes_cap,
+		    int64 *out_offs, uint8 *out_nulls, int64 cap,
+		    int64 *out_n)
+{
+	DatumStreamTypeInfo ti;
+	DatumStreamBlockRead dsr;
+	int64		pos = 0;
+	int64		nout = 0;
+	int64		bpos = 0;
+
+	memset(&dsr, 0, sizeof(dsr));
+	typeinfo_text(&ti);
+	DatumStreamBlockRead_Init(&dsr, &ti, (DatumStreamVersion) version,
+				  rle != 0, NULL, NULL, NULL, NULL);
+
+	out_offs[0] = 0;
+	while (pos < stream_len)
+	{
+		int32		sz,
+					rowcount;
+		bool		adjusted = false;
+		int32		adjustedRowCount = 0;
+		int			have;
+
+		memcpy(&sz, stream + pos, 4);
+		memcpy(&rowcount, stream + pos + 4, 4);
+		pos += 8;
+		DatumStreamBlockRead_Reset(&dsr);
+		DatumStreamBlockRead_GetReady(&dsr, (uint8 *) stream + pos, sz,
+					      1, rowcount, &adjusted,
+					      &adjustedRowCount);
+		pos += sz;
+		pos = (pos + 7) & ~(int64) 7;
+		for (have = 0; have < rowcount; have++)
+		{
+			Datum		d = 0;
+			bool		isnull = false;
+
+			if (DatumStreamBlockRead_Advance(&dsr) == 0)
+				return -2;
+			DatumStreamBlockRead_Get(&dsr, &d, &isnull);
+			if (nout >= cap)
+				return -1;
+			out_nulls[nout] = isnull ? 1 : 0;
+			if (!isnull)
+			{
+				char	   *vp = (char *) DatumGetPointer(d);
+				int64		len = REF_VARSIZE_ANY_EXHDR(vp);
+
+				if (bpos + len > bytes_cap)
+					return -1;
+				memcpy(out_bytes + bpos, REF_VARDATA_ANY(vp),
+				       len);
+				bpos += len;
+			}
+			nout++;
+			out_offs[nout] = bpos;
 		}
 	}
 	DatumStreamBlockRead_Finish(&dsr);
@@ -283,6 +498,7 @@ ref_ao_wrap_stream(const uint8 *framed, int64 framed_len, int checksums,
 				rowcount, sz);
 		firstRowNum += rowcount;
 		pos += sz;
+		pos = (pos + 7) & ~(int64) 7;
 		opos += overall;
 	}
 	*out_len = opos;
@@ -382,6 +598,7 @@ ref_ao_wrap_stream_c(const uint8 *framed, int64 framed_len, int checksums,
 			rowcount, sz, complen);
 		firstRowNum += rowcount;
 		pos += sz;
+		pos = (pos + 7) & ~(int64) 7;
 		opos += overall;
 	}
 	*out_len = opos;

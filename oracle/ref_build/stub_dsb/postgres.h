@@ -123,7 +123,13 @@ static inline void pfree(void *p) { free(p); }
 #define MAXIMUM_ALIGNOF 8
 #define MAXALIGN(LEN) TYPEALIGN(MAXIMUM_ALIGNOF, (LEN))
 
-/* ---- varlena (postgres.h 9.4, little-endian 1-byte-header forms) ---- */
+/* ---- varlena — THIS FORK'S layout (reference postgres.h:160–300):
+ * 4-byte headers stored in NETWORK byte order ("Greengage stored the
+ * 4 byte varlena header in network byte order" — postgres.h:169),
+ * flag bits in the physically first byte: 00=4B uncompressed,
+ * 01=4B compressed, 10000000=toast pointer, 1xxxxxxx=short (length
+ * includes the header byte, no shift). ---- */
+#include <arpa/inet.h>
 typedef struct varlena
 {
 	char		vl_len_[4];
@@ -132,7 +138,6 @@ typedef struct varlena
 #define VARHDRSZ ((int32) sizeof(int32))
 #define VARHDRSZ_SHORT 1
 
-/* postgres.h 9.4 varattrib_4b is a union with a compressed form */
 typedef union
 {
 	struct
@@ -155,18 +160,25 @@ typedef struct
 } varattrib_1b_stub;
 typedef varattrib_1b_stub varattrib_1b;
 
-#define VARATT_IS_4B(PTR)  ((((varattrib_1b_stub *) (PTR))->va_header & 0x03) == 0x00)
-#define VARATT_IS_4B_U(PTR) ((((varattrib_1b_stub *) (PTR))->va_header & 0x03) == 0x00)
-#define VARATT_IS_4B_C(PTR) ((((varattrib_1b_stub *) (PTR))->va_header & 0x03) == 0x02)
-#define VARATT_IS_1B(PTR)  ((((varattrib_1b_stub *) (PTR))->va_header & 0x01) == 0x01)
-#define VARATT_IS_1B_E(PTR) ((((varattrib_1b_stub *) (PTR))->va_header) == 0x01)
+#define VARATT_IS_4B(PTR) \
+	((((varattrib_1b_stub *) (PTR))->va_header & 0x80) == 0x00)
+#define VARATT_IS_4B_U(PTR) \
+	((((varattrib_1b_stub *) (PTR))->va_header & 0xC0) == 0x00)
+#define VARATT_IS_4B_C(PTR) \
+	((((varattrib_1b_stub *) (PTR))->va_header & 0xC0) == 0x40)
+#define VARATT_IS_1B(PTR) \
+	((((varattrib_1b_stub *) (PTR))->va_header & 0x80) == 0x80)
+#define VARATT_IS_1B_E(PTR) \
+	((((varattrib_1b_stub *) (PTR))->va_header) == 0x80)
+#define VARATT_NOT_PAD_BYTE(PTR) (*((uint8 *) (PTR)) != 0)
 #define VARATT_IS_COMPRESSED(PTR) VARATT_IS_4B_C(PTR)
 #define VARATT_IS_EXTERNAL(PTR) VARATT_IS_1B_E(PTR)
 #define VARATT_IS_SHORT(PTR) VARATT_IS_1B(PTR)
 #define VARATT_IS_EXTENDED(PTR) (!VARATT_IS_4B_U(PTR))
 
-#define VARSIZE_4B(PTR) ((((varattrib_4b_stub *) (PTR))->va_4byte.va_header >> 2) & 0x3FFFFFFF)
-#define VARSIZE_1B(PTR) ((((varattrib_1b_stub *) (PTR))->va_header >> 1) & 0x7F)
+#define VARSIZE_4B(PTR) \
+	(ntohl(((varattrib_4b_stub *) (PTR))->va_4byte.va_header) & 0x3FFFFFFF)
+#define VARSIZE_1B(PTR) ((((varattrib_1b_stub *) (PTR))->va_header) & 0x7F)
 #define VARSIZE(PTR) VARSIZE_4B(PTR)
 #define VARSIZE_SHORT(PTR) VARSIZE_1B(PTR)
 #define VARSIZE_ANY(PTR) \
@@ -178,11 +190,16 @@ typedef varattrib_1b_stub varattrib_1b;
 #define VARDATA_ANY(PTR) \
 	(VARATT_IS_1B(PTR) ? VARDATA_1B(PTR) : VARDATA_4B(PTR))
 #define SET_VARSIZE(PTR, len) \
-	(((varattrib_4b_stub *) (PTR))->va_4byte.va_header = (((uint32) (len)) << 2))
+	(((varattrib_4b_stub *) (PTR))->va_4byte.va_header = \
+	 htonl(((uint32) (len)) & 0x3FFFFFFF))
 #define SET_VARSIZE_SHORT(PTR, len) \
-	(((varattrib_1b_stub *) (PTR))->va_header = (((uint8) (len)) << 1) | 0x01)
-#define VARSIZE_TO_SHORT(len) ((len) - VARHDRSZ + VARHDRSZ_SHORT)
-#define VARSIZE_TO_SHORT_D(D) VARSIZE_TO_SHORT(VARSIZE(DatumGetPointer(D)))
+	(((varattrib_1b_stub *) (PTR))->va_header = ((uint8) (len)) | 0x80)
+/* tuptoaster.h:22 — produces the final short HEADER BYTE */
+#define VARSIZE_TO_SHORT(PTR) \
+	((char) (VARSIZE(PTR) - VARHDRSZ + VARHDRSZ_SHORT) | 0x80)
+#define VARSIZE_TO_SHORT_D(D) VARSIZE_TO_SHORT(DatumGetPointer(D))
+#define VARATT_CONVERTED_SHORT_SIZE(PTR) \
+	(VARSIZE(PTR) - VARHDRSZ + VARHDRSZ_SHORT)
 #define VARATT_CAN_MAKE_SHORT(PTR) \
 	(VARATT_IS_4B_U(PTR) && \
 	 (VARSIZE(PTR) - VARHDRSZ + VARHDRSZ_SHORT) <= 0x7F)
@@ -227,7 +244,7 @@ psprintf(const char *fmt, ...)
 
 /* external (toast-pointer) sizes: the wrapper never feeds external
  * datums; keep the macros compilable, trap if ever reached */
-#define VARHDRSZ_EXTERNAL 2
+#define VARHDRSZ_EXTERNAL 4	/* "In GPDB, it's 4, due to padding" */
 #define VARSIZE_EXTERNAL(PTR) (abort(), 0)
 #define VARSIZE_ANY_EXHDR(PTR) \
 	(VARATT_IS_1B_E(PTR) ? (abort(), 0) : \

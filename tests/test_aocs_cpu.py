@@ -110,7 +110,7 @@ def test_ao_wrap_smallcontent_layout():
         assert (fsz, frc) == (datalen, rowcount)
         assert np.array_equal(ao[pos + 24:pos + 24 + datalen],
                               framed[frame_pos + 8:frame_pos + 8 + fsz])
-        frame_pos += 8 + fsz
+        frame_pos = (frame_pos + 8 + fsz + 7) & ~7  # frames are 8-aligned
         pos += overall
         blocks += 1
     assert blocks >= 2
