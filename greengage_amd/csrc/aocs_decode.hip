@@ -338,6 +338,228 @@ decode_one_block(const uint8_t *buf, int32_t size, int32_t rowcount,
 	return 0;
 }
 
+/* ---- varlena (text) decode ----------------------------------------
+ * Datum walk per the reference reader (AdvanceOrig :1517-1537 /
+ * AdvanceDense :1965-1985): VARSIZE_ANY advance, then skip zero
+ * padding to the 'i' alignment.  THIS FORK's varlena layout
+ * (postgres.h:160-230): short form = first byte & 0x80, length
+ * (incl. the header byte) in the low 7 bits; 4-byte form = header in
+ * NETWORK byte order, flags in the physically first byte (00=plain,
+ * 01=compressed, exactly 0x80 = toast pointer).  Block bases are
+ * 8-aligned (AOStorage_RoundUp8), so offset-relative alignment is
+ * absolute alignment. */
+
+__device__ inline uint32_t
+bswap32(uint32_t v)
+{
+	return __builtin_bswap32(v);
+}
+
+__device__ int
+decode_one_block_text(const uint8_t *buf, int32_t size, int32_t rowcount,
+		      int version, uint8_t *pool, int64_t pool_base,
+		      unsigned long long *out_offs, uint32_t *out_lens,
+		      uint8_t *out_nulls, int64_t out_base)
+{
+	const uint8_t *p = buf;
+	int16_t flags;
+	int32_t logical_rows, physical_size;
+	BitRd nullbm{}, rlebm{};
+	bool has_null = false, has_rle = false;
+	const uint8_t *repeatp = nullptr;
+
+	if (version == 0)
+	{
+		int16_t ver, ndatum;
+		int32_t nullsz;
+
+		memcpy(&ver, p, 2);
+		memcpy(&flags, p + 2, 2);
+		memcpy(&ndatum, p + 4, 2);
+		memcpy(&nullsz, p + 8, 4);
+		memcpy(&physical_size, p + 12, 4);
+		if (ver != 0)
+			return 1;
+		logical_rows = ndatum;
+		p += 16;
+		has_null = (flags & 0x1) != 0;
+		if (has_null)
+		{
+			nullbm.init(p);
+			p += nullsz;
+		}
+	}
+	else
+	{
+		int16_t ver;
+		int32_t physical_count;
+		int32_t null_bits = 0, rle_bits = 0, rle_cnt_size = 0;
+
+		memcpy(&ver, p, 2);
+		memcpy(&flags, p + 2, 2);
+		memcpy(&logical_rows, p + 4, 4);
+		memcpy(&physical_count, p + 8, 4);
+		memcpy(&physical_size, p + 12, 4);
+		if (ver != version)
+			return 1;
+		p += 16;
+		has_rle = (flags & 0x2) != 0;
+		has_null = (flags & 0x1) != 0;
+		if (flags & 0x4)
+			return 4;	/* delta: integers only */
+		if (has_rle)
+		{
+			int32_t nb_cnt;
+
+			memcpy(&nb_cnt, p, 4);
+			memcpy(&rle_bits, p + 4, 4);
+			memcpy(&rle_cnt_size, p + 12, 4);
+			null_bits = nb_cnt;
+			p += 16;
+		}
+		if (has_null)
+		{
+			if (!has_rle)
+				null_bits = logical_rows;
+			nullbm.init(p);
+			p += (null_bits + 7) >> 3;
+		}
+		if (has_rle)
+		{
+			rlebm.init(p);
+			p += (rle_bits + 7) >> 3;
+			repeatp = p;
+			p += rle_cnt_size;
+		}
+	}
+	if (logical_rows != rowcount)
+		return 2;
+
+	{
+		uint64_t hdr = (uint64_t) (p - buf);
+		const uint8_t *datump = buf + GG_DSB_MAXALIGN(hdr);
+		const uint8_t *datum_after = datump + physical_size;
+		int64_t ppos = 0;	/* offset within datum area */
+		int32_t rle_remaining = 0;
+		bool in_repeat = false;
+		bool first = true;
+		unsigned long long cur_off = 0;
+		uint32_t cur_len = 0;
+		int64_t wpos = 0;	/* pool write cursor (block-local) */
+
+		for (int32_t nth = 0; nth < rowcount; nth++)
+		{
+			if (in_repeat)
+			{
+				if (--rle_remaining <= 0)
+					in_repeat = false;
+				out_nulls[out_base + nth] = 0;
+				out_offs[out_base + nth] = cur_off;
+				out_lens[out_base + nth] = cur_len;
+				continue;
+			}
+			if (has_null)
+			{
+				nullbm.next();
+				if (nullbm.on())
+				{
+					out_nulls[out_base + nth] = 1;
+					out_offs[out_base + nth] = 0;
+					out_lens[out_base + nth] = 0;
+					continue;
+				}
+			}
+			if (has_rle)
+			{
+				rlebm.next();
+				if (rlebm.on())
+				{
+					rle_remaining = varint_rle(repeatp);
+					in_repeat = true;
+				}
+			}
+			/* skip zero padding BEFORE this datum (except the
+			 * very first, which starts the area) */
+			if (!first && datump + ppos < datum_after &&
+			    datump[ppos] == 0)
+				ppos = (ppos + 3) & ~(int64_t) 3;
+			first = false;
+			if (datump + ppos >= datum_after)
+				return 3;
+			{
+				uint8_t b0 = datump[ppos];
+				uint32_t vlen, paylen;
+				int64_t payoff;
+
+				if (b0 & 0x80)
+				{
+					if (b0 == 0x80)
+						return 5;	/* toast ptr */
+					vlen = b0 & 0x7F;
+					paylen = vlen - 1;
+					payoff = ppos + 1;
+				}
+				else
+				{
+					uint32_t h;
+
+					if ((b0 & 0xC0) == 0x40)
+						return 6;	/* compressed */
+					memcpy(&h, datump + ppos, 4);
+					vlen = bswap32(h) & 0x3FFFFFFFu;
+					if (vlen < 4)
+						return 7;
+					paylen = vlen - 4;
+					payoff = ppos + 4;
+				}
+				if (datump + ppos + vlen > datum_after)
+					return 3;
+				for (uint32_t z = 0; z < paylen; z++)
+					pool[pool_base + wpos + z] =
+						datump[payoff + z];
+				cur_off = (unsigned long long)
+					(pool_base + wpos);
+				cur_len = paylen;
+				wpos += paylen;
+				ppos += vlen;
+				out_nulls[out_base + nth] = 0;
+				out_offs[out_base + nth] = cur_off;
+				out_lens[out_base + nth] = cur_len;
+			}
+		}
+	}
+	return 0;
+}
+
+__global__ void
+k_dsb_decode_text(const uint8_t *__restrict__ stream,
+		  const int64_t *__restrict__ offsets,
+		  const int32_t *__restrict__ sizes,
+		  const int32_t *__restrict__ rowcounts,
+		  const int64_t *__restrict__ out_offsets,
+		  const int64_t *__restrict__ pool_offsets, int32_t nblocks,
+		  int version, uint8_t *__restrict__ pool,
+		  unsigned long long *__restrict__ out_offs,
+		  uint32_t *__restrict__ out_lens,
+		  uint8_t *__restrict__ out_nulls,
+		  unsigned long long *__restrict__ err)
+{
+	int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t b = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     b < nblocks; b += stride)
+	{
+		int rc = decode_one_block_text(stream + offsets[b], sizes[b],
+					       rowcounts[b], version, pool,
+					       pool_offsets[b], out_offs,
+					       out_lens, out_nulls,
+					       out_offsets[b]);
+
+		if (rc)
+			atomicOr(err, 1ull << rc);
+	}
+}
+
 __global__ void
 k_dsb_decode(const uint8_t *__restrict__ stream,
 	     const int64_t *__restrict__ offsets,
@@ -382,6 +604,28 @@ launch_dsb_decode(hipStream_t s, const uint8_t *stream,
 			   stream, offsets, sizes, rowcounts, out_offsets,
 			   nblocks, version, datumlen, out_vals, out_nulls,
 			   out_width, err);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
+		       const int64_t *offsets, const int32_t *sizes,
+		       const int32_t *rowcounts, const int64_t *out_offsets,
+		       const int64_t *pool_offsets, int32_t nblocks,
+		       int version, uint8_t *pool,
+		       unsigned long long *out_offs, uint32_t *out_lens,
+		       uint8_t *out_nulls, unsigned long long *err)
+{
+	int blocks = (nblocks + 255) / 256;
+
+	if (blocks > 2048)
+		blocks = 2048;
+	if (blocks < 1)
+		blocks = 1;
+	hipLaunchKernelGGL(k_dsb_decode_text, dim3(blocks), dim3(256), 0, s,
+			   stream, offsets, sizes, rowcounts, out_offsets,
+			   pool_offsets, nblocks, version, pool, out_offs,
+			   out_lens, out_nulls, err);
 	return hipGetLastError();
 }
 

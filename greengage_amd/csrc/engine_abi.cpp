@@ -2520,6 +2520,146 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 				     out_nulls, cap, out_nrows);
 }
 
+/*
+ * Decode a framed stream of TEXT (varlena) datum-stream blocks on the
+ * GPU: per-row (pool offset, byte length, null flag) plus a byte pool.
+ * Rows of one block land in a contiguous pool region; regions are
+ * upper-bounded by the block content size, so pool_cap = stream_len is
+ * always enough (gaps between regions are unreferenced).
+ */
+extern "C" gg_status
+gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
+			   int version, uint64_t *out_offs,
+			   uint32_t *out_lens, uint8_t *out_nulls,
+			   int64_t cap, uint8_t *pool, int64_t pool_cap,
+			   int64_t *out_nrows, int64_t *out_pool_len)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!stream || stream_len < 0 || version < 0 || version > 2 ||
+	    !out_offs || !out_lens || !out_nulls || !pool || !out_nrows ||
+	    !out_pool_len)
+		return fail(GG_EINVAL, "bad aocs_decode_text args");
+
+	std::vector<int64_t> offs, out_off, pool_off;
+	std::vector<int32_t> sizes, rows;
+	int64_t pos = 0, total_rows = 0, pool_len = 0;
+
+	while (pos < stream_len)
+	{
+		int32_t sz, rc;
+
+		if (pos + 8 > stream_len)
+			return fail(GG_EINVAL, "truncated frame header");
+		std::memcpy(&sz, stream + pos, 4);
+		std::memcpy(&rc, stream + pos + 4, 4);
+		if (sz < 16 || pos + 8 + sz > stream_len || rc < 0)
+			return fail(GG_EINVAL, "bad frame at %lld",
+				    (long long) pos);
+		offs.push_back(pos + 8);
+		sizes.push_back(sz);
+		rows.push_back(rc);
+		out_off.push_back(total_rows);
+		pool_off.push_back(pool_len);
+		total_rows += rc;
+		pool_len += sz;	/* payload bytes <= content size */
+		pos = (pos + 8 + sz + 7) & ~(int64_t) 7;
+	}
+	if (total_rows > cap)
+		return fail(GG_EINVAL, "cap %lld < rows %lld",
+			    (long long) cap, (long long) total_rows);
+	if (pool_len > pool_cap)
+		return fail(GG_EINVAL, "pool_cap %lld < %lld",
+			    (long long) pool_cap, (long long) pool_len);
+	*out_nrows = total_rows;
+	*out_pool_len = pool_len;
+	if (total_rows == 0)
+		return GG_OK;
+
+	gg_status st = GG_OK;
+	uint8_t *d_stream = nullptr, *d_pool = nullptr, *d_nulls = nullptr;
+	int64_t *d_offs = nullptr, *d_oo = nullptr, *d_po = nullptr;
+	int32_t *d_sizes = nullptr, *d_rows = nullptr;
+	unsigned long long *d_out_offs = nullptr;
+	uint32_t *d_lens = nullptr;
+	unsigned long long *d_err = nullptr;
+	size_t nb = offs.size();
+
+	GG_HIP(hipMalloc((void **) &d_stream, (size_t) stream_len));
+	GG_HIP(hipMalloc((void **) &d_pool, (size_t) pool_len));
+	GG_HIP(hipMalloc((void **) &d_offs, nb * 8));
+	GG_HIP(hipMalloc((void **) &d_oo, nb * 8));
+	GG_HIP(hipMalloc((void **) &d_po, nb * 8));
+	GG_HIP(hipMalloc((void **) &d_sizes, nb * 4));
+	GG_HIP(hipMalloc((void **) &d_rows, nb * 4));
+	GG_HIP(hipMalloc((void **) &d_out_offs, (size_t) total_rows * 8));
+	GG_HIP(hipMalloc((void **) &d_lens, (size_t) total_rows * 4));
+	GG_HIP(hipMalloc((void **) &d_nulls, (size_t) total_rows));
+	GG_HIP(hipMalloc((void **) &d_err, 8));
+	GG_HIP(hipMemcpy(d_stream, stream, (size_t) stream_len,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_offs, offs.data(), nb * 8,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_oo, out_off.data(), nb * 8,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_po, pool_off.data(), nb * 8,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_sizes, sizes.data(), nb * 4,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_rows, rows.data(), nb * 4,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemset(d_err, 0, 8));
+	{
+		hipError_t he = launch_dsb_decode_text(
+			e.stream, d_stream, d_offs, d_sizes, d_rows, d_oo,
+			d_po, (int32_t) nb, version, d_pool, d_out_offs,
+			d_lens, d_nulls, d_err);
+
+		if (he != hipSuccess)
+			st = fail(GG_EGPU, "dsb_decode_text: %s",
+				  hipGetErrorString(he));
+	}
+	if (st == GG_OK)
+	{
+		unsigned long long herr = 0;
+
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(&herr, d_err, 8, hipMemcpyDeviceToHost));
+		if (herr)
+			st = fail(GG_EINVAL,
+				  "text block decode error mask 0x%llx",
+				  herr);
+		else
+		{
+			GG_HIP(hipMemcpy(out_offs, d_out_offs,
+					 (size_t) total_rows * 8,
+					 hipMemcpyDeviceToHost));
+			GG_HIP(hipMemcpy(out_lens, d_lens,
+					 (size_t) total_rows * 4,
+					 hipMemcpyDeviceToHost));
+			GG_HIP(hipMemcpy(out_nulls, d_nulls,
+					 (size_t) total_rows,
+					 hipMemcpyDeviceToHost));
+			GG_HIP(hipMemcpy(pool, d_pool, (size_t) pool_len,
+					 hipMemcpyDeviceToHost));
+		}
+	}
+	(void) hipFree(d_stream);
+	(void) hipFree(d_pool);
+	(void) hipFree(d_offs);
+	(void) hipFree(d_oo);
+	(void) hipFree(d_po);
+	(void) hipFree(d_sizes);
+	(void) hipFree(d_rows);
+	(void) hipFree(d_out_offs);
+	(void) hipFree(d_lens);
+	(void) hipFree(d_nulls);
+	(void) hipFree(d_err);
+	return st;
+}
+
 /* -------------- MemTuple codec (ABI surface, §8(f)3) -------------- */
 
 /*

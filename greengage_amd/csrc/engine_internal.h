@@ -472,6 +472,17 @@ hipError_t launch_dn_q3_collect(hipStream_t s,
 				gg_q3_result_row *out,
 				unsigned long long *out_count, uint64_t cap);
 
+hipError_t launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
+				  const int64_t *offsets,
+				  const int32_t *sizes,
+				  const int32_t *rowcounts,
+				  const int64_t *out_offsets,
+				  const int64_t *pool_offsets,
+				  int32_t nblocks, int version,
+				  uint8_t *pool,
+				  unsigned long long *out_offs,
+				  uint32_t *out_lens, uint8_t *out_nulls,
+				  unsigned long long *err);
 hipError_t launch_dsb_decode(hipStream_t s, const uint8_t *stream,
 			     const int64_t *offsets, const int32_t *sizes,
 			     const int32_t *rowcounts,

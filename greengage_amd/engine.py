@@ -112,6 +112,12 @@ def _load():
     lib.gg_engine_aocs_decode.argtypes = [
         ctypes.c_void_p, I64, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
         ctypes.c_int, ctypes.c_void_p, I64, ctypes.POINTER(I64)]
+    lib.gg_engine_aocs_decode_text.restype = ctypes.c_int
+    lib.gg_engine_aocs_decode_text.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.POINTER(ctypes.c_int64),
+        ctypes.POINTER(ctypes.c_int64)]
     lib.gg_engine_memtuple_binding.restype = ctypes.c_int
     lib.gg_engine_memtuple_binding.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
@@ -355,6 +361,28 @@ class Engine:
             nulls.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)), nmax,
             ctypes.byref(n)), "aocs_decode_ao")
         return vals[:n.value], nulls[:n.value]
+
+    @staticmethod
+    def aocs_decode_text(stream, version, nmax):
+        """GPU decode of TEXT datum-stream blocks; returns
+        (list of bytes, nulls)."""
+        import numpy as np
+        stream = np.ascontiguousarray(stream, np.uint8)
+        offs = np.empty(nmax, np.uint64)
+        lens = np.empty(nmax, np.uint32)
+        nulls = np.empty(nmax, np.uint8)
+        pool = np.empty(len(stream) + 64, np.uint8)
+        n, plen = I64(), I64()
+        _check(lib().gg_engine_aocs_decode_text(
+            stream.ctypes.data_as(ctypes.c_void_p), len(stream), version,
+            offs.ctypes.data_as(ctypes.c_void_p),
+            lens.ctypes.data_as(ctypes.c_void_p),
+            nulls.ctypes.data_as(ctypes.c_void_p), nmax,
+            pool.ctypes.data_as(ctypes.c_void_p), len(pool),
+            ctypes.byref(n), ctypes.byref(plen)), "aocs_decode_text")
+        vals = [bytes(pool[offs[i]:offs[i] + lens[i]])
+                for i in range(n.value)]
+        return vals, nulls[:n.value].copy()
 
     @staticmethod
     def memtuple_binding(attlen, attalign):

@@ -214,6 +214,20 @@ gg_status gg_engine_aocs_decode(const uint8_t *stream, int64_t stream_len,
 				int out_width, uint8_t *out_nulls,
 				int64_t cap, int64_t *out_nrows);
 
+/* Decode TEXT (varlena) datum-stream blocks on the GPU: arrow-style
+ * output — per-row byte offset into `pool`, byte length, null flag.
+ * Handles this fork's short (len|0x80) and network-byte-order 4-byte
+ * varlena forms, RLE, and null bitmaps; compressed/toasted datums are
+ * rejected.  pool_cap >= stream_len always suffices. */
+gg_status gg_engine_aocs_decode_text(const uint8_t *stream,
+				     int64_t stream_len, int version,
+				     uint64_t *out_offs,
+				     uint32_t *out_lens,
+				     uint8_t *out_nulls, int64_t cap,
+				     uint8_t *pool, int64_t pool_cap,
+				     int64_t *out_nrows,
+				     int64_t *out_pool_len);
+
 /* MemTuple codec (access/common/memtuple.c format, the tuple layout
  * used in executor hash tables and on the Motion wire): bulk GPU
  * conversion between column arrays and MemTuple byte streams.

@@ -243,6 +243,15 @@ def dsb_ref():
         D.ref_ao_wrap_stream.argtypes = [
             _P_U8, I64, ctypes.c_int, ctypes.c_int, _P_U8, I64,
             ctypes.POINTER(I64)]
+        D.ref_dsb_encode_text.restype = ctypes.c_int
+        D.ref_dsb_encode_text.argtypes = [
+            _P_U8, _P_I64, _P_U8, I64, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int32, _P_U8, I64, ctypes.POINTER(I64),
+            ctypes.POINTER(ctypes.c_int32)]
+        D.ref_dsb_decode_text.restype = ctypes.c_int
+        D.ref_dsb_decode_text.argtypes = [
+            _P_U8, I64, ctypes.c_int, ctypes.c_int, _P_U8, I64, _P_I64,
+            _P_U8, I64, ctypes.POINTER(I64)]
         D.ref_ao_wrap_stream_c.restype = ctypes.c_int
         D.ref_ao_wrap_stream_c.argtypes = [
             _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
@@ -347,6 +356,57 @@ def ao_probe(block, checksums=1):
                               *[ctypes.byref(o) for o in outs])
     assert rc == 0, rc
     return tuple(o.value for o in outs)
+
+
+def dsb_encode_text(values, nulls, version, rle, blocksz=32768):
+    """Encode python byte-strings with the REFERENCE text writer;
+    returns the framed stream."""
+    D = dsb_ref()
+    assert D is not None, "reference dsb codec missing"
+    n = len(values)
+    offs = np.zeros(n + 1, np.int64)
+    for i, v in enumerate(values):
+        offs[i + 1] = offs[i] + (0 if v is None else len(v))
+    blob = b"".join(v for v in values if v is not None and len(v))
+    # offsets must still be cumulative over ALL values (None -> len 0)
+    offs = np.zeros(n + 1, np.int64)
+    pos = 0
+    parts = []
+    for i, v in enumerate(values):
+        b = b"" if v is None else v
+        parts.append(b)
+        pos += len(b)
+        offs[i + 1] = pos
+    blob = b"".join(parts)
+    bts = (np.frombuffer(blob, np.uint8).copy() if blob
+           else np.zeros(1, np.uint8))
+    nl = np.ascontiguousarray(nulls, np.uint8)
+    cap = max(1 << 21, len(blob) * 4 + n * 8)
+    out = np.zeros(cap, np.uint8)
+    olen, nb = I64(), ctypes.c_int32()
+    rc = D.ref_dsb_encode_text(bts, offs, nl, n, version, rle, blocksz,
+                               out, cap, ctypes.byref(olen),
+                               ctypes.byref(nb))
+    assert rc == 0, rc
+    return out[:olen.value].copy(), nb.value
+
+
+def dsb_decode_text(stream, version, rle, nmax):
+    """Decode text with the REFERENCE reader; returns (values, nulls)
+    where values is a list of bytes (b'' for NULL rows)."""
+    D = dsb_ref()
+    assert D is not None, "reference dsb codec missing"
+    stream = np.ascontiguousarray(stream, np.uint8)
+    bcap = len(stream) * 2 + 64
+    ob = np.zeros(bcap, np.uint8)
+    oo = np.zeros(nmax + 1, np.int64)
+    on = np.zeros(nmax, np.uint8)
+    n = I64()
+    rc = D.ref_dsb_decode_text(stream, len(stream), version, rle, ob,
+                               bcap, oo, on, nmax, ctypes.byref(n))
+    assert rc == 0, rc
+    vals = [bytes(ob[oo[i]:oo[i + 1]]) for i in range(n.value)]
+    return vals, on[:n.value].copy()
 
 
 class MtSchema:
