@@ -2722,7 +2722,8 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 		return fail(GG_EINVAL, "unsupported memtuple schema (%d)",
 			    rc);
 
-	/* per-row lengths + offsets on the host (it owns the null flags) */
+	/* per-row lengths + offsets on the host (it owns the null flags
+	 * and varlena offsets); mirrors d_mt_len */
 	std::vector<int64_t> offs(nrows + 1, 0);
 
 	for (int64_t r = 0; r < nrows; r++)
@@ -2738,7 +2739,33 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 			}
 		if (hasnull)
 			len += b.null_bitmap_extra;
-		offs[r + 1] = offs[r] + ((len + 7) & ~(int64_t) 7);
+		for (int i = 0; i < natts; i++)
+		{
+			if (!b.is_var[i] ||
+			    (nulls && nulls[i] && nulls[i][r]))
+				continue;
+			const gg_text_col *tc =
+				(const gg_text_col *) cols[i];
+			int64_t paylen = tc->offs[r + 1] - tc->offs[r];
+
+			if (paylen + 1 <= 0x7F)
+				len += paylen + 1;
+			else
+			{
+				int al = b.align_of[i];
+
+				len = (len + al - 1) & ~(int64_t) (al - 1);
+				len += 4 + paylen;
+			}
+		}
+		len = (len + 7) & ~(int64_t) 7;
+		if (len > 0xFFF0)
+			return fail(GG_EINVAL,
+				    "row %lld memtuple %lld bytes: large "
+				    "(4-byte varoffset) tuples not "
+				    "supported this round",
+				    (long long) r, (long long) len);
+		offs[r + 1] = offs[r] + len;
 	}
 	int64_t total = offs[nrows];
 
@@ -2753,8 +2780,10 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 	uint8_t *d_out = nullptr;
 	int64_t *d_offs = nullptr;
 	std::vector<void *> d_cols(natts, nullptr);
+	std::vector<int64_t *> d_voffs(natts, nullptr);
 	std::vector<uint8_t *> d_nulls(natts, nullptr);
 	void **d_colp = nullptr;
+	int64_t **d_voffp = nullptr;
 	uint8_t **d_nullp = nullptr;
 
 	GG_HIP(hipMalloc((void **) &d_out, (size_t) total));
@@ -2763,11 +2792,30 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 			 hipMemcpyHostToDevice));
 	for (int i = 0; i < natts; i++)
 	{
-		size_t nb = (size_t) nrows * attlen[i];
+		if (b.is_var[i])
+		{
+			const gg_text_col *tc =
+				(const gg_text_col *) cols[i];
+			size_t nb = (size_t) tc->offs[nrows];
 
-		GG_HIP(hipMalloc(&d_cols[i], nb));
-		GG_HIP(hipMemcpy(d_cols[i], cols[i], nb,
-				 hipMemcpyHostToDevice));
+			GG_HIP(hipMalloc(&d_cols[i], nb ? nb : 1));
+			if (nb)
+				GG_HIP(hipMemcpy(d_cols[i], tc->bytes, nb,
+						 hipMemcpyHostToDevice));
+			GG_HIP(hipMalloc((void **) &d_voffs[i],
+					 (size_t) (nrows + 1) * 8));
+			GG_HIP(hipMemcpy(d_voffs[i], tc->offs,
+					 (size_t) (nrows + 1) * 8,
+					 hipMemcpyHostToDevice));
+		}
+		else
+		{
+			size_t nb = (size_t) nrows * attlen[i];
+
+			GG_HIP(hipMalloc(&d_cols[i], nb));
+			GG_HIP(hipMemcpy(d_cols[i], cols[i], nb,
+					 hipMemcpyHostToDevice));
+		}
 		if (nulls && nulls[i])
 		{
 			GG_HIP(hipMalloc((void **) &d_nulls[i],
@@ -2778,14 +2826,18 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 		}
 	}
 	GG_HIP(hipMalloc((void **) &d_colp, natts * sizeof(void *)));
+	GG_HIP(hipMalloc((void **) &d_voffp, natts * sizeof(void *)));
 	GG_HIP(hipMalloc((void **) &d_nullp, natts * sizeof(void *)));
 	GG_HIP(hipMemcpy(d_colp, d_cols.data(), natts * sizeof(void *),
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_voffp, d_voffs.data(), natts * sizeof(void *),
 			 hipMemcpyHostToDevice));
 	GG_HIP(hipMemcpy(d_nullp, d_nulls.data(), natts * sizeof(void *),
 			 hipMemcpyHostToDevice));
 	{
 		hipError_t he = launch_mt_encode(
 			e.stream, &b, (const void *const *) d_colp,
+			(const int64_t *const *) d_voffp,
 			(const uint8_t *const *) d_nullp, nrows, d_offs,
 			d_out);
 
@@ -2804,9 +2856,11 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 	for (int i = 0; i < natts; i++)
 	{
 		(void) hipFree(d_cols[i]);
+		(void) hipFree(d_voffs[i]);
 		(void) hipFree(d_nulls[i]);
 	}
 	(void) hipFree(d_colp);
+	(void) hipFree(d_voffp);
 	(void) hipFree(d_nullp);
 	return st;
 }
@@ -2872,8 +2926,12 @@ gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 	int64_t *d_offs = nullptr;
 	unsigned long long *d_err = nullptr;
 	std::vector<void *> d_cols(natts, nullptr);
+	std::vector<unsigned long long *> d_voffs(natts, nullptr);
+	std::vector<uint32_t *> d_vlens(natts, nullptr);
 	std::vector<uint8_t *> d_nulls(natts, nullptr);
 	void **d_colp = nullptr;
+	unsigned long long **d_voffp = nullptr;
+	uint32_t **d_vlenp = nullptr;
 	uint8_t **d_nullp = nullptr;
 
 	GG_HIP(hipMalloc((void **) &d_in, (size_t) stream_len));
@@ -2886,19 +2944,36 @@ gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 	GG_HIP(hipMemset(d_err, 0, 8));
 	for (int i = 0; i < natts; i++)
 	{
-		GG_HIP(hipMalloc(&d_cols[i], (size_t) nrows * attlen[i]));
+		if (b.is_var[i])
+		{
+			GG_HIP(hipMalloc((void **) &d_voffs[i],
+					 (size_t) nrows * 8));
+			GG_HIP(hipMalloc((void **) &d_vlens[i],
+					 (size_t) nrows * 4));
+		}
+		else
+			GG_HIP(hipMalloc(&d_cols[i],
+					 (size_t) nrows * attlen[i]));
 		GG_HIP(hipMalloc((void **) &d_nulls[i], (size_t) nrows));
 	}
 	GG_HIP(hipMalloc((void **) &d_colp, natts * sizeof(void *)));
+	GG_HIP(hipMalloc((void **) &d_voffp, natts * sizeof(void *)));
+	GG_HIP(hipMalloc((void **) &d_vlenp, natts * sizeof(void *)));
 	GG_HIP(hipMalloc((void **) &d_nullp, natts * sizeof(void *)));
 	GG_HIP(hipMemcpy(d_colp, d_cols.data(), natts * sizeof(void *),
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_voffp, d_voffs.data(), natts * sizeof(void *),
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_vlenp, d_vlens.data(), natts * sizeof(void *),
 			 hipMemcpyHostToDevice));
 	GG_HIP(hipMemcpy(d_nullp, d_nulls.data(), natts * sizeof(void *),
 			 hipMemcpyHostToDevice));
 	{
 		hipError_t he = launch_mt_decode(
-			e.stream, &b, d_offs, nrows, d_in,
+			e.stream, &b, d_offs, nrows, d_in, stream_len,
 			(void *const *) d_colp,
+			(unsigned long long *const *) d_voffp,
+			(uint32_t *const *) d_vlenp,
 			(uint8_t *const *) d_nullp, d_err);
 
 		if (he != hipSuccess)
@@ -2917,9 +2992,25 @@ gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 		else
 			for (int i = 0; i < natts; i++)
 			{
-				GG_HIP(hipMemcpy(cols[i], d_cols[i],
-						 (size_t) nrows * attlen[i],
-						 hipMemcpyDeviceToHost));
+				if (b.is_var[i])
+				{
+					gg_text_out *to =
+						(gg_text_out *) cols[i];
+
+					GG_HIP(hipMemcpy(to->offs,
+							 d_voffs[i],
+							 (size_t) nrows * 8,
+							 hipMemcpyDeviceToHost));
+					GG_HIP(hipMemcpy(to->lens,
+							 d_vlens[i],
+							 (size_t) nrows * 4,
+							 hipMemcpyDeviceToHost));
+				}
+				else
+					GG_HIP(hipMemcpy(cols[i], d_cols[i],
+							 (size_t) nrows *
+							 attlen[i],
+							 hipMemcpyDeviceToHost));
 				if (nulls && nulls[i])
 					GG_HIP(hipMemcpy(nulls[i],
 							 d_nulls[i],
@@ -2933,9 +3024,13 @@ gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 	for (int i = 0; i < natts; i++)
 	{
 		(void) hipFree(d_cols[i]);
+		(void) hipFree(d_voffs[i]);
+		(void) hipFree(d_vlens[i]);
 		(void) hipFree(d_nulls[i]);
 	}
 	(void) hipFree(d_colp);
+	(void) hipFree(d_voffp);
+	(void) hipFree(d_vlenp);
 	(void) hipFree(d_nullp);
 	return st;
 }

@@ -404,22 +404,29 @@ struct MtBind
 	int32_t column_align;
 	int32_t null_bitmap_extra;
 	int32_t var_start;
+	int32_t nvar;		/* number of varlena attrs */
 	int32_t offset[GG_MT_MAX_ATTS];
-	int32_t len[GG_MT_MAX_ATTS];
+	int32_t len[GG_MT_MAX_ATTS];	/* varoffset width (2) for varlena */
 	int32_t len_aligned[GG_MT_MAX_ATTS];
 	int32_t null_byte[GG_MT_MAX_ATTS];
 	uint8_t null_mask[GG_MT_MAX_ATTS];
 	int32_t phys[GG_MT_MAX_ATTS];
+	int8_t is_var[GG_MT_MAX_ATTS];
+	int8_t align_of[GG_MT_MAX_ATTS];	/* datum alignment (bytes) */
 };
 int mt_compute_binding(int natts, const int32_t *attlen,
 		       const char *attalign, MtBind *out);
 hipError_t launch_mt_encode(hipStream_t s, const MtBind *b,
 			    const void *const *cols,
+			    const int64_t *const *var_offs,
 			    const uint8_t *const *nulls, int64_t nrows,
 			    const int64_t *offs, uint8_t *out);
 hipError_t launch_mt_decode(hipStream_t s, const MtBind *b,
 			    const int64_t *offs, int64_t nrows,
-			    const uint8_t *in, void *const *cols,
+			    const uint8_t *in, int64_t in_len,
+			    void *const *cols,
+			    unsigned long long *const *var_out_offs,
+			    uint32_t *const *var_out_lens,
 			    uint8_t *const *nulls, unsigned long long *err);
 
 hipError_t launch_dn_build_orders_q5_u8(hipStream_t s, const int64_t *okey,

@@ -40,7 +40,7 @@
 namespace gg
 {
 
-static inline uint32_t
+__host__ __device__ static inline uint32_t
 mt_align(uint32_t off, int align)
 {
 	return (off + (uint32_t) align - 1) & ~((uint32_t) align - 1);
@@ -57,11 +57,17 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 {
 	if (natts < 1 || natts > GG_MT_MAX_ATTS)
 		return 1;
+	out->nvar = 0;
 	for (int i = 0; i < natts; i++)
 	{
+		if (attlen[i] == -1)
+		{		/* varlena (text/bytea/varchar) */
+			out->nvar++;
+			continue;
+		}
 		if (attlen[i] != 1 && attlen[i] != 2 && attlen[i] != 4 &&
 		    attlen[i] != 8)
-			return 2;	/* varlena: next round */
+			return 2;	/* cstring (-2): not supported */
 		char a = attalign[i];
 
 		if (a != 'c' && a != 's' && a != 'i' && a != 'd')
@@ -81,7 +87,10 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 			(int32_t) mt_align((uint32_t) (nbytes - avail),
 					   out->column_align);
 	}
-	/* four passes by alignment, as create_col_bind does */
+	/* four passes by alignment, as create_col_bind does; varlena
+	 * takes a 2-byte varoffset slot in the 's' pass (SMALL binding —
+	 * tuples over 0xFFF0 bytes would use 4-byte offsets in the 'i'
+	 * pass; the encoder rejects those, see GG_MT_FITSHORT) */
 	uint32_t cur = (out->column_align == 8) ? 8 : 4;
 	int physical = 0;
 	int prev = -1;
@@ -91,15 +100,35 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 	{
 		for (int i = 0; i < natts; i++)
 		{
-			if (attalign[i] != pass_align[pass])
-				continue;
-			char a = attalign[i];
+			bool isvar = attlen[i] == -1;
+			char a;
+			int slot;
+
+			if (isvar)
+			{
+				if (pass != 2)
+					continue;
+				a = 's';	/* slot alignment */
+				slot = 2;
+			}
+			else
+			{
+				if (attalign[i] != pass_align[pass])
+					continue;
+				a = attalign[i];
+				slot = attlen[i];
+			}
 			int al = (a == 'd') ? 8 : (a == 'i') ? 4 :
 				(a == 's') ? 2 : 1;
 
 			cur = mt_align(cur, al);
 			out->offset[i] = (int32_t) cur;
-			out->len[i] = attlen[i];
+			out->len[i] = slot;
+			out->is_var[i] = isvar ? 1 : 0;
+			out->align_of[i] = isvar ? 4 :	/* text align 'i' */
+				((attalign[i] == 'd') ? 8 :
+				 (attalign[i] == 'i') ? 4 :
+				 (attalign[i] == 's') ? 2 : 1);
 			out->null_byte[i] = physical >> 3;
 			out->null_mask[i] =
 				(uint8_t) (1u << (physical & 7));
@@ -113,7 +142,7 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 						 al);
 			prev = i;
 			physical++;
-			cur += (uint32_t) attlen[i];
+			cur += (uint32_t) slot;
 		}
 	}
 	if (prev >= 0)		/* last attr: no extra alignment ('c') */
@@ -131,10 +160,14 @@ d_mt_align8(uint32_t x)
 	return (x + 7u) & ~7u;
 }
 
-/* tuple length for one row (compute_memtuple_size_using_bind :462) */
+/* tuple length for one row (compute_memtuple_size_using_bind :462);
+ * var_offs[i] = row-offset array (n+1 prefix) for varlena attr i, or
+ * null for fixed-width attrs.  Text payloads <= 126 bytes convert to
+ * the short form (no alignment); longer keep the 4-byte header,
+ * aligned to the attr alignment (:486-500). */
 __device__ uint32_t
-d_mt_len(const MtBind &b, const uint8_t *const *nulls, int64_t row,
-	 bool *hasnull_out)
+d_mt_len(const MtBind &b, const int64_t *const *var_offs,
+	 const uint8_t *const *nulls, int64_t row, bool *hasnull_out)
 {
 	uint32_t len = (uint32_t) b.var_start;
 	bool hasnull = false;
@@ -147,12 +180,28 @@ d_mt_len(const MtBind &b, const uint8_t *const *nulls, int64_t row,
 		}
 	if (hasnull)
 		len += (uint32_t) b.null_bitmap_extra;
+	for (int i = 0; i < b.natts; i++)
+	{
+		if (!b.is_var[i] || (nulls[i] && nulls[i][row]))
+			continue;
+		uint32_t paylen = (uint32_t)
+			(var_offs[i][row + 1] - var_offs[i][row]);
+
+		if (paylen + 1 <= 0x7F)
+			len += paylen + 1;	/* short form */
+		else
+		{
+			len = mt_align(len, (int) b.align_of[i]);
+			len += 4 + paylen;
+		}
+	}
 	*hasnull_out = hasnull;
 	return d_mt_align8(len);
 }
 
 __global__ void
 k_mt_encode(MtBind b, const void *const *__restrict__ cols,
+	    const int64_t *const *__restrict__ var_offs,
 	    const uint8_t *const *__restrict__ nulls, int64_t nrows,
 	    const int64_t *__restrict__ offs, uint8_t *__restrict__ out)
 {
@@ -162,7 +211,7 @@ k_mt_encode(MtBind b, const void *const *__restrict__ cols,
 	     r < nrows; r += stride)
 	{
 		bool hasnull;
-		uint32_t len = d_mt_len(b, nulls, r, &hasnull);
+		uint32_t len = d_mt_len(b, var_offs, nulls, r, &hasnull);
 		uint8_t *tup = out + offs[r];
 
 		for (uint32_t z = 0; z < len; z++)
@@ -183,6 +232,20 @@ k_mt_encode(MtBind b, const void *const *__restrict__ cols,
 				if (nulls[i] && nulls[i][r])
 					nullp[b.null_byte[i]] |=
 						b.null_mask[i];
+		/* varlen section cursor, relative to `start`
+		 * (form_to :679-691: varlen_start = mtup + var_start -
+		 * null_save_len, then += extra when hasnull — i.e. the
+		 * section starts right after the null-compacted fixed
+		 * area, at start + (var_start - total_null_save)) */
+		uint32_t nullsave_total = 0;
+
+		if (hasnull)
+			for (int j = 0; j < b.natts; j++)
+				if (nulls[j] && nulls[j][r])
+					nullsave_total += (uint32_t)
+						b.len_aligned[j];
+		uint32_t vcur = (uint32_t) b.var_start - nullsave_total;
+
 		for (int i = 0; i < b.natts; i++)
 		{
 			if (nulls[i] && nulls[i][r])
@@ -199,6 +262,51 @@ k_mt_encode(MtBind b, const void *const *__restrict__ cols,
 							b.len_aligned[j];
 			uint8_t *dst = start + b.offset[i] - save;
 
+			if (b.is_var[i])
+			{
+				const uint8_t *bytes =
+					(const uint8_t *) cols[i];
+				int64_t o0 = var_offs[i][r];
+				uint32_t paylen = (uint32_t)
+					(var_offs[i][r + 1] - o0);
+				uint32_t attr_len;
+
+				if (paylen + 1 <= 0x7F)
+				{	/* short form, unaligned */
+					start[vcur] = (uint8_t)
+						((paylen + 1) | 0x80);
+					for (uint32_t z = 0; z < paylen; z++)
+						start[vcur + 1 + z] =
+							bytes[o0 + z];
+					attr_len = paylen + 1;
+				}
+				else
+				{	/* 4-byte network-order header,
+					 * aligned */
+					vcur = mt_align(vcur,
+							(int) b.align_of[i]);
+					uint32_t h = (paylen + 4) &
+						0x3FFFFFFFu;
+
+					start[vcur] = (uint8_t) (h >> 24);
+					start[vcur + 1] =
+						(uint8_t) (h >> 16);
+					start[vcur + 2] =
+						(uint8_t) (h >> 8);
+					start[vcur + 3] = (uint8_t) h;
+					for (uint32_t z = 0; z < paylen; z++)
+						start[vcur + 4 + z] =
+							bytes[o0 + z];
+					attr_len = paylen + 4;
+				}
+				/* 2-byte varoffset relative to start
+				 * (form_to :794) */
+				uint16_t voff = (uint16_t) vcur;
+
+				memcpy(dst, &voff, 2);
+				vcur += attr_len;
+				continue;
+			}
 			switch (b.len[i])
 			{
 				case 1:
@@ -223,7 +331,10 @@ k_mt_encode(MtBind b, const void *const *__restrict__ cols,
 
 __global__ void
 k_mt_decode(MtBind b, const int64_t *__restrict__ offs, int64_t nrows,
-	    const uint8_t *__restrict__ in, void *const *__restrict__ cols,
+	    const uint8_t *__restrict__ in, int64_t in_len,
+	    void *const *__restrict__ cols,
+	    unsigned long long *const *__restrict__ var_out_offs,
+	    uint32_t *const *__restrict__ var_out_lens,
 	    uint8_t *const *__restrict__ nulls,
 	    unsigned long long *__restrict__ err)
 {
@@ -255,20 +366,30 @@ k_mt_decode(MtBind b, const int64_t *__restrict__ offs, int64_t nrows,
 				nulls[i][r] = isnull ? 1 : 0;
 			if (isnull)
 			{
-				switch (b.len[i])
+				if (b.is_var[i])
 				{
-					case 1:
-						((uint8_t *) cols[i])[r] = 0;
-						break;
-					case 2:
-						((int16_t *) cols[i])[r] = 0;
-						break;
-					case 4:
-						((int32_t *) cols[i])[r] = 0;
-						break;
-					default:
-						((int64_t *) cols[i])[r] = 0;
+					var_out_offs[i][r] = 0;
+					var_out_lens[i][r] = 0;
 				}
+				else
+					switch (b.len[i])
+					{
+						case 1:
+							((uint8_t *)
+							 cols[i])[r] = 0;
+							break;
+						case 2:
+							((int16_t *)
+							 cols[i])[r] = 0;
+							break;
+						case 4:
+							((int32_t *)
+							 cols[i])[r] = 0;
+							break;
+						default:
+							((int64_t *)
+							 cols[i])[r] = 0;
+					}
 				continue;
 			}
 			uint32_t save = 0;
@@ -282,6 +403,54 @@ k_mt_decode(MtBind b, const int64_t *__restrict__ offs, int64_t nrows,
 							b.len_aligned[j];
 			const uint8_t *src = start + b.offset[i] - save;
 
+			if (b.is_var[i])
+			{
+				/* 2-byte varoffset relative to start
+				 * (memtuple_get_attr_data_ptr :533) */
+				uint16_t voff;
+
+				memcpy(&voff, src, 2);
+				const uint8_t *vp = start + voff;
+				uint32_t paylen;
+				int64_t payoff;
+
+				if (*vp & 0x80)
+				{
+					if (*vp == 0x80)
+					{
+						atomicOr(err, 2ull);
+						continue;
+					}
+					paylen = (uint32_t) (*vp & 0x7F) - 1;
+					payoff = (vp + 1) - in;
+				}
+				else
+				{
+					uint32_t h = ((uint32_t) vp[0] << 24) |
+						((uint32_t) vp[1] << 16) |
+						((uint32_t) vp[2] << 8) |
+						(uint32_t) vp[3];
+
+					if ((vp[0] & 0xC0) == 0x40)
+					{
+						atomicOr(err, 4ull);
+						continue;
+					}
+					paylen = (h & 0x3FFFFFFFu) - 4;
+					payoff = (vp + 4) - in;
+				}
+				if (payoff + paylen > (int64_t) in_len)
+				{
+					atomicOr(err, 8ull);
+					continue;
+				}
+				/* pool IS the input stream: report the
+				 * payload's position in it (zero-copy) */
+				var_out_offs[i][r] =
+					(unsigned long long) payoff;
+				var_out_lens[i][r] = paylen;
+				continue;
+			}
 			switch (b.len[i])
 			{
 				case 1:
@@ -319,21 +488,26 @@ mt_grid(int64_t n)
 
 hipError_t
 launch_mt_encode(hipStream_t s, const MtBind *b, const void *const *cols,
+		 const int64_t *const *var_offs,
 		 const uint8_t *const *nulls, int64_t nrows,
 		 const int64_t *offs, uint8_t *out)
 {
 	hipLaunchKernelGGL(k_mt_encode, dim3(mt_grid(nrows)), dim3(256), 0,
-			   s, *b, cols, nulls, nrows, offs, out);
+			   s, *b, cols, var_offs, nulls, nrows, offs, out);
 	return hipGetLastError();
 }
 
 hipError_t
 launch_mt_decode(hipStream_t s, const MtBind *b, const int64_t *offs,
-		 int64_t nrows, const uint8_t *in, void *const *cols,
-		 uint8_t *const *nulls, unsigned long long *err)
+		 int64_t nrows, const uint8_t *in, int64_t in_len,
+		 void *const *cols,
+		 unsigned long long *const *var_out_offs,
+		 uint32_t *const *var_out_lens, uint8_t *const *nulls,
+		 unsigned long long *err)
 {
 	hipLaunchKernelGGL(k_mt_decode, dim3(mt_grid(nrows)), dim3(256), 0,
-			   s, *b, offs, nrows, in, cols, nulls, err);
+			   s, *b, offs, nrows, in, in_len, cols,
+			   var_out_offs, var_out_lens, nulls, err);
 	return hipGetLastError();
 }
 

@@ -400,21 +400,48 @@ class Engine:
     @staticmethod
     def memtuple_encode(attlen, attalign, cols, nulls):
         """Bulk GPU encode of column arrays into a MemTuple stream.
-        cols[i]: numpy array with itemsize attlen[i]; nulls[i]: uint8
-        array or None."""
+        cols[i]: numpy array with itemsize attlen[i], or a list of
+        python bytes for attlen -1 (text); nulls[i]: uint8 array or
+        None."""
         import numpy as np
+
+        class _TC(ctypes.Structure):
+            _fields_ = [("bytes", ctypes.c_void_p),
+                        ("offs", ctypes.c_void_p)]
+
         natts = len(attlen)
         al = (ctypes.c_int32 * natts)(*attlen)
         nrows = len(cols[0])
-        carr = [np.ascontiguousarray(c) for c in cols]
+        carr, keep = [], []
+        for l, c in zip(attlen, cols):
+            if l == -1:
+                blob = b"".join(c)
+                offs = np.zeros(nrows + 1, np.int64)
+                pos = 0
+                for i, v in enumerate(c):
+                    pos += len(v)
+                    offs[i + 1] = pos
+                bts = (np.frombuffer(blob, np.uint8).copy() if blob
+                       else np.zeros(1, np.uint8))
+                tc = _TC(bts.ctypes.data_as(ctypes.c_void_p).value,
+                         offs.ctypes.data_as(ctypes.c_void_p).value)
+                keep += [bts, offs, tc]
+                carr.append(tc)
+            else:
+                carr.append(np.ascontiguousarray(c))
         colp = (ctypes.c_void_p * natts)(
-            *[c.ctypes.data_as(ctypes.c_void_p).value for c in carr])
+            *[ctypes.addressof(c) if isinstance(c, _TC)
+              else c.ctypes.data_as(ctypes.c_void_p).value
+              for c in carr])
         narr = [None if n is None else
                 np.ascontiguousarray(n, np.uint8) for n in nulls]
         nullp = (ctypes.c_void_p * natts)(
             *[0 if n is None else
               n.ctypes.data_as(ctypes.c_void_p).value for n in narr])
-        cap = nrows * (8 + sum(8 + x for x in attlen)) + 64
+        cap = nrows * (8 + sum(8 + max(x, 2) for x in attlen)) + 64
+        for l, c in zip(attlen, cols):
+            if l == -1:
+                cap += sum(len(v) + 8 for v in c)
         out = np.zeros(cap, np.uint8)
         olen = I64()
         _check(lib().gg_engine_memtuple_encode(
@@ -425,15 +452,34 @@ class Engine:
 
     @staticmethod
     def memtuple_decode(attlen, attalign, stream, cap_rows):
+        """Decode a MemTuple stream; text attrs (attlen -1) come back
+        as lists of python bytes."""
         import numpy as np
+
+        class _TO(ctypes.Structure):
+            _fields_ = [("offs", ctypes.c_void_p),
+                        ("lens", ctypes.c_void_p)]
+
         natts = len(attlen)
         al = (ctypes.c_int32 * natts)(*attlen)
         stream = np.ascontiguousarray(stream, np.uint8)
         dt = {1: np.uint8, 2: np.int16, 4: np.int32, 8: np.int64}
-        cols = [np.zeros(cap_rows, dt[l]) for l in attlen]
+        cols, keep = [], []
+        for l in attlen:
+            if l == -1:
+                offs = np.zeros(cap_rows, np.uint64)
+                lens = np.zeros(cap_rows, np.uint32)
+                to = _TO(offs.ctypes.data_as(ctypes.c_void_p).value,
+                         lens.ctypes.data_as(ctypes.c_void_p).value)
+                keep.append((offs, lens))
+                cols.append(to)
+            else:
+                cols.append(np.zeros(cap_rows, dt[l]))
         nulls = [np.zeros(cap_rows, np.uint8) for _ in attlen]
         colp = (ctypes.c_void_p * natts)(
-            *[c.ctypes.data_as(ctypes.c_void_p).value for c in cols])
+            *[ctypes.addressof(c) if isinstance(c, _TO)
+              else c.ctypes.data_as(ctypes.c_void_p).value
+              for c in cols])
         nullp = (ctypes.c_void_p * natts)(
             *[n.ctypes.data_as(ctypes.c_void_p).value for n in nulls])
         n = I64()
@@ -441,8 +487,17 @@ class Engine:
             natts, al, "".join(attalign).encode(),
             stream.ctypes.data_as(ctypes.c_void_p), len(stream), colp,
             nullp, cap_rows, ctypes.byref(n)), "memtuple_decode")
-        return ([c[:n.value] for c in cols],
-                [nl[:n.value] for nl in nulls])
+        out, ki = [], 0
+        for l, c in zip(attlen, cols):
+            if l == -1:
+                offs, lens = keep[ki]
+                ki += 1
+                out.append([bytes(stream[int(offs[i]):
+                                         int(offs[i]) + int(lens[i])])
+                            for i in range(n.value)])
+            else:
+                out.append(c[:n.value])
+        return out, [nl[:n.value] for nl in nulls]
 
     # ---- general hash group-by (arbitrary int64 keys, SUM+COUNT) ----
     @staticmethod

@@ -231,9 +231,26 @@ gg_status gg_engine_aocs_decode_text(const uint8_t *stream,
 /* MemTuple codec (access/common/memtuple.c format, the tuple layout
  * used in executor hash tables and on the Motion wire): bulk GPU
  * conversion between column arrays and MemTuple byte streams.
- * Fixed-width by-value attrs (attlen 1/2/4/8, attalign c/s/i/d),
- * nullable; cols[i] holds nrows elements of width attlen[i], nulls[i]
- * is a byte-per-row flag array or NULL for NOT NULL columns. */
+ * Fixed-width by-value attrs (attlen 1/2/4/8, attalign c/s/i/d) and
+ * varlena text (attlen -1, attalign 'i'), nullable.
+ * For fixed attrs cols[i] holds nrows elements of width attlen[i];
+ * for varlena attrs cols[i] points at a gg_text_col on encode (bytes
+ * pool + n+1 prefix offsets) and a gg_text_out on decode (per-row
+ * offset/length pairs referencing the INPUT stream buffer —
+ * zero-copy).  nulls[i] is a byte-per-row flag array or NULL.
+ * Tuples over 0xFFF0 bytes (4-byte varoffsets) are rejected. */
+typedef struct gg_text_col
+{
+	const uint8_t *bytes;
+	const int64_t *offs;	/* nrows+1 prefix offsets */
+} gg_text_col;
+
+typedef struct gg_text_out
+{
+	uint64_t   *offs;	/* per-row payload offset into the input */
+	uint32_t   *lens;	/* per-row payload length */
+} gg_text_out;
+
 gg_status gg_engine_memtuple_binding(int natts, const int32_t *attlen,
 				     const char *attalign, int32_t *out);
 gg_status gg_engine_memtuple_encode(int natts, const int32_t *attlen,

@@ -267,6 +267,10 @@ def dsb_ref():
         D.ref_mt_getattr.argtypes = [
             ctypes.c_void_p, _P_U8, ctypes.c_int, ctypes.POINTER(I64),
             ctypes.POINTER(ctypes.c_uint8)]
+        D.ref_mt_form_var.restype = ctypes.c_int32
+        D.ref_mt_form_var.argtypes = [
+            ctypes.c_void_p, _P_I64, _P_U8, _P_U8, _P_I64, _P_U8, _P_U8,
+            ctypes.c_int32]
         D.ref_mt_get_colbind.restype = ctypes.c_int32
         D.ref_mt_get_colbind.argtypes = [
             ctypes.c_void_p, ctypes.c_int, ctypes.c_int] + \
@@ -420,9 +424,10 @@ class MtSchema:
         self.natts = len(attlen)
         self.attlen = list(attlen)
         if atttypid is None:
-            atttypid = [{8: 20, 4: 23, 2: 21, 1: 18}[l] for l in attlen]
+            atttypid = [{8: 20, 4: 23, 2: 21, 1: 18, -1: 25}[l]
+                        for l in attlen]
         al = (ctypes.c_int32 * self.natts)(*attlen)
-        bv = np.ones(self.natts, np.uint8)
+        bv = np.array([0 if l == -1 else 1 for l in attlen], np.uint8)
         ti = (ctypes.c_uint32 * self.natts)(*atttypid)
         self.bind = D.ref_mt_create_binding(
             self.natts, al, bv, "".join(attalign).encode(), ti)
@@ -443,6 +448,34 @@ class MtSchema:
                                    ctypes.byref(v), ctypes.byref(isn))
         assert rc == 0
         return v.value, isn.value
+
+    def form_var(self, values, isnull):
+        """Form a tuple with mixed fixed/text attrs: text attrs take a
+        python bytes value."""
+        is_text = np.array([1 if l == -1 else 0 for l in self.attlen],
+                           np.uint8)
+        blob_parts, offs, ints = [], [0], []
+        pos = 0
+        for l, v in zip(self.attlen, values):
+            if l == -1:
+                b = v if isinstance(v, (bytes, bytearray)) else b""
+                blob_parts.append(bytes(b))
+                ints.append(len(offs) - 1)
+                pos += len(b)
+                offs.append(pos)
+            else:
+                ints.append(int(v))
+        blob = b"".join(blob_parts)
+        bts = (np.frombuffer(blob, np.uint8).copy() if blob
+               else np.zeros(1, np.uint8))
+        offs_a = np.array(offs, np.int64)
+        v = np.array(ints, np.int64)
+        nl = np.array([int(x) for x in isnull], np.uint8)
+        out = np.zeros(1 << 16, np.uint8)
+        ln = self.D.ref_mt_form_var(self.bind, v, is_text, bts, offs_a,
+                                    nl, out, 1 << 16)
+        assert ln > 0, ln
+        return out[:ln].copy()
 
     def colbind(self, attnum, large=False):
         outs = [ctypes.c_int32() for _ in range(6)]

@@ -716,6 +716,58 @@ ref_mt_form(void *pbind, const int64 *values, const uint8 *isnull,
 	return (int32) destlen;
 }
 
+/* form with mixed fixed/text attrs: for is_text[i] != 0, values[i] is
+ * the ROW INDEX into (bytes, offs) — a 4-byte-header varlena is built
+ * and passed; the writer converts to short form when it can. */
+int32
+ref_mt_form_var(void *pbind, const int64 *values, const uint8 *is_text,
+		const uint8 *bytes, const int64 *offs, const uint8 *isnull,
+		uint8 *out, int32 cap)
+{
+	MemTupleBinding *b = (MemTupleBinding *) pbind;
+	int			n = b->tupdesc->natts;
+	Datum		dvals[64];
+	bool		dnull[64];
+	uint32		destlen = 0;
+	int			i;
+	MemTuple	res;
+	static uint8 scratch[64][1 << 16];
+
+	if (n > 64)
+		return -1;
+	for (i = 0; i < n; i++)
+	{
+		dnull[i] = isnull[i] != 0;
+		if (dnull[i])
+		{
+			dvals[i] = 0;
+			continue;
+		}
+		if (is_text[i])
+		{
+			int64		row = values[i];
+			int64		len = offs[row + 1] - offs[row];
+
+			if (len + 4 > (int64) sizeof(scratch[0]))
+				return -5;
+			SET_VARSIZE(scratch[i], len + 4);
+			memcpy(scratch[i] + 4, bytes + offs[row], len);
+			dvals[i] = PointerGetDatum(scratch[i]);
+		}
+		else
+			dvals[i] = (Datum) values[i];
+	}
+	memtuple_form_to(b, dvals, dnull, NULL, &destlen, false);
+	if ((int32) destlen > cap)
+		return -2;
+	memset(out, 0, destlen);
+	res = memtuple_form_to(b, dvals, dnull, (MemTuple) out, &destlen,
+			       false);
+	if (res == NULL)
+		return -3;
+	return (int32) destlen;
+}
+
 int32
 ref_mt_getattr(void *pbind, uint8 *tup, int attnum_1based, int64 *val,
 	       uint8 *out_isnull)

@@ -101,3 +101,61 @@ def test_gpu_decode_reference_stream(eng):
         assert np.array_equal(dnulls[i] != 0, nulls[i] != 0), i
         mask = nulls[i] == 0
         assert np.array_equal(dcols[i][mask], cols[i][mask]), i
+
+
+def test_gpu_encode_text_matches_reference_bytes(eng):
+    """Mixed fixed/text tuples: GPU bulk encode must be byte-identical
+    to the reference's memtuple_form_to with text datums (short and
+    4-byte forms, nulls)."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    from greengage_amd.engine import Engine as E
+    from test_memtuple_cpu import TEXT_SCHEMAS
+    rng = np.random.default_rng(45)
+    nrows = 400
+    for attlen, attalign in TEXT_SCHEMAS:
+        attalign = list(attalign)
+        ref = pyoracle.MtSchema(attlen, attalign)
+        for null_frac in (0.0, 0.3):
+            cols, nulls = [], []
+            for l in attlen:
+                if l == -1:
+                    vals = []
+                    for r in range(nrows):
+                        ln = int(rng.integers(0, 40)) if \
+                            rng.random() < 0.9 else \
+                            int(rng.integers(120, 400))
+                        vals.append(bytes(rng.integers(65, 91, ln)
+                                          .astype(np.uint8)))
+                    cols.append(vals)
+                else:
+                    lim = 2 ** (8 * min(l, 7) - 1)
+                    cols.append(rng.integers(
+                        -lim if l > 1 else 0, lim, nrows).astype(DT[l]))
+                nulls.append((rng.random(nrows) < null_frac)
+                             .astype(np.uint8) if null_frac else None)
+            stream = E.memtuple_encode(attlen, attalign, cols, nulls)
+            parts = []
+            for r in range(nrows):
+                vals = [c[r] if attlen[i] == -1 else int(c[r])
+                        for i, c in enumerate(cols)]
+                isnull = [0 if n is None else int(n[r]) for n in nulls]
+                parts.append(ref.form_var(vals, isnull))
+            refstream = np.concatenate(parts)
+            assert len(stream) == len(refstream), (attlen, null_frac)
+            assert np.array_equal(stream, refstream), \
+                (attlen, null_frac)
+            # and the decode round-trips
+            dcols, dnulls = E.memtuple_decode(attlen, attalign, stream,
+                                              nrows + 10)
+            for i, l in enumerate(attlen):
+                if nulls[i] is not None:
+                    assert np.array_equal(dnulls[i] != 0,
+                                          nulls[i] != 0), (attlen, i)
+                for r in range(nrows):
+                    if nulls[i] is not None and nulls[i][r]:
+                        continue
+                    if l == -1:
+                        assert dcols[i][r] == cols[i][r], (attlen, i, r)
+                    else:
+                        assert dcols[i][r] == cols[i][r], (attlen, i, r)

@@ -88,3 +88,53 @@ def test_reference_form_getattr_roundtrip():
                     mask = (1 << (8 * width)) - 1
                     assert v & mask == vals[a - 1] & mask, \
                         (attlen, trial, a)
+
+
+# ---------------- varlena (text) attrs ----------------
+
+TEXT_SCHEMAS = [
+    ([8, -1, 4], "dii"),          # int8, text, int4
+    ([-1], "i"),
+    ([-1, -1, -1], "iii"),
+    ([4, -1, 2, -1, 8], "disid"),  # mixed, reordering + 2 texts
+    ([1, -1, 1], "ici"),
+]
+
+
+def test_binding_matches_reference_text():
+    """Binding parity for schemas with varlena attrs: text binds as a
+    2-byte varoffset in the 's' pass (MTB_ByRef) — offsets, aligned
+    lengths and null bits must match the reference exactly."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    for attlen, attalign in TEXT_SCHEMAS:
+        ref = pyoracle.MtSchema(attlen, list(attalign))
+        (calign, extra, var_start), per = _engine_binding(
+            attlen, list(attalign))
+        rcalign, rextra, rvar = ref.info()
+        assert (calign, extra, var_start) == (rcalign, rextra, rvar), \
+            attlen
+        for a in range(1, len(attlen) + 1):
+            roff, rlen, rlen_al, rflag, rnb, rnm = ref.colbind(a)
+            eoff, elen, elen_al, enb, enm = per[a - 1]
+            assert (eoff, elen, elen_al, enb, enm) == \
+                (roff, rlen, rlen_al, rnb, rnm), (attlen, a)
+            if attlen[a - 1] == -1:
+                assert rflag == 3, (attlen, a)  # MTB_ByRef
+
+
+def test_reference_form_var_header():
+    """Wrapper sanity for mixed tuples: formed length/flags look right
+    and short/4B text forms appear in the varlen section."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    ref = pyoracle.MtSchema([8, -1, 4], ["d", "i", "i"])
+    tup = ref.form_var([7, b"hello", -3], [0, 0, 0])
+    hdr = int(np.frombuffer(tup[:4].tobytes(), np.uint32)[0])
+    assert hdr & 0x80000000 and (hdr & 0x3FFFFFF8) == len(tup)
+    assert b"hello" in tup.tobytes()
+    # long text -> 4-byte network-order header in the tail
+    long = b"Q" * 300
+    tup2 = ref.form_var([7, long, -3], [0, 0, 0])
+    assert long in tup2.tobytes()
+    assert len(tup2) > 300
