@@ -2370,12 +2370,10 @@ extern "C" size_t ZSTD_decompress(void *dst, size_t dstCap,
 				  const void *src, size_t srcSize);
 extern "C" unsigned ZSTD_isError(size_t code);
 
-extern "C" gg_status
-gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
-			 int checksums, int ao_version, int dsb_version,
-			 int comptype, int datumlen, void *out_vals,
-			 int out_width, uint8_t *out_nulls, int64_t cap,
-			 int64_t *out_nrows)
+static gg_status
+ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
+		   int checksums, int ao_version, int comptype,
+		   std::vector<uint8_t> &framed)
 {
 	if (!stream || stream_len < 0)
 		return fail(GG_EINVAL, "bad ao stream");
@@ -2384,7 +2382,6 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 			    "ao_version %d unsupported (need >=2, Aligned64bit)",
 			    ao_version);
 
-	std::vector<uint8_t> framed;
 	framed.reserve((size_t) stream_len);
 	int64_t pos = 0;
 	int64_t expect_rownum = -1;
@@ -2514,10 +2511,51 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 				    comptype);
 		pos = overall;
 	}
+	return GG_OK;
+}
+
+extern "C" gg_status
+gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
+			 int checksums, int ao_version, int dsb_version,
+			 int comptype, int datumlen, void *out_vals,
+			 int out_width, uint8_t *out_nulls, int64_t cap,
+			 int64_t *out_nrows)
+{
+	std::vector<uint8_t> framed;
+	gg_status st = ao_parse_to_framed(stream, stream_len, checksums,
+					  ao_version, comptype, framed);
+
+	if (st != GG_OK)
+		return st;
 	return gg_engine_aocs_decode(framed.data(),
 				     (int64_t) framed.size(), dsb_version,
 				     datumlen, out_vals, out_width,
 				     out_nulls, cap, out_nrows);
+}
+
+/* same, for TEXT columns: AO layer (headers, checksums, codecs) on the
+ * host, varlena datum-stream decode on the GPU.  Output offsets
+ * reference the framed pool (see gg_engine_aocs_decode_text). */
+extern "C" gg_status
+gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
+			      int checksums, int ao_version,
+			      int dsb_version, int comptype,
+			      uint64_t *out_offs, uint32_t *out_lens,
+			      uint8_t *out_nulls, int64_t cap,
+			      uint8_t *pool, int64_t pool_cap,
+			      int64_t *out_nrows, int64_t *out_pool_len)
+{
+	std::vector<uint8_t> framed;
+	gg_status st = ao_parse_to_framed(stream, stream_len, checksums,
+					  ao_version, comptype, framed);
+
+	if (st != GG_OK)
+		return st;
+	return gg_engine_aocs_decode_text(framed.data(),
+					  (int64_t) framed.size(),
+					  dsb_version, out_offs, out_lens,
+					  out_nulls, cap, pool, pool_cap,
+					  out_nrows, out_pool_len);
 }
 
 /*

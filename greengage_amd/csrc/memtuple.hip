@@ -60,6 +60,10 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 	out->nvar = 0;
 	for (int i = 0; i < natts; i++)
 	{
+		char a = attalign[i];
+
+		if (a != 'c' && a != 's' && a != 'i' && a != 'd')
+			return 3;
 		if (attlen[i] == -1)
 		{		/* varlena (text/bytea/varchar) */
 			out->nvar++;
@@ -68,10 +72,6 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 		if (attlen[i] != 1 && attlen[i] != 2 && attlen[i] != 4 &&
 		    attlen[i] != 8)
 			return 2;	/* cstring (-2): not supported */
-		char a = attalign[i];
-
-		if (a != 'c' && a != 's' && a != 'i' && a != 'd')
-			return 3;
 	}
 	out->natts = natts;
 	out->column_align = 4;
@@ -125,10 +125,13 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 			out->offset[i] = (int32_t) cur;
 			out->len[i] = slot;
 			out->is_var[i] = isvar ? 1 : 0;
-			out->align_of[i] = isvar ? 4 :	/* text align 'i' */
-				((attalign[i] == 'd') ? 8 :
-				 (attalign[i] == 'i') ? 4 :
-				 (attalign[i] == 's') ? 2 : 1);
+			/* the DATUM alignment in the varlen section is the
+			 * declared attalign (att_align_nominal(...,
+			 * attr->attalign), form_to :788) — 'i' for text */
+			out->align_of[i] =
+				(attalign[i] == 'd') ? 8 :
+				(attalign[i] == 'i') ? 4 :
+				(attalign[i] == 's') ? 2 : 1;
 			out->null_byte[i] = physical >> 3;
 			out->null_mask[i] =
 				(uint8_t) (1u << (physical & 7));

@@ -118,6 +118,13 @@ def _load():
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
         ctypes.c_void_p, ctypes.c_int64, ctypes.POINTER(ctypes.c_int64),
         ctypes.POINTER(ctypes.c_int64)]
+    lib.gg_engine_aocs_decode_ao_text.restype = ctypes.c_int
+    lib.gg_engine_aocs_decode_ao_text.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p,
+        ctypes.c_int64, ctypes.POINTER(ctypes.c_int64),
+        ctypes.POINTER(ctypes.c_int64)]
     lib.gg_engine_memtuple_binding.restype = ctypes.c_int
     lib.gg_engine_memtuple_binding.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
@@ -380,6 +387,30 @@ class Engine:
             nulls.ctypes.data_as(ctypes.c_void_p), nmax,
             pool.ctypes.data_as(ctypes.c_void_p), len(pool),
             ctypes.byref(n), ctypes.byref(plen)), "aocs_decode_text")
+        vals = [bytes(pool[offs[i]:offs[i] + lens[i]])
+                for i in range(n.value)]
+        return vals, nulls[:n.value].copy()
+
+    @staticmethod
+    def aocs_decode_ao_text(stream, checksums, ao_version, dsb_version,
+                            nmax, comptype=0):
+        """Full AO read path for TEXT columns: header/CRC/codec layer
+        on the host, GPU varlena decode."""
+        import numpy as np
+        stream = np.ascontiguousarray(stream, np.uint8)
+        offs = np.empty(nmax, np.uint64)
+        lens = np.empty(nmax, np.uint32)
+        nulls = np.empty(nmax, np.uint8)
+        pool = np.empty(len(stream) * 8 + (1 << 16), np.uint8)
+        n, plen = I64(), I64()
+        _check(lib().gg_engine_aocs_decode_ao_text(
+            stream.ctypes.data_as(ctypes.c_void_p), len(stream),
+            checksums, ao_version, dsb_version, comptype,
+            offs.ctypes.data_as(ctypes.c_void_p),
+            lens.ctypes.data_as(ctypes.c_void_p),
+            nulls.ctypes.data_as(ctypes.c_void_p), nmax,
+            pool.ctypes.data_as(ctypes.c_void_p), len(pool),
+            ctypes.byref(n), ctypes.byref(plen)), "aocs_decode_ao_text")
         vals = [bytes(pool[offs[i]:offs[i] + lens[i]])
                 for i in range(n.value)]
         return vals, nulls[:n.value].copy()

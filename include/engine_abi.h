@@ -276,6 +276,15 @@ gg_status gg_engine_memtuple_decode(int natts, const int32_t *attlen,
  * compresstype (0=none, 1=zlib, 2=zstd — pg_compression.c registry);
  * compressed SmallContent blocks (compressedLength!=0) are
  * decompressed host-side before GPU decode. */
+gg_status gg_engine_aocs_decode_ao_text(const uint8_t *stream,
+					int64_t stream_len, int checksums,
+					int ao_version, int dsb_version,
+					int comptype, uint64_t *out_offs,
+					uint32_t *out_lens,
+					uint8_t *out_nulls, int64_t cap,
+					uint8_t *pool, int64_t pool_cap,
+					int64_t *out_nrows,
+					int64_t *out_pool_len);
 gg_status gg_engine_aocs_decode_ao(const uint8_t *stream,
 				   int64_t stream_len, int checksums,
 				   int ao_version, int dsb_version,

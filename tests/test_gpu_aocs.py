@@ -178,3 +178,34 @@ def test_gpu_decode_text_rle_heavy(eng):
     assert len(gv) == n
     assert not gn.any()
     assert all(v == b"AUTOMOBILE" for v in gv)
+
+
+def test_gpu_decode_ao_text(eng):
+    """Text columns through the FULL AO path: reference-written real
+    segfile blocks (headers + CRC32C, optionally zlib/zstd compressed)
+    -> host AO layer -> GPU varlena decode, bit-exact."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    from greengage_amd.engine import Engine as E
+    from test_aocs_cpu import _text_corpus
+    rng = np.random.default_rng(53)
+    vals, nulls = _text_corpus(rng, 10000)
+    for version in (0, 1, 2):
+        framed, nb = pyoracle.dsb_encode_text(vals, nulls, version, 0,
+                                              blocksz=8192)
+        for comptype, level, cks in ((0, 0, 1), (0, 0, 0), (1, 6, 1),
+                                     (2, 3, 1)):
+            if comptype == 0:
+                ao = pyoracle.ao_wrap(framed, checksums=cks)
+            else:
+                ao = pyoracle.ao_wrap_compressed(framed, comptype,
+                                                 level, checksums=cks)
+            gv, gn = E.aocs_decode_ao_text(ao, cks, 2, version,
+                                           len(vals) + 10,
+                                           comptype=comptype)
+            assert len(gv) == len(vals), (version, comptype)
+            assert np.array_equal(gn != 0, nulls != 0)
+            for i in range(len(vals)):
+                if nulls[i]:
+                    continue
+                assert gv[i] == vals[i], (version, comptype, i)
