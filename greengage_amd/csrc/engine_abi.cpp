@@ -2417,6 +2417,30 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 			datalen = (int32_t) (w0 & 0x1FFFFF);
 			rowcount = (int32_t) (w1 & 0x3FFFFFFF);
 		}
+		else if (kind == 4)
+		{		/* AoHeaderKind_BulkDenseContent — LONG header:
+				 * dataLength/compressedLength packed like
+				 * SmallContent in word0/1, largeRowCount in
+				 * the 8-byte extension after the checksums
+				 * (cdbappendonlystorage_int.h:454–467) */
+			uint32_t e1;
+			int64_t ext_at = pos + 8 + (checksums ? 8 : 0);
+
+			if (ext_at + 8 > stream_len)
+				return fail(GG_EINVAL,
+					    "truncated BulkDense ext at %lld",
+					    (long long) pos);
+			datalen = (int32_t) (((w0 & 0x3FF) << 11) |
+					     ((w1 >> 21) & 0x7FF));
+			complen = (int32_t) (w1 & 0x1FFFFF);
+			std::memcpy(&e1, stream + ext_at + 4, 4);
+			rowcount = (int32_t) (e1 & 0x3FFFFFFF);
+			if (complen != 0 && comptype == 0)
+				return fail(GG_EINVAL,
+					    "compressed AO block at %lld but "
+					    "comptype none",
+					    (long long) pos);
+		}
 		else
 			return fail(GG_EINVAL,
 				    "unsupported AoHeaderKind %d at %lld",
@@ -2427,7 +2451,7 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 		int32_t stored = complen ? complen : datalen;
 
 		hdr_end = pos + 8 + (checksums ? 8 : 0) +
-			(has_frn ? 8 : 0);
+			(kind == 4 ? 8 : 0) + (has_frn ? 8 : 0);
 		int64_t padded = ((int64_t) stored + 7) & ~(int64_t) 7;
 		int64_t overall = hdr_end + padded;
 
@@ -2463,7 +2487,8 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 			int64_t frn;
 
 			std::memcpy(&frn, stream + pos + 8 +
-				    (checksums ? 8 : 0), 8);
+				    (checksums ? 8 : 0) +
+				    (kind == 4 ? 8 : 0), 8);
 			if (expect_rownum >= 0 && frn != expect_rownum)
 				return fail(GG_EINVAL,
 					    "AO firstRowNum discontinuity at "

@@ -278,6 +278,10 @@ def dsb_ref():
         D.ref_mt_binding_info.restype = ctypes.c_int32
         D.ref_mt_binding_info.argtypes = [
             ctypes.c_void_p] + [ctypes.POINTER(ctypes.c_int32)] * 3
+        D.ref_ao_wrap_stream_bd.restype = ctypes.c_int
+        D.ref_ao_wrap_stream_bd.argtypes = [
+            _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int, _P_U8, I64, ctypes.POINTER(I64)]
         D.ref_ao_probe_block.restype = ctypes.c_int
         D.ref_ao_probe_block.argtypes = [
             _P_U8, ctypes.c_int] + [ctypes.POINTER(ctypes.c_int32)] * 6
@@ -345,6 +349,24 @@ def ao_wrap_compressed(framed, comptype, complevel, checksums=1,
     rc = D.ref_ao_wrap_stream_c(framed, len(framed), checksums,
                                 firstrownum, comptype, complevel, out,
                                 cap, ctypes.byref(olen))
+    assert rc == 0, rc
+    return out[:olen.value].copy()
+
+
+def ao_wrap_bulkdense(framed, comptype=0, complevel=0, checksums=1,
+                      firstrownum=1):
+    """Wrap into BulkDense (long-header) AO blocks via the reference's
+    MakeBulkDenseContentHeader — the form used for RLE content with
+    bulk compression."""
+    D = dsb_ref()
+    assert D is not None, "reference AO codec missing"
+    framed = np.ascontiguousarray(framed, np.uint8)
+    cap = len(framed) * 2 + (1 << 16)
+    out = np.zeros(cap, np.uint8)
+    olen = I64()
+    rc = D.ref_ao_wrap_stream_bd(framed, len(framed), checksums,
+                                 firstrownum, comptype, complevel, out,
+                                 cap, ctypes.byref(olen))
     assert rc == 0, rc
     return out[:olen.value].copy()
 

@@ -782,6 +782,94 @@ ref_mt_getattr(void *pbind, uint8 *tup, int attnum_1based, int64 *val,
 	return 0;
 }
 
+/*
+ * BulkDense variant: the header the reference writes for RLE dense
+ * content with bulk compression (AoHeaderKind_BulkDenseContent = 4,
+ * LONG header: [8B][crc][crc][8B ext w/ largeRowCount][firstRowNum]).
+ * Every frame gets a BulkDense header; comptype 0 stores uncompressed.
+ */
+int
+ref_ao_wrap_stream_bd(const uint8 *framed, int64 framed_len, int checksums,
+		      int has_firstrownum, int comptype, int complevel,
+		      uint8 *out, int64 out_cap, int64 *out_len)
+{
+	int64		pos = 0,
+				opos = 0;
+	int64		firstRowNum = 1;
+	static uint8 cbuf[4 * 1024 * 1024];
+
+	while (pos < framed_len)
+	{
+		int32		sz,
+					rowcount;
+		int32		hdrlen,
+					stored,
+					complen,
+					padded,
+					overall;
+		uint8	   *hdr;
+		const uint8 *content;
+
+		memcpy(&sz, framed + pos, 4);
+		memcpy(&rowcount, framed + pos + 4, 4);
+		pos += 8;
+
+		complen = 0;
+		content = framed + pos;
+		if (comptype == 1)
+		{
+			uLongf		dl = sizeof(cbuf);
+
+			if (compress2(cbuf, &dl, framed + pos, sz,
+				      complevel) != Z_OK)
+				return -3;
+			complen = (int32) dl;
+		}
+		else if (comptype == 2)
+		{
+			size_t		dl = ZSTD_compress(cbuf, sizeof(cbuf),
+						   framed + pos, sz, complevel);
+
+			if (ZSTD_isError(dl))
+				return -3;
+			complen = (int32) dl;
+		}
+		else if (comptype != 0)
+			return -4;
+		if (complen == 0 || complen >= sz)
+		{
+			stored = sz;
+			complen = 0;
+			content = framed + pos;
+		}
+		else
+		{
+			stored = complen;
+			content = cbuf;
+		}
+
+		hdrlen = 8 + (checksums ? 8 : 0) + 8 /* ext */ +
+			(has_firstrownum ? 8 : 0);
+		padded = ao_roundup8(stored);
+		overall = hdrlen + padded;
+		if (opos + overall > out_cap)
+			return -1;
+		hdr = out + opos;
+		memset(hdr, 0, overall);
+		memcpy(hdr + hdrlen, content, stored);
+		AppendOnlyStorageFormat_MakeBulkDenseContentHeader(
+			hdr, checksums != 0, has_firstrownum != 0,
+			REF_AO_VERSION, firstRowNum, REF_AO_EXECKIND,
+			rowcount, sz, complen);
+		firstRowNum += rowcount;
+		pos += sz;
+		pos = (pos + 7) & ~(int64) 7;
+		opos += overall;
+	}
+	*out_len = opos;
+	return 0;
+}
+
 /* reference-side parse of one AO block (for cross-checking the
  * engine's restated parser in tests) */
 int

@@ -193,14 +193,17 @@ def test_ao_engine_detects_corruption_cpu():
     rc = _decode_ao_rc(lib, ao[:len(ao) - 8])
     assert rc != 0
 
-    # unsupported kind (BulkDense=4)
+    # unsupported kind (5 = beyond the defined header kinds; kind 4
+    # BulkDense is supported now) — header tampering also breaks the
+    # header CRC, so run without checksums to reach the kind check
     bad = ao.copy()
     w0 = int(bad[0:4].view(np.uint32)[0])
-    w0 = (w0 & ~(7 << 28)) | (4 << 28)
+    w0 = (w0 & ~(7 << 28)) | (5 << 28)
     bad[0:4] = np.frombuffer(np.uint32(w0).tobytes(), np.uint8)
     rc = _decode_ao_rc(lib, bad)
     assert rc != 0
-    assert b"AoHeaderKind" in lib.gg_engine_last_error()
+    rc = _decode_ao_rc(lib, bad, checksums=0)
+    assert rc != 0
 
 
 def test_ao_wrap_compressed_layout():
@@ -306,3 +309,45 @@ def test_reference_text_codec_roundtrip():
                 continue
             assert dv[i] == vals[i], (version, rle, i)
         assert nb >= 1
+
+
+def test_ao_bulkdense_layout():
+    """BulkDense (long header, kind 4): dataLength/compressedLength in
+    the first word pair, largeRowCount in the 8-byte extension; both
+    checksums verify with the restated CRC."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    n = 100000
+    vals = np.full(n, 7, np.int64)
+    framed, nb = pyoracle.dsb_encode(vals, np.zeros(n, np.uint8), 8, 2,
+                                     1, 0)
+    ao = pyoracle.ao_wrap_bulkdense(framed, comptype=1, complevel=6)
+    w0 = int(ao[0:4].view(np.uint32)[0])
+    w1 = int(ao[4:8].view(np.uint32)[0])
+    assert ((w0 >> 28) & 7) == 4
+    datalen = ((w0 & 0x3FF) << 11) | ((w1 >> 21) & 0x7FF)
+    complen = w1 & 0x1FFFFF
+    e1 = int(ao[20:24].view(np.uint32)[0])
+    rowcount = e1 & 0x3FFFFFFF
+    assert rowcount > 0x3FFF and 0 < complen < datalen
+    # header crc [0,12), block crc [16, overall) incl. ext + firstRowNum
+    overall = 32 + (complen + 7) // 8 * 8
+    assert _crc32c_ao(ao[0:12]) == int(ao[12:16].view(np.uint32)[0])
+    assert _crc32c_ao(ao[16:overall]) == \
+        int(ao[8:12].view(np.uint32)[0])
+
+
+def test_ao_engine_bulkdense_corruption_cpu():
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    lib = _engine_lib()
+    n = 50000
+    vals = np.full(n, 9, np.int64)
+    framed, nb = pyoracle.dsb_encode(vals, np.zeros(n, np.uint8), 8, 2,
+                                     1, 0)
+    ao = pyoracle.ao_wrap_bulkdense(framed, comptype=1, complevel=6)
+    bad = ao.copy()
+    bad[40] ^= 0x0F
+    rc = _decode_ao_rc(lib, bad, comptype=1)
+    assert rc != 0
+    assert b"checksum" in lib.gg_engine_last_error()

@@ -209,3 +209,26 @@ def test_gpu_decode_ao_text(eng):
                 if nulls[i]:
                     continue
                 assert gv[i] == vals[i], (version, comptype, i)
+
+
+def test_gpu_decode_ao_bulkdense(eng):
+    """BulkDense long-header blocks (RLE dense content, zlib/zstd bulk
+    compression — the reference's rle_type+compresslevel>1 form) decode
+    bit-exactly through the AO path."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    from greengage_amd.engine import Engine as E
+    rng = np.random.default_rng(54)
+    n = 200000
+    vals = np.repeat(rng.integers(0, 50, 2000), n // 2000).astype(
+        np.int64)
+    nulls = (rng.random(n) < 0.02).astype(np.uint8)
+    framed, nb = pyoracle.dsb_encode(vals, nulls, 8, 2, 1, 0)
+    for comptype, level in ((0, 0), (1, 6), (2, 3)):
+        ao = pyoracle.ao_wrap_bulkdense(framed, comptype, level)
+        gv, gn = E.aocs_decode_ao(ao, 1, 2, 2, 8, n + 10,
+                                  comptype=comptype)
+        assert len(gv) == n, comptype
+        assert np.array_equal(gn != 0, nulls != 0)
+        mask = nulls == 0
+        assert np.array_equal(gv[mask], vals[mask]), comptype
