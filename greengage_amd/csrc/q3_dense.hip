@@ -102,6 +102,35 @@ dn_bit_set(unsigned long long *bm, int64_t key)
 	atomicOr(&bm[key >> 6], 1ull << (key & 63));
 }
 
+/* Wave-aggregated bitmap OR: with near-sequential keys a whole wave
+ * lands in 1–2 bitmap words, so per-lane atomics serialize.  Combine
+ * per distinct word with ballot + an or-reduction, then one atomic by
+ * the leader.  Must be executed by ALL lanes of the wave (pass
+ * active=false for lanes with nothing to set). */
+__device__ inline void
+dn_bit_set_wave(unsigned long long *bm, int64_t key, bool active)
+{
+	unsigned long long w = active ? (unsigned long long) (key >> 6)
+		: ~0ull;
+	unsigned long long bit = active ? (1ull << (key & 63)) : 0;
+	unsigned long long alive = __ballot(active);
+	int lane = (int) (threadIdx.x & 63);
+
+	while (alive)
+	{
+		int leader = __ffsll((long long) alive) - 1;
+		unsigned long long lw = __shfl(w, leader, 64);
+		bool same = active && (w == lw);
+		unsigned long long v = same ? bit : 0;
+
+		for (int off = 32; off; off >>= 1)
+			v |= __shfl_xor(v, off, 64);
+		if (lane == leader)
+			atomicOr(&bm[lw], v);
+		alive &= ~__ballot(same);
+	}
+}
+
 __device__ inline bool
 dn_bit_test(const unsigned long long *__restrict__ bm, int64_t key)
 {
@@ -126,65 +155,82 @@ k_dn_build_orders(const int64_t *__restrict__ okey,
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long matches = 0;
+	/* round the per-lane range up so every lane of a wave executes
+	 * the same number of iterations (the wave-aggregated bitmap OR
+	 * below needs a convergent wave) */
+	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	const int64_t n_up = ((n + stride - 1) / stride) * stride;
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
+	for (int64_t i = i0; i < n_up; i += stride)
 	{
-		int32_t d = dn_ld32(&odate[i]);
-		int64_t ck = dn_ld64(&ckey[i]);
-		int64_t k = dn_ld64(&okey[i]);
-		bool ok;
+		bool ok = i < n;
+		int32_t d = 0;
+		int64_t ck = 0, k = -1;
 
-		if (d >= cutoff)
-			continue;
-		if (cust_bits)
-			ok = (ck >= 0 && ck < cust_dlen &&
-			      ((cust_bits[ck >> 6] >> (ck & 63)) & 1));
-		else
+		if (ok)
 		{
-			uint32_t h = gg_hashint8(ck);
-
-			ok = true;
-			if (cust_bloom)
+			d = dn_ld32(&odate[i]);
+			ck = dn_ld64(&ckey[i]);
+			k = dn_ld64(&okey[i]);
+			ok = d < cutoff;
+		}
+		if (ok)
+		{
+			if (cust_bits)
+				ok = (ck >= 0 && ck < cust_dlen &&
+				      ((cust_bits[ck >> 6] >> (ck & 63)) & 1));
+			else
 			{
-				uint32_t h2 = h * 0x9E3779B1u;
-				uint64_t w = (uint64_t) (h >> 6) &
-					(cust_bwords - 1);
-				unsigned long long m =
-					(1ull << (h2 & 63)) |
-					(1ull << ((h2 >> 6) & 63));
+				uint32_t h = gg_hashint8(ck);
 
-				ok = (cust_bloom[w] & m) == m;
-			}
-			if (ok)
-			{
-				uint64_t pos = (uint64_t) h & (cust_slots - 1);
-
-				for (;;)
+				if (cust_bloom)
 				{
-					unsigned long long v = cust_keys[pos];
+					uint32_t h2 = h * 0x9E3779B1u;
+					uint64_t w = (uint64_t) (h >> 6) &
+						(cust_bwords - 1);
+					unsigned long long m =
+						(1ull << (h2 & 63)) |
+						(1ull << ((h2 >> 6) & 63));
 
-					if (v == (unsigned long long) ck)
-						break;
-					if (v == 0)
+					ok = (cust_bloom[w] & m) == m;
+				}
+				if (ok)
+				{
+					uint64_t pos = (uint64_t) h &
+						(cust_slots - 1);
+
+					for (;;)
 					{
-						ok = false;
-						break;
+						unsigned long long v =
+							cust_keys[pos];
+
+						if (v ==
+						    (unsigned long long) ck)
+							break;
+						if (v == 0)
+						{
+							ok = false;
+							break;
+						}
+						pos = (pos + 1) &
+							(cust_slots - 1);
 					}
-					pos = (pos + 1) & (cust_slots - 1);
 				}
 			}
 		}
-		if (!ok || k < 0 || k >= dense_len)
-			continue;
-		matches++;
-		/* NT store: skip the read-for-ownership on lines we only
-		 * partially fill (~39% of keys pass the date filter) */
-		__builtin_nontemporal_store(
-			(unsigned long long) (uint32_t) d |
-			((unsigned long long) (uint32_t) prio[i] << 32),
-			&pay[k]);
-		dn_bit_set(bloom, k);
+		ok = ok && k >= 0 && k < dense_len;
+		if (ok)
+		{
+			matches++;
+			/* NT store: skip the read-for-ownership on lines we
+			 * only partially fill (~39% of keys pass the date
+			 * filter) */
+			__builtin_nontemporal_store(
+				(unsigned long long) (uint32_t) d |
+				((unsigned long long) (uint32_t) prio[i]
+				 << 32), &pay[k]);
+		}
+		dn_bit_set_wave(bloom, k, ok);
 	}
 	for (int off = 32; off; off >>= 1)
 		matches += __shfl_down(matches, off, 64);
