@@ -367,18 +367,6 @@ hipError_t launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 				  int64_t dense_len, unsigned long long *bloom,
 				  uint64_t bwords,
 				  unsigned long long *match_count);
-hipError_t launch_dn_build_orders_q5(hipStream_t s, const int64_t *okey,
-				     const int64_t *ckey,
-				     const int32_t *odate, int64_t n,
-				     int32_t date_lo, int32_t date_hi,
-				     DeviceHashTable cust,
-				     const uint8_t *cust_dense,
-				     int64_t cust_dlen,
-				     unsigned long long *pay,
-				     int64_t dense_len,
-				     unsigned long long *bloom,
-				     uint64_t bwords,
-				     unsigned long long *match_count);
 hipError_t launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 				   const int64_t *rowpay, int64_t n,
 				   unsigned long long *pay,
@@ -452,18 +440,6 @@ hipError_t launch_dn_probe_lineitem_q5_u8(hipStream_t s,
 					  int64_t supp_dlen,
 					  unsigned long long *acc,
 					  unsigned long long *join_rows);
-hipError_t launch_dn_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
-				       const int64_t *skey,
-				       const int64_t *price,
-				       const int64_t *disc, int64_t n,
-				       unsigned long long *pay,
-				       int64_t dense_len,
-				       unsigned long long *bloom,
-				       uint64_t bwords,
-				       const uint8_t *supp_dense,
-				       int64_t supp_dlen,
-				       unsigned long long *acc,
-				       unsigned long long *join_rows);
 hipError_t launch_dn_q3_stats(hipStream_t s, const unsigned long long *pay,
 			      const unsigned long long *rev,
 			      int64_t dense_len, unsigned long long *out5);

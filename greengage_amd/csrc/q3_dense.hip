@@ -256,91 +256,6 @@ launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-/* Q5 orders build: date range + customer nation → pay = nation */
-__global__ void
-k_dn_build_orders_q5(const int64_t *__restrict__ okey,
-		     const int64_t *__restrict__ ckey,
-		     const int32_t *__restrict__ odate, int64_t n,
-		     int32_t date_lo, int32_t date_hi,
-		     const unsigned long long *__restrict__ cust_keys,
-		     const unsigned long long *__restrict__ cust_pay,
-		     uint64_t cust_slots,
-		     const uint8_t *__restrict__ cust_dense,
-		     int64_t cust_dlen, unsigned long long *__restrict__ pay,
-		     int64_t dense_len, unsigned long long *__restrict__ bloom,
-		     uint64_t bwords, unsigned long long *match_count)
-{
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-	unsigned long long matches = 0;
-
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
-	{
-		int32_t d = dn_ld32(&odate[i]);
-		int64_t ck = dn_ld64(&ckey[i]);
-		int64_t k = dn_ld64(&okey[i]);
-		unsigned long long nat = 0;
-		bool ok;
-
-		if (d < date_lo || d >= date_hi)
-			continue;
-		if (cust_dense)
-		{
-			ok = (ck >= 0 && ck < cust_dlen &&
-			      cust_dense[ck] != 255);
-			if (ok)
-				nat = cust_dense[ck];
-		}
-		else
-		{
-			uint32_t h = gg_hashint8(ck);
-			uint64_t pos = (uint64_t) h & (cust_slots - 1);
-
-			ok = false;
-			for (;;)
-			{
-				unsigned long long v = cust_keys[pos];
-
-				if (v == (unsigned long long) ck)
-				{
-					nat = cust_pay[pos];
-					ok = true;
-					break;
-				}
-				if (v == 0)
-					break;
-				pos = (pos + 1) & (cust_slots - 1);
-			}
-		}
-		if (!ok || k < 0 || k >= dense_len)
-			continue;
-		matches++;
-		pay[k] = nat;
-		dn_bloom_insert(bloom, bwords, k);
-	}
-	for (int off = 32; off; off >>= 1)
-		matches += __shfl_down(matches, off, 64);
-	if ((threadIdx.x & 63) == 0 && matches)
-		atomicAdd(match_count, matches);
-}
-
-hipError_t
-launch_dn_build_orders_q5(hipStream_t s, const int64_t *okey,
-			  const int64_t *ckey, const int32_t *odate,
-			  int64_t n, int32_t date_lo, int32_t date_hi,
-			  DeviceHashTable cust, const uint8_t *cust_dense,
-			  int64_t cust_dlen, unsigned long long *pay,
-			  int64_t dense_len, unsigned long long *bloom,
-			  uint64_t bwords, unsigned long long *match_count)
-{
-	hipLaunchKernelGGL(k_dn_build_orders_q5, dim3(dn_grid(n)),
-			   dim3(DN_THREADS), 0, s, okey, ckey, odate, n,
-			   date_lo, date_hi, cust.keys, cust.payload,
-			   cust.nslots, cust_dense, cust_dlen, pay, dense_len,
-			   bloom, bwords, match_count);
-	return hipGetLastError();
-}
-
 /* exchange-path insert of received (okey, pay) rows */
 __global__ void
 k_dn_insert_orders(const int64_t *__restrict__ okey,
@@ -424,97 +339,6 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			   dim3(DN_THREADS), 0, s, okey, shipdate, price,
 			   disc, n, cutoff, pay, rev, dense_len, bloom,
 			   bwords, join_rows);
-	return hipGetLastError();
-}
-
-/* Q5 lineitem probe: dense/hashed supplier + dense orders map */
-__global__ __launch_bounds__(DN_THREADS, 2)
-void k_dn_probe_lineitem_q5(const int64_t *__restrict__ okey,
-			    const int64_t *__restrict__ skey,
-			    const int64_t *__restrict__ price,
-			    const int64_t *__restrict__ disc, int64_t n,
-			    const unsigned long long *__restrict__ pay,
-			    int64_t dense_len,
-			    const unsigned long long *__restrict__ bloom,
-			    uint64_t bwords,
-			    const uint8_t *__restrict__ supp_dense,
-			    int64_t supp_dlen,
-			    unsigned long long *__restrict__ acc /* [25][3] */,
-			    unsigned long long *join_rows)
-{
-	__shared__ unsigned long long lds[25][2];
-
-	for (int i = threadIdx.x; i < 50; i += blockDim.x)
-		((unsigned long long *) lds)[i] = 0;
-	__syncthreads();
-
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-	unsigned long long joined = 0;
-
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
-	{
-		int64_t sk = dn_ld64(&skey[i]);
-		int64_t k = dn_ld64(&okey[i]);
-		unsigned long long snat, onat;
-
-		if (sk < 0 || sk >= supp_dlen)
-			continue;
-		snat = supp_dense[sk];
-		if (snat == 255)
-			continue;
-		if (k < 0 || k >= dense_len)
-			continue;
-		if (!dn_bloom_maybe(bloom, bwords, k))
-			continue;
-		onat = pay[k];
-		if (onat == ~0ull || snat != onat)
-			continue;
-		joined++;
-		{
-			unsigned long long rev4 = (unsigned long long)
-				(price[i] * (100 - disc[i]));
-
-			atomicAdd(&lds[snat][0], 1ull);
-			atomicAdd(&lds[snat][1], rev4);
-		}
-	}
-	for (int off = 32; off; off >>= 1)
-		joined += __shfl_down(joined, off, 64);
-	if ((threadIdx.x & 63) == 0 && joined)
-		atomicAdd(join_rows, joined);
-	__syncthreads();
-	for (int nat = threadIdx.x; nat < 25; nat += blockDim.x)
-	{
-		unsigned long long c = lds[nat][0];
-		unsigned long long r = lds[nat][1];
-
-		if (!c)
-			continue;
-		atomicAdd(&acc[nat * 3 + 0], c);
-		{
-			unsigned long long old =
-				atomicAdd(&acc[nat * 3 + 1], r);
-			if (old + r < old)
-				atomicAdd(&acc[nat * 3 + 2], 1ull);
-		}
-	}
-}
-
-hipError_t
-launch_dn_probe_lineitem_q5(hipStream_t s, const int64_t *okey,
-			    const int64_t *skey, const int64_t *price,
-			    const int64_t *disc, int64_t n,
-			    unsigned long long *pay, int64_t dense_len,
-			    unsigned long long *bloom, uint64_t bwords,
-			    const uint8_t *supp_dense, int64_t supp_dlen,
-			    unsigned long long *acc,
-			    unsigned long long *join_rows)
-{
-	hipLaunchKernelGGL(k_dn_probe_lineitem_q5, dim3(dn_grid(n)),
-			   dim3(DN_THREADS), 0, s, okey, skey, price, disc,
-			   n, pay, dense_len, bloom, bwords, supp_dense,
-			   supp_dlen, acc, join_rows);
 	return hipGetLastError();
 }
 
