@@ -92,6 +92,26 @@ static gg_status read_counter(unsigned long long *p, unsigned long long *out)
 	return GG_OK;
 }
 
+/* max over an immutable registered column, cached per pipeline */
+static gg_status
+cached_max_i64(Engine &e, Pipeline *p, const int64_t *col, int64_t n,
+	       unsigned long long *ctr, unsigned long long *out)
+{
+	for (auto &kv : p->maxk_cache)
+		if (kv.first.first == (const void *) col &&
+		    kv.first.second == n)
+		{
+			*out = kv.second;
+			return GG_OK;
+		}
+	GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+	GG_HIP(launch_max_i64(e.stream, col, n, ctr));
+	GG_HIP(hipStreamSynchronize(e.stream));
+	GG_TRY(read_counter(ctr, out));
+	p->maxk_cache.push_back({{(const void *) col, n}, *out});
+	return GG_OK;
+}
+
 /* ---------------- int128 host helpers (combine/finalize) ---------------- */
 
 typedef __int128 i128;
@@ -789,10 +809,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	{
 		unsigned long long maxk = 0;
 
-		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		GG_HIP(launch_max_i64(e.stream, c_ck, cu->nrows, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_TRY(read_counter(ctr, &maxk));
+		GG_TRY(cached_max_i64(e, p, c_ck, cu->nrows, ctr, &maxk));
 		if (cu->nrows > 0 && maxk > 0 &&
 		    (int64_t) maxk <= 8 * cu->nrows + 16)
 			cust_dlen = (int64_t) maxk + 1;
@@ -868,10 +885,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	{
 		unsigned long long maxk = 0;
 
-		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		GG_HIP(launch_max_i64(e.stream, o_ok, od->nrows, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_TRY(read_counter(ctr, &maxk));
+		GG_TRY(cached_max_i64(e, p, o_ok, od->nrows, ctr, &maxk));
 		if (od->nrows > 0 && maxk > 0 &&
 		    (int64_t) maxk <= 8 * od->nrows + 16)
 			ord_dlen = (int64_t) maxk + 1;
@@ -1459,10 +1473,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 	{
 		unsigned long long maxk = 0;
 
-		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		GG_HIP(launch_max_i64(e.stream, c_ck, cu->nrows, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_TRY(read_counter(ctr, &maxk));
+		GG_TRY(cached_max_i64(e, p, c_ck, cu->nrows, ctr, &maxk));
 		if (cu->nrows > 0 && maxk > 0 &&
 		    (int64_t) maxk <= 8 * cu->nrows + 16)
 			cust_dlen = (int64_t) maxk + 1;
@@ -1519,10 +1530,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 	{
 		unsigned long long maxk = 0;
 
-		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		GG_HIP(launch_max_i64(e.stream, s_sk, su->nrows, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_TRY(read_counter(ctr, &maxk));
+		GG_TRY(cached_max_i64(e, p, s_sk, su->nrows, ctr, &maxk));
 		if (su->nrows > 0 && maxk > 0 &&
 		    (int64_t) maxk <= 8 * su->nrows + 16)
 			supp_dense_len = (int64_t) maxk + 1;
@@ -1733,10 +1741,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 	{
 		unsigned long long maxk = 0;
 
-		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		GG_HIP(launch_max_i64(e.stream, o_ok, od->nrows, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_TRY(read_counter(ctr, &maxk));
+		GG_TRY(cached_max_i64(e, p, o_ok, od->nrows, ctr, &maxk));
 		if (od->nrows > 0 && maxk > 0 &&
 		    (int64_t) maxk <= 8 * od->nrows + 16)
 			ord_dlen = (int64_t) maxk + 1;

@@ -107,6 +107,10 @@ struct Pipeline
 	uint64_t cust_slots = 0;
 	uint64_t ord_slots = 0;
 	uint64_t supp_slots = 0;
+	/* max(key) per registered column (immutable per pipeline):
+	 * computed once for the guarded dense-map sizing, not per step */
+	std::vector<std::pair<std::pair<const void *, int64_t>,
+			      unsigned long long>> maxk_cache;
 
 	KernelStatAcc &stat(const char *name)
 	{
