@@ -2737,15 +2737,17 @@ gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
  * (3 + 5*natts int32s).  Pinned against the reference's
  * create_memtuple_binding in tests.
  */
-extern "C" gg_status
-gg_engine_memtuple_binding(int natts, const int32_t *attlen,
-			   const char *attalign, int32_t *out)
+static gg_status
+mt_binding_dump(int natts, const int32_t *attlen, const char *attalign,
+		bool large, int32_t *out)
 {
 	if (!attlen || !attalign || !out)
 		return fail(GG_EINVAL, "bad memtuple_binding args");
 
 	MtBind b;
-	int rc = mt_compute_binding(natts, attlen, attalign, &b);
+	int rc = large
+		? mt_compute_binding_large(natts, attlen, attalign, &b)
+		: mt_compute_binding(natts, attlen, attalign, &b);
 
 	if (rc)
 		return fail(GG_EINVAL, "unsupported memtuple schema (%d)",
@@ -2762,6 +2764,20 @@ gg_engine_memtuple_binding(int natts, const int32_t *attlen,
 		out[3 + i * 5 + 4] = b.null_mask[i];
 	}
 	return GG_OK;
+}
+
+extern "C" gg_status
+gg_engine_memtuple_binding_large(int natts, const int32_t *attlen,
+				 const char *attalign, int32_t *out)
+{
+	return mt_binding_dump(natts, attlen, attalign, true, out);
+}
+
+extern "C" gg_status
+gg_engine_memtuple_binding(int natts, const int32_t *attlen,
+			   const char *attalign, int32_t *out)
+{
+	return mt_binding_dump(natts, attlen, attalign, false, out);
 }
 
 /*
@@ -2783,33 +2799,33 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 	if (!attlen || !attalign || !cols || nrows < 0 || !out || !out_len)
 		return fail(GG_EINVAL, "bad memtuple_encode args");
 
-	MtBind b;
+	MtBind b, bl;
 	int rc = mt_compute_binding(natts, attlen, attalign, &b);
 
 	if (rc)
 		return fail(GG_EINVAL, "unsupported memtuple schema (%d)",
 			    rc);
+	(void) mt_compute_binding_large(natts, attlen, attalign, &bl);
 
 	/* per-row lengths + offsets on the host (it owns the null flags
-	 * and varlena offsets); mirrors d_mt_len */
-	std::vector<int64_t> offs(nrows + 1, 0);
-
-	for (int64_t r = 0; r < nrows; r++)
+	 * and varlena offsets); mirrors d_mt_len incl. the small->large
+	 * binding switch over MEMTUPLE_LEN_FITSHORT */
+	auto row_len = [&](const MtBind & bb, int64_t r) -> int64_t
 	{
-		int64_t len = b.var_start;
+		int64_t len = bb.var_start;
 		bool hasnull = false;
 
 		for (int i = 0; i < natts; i++)
 			if (nulls && nulls[i] && nulls[i][r])
 			{
 				hasnull = true;
-				len -= b.len_aligned[i];
+				len -= bb.len_aligned[i];
 			}
 		if (hasnull)
-			len += b.null_bitmap_extra;
+			len += bb.null_bitmap_extra;
 		for (int i = 0; i < natts; i++)
 		{
-			if (!b.is_var[i] ||
+			if (!bb.is_var[i] ||
 			    (nulls && nulls[i] && nulls[i][r]))
 				continue;
 			const gg_text_col *tc =
@@ -2820,19 +2836,22 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 				len += paylen + 1;
 			else
 			{
-				int al = b.align_of[i];
+				int al = bb.align_of[i];
 
 				len = (len + al - 1) & ~(int64_t) (al - 1);
 				len += 4 + paylen;
 			}
 		}
-		len = (len + 7) & ~(int64_t) 7;
+		return (len + 7) & ~(int64_t) 7;
+	};
+	std::vector<int64_t> offs(nrows + 1, 0);
+
+	for (int64_t r = 0; r < nrows; r++)
+	{
+		int64_t len = row_len(b, r);
+
 		if (len > 0xFFF0)
-			return fail(GG_EINVAL,
-				    "row %lld memtuple %lld bytes: large "
-				    "(4-byte varoffset) tuples not "
-				    "supported this round",
-				    (long long) r, (long long) len);
+			len = row_len(bl, r);
 		offs[r + 1] = offs[r] + len;
 	}
 	int64_t total = offs[nrows];
@@ -2904,7 +2923,7 @@ gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 			 hipMemcpyHostToDevice));
 	{
 		hipError_t he = launch_mt_encode(
-			e.stream, &b, (const void *const *) d_colp,
+			e.stream, &b, &bl, (const void *const *) d_colp,
 			(const int64_t *const *) d_voffp,
 			(const uint8_t *const *) d_nullp, nrows, d_offs,
 			d_out);
@@ -2948,12 +2967,13 @@ gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 	    !out_nrows)
 		return fail(GG_EINVAL, "bad memtuple_decode args");
 
-	MtBind b;
+	MtBind b, bl;
 	int rc = mt_compute_binding(natts, attlen, attalign, &b);
 
 	if (rc)
 		return fail(GG_EINVAL, "unsupported memtuple schema (%d)",
 			    rc);
+	(void) mt_compute_binding_large(natts, attlen, attalign, &bl);
 
 	/* walk tuple headers to find offsets (MEMTUP_LEN_MASK) */
 	std::vector<int64_t> offs;
@@ -3038,7 +3058,7 @@ gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 			 hipMemcpyHostToDevice));
 	{
 		hipError_t he = launch_mt_decode(
-			e.stream, &b, d_offs, nrows, d_in, stream_len,
+			e.stream, &b, &bl, d_offs, nrows, d_in, stream_len,
 			(void *const *) d_colp,
 			(unsigned long long *const *) d_voffp,
 			(uint32_t *const *) d_vlenp,

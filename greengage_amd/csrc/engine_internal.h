@@ -408,15 +408,17 @@ struct MtBind
 };
 int mt_compute_binding(int natts, const int32_t *attlen,
 		       const char *attalign, MtBind *out);
+int mt_compute_binding_large(int natts, const int32_t *attlen,
+			     const char *attalign, MtBind *out);
 hipError_t launch_mt_encode(hipStream_t s, const MtBind *b,
-			    const void *const *cols,
+			    const MtBind *bl, const void *const *cols,
 			    const int64_t *const *var_offs,
 			    const uint8_t *const *nulls, int64_t nrows,
 			    const int64_t *offs, uint8_t *out);
 hipError_t launch_mt_decode(hipStream_t s, const MtBind *b,
-			    const int64_t *offs, int64_t nrows,
-			    const uint8_t *in, int64_t in_len,
-			    void *const *cols,
+			    const MtBind *bl, const int64_t *offs,
+			    int64_t nrows, const uint8_t *in,
+			    int64_t in_len, void *const *cols,
 			    unsigned long long *const *var_out_offs,
 			    uint32_t *const *var_out_lens,
 			    uint8_t *const *nulls, unsigned long long *err);

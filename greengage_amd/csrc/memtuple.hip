@@ -51,9 +51,9 @@ mt_align(uint32_t off, int align)
  * fixed-width attrs (attlen 1/2/4/8, attalign c/s/i/d, byval).
  * Returns 0 on success, nonzero for unsupported schemas.
  */
-int
-mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
-		   MtBind *out)
+static int
+mt_compute_one(int natts, const int32_t *attlen, const char *attalign,
+	       bool islarge, MtBind *out)
 {
 	if (natts < 1 || natts > GG_MT_MAX_ATTS)
 		return 1;
@@ -88,9 +88,9 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 					   out->column_align);
 	}
 	/* four passes by alignment, as create_col_bind does; varlena
-	 * takes a 2-byte varoffset slot in the 's' pass (SMALL binding —
-	 * tuples over 0xFFF0 bytes would use 4-byte offsets in the 'i'
-	 * pass; the encoder rejects those, see GG_MT_FITSHORT) */
+	 * takes a 2-byte varoffset slot in the 's' pass (small binding)
+	 * or a 4-byte slot in the 'i' pass (large binding, tuples over
+	 * MEMTUPLE_LEN_FITSHORT=0xFFF0 — create_col_bind islarge) */
 	uint32_t cur = (out->column_align == 8) ? 8 : 4;
 	int physical = 0;
 	int prev = -1;
@@ -106,10 +106,10 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 
 			if (isvar)
 			{
-				if (pass != 2)
+				if (pass != (islarge ? 1 : 2))
 					continue;
-				a = 's';	/* slot alignment */
-				slot = 2;
+				a = islarge ? 'i' : 's';
+				slot = islarge ? 4 : 2;
 			}
 			else
 			{
@@ -154,6 +154,20 @@ mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
 	return 0;
 }
 
+int
+mt_compute_binding(int natts, const int32_t *attlen, const char *attalign,
+		   MtBind *out)
+{
+	return mt_compute_one(natts, attlen, attalign, false, out);
+}
+
+int
+mt_compute_binding_large(int natts, const int32_t *attlen,
+			 const char *attalign, MtBind *out)
+{
+	return mt_compute_one(natts, attlen, attalign, true, out);
+}
+
 namespace
 {
 
@@ -169,8 +183,8 @@ d_mt_align8(uint32_t x)
  * the short form (no alignment); longer keep the 4-byte header,
  * aligned to the attr alignment (:486-500). */
 __device__ uint32_t
-d_mt_len(const MtBind &b, const int64_t *const *var_offs,
-	 const uint8_t *const *nulls, int64_t row, bool *hasnull_out)
+d_mt_len_one(const MtBind &b, const int64_t *const *var_offs,
+	     const uint8_t *const *nulls, int64_t row, bool *hasnull_out)
 {
 	uint32_t len = (uint32_t) b.var_start;
 	bool hasnull = false;
@@ -202,8 +216,26 @@ d_mt_len(const MtBind &b, const int64_t *const *var_offs,
 	return d_mt_align8(len);
 }
 
+/* size with the small binding first; over FITSHORT the large binding
+ * applies (compute_memtuple_size :509) */
+__device__ uint32_t
+d_mt_len(const MtBind &b, const MtBind &bl,
+	 const int64_t *const *var_offs, const uint8_t *const *nulls,
+	 int64_t row, bool *hasnull_out, bool *islarge_out)
+{
+	uint32_t len = d_mt_len_one(b, var_offs, nulls, row, hasnull_out);
+
+	if (len <= 0xFFF0u)
+	{
+		*islarge_out = false;
+		return len;
+	}
+	*islarge_out = true;
+	return d_mt_len_one(bl, var_offs, nulls, row, hasnull_out);
+}
+
 __global__ void
-k_mt_encode(MtBind b, const void *const *__restrict__ cols,
+k_mt_encode(MtBind bs, MtBind blg, const void *const *__restrict__ cols,
 	    const int64_t *const *__restrict__ var_offs,
 	    const uint8_t *const *__restrict__ nulls, int64_t nrows,
 	    const int64_t *__restrict__ offs, uint8_t *__restrict__ out)
@@ -213,16 +245,18 @@ k_mt_encode(MtBind b, const void *const *__restrict__ cols,
 	for (int64_t r = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     r < nrows; r += stride)
 	{
-		bool hasnull;
-		uint32_t len = d_mt_len(b, var_offs, nulls, r, &hasnull);
+		bool hasnull, islarge;
+		uint32_t len = d_mt_len(bs, blg, var_offs, nulls, r,
+					&hasnull, &islarge);
+		const MtBind &b = islarge ? blg : bs;
 		uint8_t *tup = out + offs[r];
 
 		for (uint32_t z = 0; z < len; z++)
 			tup[z] = 0;
-		/* header: lead bit | len | hasnull (memtup.h) */
+		/* header: lead bit | len | hasnull | LARGETUP (memtup.h) */
 		{
 			uint32_t hdr = 0x80000000u | len |
-				(hasnull ? 1u : 0u);
+				(hasnull ? 1u : 0u) | (islarge ? 2u : 0u);
 
 			memcpy(tup, &hdr, 4);
 		}
@@ -302,11 +336,20 @@ k_mt_encode(MtBind b, const void *const *__restrict__ cols,
 							bytes[o0 + z];
 					attr_len = paylen + 4;
 				}
-				/* 2-byte varoffset relative to start
-				 * (form_to :794) */
-				uint16_t voff = (uint16_t) vcur;
+				/* varoffset relative to start (form_to
+				 * :794): 2 bytes small, 4 bytes large */
+				if (b.len[i] == 2)
+				{
+					uint16_t voff = (uint16_t) vcur;
 
-				memcpy(dst, &voff, 2);
+					memcpy(dst, &voff, 2);
+				}
+				else
+				{
+					uint32_t voff = vcur;
+
+					memcpy(dst, &voff, 4);
+				}
 				vcur += attr_len;
 				continue;
 			}
@@ -333,8 +376,8 @@ k_mt_encode(MtBind b, const void *const *__restrict__ cols,
 }
 
 __global__ void
-k_mt_decode(MtBind b, const int64_t *__restrict__ offs, int64_t nrows,
-	    const uint8_t *__restrict__ in, int64_t in_len,
+k_mt_decode(MtBind bs, MtBind blg, const int64_t *__restrict__ offs,
+	    int64_t nrows, const uint8_t *__restrict__ in, int64_t in_len,
 	    void *const *__restrict__ cols,
 	    unsigned long long *const *__restrict__ var_out_offs,
 	    uint32_t *const *__restrict__ var_out_lens,
@@ -356,6 +399,7 @@ k_mt_decode(MtBind b, const int64_t *__restrict__ offs, int64_t nrows,
 			continue;
 		}
 		bool hasnull = (hdr & 1u) != 0;
+		const MtBind &b = (hdr & 2u) ? blg : bs;	/* LARGETUP */
 		const uint8_t *start = tup +
 			(hasnull ? b.null_bitmap_extra : 0);
 		const uint8_t *nullp = tup + 4;
@@ -408,11 +452,19 @@ k_mt_decode(MtBind b, const int64_t *__restrict__ offs, int64_t nrows,
 
 			if (b.is_var[i])
 			{
-				/* 2-byte varoffset relative to start
+				/* varoffset relative to start
 				 * (memtuple_get_attr_data_ptr :533) */
-				uint16_t voff;
+				uint32_t voff;
 
-				memcpy(&voff, src, 2);
+				if (b.len[i] == 2)
+				{
+					uint16_t v16;
+
+					memcpy(&v16, src, 2);
+					voff = v16;
+				}
+				else
+					memcpy(&voff, src, 4);
 				const uint8_t *vp = start + voff;
 				uint32_t paylen;
 				int64_t payoff;
@@ -490,26 +542,27 @@ mt_grid(int64_t n)
 }				/* anonymous namespace */
 
 hipError_t
-launch_mt_encode(hipStream_t s, const MtBind *b, const void *const *cols,
-		 const int64_t *const *var_offs,
+launch_mt_encode(hipStream_t s, const MtBind *b, const MtBind *bl,
+		 const void *const *cols, const int64_t *const *var_offs,
 		 const uint8_t *const *nulls, int64_t nrows,
 		 const int64_t *offs, uint8_t *out)
 {
 	hipLaunchKernelGGL(k_mt_encode, dim3(mt_grid(nrows)), dim3(256), 0,
-			   s, *b, cols, var_offs, nulls, nrows, offs, out);
+			   s, *b, *bl, cols, var_offs, nulls, nrows, offs,
+			   out);
 	return hipGetLastError();
 }
 
 hipError_t
-launch_mt_decode(hipStream_t s, const MtBind *b, const int64_t *offs,
-		 int64_t nrows, const uint8_t *in, int64_t in_len,
-		 void *const *cols,
+launch_mt_decode(hipStream_t s, const MtBind *b, const MtBind *bl,
+		 const int64_t *offs, int64_t nrows, const uint8_t *in,
+		 int64_t in_len, void *const *cols,
 		 unsigned long long *const *var_out_offs,
 		 uint32_t *const *var_out_lens, uint8_t *const *nulls,
 		 unsigned long long *err)
 {
 	hipLaunchKernelGGL(k_mt_decode, dim3(mt_grid(nrows)), dim3(256), 0,
-			   s, *b, offs, nrows, in, in_len, cols,
+			   s, *b, *bl, offs, nrows, in, in_len, cols,
 			   var_out_offs, var_out_lens, nulls, err);
 	return hipGetLastError();
 }

@@ -129,6 +129,10 @@ def _load():
     lib.gg_engine_memtuple_binding.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
         ctypes.POINTER(ctypes.c_int32)]
+    lib.gg_engine_memtuple_binding_large.restype = ctypes.c_int
+    lib.gg_engine_memtuple_binding_large.argtypes = [
+        ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
+        ctypes.POINTER(ctypes.c_int32)]
     lib.gg_engine_memtuple_encode.restype = ctypes.c_int
     lib.gg_engine_memtuple_encode.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
@@ -414,6 +418,19 @@ class Engine:
         vals = [bytes(pool[offs[i]:offs[i] + lens[i]])
                 for i in range(n.value)]
         return vals, nulls[:n.value].copy()
+
+    @staticmethod
+    def memtuple_binding_large(attlen, attalign):
+        """Host-only restated LARGE binding layout (CPU-testable)."""
+        natts = len(attlen)
+        al = (ctypes.c_int32 * natts)(*attlen)
+        out = (ctypes.c_int32 * (3 + 5 * natts))()
+        _check(lib().gg_engine_memtuple_binding_large(
+            natts, al, "".join(attalign).encode(), out),
+            "memtuple_binding_large")
+        meta = (out[0], out[1], out[2])
+        per = [tuple(out[3 + i * 5: 8 + i * 5]) for i in range(natts)]
+        return meta, per
 
     @staticmethod
     def memtuple_binding(attlen, attalign):

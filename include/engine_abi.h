@@ -238,7 +238,9 @@ gg_status gg_engine_aocs_decode_text(const uint8_t *stream,
  * pool + n+1 prefix offsets) and a gg_text_out on decode (per-row
  * offset/length pairs referencing the INPUT stream buffer —
  * zero-copy).  nulls[i] is a byte-per-row flag array or NULL.
- * Tuples over 0xFFF0 bytes (4-byte varoffsets) are rejected. */
+ * Tuples over MEMTUPLE_LEN_FITSHORT (0xFFF0) switch to the large
+ * binding (4-byte varoffsets, MEMTUP_LARGETUP header flag), exactly
+ * as memtuple_form_to does. */
 typedef struct gg_text_col
 {
 	const uint8_t *bytes;
@@ -253,6 +255,10 @@ typedef struct gg_text_out
 
 gg_status gg_engine_memtuple_binding(int natts, const int32_t *attlen,
 				     const char *attalign, int32_t *out);
+gg_status gg_engine_memtuple_binding_large(int natts,
+					   const int32_t *attlen,
+					   const char *attalign,
+					   int32_t *out);
 gg_status gg_engine_memtuple_encode(int natts, const int32_t *attlen,
 				    const char *attalign,
 				    const void *const *cols,

@@ -493,9 +493,9 @@ class MtSchema:
         offs_a = np.array(offs, np.int64)
         v = np.array(ints, np.int64)
         nl = np.array([int(x) for x in isnull], np.uint8)
-        out = np.zeros(1 << 16, np.uint8)
+        out = np.zeros(1 << 20, np.uint8)
         ln = self.D.ref_mt_form_var(self.bind, v, is_text, bts, offs_a,
-                                    nl, out, 1 << 16)
+                                    nl, out, 1 << 20)
         assert ln > 0, ln
         return out[:ln].copy()
 

@@ -159,3 +159,58 @@ def test_gpu_encode_text_matches_reference_bytes(eng):
                         assert dcols[i][r] == cols[i][r], (attlen, i, r)
                     else:
                         assert dcols[i][r] == cols[i][r], (attlen, i, r)
+
+
+def test_gpu_encode_large_tuples(eng):
+    """Tuples over MEMTUPLE_LEN_FITSHORT switch to the LARGE binding
+    (4-byte varoffsets, MEMTUP_LARGETUP flag) — byte-identical to the
+    reference, mixed with small tuples in one stream."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    from greengage_amd.engine import Engine as E
+    rng = np.random.default_rng(46)
+    attlen, attalign = [8, -1, -1, 4], ["d", "i", "i", "i"]
+    ref = pyoracle.MtSchema(attlen, attalign)
+    nrows = 60
+    t1, t2, ints1, ints4 = [], [], [], []
+    for r in range(nrows):
+        if r % 3 == 0:  # large row: two ~40KB payloads
+            t1.append(bytes(rng.integers(65, 91, 40000).astype(np.uint8)))
+            t2.append(bytes(rng.integers(97, 123, 30000).astype(np.uint8)))
+        else:
+            t1.append(b"small-" + bytes([65 + r % 26]))
+            t2.append(bytes(rng.integers(65, 91, int(rng.integers(0, 200)))
+                            .astype(np.uint8)))
+        ints1.append(r * 1000)
+        ints4.append(-r)
+    cols = [np.array(ints1, np.int64), t1, t2, np.array(ints4, np.int32)]
+    nulls = [(rng.random(nrows) < 0.2).astype(np.uint8)
+             for _ in range(4)]
+    stream = E.memtuple_encode(attlen, attalign, cols, nulls)
+    parts = []
+    for r in range(nrows):
+        vals = [int(cols[0][r]), t1[r], t2[r], int(cols[3][r])]
+        isnull = [int(n[r]) for n in nulls]
+        parts.append(ref.form_var(vals, isnull))
+    refstream = np.concatenate(parts)
+    assert len(stream) == len(refstream)
+    assert np.array_equal(stream, refstream)
+    # large flag present on the big rows
+    pos = 0
+    nlarge = 0
+    while pos < len(stream):
+        hdr = int(stream[pos:pos + 4].view(np.uint32)[0])
+        if hdr & 2:
+            nlarge += 1
+        pos += hdr & 0x3FFFFFF8
+    assert nlarge > 0
+    # round trip
+    dcols, dnulls = E.memtuple_decode(attlen, attalign, stream,
+                                      nrows + 5)
+    for i, l in enumerate(attlen):
+        assert np.array_equal(dnulls[i] != 0, nulls[i] != 0), i
+        for r in range(nrows):
+            if nulls[i][r]:
+                continue
+            assert dcols[i][r] == (cols[i][r] if l == -1
+                                   else cols[i][r]), (i, r)

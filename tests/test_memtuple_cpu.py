@@ -138,3 +138,24 @@ def test_reference_form_var_header():
     tup2 = ref.form_var([7, long, -3], [0, 0, 0])
     assert long in tup2.tobytes()
     assert len(tup2) > 300
+
+
+def test_binding_matches_reference_large():
+    """LARGE binding (4-byte varoffsets, used over 0xFFF0 bytes): the
+    engine's large layout must equal the reference's large_bind."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    import sys
+    sys.path.insert(0, ".")
+    from greengage_amd.engine import Engine
+    for attlen, attalign in TEXT_SCHEMAS:
+        ref = pyoracle.MtSchema(attlen, list(attalign))
+        meta, per = Engine.memtuple_binding_large(attlen, list(attalign))
+        for a in range(1, len(attlen) + 1):
+            roff, rlen, rlen_al, rflag, rnb, rnm = ref.colbind(
+                a, large=True)
+            eoff, elen, elen_al, enb, enm = per[a - 1]
+            assert (eoff, elen, elen_al, enb, enm) == \
+                (roff, rlen, rlen_al, rnb, rnm), (attlen, a)
+            if attlen[a - 1] == -1:
+                assert elen == 4, (attlen, a)
