@@ -2390,6 +2390,12 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 	framed.reserve((size_t) stream_len);
 	int64_t pos = 0;
 	int64_t expect_rownum = -1;
+	/* LargeContent reassembly state: a kind-2 metadata block opens a
+	 * frame whose payload is the concatenated content of the
+	 * following SmallContent fragment blocks (rowCount 0, no
+	 * firstRowNum) — AppendOnlyStorageRead_Content:1240 */
+	int64_t large_remaining = 0;
+	size_t large_fpos = 0;
 
 	while (pos < stream_len)
 	{
@@ -2416,6 +2422,27 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 					    "compressed AO block at %lld but "
 					    "comptype none",
 					    (long long) pos);
+		}
+		else if (kind == 2)
+		{		/* AoHeaderKind_LargeContent: header-only
+				 * metadata (largeRowCount 25 bits split 23+2,
+				 * largeContentLength 30 bits —
+				 * cdbappendonlystorage_int.h:227) */
+			if (large_remaining > 0)
+				return fail(GG_EINVAL,
+					    "nested LargeContent at %lld",
+					    (long long) pos);
+			rowcount = (int32_t) (((w0 & 0x7FFFFF) << 2) |
+					      ((w1 >> 30) & 3));
+			datalen = 0;	/* no content of its own */
+			int32_t biglen = (int32_t) (w1 & 0x3FFFFFFF);
+
+			large_remaining = biglen;
+			large_fpos = framed.size();
+			framed.resize(large_fpos + 8);
+			std::memcpy(framed.data() + large_fpos, &biglen, 4);
+			std::memcpy(framed.data() + large_fpos + 4,
+				    &rowcount, 4);
 		}
 		else if (kind == 3)
 		{		/* AoHeaderKind_NonBulkDenseContent */
@@ -2450,6 +2477,11 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 			return fail(GG_EINVAL,
 				    "unsupported AoHeaderKind %d at %lld",
 				    kind, (long long) pos);
+
+		if (large_remaining > 0 && kind != 1 && kind != 2)
+			return fail(GG_EINVAL,
+				    "expected SmallContent fragment at %lld "
+				    "inside LargeContent", (long long) pos);
 
 		/* stored bytes = compressed when compressedLength != 0
 		 * (CompressAppend rule, cdbappendonlystoragewrite.c:1207) */
@@ -2502,18 +2534,40 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 					    (long long) expect_rownum);
 			expect_rownum = frn + rowcount;
 		}
-		size_t fpos = framed.size();
+		size_t fpos;
+		size_t dst_off;
+		bool fragment = (large_remaining > 0 && kind == 1);
 
-		framed.resize((fpos + 8 + (size_t) datalen + 7) & ~(size_t) 7);
-		std::memcpy(framed.data() + fpos, &datalen, 4);
-		std::memcpy(framed.data() + fpos + 4, &rowcount, 4);
+		if (kind == 2)
+		{		/* metadata emitted above; nothing else */
+			pos = overall;
+			continue;
+		}
+		if (fragment)
+		{
+			if (datalen > large_remaining)
+				return fail(GG_EINVAL,
+					    "LargeContent overrun at %lld",
+					    (long long) pos);
+			dst_off = framed.size();
+			framed.resize(dst_off + (size_t) datalen);
+			large_remaining -= datalen;
+		}
+		else
+		{
+			fpos = framed.size();
+			framed.resize(fpos + 8 + (size_t) datalen);
+			std::memcpy(framed.data() + fpos, &datalen, 4);
+			std::memcpy(framed.data() + fpos + 4, &rowcount, 4);
+			dst_off = fpos + 8;
+		}
 		if (complen == 0)
-			std::memcpy(framed.data() + fpos + 8,
+			std::memcpy(framed.data() + dst_off,
 				    stream + hdr_end, (size_t) datalen);
 		else if (comptype == 1)
 		{
 			uLongf dl = (uLongf) datalen;
-			int zrc = uncompress(framed.data() + fpos + 8, &dl,
+			int zrc = uncompress(framed.data() + dst_off, &dl,
 					     stream + hdr_end,
 					     (uLong) complen);
 
@@ -2526,7 +2580,7 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 		}
 		else if (comptype == 2)
 		{
-			size_t dl = ZSTD_decompress(framed.data() + fpos + 8,
+			size_t dl = ZSTD_decompress(framed.data() + dst_off,
 						    (size_t) datalen,
 						    stream + hdr_end,
 						    (size_t) complen);
@@ -2539,8 +2593,15 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 		else
 			return fail(GG_EINVAL, "unknown comptype %d",
 				    comptype);
+		/* frames stay 8-aligned (AOStorage_RoundUp8 analog) */
+		if (!fragment || large_remaining == 0)
+			framed.resize((framed.size() + 7) & ~(size_t) 7);
 		pos = overall;
 	}
+	if (large_remaining > 0)
+		return fail(GG_EINVAL,
+			    "LargeContent truncated: %lld bytes missing",
+			    (long long) large_remaining);
 	return GG_OK;
 }
 

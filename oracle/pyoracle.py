@@ -282,6 +282,10 @@ def dsb_ref():
         D.ref_ao_wrap_stream_bd.argtypes = [
             _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.c_int, _P_U8, I64, ctypes.POINTER(I64)]
+        D.ref_ao_wrap_stream_large.restype = ctypes.c_int
+        D.ref_ao_wrap_stream_large.argtypes = [
+            _P_U8, I64, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int32, _P_U8, I64, ctypes.POINTER(I64)]
         D.ref_ao_probe_block.restype = ctypes.c_int
         D.ref_ao_probe_block.argtypes = [
             _P_U8, ctypes.c_int] + [ctypes.POINTER(ctypes.c_int32)] * 6
@@ -367,6 +371,24 @@ def ao_wrap_bulkdense(framed, comptype=0, complevel=0, checksums=1,
     rc = D.ref_ao_wrap_stream_bd(framed, len(framed), checksums,
                                  firstrownum, comptype, complevel, out,
                                  cap, ctypes.byref(olen))
+    assert rc == 0, rc
+    return out[:olen.value].copy()
+
+
+def ao_wrap_large(framed, checksums=1, comptype=0, complevel=0,
+                  frag_size=8192):
+    """Wrap each frame as a LargeContent metadata block + SmallContent
+    fragments (the multi-block form for content larger than one AO
+    block), via the reference's MakeLargeContentHeader."""
+    D = dsb_ref()
+    assert D is not None, "reference AO codec missing"
+    framed = np.ascontiguousarray(framed, np.uint8)
+    cap = len(framed) * 2 + (1 << 18)
+    out = np.zeros(cap, np.uint8)
+    olen = I64()
+    rc = D.ref_ao_wrap_stream_large(framed, len(framed), checksums,
+                                    comptype, complevel, frag_size,
+                                    out, cap, ctypes.byref(olen))
     assert rc == 0, rc
     return out[:olen.value].copy()
 

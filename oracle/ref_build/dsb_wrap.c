@@ -870,6 +870,118 @@ ref_ao_wrap_stream_bd(const uint8 *framed, int64 framed_len, int checksums,
 	return 0;
 }
 
+/*
+ * LargeContent form: each frame becomes a header-only LargeContent
+ * metadata block (kind 2: largeRowCount + largeContentLength,
+ * firstRowNum) followed by SmallContent fragment blocks with
+ * rowCount=0 and no firstRowNum — exactly the
+ * AppendOnlyStorageWrite_LargeContent layout
+ * (cdbappendonlystoragewrite.c:1600-1720).  frag_size bounds each
+ * fragment's uncompressed content; comptype compresses per fragment.
+ */
+int
+ref_ao_wrap_stream_large(const uint8 *framed, int64 framed_len,
+			 int checksums, int comptype, int complevel,
+			 int32 frag_size, uint8 *out, int64 out_cap,
+			 int64 *out_len)
+{
+	int64		pos = 0,
+				opos = 0;
+	int64		firstRowNum = 1;
+	static uint8 cbuf[4 * 1024 * 1024];
+
+	while (pos < framed_len)
+	{
+		int32		sz,
+					rowcount;
+		int32		hdrlen,
+					remaining;
+		const uint8 *content;
+		uint8	   *hdr;
+
+		memcpy(&sz, framed + pos, 4);
+		memcpy(&rowcount, framed + pos + 4, 4);
+		pos += 8;
+		content = framed + pos;
+
+		/* metadata block: header only (+FRN), no content */
+		hdrlen = 8 + (checksums ? 8 : 0) + 8 /* firstRowNum */ ;
+		if (opos + hdrlen > out_cap)
+			return -1;
+		hdr = out + opos;
+		memset(hdr, 0, hdrlen);
+		AppendOnlyStorageFormat_MakeLargeContentHeader(
+			hdr, checksums != 0, true, REF_AO_VERSION, firstRowNum,
+			REF_AO_EXECKIND, rowcount, sz);
+		firstRowNum += rowcount;
+		opos += hdrlen;
+
+		/* fragments */
+		remaining = sz;
+		while (remaining > 0)
+		{
+			int32		flen = remaining < frag_size
+				? remaining : frag_size;
+			int32		complen = 0,
+						stored,
+						fh,
+						padded;
+			const uint8 *src = content + (sz - remaining);
+			const uint8 *body = src;
+
+			if (comptype == 1)
+			{
+				uLongf		dl = sizeof(cbuf);
+
+				if (compress2(cbuf, &dl, src, flen,
+					      complevel) != Z_OK)
+					return -3;
+				complen = (int32) dl;
+			}
+			else if (comptype == 2)
+			{
+				size_t		dl = ZSTD_compress(cbuf,
+							   sizeof(cbuf), src, flen,
+							   complevel);
+
+				if (ZSTD_isError(dl))
+					return -3;
+				complen = (int32) dl;
+			}
+			else if (comptype != 0)
+				return -4;
+			if (complen == 0 || complen >= flen)
+			{
+				stored = flen;
+				complen = 0;
+				body = src;
+			}
+			else
+			{
+				stored = complen;
+				body = cbuf;
+			}
+			fh = 8 + (checksums ? 8 : 0);	/* no firstRowNum */
+			padded = ao_roundup8(stored);
+			if (opos + fh + padded > out_cap)
+				return -1;
+			hdr = out + opos;
+			memset(hdr, 0, fh + padded);
+			memcpy(hdr + fh, body, stored);
+			AppendOnlyStorageFormat_MakeSmallContentHeader(
+				hdr, checksums != 0, false /* no FRN */ ,
+				REF_AO_VERSION, 0, REF_AO_EXECKIND,
+				0 /* rowCount */ , flen, complen);
+			opos += fh + padded;
+			remaining -= flen;
+		}
+		pos += sz;
+		pos = (pos + 7) & ~(int64) 7;
+	}
+	*out_len = opos;
+	return 0;
+}
+
 /* reference-side parse of one AO block (for cross-checking the
  * engine's restated parser in tests) */
 int

@@ -351,3 +351,33 @@ def test_ao_engine_bulkdense_corruption_cpu():
     rc = _decode_ao_rc(lib, bad, comptype=1)
     assert rc != 0
     assert b"checksum" in lib.gg_engine_last_error()
+
+
+def test_ao_largecontent_layout():
+    """LargeContent (kind 2): header-only metadata with largeRowCount
+    (25 bits split 23+2) and largeContentLength, followed by rowCount=0
+    SmallContent fragments."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    rng = np.random.default_rng(25)
+    vals = rng.integers(-2**30, 2**30, 30000).astype(np.int64)
+    framed, nb = pyoracle.dsb_encode(vals, np.zeros(30000, np.uint8), 8,
+                                     2, 0, 0, blocksz=1 << 20)
+    assert nb == 1  # one big frame
+    ao = pyoracle.ao_wrap_large(framed, frag_size=4096)
+    w0 = int(ao[0:4].view(np.uint32)[0])
+    w1 = int(ao[4:8].view(np.uint32)[0])
+    assert ((w0 >> 28) & 7) == 2
+    rowcount = ((w0 & 0x7FFFFF) << 2) | ((w1 >> 30) & 3)
+    biglen = w1 & 0x3FFFFFFF
+    fsz = int(framed[0:4].view(np.int32)[0])
+    frc = int(framed[4:8].view(np.int32)[0])
+    assert (rowcount, biglen) == (frc, fsz)
+    # metadata block: header + crcs + firstRowNum only
+    assert _crc32c_ao(ao[0:12]) == int(ao[12:16].view(np.uint32)[0])
+    # first fragment right after: SmallContent, rowCount 0, no FRN
+    f0 = 24
+    fw0 = int(ao[f0:f0 + 4].view(np.uint32)[0])
+    assert ((fw0 >> 28) & 7) == 1
+    assert ((fw0 >> 27) & 1) == 0  # no firstRowNum
+    assert ((fw0 >> 10) & 0x3FFF) == 0  # rowCount 0

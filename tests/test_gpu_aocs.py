@@ -232,3 +232,37 @@ def test_gpu_decode_ao_bulkdense(eng):
         assert np.array_equal(gn != 0, nulls != 0)
         mask = nulls == 0
         assert np.array_equal(gv[mask], vals[mask]), comptype
+
+
+def test_gpu_decode_ao_largecontent(eng):
+    """LargeContent reassembly: metadata + fragments (plain and
+    compressed) rebuild the original datum-stream block and decode
+    bit-exactly on the GPU."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    from greengage_amd.engine import Engine as E
+    rng = np.random.default_rng(55)
+    n = 60000
+    vals = rng.integers(-2**40, 2**40, n).astype(np.int64)
+    nulls = (rng.random(n) < 0.1).astype(np.uint8)
+    framed, nb = pyoracle.dsb_encode(vals, nulls, 8, 2, 0, 0,
+                                     blocksz=1 << 21)
+    for comptype, level in ((0, 0), (1, 6), (2, 3)):
+        ao = pyoracle.ao_wrap_large(framed, comptype=comptype,
+                                    complevel=level, frag_size=8192)
+        gv, gn = E.aocs_decode_ao(ao, 1, 2, 2, 8, n + 10,
+                                  comptype=comptype)
+        assert len(gv) == n, comptype
+        assert np.array_equal(gn != 0, nulls != 0)
+        mask = nulls == 0
+        assert np.array_equal(gv[mask], vals[mask]), comptype
+    # mixed stream: large + regular small blocks behind one another
+    framed2, nb2 = pyoracle.dsb_encode(vals[:5000],
+                                       nulls[:5000], 8, 2, 0, 0)
+    ao_mixed = np.concatenate([
+        pyoracle.ao_wrap_large(framed, frag_size=8192),
+        pyoracle.ao_wrap(framed2, firstrownum=0)])  # no FRN: appended
+    # streams would otherwise trip the firstRowNum continuity check
+    gv, gn = E.aocs_decode_ao(ao_mixed, 1, 2, 2, 8, n + 5010)
+    assert len(gv) == n + 5000
+    assert np.array_equal(gv[:n][nulls == 0], vals[nulls == 0])
