@@ -298,7 +298,7 @@ def dsb_encode(vals, nulls, datumlen, version, rle, delta, blocksz=32768):
     assert D is not None, "reference dsb codec missing"
     v = np.ascontiguousarray(vals, np.int64)
     nl = np.ascontiguousarray(nulls, np.uint8)
-    cap = max(1 << 21, len(v) * 16)
+    cap = max(1 << 21, len(v) * 16) + blocksz + 64
     out = np.zeros(cap, np.uint8)
     olen = I64()
     nb = ctypes.c_int32()
@@ -429,7 +429,7 @@ def dsb_encode_text(values, nulls, version, rle, blocksz=32768):
     bts = (np.frombuffer(blob, np.uint8).copy() if blob
            else np.zeros(1, np.uint8))
     nl = np.ascontiguousarray(nulls, np.uint8)
-    cap = max(1 << 21, len(blob) * 4 + n * 8)
+    cap = max(1 << 21, len(blob) * 4 + n * 8) + blocksz + 64
     out = np.zeros(cap, np.uint8)
     olen, nb = I64(), ctypes.c_int32()
     rc = D.ref_dsb_encode_text(bts, offs, nl, n, version, rle, blocksz,
