@@ -3184,6 +3184,182 @@ gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 	return st;
 }
 
+/* ---------- Motion wire chunk framing (ABI surface, §8(f)3) ---------- */
+
+/*
+ * The interconnect carries tuples as 4-byte-headed chunks
+ * ([u16 payload size excl header][u16 TupleChunkType], native endian —
+ * tupchunk.h:34-49,80-84): a MemTuple is sent TC_WHOLE padded to
+ * TUPLE_CHUNK_ALIGN(4), or split into TC_PARTIAL_START/MID/END when it
+ * exceeds the max chunk length (SerializeTuple, tupser.c:400-612;
+ * splitting rule addByteStringToChunkList :230).  TC_END_OF_STREAM
+ * closes a sender's stream.  These two host-side helpers convert
+ * between a MemTuple stream (gg_engine_memtuple_encode's output) and
+ * the exact byte stream a CPU segment's Motion sender/receiver uses.
+ */
+#define GG_TC_WHOLE 0
+#define GG_TC_PARTIAL_START 1
+#define GG_TC_PARTIAL_MID 2
+#define GG_TC_PARTIAL_END 3
+#define GG_TC_END_OF_STREAM 4
+
+extern "C" gg_status
+gg_engine_motion_chunkify(const uint8_t *tuples, int64_t tuples_len,
+			  int32_t max_chunk, int append_eos, uint8_t *out,
+			  int64_t cap, int64_t *out_len)
+{
+	if (!tuples || tuples_len < 0 || !out || !out_len ||
+	    max_chunk < 16 || max_chunk > 0xFFFF)
+		return fail(GG_EINVAL, "bad chunkify args");
+
+	int64_t pos = 0, opos = 0;
+
+	auto emit = [&](uint16_t type, const uint8_t *data, int32_t n,
+			int32_t pad) -> bool
+	{
+		if (opos + 4 + n + pad > cap)
+			return false;
+		uint16_t sz = (uint16_t) (n + pad);
+
+		std::memcpy(out + opos, &sz, 2);
+		std::memcpy(out + opos + 2, &type, 2);
+		if (n)
+			std::memcpy(out + opos + 4, data, (size_t) n);
+		if (pad)
+			std::memset(out + opos + 4 + n, 0, (size_t) pad);
+		opos += 4 + n + pad;
+		return true;
+	};
+
+	while (pos < tuples_len)
+	{
+		uint32_t hdr;
+
+		if (pos + 8 > tuples_len)
+			return fail(GG_EINVAL, "truncated memtuple at %lld",
+				    (long long) pos);
+		std::memcpy(&hdr, tuples + pos, 4);
+		if (!(hdr & 0x80000000u))
+			return fail(GG_EINVAL, "lead bit clear at %lld",
+				    (long long) pos);
+		int64_t tlen = (int64_t) (hdr & 0x3FFFFFF8u);
+
+		if (tlen < 8 || pos + tlen > tuples_len)
+			return fail(GG_EINVAL, "bad memtuple len at %lld",
+				    (long long) pos);
+		/* memtuple sizes are 8-aligned, so the 4-align pad is 0 */
+		int32_t payload_max = max_chunk - 4;
+
+		if (tlen <= payload_max)
+		{
+			if (!emit(GG_TC_WHOLE, tuples + pos,
+				  (int32_t) tlen, 0))
+				return fail(GG_EINVAL, "chunk cap");
+		}
+		else
+		{
+			int64_t off = 0;
+			int nchunk = 0;
+
+			while (off < tlen)
+			{
+				int32_t n = (int32_t)
+					((tlen - off < payload_max)
+					 ? tlen - off : payload_max);
+				uint16_t ty = nchunk == 0
+					? GG_TC_PARTIAL_START
+					: (off + n == tlen
+					   ? GG_TC_PARTIAL_END
+					   : GG_TC_PARTIAL_MID);
+
+				if (!emit(ty, tuples + pos + off, n, 0))
+					return fail(GG_EINVAL, "chunk cap");
+				off += n;
+				nchunk++;
+			}
+		}
+		pos += tlen;
+	}
+	if (append_eos && !emit(GG_TC_END_OF_STREAM, nullptr, 0, 0))
+		return fail(GG_EINVAL, "chunk cap");
+	*out_len = opos;
+	return GG_OK;
+}
+
+extern "C" gg_status
+gg_engine_motion_dechunkify(const uint8_t *chunks, int64_t chunks_len,
+			    uint8_t *out, int64_t cap, int64_t *out_len,
+			    int *saw_eos)
+{
+	if (!chunks || chunks_len < 0 || !out || !out_len || !saw_eos)
+		return fail(GG_EINVAL, "bad dechunkify args");
+
+	int64_t pos = 0, opos = 0;
+	bool in_partial = false;
+
+	*saw_eos = 0;
+	while (pos < chunks_len)
+	{
+		uint16_t sz, ty;
+
+		if (pos + 4 > chunks_len)
+			return fail(GG_EINVAL, "truncated chunk at %lld",
+				    (long long) pos);
+		std::memcpy(&sz, chunks + pos, 2);
+		std::memcpy(&ty, chunks + pos + 2, 2);
+		if (pos + 4 + sz > chunks_len)
+			return fail(GG_EINVAL, "chunk overruns at %lld",
+				    (long long) pos);
+		switch (ty)
+		{
+			case GG_TC_WHOLE:
+				if (in_partial)
+					return fail(GG_EINVAL,
+						    "WHOLE inside partial "
+						    "at %lld",
+						    (long long) pos);
+				break;
+			case GG_TC_PARTIAL_START:
+				if (in_partial)
+					return fail(GG_EINVAL,
+						    "nested PARTIAL_START "
+						    "at %lld",
+						    (long long) pos);
+				in_partial = true;
+				break;
+			case GG_TC_PARTIAL_MID:
+			case GG_TC_PARTIAL_END:
+				if (!in_partial)
+					return fail(GG_EINVAL,
+						    "PARTIAL_%s without "
+						    "START at %lld",
+						    ty == GG_TC_PARTIAL_END
+						    ? "END" : "MID",
+						    (long long) pos);
+				if (ty == GG_TC_PARTIAL_END)
+					in_partial = false;
+				break;
+			case GG_TC_END_OF_STREAM:
+				*saw_eos = 1;
+				pos += 4 + sz;
+				continue;
+			default:
+				return fail(GG_EINVAL,
+					    "bad chunk type %d at %lld",
+					    (int) ty, (long long) pos);
+		}
+		if (opos + sz > cap)
+			return fail(GG_EINVAL, "dechunk cap");
+		std::memcpy(out + opos, chunks + pos + 4, sz);
+		opos += sz;
+		pos += 4 + sz;
+	}
+	if (in_partial)
+		return fail(GG_EINVAL, "partial tuple unterminated");
+	*out_len = opos;
+	return GG_OK;
+}
+
 /* ---------------- general hash group-by (ABI surface) ---------------- */
 
 extern "C" gg_status

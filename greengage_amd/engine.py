@@ -125,6 +125,15 @@ def _load():
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p,
         ctypes.c_int64, ctypes.POINTER(ctypes.c_int64),
         ctypes.POINTER(ctypes.c_int64)]
+    lib.gg_engine_motion_chunkify.restype = ctypes.c_int
+    lib.gg_engine_motion_chunkify.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int32, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.POINTER(ctypes.c_int64)]
+    lib.gg_engine_motion_dechunkify.restype = ctypes.c_int
+    lib.gg_engine_motion_dechunkify.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p,
+        ctypes.c_int64, ctypes.POINTER(ctypes.c_int64),
+        ctypes.POINTER(ctypes.c_int)]
     lib.gg_engine_memtuple_binding.restype = ctypes.c_int
     lib.gg_engine_memtuple_binding.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int32), ctypes.c_char_p,
@@ -418,6 +427,33 @@ class Engine:
         vals = [bytes(pool[offs[i]:offs[i] + lens[i]])
                 for i in range(n.value)]
         return vals, nulls[:n.value].copy()
+
+    @staticmethod
+    def motion_chunkify(tuples, max_chunk=8192, append_eos=True):
+        """Frame a MemTuple stream as interconnect tuple chunks."""
+        import numpy as np
+        tuples = np.ascontiguousarray(tuples, np.uint8)
+        cap = len(tuples) + (len(tuples) // 16 + 16) * 8 + 64
+        out = np.zeros(cap, np.uint8)
+        n = I64()
+        _check(lib().gg_engine_motion_chunkify(
+            tuples.ctypes.data_as(ctypes.c_void_p), len(tuples),
+            max_chunk, int(append_eos),
+            out.ctypes.data_as(ctypes.c_void_p), cap, ctypes.byref(n)),
+            "motion_chunkify")
+        return out[:n.value].copy()
+
+    @staticmethod
+    def motion_dechunkify(chunks):
+        import numpy as np
+        chunks = np.ascontiguousarray(chunks, np.uint8)
+        out = np.zeros(len(chunks) + 64, np.uint8)
+        n, eos = I64(), ctypes.c_int()
+        _check(lib().gg_engine_motion_dechunkify(
+            chunks.ctypes.data_as(ctypes.c_void_p), len(chunks),
+            out.ctypes.data_as(ctypes.c_void_p), len(out),
+            ctypes.byref(n), ctypes.byref(eos)), "motion_dechunkify")
+        return out[:n.value].copy(), bool(eos.value)
 
     @staticmethod
     def memtuple_binding_large(attlen, attalign):

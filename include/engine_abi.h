@@ -272,6 +272,19 @@ gg_status gg_engine_memtuple_decode(int natts, const int32_t *attlen,
 				    uint8_t *const *nulls,
 				    int64_t cap_rows, int64_t *out_nrows);
 
+/* Motion wire chunk framing (tupchunk.h:34-84, tupser.c:400): convert
+ * between a MemTuple stream and the interconnect's 4-byte-headed
+ * tuple chunks (TC_WHOLE / TC_PARTIAL_* splitting at max_chunk,
+ * TC_END_OF_STREAM).  Host-side. */
+gg_status gg_engine_motion_chunkify(const uint8_t *tuples,
+				    int64_t tuples_len, int32_t max_chunk,
+				    int append_eos, uint8_t *out,
+				    int64_t cap, int64_t *out_len);
+gg_status gg_engine_motion_dechunkify(const uint8_t *chunks,
+				      int64_t chunks_len, uint8_t *out,
+				      int64_t cap, int64_t *out_len,
+				      int *saw_eos);
+
 /* Decode REAL Append-Only storage blocks (headers + CRC32C verify +
  * datum-stream content on the GPU).  Replaces the AO read path's
  * header/checksum layer (cdbappendonlystorageformat.c:125,1661 and

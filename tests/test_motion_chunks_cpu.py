@@ -1,0 +1,81 @@
+"""Motion wire chunk framing (host-only): the 4-byte-headed tuple
+chunks the interconnect carries (tupchunk.h:34-84; splitting rule
+SerializeTuple/addByteStringToChunkList, tupser.c:400/:230).  Uses
+reference-formed MemTuples as payloads."""
+import numpy as np
+import pytest
+
+import pyoracle
+
+
+def _engine():
+    import sys
+    sys.path.insert(0, ".")
+    from greengage_amd.engine import Engine
+    return Engine
+
+
+def _mt_stream(n=50):
+    ref = pyoracle.MtSchema([8, 4, 4], ["d", "i", "i"])
+    parts = [ref.form([r, r * 2, -r], [0, 0, 0]) for r in range(n)]
+    return np.concatenate(parts), parts
+
+
+def test_chunk_roundtrip_whole():
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    E = _engine()
+    stream, parts = _mt_stream()
+    chunks = E.motion_chunkify(stream, max_chunk=8192)
+    # one TC_WHOLE per tuple + one TC_END_OF_STREAM
+    pos, types = 0, []
+    while pos < len(chunks):
+        sz = int(chunks[pos:pos + 2].view(np.uint16)[0])
+        ty = int(chunks[pos + 2:pos + 4].view(np.uint16)[0])
+        types.append(ty)
+        pos += 4 + sz
+    assert types == [0] * len(parts) + [4]  # WHOLE..., END_OF_STREAM
+    out, eos = E.motion_dechunkify(chunks)
+    assert eos
+    assert np.array_equal(out, stream)
+
+
+def test_chunk_split_large_tuple():
+    """A tuple larger than max_chunk splits into
+    PARTIAL_START/MID/END with max-size middle chunks."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    E = _engine()
+    ref = pyoracle.MtSchema([8, -1], ["d", "i"])
+    big = bytes(np.random.default_rng(0).integers(65, 91, 3000)
+                .astype(np.uint8))
+    tup = ref.form_var([1, big], [0, 0])
+    chunks = E.motion_chunkify(tup, max_chunk=1024, append_eos=False)
+    pos, seq = 0, []
+    while pos < len(chunks):
+        sz = int(chunks[pos:pos + 2].view(np.uint16)[0])
+        ty = int(chunks[pos + 2:pos + 4].view(np.uint16)[0])
+        seq.append((ty, sz))
+        pos += 4 + sz
+    assert seq[0][0] == 1 and seq[-1][0] == 3  # START ... END
+    assert all(t == 2 for t, _ in seq[1:-1])   # MIDs
+    assert all(s == 1020 for _, s in seq[:-1])  # filled to max
+    out, eos = E.motion_dechunkify(chunks)
+    assert not eos
+    assert np.array_equal(out, tup)
+
+
+def test_chunk_error_paths():
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    E = _engine()
+    stream, _ = _mt_stream(3)
+    chunks = E.motion_chunkify(stream)
+    # corrupt a type
+    bad = chunks.copy()
+    bad[2] = 9
+    with pytest.raises(RuntimeError):
+        E.motion_dechunkify(bad)
+    # truncated
+    with pytest.raises(RuntimeError):
+        E.motion_dechunkify(chunks[:len(chunks) - 3])
