@@ -2509,8 +2509,14 @@ ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
 					    "%lld: stored 0x%08X computed "
 					    "0x%08X", (long long) pos,
 					    stored_hdr, comp_hdr);
+			/* LargeContent metadata: the writer computes the
+			 * block CRC over ZERO bytes (MakeLargeContentHeader
+			 * passes hasFirstRowNum=false to
+			 * AddBlockHeaderChecksums, so overall there is just
+			 * the 16 header bytes) — mirror that */
+			int64_t blk_end = (kind == 2) ? pos + 16 : overall;
 			uint32_t comp_blk = ao_crc32c(stream + pos + 16,
-						      overall - pos - 16);
+						      blk_end - pos - 16);
 
 			if (comp_blk != stored_blk)
 				return fail(GG_EINVAL,
@@ -3361,6 +3367,167 @@ gg_engine_motion_dechunkify(const uint8_t *chunks, int64_t chunks_len,
 }
 
 /* ---------------- general hash group-by (ABI surface) ---------------- */
+
+/*
+ * Spill-tier hash group-by: when the input exceeds the
+ * device budget, hash-range-partition it on the GPU in chunks, stage
+ * the partitions in HOST memory (the spill arena — the reference
+ * writes hash-range partitions to spill files, spill_hash_table
+ * execHHashagg.c:1350), then reload and aggregate one partition at a
+ * time (agg_hash_reload :1852).  Partitions are disjoint by key, so
+ * the result is the concatenation of per-partition results.
+ * Partition id = top hash bits; the group table uses the low bits, so
+ * reloaded partitions still hash uniformly.
+ */
+extern "C" gg_status
+gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
+				 int64_t n, int64_t budget_bytes,
+				 int64_t *out_keys, int64_t *out_sums,
+				 int64_t *out_counts, int64_t cap,
+				 int64_t *out_ngroups,
+				 int32_t *out_npartitions)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!keys || !vals || n < 0 || !out_keys || !out_sums ||
+	    !out_counts || !out_ngroups || !out_npartitions ||
+	    budget_bytes < (1 << 20))
+		return fail(GG_EINVAL, "bad groupby_spill args");
+
+	/* in-memory fast path: input pairs + 2x table + compact buffers
+	 * ≈ 7 x n x 8 bytes */
+	if (n * 8 * 7 <= budget_bytes)
+	{
+		*out_npartitions = 1;
+		return gg_engine_hash_groupby_i64(keys, vals, n, out_keys,
+						  out_sums, out_counts, cap,
+						  out_ngroups);
+	}
+
+	/* number of hash-range partitions: each must fit the budget */
+	uint64_t P = next_pow2((uint64_t) ((n * 8 * 7) / budget_bytes) + 1);
+
+	if (P > 4096)
+		return fail(GG_EINVAL, "input needs %llu partitions (>4096)",
+			    (unsigned long long) P);
+	int shift = 32;
+
+	for (uint64_t t = P; t > 1; t >>= 1)
+		shift--;
+
+	/* chunked partition pass: GPU hashes + scatters, host stages */
+	int64_t chunk = budget_bytes / (8 * 4);	/* in+out pairs resident */
+
+	if (chunk < 1024)
+		chunk = 1024;
+	std::vector<std::vector<int64_t>> part_k(P), part_v(P);
+	int64_t *dk = nullptr, *dv = nullptr, *sk = nullptr, *sv = nullptr;
+	unsigned long long *dcnt = nullptr;
+	gg_status st = GG_OK;
+
+	GG_HIP(hipMalloc((void **) &dk, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &dv, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &sk, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &sv, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &dcnt, (P + 1) * 8));
+
+	std::vector<unsigned long long> counts(P), curs(P + 1);
+
+	for (int64_t base = 0; st == GG_OK && base < n; base += chunk)
+	{
+		int64_t m = (n - base < chunk) ? n - base : chunk;
+
+		GG_HIP(hipMemcpy(dk, keys + base, (size_t) m * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(hipMemcpy(dv, vals + base, (size_t) m * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(hipMemsetAsync(dcnt, 0, P * 8, e.stream));
+		GG_HIP(launch_gb_part_count(e.stream, dk, m, shift, dcnt));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(counts.data(), dcnt, P * 8,
+				 hipMemcpyDeviceToHost));
+		curs[0] = 0;
+		for (uint64_t p2 = 0; p2 < P; p2++)
+			curs[p2 + 1] = curs[p2] + counts[p2];
+		GG_HIP(hipMemcpy(dcnt, curs.data(), P * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(launch_gb_part_scatter(e.stream, dk, dv, m, shift,
+					      dcnt, sk, sv));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		for (uint64_t p2 = 0; p2 < P; p2++)
+		{
+			if (!counts[p2])
+				continue;
+			size_t old = part_k[p2].size();
+
+			part_k[p2].resize(old + counts[p2]);
+			part_v[p2].resize(old + counts[p2]);
+			GG_HIP(hipMemcpy(part_k[p2].data() + old,
+					 sk + curs[p2],
+					 counts[p2] * 8,
+					 hipMemcpyDeviceToHost));
+			GG_HIP(hipMemcpy(part_v[p2].data() + old,
+					 sv + curs[p2],
+					 counts[p2] * 8,
+					 hipMemcpyDeviceToHost));
+		}
+	}
+	(void) hipFree(dk);
+	(void) hipFree(dv);
+	(void) hipFree(sk);
+	(void) hipFree(sv);
+	(void) hipFree(dcnt);
+	if (st != GG_OK)
+		return st;
+
+	/* reload + aggregate partition by partition */
+	int64_t ng_total = 0;
+
+	for (uint64_t p2 = 0; p2 < P; p2++)
+	{
+		if (part_k[p2].empty())
+			continue;
+		int64_t ng = 0;
+
+		st = gg_engine_hash_groupby_i64(
+			part_k[p2].data(), part_v[p2].data(),
+			(int64_t) part_k[p2].size(), out_keys + ng_total,
+			out_sums + ng_total, out_counts + ng_total,
+			cap - ng_total, &ng);
+		if (st != GG_OK)
+			return st;
+		ng_total += ng;
+		part_k[p2].clear();
+		part_k[p2].shrink_to_fit();
+		part_v[p2].clear();
+		part_v[p2].shrink_to_fit();
+	}
+
+	/* per-partition results are key-sorted; merge to global order */
+	{
+		std::vector<size_t> idx(ng_total);
+		std::vector<int64_t> mk(out_keys, out_keys + ng_total);
+		std::vector<int64_t> ms(out_sums, out_sums + ng_total);
+		std::vector<int64_t> mc(out_counts, out_counts + ng_total);
+
+		for (int64_t i = 0; i < ng_total; i++)
+			idx[i] = (size_t) i;
+		std::sort(idx.begin(), idx.end(),
+			  [&](size_t a2, size_t b2)
+			  { return mk[a2] < mk[b2]; });
+		for (int64_t i = 0; i < ng_total; i++)
+		{
+			out_keys[i] = mk[idx[i]];
+			out_sums[i] = ms[idx[i]];
+			out_counts[i] = mc[idx[i]];
+		}
+	}
+	*out_ngroups = ng_total;
+	*out_npartitions = (int32_t) P;
+	return GG_OK;
+}
 
 extern "C" gg_status
 gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,

@@ -1853,6 +1853,60 @@ launch_gen_supplier(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
 
 #define GB_EMPTY 0x8000000000000000ull	/* INT64_MIN as the empty slot */
 
+/* hash-range partitioning for the spill tier (execHHashagg.c:1350
+ * spill_hash_table semantics): partition id = TOP bits of the key
+ * hash, decoupled from the group table's slot index (low bits) */
+__global__ void
+k_gb_part_count(const int64_t *__restrict__ keys, int64_t n, int shift,
+		unsigned long long *__restrict__ counts)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		atomicAdd(&counts[gg_hashint8(keys[i]) >> shift], 1ull);
+}
+
+__global__ void
+k_gb_part_scatter(const int64_t *__restrict__ keys,
+		  const int64_t *__restrict__ vals, int64_t n, int shift,
+		  unsigned long long *__restrict__ cursors,
+		  int64_t *__restrict__ out_k, int64_t *__restrict__ out_v)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		uint32_t p = gg_hashint8(keys[i]) >> shift;
+		unsigned long long at = atomicAdd(&cursors[p], 1ull);
+
+		out_k[at] = keys[i];
+		out_v[at] = vals[i];
+	}
+}
+
+hipError_t
+launch_gb_part_count(hipStream_t s, const int64_t *keys, int64_t n,
+		     int shift, unsigned long long *counts)
+{
+	hipLaunchKernelGGL(k_gb_part_count, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, keys, n, shift, counts);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_gb_part_scatter(hipStream_t s, const int64_t *keys,
+		       const int64_t *vals, int64_t n, int shift,
+		       unsigned long long *cursors, int64_t *out_k,
+		       int64_t *out_v)
+{
+	hipLaunchKernelGGL(k_gb_part_scatter, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, keys, vals, n, shift,
+			   cursors, out_k, out_v);
+	return hipGetLastError();
+}
+
 __global__ void
 k_groupby_build(const int64_t *__restrict__ keys,
 		const int64_t *__restrict__ vals, int64_t n,

@@ -125,6 +125,12 @@ def _load():
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p,
         ctypes.c_int64, ctypes.POINTER(ctypes.c_int64),
         ctypes.POINTER(ctypes.c_int64)]
+    lib.gg_engine_hash_groupby_i64_spill.restype = ctypes.c_int
+    lib.gg_engine_hash_groupby_i64_spill.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+        ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32)]
     lib.gg_engine_motion_chunkify.restype = ctypes.c_int
     lib.gg_engine_motion_chunkify.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_int32, ctypes.c_int,
@@ -427,6 +433,30 @@ class Engine:
         vals = [bytes(pool[offs[i]:offs[i] + lens[i]])
                 for i in range(n.value)]
         return vals, nulls[:n.value].copy()
+
+    @staticmethod
+    def hash_groupby_spill(keys, vals, budget_bytes):
+        """Spill-tier group-by: partitions staged to host when the
+        input exceeds budget_bytes of device memory; returns
+        (keys, sums, counts, npartitions)."""
+        import numpy as np
+        keys = np.ascontiguousarray(keys, np.int64)
+        vals = np.ascontiguousarray(vals, np.int64)
+        n = len(keys)
+        ok = np.zeros(n + 1, np.int64)
+        os_ = np.zeros(n + 1, np.int64)
+        oc = np.zeros(n + 1, np.int64)
+        ng = I64()
+        np_ = ctypes.c_int32()
+        _check(lib().gg_engine_hash_groupby_i64_spill(
+            keys.ctypes.data_as(ctypes.c_void_p),
+            vals.ctypes.data_as(ctypes.c_void_p), n, budget_bytes,
+            ok.ctypes.data_as(ctypes.c_void_p),
+            os_.ctypes.data_as(ctypes.c_void_p),
+            oc.ctypes.data_as(ctypes.c_void_p), n + 1,
+            ctypes.byref(ng), ctypes.byref(np_)), "groupby_spill")
+        g = ng.value
+        return ok[:g], os_[:g], oc[:g], np_.value
 
     @staticmethod
     def motion_chunkify(tuples, max_chunk=8192, append_eos=True):

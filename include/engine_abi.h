@@ -323,6 +323,22 @@ gg_status gg_engine_hash_groupby_i64(const int64_t *keys,
 				     int64_t *out_counts, int64_t cap,
 				     int64_t *out_ngroups);
 
+/* Spill-tier variant: inputs larger than budget_bytes of device
+ * memory are hash-range partitioned on the GPU, staged in host
+ * memory, and aggregated partition by partition (the reference's
+ * spill_hash_table/agg_hash_reload semantics,
+ * execHHashagg.c:1350/:1852).  Results identical to the in-memory
+ * path; out_npartitions reports the fan-out (1 = no spill). */
+gg_status gg_engine_hash_groupby_i64_spill(const int64_t *keys,
+					   const int64_t *vals, int64_t n,
+					   int64_t budget_bytes,
+					   int64_t *out_keys,
+					   int64_t *out_sums,
+					   int64_t *out_counts,
+					   int64_t cap,
+					   int64_t *out_ngroups,
+					   int32_t *out_npartitions);
+
 /* General ORDER BY operator: stable LSB radix sort of (u64 key, u64
  * payload) pairs on the GPU (nodeSort.c:48 / tuplesort.c semantics for
  * unbounded sorts; the LIMIT-k case uses histogram select instead,
