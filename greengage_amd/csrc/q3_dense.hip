@@ -563,7 +563,7 @@ k_dn_q3_stats(const unsigned long long *__restrict__ pay,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < dense_len; i += stride)
 	{
-		unsigned long long r = rev[i];
+		unsigned long long r = __builtin_nontemporal_load(&rev[i]);
 
 		if (!r)
 			continue;
