@@ -24,6 +24,8 @@
 
 #include "../../include/gg_pg_hash.h"
 #include "../../include/gg_checksum.h"
+#include <cstdlib>
+
 #include "engine_internal.h"
 
 namespace gg
@@ -327,6 +329,25 @@ void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 		atomicAdd(join_rows, joined);
 }
 
+static int
+dn_grid_env(int64_t n, const char *env)
+{
+	const char *v = getenv(env);
+
+	if (v && *v)
+	{
+		int g = atoi(v);
+
+		if (g >= 1 && g <= 65535)
+		{
+			int64_t need = (n + DN_THREADS - 1) / DN_THREADS;
+
+			return (int) (need < g ? (need < 1 ? 1 : need) : g);
+		}
+	}
+	return dn_grid(n);
+}
+
 hipError_t
 launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 const int32_t *shipdate, const int64_t *price,
@@ -335,7 +356,7 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 int64_t dense_len, unsigned long long *bloom,
 			 uint64_t bwords, unsigned long long *join_rows)
 {
-	hipLaunchKernelGGL(k_dn_probe_lineitem, dim3(dn_grid(n)),
+	hipLaunchKernelGGL(k_dn_probe_lineitem, dim3(dn_grid_env(n, "GG_Q3_PROBE_GRID")),
 			   dim3(DN_THREADS), 0, s, okey, shipdate, price,
 			   disc, n, cutoff, pay, rev, dense_len, bloom,
 			   bwords, join_rows);
@@ -544,7 +565,8 @@ launch_dn_probe_lineitem_q5_u8(hipStream_t s, const int64_t *okey,
 			       unsigned long long *acc,
 			       unsigned long long *join_rows)
 {
-	hipLaunchKernelGGL(k_dn_probe_lineitem_q5_u8, dim3(dn_grid(n)),
+	hipLaunchKernelGGL(k_dn_probe_lineitem_q5_u8,
+			   dim3(dn_grid_env(n, "GG_Q5_PROBE_GRID")),
 			   dim3(DN_THREADS), 0, s, okey, skey, price, disc,
 			   n, pay8, dense_len, supp_dense, supp_dlen, acc,
 			   join_rows);
