@@ -140,8 +140,8 @@ dn_bit_test(const unsigned long long *__restrict__ bm, int64_t key)
 }
 
 /* Q3 orders build: date filter + customer membership → pay store */
-__global__ void
-k_dn_build_orders(const int64_t *__restrict__ okey,
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_build_orders(const int64_t *__restrict__ okey,
 		  const int64_t *__restrict__ ckey,
 		  const int32_t *__restrict__ odate,
 		  const int32_t *__restrict__ prio, int64_t n, int32_t cutoff,
@@ -291,8 +291,11 @@ launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-/* Q3 lineitem probe against the dense orders map */
-__global__ __launch_bounds__(DN_THREADS, 2)
+/* Q3 lineitem probe against the dense orders map.  launch_bounds
+ * occupancy 8 blocks/CU (32 waves): the probe moves ~7.5 GB per launch
+ * (PMC) but at 2-block occupancy ran latency-bound at 3.6 TB/s —
+ * more resident waves hide the NT-load latency. */
+__global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 			 const int32_t *__restrict__ shipdate,
 			 const int64_t *__restrict__ price,
@@ -369,8 +372,8 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
  * random probe traffic than the u64 map, and it doubles as its own
  * membership filter, so the probe needs no Bloom reads at all. */
 
-__global__ void
-k_dn_build_orders_q5_u8(const int64_t *__restrict__ okey,
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_build_orders_q5_u8(const int64_t *__restrict__ okey,
 			const int64_t *__restrict__ ckey,
 			const int32_t *__restrict__ odate, int64_t n,
 			int32_t date_lo, int32_t date_hi,
@@ -481,8 +484,8 @@ launch_dn_insert_orders_q5_u8(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-__global__ void
-k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 			  const int64_t *__restrict__ skey,
 			  const int64_t *__restrict__ price,
 			  const int64_t *__restrict__ disc, int64_t n,
@@ -574,8 +577,8 @@ launch_dn_probe_lineitem_q5_u8(hipStream_t s, const int64_t *okey,
 }
 
 /* stats (+max) over the dense group arrays (orderkey = index) */
-__global__ void
-k_dn_q3_stats(const unsigned long long *__restrict__ pay,
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
 	      const unsigned long long *__restrict__ rev, int64_t dense_len,
 	      unsigned long long *out5)
 {
