@@ -291,10 +291,11 @@ launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-/* Q3 lineitem probe against the dense orders map.  launch_bounds
- * occupancy 8 blocks/CU (32 waves): the probe moves ~7.5 GB per launch
- * (PMC) but at 2-block occupancy ran latency-bound at 3.6 TB/s —
- * more resident waves hide the NT-load latency. */
+/* Q3 lineitem probe against the dense orders map.  PMC shows ~7.5 GB
+ * moved per launch at ~3.6 TB/s effective — below the 6.3 TB/s
+ * streaming limit because of the random-access components; occupancy
+ * hints (2 vs 8 blocks/CU) measured identical, so the limiter is the
+ * access pattern, not resident-wave count. */
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 			 const int32_t *__restrict__ shipdate,
