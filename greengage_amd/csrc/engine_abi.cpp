@@ -2023,7 +2023,34 @@ orders_done:;
 	{
 		Timed tm(e.stream);
 
-		if (ord_dlen)
+		if (ord_dlen && getenv("GG_Q5_TWOPASS"))
+		{
+			/* two-pass: compact order-matching rows (streams
+			 * okey + u8 map only), then gather the wide
+			 * columns for the ~3% survivors */
+			unsigned long long *comp = (unsigned long long *)
+				p->sget("q5.comp",
+					(size_t) (li->nrows + 64) * 8);
+			unsigned long long *ctr2 = (unsigned long long *)
+				p->sget("q5.ctr2", 8);
+
+			if (!comp || !ctr2)
+				return fail(GG_ENOMEM, "q5 compact");
+			GG_HIP(hipMemsetAsync(ctr2, 0, 8, e.stream));
+			GG_HIP(launch_dn_q5_compact(e.stream, l_ok,
+						    li->nrows, ordd_pay8,
+						    ord_dlen, comp, ctr2));
+			unsigned long long m = 0;
+
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_TRY(read_counter(ctr2, &m));
+			GG_HIP(launch_dn_q5_gather(e.stream, comp,
+						   (int64_t) m, l_sk, l_pc,
+						   l_dc, supp_dense,
+						   supp_dense_len, acc,
+						   ctr));
+		}
+		else if (ord_dlen)
 			GG_HIP(launch_dn_probe_lineitem_q5_u8(
 				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,
 				ordd_pay8, ord_dlen, supp_dense,

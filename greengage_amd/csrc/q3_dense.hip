@@ -577,6 +577,133 @@ launch_dn_probe_lineitem_q5_u8(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
+/* Two-pass Q5 probe: pass A streams only okey + the u8 orders map and
+ * compacts the ~3%% of rows with a matching order (wave-aggregated
+ * append); pass B gathers skey/price/disc just for those rows (their
+ * indices are ascending, so the gathers still coalesce by line) and
+ * finishes the supplier-nation join + aggregation.  Avoids fetching
+ * the wide columns' lines for non-matching lanes. */
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_q5_compact(const int64_t *__restrict__ okey, int64_t n,
+		     const uint8_t *__restrict__ pay8, int64_t dense_len,
+		     unsigned long long *__restrict__ out /* idx<<8|nat */ ,
+		     unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	const int64_t n_up = ((n + stride - 1) / stride) * stride;
+
+	for (int64_t i = i0; i < n_up; i += stride)
+	{
+		bool take = false;
+		unsigned nat = 255;
+
+		if (i < n)
+		{
+			int64_t k = dn_ld64(&okey[i]);
+
+			if (k >= 0 && k < dense_len)
+			{
+				nat = pay8[k];
+				take = nat != 255;
+			}
+		}
+		unsigned long long at = dn_wave_append(out_count, take);
+
+		if (take)
+			out[at] = ((unsigned long long) i << 8) | nat;
+	}
+}
+
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_q5_gather(const unsigned long long *__restrict__ comp,
+		    int64_t m, const int64_t *__restrict__ skey,
+		    const int64_t *__restrict__ price,
+		    const int64_t *__restrict__ disc,
+		    const uint8_t *__restrict__ supp_dense,
+		    int64_t supp_dlen,
+		    unsigned long long *__restrict__ acc /* [25][3] */ ,
+		    unsigned long long *join_rows)
+{
+	__shared__ unsigned long long lds[25][2];
+
+	for (int i = threadIdx.x; i < 50; i += blockDim.x)
+		((unsigned long long *) lds)[i] = 0;
+	__syncthreads();
+
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long joined = 0;
+
+	for (int64_t j = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     j < m; j += stride)
+	{
+		unsigned long long e = comp[j];
+		int64_t i = (int64_t) (e >> 8);
+		unsigned onat = (unsigned) (e & 0xFF);
+		int64_t sk = skey[i];
+
+		if (sk < 0 || sk >= supp_dlen)
+			continue;
+		unsigned snat = supp_dense[sk];
+
+		if (snat != onat)
+			continue;
+		joined++;
+		{
+			unsigned long long rev4 = (unsigned long long)
+				(price[i] * (100 - disc[i]));
+
+			atomicAdd(&lds[onat][0], 1ull);
+			atomicAdd(&lds[onat][1], rev4);
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+		joined += __shfl_down(joined, off, 64);
+	if ((threadIdx.x & 63) == 0 && joined)
+		atomicAdd(join_rows, joined);
+	__syncthreads();
+	for (int nat = threadIdx.x; nat < 25; nat += blockDim.x)
+	{
+		unsigned long long c = lds[nat][0];
+		unsigned long long r = lds[nat][1];
+
+		if (!c)
+			continue;
+		atomicAdd(&acc[nat * 3 + 0], c);
+		{
+			unsigned long long old =
+				atomicAdd(&acc[nat * 3 + 1], r);
+
+			if (old + r < old)
+				atomicAdd(&acc[nat * 3 + 2], 1ull);
+		}
+	}
+}
+
+hipError_t
+launch_dn_q5_compact(hipStream_t s, const int64_t *okey, int64_t n,
+		     const uint8_t *pay8, int64_t dense_len,
+		     unsigned long long *out, unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_dn_q5_compact, dim3(dn_grid(n)),
+			   dim3(DN_THREADS), 0, s, okey, n, pay8, dense_len,
+			   out, out_count);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_dn_q5_gather(hipStream_t s, const unsigned long long *comp,
+		    int64_t m, const int64_t *skey, const int64_t *price,
+		    const int64_t *disc, const uint8_t *supp_dense,
+		    int64_t supp_dlen, unsigned long long *acc,
+		    unsigned long long *join_rows)
+{
+	hipLaunchKernelGGL(k_dn_q5_gather, dim3(dn_grid(m)),
+			   dim3(DN_THREADS), 0, s, comp, m, skey, price,
+			   disc, supp_dense, supp_dlen, acc, join_rows);
+	return hipGetLastError();
+}
+
 /* stats (+max) over the dense group arrays (orderkey = index) */
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
