@@ -2023,34 +2023,63 @@ orders_done:;
 	{
 		Timed tm(e.stream);
 
-		if (ord_dlen && getenv("GG_Q5_TWOPASS"))
+		bool twopass_done = false;
+
+		if (ord_dlen && !getenv("GG_Q5_ONEPASS"))
 		{
-			/* two-pass: compact order-matching rows (streams
-			 * okey + u8 map only), then gather the wide
-			 * columns for the ~3% survivors */
+			/* two-pass: compact order-matching rows into
+			 * per-block regions (streams okey + the u8 map
+			 * only), then gather the wide columns for the ~3%
+			 * survivors.  Region sized for ~8x the expected
+			 * selectivity; overflow flips a flag and we fall
+			 * back to the fused one-pass probe. */
+			/* must equal the launcher's dn_grid(n) */
+			int grid = (int) ((li->nrows + 255) / 256);
+
+			if (grid > 2048)
+				grid = 2048;
+			if (grid < 1)
+				grid = 1;
+			int64_t per_block =
+				(li->nrows + (int64_t) grid * 256 - 1) /
+				((int64_t) grid * 256) * 256;
+			int64_t region = per_block / 4 + 4096;
 			unsigned long long *comp = (unsigned long long *)
 				p->sget("q5.comp",
-					(size_t) (li->nrows + 64) * 8);
-			unsigned long long *ctr2 = (unsigned long long *)
-				p->sget("q5.ctr2", 8);
+					(size_t) grid * region * 8);
+			unsigned long long *ccnt = (unsigned long long *)
+				p->sget("q5.ccnt", (size_t) grid * 8 + 8);
 
-			if (!comp || !ctr2)
+			if (!comp || !ccnt)
 				return fail(GG_ENOMEM, "q5 compact");
-			GG_HIP(hipMemsetAsync(ctr2, 0, 8, e.stream));
+			unsigned long long *ovf = ccnt + grid;
+
+			GG_HIP(hipMemsetAsync(ovf, 0, 8, e.stream));
 			GG_HIP(launch_dn_q5_compact(e.stream, l_ok,
 						    li->nrows, ordd_pay8,
-						    ord_dlen, comp, ctr2));
-			unsigned long long m = 0;
+						    ord_dlen, region, comp,
+						    ccnt));
+			GG_HIP(launch_dn_q5_gather(e.stream, comp, ccnt,
+						   region, grid, l_sk, l_pc,
+						   l_dc, supp_dense,
+						   supp_dense_len, acc, ctr,
+						   ovf));
+			unsigned long long hovf = 0;
 
 			GG_HIP(hipStreamSynchronize(e.stream));
-			GG_TRY(read_counter(ctr2, &m));
-			GG_HIP(launch_dn_q5_gather(e.stream, comp,
-						   (int64_t) m, l_sk, l_pc,
-						   l_dc, supp_dense,
-						   supp_dense_len, acc,
-						   ctr));
+			GG_TRY(read_counter(ovf, &hovf));
+			if (!hovf)
+				twopass_done = true;
+			else
+			{	/* region overflow: redo with the fused
+				 * one-pass probe */
+				GG_HIP(hipMemsetAsync(acc, 0,
+						      GG_NNATIONS * 3 * 8,
+						      e.stream));
+				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+			}
 		}
-		else if (ord_dlen)
+		if (ord_dlen && !twopass_done)
 			GG_HIP(launch_dn_probe_lineitem_q5_u8(
 				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,
 				ordd_pay8, ord_dlen, supp_dense,

@@ -441,15 +441,19 @@ hipError_t launch_dn_build_orders_q5_u8(hipStream_t s, const int64_t *okey,
 					 unsigned long long *match_count);
 hipError_t launch_dn_q5_compact(hipStream_t s, const int64_t *okey,
 				int64_t n, const uint8_t *pay8,
-				int64_t dense_len, unsigned long long *out,
-				unsigned long long *out_count);
+				int64_t dense_len, int64_t region,
+				unsigned long long *out,
+				unsigned long long *counts);
 hipError_t launch_dn_q5_gather(hipStream_t s,
-			       const unsigned long long *comp, int64_t m,
+			       const unsigned long long *comp,
+			       const unsigned long long *counts,
+			       int64_t region, int64_t nregions,
 			       const int64_t *skey, const int64_t *price,
 			       const int64_t *disc,
 			       const uint8_t *supp_dense, int64_t supp_dlen,
 			       unsigned long long *acc,
-			       unsigned long long *join_rows);
+			       unsigned long long *join_rows,
+			       unsigned long long *overflow);
 hipError_t launch_dn_insert_orders_q5_u8(hipStream_t s, const int64_t *okey,
 					 const int64_t *rownat, int64_t n,
 					 uint8_t *pay8, int64_t dense_len);

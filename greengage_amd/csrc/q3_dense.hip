@@ -577,53 +577,65 @@ launch_dn_probe_lineitem_q5_u8(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-/* Two-pass Q5 probe: pass A streams only okey + the u8 orders map and
- * compacts the ~3%% of rows with a matching order (wave-aggregated
- * append); pass B gathers skey/price/disc just for those rows (their
- * indices are ascending, so the gathers still coalesce by line) and
- * finishes the supplier-nation join + aggregation.  Avoids fetching
- * the wide columns' lines for non-matching lanes. */
+/* Two-pass Q5 probe.  Pass A streams only okey + the u8 orders map
+ * and compacts the ~3%% of rows with a matching order into PER-BLOCK
+ * regions of the output (positions from an LDS counter — no global
+ * atomics: a single returning global counter saturates at ~88
+ * atomics/us, which made a wave-append version 45x slower).  Pass B
+ * gathers skey/price/disc only for the survivors (indices ascending
+ * within a region, so lines still coalesce) and finishes the
+ * supplier-nation join + aggregation; gather block b walks compact
+ * region b.
+ */
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q5_compact(const int64_t *__restrict__ okey, int64_t n,
 		     const uint8_t *__restrict__ pay8, int64_t dense_len,
+		     int64_t region, /* per-block output capacity */
 		     unsigned long long *__restrict__ out /* idx<<8|nat */ ,
-		     unsigned long long *out_count)
+		     unsigned long long *__restrict__ counts /* [grid] */ )
 {
+	__shared__ unsigned long long lcnt;
+
+	if (threadIdx.x == 0)
+		lcnt = 0;
+	__syncthreads();
+
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	const int64_t n_up = ((n + stride - 1) / stride) * stride;
+	unsigned long long *mine = out + (int64_t) blockIdx.x * region;
 
-	for (int64_t i = i0; i < n_up; i += stride)
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
 	{
-		bool take = false;
-		unsigned nat = 255;
+		int64_t k = dn_ld64(&okey[i]);
 
-		if (i < n)
-		{
-			int64_t k = dn_ld64(&okey[i]);
+		if (k < 0 || k >= dense_len)
+			continue;
+		unsigned nat = pay8[k];
 
-			if (k >= 0 && k < dense_len)
-			{
-				nat = pay8[k];
-				take = nat != 255;
-			}
-		}
-		unsigned long long at = dn_wave_append(out_count, take);
+		if (nat == 255)
+			continue;
+		unsigned long long at = atomicAdd(&lcnt, 1ull);
 
-		if (take)
-			out[at] = ((unsigned long long) i << 8) | nat;
+		if (at < (unsigned long long) region)
+			mine[at] = ((unsigned long long) i << 8) | nat;
 	}
+	__syncthreads();
+	if (threadIdx.x == 0)
+		counts[blockIdx.x] = lcnt;
 }
 
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q5_gather(const unsigned long long *__restrict__ comp,
-		    int64_t m, const int64_t *__restrict__ skey,
+		    const unsigned long long *__restrict__ counts,
+		    int64_t region, int64_t nregions,
+		    const int64_t *__restrict__ skey,
 		    const int64_t *__restrict__ price,
 		    const int64_t *__restrict__ disc,
 		    const uint8_t *__restrict__ supp_dense,
 		    int64_t supp_dlen,
 		    unsigned long long *__restrict__ acc /* [25][3] */ ,
-		    unsigned long long *join_rows)
+		    unsigned long long *join_rows,
+		    unsigned long long *overflow)
 {
 	__shared__ unsigned long long lds[25][2];
 
@@ -631,30 +643,43 @@ void k_dn_q5_gather(const unsigned long long *__restrict__ comp,
 		((unsigned long long *) lds)[i] = 0;
 	__syncthreads();
 
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long joined = 0;
 
-	for (int64_t j = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     j < m; j += stride)
+	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
 	{
-		unsigned long long e = comp[j];
-		int64_t i = (int64_t) (e >> 8);
-		unsigned onat = (unsigned) (e & 0xFF);
-		int64_t sk = skey[i];
+		unsigned long long m = counts[b];
 
-		if (sk < 0 || sk >= supp_dlen)
-			continue;
-		unsigned snat = supp_dense[sk];
-
-		if (snat != onat)
-			continue;
-		joined++;
+		if (m > (unsigned long long) region)
 		{
-			unsigned long long rev4 = (unsigned long long)
-				(price[i] * (100 - disc[i]));
+			if (threadIdx.x == 0)
+				atomicOr(overflow, 1ull);
+			m = (unsigned long long) region;
+		}
+		const unsigned long long *seg = comp + b * region;
 
-			atomicAdd(&lds[onat][0], 1ull);
-			atomicAdd(&lds[onat][1], rev4);
+		for (unsigned long long j = threadIdx.x; j < m;
+		     j += blockDim.x)
+		{
+			unsigned long long e = seg[j];
+			int64_t i = (int64_t) (e >> 8);
+			unsigned onat = (unsigned) (e & 0xFF);
+			int64_t sk = skey[i];
+
+			if (sk < 0 || sk >= supp_dlen)
+				continue;
+			unsigned snat = supp_dense[sk];
+
+			if (snat != onat)
+				continue;
+			joined++;
+			{
+				unsigned long long rev4 =
+					(unsigned long long)
+					(price[i] * (100 - disc[i]));
+
+				atomicAdd(&lds[onat][0], 1ull);
+				atomicAdd(&lds[onat][1], rev4);
+			}
 		}
 	}
 	for (int off = 32; off; off >>= 1)
@@ -682,25 +707,28 @@ void k_dn_q5_gather(const unsigned long long *__restrict__ comp,
 
 hipError_t
 launch_dn_q5_compact(hipStream_t s, const int64_t *okey, int64_t n,
-		     const uint8_t *pay8, int64_t dense_len,
-		     unsigned long long *out, unsigned long long *out_count)
+		     const uint8_t *pay8, int64_t dense_len, int64_t region,
+		     unsigned long long *out, unsigned long long *counts)
 {
 	hipLaunchKernelGGL(k_dn_q5_compact, dim3(dn_grid(n)),
 			   dim3(DN_THREADS), 0, s, okey, n, pay8, dense_len,
-			   out, out_count);
+			   region, out, counts);
 	return hipGetLastError();
 }
 
 hipError_t
 launch_dn_q5_gather(hipStream_t s, const unsigned long long *comp,
-		    int64_t m, const int64_t *skey, const int64_t *price,
-		    const int64_t *disc, const uint8_t *supp_dense,
-		    int64_t supp_dlen, unsigned long long *acc,
-		    unsigned long long *join_rows)
+		    const unsigned long long *counts, int64_t region,
+		    int64_t nregions, const int64_t *skey,
+		    const int64_t *price, const int64_t *disc,
+		    const uint8_t *supp_dense, int64_t supp_dlen,
+		    unsigned long long *acc, unsigned long long *join_rows,
+		    unsigned long long *overflow)
 {
-	hipLaunchKernelGGL(k_dn_q5_gather, dim3(dn_grid(m)),
-			   dim3(DN_THREADS), 0, s, comp, m, skey, price,
-			   disc, supp_dense, supp_dlen, acc, join_rows);
+	hipLaunchKernelGGL(k_dn_q5_gather, dim3((int) nregions),
+			   dim3(DN_THREADS), 0, s, comp, counts, region,
+			   nregions, skey, price, disc, supp_dense,
+			   supp_dlen, acc, join_rows, overflow);
 	return hipGetLastError();
 }
 
