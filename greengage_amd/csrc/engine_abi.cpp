@@ -2079,11 +2079,15 @@ orders_done:;
 				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 			}
 		}
-		if (ord_dlen && !twopass_done)
-			GG_HIP(launch_dn_probe_lineitem_q5_u8(
-				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,
-				ordd_pay8, ord_dlen, supp_dense,
-				supp_dense_len, acc, ctr));
+		if (ord_dlen)
+		{
+			if (!twopass_done)
+				GG_HIP(launch_dn_probe_lineitem_q5_u8(
+					e.stream, l_ok, l_sk, l_pc, l_dc,
+					li->nrows, ordd_pay8, ord_dlen,
+					supp_dense, supp_dense_len, acc,
+					ctr));
+		}
 		else
 			GG_HIP(launch_probe_lineitem_q5(
 				e.stream, l_ok, l_sk, l_pc, l_dc, li->nrows,
