@@ -2025,14 +2025,17 @@ orders_done:;
 
 		bool twopass_done = false;
 
-		if (ord_dlen && !getenv("GG_Q5_ONEPASS"))
+		if (ord_dlen && getenv("GG_Q5_TWOPASS"))
 		{
-			/* two-pass: compact order-matching rows into
-			 * per-block regions (streams okey + the u8 map
-			 * only), then gather the wide columns for the ~3%
-			 * survivors.  Region sized for ~8x the expected
-			 * selectivity; overflow flips a flag and we fall
-			 * back to the fused one-pass probe. */
+			/* EXPERIMENT (measured equal to the fused probe at
+			 * SF100: 2.38 vs 2.32 ms — the compact pass's 4.8 GB
+			 * stream costs what the skipped wide-column lines
+			 * saved; kept behind GG_Q5_TWOPASS as evidence):
+			 * compact order-matching rows into per-block
+			 * regions (LDS counters, no global returning
+			 * atomics), then gather the wide columns for the
+			 * ~3% survivors.  Overflow falls back to the fused
+			 * probe. */
 			/* must equal the launcher's dn_grid(n) */
 			int grid = (int) ((li->nrows + 255) / 256);
 
