@@ -381,3 +381,44 @@ def test_ao_largecontent_layout():
     assert ((fw0 >> 28) & 7) == 1
     assert ((fw0 >> 27) & 1) == 0  # no firstRowNum
     assert ((fw0 >> 10) & 0x3FFF) == 0  # rowCount 0
+
+
+def test_ao_parser_corruption_fuzz():
+    """Robustness: random byte flips / truncations of AO streams must
+    produce a clean engine error or (if the flip misses anything
+    load-bearing) succeed — never crash.  Host-side parse only.
+    The reference treats corrupt blocks the same way (header/block
+    checksum ereports in cdbappendonlystorageformat.c)."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    lib = _engine_lib()
+    rng = np.random.default_rng(77)
+    vals = rng.integers(-2**30, 2**30, 4000).astype(np.int64)
+    framed, nb = pyoracle.dsb_encode(vals, np.zeros(4000, np.uint8), 8,
+                                     2, 0, 0, blocksz=2048)
+    streams = {
+        "small": pyoracle.ao_wrap(framed),
+        "compressed": pyoracle.ao_wrap_compressed(framed, 1, 6),
+        "large": pyoracle.ao_wrap_large(framed, frag_size=1024),
+        "bulkdense": pyoracle.ao_wrap_bulkdense(framed, comptype=2,
+                                                complevel=3),
+    }
+    comptypes = {"small": 0, "compressed": 1, "large": 0,
+                 "bulkdense": 2}
+    for name, ao in streams.items():
+        ct = comptypes[name]
+        for trial in range(300):
+            bad = ao.copy()
+            mode = trial % 3
+            if mode == 0:      # flip a random byte
+                pos = int(rng.integers(0, len(bad)))
+                bad[pos] ^= int(rng.integers(1, 256))
+            elif mode == 1:    # truncate
+                bad = bad[:int(rng.integers(0, len(bad)))]
+            else:              # flip a random bit in a header word
+                pos = int(rng.integers(0, min(32, len(bad))))
+                bad[pos] ^= 1 << int(rng.integers(0, 8))
+            # must not crash; either clean error or GG_ESTATE
+            # ("engine not initialized", CPU) after a clean parse
+            rc = _decode_ao_rc(lib, bad, comptype=ct)
+            assert rc != 0  # no GPU here: success is impossible
