@@ -79,3 +79,25 @@ def test_chunk_error_paths():
     # truncated
     with pytest.raises(RuntimeError):
         E.motion_dechunkify(chunks[:len(chunks) - 3])
+
+
+def test_wire_fuzz_no_crash():
+    """Random corruption of chunk streams and MemTuple streams must
+    error cleanly or succeed, never crash (host-side paths)."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference memtuple codec not built")
+    E = _engine()
+    stream, _ = _mt_stream(20)
+    chunks = E.motion_chunkify(stream)
+    rng = np.random.default_rng(78)
+    for trial in range(400):
+        bad = chunks.copy()
+        if trial % 2 == 0:
+            pos = int(rng.integers(0, len(bad)))
+            bad[pos] ^= int(rng.integers(1, 256))
+        else:
+            bad = bad[:int(rng.integers(0, len(bad)))]
+        try:
+            out, eos = E.motion_dechunkify(bad)
+        except RuntimeError:
+            continue  # clean error
