@@ -3432,6 +3432,210 @@ gg_engine_motion_dechunkify(const uint8_t *chunks, int64_t chunks_len,
 /* ---------------- general hash group-by (ABI surface) ---------------- */
 
 /*
+ * Spill-tier hash JOIN (the nodeHash.c:713 batching
+ * side): both inputs are hash-range partitioned on the GPU in chunks
+ * and staged in host memory (the reference writes batch files), then
+ * each partition pair is built+probed on the GPU.  Build keys must be
+ * unique (PK-side build, as in the engine's pipeline joins); output
+ * is (probe_row_index, build_val) for every probe row with a match.
+ */
+extern "C" gg_status
+gg_engine_hash_join_i64_spill(const int64_t *build_keys,
+			      const int64_t *build_vals, int64_t nb,
+			      const int64_t *probe_keys, int64_t np,
+			      int64_t budget_bytes, int64_t *out_probe_idx,
+			      int64_t *out_vals, int64_t cap,
+			      int64_t *out_nmatch, int32_t *out_npartitions)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!build_keys || !build_vals || nb < 0 || !probe_keys || np < 0 ||
+	    !out_probe_idx || !out_vals || !out_nmatch || !out_npartitions ||
+	    budget_bytes < (1 << 20))
+		return fail(GG_EINVAL, "bad join_spill args");
+	*out_nmatch = 0;
+
+	/* footprint ≈ build pairs + 2x table + probe pairs + outputs */
+	int64_t need = nb * 8 * 6 + np * 8 * 4;
+	uint64_t P = 1;
+	int shift = 32;
+
+	if (need > budget_bytes)
+	{
+		P = next_pow2((uint64_t) (need / budget_bytes) + 1);
+		if (P > 4096)
+			return fail(GG_EINVAL,
+				    "join needs %llu partitions (>4096)",
+				    (unsigned long long) P);
+		for (uint64_t t = P; t > 1; t >>= 1)
+			shift--;
+	}
+	*out_npartitions = (int32_t) P;
+
+	/* stage both sides into host partitions (P==1: single "partition"
+	 * passthrough still goes through the same code path) */
+	int64_t chunk = budget_bytes / (8 * 4);
+
+	if (chunk < 1024)
+		chunk = 1024;
+	std::vector<std::vector<int64_t>> bk(P), bv(P), pk(P), pi(P);
+	int64_t *dk = nullptr, *dv = nullptr, *sk = nullptr, *sv = nullptr;
+	unsigned long long *dcnt = nullptr;
+	gg_status st = GG_OK;
+
+	GG_HIP(hipMalloc((void **) &dk, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &dv, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &sk, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &sv, (size_t) chunk * 8));
+	GG_HIP(hipMalloc((void **) &dcnt, (P + 1) * 8));
+
+	std::vector<unsigned long long> counts(P), curs(P + 1);
+	auto stage = [&](const int64_t *keys, const int64_t *vals,
+			 int64_t n, bool idx_side,
+			 std::vector<std::vector<int64_t>> &out_k,
+			 std::vector<std::vector<int64_t>> &out_v)
+		-> gg_status
+	{
+		for (int64_t base = 0; base < n; base += chunk)
+		{
+			int64_t m = (n - base < chunk) ? n - base : chunk;
+
+			GG_HIP(hipMemcpy(dk, keys + base, (size_t) m * 8,
+					 hipMemcpyHostToDevice));
+			if (!idx_side)
+				GG_HIP(hipMemcpy(dv, vals + base,
+						 (size_t) m * 8,
+						 hipMemcpyHostToDevice));
+			GG_HIP(hipMemsetAsync(dcnt, 0, P * 8, e.stream));
+			GG_HIP(launch_gb_part_count(e.stream, dk, m, shift,
+						    dcnt));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_HIP(hipMemcpy(counts.data(), dcnt, P * 8,
+					 hipMemcpyDeviceToHost));
+			curs[0] = 0;
+			for (uint64_t p2 = 0; p2 < P; p2++)
+				curs[p2 + 1] = curs[p2] + counts[p2];
+			GG_HIP(hipMemcpy(dcnt, curs.data(), P * 8,
+					 hipMemcpyHostToDevice));
+			if (idx_side)
+				GG_HIP(launch_gb_part_scatter_idx(
+					e.stream, dk, m, base, shift, dcnt,
+					sk, sv));
+			else
+				GG_HIP(launch_gb_part_scatter(
+					e.stream, dk, dv, m, shift, dcnt,
+					sk, sv));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			for (uint64_t p2 = 0; p2 < P; p2++)
+			{
+				if (!counts[p2])
+					continue;
+				size_t old = out_k[p2].size();
+
+				out_k[p2].resize(old + counts[p2]);
+				out_v[p2].resize(old + counts[p2]);
+				GG_HIP(hipMemcpy(out_k[p2].data() + old,
+						 sk + curs[p2],
+						 counts[p2] * 8,
+						 hipMemcpyDeviceToHost));
+				GG_HIP(hipMemcpy(out_v[p2].data() + old,
+						 sv + curs[p2],
+						 counts[p2] * 8,
+						 hipMemcpyDeviceToHost));
+			}
+		}
+		return GG_OK;
+	};
+	st = stage(build_keys, build_vals, nb, false, bk, bv);
+	if (st == GG_OK)
+		st = stage(probe_keys, nullptr, np, true, pk, pi);
+	(void) hipFree(dk);
+	(void) hipFree(dv);
+	(void) hipFree(sk);
+	(void) hipFree(sv);
+	(void) hipFree(dcnt);
+	if (st != GG_OK)
+		return st;
+
+	/* per-partition build + probe */
+	int64_t total = 0;
+
+	for (uint64_t p2 = 0; p2 < P && st == GG_OK; p2++)
+	{
+		int64_t b_n = (int64_t) bk[p2].size();
+		int64_t p_n = (int64_t) pk[p2].size();
+
+		if (b_n == 0 || p_n == 0)
+			continue;
+		uint64_t nslots = next_pow2(2 * (uint64_t) b_n);
+		int64_t *d_bk = nullptr, *d_bv = nullptr, *d_pk = nullptr,
+			*d_pi = nullptr, *d_oi = nullptr, *d_ov = nullptr;
+		unsigned long long *tk = nullptr, *tv = nullptr,
+			*ctr = nullptr;
+
+		GG_HIP(hipMalloc((void **) &d_bk, (size_t) b_n * 8));
+		GG_HIP(hipMalloc((void **) &d_bv, (size_t) b_n * 8));
+		GG_HIP(hipMalloc((void **) &d_pk, (size_t) p_n * 8));
+		GG_HIP(hipMalloc((void **) &d_pi, (size_t) p_n * 8));
+		GG_HIP(hipMalloc((void **) &d_oi, (size_t) p_n * 8));
+		GG_HIP(hipMalloc((void **) &d_ov, (size_t) p_n * 8));
+		GG_HIP(hipMalloc((void **) &tk, nslots * 8));
+		GG_HIP(hipMalloc((void **) &tv, nslots * 8));
+		GG_HIP(hipMalloc((void **) &ctr, 8));
+		GG_HIP(hipMemcpy(d_bk, bk[p2].data(), (size_t) b_n * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(hipMemcpy(d_bv, bv[p2].data(), (size_t) b_n * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(hipMemcpy(d_pk, pk[p2].data(), (size_t) p_n * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(hipMemcpy(d_pi, pi[p2].data(), (size_t) p_n * 8,
+				 hipMemcpyHostToDevice));
+		GG_HIP(hipMemsetAsync(tk, 0, nslots * 8, e.stream));
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_sj_build(e.stream, d_bk, d_bv, b_n, tk, tv,
+				       nslots));
+		GG_HIP(launch_sj_probe(e.stream, d_pk, d_pi, p_n, tk, tv,
+				       nslots, d_oi, d_ov, ctr));
+		GG_HIP(hipStreamSynchronize(e.stream));
+
+		unsigned long long nm = 0;
+
+		GG_HIP(hipMemcpy(&nm, ctr, 8, hipMemcpyDeviceToHost));
+		if (total + (int64_t) nm > cap)
+			st = fail(GG_EINVAL, "join cap %lld < %lld",
+				  (long long) cap,
+				  (long long) (total + (int64_t) nm));
+		else if (nm)
+		{
+			GG_HIP(hipMemcpy(out_probe_idx + total, d_oi,
+					 nm * 8, hipMemcpyDeviceToHost));
+			GG_HIP(hipMemcpy(out_vals + total, d_ov, nm * 8,
+					 hipMemcpyDeviceToHost));
+			total += (int64_t) nm;
+		}
+		(void) hipFree(d_bk);
+		(void) hipFree(d_bv);
+		(void) hipFree(d_pk);
+		(void) hipFree(d_pi);
+		(void) hipFree(d_oi);
+		(void) hipFree(d_ov);
+		(void) hipFree(tk);
+		(void) hipFree(tv);
+		(void) hipFree(ctr);
+		bk[p2].clear(); bk[p2].shrink_to_fit();
+		bv[p2].clear(); bv[p2].shrink_to_fit();
+		pk[p2].clear(); pk[p2].shrink_to_fit();
+		pi[p2].clear(); pi[p2].shrink_to_fit();
+	}
+	if (st != GG_OK)
+		return st;
+	*out_nmatch = total;
+	return GG_OK;
+}
+
+/*
  * Spill-tier hash group-by: when the input exceeds the
  * device budget, hash-range-partition it on the GPU in chunks, stage
  * the partitions in HOST memory (the spill arena — the reference

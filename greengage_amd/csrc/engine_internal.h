@@ -349,6 +349,20 @@ hipError_t launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n,
 
 hipError_t launch_fill_u64(hipStream_t s, unsigned long long *p,
 			   uint64_t n, unsigned long long v);
+hipError_t launch_sj_build(hipStream_t s, const int64_t *keys,
+			   const int64_t *vals, int64_t n,
+			   unsigned long long *tkeys,
+			   unsigned long long *tvals, uint64_t nslots);
+hipError_t launch_sj_probe(hipStream_t s, const int64_t *keys,
+			   const int64_t *idxs, int64_t n,
+			   const unsigned long long *tkeys,
+			   const unsigned long long *tvals, uint64_t nslots,
+			   int64_t *out_idx, int64_t *out_val,
+			   unsigned long long *out_count);
+hipError_t launch_gb_part_scatter_idx(hipStream_t s, const int64_t *keys,
+				      int64_t n, int64_t base, int shift,
+				      unsigned long long *cursors,
+				      int64_t *out_k, int64_t *out_v);
 hipError_t launch_gb_part_count(hipStream_t s, const int64_t *keys,
 				int64_t n, int shift,
 				unsigned long long *counts);

@@ -1853,6 +1853,147 @@ launch_gen_supplier(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
 
 #define GB_EMPTY 0x8000000000000000ull	/* INT64_MIN as the empty slot */
 
+/* generic i64-key hash-join kernels for the spill tier
+ * (ExecHashJoin batching semantics, nodeHash.c:713): CAS-insert build
+ * (key 0 reserved as empty, like the pipeline tables), probe emits
+ * (probe_row_index, build_val) matches via wave-aggregated append */
+__global__ void
+k_sj_build(const int64_t *__restrict__ keys,
+	   const int64_t *__restrict__ vals, int64_t n,
+	   unsigned long long *__restrict__ tkeys,
+	   unsigned long long *__restrict__ tvals, uint64_t nslots)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t kk = keys[i];
+		uint64_t pos = (uint64_t) gg_hashint8(kk) & (nslots - 1);
+
+		for (;;)
+		{
+			unsigned long long prev =
+				atomicCAS(&tkeys[pos], 0ull,
+					  (unsigned long long) kk);
+
+			if (prev == 0 || prev == (unsigned long long) kk)
+			{
+				tvals[pos] = (unsigned long long) vals[i];
+				break;
+			}
+			pos = (pos + 1) & (nslots - 1);
+		}
+	}
+}
+
+__global__ void
+k_sj_probe(const int64_t *__restrict__ keys,
+	   const int64_t *__restrict__ idxs, int64_t n,
+	   const unsigned long long *__restrict__ tkeys,
+	   const unsigned long long *__restrict__ tvals, uint64_t nslots,
+	   int64_t *__restrict__ out_idx, int64_t *__restrict__ out_val,
+	   unsigned long long *out_count)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	const int64_t n_up = ((n + stride - 1) / stride) * stride;
+
+	for (int64_t i = i0; i < n_up; i += stride)
+	{
+		bool hit = false;
+		unsigned long long v = 0;
+		int64_t kk = 0;
+
+		if (i < n)
+		{
+			kk = keys[i];
+			uint64_t pos = (uint64_t) gg_hashint8(kk) &
+				(nslots - 1);
+
+			for (;;)
+			{
+				unsigned long long cur = tkeys[pos];
+
+				if (cur == (unsigned long long) kk)
+				{
+					v = tvals[pos];
+					hit = true;
+					break;
+				}
+				if (cur == 0)
+					break;
+				pos = (pos + 1) & (nslots - 1);
+			}
+		}
+		unsigned long long at = wave_append(out_count, hit);
+
+		if (hit)
+		{
+			out_idx[at] = idxs[i];
+			out_val[at] = (int64_t) v;
+		}
+	}
+}
+
+hipError_t
+launch_sj_build(hipStream_t s, const int64_t *keys, const int64_t *vals,
+		int64_t n, unsigned long long *tkeys,
+		unsigned long long *tvals, uint64_t nslots)
+{
+	hipLaunchKernelGGL(k_sj_build, dim3(grid_for(n)), dim3(THREADS), 0,
+			   s, keys, vals, n, tkeys, tvals, nslots);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_sj_probe(hipStream_t s, const int64_t *keys, const int64_t *idxs,
+		int64_t n, const unsigned long long *tkeys,
+		const unsigned long long *tvals, uint64_t nslots,
+		int64_t *out_idx, int64_t *out_val,
+		unsigned long long *out_count)
+{
+	hipLaunchKernelGGL(k_sj_probe, dim3(grid_for(n)), dim3(THREADS), 0,
+			   s, keys, idxs, n, tkeys, tvals, nslots, out_idx,
+			   out_val, out_count);
+	return hipGetLastError();
+}
+
+/* variant of the partition scatter that carries the ORIGINAL row index
+ * (base + i) as the value — used to keep probe-row identity through
+ * the spill partitioning without materializing an iota on the host */
+__global__ void
+k_gb_part_scatter_idx(const int64_t *__restrict__ keys, int64_t n,
+		      int64_t base, int shift,
+		      unsigned long long *__restrict__ cursors,
+		      int64_t *__restrict__ out_k,
+		      int64_t *__restrict__ out_v)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		uint32_t p = gg_hashint8(keys[i]) >> shift;
+		unsigned long long at = atomicAdd(&cursors[p], 1ull);
+
+		out_k[at] = keys[i];
+		out_v[at] = base + i;
+	}
+}
+
+hipError_t
+launch_gb_part_scatter_idx(hipStream_t s, const int64_t *keys, int64_t n,
+			   int64_t base, int shift,
+			   unsigned long long *cursors, int64_t *out_k,
+			   int64_t *out_v)
+{
+	hipLaunchKernelGGL(k_gb_part_scatter_idx, dim3(grid_for(n)),
+			   dim3(THREADS), 0, s, keys, n, base, shift,
+			   cursors, out_k, out_v);
+	return hipGetLastError();
+}
+
 /* hash-range partitioning for the spill tier (execHHashagg.c:1350
  * spill_hash_table semantics): partition id = TOP bits of the key
  * hash, decoupled from the group table's slot index (low bits) */

@@ -125,6 +125,12 @@ def _load():
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p,
         ctypes.c_int64, ctypes.POINTER(ctypes.c_int64),
         ctypes.POINTER(ctypes.c_int64)]
+    lib.gg_engine_hash_join_i64_spill.restype = ctypes.c_int
+    lib.gg_engine_hash_join_i64_spill.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32)]
     lib.gg_engine_hash_groupby_i64_spill.restype = ctypes.c_int
     lib.gg_engine_hash_groupby_i64_spill.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
@@ -433,6 +439,29 @@ class Engine:
         vals = [bytes(pool[offs[i]:offs[i] + lens[i]])
                 for i in range(n.value)]
         return vals, nulls[:n.value].copy()
+
+    @staticmethod
+    def hash_join_spill(build_keys, build_vals, probe_keys,
+                        budget_bytes):
+        """Spill-tier hash join (unique build keys); returns
+        (probe_idx, matched_vals, npartitions)."""
+        import numpy as np
+        bk = np.ascontiguousarray(build_keys, np.int64)
+        bv = np.ascontiguousarray(build_vals, np.int64)
+        pk = np.ascontiguousarray(probe_keys, np.int64)
+        cap = len(pk) + 1
+        oi = np.zeros(cap, np.int64)
+        ov = np.zeros(cap, np.int64)
+        nm = I64()
+        np_ = ctypes.c_int32()
+        _check(lib().gg_engine_hash_join_i64_spill(
+            bk.ctypes.data_as(ctypes.c_void_p),
+            bv.ctypes.data_as(ctypes.c_void_p), len(bk),
+            pk.ctypes.data_as(ctypes.c_void_p), len(pk), budget_bytes,
+            oi.ctypes.data_as(ctypes.c_void_p),
+            ov.ctypes.data_as(ctypes.c_void_p), cap, ctypes.byref(nm),
+            ctypes.byref(np_)), "join_spill")
+        return oi[:nm.value], ov[:nm.value], np_.value
 
     @staticmethod
     def hash_groupby_spill(keys, vals, budget_bytes):

@@ -329,6 +329,21 @@ gg_status gg_engine_hash_groupby_i64(const int64_t *keys,
  * spill_hash_table/agg_hash_reload semantics,
  * execHHashagg.c:1350/:1852).  Results identical to the in-memory
  * path; out_npartitions reports the fan-out (1 = no spill). */
+/* Spill-tier hash JOIN (nodeHash.c:713 batching): both sides hash-
+ * range partitioned on the GPU, staged in host memory, each partition
+ * pair built+probed on the GPU.  Build keys must be unique (PK-side
+ * build).  Key 0 is reserved (table empty sentinel), as in the
+ * pipeline joins.  Output: (probe_row_index, build_val) per match. */
+gg_status gg_engine_hash_join_i64_spill(const int64_t *build_keys,
+					const int64_t *build_vals,
+					int64_t nb,
+					const int64_t *probe_keys,
+					int64_t np, int64_t budget_bytes,
+					int64_t *out_probe_idx,
+					int64_t *out_vals, int64_t cap,
+					int64_t *out_nmatch,
+					int32_t *out_npartitions);
+
 gg_status gg_engine_hash_groupby_i64_spill(const int64_t *keys,
 					   const int64_t *vals, int64_t n,
 					   int64_t budget_bytes,
