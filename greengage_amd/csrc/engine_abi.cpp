@@ -3477,7 +3477,10 @@ gg_engine_hash_join_i64_spill(const int64_t *build_keys,
 	/* stage both sides into host partitions (P==1: single "partition"
 	 * passthrough still goes through the same code path) */
 	int64_t chunk = budget_bytes / (8 * 4);
+	int64_t biggest = nb > np ? nb : np;
 
+	if (chunk > biggest)
+		chunk = biggest;	/* never allocate beyond the input */
 	if (chunk < 1024)
 		chunk = 1024;
 	std::vector<std::vector<int64_t>> bk(P), bv(P), pk(P), pi(P);
@@ -3575,46 +3578,67 @@ gg_engine_hash_join_i64_spill(const int64_t *build_keys,
 		unsigned long long *tk = nullptr, *tv = nullptr,
 			*ctr = nullptr;
 
-		GG_HIP(hipMalloc((void **) &d_bk, (size_t) b_n * 8));
-		GG_HIP(hipMalloc((void **) &d_bv, (size_t) b_n * 8));
-		GG_HIP(hipMalloc((void **) &d_pk, (size_t) p_n * 8));
-		GG_HIP(hipMalloc((void **) &d_pi, (size_t) p_n * 8));
-		GG_HIP(hipMalloc((void **) &d_oi, (size_t) p_n * 8));
-		GG_HIP(hipMalloc((void **) &d_ov, (size_t) p_n * 8));
-		GG_HIP(hipMalloc((void **) &tk, nslots * 8));
-		GG_HIP(hipMalloc((void **) &tv, nslots * 8));
-		GG_HIP(hipMalloc((void **) &ctr, 8));
-		GG_HIP(hipMemcpy(d_bk, bk[p2].data(), (size_t) b_n * 8,
-				 hipMemcpyHostToDevice));
-		GG_HIP(hipMemcpy(d_bv, bv[p2].data(), (size_t) b_n * 8,
-				 hipMemcpyHostToDevice));
-		GG_HIP(hipMemcpy(d_pk, pk[p2].data(), (size_t) p_n * 8,
-				 hipMemcpyHostToDevice));
-		GG_HIP(hipMemcpy(d_pi, pi[p2].data(), (size_t) p_n * 8,
-				 hipMemcpyHostToDevice));
-		GG_HIP(hipMemsetAsync(tk, 0, nslots * 8, e.stream));
-		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-		GG_HIP(launch_sj_build(e.stream, d_bk, d_bv, b_n, tk, tv,
-				       nslots));
-		GG_HIP(launch_sj_probe(e.stream, d_pk, d_pi, p_n, tk, tv,
-				       nslots, d_oi, d_ov, ctr));
-		GG_HIP(hipStreamSynchronize(e.stream));
+		/* no-leak variant of GG_HIP: errors fall through to the
+		 * frees below instead of returning */
+#define GG_HIP_BRK(x) \
+		{ hipError_t e_ = (x); \
+		  if (st == GG_OK && e_ != hipSuccess) \
+			st = fail(GG_EGPU, "join_spill: %s", \
+				  hipGetErrorString(e_)); }
+		GG_HIP_BRK(hipMalloc((void **) &d_bk, (size_t) b_n * 8));
+		GG_HIP_BRK(hipMalloc((void **) &d_bv, (size_t) b_n * 8));
+		GG_HIP_BRK(hipMalloc((void **) &d_pk, (size_t) p_n * 8));
+		GG_HIP_BRK(hipMalloc((void **) &d_pi, (size_t) p_n * 8));
+		GG_HIP_BRK(hipMalloc((void **) &d_oi, (size_t) p_n * 8));
+		GG_HIP_BRK(hipMalloc((void **) &d_ov, (size_t) p_n * 8));
+		GG_HIP_BRK(hipMalloc((void **) &tk, nslots * 8));
+		GG_HIP_BRK(hipMalloc((void **) &tv, nslots * 8));
+		GG_HIP_BRK(hipMalloc((void **) &ctr, 8));
+		if (st == GG_OK)
+		{
+			GG_HIP_BRK(hipMemcpy(d_bk, bk[p2].data(),
+					     (size_t) b_n * 8,
+					     hipMemcpyHostToDevice));
+			GG_HIP_BRK(hipMemcpy(d_bv, bv[p2].data(),
+					     (size_t) b_n * 8,
+					     hipMemcpyHostToDevice));
+			GG_HIP_BRK(hipMemcpy(d_pk, pk[p2].data(),
+					     (size_t) p_n * 8,
+					     hipMemcpyHostToDevice));
+			GG_HIP_BRK(hipMemcpy(d_pi, pi[p2].data(),
+					     (size_t) p_n * 8,
+					     hipMemcpyHostToDevice));
+			GG_HIP_BRK(hipMemsetAsync(tk, 0, nslots * 8,
+						  e.stream));
+			GG_HIP_BRK(hipMemsetAsync(ctr, 0, 8, e.stream));
+			GG_HIP_BRK(launch_sj_build(e.stream, d_bk, d_bv,
+						   b_n, tk, tv, nslots));
+			GG_HIP_BRK(launch_sj_probe(e.stream, d_pk, d_pi,
+						   p_n, tk, tv, nslots,
+						   d_oi, d_ov, ctr));
+			GG_HIP_BRK(hipStreamSynchronize(e.stream));
+		}
 
 		unsigned long long nm = 0;
 
-		GG_HIP(hipMemcpy(&nm, ctr, 8, hipMemcpyDeviceToHost));
-		if (total + (int64_t) nm > cap)
+		if (st == GG_OK)
+			GG_HIP_BRK(hipMemcpy(&nm, ctr, 8,
+					     hipMemcpyDeviceToHost));
+		if (st == GG_OK && total + (int64_t) nm > cap)
 			st = fail(GG_EINVAL, "join cap %lld < %lld",
 				  (long long) cap,
 				  (long long) (total + (int64_t) nm));
-		else if (nm)
+		else if (st == GG_OK && nm)
 		{
-			GG_HIP(hipMemcpy(out_probe_idx + total, d_oi,
-					 nm * 8, hipMemcpyDeviceToHost));
-			GG_HIP(hipMemcpy(out_vals + total, d_ov, nm * 8,
-					 hipMemcpyDeviceToHost));
+			GG_HIP_BRK(hipMemcpy(out_probe_idx + total, d_oi,
+					     nm * 8,
+					     hipMemcpyDeviceToHost));
+			GG_HIP_BRK(hipMemcpy(out_vals + total, d_ov,
+					     nm * 8,
+					     hipMemcpyDeviceToHost));
 			total += (int64_t) nm;
 		}
+#undef GG_HIP_BRK
 		(void) hipFree(d_bk);
 		(void) hipFree(d_bv);
 		(void) hipFree(d_pk);
@@ -3702,28 +3726,38 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 
 	std::vector<unsigned long long> counts(P), curs(P + 1);
 
+	/* no-leak variant: errors fall through to the frees below */
+#define GG_HIP_GBS(x) \
+	{ hipError_t e_ = (x); \
+	  if (st == GG_OK && e_ != hipSuccess) \
+		st = fail(GG_EGPU, "groupby_spill: %s", \
+			  hipGetErrorString(e_)); \
+	  if (st != GG_OK) break; }
 	for (int64_t base = 0; st == GG_OK && base < n; base += chunk)
 	{
 		int64_t m = (n - base < chunk) ? n - base : chunk;
 
-		GG_HIP(hipMemcpy(dk, keys + base, (size_t) m * 8,
-				 hipMemcpyHostToDevice));
-		GG_HIP(hipMemcpy(dv, vals + base, (size_t) m * 8,
-				 hipMemcpyHostToDevice));
-		GG_HIP(hipMemsetAsync(dcnt, 0, P * 8, e.stream));
-		GG_HIP(launch_gb_part_count(e.stream, dk, m, shift, dcnt));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_HIP(hipMemcpy(counts.data(), dcnt, P * 8,
-				 hipMemcpyDeviceToHost));
+		GG_HIP_GBS(hipMemcpy(dk, keys + base, (size_t) m * 8,
+				     hipMemcpyHostToDevice));
+		GG_HIP_GBS(hipMemcpy(dv, vals + base, (size_t) m * 8,
+				     hipMemcpyHostToDevice));
+		GG_HIP_GBS(hipMemsetAsync(dcnt, 0, P * 8, e.stream));
+		GG_HIP_GBS(launch_gb_part_count(e.stream, dk, m, shift,
+						dcnt));
+		GG_HIP_GBS(hipStreamSynchronize(e.stream));
+		GG_HIP_GBS(hipMemcpy(counts.data(), dcnt, P * 8,
+				     hipMemcpyDeviceToHost));
 		curs[0] = 0;
 		for (uint64_t p2 = 0; p2 < P; p2++)
 			curs[p2 + 1] = curs[p2] + counts[p2];
-		GG_HIP(hipMemcpy(dcnt, curs.data(), P * 8,
-				 hipMemcpyHostToDevice));
-		GG_HIP(launch_gb_part_scatter(e.stream, dk, dv, m, shift,
-					      dcnt, sk, sv));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		for (uint64_t p2 = 0; p2 < P; p2++)
+		GG_HIP_GBS(hipMemcpy(dcnt, curs.data(), P * 8,
+				     hipMemcpyHostToDevice));
+		GG_HIP_GBS(launch_gb_part_scatter(e.stream, dk, dv, m,
+						  shift, dcnt, sk, sv));
+		GG_HIP_GBS(hipStreamSynchronize(e.stream));
+		bool bad = false;
+
+		for (uint64_t p2 = 0; p2 < P && !bad; p2++)
 		{
 			if (!counts[p2])
 				continue;
@@ -3731,16 +3765,23 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 
 			part_k[p2].resize(old + counts[p2]);
 			part_v[p2].resize(old + counts[p2]);
-			GG_HIP(hipMemcpy(part_k[p2].data() + old,
-					 sk + curs[p2],
-					 counts[p2] * 8,
-					 hipMemcpyDeviceToHost));
-			GG_HIP(hipMemcpy(part_v[p2].data() + old,
-					 sv + curs[p2],
-					 counts[p2] * 8,
-					 hipMemcpyDeviceToHost));
+			hipError_t e1 = hipMemcpy(part_k[p2].data() + old,
+						  sk + curs[p2],
+						  counts[p2] * 8,
+						  hipMemcpyDeviceToHost);
+			hipError_t e2 = hipMemcpy(part_v[p2].data() + old,
+						  sv + curs[p2],
+						  counts[p2] * 8,
+						  hipMemcpyDeviceToHost);
+
+			if (e1 != hipSuccess || e2 != hipSuccess)
+			{
+				st = fail(GG_EGPU, "groupby_spill stage");
+				bad = true;
+			}
 		}
 	}
+#undef GG_HIP_GBS
 	(void) hipFree(dk);
 	(void) hipFree(dv);
 	(void) hipFree(sk);
