@@ -1974,7 +1974,7 @@ k_gb_part_scatter_idx(const int64_t *__restrict__ keys, int64_t n,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		uint32_t p = gg_hashint8(keys[i]) >> shift;
+		uint64_t p = (uint64_t) gg_hashint8(keys[i]) >> shift;
 		unsigned long long at = atomicAdd(&cursors[p], 1ull);
 
 		out_k[at] = keys[i];
@@ -2005,7 +2005,9 @@ k_gb_part_count(const int64_t *__restrict__ keys, int64_t n, int shift,
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
-		atomicAdd(&counts[gg_hashint8(keys[i]) >> shift], 1ull);
+		/* 64-bit shift: shift==32 (single partition) must yield 0 */
+		atomicAdd(&counts[(uint64_t) gg_hashint8(keys[i]) >> shift],
+			  1ull);
 }
 
 __global__ void
@@ -2019,7 +2021,7 @@ k_gb_part_scatter(const int64_t *__restrict__ keys,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
-		uint32_t p = gg_hashint8(keys[i]) >> shift;
+		uint64_t p = (uint64_t) gg_hashint8(keys[i]) >> shift;
 		unsigned long long at = atomicAdd(&cursors[p], 1ull);
 
 		out_k[at] = keys[i];
