@@ -133,3 +133,39 @@ def test_join_spill_matches_numpy(eng):
         order = np.argsort(oi, kind="stable")
         assert np.array_equal(oi[order], exp_idx)
         assert np.array_equal(ov[order], exp_vals)
+
+
+def test_spill_paths_at_scale(eng):
+    """Larger staging volumes: 50M-row group-by and a 20M-row probe
+    join, both forced through many host-staged partitions."""
+    import numpy as np
+    from greengage_amd.engine import Engine as E
+    rng = np.random.default_rng(63)
+    n = 50_000_000
+    keys = rng.integers(0, 1_000_000, n).astype(np.int64)
+    vals = rng.integers(-100, 100, n).astype(np.int64)
+    k1, s1, c1, nparts = E.hash_groupby_spill(keys, vals, 64 << 20)
+    assert nparts >= 32, nparts
+    sums = np.zeros(1_000_000, np.int64)
+    cnts = np.zeros(1_000_000, np.int64)
+    np.add.at(sums, keys, vals)
+    np.add.at(cnts, keys, 1)
+    present = np.nonzero(cnts)[0]
+    assert np.array_equal(k1, present)
+    assert np.array_equal(s1, sums[present])
+    assert np.array_equal(c1, cnts[present])
+
+    nb, np_rows = 2_000_000, 20_000_000
+    bk = rng.permutation(np.arange(1, nb + 1)).astype(np.int64)
+    bv = rng.integers(0, 10**9, nb).astype(np.int64)
+    pk = rng.integers(1, 3 * nb, np_rows).astype(np.int64)
+    oi, ov, jparts = E.hash_join_spill(bk, bv, pk, 32 << 20)
+    assert jparts >= 8, jparts
+    val_of = np.zeros(3 * nb + 1, np.int64)
+    present2 = np.zeros(3 * nb + 1, bool)
+    val_of[bk] = bv
+    present2[bk] = True
+    exp_idx = np.nonzero(present2[pk])[0]
+    order = np.argsort(oi, kind="stable")
+    assert np.array_equal(oi[order], exp_idx)
+    assert np.array_equal(ov[order], val_of[pk[exp_idx]])
