@@ -2858,6 +2858,188 @@ gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
 	return st;
 }
 
+/* --------- text dictionary encode (ABI surface, §8 a4) --------- */
+
+/*
+ * Dictionary-encode a categorical text column (the arrow-style shape
+ * gg_engine_aocs_decode_text outputs) into per-row int32 codes + a
+ * lexicographically sorted dictionary: codes are deterministic across
+ * shards/runs, which the multi-GPU exchange requires.  NULL rows get
+ * code -1.  max_dict bounds the cardinality (error beyond — this is
+ * for categorical columns like c_mktsegment, not free text).
+ */
+extern "C" gg_status
+gg_engine_text_dict_encode(const uint8_t *pool, const uint64_t *offs,
+			   const uint32_t *lens, const uint8_t *nulls,
+			   int64_t n, int32_t max_dict, int32_t *out_codes,
+			   uint8_t *dict_bytes, int64_t dict_cap,
+			   int64_t *dict_offs /* max_dict+1 */ ,
+			   int32_t *out_ndict)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!pool || !offs || !lens || n < 0 || max_dict < 1 ||
+	    max_dict > (1 << 20) || !out_codes || !dict_bytes ||
+	    !dict_offs || !out_ndict)
+		return fail(GG_EINVAL, "bad text_dict args");
+	*out_ndict = 0;
+	if (n == 0)
+	{
+		dict_offs[0] = 0;
+		return GG_OK;
+	}
+
+	int64_t pool_len = 0;
+
+	for (int64_t i = 0; i < n; i++)
+		if ((!nulls || !nulls[i]) &&
+		    (int64_t) (offs[i] + lens[i]) > pool_len)
+			pool_len = (int64_t) (offs[i] + lens[i]);
+
+	uint64_t nslots = next_pow2(4 * (uint64_t) max_dict);
+	gg_status st = GG_OK;
+	uint8_t *d_pool = nullptr, *d_nulls = nullptr;
+	unsigned long long *d_offs = nullptr, *d_slots = nullptr,
+		*d_err = nullptr;
+	uint32_t *d_lens = nullptr, *d_rowslot = nullptr;
+	int32_t *d_map = nullptr, *d_codes = nullptr;
+
+	GG_HIP(hipMalloc((void **) &d_pool, (size_t) (pool_len + 1)));
+	GG_HIP(hipMalloc((void **) &d_offs, (size_t) n * 8));
+	GG_HIP(hipMalloc((void **) &d_lens, (size_t) n * 4));
+	GG_HIP(hipMalloc((void **) &d_rowslot, (size_t) n * 4));
+	GG_HIP(hipMalloc((void **) &d_codes, (size_t) n * 4));
+	GG_HIP(hipMalloc((void **) &d_slots, nslots * 8));
+	GG_HIP(hipMalloc((void **) &d_map, nslots * 4));
+	GG_HIP(hipMalloc((void **) &d_err, 8));
+	if (pool_len)
+		GG_HIP(hipMemcpy(d_pool, pool, (size_t) pool_len,
+				 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_offs, offs, (size_t) n * 8,
+			 hipMemcpyHostToDevice));
+	GG_HIP(hipMemcpy(d_lens, lens, (size_t) n * 4,
+			 hipMemcpyHostToDevice));
+	if (nulls)
+	{
+		GG_HIP(hipMalloc((void **) &d_nulls, (size_t) n));
+		GG_HIP(hipMemcpy(d_nulls, nulls, (size_t) n,
+				 hipMemcpyHostToDevice));
+	}
+	GG_HIP(hipMemsetAsync(d_slots, 0, nslots * 8, e.stream));
+	GG_HIP(hipMemsetAsync(d_err, 0, 8, e.stream));
+	{
+		hipError_t he = launch_td_insert(
+			e.stream, d_pool, d_offs, d_lens, d_nulls, n,
+			d_slots, nslots, d_rowslot, d_err);
+
+		if (he != hipSuccess)
+			st = fail(GG_EGPU, "td_insert: %s",
+				  hipGetErrorString(he));
+	}
+	std::vector<unsigned long long> hslots;
+
+	if (st == GG_OK)
+	{
+		unsigned long long herr = 0;
+
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(&herr, d_err, 8, hipMemcpyDeviceToHost));
+		if (herr)
+			st = fail(GG_EINVAL, "dictionary overflow "
+				  "(> %d distinct values)", (int) max_dict);
+	}
+	if (st == GG_OK)
+	{
+		hslots.resize(nslots);
+		GG_HIP(hipMemcpy(hslots.data(), d_slots, nslots * 8,
+				 hipMemcpyDeviceToHost));
+
+		/* unique strings -> lexicographic ids (bytewise compare,
+		 * text_lt with C collation) */
+		struct Ent
+		{
+			uint64_t slot;
+			int64_t row;
+		};
+		std::vector<Ent> ents;
+
+		for (uint64_t sl = 0; sl < nslots; sl++)
+			if (hslots[sl])
+				ents.push_back({sl,
+						(int64_t) hslots[sl] - 1});
+		if ((int32_t) ents.size() > max_dict)
+			st = fail(GG_EINVAL, "dictionary overflow "
+				  "(%zu > %d)", ents.size(),
+				  (int) max_dict);
+		if (st == GG_OK)
+		{
+			std::sort(ents.begin(), ents.end(),
+				  [&](const Ent & a, const Ent & b)
+				  {
+				  const uint8_t *sa = pool + offs[a.row];
+				  const uint8_t *sb = pool + offs[b.row];
+				  uint32_t la = lens[a.row],
+				  lb = lens[b.row];
+				  uint32_t m = la < lb ? la : lb;
+
+				  for (uint32_t z = 0; z < m; z++)
+				  if (sa[z] != sb[z])
+				  return sa[z] < sb[z];
+				  return la < lb;
+				  });
+			std::vector<int32_t> slot_to_id(nslots, -1);
+			int64_t dpos = 0;
+
+			dict_offs[0] = 0;
+			for (size_t id = 0; id < ents.size(); id++)
+			{
+				slot_to_id[ents[id].slot] = (int32_t) id;
+				uint32_t l = lens[ents[id].row];
+
+				if (dpos + l > dict_cap)
+					return fail(GG_EINVAL,
+						    "dict_cap too small");
+				std::memcpy(dict_bytes + dpos,
+					    pool + offs[ents[id].row], l);
+				dpos += l;
+				dict_offs[id + 1] = dpos;
+			}
+			*out_ndict = (int32_t) ents.size();
+			GG_HIP(hipMemcpy(d_map, slot_to_id.data(),
+					 nslots * 4,
+					 hipMemcpyHostToDevice));
+			{
+				hipError_t he = launch_td_map(
+					e.stream, d_rowslot, n, d_map,
+					d_codes);
+
+				if (he != hipSuccess)
+					st = fail(GG_EGPU, "td_map: %s",
+						  hipGetErrorString(he));
+			}
+			if (st == GG_OK)
+			{
+				GG_HIP(hipStreamSynchronize(e.stream));
+				GG_HIP(hipMemcpy(out_codes, d_codes,
+						 (size_t) n * 4,
+						 hipMemcpyDeviceToHost));
+			}
+		}
+	}
+	(void) hipFree(d_pool);
+	(void) hipFree(d_offs);
+	(void) hipFree(d_lens);
+	(void) hipFree(d_rowslot);
+	(void) hipFree(d_codes);
+	(void) hipFree(d_slots);
+	(void) hipFree(d_map);
+	(void) hipFree(d_err);
+	(void) hipFree(d_nulls);
+	return st;
+}
+
 /* -------------- MemTuple codec (ABI surface, §8(f)3) -------------- */
 
 /*

@@ -408,6 +408,16 @@ hipError_t launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 				    unsigned long long *bloom,
 				    uint64_t bwords,
 				    unsigned long long *join_rows);
+hipError_t launch_td_insert(hipStream_t s, const uint8_t *pool,
+			    const unsigned long long *offs,
+			    const uint32_t *lens, const uint8_t *nulls,
+			    int64_t n, unsigned long long *slots,
+			    uint64_t nslots, uint32_t *rowslot,
+			    unsigned long long *err);
+hipError_t launch_td_map(hipStream_t s, const uint32_t *rowslot,
+			 int64_t n, const int32_t *slot_to_id,
+			 int32_t *codes);
+
 #define GG_MT_MAX_ATTS 32
 
 /* memtuple.hip — MemTuple codec */

@@ -137,6 +137,12 @@ def _load():
         ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int64,
         ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32)]
+    lib.gg_engine_text_dict_encode.restype = ctypes.c_int
+    lib.gg_engine_text_dict_encode.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int32,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_int32)]
     lib.gg_engine_motion_chunkify.restype = ctypes.c_int
     lib.gg_engine_motion_chunkify.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_int32, ctypes.c_int,
@@ -486,6 +492,43 @@ class Engine:
             ctypes.byref(ng), ctypes.byref(np_)), "groupby_spill")
         g = ng.value
         return ok[:g], os_[:g], oc[:g], np_.value
+
+    @staticmethod
+    def text_dict_encode(values, nulls=None, max_dict=4096):
+        """GPU dictionary-encode a list of bytes values; returns
+        (codes int32, dict list of bytes).  NULLs -> -1."""
+        import numpy as np
+        n = len(values)
+        blob = b"".join(values)
+        offs = np.zeros(n, np.uint64)
+        lens = np.zeros(n, np.uint32)
+        pos = 0
+        for i, v in enumerate(values):
+            offs[i] = pos
+            lens[i] = len(v)
+            pos += len(v)
+        pool = (np.frombuffer(blob, np.uint8).copy() if blob
+                else np.zeros(1, np.uint8))
+        nl = (None if nulls is None
+              else np.ascontiguousarray(nulls, np.uint8))
+        codes = np.zeros(n, np.int32)
+        dcap = pos + 16
+        dbytes = np.zeros(dcap, np.uint8)
+        doffs = np.zeros(max_dict + 1, np.int64)
+        nd = ctypes.c_int32()
+        _check(lib().gg_engine_text_dict_encode(
+            pool.ctypes.data_as(ctypes.c_void_p),
+            offs.ctypes.data_as(ctypes.c_void_p),
+            lens.ctypes.data_as(ctypes.c_void_p),
+            None if nl is None else
+            nl.ctypes.data_as(ctypes.c_void_p), n, max_dict,
+            codes.ctypes.data_as(ctypes.c_void_p),
+            dbytes.ctypes.data_as(ctypes.c_void_p), dcap,
+            doffs.ctypes.data_as(ctypes.c_void_p),
+            ctypes.byref(nd)), "text_dict_encode")
+        d = [bytes(dbytes[doffs[i]:doffs[i + 1]])
+             for i in range(nd.value)]
+        return codes, d
 
     @staticmethod
     def motion_chunkify(tuples, max_chunk=8192, append_eos=True):

@@ -228,6 +228,22 @@ gg_status gg_engine_aocs_decode_text(const uint8_t *stream,
 				     int64_t *out_nrows,
 				     int64_t *out_pool_len);
 
+/* Dictionary-encode a categorical text column (arrow-style inputs,
+ * e.g. gg_engine_aocs_decode_text's output) into int32 codes + a
+ * lexicographically sorted dictionary — deterministic across shards,
+ * as the exchange requires.  NULLs code as -1; cardinality bounded by
+ * max_dict. */
+gg_status gg_engine_text_dict_encode(const uint8_t *pool,
+				     const uint64_t *offs,
+				     const uint32_t *lens,
+				     const uint8_t *nulls, int64_t n,
+				     int32_t max_dict,
+				     int32_t *out_codes,
+				     uint8_t *dict_bytes,
+				     int64_t dict_cap,
+				     int64_t *dict_offs,
+				     int32_t *out_ndict);
+
 /* MemTuple codec (access/common/memtuple.c format, the tuple layout
  * used in executor hash tables and on the Motion wire): bulk GPU
  * conversion between column arrays and MemTuple byte streams.
