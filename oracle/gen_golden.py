@@ -392,6 +392,18 @@ def make_small_inputs():
     # encode segment as the engine's dict code: MACHINERY = GG_MKTSEG_MACHINERY
     arr["c_mktseg"] = np.array([2 if s else 0 for s in seg], np.uint8)
     arr["c_nationkey"] = np.array(cnat, np.uint8)
+    # the RAW text too (blob + offsets), for the text-predicate path
+    segtext = [line.rstrip("\n").split("|")[6] for line in
+               open(os.path.join(REF,
+                                 "src/test/regress/data/customer.csv"))]
+    blob = "".join(segtext).encode()
+    offs = np.zeros(len(segtext) + 1, np.int64)
+    p = 0
+    for i, s in enumerate(segtext):
+        p += len(s)
+        offs[i + 1] = p
+    arr["c_mktseg_text_blob"] = np.frombuffer(blob, np.uint8).copy()
+    arr["c_mktseg_text_offs"] = offs
     sk, snat = [], []
     for line in open(os.path.join(REF, "src/test/regress/data/supplier.csv")):
         f = line.rstrip("\n").split("|")
