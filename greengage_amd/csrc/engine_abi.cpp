@@ -86,6 +86,11 @@ static gg_status dev_counter(unsigned long long **p)
 	return GG_OK;
 }
 
+static gg_status ao_parse_to_framed(const uint8_t *stream,
+				    int64_t stream_len, int checksums,
+				    int ao_version, int comptype,
+				    std::vector<uint8_t> &framed);
+
 static gg_status read_counter(unsigned long long *p, unsigned long long *out)
 {
 	GG_HIP(hipMemcpy(out, p, sizeof(*out), hipMemcpyDeviceToHost));
@@ -240,6 +245,207 @@ gg_engine_register_table(const char *name, const gg_column_desc *cols,
 			}
 		}
 	}
+	e.tables.push_back(t);
+	*out = (gg_table) (e.tables.size() - 1);
+	return GG_OK;
+}
+
+/*
+ * Mount a REAL AO table: one AO segfile byte stream per column, the
+ * way cdbbufferedread.c hands blocks up.  Each column goes through
+ * the full storage layer (headers + CRC32C + zlib/zstd + datum-stream
+ * decode) and lands DIRECTLY in a device-resident column — no host
+ * round trip — then the table registers like any other.  Columns must
+ * be NOT NULL (the hot path's columns are); a NULL anywhere errors.
+ */
+extern "C" gg_status
+gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
+			    int ncols, gg_table *out)
+{
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+	if (!name || !cols || ncols <= 0 || !out)
+		return fail(GG_EINVAL, "bad register_table_ao args");
+
+	Table *t = new Table();
+
+	t->name = name;
+	t->nrows = -1;
+
+	gg_status st = GG_OK;
+
+	for (int ci = 0; ci < ncols && st == GG_OK; ci++)
+	{
+		const gg_ao_column &ac = cols[ci];
+		int datumlen = (ac.type == GG_COL_INT64 ||
+				ac.type == GG_COL_DEC64_S2) ? 8 : 4;
+		int out_width = datumlen;
+		std::vector<uint8_t> framed;
+
+		st = ao_parse_to_framed(ac.stream, ac.stream_len,
+					ac.checksums, ac.ao_version,
+					ac.comptype, framed);
+		if (st != GG_OK)
+			break;
+
+		/* frame descriptors */
+		std::vector<int64_t> offs, out_offs;
+		std::vector<int32_t> sizes, rows;
+		int64_t pos = 0, total_rows = 0;
+
+		while (pos < (int64_t) framed.size())
+		{
+			int32_t sz, rc;
+
+			std::memcpy(&sz, framed.data() + pos, 4);
+			std::memcpy(&rc, framed.data() + pos + 4, 4);
+			offs.push_back(pos + 8);
+			sizes.push_back(sz);
+			rows.push_back(rc);
+			out_offs.push_back(total_rows);
+			total_rows += rc;
+			pos = (pos + 8 + sz + 7) & ~(int64_t) 7;
+		}
+		if (t->nrows < 0)
+			t->nrows = total_rows;
+		else if (t->nrows != total_rows)
+		{
+			st = fail(GG_EINVAL, "column %s has %lld rows, "
+				  "table has %lld", ac.name,
+				  (long long) total_rows,
+				  (long long) t->nrows);
+			break;
+		}
+
+		uint8_t *d_stream = nullptr, *d_nulls = nullptr;
+		int64_t *d_offs = nullptr, *d_oo = nullptr;
+		int32_t *d_sizes = nullptr, *d_rows = nullptr;
+		void *d_vals = nullptr;
+		unsigned long long *d_err = nullptr;
+		size_t nb = offs.size();
+
+#define GG_HIP_AO(x) \
+		{ hipError_t e_ = (x); \
+		  if (st == GG_OK && e_ != hipSuccess) \
+			st = fail(GG_EGPU, "register_table_ao: %s", \
+				  hipGetErrorString(e_)); }
+		GG_HIP_AO(hipMalloc((void **) &d_stream,
+				    framed.size() ? framed.size() : 1));
+		GG_HIP_AO(hipMalloc((void **) &d_offs, (nb + 1) * 8));
+		GG_HIP_AO(hipMalloc((void **) &d_oo, (nb + 1) * 8));
+		GG_HIP_AO(hipMalloc((void **) &d_sizes, (nb + 1) * 4));
+		GG_HIP_AO(hipMalloc((void **) &d_rows, (nb + 1) * 4));
+		GG_HIP_AO(hipMalloc(&d_vals,
+				    (size_t) total_rows * out_width + 1));
+		GG_HIP_AO(hipMalloc((void **) &d_nulls,
+				    (size_t) total_rows + 1));
+		GG_HIP_AO(hipMalloc((void **) &d_err, 8));
+		if (st == GG_OK && nb)
+		{
+			GG_HIP_AO(hipMemcpy(d_stream, framed.data(),
+					    framed.size(),
+					    hipMemcpyHostToDevice));
+			GG_HIP_AO(hipMemcpy(d_offs, offs.data(), nb * 8,
+					    hipMemcpyHostToDevice));
+			GG_HIP_AO(hipMemcpy(d_oo, out_offs.data(), nb * 8,
+					    hipMemcpyHostToDevice));
+			GG_HIP_AO(hipMemcpy(d_sizes, sizes.data(), nb * 4,
+					    hipMemcpyHostToDevice));
+			GG_HIP_AO(hipMemcpy(d_rows, rows.data(), nb * 4,
+					    hipMemcpyHostToDevice));
+			GG_HIP_AO(hipMemset(d_err, 0, 8));
+			GG_HIP_AO(launch_dsb_decode(
+				e.stream, d_stream, d_offs, d_sizes,
+				d_rows, d_oo, (int32_t) nb, ac.dsb_version,
+				datumlen, d_vals, d_nulls, out_width,
+				d_err));
+			GG_HIP_AO(hipStreamSynchronize(e.stream));
+			if (st == GG_OK)
+			{
+				unsigned long long herr = 0;
+
+				GG_HIP_AO(hipMemcpy(&herr, d_err, 8,
+						    hipMemcpyDeviceToHost));
+				if (st == GG_OK && herr)
+					st = fail(GG_EINVAL,
+						  "column %s decode error "
+						  "0x%llx", ac.name, herr);
+			}
+			/* NOT NULL check */
+			if (st == GG_OK && total_rows)
+			{
+				std::vector<uint8_t> hn(total_rows);
+
+				GG_HIP_AO(hipMemcpy(hn.data(), d_nulls,
+						    (size_t) total_rows,
+						    hipMemcpyDeviceToHost));
+				for (int64_t r = 0;
+				     st == GG_OK && r < total_rows; r++)
+					if (hn[r])
+						st = fail(GG_EINVAL,
+							  "column %s has "
+							  "NULL at row %lld",
+							  ac.name,
+							  (long long) r);
+			}
+		}
+		/* attach the decoded column (narrowing for char1) */
+		if (st == GG_OK)
+		{
+			Table::Col c;
+
+			c.name = ac.name;
+			c.type = ac.type;
+			if (ac.type == GG_COL_CHAR1)
+			{
+				uint8_t *d_u8 = nullptr;
+
+				GG_HIP_AO(hipMalloc((void **) &d_u8,
+						    (size_t) total_rows
+						    + 1));
+				GG_HIP_AO(launch_narrow_i32_u8(
+					e.stream, (const int32_t *) d_vals,
+					total_rows, d_u8));
+				GG_HIP_AO(hipStreamSynchronize(e.stream));
+				c.dev = d_u8;
+				c.bytes = (size_t) total_rows;
+				(void) hipFree(d_vals);
+				d_vals = nullptr;
+				if (st != GG_OK)
+					(void) hipFree(d_u8);
+			}
+			else
+			{
+				c.dev = d_vals;
+				c.bytes = (size_t) total_rows * out_width;
+				d_vals = nullptr;	/* owned by table */
+			}
+			if (st == GG_OK)
+				t->cols.push_back(c);
+		}
+#undef GG_HIP_AO
+		(void) hipFree(d_stream);
+		(void) hipFree(d_offs);
+		(void) hipFree(d_oo);
+		(void) hipFree(d_sizes);
+		(void) hipFree(d_rows);
+		(void) hipFree(d_nulls);
+		(void) hipFree(d_err);
+		if (d_vals)
+			(void) hipFree(d_vals);
+	}
+	if (st != GG_OK)
+	{
+		for (auto &c : t->cols)
+			if (c.bytes)
+				(void) hipFree(c.dev);
+		delete t;
+		return st;
+	}
+	if (t->nrows < 0)
+		t->nrows = 0;
 	e.tables.push_back(t);
 	*out = (gg_table) (e.tables.size() - 1);
 	return GG_OK;

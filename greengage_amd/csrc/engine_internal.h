@@ -349,6 +349,8 @@ hipError_t launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n,
 
 hipError_t launch_fill_u64(hipStream_t s, unsigned long long *p,
 			   uint64_t n, unsigned long long v);
+hipError_t launch_narrow_i32_u8(hipStream_t s, const int32_t *in,
+				int64_t n, uint8_t *out);
 hipError_t launch_sj_build(hipStream_t s, const int64_t *keys,
 			   const int64_t *vals, int64_t n,
 			   unsigned long long *tkeys,

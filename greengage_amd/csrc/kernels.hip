@@ -1853,6 +1853,27 @@ launch_gen_supplier(hipStream_t s, uint64_t seed, int64_t row_lo, int64_t n,
 
 #define GB_EMPTY 0x8000000000000000ull	/* INT64_MIN as the empty slot */
 
+/* narrow decoded int32 values into a u8 column (char1/dict codes) */
+__global__ void
+k_narrow_i32_u8(const int32_t *__restrict__ in, int64_t n,
+		uint8_t *__restrict__ out)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		out[i] = (uint8_t) in[i];
+}
+
+hipError_t
+launch_narrow_i32_u8(hipStream_t s, const int32_t *in, int64_t n,
+		     uint8_t *out)
+{
+	hipLaunchKernelGGL(k_narrow_i32_u8, dim3(grid_for(n)), dim3(THREADS),
+			   0, s, in, n, out);
+	return hipGetLastError();
+}
+
 /* generic i64-key hash-join kernels for the spill tier
  * (ExecHashJoin batching semantics, nodeHash.c:713): CAS-insert build
  * (key 0 reserved as empty, like the pipeline tables), probe emits

@@ -137,6 +137,20 @@ def _load():
         ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int64,
         ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32)]
+    class _AOCol(ctypes.Structure):
+        _fields_ = [("name", ctypes.c_char_p),
+                    ("type", ctypes.c_int),
+                    ("stream", ctypes.c_void_p),
+                    ("stream_len", ctypes.c_int64),
+                    ("checksums", ctypes.c_int),
+                    ("ao_version", ctypes.c_int),
+                    ("dsb_version", ctypes.c_int),
+                    ("comptype", ctypes.c_int)]
+    lib.gg_AOCol = _AOCol
+    lib.gg_engine_register_table_ao.restype = ctypes.c_int
+    lib.gg_engine_register_table_ao.argtypes = [
+        ctypes.c_char_p, ctypes.POINTER(_AOCol), ctypes.c_int,
+        ctypes.POINTER(ctypes.c_int32)]
     lib.gg_engine_text_dict_encode.restype = ctypes.c_int
     lib.gg_engine_text_dict_encode.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
@@ -251,6 +265,32 @@ class Engine:
                                               len(cols), nrows,
                                               ctypes.byref(h)),
                f"register_table({name})")
+        return h.value
+
+    def register_table_ao(self, name, cols):
+        """Mount a real AO table: cols = list of (name, typename,
+        ao_bytes, checksums, ao_version, dsb_version, comptype)."""
+        import numpy as np
+        L = lib()
+        descs = (L.gg_AOCol * len(cols))()
+        keep = []
+        for i, (cname, ctype, ao, cks, aov, dsbv, ct) in \
+                enumerate(cols):
+            ao = np.ascontiguousarray(ao, np.uint8)
+            keep.append(ao)
+            descs[i].name = cname.encode()
+            descs[i].type = COLTYPE[ctype]
+            descs[i].stream = ao.ctypes.data_as(ctypes.c_void_p)
+            descs[i].stream_len = len(ao)
+            descs[i].checksums = cks
+            descs[i].ao_version = aov
+            descs[i].dsb_version = dsbv
+            descs[i].comptype = ct
+        h = I32()
+        _check(L.gg_engine_register_table_ao(name.encode(), descs,
+                                             len(cols),
+                                             ctypes.byref(h)),
+               f"register_table_ao({name})")
         return h.value
 
     def table_nrows(self, h):

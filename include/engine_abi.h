@@ -228,6 +228,28 @@ gg_status gg_engine_aocs_decode_text(const uint8_t *stream,
 				     int64_t *out_nrows,
 				     int64_t *out_pool_len);
 
+/* Mount a REAL AO table: one AO segfile byte stream per column (the
+ * form cdbbufferedread.c hands up); each column runs through the full
+ * storage layer (headers + CRC32C + codecs + datum-stream decode)
+ * straight into a device-resident column.  Columns must be NOT NULL.
+ * The table is then usable by every pipeline like any registered
+ * table. */
+typedef struct gg_ao_column
+{
+	const char *name;
+	gg_coltype	type;
+	const uint8_t *stream;	/* raw AO segfile bytes */
+	int64_t		stream_len;
+	int			checksums;
+	int			ao_version;	/* >= 2 */
+	int			dsb_version;	/* 0/1/2 */
+	int			comptype;	/* 0 none, 1 zlib, 2 zstd */
+} gg_ao_column;
+
+gg_status gg_engine_register_table_ao(const char *name,
+				      const gg_ao_column *cols, int ncols,
+				      gg_table *out);
+
 /* Dictionary-encode a categorical text column (arrow-style inputs,
  * e.g. gg_engine_aocs_decode_text's output) into int32 codes + a
  * lexicographically sorted dictionary — deterministic across shards,

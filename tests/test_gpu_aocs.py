@@ -422,3 +422,49 @@ def test_q3_with_text_mktsegment_from_ao(eng):
     for got, exp in zip(rows, g["rows"][:10]):
         assert got["orderkey"] == exp["orderkey"]
         assert got["revenue4"] == exp["revenue4"]
+
+
+def test_register_table_ao_and_run_q1(eng):
+    """gg_engine_register_table_ao mounts a whole lineitem table from
+    per-column AO segfile byte streams (mixed codecs) straight into
+    device-resident columns; Q1 on it equals the golden."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    import json
+    import os
+    from conftest import REPO
+    from greengage_amd.engine import Engine as E, PIPE_Q1
+    small = dict(np.load(os.path.join(REPO, "tests", "golden",
+                                      "small_inputs.npz")))
+    g = json.load(open(os.path.join(REPO, "tests", "golden",
+                                    "q1_small.json")))
+
+    def ao_col(arr, width, comptype=1, level=6, version=2):
+        vals = np.ascontiguousarray(arr).astype(np.int64)
+        framed, nb = pyoracle.dsb_encode(
+            vals, np.zeros(len(vals), np.uint8), width, version, 0, 0,
+            blocksz=4096)
+        if comptype:
+            return pyoracle.ao_wrap_compressed(framed, comptype, level)
+        return pyoracle.ao_wrap(framed)
+
+    n = len(small["li_orderkey"])
+    t = eng.register_table_ao("lineitem_mounted", [
+        ("orderkey", "int64", ao_col(small["li_orderkey"], 8), 1, 2, 2, 1),
+        ("qty", "dec64", ao_col(small["li_qty_c"], 8, 2, 3), 1, 2, 2, 2),
+        ("price", "dec64", ao_col(small["li_price_c"], 8), 1, 2, 2, 1),
+        ("disc", "dec64", ao_col(small["li_disc_c"], 8, 0), 1, 2, 2, 0),
+        ("tax", "dec64", ao_col(small["li_tax_c"], 8), 1, 2, 2, 1),
+        ("shipdate", "int32", ao_col(small["li_shipdate"], 4), 1, 2, 2, 1),
+        ("rflag", "char1", ao_col(small["li_rflag"], 4, 2, 3), 1, 2, 2, 2),
+        ("lstatus", "char1", ao_col(small["li_lstatus"], 4), 1, 2, 2, 1),
+    ])
+    assert eng.table_nrows(t) == n
+    p = eng.compile(PIPE_Q1, lineitem=t, cutoff_date=g["cutoff_pgdate"])
+    groups = eng.execute_q1(p)
+    assert len(groups) == len(g["rows"])
+    for got, exp in zip(groups, g["rows"]):
+        assert got["returnflag"] == exp["l_returnflag"]
+        assert got["count"] == exp["count_order"]
+        assert got["sum_qty_c"] == exp["sum_qty_c"]
+        assert got["sum_charge6"] == exp["sum_charge6"]
