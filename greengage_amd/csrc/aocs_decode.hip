@@ -560,8 +560,13 @@ k_dsb_decode_text(const uint8_t *__restrict__ stream,
 	}
 }
 
+/* offsets[b] >= 0 address `stream`; negative offsets address the
+ * spill buffer at (-off - 1) — the zero-copy AO path leaves
+ * uncompressed block content in the original segfile bytes and only
+ * materializes decompressed/reassembled content into the spill */
 __global__ void
 k_dsb_decode(const uint8_t *__restrict__ stream,
+	     const uint8_t *__restrict__ spill,
 	     const int64_t *__restrict__ offsets,
 	     const int32_t *__restrict__ sizes,
 	     const int32_t *__restrict__ rowcounts,
@@ -575,7 +580,10 @@ k_dsb_decode(const uint8_t *__restrict__ stream,
 	for (int64_t b = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     b < nblocks; b += stride)
 	{
-		int rc = decode_one_block(stream + offsets[b], sizes[b],
+		int64_t off = offsets[b];
+		const uint8_t *src = off >= 0 ? stream + off
+			: spill + (-off - 1);
+		int rc = decode_one_block(src, sizes[b],
 					  rowcounts[b], version, datumlen,
 					  out_vals, out_nulls, out_width,
 					  out_offsets[b]);
@@ -594,6 +602,20 @@ launch_dsb_decode(hipStream_t s, const uint8_t *stream,
 		  int32_t nblocks, int version, int datumlen, void *out_vals,
 		  uint8_t *out_nulls, int out_width, unsigned long long *err)
 {
+	return launch_dsb_decode2(s, stream, nullptr, offsets, sizes,
+				  rowcounts, out_offsets, nblocks, version,
+				  datumlen, out_vals, out_nulls, out_width,
+				  err);
+}
+
+hipError_t
+launch_dsb_decode2(hipStream_t s, const uint8_t *stream,
+		   const uint8_t *spill, const int64_t *offsets,
+		   const int32_t *sizes, const int32_t *rowcounts,
+		   const int64_t *out_offsets, int32_t nblocks, int version,
+		   int datumlen, void *out_vals, uint8_t *out_nulls,
+		   int out_width, unsigned long long *err)
+{
 	int blocks = (nblocks + 255) / 256;
 
 	if (blocks > 2048)
@@ -601,9 +623,9 @@ launch_dsb_decode(hipStream_t s, const uint8_t *stream,
 	if (blocks < 1)
 		blocks = 1;
 	hipLaunchKernelGGL(k_dsb_decode, dim3(blocks), dim3(256), 0, s,
-			   stream, offsets, sizes, rowcounts, out_offsets,
-			   nblocks, version, datumlen, out_vals, out_nulls,
-			   out_width, err);
+			   stream, spill, offsets, sizes, rowcounts,
+			   out_offsets, nblocks, version, datumlen, out_vals,
+			   out_nulls, out_width, err);
 	return hipGetLastError();
 }
 

@@ -8,10 +8,18 @@
  * Greengage deployment — by the segment shim.
  */
 #include <algorithm>
+#include <atomic>
 #include <cstdarg>
 #include <cstdio>
 #include <cstring>
 #include <vector>
+
+/* zlib (pg_compression.c:253 binds compress2/uncompress) */
+#include <zlib.h>
+/* zstd public API — libzstd.so.1 ships without dev headers here */
+extern "C" size_t ZSTD_decompress(void *dst, size_t dstCap,
+				  const void *src, size_t srcSize);
+extern "C" unsigned ZSTD_isError(size_t code);
 
 #include "engine_internal.h"
 #include "../../include/gg_gen.h"
@@ -85,6 +93,8 @@ static gg_status dev_counter(unsigned long long **p)
 	GG_HIP(hipMemset(*p, 0, sizeof(unsigned long long)));
 	return GG_OK;
 }
+
+static uint32_t ao_crc32c(const uint8_t *p, int64_t len);
 
 static gg_status ao_parse_to_framed(const uint8_t *stream,
 				    int64_t stream_len, int checksums,
@@ -250,6 +260,370 @@ gg_engine_register_table(const char *name, const gg_column_desc *cols,
 	return GG_OK;
 }
 
+/* ---- zero-copy AO block parse (the fast path for the int decoders):
+ * descriptors point INTO the original segfile bytes for uncompressed
+ * Small/NonBulkDense/BulkDense blocks; only decompressed content and
+ * reassembled LargeContent goes to a spill buffer.  Checksums verify
+ * and codecs run OpenMP-parallel across blocks — the sequential
+ * single-thread walk only reads headers. ---- */
+
+struct AoDesc
+{
+	int64_t off;		/* >=0: stream offset; <0: spill (-off-1) */
+	int32_t size;
+	int32_t rowcount;
+};
+
+struct AoJob
+{
+	int64_t src;		/* stream offset of stored bytes */
+	int32_t stored;		/* compressed (or raw) length */
+	int32_t datalen;	/* uncompressed length */
+	int64_t dst;		/* spill offset */
+	int comptype;		/* 0 = plain copy (fragment), 1/2 codec */
+};
+
+static gg_status
+ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
+		int ao_version, int comptype, std::vector<AoDesc> &descs,
+		std::vector<uint8_t> &spill)
+{
+	if (!stream || stream_len < 0)
+		return fail(GG_EINVAL, "bad ao stream");
+	if (ao_version < 2)
+		return fail(GG_EINVAL, "ao_version %d unsupported",
+			    ao_version);
+
+	struct Blk
+	{
+		int64_t pos;
+		int64_t overall;
+		int kind;
+	};
+	std::vector<Blk> blks;
+	std::vector<AoJob> jobs;
+	int64_t pos = 0, expect_rownum = -1, spill_len = 0;
+	int64_t large_remaining = 0, large_dst = 0;
+
+	while (pos < stream_len)
+	{
+		uint32_t w0, w1;
+
+		if (pos + 8 > stream_len)
+			return fail(GG_EINVAL, "truncated AO header at %lld",
+				    (long long) pos);
+		std::memcpy(&w0, stream + pos, 4);
+		std::memcpy(&w1, stream + pos + 4, 4);
+		int kind = (int) ((w0 >> 28) & 7);
+		int has_frn = (int) ((w0 >> 27) & 1);
+		int32_t rowcount, datalen, complen = 0;
+
+		if (kind == 1)
+		{
+			rowcount = (int32_t) ((w0 >> 10) & 0x3FFF);
+			datalen = (int32_t) (((w0 & 0x3FF) << 11) |
+					     ((w1 >> 21) & 0x7FF));
+			complen = (int32_t) (w1 & 0x1FFFFF);
+		}
+		else if (kind == 2)
+		{
+			if (large_remaining > 0)
+				return fail(GG_EINVAL,
+					    "nested LargeContent at %lld",
+					    (long long) pos);
+			rowcount = (int32_t) (((w0 & 0x7FFFFF) << 2) |
+					      ((w1 >> 30) & 3));
+			datalen = 0;
+			int32_t biglen = (int32_t) (w1 & 0x3FFFFFFF);
+
+			large_remaining = biglen;
+			large_dst = spill_len;
+			descs.push_back({-(spill_len + 1), biglen,
+					 rowcount});
+			spill_len += biglen;
+		}
+		else if (kind == 3)
+		{
+			datalen = (int32_t) (w0 & 0x1FFFFF);
+			rowcount = (int32_t) (w1 & 0x3FFFFFFF);
+		}
+		else if (kind == 4)
+		{
+			uint32_t e1;
+			int64_t ext_at = pos + 8 + (checksums ? 8 : 0);
+
+			if (ext_at + 8 > stream_len)
+				return fail(GG_EINVAL,
+					    "truncated BulkDense at %lld",
+					    (long long) pos);
+			datalen = (int32_t) (((w0 & 0x3FF) << 11) |
+					     ((w1 >> 21) & 0x7FF));
+			complen = (int32_t) (w1 & 0x1FFFFF);
+			std::memcpy(&e1, stream + ext_at + 4, 4);
+			rowcount = (int32_t) (e1 & 0x3FFFFFFF);
+		}
+		else
+			return fail(GG_EINVAL,
+				    "unsupported AoHeaderKind %d at %lld",
+				    kind, (long long) pos);
+		if (complen != 0 && comptype == 0)
+			return fail(GG_EINVAL,
+				    "compressed AO block at %lld but "
+				    "comptype none", (long long) pos);
+		if (large_remaining > 0 && kind != 1 && kind != 2)
+			return fail(GG_EINVAL,
+				    "expected SmallContent fragment at "
+				    "%lld", (long long) pos);
+
+		int32_t stored = complen ? complen : datalen;
+		int64_t hdr_end = pos + 8 + (checksums ? 8 : 0) +
+			(kind == 4 ? 8 : 0) + (has_frn ? 8 : 0);
+		int64_t padded = ((int64_t) stored + 7) & ~(int64_t) 7;
+		int64_t overall = hdr_end + padded;
+
+		if (overall > stream_len || datalen < 0 || rowcount < 0)
+			return fail(GG_EINVAL, "AO block at %lld overruns",
+				    (long long) pos);
+		blks.push_back({pos, overall, kind});
+		if (has_frn)
+		{
+			int64_t frn;
+
+			std::memcpy(&frn, stream + pos + 8 +
+				    (checksums ? 8 : 0) +
+				    (kind == 4 ? 8 : 0), 8);
+			if (expect_rownum >= 0 && frn != expect_rownum)
+				return fail(GG_EINVAL,
+					    "firstRowNum discontinuity at "
+					    "%lld", (long long) pos);
+			expect_rownum = frn + rowcount;
+		}
+		bool fragment = (large_remaining > 0 && kind == 1);
+
+		if (kind == 2)
+			;	/* frame opened above */
+		else if (fragment)
+		{
+			if (datalen > large_remaining)
+				return fail(GG_EINVAL,
+					    "LargeContent overrun at %lld",
+					    (long long) pos);
+			jobs.push_back({hdr_end, stored, datalen, large_dst,
+					complen ? comptype : 0});
+			large_dst += datalen;
+			large_remaining -= datalen;
+		}
+		else if (complen == 0)
+			descs.push_back({hdr_end, datalen, rowcount});
+		else
+		{
+			jobs.push_back({hdr_end, stored, datalen, spill_len,
+					comptype});
+			descs.push_back({-(spill_len + 1), datalen,
+					 rowcount});
+			spill_len += datalen;
+		}
+		pos = overall;
+	}
+	if (large_remaining > 0)
+		return fail(GG_EINVAL, "LargeContent truncated");
+
+	/* parallel checksum verify */
+	if (checksums)
+	{
+		std::atomic<int64_t> bad(-1);
+
+#pragma omp parallel for schedule(dynamic, 64)
+		for (int64_t b = 0; b < (int64_t) blks.size(); b++)
+		{
+			const Blk &bl = blks[b];
+			uint32_t stored_hdr, stored_blk;
+
+			std::memcpy(&stored_blk, stream + bl.pos + 8, 4);
+			std::memcpy(&stored_hdr, stream + bl.pos + 12, 4);
+			if (ao_crc32c(stream + bl.pos, 12) != stored_hdr)
+				bad.store(bl.pos);
+			else
+			{
+				int64_t end = bl.kind == 2 ? bl.pos + 16
+					: bl.overall;
+
+				if (ao_crc32c(stream + bl.pos + 16,
+					      end - bl.pos - 16) !=
+				    stored_blk)
+					bad.store(bl.pos);
+			}
+		}
+		if (bad.load() >= 0)
+			return fail(GG_EINVAL, "AO checksum mismatch at "
+				    "%lld", (long long) bad.load());
+	}
+
+	/* parallel decompress / fragment copy into the spill */
+	spill.resize((size_t) spill_len);
+	{
+		std::atomic<int64_t> bad(-1);
+
+#pragma omp parallel for schedule(dynamic, 16)
+		for (int64_t j = 0; j < (int64_t) jobs.size(); j++)
+		{
+			const AoJob &jb = jobs[j];
+			bool ok = true;
+
+			if (jb.comptype == 0)
+				std::memcpy(spill.data() + jb.dst,
+					    stream + jb.src,
+					    (size_t) jb.datalen);
+			else if (jb.comptype == 1)
+			{
+				uLongf dl = (uLongf) jb.datalen;
+
+				ok = uncompress(spill.data() + jb.dst, &dl,
+						stream + jb.src,
+						(uLong) jb.stored) == Z_OK
+					&& dl == (uLongf) jb.datalen;
+			}
+			else
+			{
+				size_t dl = ZSTD_decompress(
+					spill.data() + jb.dst,
+					(size_t) jb.datalen,
+					stream + jb.src,
+					(size_t) jb.stored);
+
+				ok = !ZSTD_isError(dl) &&
+					dl == (size_t) jb.datalen;
+			}
+			if (!ok)
+				bad.store(jb.src);
+		}
+		if (bad.load() >= 0)
+			return fail(GG_EINVAL, "AO decompress failed at "
+				    "%lld", (long long) bad.load());
+	}
+	return GG_OK;
+}
+
+/* upload + device datum-stream decode of parsed blocks; on success the
+ * caller owns *d_vals (device) — *d_nulls is checked NOT NULL here */
+static gg_status
+ao_decode_to_device(Engine &e, const uint8_t *stream, int64_t stream_len,
+		    const std::vector<AoDesc> &descs,
+		    const std::vector<uint8_t> &spill, int dsb_version,
+		    int datumlen, int out_width, void **out_d_vals,
+		    int64_t *out_rows)
+{
+	size_t nb = descs.size();
+	int64_t total_rows = 0;
+	std::vector<int64_t> offs(nb), out_offs(nb);
+	std::vector<int32_t> sizes(nb), rows(nb);
+
+	for (size_t i = 0; i < nb; i++)
+	{
+		offs[i] = descs[i].off;
+		sizes[i] = descs[i].size;
+		rows[i] = descs[i].rowcount;
+		out_offs[i] = total_rows;
+		total_rows += descs[i].rowcount;
+	}
+	*out_rows = total_rows;
+	*out_d_vals = nullptr;
+	if (total_rows == 0)
+		return GG_OK;
+
+	gg_status st = GG_OK;
+	uint8_t *d_stream = nullptr, *d_spill = nullptr, *d_nulls = nullptr;
+	int64_t *d_offs = nullptr, *d_oo = nullptr;
+	int32_t *d_sizes = nullptr, *d_rows = nullptr;
+	void *d_vals = nullptr;
+	unsigned long long *d_err = nullptr;
+
+#define GG_HIP_AD(x) \
+	{ hipError_t e_ = (x); \
+	  if (st == GG_OK && e_ != hipSuccess) \
+		st = fail(GG_EGPU, "ao_decode: %s", \
+			  hipGetErrorString(e_)); }
+	GG_HIP_AD(hipMalloc((void **) &d_stream,
+			    stream_len ? (size_t) stream_len : 1));
+	GG_HIP_AD(hipMalloc((void **) &d_spill,
+			    spill.size() ? spill.size() : 1));
+	GG_HIP_AD(hipMalloc((void **) &d_offs, nb * 8));
+	GG_HIP_AD(hipMalloc((void **) &d_oo, nb * 8));
+	GG_HIP_AD(hipMalloc((void **) &d_sizes, nb * 4));
+	GG_HIP_AD(hipMalloc((void **) &d_rows, nb * 4));
+	GG_HIP_AD(hipMalloc(&d_vals, (size_t) total_rows * out_width));
+	GG_HIP_AD(hipMalloc((void **) &d_nulls, (size_t) total_rows));
+	GG_HIP_AD(hipMalloc((void **) &d_err, 8));
+	if (st == GG_OK)
+	{
+		if (stream_len)
+			GG_HIP_AD(hipMemcpyAsync(d_stream, stream,
+						 (size_t) stream_len,
+						 hipMemcpyHostToDevice,
+						 e.stream));
+		if (spill.size())
+			GG_HIP_AD(hipMemcpyAsync(d_spill, spill.data(),
+						 spill.size(),
+						 hipMemcpyHostToDevice,
+						 e.stream));
+		GG_HIP_AD(hipMemcpyAsync(d_offs, offs.data(), nb * 8,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AD(hipMemcpyAsync(d_oo, out_offs.data(), nb * 8,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AD(hipMemcpyAsync(d_sizes, sizes.data(), nb * 4,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AD(hipMemcpyAsync(d_rows, rows.data(), nb * 4,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AD(hipMemsetAsync(d_err, 0, 8, e.stream));
+		GG_HIP_AD(launch_dsb_decode2(e.stream, d_stream, d_spill,
+					     d_offs, d_sizes, d_rows, d_oo,
+					     (int32_t) nb, dsb_version,
+					     datumlen, d_vals, d_nulls,
+					     out_width, d_err));
+		GG_HIP_AD(hipStreamSynchronize(e.stream));
+	}
+	if (st == GG_OK)
+	{
+		unsigned long long herr = 0;
+
+		GG_HIP_AD(hipMemcpy(&herr, d_err, 8,
+				    hipMemcpyDeviceToHost));
+		if (st == GG_OK && herr)
+			st = fail(GG_EINVAL, "block decode error mask "
+				  "0x%llx", herr);
+	}
+	if (st == GG_OK)
+	{
+		/* NOT NULL check (hot-path columns) */
+		std::vector<uint8_t> hn(total_rows);
+
+		GG_HIP_AD(hipMemcpy(hn.data(), d_nulls,
+				    (size_t) total_rows,
+				    hipMemcpyDeviceToHost));
+		for (int64_t r = 0; st == GG_OK && r < total_rows; r++)
+			if (hn[r])
+				st = fail(GG_EINVAL,
+					  "NULL at row %lld in a NOT NULL "
+					  "column", (long long) r);
+	}
+#undef GG_HIP_AD
+	(void) hipFree(d_stream);
+	(void) hipFree(d_spill);
+	(void) hipFree(d_offs);
+	(void) hipFree(d_oo);
+	(void) hipFree(d_sizes);
+	(void) hipFree(d_rows);
+	(void) hipFree(d_nulls);
+	(void) hipFree(d_err);
+	if (st != GG_OK)
+	{
+		(void) hipFree(d_vals);
+		return st;
+	}
+	*out_d_vals = d_vals;
+	return GG_OK;
+}
+
 /*
  * Mount a REAL AO table: one AO segfile byte stream per column, the
  * way cdbbufferedread.c hands blocks up.  Each column goes through
@@ -281,37 +655,29 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 		const gg_ao_column &ac = cols[ci];
 		int datumlen = (ac.type == GG_COL_INT64 ||
 				ac.type == GG_COL_DEC64_S2) ? 8 : 4;
-		int out_width = datumlen;
-		std::vector<uint8_t> framed;
+		std::vector<AoDesc> descs;
+		std::vector<uint8_t> spill;
 
-		st = ao_parse_to_framed(ac.stream, ac.stream_len,
-					ac.checksums, ac.ao_version,
-					ac.comptype, framed);
+		st = ao_parse_blocks(ac.stream, ac.stream_len, ac.checksums,
+				     ac.ao_version, ac.comptype, descs,
+				     spill);
 		if (st != GG_OK)
 			break;
 
-		/* frame descriptors */
-		std::vector<int64_t> offs, out_offs;
-		std::vector<int32_t> sizes, rows;
-		int64_t pos = 0, total_rows = 0;
+		void *d_vals = nullptr;
+		int64_t total_rows = 0;
 
-		while (pos < (int64_t) framed.size())
-		{
-			int32_t sz, rc;
-
-			std::memcpy(&sz, framed.data() + pos, 4);
-			std::memcpy(&rc, framed.data() + pos + 4, 4);
-			offs.push_back(pos + 8);
-			sizes.push_back(sz);
-			rows.push_back(rc);
-			out_offs.push_back(total_rows);
-			total_rows += rc;
-			pos = (pos + 8 + sz + 7) & ~(int64_t) 7;
-		}
+		st = ao_decode_to_device(e, ac.stream, ac.stream_len,
+					 descs, spill, ac.dsb_version,
+					 datumlen, datumlen, &d_vals,
+					 &total_rows);
+		if (st != GG_OK)
+			break;
 		if (t->nrows < 0)
 			t->nrows = total_rows;
 		else if (t->nrows != total_rows)
 		{
+			(void) hipFree(d_vals);
 			st = fail(GG_EINVAL, "column %s has %lld rows, "
 				  "table has %lld", ac.name,
 				  (long long) total_rows,
@@ -319,122 +685,39 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 			break;
 		}
 
-		uint8_t *d_stream = nullptr, *d_nulls = nullptr;
-		int64_t *d_offs = nullptr, *d_oo = nullptr;
-		int32_t *d_sizes = nullptr, *d_rows = nullptr;
-		void *d_vals = nullptr;
-		unsigned long long *d_err = nullptr;
-		size_t nb = offs.size();
+		Table::Col c;
 
-#define GG_HIP_AO(x) \
-		{ hipError_t e_ = (x); \
-		  if (st == GG_OK && e_ != hipSuccess) \
-			st = fail(GG_EGPU, "register_table_ao: %s", \
-				  hipGetErrorString(e_)); }
-		GG_HIP_AO(hipMalloc((void **) &d_stream,
-				    framed.size() ? framed.size() : 1));
-		GG_HIP_AO(hipMalloc((void **) &d_offs, (nb + 1) * 8));
-		GG_HIP_AO(hipMalloc((void **) &d_oo, (nb + 1) * 8));
-		GG_HIP_AO(hipMalloc((void **) &d_sizes, (nb + 1) * 4));
-		GG_HIP_AO(hipMalloc((void **) &d_rows, (nb + 1) * 4));
-		GG_HIP_AO(hipMalloc(&d_vals,
-				    (size_t) total_rows * out_width + 1));
-		GG_HIP_AO(hipMalloc((void **) &d_nulls,
-				    (size_t) total_rows + 1));
-		GG_HIP_AO(hipMalloc((void **) &d_err, 8));
-		if (st == GG_OK && nb)
+		c.name = ac.name;
+		c.type = ac.type;
+		if (ac.type == GG_COL_CHAR1 && total_rows)
 		{
-			GG_HIP_AO(hipMemcpy(d_stream, framed.data(),
-					    framed.size(),
-					    hipMemcpyHostToDevice));
-			GG_HIP_AO(hipMemcpy(d_offs, offs.data(), nb * 8,
-					    hipMemcpyHostToDevice));
-			GG_HIP_AO(hipMemcpy(d_oo, out_offs.data(), nb * 8,
-					    hipMemcpyHostToDevice));
-			GG_HIP_AO(hipMemcpy(d_sizes, sizes.data(), nb * 4,
-					    hipMemcpyHostToDevice));
-			GG_HIP_AO(hipMemcpy(d_rows, rows.data(), nb * 4,
-					    hipMemcpyHostToDevice));
-			GG_HIP_AO(hipMemset(d_err, 0, 8));
-			GG_HIP_AO(launch_dsb_decode(
-				e.stream, d_stream, d_offs, d_sizes,
-				d_rows, d_oo, (int32_t) nb, ac.dsb_version,
-				datumlen, d_vals, d_nulls, out_width,
-				d_err));
-			GG_HIP_AO(hipStreamSynchronize(e.stream));
-			if (st == GG_OK)
-			{
-				unsigned long long herr = 0;
+			uint8_t *d_u8 = nullptr;
+			hipError_t he = hipMalloc((void **) &d_u8,
+						  (size_t) total_rows);
 
-				GG_HIP_AO(hipMemcpy(&herr, d_err, 8,
-						    hipMemcpyDeviceToHost));
-				if (st == GG_OK && herr)
-					st = fail(GG_EINVAL,
-						  "column %s decode error "
-						  "0x%llx", ac.name, herr);
-			}
-			/* NOT NULL check */
-			if (st == GG_OK && total_rows)
-			{
-				std::vector<uint8_t> hn(total_rows);
-
-				GG_HIP_AO(hipMemcpy(hn.data(), d_nulls,
-						    (size_t) total_rows,
-						    hipMemcpyDeviceToHost));
-				for (int64_t r = 0;
-				     st == GG_OK && r < total_rows; r++)
-					if (hn[r])
-						st = fail(GG_EINVAL,
-							  "column %s has "
-							  "NULL at row %lld",
-							  ac.name,
-							  (long long) r);
-			}
-		}
-		/* attach the decoded column (narrowing for char1) */
-		if (st == GG_OK)
-		{
-			Table::Col c;
-
-			c.name = ac.name;
-			c.type = ac.type;
-			if (ac.type == GG_COL_CHAR1)
-			{
-				uint8_t *d_u8 = nullptr;
-
-				GG_HIP_AO(hipMalloc((void **) &d_u8,
-						    (size_t) total_rows
-						    + 1));
-				GG_HIP_AO(launch_narrow_i32_u8(
+			if (he == hipSuccess)
+				he = launch_narrow_i32_u8(
 					e.stream, (const int32_t *) d_vals,
-					total_rows, d_u8));
-				GG_HIP_AO(hipStreamSynchronize(e.stream));
-				c.dev = d_u8;
-				c.bytes = (size_t) total_rows;
-				(void) hipFree(d_vals);
-				d_vals = nullptr;
-				if (st != GG_OK)
-					(void) hipFree(d_u8);
-			}
-			else
-			{
-				c.dev = d_vals;
-				c.bytes = (size_t) total_rows * out_width;
-				d_vals = nullptr;	/* owned by table */
-			}
-			if (st == GG_OK)
-				t->cols.push_back(c);
-		}
-#undef GG_HIP_AO
-		(void) hipFree(d_stream);
-		(void) hipFree(d_offs);
-		(void) hipFree(d_oo);
-		(void) hipFree(d_sizes);
-		(void) hipFree(d_rows);
-		(void) hipFree(d_nulls);
-		(void) hipFree(d_err);
-		if (d_vals)
+					total_rows, d_u8);
+			if (he == hipSuccess)
+				he = hipStreamSynchronize(e.stream);
 			(void) hipFree(d_vals);
+			if (he != hipSuccess)
+			{
+				(void) hipFree(d_u8);
+				st = fail(GG_EGPU, "narrow: %s",
+					  hipGetErrorString(he));
+				break;
+			}
+			c.dev = d_u8;
+			c.bytes = (size_t) total_rows;
+		}
+		else
+		{
+			c.dev = d_vals;
+			c.bytes = (size_t) total_rows * datumlen;
+		}
+		t->cols.push_back(c);
 	}
 	if (st != GG_OK)
 	{
@@ -2637,12 +2920,6 @@ ao_crc32c(const uint8_t *p, int64_t len)
  * via gg_engine_aocs_decode.  Compressed blocks (compressedLength!=0)
  * and ao_version<2 are out of scope this round.
  */
-/* zlib (pg_compression.c:253 binds compress2/uncompress) */
-#include <zlib.h>
-/* zstd public API — libzstd.so.1 ships without dev headers here */
-extern "C" size_t ZSTD_decompress(void *dst, size_t dstCap,
-				  const void *src, size_t srcSize);
-extern "C" unsigned ZSTD_isError(size_t code);
 
 static gg_status
 ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
@@ -2887,16 +3164,56 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 			 int out_width, uint8_t *out_nulls, int64_t cap,
 			 int64_t *out_nrows)
 {
-	std::vector<uint8_t> framed;
-	gg_status st = ao_parse_to_framed(stream, stream_len, checksums,
-					  ao_version, comptype, framed);
+	if (!out_vals || !out_nulls || !out_nrows ||
+	    (out_width != 4 && out_width != 8) ||
+	    (datumlen != 4 && datumlen != 8))
+		return fail(GG_EINVAL, "bad aocs_decode_ao args");
+
+	/* zero-copy parse (host-only — corruption is detected here even
+	 * without a GPU): uncompressed content decodes straight from the
+	 * uploaded segfile bytes; only decompressed/reassembled content
+	 * travels through the spill */
+	std::vector<AoDesc> descs;
+	std::vector<uint8_t> spill;
+	gg_status st = ao_parse_blocks(stream, stream_len, checksums,
+				       ao_version, comptype, descs, spill);
 
 	if (st != GG_OK)
 		return st;
-	return gg_engine_aocs_decode(framed.data(),
-				     (int64_t) framed.size(), dsb_version,
-				     datumlen, out_vals, out_width,
-				     out_nulls, cap, out_nrows);
+
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+
+	void *d_vals = nullptr;
+	int64_t total_rows = 0;
+
+	st = ao_decode_to_device(e, stream, stream_len, descs, spill,
+				 dsb_version, datumlen, out_width, &d_vals,
+				 &total_rows);
+	if (st != GG_OK)
+		return st;
+	if (total_rows > cap)
+	{
+		(void) hipFree(d_vals);
+		return fail(GG_EINVAL, "cap %lld < rows %lld",
+			    (long long) cap, (long long) total_rows);
+	}
+	*out_nrows = total_rows;
+	if (total_rows)
+	{
+		hipError_t he = hipMemcpy(out_vals, d_vals,
+					  (size_t) total_rows * out_width,
+					  hipMemcpyDeviceToHost);
+
+		std::memset(out_nulls, 0, (size_t) total_rows);
+		if (he != hipSuccess)
+			st = fail(GG_EGPU, "decode_ao copyback: %s",
+				  hipGetErrorString(he));
+	}
+	(void) hipFree(d_vals);
+	return st;
 }
 
 /* same, for TEXT columns: AO layer (headers, checksums, codecs) on the

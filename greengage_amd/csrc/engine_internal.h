@@ -520,6 +520,15 @@ hipError_t launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
 				  unsigned long long *out_offs,
 				  uint32_t *out_lens, uint8_t *out_nulls,
 				  unsigned long long *err);
+hipError_t launch_dsb_decode2(hipStream_t s, const uint8_t *stream,
+			      const uint8_t *spill,
+			      const int64_t *offsets,
+			      const int32_t *sizes,
+			      const int32_t *rowcounts,
+			      const int64_t *out_offsets, int32_t nblocks,
+			      int version, int datumlen, void *out_vals,
+			      uint8_t *out_nulls, int out_width,
+			      unsigned long long *err);
 hipError_t launch_dsb_decode(hipStream_t s, const uint8_t *stream,
 			     const int64_t *offsets, const int32_t *sizes,
 			     const int32_t *rowcounts,
