@@ -9,6 +9,7 @@
  */
 #include <algorithm>
 #include <atomic>
+#include <thread>
 #include <cstdarg>
 #include <cstdio>
 #include <cstring>
@@ -267,6 +268,43 @@ gg_engine_register_table(const char *name, const gg_column_desc *cols,
  * and codecs run OpenMP-parallel across blocks — the sequential
  * single-thread walk only reads headers. ---- */
 
+/* simple static-partition parallel-for over host threads */
+template <typename F>
+static void
+ao_parallel_for(int64_t n, F fn)
+{
+	unsigned nt = std::thread::hardware_concurrency();
+
+	if (nt > 32)
+		nt = 32;
+	if (nt < 2 || n < 64)
+	{
+		for (int64_t i = 0; i < n; i++)
+			fn(i);
+		return;
+	}
+	std::vector<std::thread> ths;
+	std::atomic<int64_t> next(0);
+
+	for (unsigned t = 0; t < nt; t++)
+		ths.emplace_back([&]()
+		{
+			for (;;)
+			{
+				int64_t i = next.fetch_add(64);
+
+				if (i >= n)
+					return;
+				int64_t hi = i + 64 < n ? i + 64 : n;
+
+				for (; i < hi; i++)
+					fn(i);
+			}
+		});
+	for (auto &th : ths)
+		th.join();
+}
+
 struct AoDesc
 {
 	int64_t off;		/* >=0: stream offset; <0: spill (-off-1) */
@@ -428,13 +466,13 @@ ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
 	if (large_remaining > 0)
 		return fail(GG_EINVAL, "LargeContent truncated");
 
-	/* parallel checksum verify */
+	/* parallel checksum verify (std::thread — hipcc must not see
+	 * -fopenmp, it leaks into device codegen) */
 	if (checksums)
 	{
 		std::atomic<int64_t> bad(-1);
 
-#pragma omp parallel for schedule(dynamic, 64)
-		for (int64_t b = 0; b < (int64_t) blks.size(); b++)
+		ao_parallel_for((int64_t) blks.size(), [&](int64_t b)
 		{
 			const Blk &bl = blks[b];
 			uint32_t stored_hdr, stored_blk;
@@ -453,7 +491,7 @@ ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
 				    stored_blk)
 					bad.store(bl.pos);
 			}
-		}
+		});
 		if (bad.load() >= 0)
 			return fail(GG_EINVAL, "AO checksum mismatch at "
 				    "%lld", (long long) bad.load());
@@ -464,8 +502,7 @@ ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
 	{
 		std::atomic<int64_t> bad(-1);
 
-#pragma omp parallel for schedule(dynamic, 16)
-		for (int64_t j = 0; j < (int64_t) jobs.size(); j++)
+		ao_parallel_for((int64_t) jobs.size(), [&](int64_t j)
 		{
 			const AoJob &jb = jobs[j];
 			bool ok = true;
@@ -496,7 +533,7 @@ ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
 			}
 			if (!ok)
 				bad.store(jb.src);
-		}
+		});
 		if (bad.load() >= 0)
 			return fail(GG_EINVAL, "AO decompress failed at "
 				    "%lld", (long long) bad.load());
