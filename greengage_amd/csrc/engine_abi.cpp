@@ -548,7 +548,7 @@ ao_decode_to_device(Engine &e, const uint8_t *stream, int64_t stream_len,
 		    const std::vector<AoDesc> &descs,
 		    const std::vector<uint8_t> &spill, int dsb_version,
 		    int datumlen, int out_width, void **out_d_vals,
-		    int64_t *out_rows)
+		    uint8_t **out_d_nulls, int64_t *out_rows)
 {
 	size_t nb = descs.size();
 	int64_t total_rows = 0;
@@ -629,9 +629,9 @@ ao_decode_to_device(Engine &e, const uint8_t *stream, int64_t stream_len,
 			st = fail(GG_EINVAL, "block decode error mask "
 				  "0x%llx", herr);
 	}
-	if (st == GG_OK)
+	if (st == GG_OK && !out_d_nulls)
 	{
-		/* NOT NULL check (hot-path columns) */
+		/* caller wants a NOT NULL column (pipeline mounts) */
 		std::vector<uint8_t> hn(total_rows);
 
 		GG_HIP_AD(hipMemcpy(hn.data(), d_nulls,
@@ -650,14 +650,18 @@ ao_decode_to_device(Engine &e, const uint8_t *stream, int64_t stream_len,
 	(void) hipFree(d_oo);
 	(void) hipFree(d_sizes);
 	(void) hipFree(d_rows);
-	(void) hipFree(d_nulls);
 	(void) hipFree(d_err);
 	if (st != GG_OK)
 	{
 		(void) hipFree(d_vals);
+		(void) hipFree(d_nulls);
 		return st;
 	}
 	*out_d_vals = d_vals;
+	if (out_d_nulls)
+		*out_d_nulls = d_nulls;
+	else
+		(void) hipFree(d_nulls);
 	return GG_OK;
 }
 
@@ -707,6 +711,7 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 		st = ao_decode_to_device(e, ac.stream, ac.stream_len,
 					 descs, spill, ac.dsb_version,
 					 datumlen, datumlen, &d_vals,
+					 nullptr /* NOT NULL required */ ,
 					 &total_rows);
 		if (st != GG_OK)
 			break;
@@ -3224,16 +3229,18 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 		return fail(GG_ESTATE, "engine not initialized");
 
 	void *d_vals = nullptr;
+	uint8_t *d_nulls = nullptr;
 	int64_t total_rows = 0;
 
 	st = ao_decode_to_device(e, stream, stream_len, descs, spill,
 				 dsb_version, datumlen, out_width, &d_vals,
-				 &total_rows);
+				 &d_nulls, &total_rows);
 	if (st != GG_OK)
 		return st;
 	if (total_rows > cap)
 	{
 		(void) hipFree(d_vals);
+		(void) hipFree(d_nulls);
 		return fail(GG_EINVAL, "cap %lld < rows %lld",
 			    (long long) cap, (long long) total_rows);
 	}
@@ -3244,12 +3251,16 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 					  (size_t) total_rows * out_width,
 					  hipMemcpyDeviceToHost);
 
-		std::memset(out_nulls, 0, (size_t) total_rows);
+		if (he == hipSuccess)
+			he = hipMemcpy(out_nulls, d_nulls,
+				       (size_t) total_rows,
+				       hipMemcpyDeviceToHost);
 		if (he != hipSuccess)
 			st = fail(GG_EGPU, "decode_ao copyback: %s",
 				  hipGetErrorString(he));
 	}
 	(void) hipFree(d_vals);
+	(void) hipFree(d_nulls);
 	return st;
 }
 
