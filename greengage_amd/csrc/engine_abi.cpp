@@ -470,7 +470,7 @@ ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
 	 * -fopenmp, it leaks into device codegen) */
 	if (checksums)
 	{
-		std::atomic<int64_t> bad(-1);
+		std::atomic<int64_t> bad_hdr(-1), bad_blk(-1);
 
 		ao_parallel_for((int64_t) blks.size(), [&](int64_t b)
 		{
@@ -480,7 +480,7 @@ ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
 			std::memcpy(&stored_blk, stream + bl.pos + 8, 4);
 			std::memcpy(&stored_hdr, stream + bl.pos + 12, 4);
 			if (ao_crc32c(stream + bl.pos, 12) != stored_hdr)
-				bad.store(bl.pos);
+				bad_hdr.store(bl.pos);
 			else
 			{
 				int64_t end = bl.kind == 2 ? bl.pos + 16
@@ -489,12 +489,17 @@ ao_parse_blocks(const uint8_t *stream, int64_t stream_len, int checksums,
 				if (ao_crc32c(stream + bl.pos + 16,
 					      end - bl.pos - 16) !=
 				    stored_blk)
-					bad.store(bl.pos);
+					bad_blk.store(bl.pos);
 			}
 		});
-		if (bad.load() >= 0)
-			return fail(GG_EINVAL, "AO checksum mismatch at "
-				    "%lld", (long long) bad.load());
+		if (bad_hdr.load() >= 0)
+			return fail(GG_EINVAL, "AO header checksum "
+				    "mismatch at %lld",
+				    (long long) bad_hdr.load());
+		if (bad_blk.load() >= 0)
+			return fail(GG_EINVAL, "AO block checksum "
+				    "mismatch at %lld",
+				    (long long) bad_blk.load());
 	}
 
 	/* parallel decompress / fragment copy into the spill */
