@@ -533,6 +533,7 @@ decode_one_block_text(const uint8_t *buf, int32_t size, int32_t rowcount,
 
 __global__ void
 k_dsb_decode_text(const uint8_t *__restrict__ stream,
+		  const uint8_t *__restrict__ spill,
 		  const int64_t *__restrict__ offsets,
 		  const int32_t *__restrict__ sizes,
 		  const int32_t *__restrict__ rowcounts,
@@ -549,7 +550,10 @@ k_dsb_decode_text(const uint8_t *__restrict__ stream,
 	for (int64_t b = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     b < nblocks; b += stride)
 	{
-		int rc = decode_one_block_text(stream + offsets[b], sizes[b],
+		int64_t off = offsets[b];
+		const uint8_t *src = off >= 0 ? stream + off
+			: spill + (-off - 1);
+		int rc = decode_one_block_text(src, sizes[b],
 					       rowcounts[b], version, pool,
 					       pool_offsets[b], out_offs,
 					       out_lens, out_nulls,
@@ -631,8 +635,9 @@ launch_dsb_decode2(hipStream_t s, const uint8_t *stream,
 
 hipError_t
 launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
-		       const int64_t *offsets, const int32_t *sizes,
-		       const int32_t *rowcounts, const int64_t *out_offsets,
+		       const uint8_t *spill, const int64_t *offsets,
+		       const int32_t *sizes, const int32_t *rowcounts,
+		       const int64_t *out_offsets,
 		       const int64_t *pool_offsets, int32_t nblocks,
 		       int version, uint8_t *pool,
 		       unsigned long long *out_offs, uint32_t *out_lens,
@@ -645,9 +650,9 @@ launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
 	if (blocks < 1)
 		blocks = 1;
 	hipLaunchKernelGGL(k_dsb_decode_text, dim3(blocks), dim3(256), 0, s,
-			   stream, offsets, sizes, rowcounts, out_offsets,
-			   pool_offsets, nblocks, version, pool, out_offs,
-			   out_lens, out_nulls, err);
+			   stream, spill, offsets, sizes, rowcounts,
+			   out_offsets, pool_offsets, nblocks, version, pool,
+			   out_offs, out_lens, out_nulls, err);
 	return hipGetLastError();
 }
 

@@ -97,11 +97,6 @@ static gg_status dev_counter(unsigned long long **p)
 
 static uint32_t ao_crc32c(const uint8_t *p, int64_t len);
 
-static gg_status ao_parse_to_framed(const uint8_t *stream,
-				    int64_t stream_len, int checksums,
-				    int ao_version, int comptype,
-				    std::vector<uint8_t> &framed);
-
 static gg_status read_counter(unsigned long long *p, unsigned long long *out)
 {
 	GG_HIP(hipMemcpy(out, p, sizeof(*out), hipMemcpyDeviceToHost));
@@ -2968,242 +2963,6 @@ ao_crc32c(const uint8_t *p, int64_t len)
  * and ao_version<2 are out of scope this round.
  */
 
-static gg_status
-ao_parse_to_framed(const uint8_t *stream, int64_t stream_len,
-		   int checksums, int ao_version, int comptype,
-		   std::vector<uint8_t> &framed)
-{
-	if (!stream || stream_len < 0)
-		return fail(GG_EINVAL, "bad ao stream");
-	if (ao_version < 2)
-		return fail(GG_EINVAL,
-			    "ao_version %d unsupported (need >=2, Aligned64bit)",
-			    ao_version);
-
-	framed.reserve((size_t) stream_len);
-	int64_t pos = 0;
-	int64_t expect_rownum = -1;
-	/* LargeContent reassembly state: a kind-2 metadata block opens a
-	 * frame whose payload is the concatenated content of the
-	 * following SmallContent fragment blocks (rowCount 0, no
-	 * firstRowNum) — AppendOnlyStorageRead_Content:1240 */
-	int64_t large_remaining = 0;
-	size_t large_fpos = 0;
-
-	while (pos < stream_len)
-	{
-		uint32_t w0, w1;
-		int64_t hdr_end;
-
-		if (pos + 8 > stream_len)
-			return fail(GG_EINVAL, "truncated AO header at %lld",
-				    (long long) pos);
-		std::memcpy(&w0, stream + pos, 4);
-		std::memcpy(&w1, stream + pos + 4, 4);
-		int kind = (int) ((w0 >> 28) & 7);
-		int has_frn = (int) ((w0 >> 27) & 1);
-		int32_t rowcount, datalen, complen = 0;
-
-		if (kind == 1)
-		{		/* AoHeaderKind_SmallContent */
-			rowcount = (int32_t) ((w0 >> 10) & 0x3FFF);
-			datalen = (int32_t) (((w0 & 0x3FF) << 11) |
-					     ((w1 >> 21) & 0x7FF));
-			complen = (int32_t) (w1 & 0x1FFFFF);
-			if (complen != 0 && comptype == 0)
-				return fail(GG_EINVAL,
-					    "compressed AO block at %lld but "
-					    "comptype none",
-					    (long long) pos);
-		}
-		else if (kind == 2)
-		{		/* AoHeaderKind_LargeContent: header-only
-				 * metadata (largeRowCount 25 bits split 23+2,
-				 * largeContentLength 30 bits —
-				 * cdbappendonlystorage_int.h:227) */
-			if (large_remaining > 0)
-				return fail(GG_EINVAL,
-					    "nested LargeContent at %lld",
-					    (long long) pos);
-			rowcount = (int32_t) (((w0 & 0x7FFFFF) << 2) |
-					      ((w1 >> 30) & 3));
-			datalen = 0;	/* no content of its own */
-			int32_t biglen = (int32_t) (w1 & 0x3FFFFFFF);
-
-			large_remaining = biglen;
-			large_fpos = framed.size();
-			framed.resize(large_fpos + 8);
-			std::memcpy(framed.data() + large_fpos, &biglen, 4);
-			std::memcpy(framed.data() + large_fpos + 4,
-				    &rowcount, 4);
-		}
-		else if (kind == 3)
-		{		/* AoHeaderKind_NonBulkDenseContent */
-			datalen = (int32_t) (w0 & 0x1FFFFF);
-			rowcount = (int32_t) (w1 & 0x3FFFFFFF);
-		}
-		else if (kind == 4)
-		{		/* AoHeaderKind_BulkDenseContent — LONG header:
-				 * dataLength/compressedLength packed like
-				 * SmallContent in word0/1, largeRowCount in
-				 * the 8-byte extension after the checksums
-				 * (cdbappendonlystorage_int.h:454–467) */
-			uint32_t e1;
-			int64_t ext_at = pos + 8 + (checksums ? 8 : 0);
-
-			if (ext_at + 8 > stream_len)
-				return fail(GG_EINVAL,
-					    "truncated BulkDense ext at %lld",
-					    (long long) pos);
-			datalen = (int32_t) (((w0 & 0x3FF) << 11) |
-					     ((w1 >> 21) & 0x7FF));
-			complen = (int32_t) (w1 & 0x1FFFFF);
-			std::memcpy(&e1, stream + ext_at + 4, 4);
-			rowcount = (int32_t) (e1 & 0x3FFFFFFF);
-			if (complen != 0 && comptype == 0)
-				return fail(GG_EINVAL,
-					    "compressed AO block at %lld but "
-					    "comptype none",
-					    (long long) pos);
-		}
-		else
-			return fail(GG_EINVAL,
-				    "unsupported AoHeaderKind %d at %lld",
-				    kind, (long long) pos);
-
-		if (large_remaining > 0 && kind != 1 && kind != 2)
-			return fail(GG_EINVAL,
-				    "expected SmallContent fragment at %lld "
-				    "inside LargeContent", (long long) pos);
-
-		/* stored bytes = compressed when compressedLength != 0
-		 * (CompressAppend rule, cdbappendonlystoragewrite.c:1207) */
-		int32_t stored = complen ? complen : datalen;
-
-		hdr_end = pos + 8 + (checksums ? 8 : 0) +
-			(kind == 4 ? 8 : 0) + (has_frn ? 8 : 0);
-		int64_t padded = ((int64_t) stored + 7) & ~(int64_t) 7;
-		int64_t overall = hdr_end + padded;
-
-		if (overall > stream_len || datalen < 0 || rowcount < 0)
-			return fail(GG_EINVAL, "AO block at %lld overruns "
-				    "stream", (long long) pos);
-		if (checksums)
-		{
-			uint32_t stored_hdr, stored_blk;
-
-			std::memcpy(&stored_blk, stream + pos + 8, 4);
-			std::memcpy(&stored_hdr, stream + pos + 12, 4);
-			uint32_t comp_hdr = ao_crc32c(stream + pos, 12);
-
-			if (comp_hdr != stored_hdr)
-				return fail(GG_EINVAL,
-					    "AO header checksum mismatch at "
-					    "%lld: stored 0x%08X computed "
-					    "0x%08X", (long long) pos,
-					    stored_hdr, comp_hdr);
-			/* LargeContent metadata: the writer computes the
-			 * block CRC over ZERO bytes (MakeLargeContentHeader
-			 * passes hasFirstRowNum=false to
-			 * AddBlockHeaderChecksums, so overall there is just
-			 * the 16 header bytes) — mirror that */
-			int64_t blk_end = (kind == 2) ? pos + 16 : overall;
-			uint32_t comp_blk = ao_crc32c(stream + pos + 16,
-						      blk_end - pos - 16);
-
-			if (comp_blk != stored_blk)
-				return fail(GG_EINVAL,
-					    "AO block checksum mismatch at "
-					    "%lld: stored 0x%08X computed "
-					    "0x%08X", (long long) pos,
-					    stored_blk, comp_blk);
-		}
-		if (has_frn)
-		{
-			int64_t frn;
-
-			std::memcpy(&frn, stream + pos + 8 +
-				    (checksums ? 8 : 0) +
-				    (kind == 4 ? 8 : 0), 8);
-			if (expect_rownum >= 0 && frn != expect_rownum)
-				return fail(GG_EINVAL,
-					    "AO firstRowNum discontinuity at "
-					    "%lld: got %lld expected %lld",
-					    (long long) pos, (long long) frn,
-					    (long long) expect_rownum);
-			expect_rownum = frn + rowcount;
-		}
-		size_t fpos;
-		size_t dst_off;
-		bool fragment = (large_remaining > 0 && kind == 1);
-
-		if (kind == 2)
-		{		/* metadata emitted above; nothing else */
-			pos = overall;
-			continue;
-		}
-		if (fragment)
-		{
-			if (datalen > large_remaining)
-				return fail(GG_EINVAL,
-					    "LargeContent overrun at %lld",
-					    (long long) pos);
-			dst_off = framed.size();
-			framed.resize(dst_off + (size_t) datalen);
-			large_remaining -= datalen;
-		}
-		else
-		{
-			fpos = framed.size();
-			framed.resize(fpos + 8 + (size_t) datalen);
-			std::memcpy(framed.data() + fpos, &datalen, 4);
-			std::memcpy(framed.data() + fpos + 4, &rowcount, 4);
-			dst_off = fpos + 8;
-		}
-		if (complen == 0)
-			std::memcpy(framed.data() + dst_off,
-				    stream + hdr_end, (size_t) datalen);
-		else if (comptype == 1)
-		{
-			uLongf dl = (uLongf) datalen;
-			int zrc = uncompress(framed.data() + dst_off, &dl,
-					     stream + hdr_end,
-					     (uLong) complen);
-
-			if (zrc != Z_OK || dl != (uLongf) datalen)
-				return fail(GG_EINVAL,
-					    "zlib decompress failed at %lld "
-					    "(rc %d, %lu of %d bytes)",
-					    (long long) pos, zrc,
-					    (unsigned long) dl, datalen);
-		}
-		else if (comptype == 2)
-		{
-			size_t dl = ZSTD_decompress(framed.data() + dst_off,
-						    (size_t) datalen,
-						    stream + hdr_end,
-						    (size_t) complen);
-
-			if (ZSTD_isError(dl) || dl != (size_t) datalen)
-				return fail(GG_EINVAL,
-					    "zstd decompress failed at %lld",
-					    (long long) pos);
-		}
-		else
-			return fail(GG_EINVAL, "unknown comptype %d",
-				    comptype);
-		/* frames stay 8-aligned (AOStorage_RoundUp8 analog) */
-		if (!fragment || large_remaining == 0)
-			framed.resize((framed.size() + 7) & ~(size_t) 7);
-		pos = overall;
-	}
-	if (large_remaining > 0)
-		return fail(GG_EINVAL,
-			    "LargeContent truncated: %lld bytes missing",
-			    (long long) large_remaining);
-	return GG_OK;
-}
-
 extern "C" gg_status
 gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 			 int checksums, int ao_version, int dsb_version,
@@ -3270,8 +3029,8 @@ gg_engine_aocs_decode_ao(const uint8_t *stream, int64_t stream_len,
 }
 
 /* same, for TEXT columns: AO layer (headers, checksums, codecs) on the
- * host, varlena datum-stream decode on the GPU.  Output offsets
- * reference the framed pool (see gg_engine_aocs_decode_text). */
+ * host via the zero-copy parallel parser, varlena datum-stream decode
+ * on the GPU.  Output offsets reference `pool`. */
 extern "C" gg_status
 gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 			      int checksums, int ao_version,
@@ -3281,17 +3040,145 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 			      uint8_t *pool, int64_t pool_cap,
 			      int64_t *out_nrows, int64_t *out_pool_len)
 {
-	std::vector<uint8_t> framed;
-	gg_status st = ao_parse_to_framed(stream, stream_len, checksums,
-					  ao_version, comptype, framed);
+	if (!out_offs || !out_lens || !out_nulls || !pool || !out_nrows ||
+	    !out_pool_len)
+		return fail(GG_EINVAL, "bad aocs_decode_ao_text args");
+
+	std::vector<AoDesc> descs;
+	std::vector<uint8_t> spill;
+	gg_status st = ao_parse_blocks(stream, stream_len, checksums,
+				       ao_version, comptype, descs, spill);
 
 	if (st != GG_OK)
 		return st;
-	return gg_engine_aocs_decode_text(framed.data(),
-					  (int64_t) framed.size(),
-					  dsb_version, out_offs, out_lens,
-					  out_nulls, cap, pool, pool_cap,
-					  out_nrows, out_pool_len);
+
+	Engine &e = engine();
+
+	if (!e.inited)
+		return fail(GG_ESTATE, "engine not initialized");
+
+	size_t nb = descs.size();
+	int64_t total_rows = 0, pool_len = 0;
+	std::vector<int64_t> offs(nb), out_off(nb), pool_off(nb);
+	std::vector<int32_t> sizes(nb), rows(nb);
+
+	for (size_t i = 0; i < nb; i++)
+	{
+		offs[i] = descs[i].off;
+		sizes[i] = descs[i].size;
+		rows[i] = descs[i].rowcount;
+		out_off[i] = total_rows;
+		pool_off[i] = pool_len;
+		total_rows += descs[i].rowcount;
+		pool_len += descs[i].size;
+	}
+	if (total_rows > cap)
+		return fail(GG_EINVAL, "cap %lld < rows %lld",
+			    (long long) cap, (long long) total_rows);
+	if (pool_len > pool_cap)
+		return fail(GG_EINVAL, "pool_cap %lld < %lld",
+			    (long long) pool_cap, (long long) pool_len);
+	*out_nrows = total_rows;
+	*out_pool_len = pool_len;
+	if (total_rows == 0)
+		return GG_OK;
+
+	uint8_t *d_stream = nullptr, *d_spill = nullptr, *d_pool = nullptr,
+		*d_nulls = nullptr;
+	int64_t *d_offs = nullptr, *d_oo = nullptr, *d_po = nullptr;
+	int32_t *d_sizes = nullptr, *d_rows = nullptr;
+	unsigned long long *d_out_offs = nullptr;
+	uint32_t *d_lens = nullptr;
+	unsigned long long *d_err = nullptr;
+
+#define GG_HIP_AT(x) \
+	{ hipError_t e_ = (x); \
+	  if (st == GG_OK && e_ != hipSuccess) \
+		st = fail(GG_EGPU, "decode_ao_text: %s", \
+			  hipGetErrorString(e_)); }
+	GG_HIP_AT(hipMalloc((void **) &d_stream,
+			    stream_len ? (size_t) stream_len : 1));
+	GG_HIP_AT(hipMalloc((void **) &d_spill,
+			    spill.size() ? spill.size() : 1));
+	GG_HIP_AT(hipMalloc((void **) &d_pool, (size_t) pool_len));
+	GG_HIP_AT(hipMalloc((void **) &d_offs, nb * 8));
+	GG_HIP_AT(hipMalloc((void **) &d_oo, nb * 8));
+	GG_HIP_AT(hipMalloc((void **) &d_po, nb * 8));
+	GG_HIP_AT(hipMalloc((void **) &d_sizes, nb * 4));
+	GG_HIP_AT(hipMalloc((void **) &d_rows, nb * 4));
+	GG_HIP_AT(hipMalloc((void **) &d_out_offs,
+			    (size_t) total_rows * 8));
+	GG_HIP_AT(hipMalloc((void **) &d_lens, (size_t) total_rows * 4));
+	GG_HIP_AT(hipMalloc((void **) &d_nulls, (size_t) total_rows));
+	GG_HIP_AT(hipMalloc((void **) &d_err, 8));
+	if (st == GG_OK)
+	{
+		if (stream_len)
+			GG_HIP_AT(hipMemcpyAsync(d_stream, stream,
+						 (size_t) stream_len,
+						 hipMemcpyHostToDevice,
+						 e.stream));
+		if (spill.size())
+			GG_HIP_AT(hipMemcpyAsync(d_spill, spill.data(),
+						 spill.size(),
+						 hipMemcpyHostToDevice,
+						 e.stream));
+		GG_HIP_AT(hipMemcpyAsync(d_offs, offs.data(), nb * 8,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AT(hipMemcpyAsync(d_oo, out_off.data(), nb * 8,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AT(hipMemcpyAsync(d_po, pool_off.data(), nb * 8,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AT(hipMemcpyAsync(d_sizes, sizes.data(), nb * 4,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AT(hipMemcpyAsync(d_rows, rows.data(), nb * 4,
+					 hipMemcpyHostToDevice, e.stream));
+		GG_HIP_AT(hipMemsetAsync(d_err, 0, 8, e.stream));
+		GG_HIP_AT(launch_dsb_decode_text(
+			e.stream, d_stream, d_spill, d_offs, d_sizes,
+			d_rows, d_oo, d_po, (int32_t) nb, dsb_version,
+			d_pool, d_out_offs, d_lens, d_nulls, d_err));
+		GG_HIP_AT(hipStreamSynchronize(e.stream));
+	}
+	if (st == GG_OK)
+	{
+		unsigned long long herr = 0;
+
+		GG_HIP_AT(hipMemcpy(&herr, d_err, 8,
+				    hipMemcpyDeviceToHost));
+		if (st == GG_OK && herr)
+			st = fail(GG_EINVAL,
+				  "text block decode error mask 0x%llx",
+				  herr);
+	}
+	if (st == GG_OK)
+	{
+		GG_HIP_AT(hipMemcpy(out_offs, d_out_offs,
+				    (size_t) total_rows * 8,
+				    hipMemcpyDeviceToHost));
+		GG_HIP_AT(hipMemcpy(out_lens, d_lens,
+				    (size_t) total_rows * 4,
+				    hipMemcpyDeviceToHost));
+		GG_HIP_AT(hipMemcpy(out_nulls, d_nulls,
+				    (size_t) total_rows,
+				    hipMemcpyDeviceToHost));
+		GG_HIP_AT(hipMemcpy(pool, d_pool, (size_t) pool_len,
+				    hipMemcpyDeviceToHost));
+	}
+#undef GG_HIP_AT
+	(void) hipFree(d_stream);
+	(void) hipFree(d_spill);
+	(void) hipFree(d_pool);
+	(void) hipFree(d_offs);
+	(void) hipFree(d_oo);
+	(void) hipFree(d_po);
+	(void) hipFree(d_sizes);
+	(void) hipFree(d_rows);
+	(void) hipFree(d_out_offs);
+	(void) hipFree(d_lens);
+	(void) hipFree(d_nulls);
+	(void) hipFree(d_err);
+	return st;
 }
 
 /*
@@ -3387,9 +3274,9 @@ gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
 	GG_HIP(hipMemset(d_err, 0, 8));
 	{
 		hipError_t he = launch_dsb_decode_text(
-			e.stream, d_stream, d_offs, d_sizes, d_rows, d_oo,
-			d_po, (int32_t) nb, version, d_pool, d_out_offs,
-			d_lens, d_nulls, d_err);
+			e.stream, d_stream, nullptr, d_offs, d_sizes,
+			d_rows, d_oo, d_po, (int32_t) nb, version, d_pool,
+			d_out_offs, d_lens, d_nulls, d_err);
 
 		if (he != hipSuccess)
 			st = fail(GG_EGPU, "dsb_decode_text: %s",

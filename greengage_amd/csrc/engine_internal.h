@@ -510,6 +510,7 @@ hipError_t launch_dn_q3_collect(hipStream_t s,
 				unsigned long long *out_count, uint64_t cap);
 
 hipError_t launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
+				  const uint8_t *spill,
 				  const int64_t *offsets,
 				  const int32_t *sizes,
 				  const int32_t *rowcounts,
