@@ -694,6 +694,124 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 	for (int ci = 0; ci < ncols && st == GG_OK; ci++)
 	{
 		const gg_ao_column &ac = cols[ci];
+
+		if (ac.text_dict)
+		{
+			/* TEXT column: decode varlena + GPU dictionary
+			 * encode; codes become the u8 column, the sorted
+			 * dictionary stays on the Table::Col for
+			 * predicate-constant lookup */
+			if (ac.type != GG_COL_CHAR1)
+			{
+				st = fail(GG_EINVAL, "text_dict column %s "
+					  "must be char1", ac.name);
+				break;
+			}
+			std::vector<uint64_t> toffs;
+			std::vector<uint32_t> tlens;
+			std::vector<uint8_t> tnulls, tpool;
+			int64_t trows = 0, tpool_len = 0;
+
+			{	/* size via a parse pass */
+				std::vector<AoDesc> d2;
+				std::vector<uint8_t> s2;
+
+				st = ao_parse_blocks(ac.stream,
+						     ac.stream_len,
+						     ac.checksums,
+						     ac.ao_version,
+						     ac.comptype, d2, s2);
+				if (st != GG_OK)
+					break;
+				for (auto &d3 : d2)
+				{
+					trows += d3.rowcount;
+					tpool_len += d3.size;
+				}
+			}
+			toffs.resize(trows + 1);
+			tlens.resize(trows + 1);
+			tnulls.resize(trows + 1);
+			tpool.resize(tpool_len + 1);
+			int64_t got_rows = 0, got_pool = 0;
+
+			st = gg_engine_aocs_decode_ao_text(
+				ac.stream, ac.stream_len, ac.checksums,
+				ac.ao_version, ac.dsb_version, ac.comptype,
+				toffs.data(), tlens.data(), tnulls.data(),
+				trows + 1, tpool.data(), tpool_len + 1,
+				&got_rows, &got_pool);
+			if (st != GG_OK)
+				break;
+
+			std::vector<int32_t> codes(got_rows);
+			std::vector<uint8_t> dbytes(got_pool + 16);
+			std::vector<int64_t> doffs(257);
+			int32_t nd = 0;
+
+			st = gg_engine_text_dict_encode(
+				tpool.data(), toffs.data(), tlens.data(),
+				tnulls.data(), got_rows,
+				255 /* u8 codes */ , codes.data(),
+				dbytes.data(), (int64_t) dbytes.size(),
+				doffs.data(), &nd);
+			if (st != GG_OK)
+				break;
+			for (int64_t r = 0; r < got_rows; r++)
+				if (codes[r] < 0)
+				{
+					st = fail(GG_EINVAL,
+						  "column %s has NULL at "
+						  "row %lld", ac.name,
+						  (long long) r);
+					break;
+				}
+			if (st != GG_OK)
+				break;
+			if (t->nrows < 0)
+				t->nrows = got_rows;
+			else if (t->nrows != got_rows)
+			{
+				st = fail(GG_EINVAL, "column %s has %lld "
+					  "rows, table has %lld", ac.name,
+					  (long long) got_rows,
+					  (long long) t->nrows);
+				break;
+			}
+			std::vector<uint8_t> u8codes(got_rows);
+
+			for (int64_t r = 0; r < got_rows; r++)
+				u8codes[r] = (uint8_t) codes[r];
+			Table::Col c;
+
+			c.name = ac.name;
+			c.type = GG_COL_CHAR1;
+			c.bytes = (size_t) got_rows;
+			{
+				hipError_t he = hipMalloc(&c.dev,
+							  c.bytes ? c.bytes
+							  : 1);
+
+				if (he == hipSuccess && c.bytes)
+					he = hipMemcpy(c.dev,
+						       u8codes.data(),
+						       c.bytes,
+						       hipMemcpyHostToDevice);
+				if (he != hipSuccess)
+				{
+					st = fail(GG_EGPU, "dict col: %s",
+						  hipGetErrorString(he));
+					break;
+				}
+			}
+			c.dict_bytes.assign(dbytes.begin(),
+					    dbytes.begin() + doffs[nd]);
+			c.dict_offs.assign(doffs.begin(),
+					   doffs.begin() + nd + 1);
+			t->cols.push_back(c);
+			continue;
+		}
+
 		int datumlen = (ac.type == GG_COL_INT64 ||
 				ac.type == GG_COL_DEC64_S2) ? 8 : 4;
 		std::vector<AoDesc> descs;
@@ -783,6 +901,40 @@ static Table *get_table(gg_table h)
 	if (h < 0 || (size_t) h >= e.tables.size())
 		return nullptr;
 	return e.tables[h];
+}
+
+extern "C" gg_status
+gg_engine_table_text_dict(gg_table h, const char *col, uint8_t *out_bytes,
+			  int64_t cap, int64_t *out_offs,
+			  int32_t max_entries, int32_t *out_n)
+{
+	Table *t = get_table(h);
+
+	if (!t || !col || !out_bytes || !out_offs || !out_n)
+		return fail(GG_EINVAL, "bad table_text_dict args");
+	for (auto &c : t->cols)
+		if (c.name == col)
+		{
+			if (c.dict_offs.empty())
+				return fail(GG_EINVAL,
+					    "column %s has no dictionary",
+					    col);
+			int32_t n = (int32_t) c.dict_offs.size() - 1;
+
+			if (n > max_entries)
+				return fail(GG_EINVAL,
+					    "dict has %d entries > %d",
+					    n, max_entries);
+			if ((int64_t) c.dict_bytes.size() > cap)
+				return fail(GG_EINVAL, "dict cap");
+			std::memcpy(out_bytes, c.dict_bytes.data(),
+				    c.dict_bytes.size());
+			for (int32_t i = 0; i <= n; i++)
+				out_offs[i] = c.dict_offs[i];
+			*out_n = n;
+			return GG_OK;
+		}
+	return fail(GG_EINVAL, "no column %s", col);
 }
 
 extern "C" gg_status gg_engine_table_nrows(gg_table h, int64_t *out)

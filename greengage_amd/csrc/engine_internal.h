@@ -70,6 +70,10 @@ struct Table
 		gg_coltype type;
 		void *dev = nullptr;
 		size_t bytes = 0;
+		/* for dictionary-encoded text columns: the sorted dict
+		 * (host-side; categorical, small) */
+		std::vector<uint8_t> dict_bytes;
+		std::vector<int64_t> dict_offs;
 	};
 	std::vector<Col> cols;
 

@@ -145,8 +145,14 @@ def _load():
                     ("checksums", ctypes.c_int),
                     ("ao_version", ctypes.c_int),
                     ("dsb_version", ctypes.c_int),
-                    ("comptype", ctypes.c_int)]
+                    ("comptype", ctypes.c_int),
+                    ("text_dict", ctypes.c_int)]
     lib.gg_AOCol = _AOCol
+    lib.gg_engine_table_text_dict.restype = ctypes.c_int
+    lib.gg_engine_table_text_dict.argtypes = [
+        ctypes.c_int32, ctypes.c_char_p, ctypes.c_void_p,
+        ctypes.c_int64, ctypes.c_void_p, ctypes.c_int32,
+        ctypes.POINTER(ctypes.c_int32)]
     lib.gg_engine_register_table_ao.restype = ctypes.c_int
     lib.gg_engine_register_table_ao.argtypes = [
         ctypes.c_char_p, ctypes.POINTER(_AOCol), ctypes.c_int,
@@ -274,8 +280,8 @@ class Engine:
         L = lib()
         descs = (L.gg_AOCol * len(cols))()
         keep = []
-        for i, (cname, ctype, ao, cks, aov, dsbv, ct) in \
-                enumerate(cols):
+        for i, col in enumerate(cols):
+            cname, ctype, ao, cks, aov, dsbv, ct = col[:7]
             ao = np.ascontiguousarray(ao, np.uint8)
             keep.append(ao)
             descs[i].name = cname.encode()
@@ -286,12 +292,25 @@ class Engine:
             descs[i].ao_version = aov
             descs[i].dsb_version = dsbv
             descs[i].comptype = ct
+            descs[i].text_dict = col[7] if len(col) > 7 else 0
         h = I32()
         _check(L.gg_engine_register_table_ao(name.encode(), descs,
                                              len(cols),
                                              ctypes.byref(h)),
                f"register_table_ao({name})")
         return h.value
+
+    def table_text_dict(self, h, col):
+        """Sorted dictionary of a text_dict-mounted column."""
+        import numpy as np
+        out = np.zeros(1 << 16, np.uint8)
+        offs = np.zeros(257, np.int64)
+        n = ctypes.c_int32()
+        _check(lib().gg_engine_table_text_dict(
+            h, col.encode(), out.ctypes.data_as(ctypes.c_void_p),
+            len(out), offs.ctypes.data_as(ctypes.c_void_p), 256,
+            ctypes.byref(n)), "table_text_dict")
+        return [bytes(out[offs[i]:offs[i + 1]]) for i in range(n.value)]
 
     def table_nrows(self, h):
         n = I64()

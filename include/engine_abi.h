@@ -244,11 +244,23 @@ typedef struct gg_ao_column
 	int			ao_version;	/* >= 2 */
 	int			dsb_version;	/* 0/1/2 */
 	int			comptype;	/* 0 none, 1 zlib, 2 zstd */
+	/* nonzero for a TEXT column to mount as dictionary codes
+	 * (type must be GG_COL_CHAR1; <= 255 distinct values).  The
+	 * lexicographically-sorted dictionary is retrievable with
+	 * gg_engine_table_text_dict for predicate-constant lookup. */
+	int			text_dict;
 } gg_ao_column;
 
 gg_status gg_engine_register_table_ao(const char *name,
 				      const gg_ao_column *cols, int ncols,
 				      gg_table *out);
+
+/* dictionary of a text_dict-mounted column: bytes + n+1 offsets */
+gg_status gg_engine_table_text_dict(gg_table h, const char *col,
+				    uint8_t *out_bytes, int64_t cap,
+				    int64_t *out_offs,
+				    int32_t max_entries,
+				    int32_t *out_n);
 
 /* Dictionary-encode a categorical text column (arrow-style inputs,
  * e.g. gg_engine_aocs_decode_text's output) into int32 codes + a

@@ -468,3 +468,54 @@ def test_register_table_ao_and_run_q1(eng):
         assert got["count"] == exp["count_order"]
         assert got["sum_qty_c"] == exp["sum_qty_c"]
         assert got["sum_charge6"] == exp["sum_charge6"]
+
+
+def test_mount_table_with_text_dict_column(eng):
+    """One-call mount of a table whose text column dictionary-encodes
+    on the GPU: Q3 runs on the mounted customer with the predicate
+    constant resolved through the table's dictionary — golden-exact."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    import json
+    import os
+    from conftest import REPO
+    from greengage_amd.engine import PIPE_Q3
+    small = dict(np.load(os.path.join(REPO, "tests", "golden",
+                                      "small_inputs.npz")))
+    g = json.load(open(os.path.join(REPO, "tests", "golden",
+                                    "q3_small.json")))
+    blob = small["c_mktseg_text_blob"]
+    offs = small["c_mktseg_text_offs"]
+    texts = [bytes(blob[offs[i]:offs[i + 1]])
+             for i in range(len(offs) - 1)]
+    nulls = np.zeros(len(texts), np.uint8)
+    framed_t, _ = pyoracle.dsb_encode_text(texts, nulls, 2, 1,
+                                           blocksz=4096)
+    ao_t = pyoracle.ao_wrap_compressed(framed_t, 1, 6)
+    framed_k, _ = pyoracle.dsb_encode(
+        small["c_custkey"], np.zeros(len(texts), np.uint8), 8, 2, 0, 0,
+        blocksz=4096)
+    ao_k = pyoracle.ao_wrap(framed_k)
+
+    cu = eng.register_table_ao("customer_dictmount", [
+        ("custkey", "int64", ao_k, 1, 2, 2, 0),
+        ("mktseg", "char1", ao_t, 1, 2, 2, 1, 1),  # text_dict
+    ])
+    d = eng.table_text_dict(cu, "mktseg")
+    segcode = d.index(b"MACHINERY")
+
+    from test_gpu_engine import register_lineitem_small
+    li = register_lineitem_small(eng, small)
+    od = eng.register_table("orders_small_dm", [
+        ("orderkey", "int64", small["o_orderkey"]),
+        ("custkey", "int64", small["o_custkey"]),
+        ("orderdate", "int32", small["o_orderdate"]),
+        ("shippriority", "int32", small["o_shippriority"]),
+    ], len(small["o_orderkey"]))
+    p = eng.compile(PIPE_Q3, lineitem=li, orders=od, customer=cu,
+                    cutoff_date=g["cutoff_pgdate"], mktsegment=segcode,
+                    limit_k=10)
+    rows, hdr = eng.execute_q3(p)
+    assert hdr["n_groups"] == g["n_groups"]
+    assert hdr["rev_sum4"] == g["rev_sum4"]
+    assert hdr["group_checksum"] == g["group_checksum"]
