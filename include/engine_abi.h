@@ -343,8 +343,10 @@ gg_status gg_engine_motion_dechunkify(const uint8_t *chunks,
  * dsb_version the datum-stream block version
  * (0=Orig,1=Dense,2=Dense_Enhanced), comptype the catalog's
  * compresstype (0=none, 1=zlib, 2=zstd — pg_compression.c registry);
- * compressed SmallContent blocks (compressedLength!=0) are
- * decompressed host-side before GPU decode. */
+ * all four AO header kinds parse (Small, NonBulkDense, BulkDense
+ * long headers, LargeContent metadata+fragments); checksums verify
+ * and codecs decompress host-parallel, uncompressed content decodes
+ * zero-copy from the segfile bytes. */
 gg_status gg_engine_aocs_decode_ao_text(const uint8_t *stream,
 					int64_t stream_len, int checksums,
 					int ao_version, int dsb_version,
