@@ -14,10 +14,10 @@ import os
 import pytest
 
 
-def _worker(rank, world, q):
+def _worker(rank, world, q, port=29511):
     import torch.distributed as dist
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = "29511"
+    os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
     import sys
     sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
@@ -85,6 +85,23 @@ def test_gloo_world2_distributed_identities():
     for p in procs:
         p.start()
     results = [q.get(timeout=150) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", (rank, status)
+
+
+@pytest.mark.timeout(240)
+def test_gloo_world4_distributed_identities():
+    """Same identities at world size 4 — the driver's 4/8-GPU scale
+    runs exercise the same sharding/transpose/merge rules."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, 4, q, 29517))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=200) for _ in range(4)]
     for p in procs:
         p.join(timeout=30)
     for rank, status in results:
