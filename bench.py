@@ -256,7 +256,7 @@ def main():
             "dtype": "int64",
             "data": "synthetic",
             "config": {
-                "workload": "tpch_q1_sf100",
+                "workload": f"tpch_q1_sf{SF}",
                 "sf": SF,
                 "query": "mpph1 (Q1, date - 108 days)",
                 "rows": total_rows,
