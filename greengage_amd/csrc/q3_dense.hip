@@ -741,13 +741,13 @@ void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long ng = 0, revsum = 0, carry = 0, ck = 0, mx = 0;
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < dense_len; i += stride)
+	/* 4-way unrolled: a single-stream scan needs explicit
+	 * memory-level parallelism to hide NT-load latency (one load per
+	 * iteration measured 1.9 TB/s; Q1's five streams hit 6.4) */
+	auto body = [&](int64_t i, unsigned long long r)
 	{
-		unsigned long long r = __builtin_nontemporal_load(&rev[i]);
-
 		if (!r)
-			continue;
+			return;
 		ng++;
 		mx = max(mx, r);
 		{
@@ -763,7 +763,27 @@ void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
 
 			ck += gg_group_hash(i, r, 0, date, prio);
 		}
+	};
+	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	int64_t i = i0;
+
+	for (; i + 3 * stride < dense_len; i += 4 * stride)
+	{
+		unsigned long long r0 = __builtin_nontemporal_load(&rev[i]);
+		unsigned long long r1 =
+			__builtin_nontemporal_load(&rev[i + stride]);
+		unsigned long long r2 =
+			__builtin_nontemporal_load(&rev[i + 2 * stride]);
+		unsigned long long r3 =
+			__builtin_nontemporal_load(&rev[i + 3 * stride]);
+
+		body(i, r0);
+		body(i + stride, r1);
+		body(i + 2 * stride, r2);
+		body(i + 3 * stride, r3);
 	}
+	for (; i < dense_len; i += stride)
+		body(i, __builtin_nontemporal_load(&rev[i]));
 	for (int off = 32; off; off >>= 1)
 	{
 		unsigned long long orev = revsum;
