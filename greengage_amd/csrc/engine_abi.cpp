@@ -62,7 +62,17 @@ gg_status comm_alltoallv_i64(const int64_t *send_base,
 
 static uint64_t next_pow2(uint64_t v)
 {
-	uint64_t p = 1024;
+	uint64_t p = 1024;	/* floor tuned for hash-table slot counts */
+
+	while (p < v)
+		p <<= 1;
+	return p;
+}
+
+/* exact power-of-two ceiling (no floor) — partition fan-outs */
+static uint64_t pow2_ceil(uint64_t v)
+{
+	uint64_t p = 1;
 
 	while (p < v)
 		p <<= 1;
@@ -4261,7 +4271,7 @@ gg_engine_hash_join_i64_spill(const int64_t *build_keys,
 
 	if (need > budget_bytes)
 	{
-		P = next_pow2((uint64_t) (need / budget_bytes) + 1);
+		P = pow2_ceil((uint64_t) (need / budget_bytes) + 1);
 		if (P > 4096)
 			return fail(GG_EINVAL,
 				    "join needs %llu partitions (>4096)",
@@ -4495,7 +4505,7 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 	}
 
 	/* number of hash-range partitions: each must fit the budget */
-	uint64_t P = next_pow2((uint64_t) ((n * 8 * 7) / budget_bytes) + 1);
+	uint64_t P = pow2_ceil((uint64_t) ((n * 8 * 7) / budget_bytes) + 1);
 
 	if (P > 4096)
 		return fail(GG_EINVAL, "input needs %llu partitions (>4096)",
