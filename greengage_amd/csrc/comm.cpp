@@ -96,19 +96,26 @@ gg_status comm_alltoallv_i64(const int64_t *send_base,
 	Engine &e = engine();
 	int n = e.cfg.n_segments;
 
+	/* an error between GroupStart and GroupEnd must still close the
+	 * group, or the communicator is left group-started and the next
+	 * collective on it misbehaves */
 	GG_NCCL(ncclGroupStart());
-	for (int p = 0; p < n; p++)
+	ncclResult_t r = ncclSuccess;
+
+	for (int p = 0; p < n && r == ncclSuccess; p++)
 	{
 		if (send_cnts[p])
-			GG_NCCL(ncclSend(send_base + send_offs[p],
-					 send_cnts[p], ncclInt64, p,
-					 comm_handle(), e.stream));
-		if (recv_cnts[p])
-			GG_NCCL(ncclRecv(recv_base + recv_offs[p],
-					 recv_cnts[p], ncclInt64, p,
-					 comm_handle(), e.stream));
+			r = ncclSend(send_base + send_offs[p], send_cnts[p],
+				     ncclInt64, p, comm_handle(), e.stream);
+		if (recv_cnts[p] && r == ncclSuccess)
+			r = ncclRecv(recv_base + recv_offs[p], recv_cnts[p],
+				     ncclInt64, p, comm_handle(), e.stream);
 	}
-	GG_NCCL(ncclGroupEnd());
+	ncclResult_t rend = ncclGroupEnd();
+
+	if (r != ncclSuccess || rend != ncclSuccess)
+		return fail(GG_ECOMM, "alltoallv RCCL error %s",
+			    ncclGetErrorString(r != ncclSuccess ? r : rend));
 	GG_HIP(hipStreamSynchronize(e.stream));
 	return GG_OK;
 }

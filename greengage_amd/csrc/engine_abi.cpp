@@ -1903,7 +1903,11 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.hbm_bytes += li->nrows * 28;	/* SURVEY §8(d) */
 	}
 
-	/* 4. stats (+max) in one pass, then histogram-select top-k */
+	/* 4. top-k select.  Dense path: ONE fused sweep of rev[] (stats +
+	 * exponent histogram + survivor compaction) → device threshold →
+	 * finish over the ~ngroups survivors only — no host round trip
+	 * between the three kernels.  Hash-table path (and any fused-path
+	 * overflow): the original stats→hist→collect sweeps. */
 	unsigned long long hstats[5] = {0, 0, 0, 0, 0};
 	unsigned long long *stats5 =
 		(unsigned long long *) p->sget("stats5", 5 * 8);
@@ -1912,72 +1916,178 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		return fail(GG_ENOMEM, "stats scratch");
 	{
 		Timed tm(e.stream);
-
-		GG_HIP(hipMemsetAsync(stats5, 0, 5 * 8, e.stream));
-		if (ord_dlen)
-			GG_HIP(launch_dn_q3_stats(e.stream, ordd_pay,
-						  ordd_rev, ord_dlen, stats5));
-		else
-			GG_HIP(launch_q3_stats(e.stream, ord, stats5));
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_HIP(hipMemcpy(hstats, stats5, 40, hipMemcpyDeviceToHost));
-
-		unsigned long long maxrev = hstats[4];
 		std::vector<gg_q3_result_row> cand;
+		bool need_old = !ord_dlen;
 
-		if (maxrev > 0)
+		if (ord_dlen)
 		{
-			/* device-side threshold select: histogram + suffix
-			 * walk + collect without host round trips */
+			/* grid must match launch_dn_q3_pass1's dn_grid() */
+			int64_t pgrid = (ord_dlen + 255) / 256;
+
+			if (pgrid > 2048)
+				pgrid = 2048;
+			if (pgrid < 1)
+				pgrid = 1;
+			/* survivors (groups) <= nmatch (matched orders);
+			 * grid-stride sampling keeps per-block counts near
+			 * uniform — 2x expected + slack */
+			int64_t region =
+				2 * ((int64_t) nmatch / pgrid) + 256;
+			uint64_t cap = 4 * (uint64_t) k + 65536;
 			unsigned int *dhist =
 				(unsigned int *) p->sget("hist", 65536 * 4);
 			unsigned long long *dthr =
 				(unsigned long long *) p->sget("thr", 8);
-			uint64_t cap = hstats[0] + 1;	/* <= n_groups */
+			unsigned long long *dsurv = (unsigned long long *)
+				p->sget("surv", (size_t) pgrid * region * 8);
+			unsigned long long *dcnts = (unsigned long long *)
+				p->sget("surv.cnts", (size_t) pgrid * 8);
+			unsigned long long *dovf =
+				(unsigned long long *) p->sget("topk.ovf", 8);
 			gg_q3_result_row *dout = (gg_q3_result_row *)
 				p->sget("cand", cap * sizeof(gg_q3_result_row));
-			unsigned long long ncand = 0;
+			unsigned long long ncand = 0, ovf = 0;
+			int pass1_grid = 0;
 
-			if (!dhist || !dthr || !dout)
+			if (!dhist || !dthr || !dsurv || !dcnts || !dovf ||
+			    !dout)
 				return fail(GG_ENOMEM, "topk scratch");
+			GG_HIP(hipMemsetAsync(stats5, 0, 5 * 8, e.stream));
 			GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4, e.stream));
+			GG_HIP(hipMemsetAsync(dovf, 0, 8, e.stream));
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-			if (ord_dlen)
-			{
-				GG_HIP(launch_dn_q3_hist(e.stream, ordd_rev,
-							 ord_dlen, stats5,
-							 dhist));
-				GG_HIP(launch_q3_threshold(e.stream, dhist,
-							   stats5, k, dthr));
-				GG_HIP(launch_dn_q3_collect(e.stream, ordd_pay,
-							    ordd_rev, ord_dlen,
-							    dthr, dout, ctr,
-							    cap));
-			}
-			else
-			{
-				GG_HIP(launch_q3_hist(e.stream, ord, stats5,
-						      dhist));
-				GG_HIP(launch_q3_threshold(e.stream, dhist,
-							   stats5, k, dthr));
-				GG_HIP(launch_q3_collect(e.stream, ord, dthr,
-							 dout, ctr, cap));
-			}
+			GG_HIP(launch_dn_q3_pass1(e.stream, ordd_rev,
+						  ord_dlen, stats5, dhist,
+						  dsurv, region, dcnts,
+						  &pass1_grid));
+			if (pass1_grid != (int) pgrid)
+				return fail(GG_ESTATE,
+					    "pass1 grid mismatch %d vs %lld",
+					    pass1_grid, (long long) pgrid);
+			GG_HIP(launch_dn_q3_threshold2(e.stream, dhist, k,
+						       dthr));
+			GG_HIP(launch_dn_q3_finish(e.stream, dsurv, dcnts,
+						   region, pgrid, ordd_pay,
+						   ordd_rev, dthr, stats5,
+						   dout, ctr, cap, dovf));
 			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_HIP(hipMemcpy(hstats, stats5, 40,
+					 hipMemcpyDeviceToHost));
 			GG_TRY(read_counter(ctr, &ncand));
-			if (ncand > cap)
-				return fail(GG_EINVAL,
-					    "top-k candidate overflow (%llu)",
-					    ncand);
-			cand.resize(ncand);
-			if (ncand)
+			GG_HIP(hipMemcpy(&ovf, dovf, 8,
+					 hipMemcpyDeviceToHost));
+			if (ovf)
+				need_old = true;	/* region spill */
+			else if (ncand > cap)
+			{
+				/* tie-heavy threshold bin: retry the finish
+				 * sweep alone with an exact-size buffer */
+				cap = hstats[0] + 1;
+				dout = (gg_q3_result_row *) p->sget(
+					"cand",
+					cap * sizeof(gg_q3_result_row));
+				if (!dout)
+					return fail(GG_ENOMEM, "topk retry");
+				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+				/* reset the checksum word the retry re-adds */
+				GG_HIP(hipMemsetAsync(stats5 + 3, 0, 8,
+						      e.stream));
+				GG_HIP(launch_dn_q3_finish(
+					e.stream, dsurv, dcnts, region,
+					pgrid, ordd_pay, ordd_rev, dthr,
+					stats5, dout, ctr, cap, dovf));
+				GG_HIP(hipStreamSynchronize(e.stream));
+				GG_HIP(hipMemcpy(hstats, stats5, 40,
+						 hipMemcpyDeviceToHost));
+				GG_TRY(read_counter(ctr, &ncand));
+				if (ncand > cap)
+					return fail(GG_EINVAL,
+						    "top-k overflow (%llu)",
+						    ncand);
+			}
+			if (!need_old && ncand)
+			{
+				cand.resize(ncand);
 				GG_HIP(hipMemcpy(cand.data(), dout,
 						 ncand * sizeof(gg_q3_result_row),
 						 hipMemcpyDeviceToHost));
-			std::sort(cand.begin(), cand.end(), Q3TopkCmp());
-			if ((int64_t) cand.size() > k)
-				cand.resize(k);
+			}
 		}
+
+		if (need_old)
+		{
+			GG_HIP(hipMemsetAsync(stats5, 0, 5 * 8, e.stream));
+			if (ord_dlen)
+				GG_HIP(launch_dn_q3_stats(e.stream, ordd_pay,
+							  ordd_rev, ord_dlen,
+							  stats5));
+			else
+				GG_HIP(launch_q3_stats(e.stream, ord, stats5));
+			GG_HIP(hipStreamSynchronize(e.stream));
+			GG_HIP(hipMemcpy(hstats, stats5, 40,
+					 hipMemcpyDeviceToHost));
+
+			unsigned long long maxrev = hstats[4];
+
+			cand.clear();
+			if (maxrev > 0)
+			{
+				unsigned int *dhist = (unsigned int *)
+					p->sget("hist", 65536 * 4);
+				unsigned long long *dthr =
+					(unsigned long long *) p->sget("thr", 8);
+				uint64_t cap = hstats[0] + 1;	/* <= n_groups */
+				gg_q3_result_row *dout = (gg_q3_result_row *)
+					p->sget("cand",
+						cap * sizeof(gg_q3_result_row));
+				unsigned long long ncand = 0;
+
+				if (!dhist || !dthr || !dout)
+					return fail(GG_ENOMEM, "topk scratch");
+				GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4,
+						      e.stream));
+				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+				if (ord_dlen)
+				{
+					GG_HIP(launch_dn_q3_hist(
+						e.stream, ordd_rev, ord_dlen,
+						stats5, dhist));
+					GG_HIP(launch_q3_threshold(
+						e.stream, dhist, stats5, k,
+						dthr));
+					GG_HIP(launch_dn_q3_collect(
+						e.stream, ordd_pay, ordd_rev,
+						ord_dlen, dthr, dout, ctr,
+						cap));
+				}
+				else
+				{
+					GG_HIP(launch_q3_hist(e.stream, ord,
+							      stats5, dhist));
+					GG_HIP(launch_q3_threshold(
+						e.stream, dhist, stats5, k,
+						dthr));
+					GG_HIP(launch_q3_collect(
+						e.stream, ord, dthr, dout,
+						ctr, cap));
+				}
+				GG_HIP(hipStreamSynchronize(e.stream));
+				GG_TRY(read_counter(ctr, &ncand));
+				if (ncand > cap)
+					return fail(GG_EINVAL,
+						    "top-k candidate overflow (%llu)",
+						    ncand);
+				cand.resize(ncand);
+				if (ncand)
+					GG_HIP(hipMemcpy(
+						cand.data(), dout,
+						ncand * sizeof(gg_q3_result_row),
+						hipMemcpyDeviceToHost));
+			}
+		}
+		std::sort(cand.begin(), cand.end(), Q3TopkCmp());
+		if ((int64_t) cand.size() > k)
+			cand.resize(k);
 
 		double ms = tm.stop();
 		KernelStatAcc &st = p->stat("q3_topk");
@@ -4748,64 +4858,82 @@ gg_engine_radix_sort_u64(uint64_t *keys, uint64_t *payload_or_null,
 
 	/* stable descending = stable ascending on bit-inverted keys
 	 * (inverted only within the sorted byte range so higher bytes
-	 * stay untouched) */
-	uint64_t inv = 0;
-
-	if (descending)
-	{
-		inv = (key_bytes == 8) ? ~0ull
-			: ((1ull << (8 * key_bytes)) - 1);
-		for (int64_t i = 0; i < n; i++)
-			keys[i] ^= inv;
-	}
+	 * stay untouched).  The caller's array is only touched after
+	 * every allocation succeeded, so error paths leave it intact;
+	 * all post-alloc failures route through the shared free block. */
+	uint64_t inv = !descending ? 0
+		: (key_bytes == 8) ? ~0ull : ((1ull << (8 * key_bytes)) - 1);
 
 	unsigned long long *d[2] = {nullptr, nullptr};
 	unsigned long long *dp[2] = {nullptr, nullptr};
 	unsigned int *hist = nullptr;
 	int nblocks = radix_sort_nblocks(n);
 	bool pay = payload_or_null != nullptr;
+	bool inverted = false;
 	gg_status st = GG_OK;
 
-	GG_HIP(hipMalloc((void **) &d[0], (size_t) n * 8));
-	GG_HIP(hipMalloc((void **) &d[1], (size_t) n * 8));
+#define RS_HIP(call) \
+	do { \
+		hipError_t _e = (call); \
+		if (_e != hipSuccess) { \
+			st = fail(GG_EGPU, "radix_sort: %s in %s", \
+				  hipGetErrorString(_e), #call); \
+			goto done; \
+		} \
+	} while (0)
+
+	RS_HIP(hipMalloc((void **) &d[0], (size_t) n * 8));
+	RS_HIP(hipMalloc((void **) &d[1], (size_t) n * 8));
 	if (pay)
 	{
-		GG_HIP(hipMalloc((void **) &dp[0], (size_t) n * 8));
-		GG_HIP(hipMalloc((void **) &dp[1], (size_t) n * 8));
+		RS_HIP(hipMalloc((void **) &dp[0], (size_t) n * 8));
+		RS_HIP(hipMalloc((void **) &dp[1], (size_t) n * 8));
 	}
-	GG_HIP(hipMalloc((void **) &hist, (size_t) 256 * nblocks * 4));
-	GG_HIP(hipMemcpy(d[0], keys, (size_t) n * 8, hipMemcpyHostToDevice));
+	RS_HIP(hipMalloc((void **) &hist, (size_t) 256 * nblocks * 4));
+	if (descending)
+	{
+		inverted = true;
+		for (int64_t i = 0; i < n; i++)
+			keys[i] ^= inv;
+	}
+	RS_HIP(hipMemcpy(d[0], keys, (size_t) n * 8, hipMemcpyHostToDevice));
 	if (pay)
-		GG_HIP(hipMemcpy(dp[0], payload_or_null, (size_t) n * 8,
+		RS_HIP(hipMemcpy(dp[0], payload_or_null, (size_t) n * 8,
 				 hipMemcpyHostToDevice));
 
-	int cur = 0;
-
-	for (int pass = 0; pass < key_bytes && st == GG_OK; pass++)
 	{
-		hipError_t he = launch_radix_sort_pass(
-			e.stream, d[cur], pay ? dp[cur] : nullptr, n,
-			8 * pass, hist, nblocks, d[1 - cur],
-			pay ? dp[1 - cur] : nullptr);
+		int cur = 0;
 
-		if (he != hipSuccess)
-			st = fail(GG_EGPU, "radix pass: %s",
-				  hipGetErrorString(he));
-		cur = 1 - cur;
-	}
-	if (st == GG_OK)
-	{
-		GG_HIP(hipStreamSynchronize(e.stream));
-		GG_HIP(hipMemcpy(keys, d[cur], (size_t) n * 8,
-				 hipMemcpyDeviceToHost));
-		if (pay)
-			GG_HIP(hipMemcpy(payload_or_null, dp[cur],
-					 (size_t) n * 8,
+		for (int pass = 0; pass < key_bytes && st == GG_OK; pass++)
+		{
+			hipError_t he = launch_radix_sort_pass(
+				e.stream, d[cur], pay ? dp[cur] : nullptr, n,
+				8 * pass, hist, nblocks, d[1 - cur],
+				pay ? dp[1 - cur] : nullptr);
+
+			if (he != hipSuccess)
+				st = fail(GG_EGPU, "radix pass: %s",
+					  hipGetErrorString(he));
+			cur = 1 - cur;
+		}
+		if (st == GG_OK)
+		{
+			RS_HIP(hipStreamSynchronize(e.stream));
+			RS_HIP(hipMemcpy(keys, d[cur], (size_t) n * 8,
 					 hipMemcpyDeviceToHost));
-		if (descending)
-			for (int64_t i = 0; i < n; i++)
-				keys[i] ^= inv;
+			if (pay)
+				RS_HIP(hipMemcpy(payload_or_null, dp[cur],
+						 (size_t) n * 8,
+						 hipMemcpyDeviceToHost));
+		}
 	}
+done:
+	/* keys were inverted only after allocations succeeded; un-invert
+	 * on every exit so the caller's input is never silently mangled */
+	if (inverted)
+		for (int64_t i = 0; i < n; i++)
+			keys[i] ^= inv;
+#undef RS_HIP
 	(void) hipFree(d[0]);
 	(void) hipFree(d[1]);
 	(void) hipFree(dp[0]);

@@ -498,6 +498,25 @@ hipError_t launch_dn_probe_lineitem_q5_u8(hipStream_t s,
 					  int64_t supp_dlen,
 					  unsigned long long *acc,
 					  unsigned long long *join_rows);
+hipError_t launch_dn_q3_pass1(hipStream_t s, const unsigned long long *rev,
+			      int64_t dense_len, unsigned long long *out5,
+			      unsigned int *hist64k,
+			      unsigned long long *surv, int64_t region,
+			      unsigned long long *counts, int *out_grid);
+hipError_t launch_dn_q3_threshold2(hipStream_t s,
+				   const unsigned int *hist64k, int64_t k,
+				   unsigned long long *out_thr);
+hipError_t launch_dn_q3_finish(hipStream_t s,
+			       const unsigned long long *surv,
+			       const unsigned long long *counts,
+			       int64_t region, int64_t nregions,
+			       const unsigned long long *pay,
+			       const unsigned long long *rev,
+			       const unsigned long long *thr_ptr,
+			       unsigned long long *out5,
+			       gg_q3_result_row *out,
+			       unsigned long long *out_count, uint64_t cap,
+			       unsigned long long *overflow);
 hipError_t launch_dn_q3_stats(hipStream_t s, const unsigned long long *pay,
 			      const unsigned long long *rev,
 			      int64_t dense_len, unsigned long long *out5);

@@ -190,10 +190,19 @@ launch_q1(hipStream_t s, const int32_t *shipdate, const uint8_t *rflag,
 	if (gs)
 	{
 		int64_t cap = (n + THREADS - 1) / THREADS;
+		/* per-block partials are u64; rows-per-block × max charge6
+		 * (~1.19e11, scale 6) must stay < 2^64 (DESIGN.md overflow
+		 * budget), so a forced grid may not drop below this floor */
+		int64_t floor_g = (int64_t) (((unsigned __int128) n *
+					      120000000000ull) / ~0ull) + 1;
 		int g = atoi(gs);
 
 		if (g > 0)
+		{
+			if (g < floor_g)
+				g = (int) floor_g;
 			grid = (int) (g < cap ? g : cap);
+		}
 	}
 	if (nt)
 		hipLaunchKernelGGL(k_q1_agg<true>, dim3(grid), dim3(THREADS),

@@ -41,6 +41,44 @@ static inline int dn_grid(int64_t n)
 	return (int) (b < 1 ? 1 : (b > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : b));
 }
 
+/* grid for XCD-chunked kernels: multiple of 8 so each XCD gets the
+ * same number of blocks */
+static inline int dn_grid8(int64_t n)
+{
+	int g = dn_grid(n);
+
+	return (g + 7) & ~7;
+}
+
+/* Contiguous per-XCD chunking (microarch guide: block b runs on XCD
+ * b % 8, each XCD has a private 4 MiB L2).  Both lineitem and the
+ * dense orderkey-indexed arrays are orderkey-ordered, so giving XCD x
+ * the contiguous row range [x*chunk, (x+1)*chunk) makes its slice of
+ * the membership bitmap (2.3 MB at SF100) L2-resident and keeps the
+ * rev[] atomics XCD-local — a grid-stride loop instead drags the whole
+ * 18.75 MB bitmap through every XCD's L2. */
+struct DnChunk
+{
+	int64_t base;		/* first index of this block's lane 0 */
+	int64_t hi;		/* this XCD chunk's end (exclusive) */
+	int64_t stride;		/* per-iteration advance */
+};
+
+__device__ inline DnChunk dn_chunk(int64_t n)
+{
+	const int xcd = blockIdx.x & 7;
+	const int lb = blockIdx.x >> 3;
+	const int nlb = gridDim.x >> 3;	/* grid is a multiple of 8 */
+	const int64_t chunk = (n + 7) >> 3;
+	const int64_t lo = (int64_t) xcd * chunk;
+	DnChunk c;
+
+	c.hi = lo + chunk < n ? lo + chunk : n;
+	c.base = lo + (int64_t) lb * blockDim.x;
+	c.stride = (int64_t) nlb * blockDim.x;
+	return c;
+}
+
 __device__ inline int64_t dn_ld64(const int64_t *p)
 {
 	return __builtin_nontemporal_load(p);
@@ -155,27 +193,16 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 		  unsigned long long *__restrict__ bloom, uint64_t bwords,
 		  unsigned long long *match_count)
 {
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	DnChunk c = dn_chunk(n);
 	unsigned long long matches = 0;
-	/* round the per-lane range up so every lane of a wave executes
-	 * the same number of iterations (the wave-aggregated bitmap OR
-	 * below needs a convergent wave) */
-	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	const int64_t n_up = ((n + stride - 1) / stride) * stride;
 
-	for (int64_t i = i0; i < n_up; i += stride)
+	/* one fully-evaluated row; returns ok so the caller can do the
+	 * wave-aggregated bitmap OR convergently */
+	auto row = [&](int64_t i, bool live, int32_t d, int64_t ck,
+		       int64_t k) -> bool
 	{
-		bool ok = i < n;
-		int32_t d = 0;
-		int64_t ck = 0, k = -1;
+		bool ok = live && d < cutoff;
 
-		if (ok)
-		{
-			d = dn_ld32(&odate[i]);
-			ck = dn_ld64(&ckey[i]);
-			k = dn_ld64(&okey[i]);
-			ok = d < cutoff;
-		}
 		if (ok)
 		{
 			if (cust_bits)
@@ -232,7 +259,45 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 				((unsigned long long) (uint32_t) prio[i]
 				 << 32), &pay[k]);
 		}
-		dn_bit_set_wave(bloom, k, ok);
+		return ok;
+	};
+	const int64_t S = c.stride;
+	int64_t base = c.base;
+
+	/* lane-uniform loop bounds keep waves convergent for the
+	 * wave-aggregated bitmap OR; per-lane validity is the live flag */
+	for (; base + 3 * S + blockDim.x <= c.hi; base += 4 * S)
+	{
+		int64_t i0 = base + threadIdx.x;
+		int32_t d0 = dn_ld32(&odate[i0]);
+		int32_t d1 = dn_ld32(&odate[i0 + S]);
+		int32_t d2 = dn_ld32(&odate[i0 + 2 * S]);
+		int32_t d3 = dn_ld32(&odate[i0 + 3 * S]);
+		int64_t c0 = dn_ld64(&ckey[i0]);
+		int64_t c1 = dn_ld64(&ckey[i0 + S]);
+		int64_t c2 = dn_ld64(&ckey[i0 + 2 * S]);
+		int64_t c3 = dn_ld64(&ckey[i0 + 3 * S]);
+		int64_t k0 = dn_ld64(&okey[i0]);
+		int64_t k1 = dn_ld64(&okey[i0 + S]);
+		int64_t k2 = dn_ld64(&okey[i0 + 2 * S]);
+		int64_t k3 = dn_ld64(&okey[i0 + 3 * S]);
+
+		dn_bit_set_wave(bloom, k0, row(i0, true, d0, c0, k0));
+		dn_bit_set_wave(bloom, k1, row(i0 + S, true, d1, c1, k1));
+		dn_bit_set_wave(bloom, k2,
+				row(i0 + 2 * S, true, d2, c2, k2));
+		dn_bit_set_wave(bloom, k3,
+				row(i0 + 3 * S, true, d3, c3, k3));
+	}
+	for (; base < c.hi; base += S)
+	{
+		int64_t i = base + threadIdx.x;
+		bool live = i < c.hi;
+		int32_t d = live ? dn_ld32(&odate[i]) : 0;
+		int64_t ck = live ? dn_ld64(&ckey[i]) : 0;
+		int64_t k = live ? dn_ld64(&okey[i]) : -1;
+
+		dn_bit_set_wave(bloom, k, row(i, live, d, ck, k));
 	}
 	for (int off = 32; off; off >>= 1)
 		matches += __shfl_down(matches, off, 64);
@@ -250,7 +315,7 @@ launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 		       int64_t dense_len, unsigned long long *bloom,
 		       uint64_t bwords, unsigned long long *match_count)
 {
-	hipLaunchKernelGGL(k_dn_build_orders, dim3(dn_grid(n)),
+	hipLaunchKernelGGL(k_dn_build_orders, dim3(dn_grid8(n)),
 			   dim3(DN_THREADS), 0, s, okey, ckey, odate, prio, n,
 			   cutoff, cust.keys, cust.nslots, cust.bloom,
 			   cust.bloom_words, cust_bits, cust_dlen, pay,
@@ -291,11 +356,12 @@ launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-/* Q3 lineitem probe against the dense orders map.  PMC shows ~7.5 GB
- * moved per launch at ~3.6 TB/s effective — below the 6.3 TB/s
- * streaming limit because of the random-access components; occupancy
- * hints (2 vs 8 blocks/CU) measured identical, so the limiter is the
- * access pattern, not resident-wave count. */
+/* Q3 lineitem probe against the dense orders map.  Round-1 PMC showed
+ * this kernel at ~3.5 TB/s effective with only ~12 B/row of sequential
+ * traffic (okey+shipdate): the limiter was memory-level parallelism
+ * (2 NT streams/iteration) and cross-XCD bitmap traffic, not occupancy.
+ * v2: 4-way unrolled streams (8 loads in flight per lane) + XCD
+ * chunking so each XCD's bitmap/rev slice stays in its private L2. */
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 			 const int32_t *__restrict__ shipdate,
@@ -308,25 +374,42 @@ void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 			 const unsigned long long *__restrict__ bloom,
 			 uint64_t bwords, unsigned long long *join_rows)
 {
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	DnChunk c = dn_chunk(n);
 	unsigned long long joined = 0;
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
+	auto body = [&](int64_t i, int32_t sd, int64_t k)
 	{
-		int32_t sd = dn_ld32(&shipdate[i]);
-		int64_t k = dn_ld64(&okey[i]);
-
 		if (sd <= cutoff)	/* qual: l_shipdate > cutoff */
-			continue;
+			return;
 		if (k < 0 || k >= dense_len)
-			continue;
+			return;
 		if (!dn_bit_test(bloom, k))	/* exact: no false positives */
-			continue;
+			return;
 		joined++;
 		atomicAdd(&rev[k],
 			  (unsigned long long) (price[i] * (100 - disc[i])));
+	};
+	const int64_t S = c.stride;
+	int64_t i = c.base + threadIdx.x;
+
+	for (; i + 3 * S < c.hi; i += 4 * S)
+	{
+		int32_t sd0 = dn_ld32(&shipdate[i]);
+		int32_t sd1 = dn_ld32(&shipdate[i + S]);
+		int32_t sd2 = dn_ld32(&shipdate[i + 2 * S]);
+		int32_t sd3 = dn_ld32(&shipdate[i + 3 * S]);
+		int64_t k0 = dn_ld64(&okey[i]);
+		int64_t k1 = dn_ld64(&okey[i + S]);
+		int64_t k2 = dn_ld64(&okey[i + 2 * S]);
+		int64_t k3 = dn_ld64(&okey[i + 3 * S]);
+
+		body(i, sd0, k0);
+		body(i + S, sd1, k1);
+		body(i + 2 * S, sd2, k2);
+		body(i + 3 * S, sd3, k3);
 	}
+	for (; i < c.hi; i += S)
+		body(i, dn_ld32(&shipdate[i]), dn_ld64(&okey[i]));
 	for (int off = 32; off; off >>= 1)
 		joined += __shfl_down(joined, off, 64);
 	if ((threadIdx.x & 63) == 0 && joined)
@@ -360,7 +443,9 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 int64_t dense_len, unsigned long long *bloom,
 			 uint64_t bwords, unsigned long long *join_rows)
 {
-	hipLaunchKernelGGL(k_dn_probe_lineitem, dim3(dn_grid_env(n, "GG_Q3_PROBE_GRID")),
+	int g = (dn_grid_env(n, "GG_Q3_PROBE_GRID") + 7) & ~7;
+
+	hipLaunchKernelGGL(k_dn_probe_lineitem, dim3(g),
 			   dim3(DN_THREADS), 0, s, okey, shipdate, price,
 			   disc, n, cutoff, pay, rev, dense_len, bloom,
 			   bwords, join_rows);
@@ -503,32 +588,31 @@ void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 		((unsigned long long *) lds)[i] = 0;
 	__syncthreads();
 
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	DnChunk c = dn_chunk(n);
 	unsigned long long joined = 0;
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < n; i += stride)
+	/* filter order: the orders map passes ~3% of rows (date range x
+	 * in-region customer) and its reads are nearly sequential
+	 * (lineitem is orderkey-ordered; XCD chunking keeps each XCD's
+	 * map slice local), so check it before touching any other column
+	 * — skey/price/disc lines are then only fetched for passing
+	 * lanes.  4-way unrolled okey stream for load-level parallelism. */
+	auto body = [&](int64_t i, int64_t k)
 	{
-		/* filter order: the orders map passes ~3%% of rows (date
-		 * range x in-region customer) and its reads are nearly
-		 * sequential (lineitem is orderkey-ordered), so check it
-		 * before touching any other column — skey/price/disc lines
-		 * are then only fetched for passing lanes. */
-		int64_t k = dn_ld64(&okey[i]);
 		unsigned snat, onat;
 
 		if (k < 0 || k >= dense_len)
-			continue;
+			return;
 		onat = pay8[k];
 		if (onat == 255)
-			continue;
+			return;
 		int64_t sk = skey[i];
 
 		if (sk < 0 || sk >= supp_dlen)
-			continue;
+			return;
 		snat = supp_dense[sk];
 		if (snat != onat)	/* 255 (absent) never equals 0..24 */
-			continue;
+			return;
 		joined++;
 		{
 			unsigned long long rev4 = (unsigned long long)
@@ -537,7 +621,24 @@ void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 			atomicAdd(&lds[snat][0], 1ull);
 			atomicAdd(&lds[snat][1], rev4);
 		}
+	};
+	const int64_t S = c.stride;
+	int64_t i = c.base + threadIdx.x;
+
+	for (; i + 3 * S < c.hi; i += 4 * S)
+	{
+		int64_t k0 = dn_ld64(&okey[i]);
+		int64_t k1 = dn_ld64(&okey[i + S]);
+		int64_t k2 = dn_ld64(&okey[i + 2 * S]);
+		int64_t k3 = dn_ld64(&okey[i + 3 * S]);
+
+		body(i, k0);
+		body(i + S, k1);
+		body(i + 2 * S, k2);
+		body(i + 3 * S, k3);
 	}
+	for (; i < c.hi; i += S)
+		body(i, dn_ld64(&okey[i]));
 	for (int off = 32; off; off >>= 1)
 		joined += __shfl_down(joined, off, 64);
 	if ((threadIdx.x & 63) == 0 && joined)
@@ -570,7 +671,7 @@ launch_dn_probe_lineitem_q5_u8(hipStream_t s, const int64_t *okey,
 			       unsigned long long *join_rows)
 {
 	hipLaunchKernelGGL(k_dn_probe_lineitem_q5_u8,
-			   dim3(dn_grid_env(n, "GG_Q5_PROBE_GRID")),
+			   dim3((dn_grid_env(n, "GG_Q5_PROBE_GRID") + 7) & ~7),
 			   dim3(DN_THREADS), 0, s, okey, skey, price, disc,
 			   n, pay8, dense_len, supp_dense, supp_dlen, acc,
 			   join_rows);
@@ -729,6 +830,278 @@ launch_dn_q5_gather(hipStream_t s, const unsigned long long *comp,
 			   dim3(DN_THREADS), 0, s, comp, counts, region,
 			   nregions, skey, price, disc, supp_dense,
 			   supp_dlen, acc, join_rows, overflow);
+	return hipGetLastError();
+}
+
+/* ---- fused single-sweep top-k (v2) --------------------------------
+ * Replaces the stats→hist→collect triple sweep of the 1.2 GB rev[]
+ * array (the reference's bounded-heap switch, tuplesort.c:1360–1377)
+ * with ONE sweep: k_dn_q3_pass1 computes ngroups/revsum/max, builds a
+ * 16-bit monotonic exponent-mantissa histogram (no dependence on max,
+ * so no pre-pass), and compacts survivor indices into per-block
+ * regions (LDS counter — a single global counter saturates at ~88
+ * returning atomics/µs).  k_dn_q3_threshold2 picks the revenue
+ * threshold from the histogram; k_dn_q3_finish walks only the ~ngroups
+ * survivors (not the dense array) to compute the group checksum and
+ * collect candidates ≥ threshold. */
+
+/* monotonic 16-bit code of a u64 (r > 0): 6-bit exponent bucket + 10
+ * mantissa bits; order-preserving, max code 55 295 < 65 536 */
+__device__ inline unsigned dn_code16(unsigned long long r)
+{
+	int e = 63 - __clzll(r);
+
+	if (e <= 9)
+		return (unsigned) r;
+	return (unsigned) (((unsigned) (e - 9) << 10) |
+			   ((r >> (e - 10)) & 1023));
+}
+
+/* smallest u64 whose code is `code` (threshold decode) */
+__device__ inline unsigned long long dn_code16_lo(unsigned code)
+{
+	if (code < 1024)
+		return code;
+	{
+		unsigned b = code >> 10;
+		unsigned m = code & 1023;
+
+		return (1024ull + m) << (b - 1);
+	}
+}
+
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
+		   int64_t dense_len, unsigned long long *__restrict__ out5,
+		   unsigned int *__restrict__ hist64k,
+		   unsigned long long *__restrict__ surv, int64_t region,
+		   unsigned long long *__restrict__ counts)
+{
+	__shared__ unsigned long long lcnt;
+
+	if (threadIdx.x == 0)
+		lcnt = 0;
+	__syncthreads();
+
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long ng = 0, revsum = 0, carry = 0, mx = 0;
+	unsigned long long *mine = surv + (int64_t) blockIdx.x * region;
+
+	auto body = [&](int64_t i, unsigned long long r)
+	{
+		if (!r)
+			return;
+		ng++;
+		mx = max(mx, r);
+		{
+			unsigned long long old = revsum;
+
+			revsum += r;
+			carry += (revsum < old);
+		}
+		atomicAdd(&hist64k[dn_code16(r)], 1u);
+		{
+			unsigned long long at = atomicAdd(&lcnt, 1ull);
+
+			if (at < (unsigned long long) region)
+				mine[at] = (unsigned long long) i;
+		}
+	};
+	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	int64_t i = i0;
+
+	for (; i + 3 * stride < dense_len; i += 4 * stride)
+	{
+		unsigned long long r0 = __builtin_nontemporal_load(&rev[i]);
+		unsigned long long r1 =
+			__builtin_nontemporal_load(&rev[i + stride]);
+		unsigned long long r2 =
+			__builtin_nontemporal_load(&rev[i + 2 * stride]);
+		unsigned long long r3 =
+			__builtin_nontemporal_load(&rev[i + 3 * stride]);
+
+		body(i, r0);
+		body(i + stride, r1);
+		body(i + 2 * stride, r2);
+		body(i + 3 * stride, r3);
+	}
+	for (; i < dense_len; i += stride)
+		body(i, __builtin_nontemporal_load(&rev[i]));
+	for (int off = 32; off; off >>= 1)
+	{
+		unsigned long long orev = revsum;
+
+		ng += __shfl_down(ng, off, 64);
+		revsum += __shfl_down(revsum, off, 64);
+		carry += __shfl_down(carry, off, 64) + (revsum < orev);
+		mx = max(mx, __shfl_down(mx, off, 64));
+	}
+	if ((threadIdx.x & 63) == 0 && ng)
+	{
+		atomicAdd(&out5[0], ng);
+		{
+			unsigned long long old = atomicAdd(&out5[1], revsum);
+
+			if (old + revsum < old)
+				atomicAdd(&out5[2], 1ull);
+			atomicAdd(&out5[2], carry);
+		}
+		atomicMax(&out5[4], mx);
+	}
+	__syncthreads();
+	if (threadIdx.x == 0)
+		counts[blockIdx.x] = lcnt;
+}
+
+hipError_t
+launch_dn_q3_pass1(hipStream_t s, const unsigned long long *rev,
+		   int64_t dense_len, unsigned long long *out5,
+		   unsigned int *hist64k, unsigned long long *surv,
+		   int64_t region, unsigned long long *counts, int *out_grid)
+{
+	int g = dn_grid(dense_len);
+
+	*out_grid = g;
+	hipLaunchKernelGGL(k_dn_q3_pass1, dim3(g), dim3(DN_THREADS), 0, s,
+			   rev, dense_len, out5, hist64k, surv, region,
+			   counts);
+	return hipGetLastError();
+}
+
+/* one block: walk the 16-bit-code histogram from the top until >= k
+ * candidates; emit the decoded revenue threshold */
+__global__ __launch_bounds__(1024)
+void k_dn_q3_threshold2(const unsigned int *__restrict__ hist64k, int64_t k,
+			unsigned long long *__restrict__ out_thr)
+{
+	__shared__ unsigned long long partial[1024];
+	unsigned long long s = 0;
+
+	for (int b = 0; b < 64; b++)
+		s += hist64k[threadIdx.x * 64 + b];
+	partial[threadIdx.x] = s;
+	__syncthreads();
+	if (threadIdx.x == 0)
+	{
+		unsigned long long cum = 0;
+		int chunk = 1023;
+
+		for (; chunk > 0; chunk--)
+		{
+			if (cum + partial[chunk] >= (unsigned long long) k)
+				break;
+			cum += partial[chunk];
+		}
+		{
+			int b = 63;
+
+			for (; b > 0; b--)
+			{
+				unsigned long long c =
+					hist64k[chunk * 64 + b];
+
+				if (cum + c >= (unsigned long long) k)
+					break;
+				cum += c;
+			}
+			{
+				unsigned long long thr = dn_code16_lo(
+					(unsigned) (chunk * 64 + b));
+
+				*out_thr = thr ? thr : 1;
+			}
+		}
+	}
+}
+
+hipError_t
+launch_dn_q3_threshold2(hipStream_t s, const unsigned int *hist64k, int64_t k,
+			unsigned long long *out_thr)
+{
+	hipLaunchKernelGGL(k_dn_q3_threshold2, dim3(1), dim3(1024), 0, s,
+			   hist64k, k, out_thr);
+	return hipGetLastError();
+}
+
+/* walk the compacted survivors: group checksum (gathers pay only for
+ * real groups) + threshold collect.  overflow is set when any region
+ * spilled (host falls back to the full-sweep path). */
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_q3_finish(const unsigned long long *__restrict__ surv,
+		    const unsigned long long *__restrict__ counts,
+		    int64_t region, int64_t nregions,
+		    const unsigned long long *__restrict__ pay,
+		    const unsigned long long *__restrict__ rev,
+		    const unsigned long long *__restrict__ thr_ptr,
+		    unsigned long long *__restrict__ out5,
+		    gg_q3_result_row *__restrict__ out,
+		    unsigned long long *out_count, uint64_t cap,
+		    unsigned long long *overflow)
+{
+	const unsigned long long threshold = *thr_ptr;
+	unsigned long long ck = 0;
+
+	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
+	{
+		unsigned long long m = counts[b];
+
+		if (m > (unsigned long long) region)
+		{
+			if (threadIdx.x == 0)
+				atomicOr(overflow, 1ull);
+			m = (unsigned long long) region;
+		}
+		const unsigned long long *seg = surv + b * region;
+
+		for (unsigned long long j = threadIdx.x; j < m;
+		     j += blockDim.x)
+		{
+			int64_t i = (int64_t) seg[j];
+			unsigned long long r = rev[i];
+			unsigned long long p = pay[i];
+			int32_t date = (int32_t) (uint32_t) p;
+			int32_t prio = (int32_t) (uint32_t) (p >> 32);
+
+			ck += gg_group_hash((unsigned long long) i, r, 0,
+					    date, prio);
+			{
+				bool take = r >= threshold;
+				unsigned long long idx =
+					dn_wave_append(out_count, take);
+
+				if (take && idx < cap)
+				{
+					out[idx].orderkey = i;
+					out[idx].rev_lo = r;
+					out[idx].rev_hi = 0;
+					out[idx].orderdate = date;
+					out[idx].shippriority = prio;
+				}
+			}
+		}
+	}
+	for (int off = 32; off; off >>= 1)
+		ck += __shfl_down(ck, off, 64);
+	if ((threadIdx.x & 63) == 0 && ck)
+		atomicAdd(&out5[3], ck);
+}
+
+hipError_t
+launch_dn_q3_finish(hipStream_t s, const unsigned long long *surv,
+		    const unsigned long long *counts, int64_t region,
+		    int64_t nregions, const unsigned long long *pay,
+		    const unsigned long long *rev,
+		    const unsigned long long *thr_ptr,
+		    unsigned long long *out5, gg_q3_result_row *out,
+		    unsigned long long *out_count, uint64_t cap,
+		    unsigned long long *overflow)
+{
+	int g = (int) (nregions < 1 ? 1 :
+		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
+
+	hipLaunchKernelGGL(k_dn_q3_finish, dim3(g), dim3(DN_THREADS), 0, s,
+			   surv, counts, region, nregions, pay, rev, thr_ptr,
+			   out5, out, out_count, cap, overflow);
 	return hipGetLastError();
 }
 
