@@ -1507,6 +1507,9 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		    (int64_t) maxk <= 8 * cu->nrows + 16)
 			cust_dlen = (int64_t) maxk + 1;
 	}
+	/* record which build-side layout ran (nodeHash.c:450 sizing
+	 * decision analog) so EXPLAIN-style stats show the taken path */
+	p->stat(cust_dlen ? "path_cust_bitmap" : "path_cust_hash").launches++;
 	{
 		Timed tm(e.stream);
 
@@ -1583,6 +1586,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		    (int64_t) maxk <= 8 * od->nrows + 16)
 			ord_dlen = (int64_t) maxk + 1;
 	}
+	p->stat(ord_dlen ? "path_orders_dense" : "path_orders_hash").launches++;
 	if (ord_dlen)
 	{
 		/* exact membership bitmap over dense keys */
@@ -2281,6 +2285,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		    (int64_t) maxk <= 8 * cu->nrows + 16)
 			cust_dlen = (int64_t) maxk + 1;
 	}
+	p->stat(cust_dlen ? "path_cust_dense" : "path_cust_hash").launches++;
 	{
 		Timed tm(e.stream);
 
@@ -2549,6 +2554,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		    (int64_t) maxk <= 8 * od->nrows + 16)
 			ord_dlen = (int64_t) maxk + 1;
 	}
+	p->stat(ord_dlen ? "path_orders_dense" : "path_orders_hash").launches++;
 	if (ord_dlen)
 	{
 		/* nation fits a byte; 255 = no matching order.  The u8 map

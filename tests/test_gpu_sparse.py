@@ -1,0 +1,105 @@
+"""Sparse-key hash-table fallback.
+
+The Q3/Q5 pipelines collapse to dense direct maps when max(key) <=
+8x rows (nodeHash.c:450 sizing analog).  These tests feed keys ~1000x
+sparser than rows so the guard REJECTS the dense map and the
+open-addressing CAS hash tables (k_build_set / k_build_orders /
+k_probe_lineitem, q3 stats/hist/collect on DeviceHashTable) run —
+asserted bit-exact against the CPU oracle, with the taken path
+reported by gg_engine_stats.
+"""
+import numpy as np
+import pytest
+
+import pyoracle
+
+pytestmark = pytest.mark.gpu
+
+SEED = 7
+
+
+@pytest.fixture(scope="module")
+def eng():
+    from greengage_amd import Engine
+    e = Engine(device=0, n_segments=1, segment_id=0)
+    yield e
+    e.shutdown()
+
+
+def make_sparse(ncust=5_000, nord=40_000, lines_per_order=4, stride=1009):
+    """TPC-H-shaped tables whose keys are `stride`x sparser than rows
+    (stride is prime so key sets stay collision-free)."""
+    rng = np.random.default_rng(SEED)
+    c_custkey = (np.arange(ncust, dtype=np.int64) + 1) * stride
+    c_mktseg = rng.integers(0, 5, ncust).astype(np.uint8)
+    o_orderkey = (np.arange(nord, dtype=np.int64) + 1) * stride
+    o_custkey = c_custkey[rng.integers(0, ncust, nord)]
+    o_orderdate = rng.integers(-2000, 500, nord).astype(np.int32)
+    o_prio = rng.integers(0, 2, nord).astype(np.int32)
+    nli = nord * lines_per_order
+    l_orderkey = np.repeat(o_orderkey, lines_per_order)
+    l_shipdate = rng.integers(-2000, 500, nli).astype(np.int32)
+    l_price = rng.integers(90100, 10494950, nli).astype(np.int64)
+    l_disc = rng.integers(0, 11, nli).astype(np.int64)
+    return (c_custkey, c_mktseg, o_orderkey, o_custkey, o_orderdate,
+            o_prio, l_orderkey, l_shipdate, l_price, l_disc)
+
+
+def test_q3_sparse_keys_hash_fallback(eng):
+    from greengage_amd.engine import PIPE_Q3
+    (c_ck, c_ms, o_ok, o_ck, o_od, o_pr,
+     l_ok, l_sd, l_pc, l_dc) = make_sparse()
+
+    cu = eng.register_table("cust_sparse", [
+        ("custkey", "int64", c_ck), ("mktseg", "char1", c_ms)], len(c_ck))
+    od = eng.register_table("ord_sparse", [
+        ("orderkey", "int64", o_ok), ("custkey", "int64", o_ck),
+        ("orderdate", "int32", o_od), ("shippriority", "int32", o_pr)],
+        len(o_ok))
+    li = eng.register_table("li_sparse", [
+        ("orderkey", "int64", l_ok), ("shipdate", "int32", l_sd),
+        ("price", "dec64", l_pc), ("disc", "dec64", l_dc)], len(l_ok))
+
+    cutoff = -800
+    p = eng.compile(PIPE_Q3, lineitem=li, orders=od, customer=cu,
+                    cutoff_date=cutoff, mktsegment=2, limit_k=10)
+    rows, hdr = eng.execute_q3(p)
+
+    # the guard must have rejected the dense map on BOTH sides
+    paths = {s["name"] for s in eng.stats(p) if s["name"].startswith("path")}
+    assert "path_cust_hash" in paths, paths
+    assert "path_orders_hash" in paths, paths
+    assert "path_cust_bitmap" not in paths
+    assert "path_orders_dense" not in paths
+
+    topk, res = pyoracle.q3_arrays(
+        c_ck, c_ms, 2, o_ok, o_ck, o_od, o_pr,
+        l_ok, l_sd, l_pc, l_dc, cutoff)
+    assert hdr["n_groups"] == res["n_groups"]
+    assert hdr["n_join_rows"] == res["n_join_rows"]
+    assert hdr["rev_sum4"] == res["rev_sum4"]
+    assert hdr["group_checksum"] == res["group_checksum"]
+    assert [(r["orderkey"], r["revenue4"], r["orderdate"],
+             r["shippriority"]) for r in rows] == \
+           [(r["orderkey"], r["revenue4"], r["orderdate"],
+             r["shippriority"]) for r in topk]
+
+
+def test_q3_dense_keys_report_dense_path(eng):
+    """Control: dbgen-shaped dense keys must keep the dense map (and
+    still match the oracle), so the guard is pinned from both sides."""
+    from greengage_amd import PGDate
+    from greengage_amd.engine import PIPE_Q3
+    li = eng.register_synth("lineitem", seed=42, sf=1)
+    od = eng.register_synth("orders", seed=42, sf=1)
+    cu = eng.register_synth("customer", seed=42, sf=1)
+    cutoff = PGDate("1995-03-15")
+    p = eng.compile(PIPE_Q3, lineitem=li, orders=od, customer=cu,
+                    cutoff_date=cutoff, mktsegment=2, limit_k=10)
+    rows, hdr = eng.execute_q3(p)
+    paths = {s["name"] for s in eng.stats(p) if s["name"].startswith("path")}
+    assert "path_cust_bitmap" in paths, paths
+    assert "path_orders_dense" in paths, paths
+    _, res = pyoracle.q3_synth(42, 1, cutoff, k=10)
+    assert hdr["n_groups"] == res["n_groups"]
+    assert hdr["group_checksum"] == res["group_checksum"]
