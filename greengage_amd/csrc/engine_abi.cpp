@@ -2100,6 +2100,10 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.total_ms += ms;
 		st.rows_in += (int64_t) hstats[0];
 		st.rows_out += (int64_t) cand.size();
+		/* dominant traffic: one 8 B/slot sweep of the dense rev
+		 * array (survivor finish pass is ~ngroups gathers) */
+		st.hbm_bytes += ord_dlen ? ord_dlen * 8
+			: (int64_t) ord.nslots * 16;
 
 		/* 5. global combine (the Gather Motion to the QD:
 		 * nodeMotion gather + final-stage combine) */
@@ -2535,6 +2539,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		st.launches++;
 		st.total_ms += ms;
 		st.rows_in += su->nrows;
+		st.hbm_bytes += su->nrows * 9;	/* suppkey 8 + nation 1 */
 	}
 
 	/* 3. orders → okey → c_nationkey map.  Dense o_orderkey (with a
@@ -2587,6 +2592,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 			st0.total_ms += ms0;
 			st0.rows_in += od->nrows;
 			st0.rows_out += (int64_t) nmatch;
+			st0.hbm_bytes += od->nrows * 20;  /* okey,ckey,odate */
 			goto orders_done;
 		}
 		if (!p->ord_slots)
@@ -2629,6 +2635,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 			st.total_ms += ms;
 			st.rows_in += od->nrows;
 			st.rows_out += (int64_t) nmatch;
+			st.hbm_bytes += od->nrows * 20;  /* okey,ckey,odate */
 		}
 orders_done:;
 	}

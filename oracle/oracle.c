@@ -10,6 +10,7 @@
  */
 #include <stdio.h>
 #include <stdlib.h>
+#include <stddef.h>
 #include <string.h>
 
 #include "../include/gg_pg_hash.h"
@@ -230,6 +231,270 @@ gg_oracle_q1_synth_segment(uint64_t seed, int64_t sf, int32_t nseg,
 {
 	return q1_synth_range(seed, 0, gg_n_lineitem(sf), cutoff, nseg, seg,
 			      out, nthreads);
+}
+
+/* ---------------- Q1, Volcano restatement (the "cpu-ref" leg) --------
+ * Models the reference's tuple-at-a-time executor for the Q1 slice
+ * (BASELINE.md cpu-ref): one ExecProcNode-style INDIRECT call per
+ * tuple (execProcnode.c:925), slot deform into a Datum array
+ * (slot_deform_tuple heaptuple.c:1119), an interpreted expression
+ * walk per row for the qual (ExecQual execQual.c:6260) and each agg
+ * input (ExecTargetList :6369), a per-row group hash + bucket-chain
+ * compare (execHHashagg.c:157 calc_hash_value / :456
+ * lookup_agg_hash_entry), and one indirect transition-function call
+ * per agg per row (invoke_agg_trans_func nodeAgg.c:413) — the
+ * structure whose per-row costs the batched GPU engine removes.
+ * Results are bit-identical to the vectorized oracle leg. */
+
+enum
+{
+	VE_VAR,			/* slot[attno] */
+	VE_CONST,
+	VE_LE,			/* l <= r */
+	VE_SUB100_RHS,		/* l * (100 - r): mul_var scale 2+2 */
+	VE_ADD100_RHS		/* l * (100 + r): mul_var scale 4+2 */
+};
+
+typedef struct vexpr
+{
+	int op;
+	int attno;
+	int64_t cval;
+	const struct vexpr *l, *r;
+} vexpr;
+
+static int64_t
+vexpr_eval(const vexpr *e, const int64_t *slot)
+{
+	switch (e->op)
+	{
+		case VE_VAR:
+			return slot[e->attno];
+		case VE_CONST:
+			return e->cval;
+		case VE_LE:
+			return vexpr_eval(e->l, slot) <= vexpr_eval(e->r, slot);
+		case VE_SUB100_RHS:
+			return vexpr_eval(e->l, slot) *
+				(100 - vexpr_eval(e->r, slot));
+		case VE_ADD100_RHS:
+			return vexpr_eval(e->l, slot) *
+				(100 + vexpr_eval(e->r, slot));
+	}
+	return 0;
+}
+
+/* slot layout (deformed tuple) */
+enum { VA_QTY, VA_PRICE, VA_DISC, VA_TAX, VA_SHIPDATE, VA_RFLAG, VA_LSTATUS,
+       VA_NATTS };
+
+typedef struct vscan
+{
+	uint64_t seed;
+	int64_t next, hi;
+	const vexpr *qual;
+	int64_t slot[VA_NATTS];
+} vscan;
+
+/* SeqNext + ExecQual loop (execScan.c:110–214): returns the deformed
+ * slot of the next qual-passing tuple, NULL at end of scan */
+static const int64_t *
+vscan_next(void *p)
+{
+	vscan *s = (vscan *) p;
+
+	while (s->next < s->hi)
+	{
+		gg_lineitem_row r;
+
+		gg_gen_lineitem(s->seed, s->next++, &r);
+		s->slot[VA_QTY] = r.l_quantity_c;
+		s->slot[VA_PRICE] = r.l_extendedprice_c;
+		s->slot[VA_DISC] = r.l_discount_c;
+		s->slot[VA_TAX] = r.l_tax_c;
+		s->slot[VA_SHIPDATE] = r.l_shipdate;
+		s->slot[VA_RFLAG] = r.l_returnflag;
+		s->slot[VA_LSTATUS] = r.l_linestatus;
+		if (vexpr_eval(s->qual, s->slot))
+			return s->slot;
+	}
+	return NULL;
+}
+
+/* per-agg transition state + indirect transition fn (nodeAgg.c:413) */
+typedef void (*vtrans_fn)(void *state, int64_t datum);
+
+static void vt_count(void *s, int64_t d) { (void) d; *(int64_t *) s += 1; }
+static void vt_sum64(void *s, int64_t d) { *(int64_t *) s += d; }
+static void vt_sum128(void *s, int64_t d) { *(i128 *) s += d; }
+
+typedef struct vagg
+{
+	vtrans_fn fn;
+	const vexpr *arg;	/* NULL for count(*) */
+	size_t state_off;	/* into the group entry */
+} vagg;
+
+typedef struct vgroup
+{
+	uint8_t rf, ls;
+	int used;
+	int64_t count, sum_qty, sum_base, sum_dcol;
+	i128 sum_disc, sum_charge;
+	struct vgroup *next;	/* bucket chain */
+} vgroup;
+
+#define VHASH_NBUCKETS 16
+
+typedef struct vhash
+{
+	vgroup *buckets[VHASH_NBUCKETS];
+	vgroup pool[GG_Q1_NGROUPS + 2];
+	int npool;
+} vhash;
+
+/* calc_hash_value (execHHashagg.c:157): FunctionCall1-per-key hashes
+ * combined with the rotate-left-xor the reference applies per column */
+static vgroup *
+vhash_lookup(vhash *h, uint8_t rf, uint8_t ls)
+{
+	uint32_t hv = gg_hashchar((char) rf);
+
+	hv = ((hv << 1) | (hv >> 31)) ^ gg_hashchar((char) ls);
+	{
+		vgroup *g = h->buckets[hv & (VHASH_NBUCKETS - 1)];
+
+		for (; g; g = g->next)
+			if (g->rf == rf && g->ls == ls)
+				return g;
+		if (h->npool >= GG_Q1_NGROUPS + 2)
+			return NULL;
+		g = &h->pool[h->npool++];
+		memset(g, 0, sizeof(*g));
+		g->rf = rf;
+		g->ls = ls;
+		g->used = 1;
+		g->next = h->buckets[hv & (VHASH_NBUCKETS - 1)];
+		h->buckets[hv & (VHASH_NBUCKETS - 1)] = g;
+		return g;
+	}
+}
+
+static int
+q1_volcano_range(uint64_t seed, int64_t row_lo, int64_t row_hi,
+		 int32_t cutoff, q1_acc *acc)
+{
+	/* plan setup (ExecutorStart analog): qual + agg input exprs */
+	vexpr v_ship = {VE_VAR, VA_SHIPDATE, 0, NULL, NULL};
+	vexpr v_cut = {VE_CONST, 0, cutoff, NULL, NULL};
+	vexpr qual = {VE_LE, 0, 0, &v_ship, &v_cut};
+	vexpr v_qty = {VE_VAR, VA_QTY, 0, NULL, NULL};
+	vexpr v_price = {VE_VAR, VA_PRICE, 0, NULL, NULL};
+	vexpr v_disc = {VE_VAR, VA_DISC, 0, NULL, NULL};
+	vexpr v_tax = {VE_VAR, VA_TAX, 0, NULL, NULL};
+	vexpr e_disc4 = {VE_SUB100_RHS, 0, 0, &v_price, &v_disc};
+	vexpr e_charge6 = {VE_ADD100_RHS, 0, 0, &e_disc4, &v_tax};
+	vagg aggs[6] = {
+		{vt_count, NULL, offsetof(vgroup, count)},
+		{vt_sum64, &v_qty, offsetof(vgroup, sum_qty)},
+		{vt_sum64, &v_price, offsetof(vgroup, sum_base)},
+		{vt_sum64, &v_disc, offsetof(vgroup, sum_dcol)},
+		{vt_sum128, &e_disc4, offsetof(vgroup, sum_disc)},
+		{vt_sum128, &e_charge6, offsetof(vgroup, sum_charge)},
+	};
+	vscan scan;
+	vhash hash;
+	/* indirect child call per pull, as ExecProcNode dispatches */
+	const int64_t *(*volatile child)(void *) = vscan_next;
+
+	memset(&hash, 0, sizeof(hash));
+	scan.seed = seed;
+	scan.next = row_lo;
+	scan.hi = row_hi;
+	scan.qual = &qual;
+
+	for (;;)
+	{
+		const int64_t *slot = child(&scan);
+
+		if (!slot)
+			break;
+		{
+			vgroup *g = vhash_lookup(&hash,
+						 (uint8_t) slot[VA_RFLAG],
+						 (uint8_t) slot[VA_LSTATUS]);
+
+			if (!g)
+				return -1;
+			for (int a = 0; a < 6; a++)
+			{
+				int64_t d = aggs[a].arg
+					? vexpr_eval(aggs[a].arg, slot) : 0;
+
+				aggs[a].fn((char *) g + aggs[a].state_off, d);
+			}
+		}
+	}
+	for (int i = 0; i < hash.npool; i++)
+	{
+		vgroup *g = &hash.pool[i];
+		int s = q1_slot(g->rf, g->ls);
+
+		if (s < 0)
+			return -1;
+		acc->count[s] += g->count;
+		acc->sum_qty[s] += g->sum_qty;
+		acc->sum_base[s] += g->sum_base;
+		acc->sum_dcol[s] += g->sum_dcol;
+		acc->sum_disc[s] += g->sum_disc;
+		acc->sum_charge[s] += g->sum_charge;
+	}
+	return 0;
+}
+
+int
+gg_oracle_q1_volcano_synth(uint64_t seed, int64_t sf, int64_t row_lo,
+			   int64_t row_hi, int32_t cutoff,
+			   gg_q1_group out[GG_Q1_NGROUPS], int nthreads)
+{
+	q1_acc total;
+	int bad = 0;
+
+	if (row_hi < 0)
+		row_hi = gg_n_lineitem(sf);
+	memset(&total, 0, sizeof(total));
+	set_threads(nthreads);
+	if (nthreads == 1)
+		bad = q1_volcano_range(seed, row_lo, row_hi, cutoff, &total);
+	else
+	{
+		/* parallel variant: one Volcano pipeline per thread over a
+		 * contiguous row range (1 segment per worker), partials
+		 * merged — the reference's N-segments-on-one-host shape */
+#pragma omp parallel reduction(|:bad)
+		{
+			q1_acc local;
+			int64_t nth = 1, tid = 0;
+
+			memset(&local, 0, sizeof(local));
+#ifdef _OPENMP
+			nth = omp_get_num_threads();
+			tid = omp_get_thread_num();
+#endif
+			{
+				int64_t n = row_hi - row_lo;
+				int64_t lo = row_lo + n * tid / nth;
+				int64_t hi = row_lo + n * (tid + 1) / nth;
+
+				bad |= q1_volcano_range(seed, lo, hi, cutoff,
+							&local);
+			}
+#pragma omp critical
+			q1_acc_merge(&total, &local);
+		}
+	}
+	q1_acc_out(&total, out);
+	return bad ? -1 : 0;
 }
 
 /* ---------------- numeric formatting / AVG ---------------- */

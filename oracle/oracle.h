@@ -78,6 +78,16 @@ int gg_oracle_q1_synth_segment(uint64_t seed, int64_t sf, int32_t nseg,
 			       int32_t seg, int32_t cutoff,
 			       gg_q1_group out[GG_Q1_NGROUPS], int nthreads);
 
+/* Tuple-at-a-time Volcano restatement of the same slice (the
+ * BASELINE.md "cpu-ref" executor leg): per-row indirect ExecProcNode
+ * call, interpreted quals/projections, per-row group hash + chain
+ * compare, indirect transition call per agg per row.  nthreads == 1
+ * runs one pipeline; otherwise one pipeline per OpenMP thread over
+ * contiguous row ranges, partials merged.  Bit-identical results. */
+int gg_oracle_q1_volcano_synth(uint64_t seed, int64_t sf, int64_t row_lo,
+			       int64_t row_hi, int32_t cutoff,
+			       gg_q1_group out[GG_Q1_NGROUPS], int nthreads);
+
 /* ---- numeric finalization (reference numeric.c semantics) ---- */
 
 /* "1234.5600" for int128 {lo,hi} at scale; buf >= 64 bytes */

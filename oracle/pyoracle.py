@@ -121,6 +121,9 @@ def lib():
         L.gg_oracle_q1_synth.restype = ctypes.c_int
         L.gg_oracle_q1_synth.argtypes = [
             U64, I64, I64, I64, I32, ctypes.POINTER(Q1Group), ctypes.c_int]
+        L.gg_oracle_q1_volcano_synth.restype = ctypes.c_int
+        L.gg_oracle_q1_volcano_synth.argtypes = [
+            U64, I64, I64, I64, I32, ctypes.POINTER(Q1Group), ctypes.c_int]
         L.gg_oracle_q1_synth_segment.restype = ctypes.c_int
         L.gg_oracle_q1_synth_segment.argtypes = [
             U64, I64, I32, I32, I32, ctypes.POINTER(Q1Group), ctypes.c_int]
@@ -569,6 +572,15 @@ def q1_synth(seed, sf, cutoff, row_lo=0, row_hi=-1, nthreads=0):
     out = (Q1Group * 6)()
     rc = lib().gg_oracle_q1_synth(seed, sf, row_lo, row_hi, cutoff, out,
                                   nthreads)
+    assert rc == 0, rc
+    return [g.as_dict() for g in out]
+
+
+def q1_volcano_synth(seed, sf, cutoff, row_lo=0, row_hi=-1, nthreads=1):
+    """Tuple-at-a-time Volcano restatement (BASELINE.md cpu-ref leg)."""
+    out = (Q1Group * 6)()
+    rc = lib().gg_oracle_q1_volcano_synth(seed, sf, row_lo, row_hi, cutoff,
+                                          out, nthreads)
     assert rc == 0, rc
     return [g.as_dict() for g in out]
 
