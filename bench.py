@@ -75,9 +75,11 @@ def cpu_baseline_leg(cutoff):
         lambda hi: pyoracle.q1_volcano_synth(SEED, SF, cutoff, 0, hi,
                                              nthreads=1),
         cutoff, 6.0, 1_000_000)
+    # explicit thread count: omp_set_num_threads is process-global, so
+    # the 1-thread leg above would otherwise pin this one to 1 thread
     volmt_rate, volmt_sample, volmt_dt = _timed_leg(
         lambda hi: pyoracle.q1_volcano_synth(SEED, SF, cutoff, 0, hi,
-                                             nthreads=0),
+                                             nthreads=cores),
         cutoff, 6.0, 4_000_000)
     return {
         "value": vec_rate,

@@ -1599,10 +1599,20 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			p->sget("ordd.bloom", ordd_bwords * 8);
 		if (!ordd_pay || !ordd_rev || !ordd_bloom)
 			return fail(GG_ENOMEM, "ord dense");
-		GG_HIP(hipMemsetAsync(ordd_pay, 0xff, (size_t) ord_dlen * 8,
-				      e.stream));
-		GG_HIP(hipMemsetAsync(ordd_rev, 0, (size_t) ord_dlen * 8,
-				      e.stream));
+		/* pay[] needs NO init: entries are read only behind the
+		 * exact membership bitmap, which is rebuilt every pass,
+		 * so any read slot was stored this pass.  rev[] is fully
+		 * zeroed only on first use / after a fallback pass; the
+		 * fused top-k otherwise zeroes exactly the entries the
+		 * probe touched (k_dn_q3_clear over the survivor list). */
+		if (!p->q3_rev_inited || p->q3_rev_dirty)
+		{
+			GG_HIP(hipMemsetAsync(ordd_rev, 0,
+					      (size_t) ord_dlen * 8,
+					      e.stream));
+			p->q3_rev_inited = true;
+		}
+		p->q3_rev_dirty = true;	/* until the sparse clear runs */
 		GG_HIP(hipMemsetAsync(ordd_bloom, 0, ordd_bwords * 8,
 				      e.stream));
 	}
@@ -2015,6 +2025,17 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				GG_HIP(hipMemcpy(cand.data(), dout,
 						 ncand * sizeof(gg_q3_result_row),
 						 hipMemcpyDeviceToHost));
+			}
+			if (!need_old)
+			{
+				/* zero only the touched rev[] entries so
+				 * the next execute skips the 1.2 GB
+				 * memset (runs async; later work queues
+				 * behind it on the engine stream) */
+				GG_HIP(launch_dn_q3_clear(
+					e.stream, dsurv, dcnts, region,
+					pgrid, ordd_rev));
+				p->q3_rev_dirty = false;
 			}
 		}
 

@@ -115,6 +115,12 @@ struct Pipeline
 	 * computed once for the guarded dense-map sizing, not per step */
 	std::vector<std::pair<std::pair<const void *, int64_t>,
 			      unsigned long long>> maxk_cache;
+	/* Q3 dense rev[] lifecycle: the 1.2 GB accumulator is fully
+	 * memset only when freshly allocated or left dirty (fallback
+	 * path); otherwise the previous execute's survivor sweep zeroed
+	 * exactly the touched entries */
+	bool q3_rev_inited = false;
+	bool q3_rev_dirty = false;
 
 	KernelStatAcc &stat(const char *name)
 	{
@@ -506,6 +512,11 @@ hipError_t launch_dn_q3_pass1(hipStream_t s, const unsigned long long *rev,
 hipError_t launch_dn_q3_threshold2(hipStream_t s,
 				   const unsigned int *hist64k, int64_t k,
 				   unsigned long long *out_thr);
+hipError_t launch_dn_q3_clear(hipStream_t s,
+			      const unsigned long long *surv,
+			      const unsigned long long *counts,
+			      int64_t region, int64_t nregions,
+			      unsigned long long *rev);
 hipError_t launch_dn_q3_finish(hipStream_t s,
 			       const unsigned long long *surv,
 			       const unsigned long long *counts,

@@ -79,6 +79,44 @@ __device__ inline DnChunk dn_chunk(int64_t n)
 	return c;
 }
 
+/* Quad variant: each lane owns 4 CONSECUTIVE rows so the column
+ * streams are read with 16-byte vector loads (the microarch guide
+ * measures 8-B NT accesses at only 0.54–0.70x the 16-B rate; a 4-B
+ * int32 stream is worse still).  Chunk starts are 4-row aligned so
+ * longlong2/int4 loads stay 16-B aligned. */
+__device__ inline DnChunk dn_chunk4(int64_t n)
+{
+	const int xcd = blockIdx.x & 7;
+	const int lb = blockIdx.x >> 3;
+	const int nlb = gridDim.x >> 3;
+	const int64_t chunk = (((n + 7) >> 3) + 3) & ~(int64_t) 3;
+	const int64_t lo = (int64_t) xcd * chunk;
+	DnChunk c;
+
+	c.hi = lo + chunk < n ? lo + chunk : n;
+	if (c.hi < lo)
+		c.hi = lo;
+	c.base = lo + (int64_t) lb * blockDim.x * 4;
+	c.stride = (int64_t) nlb * blockDim.x * 4;
+	return c;
+}
+
+/* clang ext_vector types (HIP_vector_type is a class the nontemporal
+ * builtin rejects) */
+typedef long long gg_ll2 __attribute__((ext_vector_type(2)));
+typedef int gg_i4 __attribute__((ext_vector_type(4)));
+typedef unsigned long long gg_ull2 __attribute__((ext_vector_type(2)));
+
+__device__ inline gg_ll2 dn_ld2(const int64_t *p)
+{
+	return __builtin_nontemporal_load((const gg_ll2 *) p);
+}
+
+__device__ inline gg_i4 dn_ld4x32(const int32_t *p)
+{
+	return __builtin_nontemporal_load((const gg_i4 *) p);
+}
+
 __device__ inline int64_t dn_ld64(const int64_t *p)
 {
 	return __builtin_nontemporal_load(p);
@@ -148,20 +186,20 @@ dn_bit_set(unsigned long long *bm, int64_t key)
  * the leader.  Must be executed by ALL lanes of the wave (pass
  * active=false for lanes with nothing to set). */
 __device__ inline void
-dn_bit_set_wave(unsigned long long *bm, int64_t key, bool active)
+dn_bit_set_wave_mask(unsigned long long *bm, unsigned long long w,
+		     unsigned long long mask, bool active)
 {
-	unsigned long long w = active ? (unsigned long long) (key >> 6)
-		: ~0ull;
-	unsigned long long bit = active ? (1ull << (key & 63)) : 0;
 	unsigned long long alive = __ballot(active);
 	int lane = (int) (threadIdx.x & 63);
 
+	if (!active)
+		w = ~0ull;
 	while (alive)
 	{
 		int leader = __ffsll((long long) alive) - 1;
 		unsigned long long lw = __shfl(w, leader, 64);
 		bool same = active && (w == lw);
-		unsigned long long v = same ? bit : 0;
+		unsigned long long v = same ? mask : 0;
 
 		for (int off = 32; off; off >>= 1)
 			v |= __shfl_xor(v, off, 64);
@@ -171,6 +209,13 @@ dn_bit_set_wave(unsigned long long *bm, int64_t key, bool active)
 	}
 }
 
+__device__ inline void
+dn_bit_set_wave(unsigned long long *bm, int64_t key, bool active)
+{
+	dn_bit_set_wave_mask(bm, (unsigned long long) (key >> 6),
+			     active ? (1ull << (key & 63)) : 0, active);
+}
+
 __device__ inline bool
 dn_bit_test(const unsigned long long *__restrict__ bm, int64_t key)
 {
@@ -178,7 +223,7 @@ dn_bit_test(const unsigned long long *__restrict__ bm, int64_t key)
 }
 
 /* Q3 orders build: date filter + customer membership → pay store */
-__global__ __launch_bounds__(DN_THREADS, 8)
+__global__ __launch_bounds__(DN_THREADS, 4)
 void k_dn_build_orders(const int64_t *__restrict__ okey,
 		  const int64_t *__restrict__ ckey,
 		  const int32_t *__restrict__ odate,
@@ -193,7 +238,7 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 		  unsigned long long *__restrict__ bloom, uint64_t bwords,
 		  unsigned long long *match_count)
 {
-	DnChunk c = dn_chunk(n);
+	DnChunk c = dn_chunk4(n);
 	unsigned long long matches = 0;
 
 	/* one fully-evaluated row; returns ok so the caller can do the
@@ -262,42 +307,75 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 		return ok;
 	};
 	const int64_t S = c.stride;
-	int64_t base = c.base;
 
-	/* lane-uniform loop bounds keep waves convergent for the
-	 * wave-aggregated bitmap OR; per-lane validity is the live flag */
-	for (; base + 3 * S + blockDim.x <= c.hi; base += 4 * S)
+	/* block-uniform outer loop keeps every wave converged for the
+	 * wave-aggregated bitmap OR (its butterfly reduction reads all
+	 * 64 lanes); per-lane/element validity is the live flag.  Each
+	 * lane owns 4 consecutive rows → 16-B vector loads, and its
+	 * quad's bitmap bits land in at most 2 words, merged lane-
+	 * locally before ONE wave aggregation per word. */
+	for (int64_t rb = c.base; rb < c.hi; rb += S)
 	{
-		int64_t i0 = base + threadIdx.x;
-		int32_t d0 = dn_ld32(&odate[i0]);
-		int32_t d1 = dn_ld32(&odate[i0 + S]);
-		int32_t d2 = dn_ld32(&odate[i0 + 2 * S]);
-		int32_t d3 = dn_ld32(&odate[i0 + 3 * S]);
-		int64_t c0 = dn_ld64(&ckey[i0]);
-		int64_t c1 = dn_ld64(&ckey[i0 + S]);
-		int64_t c2 = dn_ld64(&ckey[i0 + 2 * S]);
-		int64_t c3 = dn_ld64(&ckey[i0 + 3 * S]);
-		int64_t k0 = dn_ld64(&okey[i0]);
-		int64_t k1 = dn_ld64(&okey[i0 + S]);
-		int64_t k2 = dn_ld64(&okey[i0 + 2 * S]);
-		int64_t k3 = dn_ld64(&okey[i0 + 3 * S]);
+		int64_t r = rb + (int64_t) threadIdx.x * 4;
+		bool quad = r + 3 < c.hi;
+		int32_t d[4];
+		int64_t ck[4], k[4];
+		int nrows;
 
-		dn_bit_set_wave(bloom, k0, row(i0, true, d0, c0, k0));
-		dn_bit_set_wave(bloom, k1, row(i0 + S, true, d1, c1, k1));
-		dn_bit_set_wave(bloom, k2,
-				row(i0 + 2 * S, true, d2, c2, k2));
-		dn_bit_set_wave(bloom, k3,
-				row(i0 + 3 * S, true, d3, c3, k3));
-	}
-	for (; base < c.hi; base += S)
-	{
-		int64_t i = base + threadIdx.x;
-		bool live = i < c.hi;
-		int32_t d = live ? dn_ld32(&odate[i]) : 0;
-		int64_t ck = live ? dn_ld64(&ckey[i]) : 0;
-		int64_t k = live ? dn_ld64(&okey[i]) : -1;
+		if (quad)
+		{
+			gg_i4 dd = dn_ld4x32(&odate[r]);
+			gg_ll2 c01 = dn_ld2(&ckey[r]);
+			gg_ll2 c23 = dn_ld2(&ckey[r + 2]);
+			gg_ll2 k01 = dn_ld2(&okey[r]);
+			gg_ll2 k23 = dn_ld2(&okey[r + 2]);
 
-		dn_bit_set_wave(bloom, k, row(i, live, d, ck, k));
+			d[0] = dd.x; d[1] = dd.y; d[2] = dd.z; d[3] = dd.w;
+			ck[0] = c01.x; ck[1] = c01.y;
+			ck[2] = c23.x; ck[3] = c23.y;
+			k[0] = k01.x; k[1] = k01.y;
+			k[2] = k23.x; k[3] = k23.y;
+			nrows = 4;
+		}
+		else
+		{
+			nrows = (int) (r < c.hi ? c.hi - r : 0);
+			for (int j = 0; j < nrows; j++)
+			{
+				d[j] = dn_ld32(&odate[r + j]);
+				ck[j] = dn_ld64(&ckey[r + j]);
+				k[j] = dn_ld64(&okey[r + j]);
+			}
+		}
+		{
+			unsigned long long w0 = ~0ull, m0 = 0;
+			unsigned long long w1 = ~0ull, m1 = 0;
+
+			for (int j = 0; j < nrows; j++)
+				if (row(r + j, true, d[j], ck[j], k[j]))
+				{
+					unsigned long long w =
+						(unsigned long long)
+						(k[j] >> 6);
+					unsigned long long bit =
+						1ull << (k[j] & 63);
+
+					if (!m0 || w0 == w)
+					{
+						w0 = w;
+						m0 |= bit;
+					}
+					else if (!m1 || w1 == w)
+					{
+						w1 = w;
+						m1 |= bit;
+					}
+					else	/* >2 words per quad: rare */
+						dn_bit_set(bloom, k[j]);
+				}
+			dn_bit_set_wave_mask(bloom, w0, m0, m0 != 0);
+			dn_bit_set_wave_mask(bloom, w1, m1, m1 != 0);
+		}
 	}
 	for (int off = 32; off; off >>= 1)
 		matches += __shfl_down(matches, off, 64);
@@ -359,10 +437,10 @@ launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 /* Q3 lineitem probe against the dense orders map.  Round-1 PMC showed
  * this kernel at ~3.5 TB/s effective with only ~12 B/row of sequential
  * traffic (okey+shipdate): the limiter was memory-level parallelism
- * (2 NT streams/iteration) and cross-XCD bitmap traffic, not occupancy.
- * v2: 4-way unrolled streams (8 loads in flight per lane) + XCD
+ * and narrow NT loads, not occupancy.  v2: 4 consecutive rows per lane
+ * read with 16-B vector NT loads (3 loads per quad instead of 8) + XCD
  * chunking so each XCD's bitmap/rev slice stays in its private L2. */
-__global__ __launch_bounds__(DN_THREADS, 8)
+__global__ __launch_bounds__(DN_THREADS, 4)
 void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 			 const int32_t *__restrict__ shipdate,
 			 const int64_t *__restrict__ price,
@@ -374,7 +452,7 @@ void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 			 const unsigned long long *__restrict__ bloom,
 			 uint64_t bwords, unsigned long long *join_rows)
 {
-	DnChunk c = dn_chunk(n);
+	DnChunk c = dn_chunk4(n);
 	unsigned long long joined = 0;
 
 	auto body = [&](int64_t i, int32_t sd, int64_t k)
@@ -390,26 +468,26 @@ void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 			  (unsigned long long) (price[i] * (100 - disc[i])));
 	};
 	const int64_t S = c.stride;
-	int64_t i = c.base + threadIdx.x;
 
-	for (; i + 3 * S < c.hi; i += 4 * S)
+	for (int64_t r = c.base + (int64_t) threadIdx.x * 4; r < c.hi;
+	     r += S)
 	{
-		int32_t sd0 = dn_ld32(&shipdate[i]);
-		int32_t sd1 = dn_ld32(&shipdate[i + S]);
-		int32_t sd2 = dn_ld32(&shipdate[i + 2 * S]);
-		int32_t sd3 = dn_ld32(&shipdate[i + 3 * S]);
-		int64_t k0 = dn_ld64(&okey[i]);
-		int64_t k1 = dn_ld64(&okey[i + S]);
-		int64_t k2 = dn_ld64(&okey[i + 2 * S]);
-		int64_t k3 = dn_ld64(&okey[i + 3 * S]);
+		if (r + 3 < c.hi)
+		{
+			gg_i4 sd = dn_ld4x32(&shipdate[r]);
+			gg_ll2 k01 = dn_ld2(&okey[r]);
+			gg_ll2 k23 = dn_ld2(&okey[r + 2]);
 
-		body(i, sd0, k0);
-		body(i + S, sd1, k1);
-		body(i + 2 * S, sd2, k2);
-		body(i + 3 * S, sd3, k3);
+			body(r, sd.x, k01.x);
+			body(r + 1, sd.y, k01.y);
+			body(r + 2, sd.z, k23.x);
+			body(r + 3, sd.w, k23.y);
+		}
+		else
+			for (int64_t j = r; j < c.hi; j++)
+				body(j, dn_ld32(&shipdate[j]),
+				     dn_ld64(&okey[j]));
 	}
-	for (; i < c.hi; i += S)
-		body(i, dn_ld32(&shipdate[i]), dn_ld64(&okey[i]));
 	for (int off = 32; off; off >>= 1)
 		joined += __shfl_down(joined, off, 64);
 	if ((threadIdx.x & 63) == 0 && joined)
@@ -570,7 +648,7 @@ launch_dn_insert_orders_q5_u8(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-__global__ __launch_bounds__(DN_THREADS, 8)
+__global__ __launch_bounds__(DN_THREADS, 4)
 void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 			  const int64_t *__restrict__ skey,
 			  const int64_t *__restrict__ price,
@@ -588,7 +666,7 @@ void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 		((unsigned long long *) lds)[i] = 0;
 	__syncthreads();
 
-	DnChunk c = dn_chunk(n);
+	DnChunk c = dn_chunk4(n);
 	unsigned long long joined = 0;
 
 	/* filter order: the orders map passes ~3% of rows (date range x
@@ -596,7 +674,7 @@ void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 	 * (lineitem is orderkey-ordered; XCD chunking keeps each XCD's
 	 * map slice local), so check it before touching any other column
 	 * — skey/price/disc lines are then only fetched for passing
-	 * lanes.  4-way unrolled okey stream for load-level parallelism. */
+	 * lanes.  4 consecutive rows per lane → 16-B vector loads. */
 	auto body = [&](int64_t i, int64_t k)
 	{
 		unsigned snat, onat;
@@ -623,22 +701,24 @@ void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 		}
 	};
 	const int64_t S = c.stride;
-	int64_t i = c.base + threadIdx.x;
 
-	for (; i + 3 * S < c.hi; i += 4 * S)
+	for (int64_t r = c.base + (int64_t) threadIdx.x * 4; r < c.hi;
+	     r += S)
 	{
-		int64_t k0 = dn_ld64(&okey[i]);
-		int64_t k1 = dn_ld64(&okey[i + S]);
-		int64_t k2 = dn_ld64(&okey[i + 2 * S]);
-		int64_t k3 = dn_ld64(&okey[i + 3 * S]);
+		if (r + 3 < c.hi)
+		{
+			gg_ll2 k01 = dn_ld2(&okey[r]);
+			gg_ll2 k23 = dn_ld2(&okey[r + 2]);
 
-		body(i, k0);
-		body(i + S, k1);
-		body(i + 2 * S, k2);
-		body(i + 3 * S, k3);
+			body(r, k01.x);
+			body(r + 1, k01.y);
+			body(r + 2, k23.x);
+			body(r + 3, k23.y);
+		}
+		else
+			for (int64_t j = r; j < c.hi; j++)
+				body(j, dn_ld64(&okey[j]));
 	}
-	for (; i < c.hi; i += S)
-		body(i, dn_ld64(&okey[i]));
 	for (int off = 32; off; off >>= 1)
 		joined += __shfl_down(joined, off, 64);
 	if ((threadIdx.x & 63) == 0 && joined)
@@ -870,7 +950,7 @@ __device__ inline unsigned long long dn_code16_lo(unsigned code)
 	}
 }
 
-__global__ __launch_bounds__(DN_THREADS, 8)
+__global__ __launch_bounds__(DN_THREADS, 4)
 void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
 		   int64_t dense_len, unsigned long long *__restrict__ out5,
 		   unsigned int *__restrict__ hist64k,
@@ -907,26 +987,29 @@ void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
 				mine[at] = (unsigned long long) i;
 		}
 	};
-	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	int64_t i = i0;
+	/* 4 consecutive slots per lane → two 16-B NT loads per quad */
+	const int64_t qstride = stride * 4;
 
-	for (; i + 3 * stride < dense_len; i += 4 * stride)
+	for (int64_t r = ((int64_t) blockIdx.x * blockDim.x +
+			  threadIdx.x) * 4;
+	     r < dense_len; r += qstride)
 	{
-		unsigned long long r0 = __builtin_nontemporal_load(&rev[i]);
-		unsigned long long r1 =
-			__builtin_nontemporal_load(&rev[i + stride]);
-		unsigned long long r2 =
-			__builtin_nontemporal_load(&rev[i + 2 * stride]);
-		unsigned long long r3 =
-			__builtin_nontemporal_load(&rev[i + 3 * stride]);
+		if (r + 3 < dense_len)
+		{
+			gg_ull2 v01 = __builtin_nontemporal_load(
+				(const gg_ull2 *) &rev[r]);
+			gg_ull2 v23 = __builtin_nontemporal_load(
+				(const gg_ull2 *) &rev[r + 2]);
 
-		body(i, r0);
-		body(i + stride, r1);
-		body(i + 2 * stride, r2);
-		body(i + 3 * stride, r3);
+			body(r, v01.x);
+			body(r + 1, v01.y);
+			body(r + 2, v23.x);
+			body(r + 3, v23.y);
+		}
+		else
+			for (int64_t j = r; j < dense_len; j++)
+				body(j, __builtin_nontemporal_load(&rev[j]));
 	}
-	for (; i < dense_len; i += stride)
-		body(i, __builtin_nontemporal_load(&rev[i]));
 	for (int off = 32; off; off >>= 1)
 	{
 		unsigned long long orev = revsum;
@@ -1102,6 +1185,43 @@ launch_dn_q3_finish(hipStream_t s, const unsigned long long *surv,
 	hipLaunchKernelGGL(k_dn_q3_finish, dim3(g), dim3(DN_THREADS), 0, s,
 			   surv, counts, region, nregions, pay, rev, thr_ptr,
 			   out5, out, out_count, cap, overflow);
+	return hipGetLastError();
+}
+
+/* zero exactly the rev[] entries this pass touched (the survivor
+ * list), so the next execute skips the full 1.2 GB memset */
+__global__ __launch_bounds__(DN_THREADS, 4)
+void k_dn_q3_clear(const unsigned long long *__restrict__ surv,
+		   const unsigned long long *__restrict__ counts,
+		   int64_t region, int64_t nregions,
+		   unsigned long long *__restrict__ rev)
+{
+	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
+	{
+		unsigned long long m = counts[b];
+
+		if (m > (unsigned long long) region)
+			m = (unsigned long long) region;
+		{
+			const unsigned long long *seg = surv + b * region;
+
+			for (unsigned long long j = threadIdx.x; j < m;
+			     j += blockDim.x)
+				rev[seg[j]] = 0;
+		}
+	}
+}
+
+hipError_t
+launch_dn_q3_clear(hipStream_t s, const unsigned long long *surv,
+		   const unsigned long long *counts, int64_t region,
+		   int64_t nregions, unsigned long long *rev)
+{
+	int g = (int) (nregions < 1 ? 1 :
+		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
+
+	hipLaunchKernelGGL(k_dn_q3_clear, dim3(g), dim3(DN_THREADS), 0, s,
+			   surv, counts, region, nregions, rev);
 	return hipGetLastError();
 }
 

@@ -415,6 +415,124 @@ gg_status gg_engine_hash_groupby_i64_spill(const int64_t *keys,
 gg_status gg_engine_radix_sort_u64(uint64_t *keys, uint64_t *payload_or_null,
 				   int64_t n, int key_bytes, int descending);
 
+/* ---- generalized pipeline descriptor (v2) ----------------------------
+ * The compositional form of the subtree a shim extracts from the
+ * PlanState tree (execProcnode.c:925–1148): a driving scan with
+ * conjunctive range predicates (ExecQual execQual.c:6260 over the
+ * strategy-number operator classes), zero or more hash SEMI-join
+ * filters against filtered build tables (nodeHash.c:88 build /
+ * nodeHashjoin.c:78 probe, join-qual = key equality), a hash group-by
+ * over 0–2 key columns (execHHashagg.c:905), and COUNT/SUM aggregate
+ * transitions whose inputs are products of up to three column factors
+ * with the (100±col) scaled-decimal modifiers TPC-H money expressions
+ * need (numeric.c:1735 mul_var dscale rule; ExecTargetList
+ * execQual.c:6369).  Q1/Q3/Q5-shaped descriptors keep their
+ * specialized kernels via gg_engine_compile_pipeline; everything else
+ * — mpph6 (Q6), changed predicate columns, ad-hoc scans — compiles
+ * here with ZERO query-specific kernels.  Unsupported shapes return
+ * GG_ENOTSUP (shim falls back to standard_ExecutorRun). */
+
+#define GG_PLAN_MAX_PREDS 8
+#define GG_PLAN_MAX_JOINS 2
+#define GG_PLAN_MAX_AGGS 8
+
+/* conjunct: lo <= col < hi (INT64_MIN/INT64_MAX = one-sided; equality
+ * = [v, v+1)).  A NULL column value fails the qual (SQL three-valued
+ * logic: WHERE discards non-true, execScan.c:185). */
+typedef struct gg_plan_pred
+{
+	const char *col;
+	int64_t lo;		/* inclusive */
+	int64_t hi;		/* exclusive */
+} gg_plan_pred;
+
+typedef struct gg_plan_scan
+{
+	gg_table table;
+	gg_plan_pred preds[GG_PLAN_MAX_PREDS];
+	int npreds;
+} gg_plan_scan;
+
+/* hash semi-join filter: driving rows survive iff probe_key matches
+ * the build_key of some build-side row passing the build preds */
+typedef struct gg_plan_join
+{
+	gg_plan_scan build;
+	const char *build_key;
+	const char *probe_key;	/* column of the driving table */
+} gg_plan_join;
+
+/* aggregate: COUNT(*) / COUNT(col) (non-NULL, nodeAgg strict rules) /
+ * SUM(f0*f1*f2), factor = col, (100-col) or (100+col); SUM skips rows
+ * whose factor inputs are NULL (strict transition, nodeAgg.c:413–460) */
+typedef enum gg_plan_aggkind
+{
+	GG_AGG_COUNT_STAR = 0,
+	GG_AGG_COUNT_COL = 1,
+	GG_AGG_SUM = 2
+} gg_plan_aggkind;
+
+typedef enum gg_plan_fmod
+{
+	GG_FMOD_ID = 0,		/* col */
+	GG_FMOD_SUB100 = 1,	/* 100 - col */
+	GG_FMOD_ADD100 = 2	/* 100 + col */
+} gg_plan_fmod;
+
+typedef struct gg_plan_agg
+{
+	int kind;		/* gg_plan_aggkind */
+	int nfactors;		/* 0 for COUNT(*); 1 for COUNT(col) */
+	const char *col[3];
+	int8_t mod[3];		/* gg_plan_fmod */
+} gg_plan_agg;
+
+typedef struct gg_plan_desc
+{
+	gg_plan_scan scan;	/* driving (probe-side) scan */
+	gg_plan_join joins[GG_PLAN_MAX_JOINS];
+	int njoins;
+	/* 0 = one global group; 1 = any char1/int32/int64 column;
+	 * 2 = two char1 columns (code = c0*256 + c1).  Group keys of
+	 * INT64_MIN are rejected (open-addressing sentinel); NULL keys
+	 * group together (execHHashagg.c:531), reported as
+	 * GG_PLAN_NULL_KEY. */
+	const char *group_cols[2];
+	int ngroup;
+	gg_plan_agg aggs[GG_PLAN_MAX_AGGS];
+	int naggs;
+} gg_plan_desc;
+
+#define GG_PLAN_NULL_KEY (-9223372036854775807LL)	/* INT64_MIN+1 */
+
+/* result arena layout (little-endian):
+ *   int64 n_groups; int32 naggs; int32 ngroup;
+ *   then n_groups rows, sorted by (key0, key1) ascending:
+ *     int64 key0, int64 key1, then naggs x (uint64 lo, int64 hi)
+ *     (each aggregate as a signed 128-bit value) */
+gg_status gg_engine_compile_plan(const gg_plan_desc *desc, gg_pipeline *out);
+
+/* Attach a NULL-flag array (1 byte per row, nonzero = NULL) to a
+ * registered column — the nullable-scan surface (AO null bitmaps land
+ * here; aocsam.c:699 per-column null arrays).  Predicates, group keys
+ * and aggregate transitions then follow the strict-transition /
+ * NULL-key rules above. */
+gg_status gg_engine_table_set_nulls(gg_table t, const char *col,
+				    const uint8_t *host_nulls);
+
+/* NULL-aware general hash group-by (execHHashagg find-or-create +
+ * nodeAgg.c:413 strict transitions): NULL keys form one group
+ * (reported as GG_PLAN_NULL_KEY); NULL vals are skipped by both SUM
+ * and COUNT (the strict SUM/AVG transition pair).  key_nulls/
+ * val_nulls may be NULL (= no NULLs).  Keys INT64_MIN rejected. */
+gg_status gg_engine_hash_groupby_i64_n(const int64_t *keys,
+				       const uint8_t *key_nulls,
+				       const int64_t *vals,
+				       const uint8_t *val_nulls, int64_t n,
+				       int64_t *out_keys, int64_t *out_sums,
+				       int64_t *out_counts, int64_t cap,
+				       int64_t *out_ngroups);
+
 /* build info: "gfx950" etc. — lets callers assert the native path */
 const char *gg_engine_build_info(void);
 
