@@ -133,6 +133,14 @@ cached_max_i64(Engine &e, Pipeline *p, const int64_t *col, int64_t n,
 	return GG_OK;
 }
 
+/* shared with plan.cpp */
+gg_status
+engine_cached_max_i64(Engine &e, Pipeline *p, const int64_t *col, int64_t n,
+		      unsigned long long *ctr, unsigned long long *out)
+{
+	return cached_max_i64(e, p, col, n, ctr, out);
+}
+
 /* ---------------- int128 host helpers (combine/finalize) ---------------- */
 
 typedef __int128 i128;
@@ -911,6 +919,12 @@ static Table *get_table(gg_table h)
 	if (h < 0 || (size_t) h >= e.tables.size())
 		return nullptr;
 	return e.tables[h];
+}
+
+/* shared with plan.cpp */
+Table *engine_table(gg_table h)
+{
+	return get_table(h);
 }
 
 extern "C" gg_status
@@ -3028,6 +3042,8 @@ gg_engine_execute(gg_pipeline h, void *arena, size_t bytes, size_t *written)
 		return fail(GG_EINVAL, "null arena");
 	Pipeline *p = e.pipelines[h];
 
+	if ((int) p->desc.kind == GG_PIPE_PLAN_INTERNAL)
+		return exec_plan(p, arena, bytes, written);
 	switch (p->desc.kind)
 	{
 		case GG_PIPE_Q1:

@@ -8,6 +8,7 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <memory>
 #include <string>
 #include <vector>
 
@@ -70,6 +71,9 @@ struct Table
 		gg_coltype type;
 		void *dev = nullptr;
 		size_t bytes = 0;
+		/* optional device NULL-flag array (byte per row) —
+		 * gg_engine_table_set_nulls / nullable AO mounts */
+		void *nulls = nullptr;
 		/* for dictionary-encoded text columns: the sorted dict
 		 * (host-side; categorical, small) */
 		std::vector<uint8_t> dict_bytes;
@@ -84,6 +88,14 @@ struct Table
 				return c.dev;
 		return nullptr;
 	}
+
+	Col *find(const char *n)
+	{
+		for (auto &c : cols)
+			if (c.name == n)
+				return &c;
+		return nullptr;
+	}
 };
 
 struct KernelStatAcc
@@ -96,9 +108,15 @@ struct KernelStatAcc
 	int64_t hbm_bytes = 0;
 };
 
+/* internal pipeline kind for generic compiled plans */
+#define GG_PIPE_PLAN_INTERNAL 99
+
 struct Pipeline
 {
 	gg_pipeline_desc desc;
+	/* resolved generic plan (plan.cpp PlanResolved) when desc.kind
+	 * == GG_PIPE_PLAN_INTERNAL; shared_ptr carries the deleter */
+	std::shared_ptr<void> plan;
 	std::vector<KernelStatAcc> stats;
 
 	/* named device scratch reused across executes (hash tables,
@@ -429,6 +447,86 @@ hipError_t launch_td_insert(hipStream_t s, const uint8_t *pool,
 hipError_t launch_td_map(hipStream_t s, const uint32_t *rowslot,
 			 int64_t n, const int32_t *slot_to_id,
 			 int32_t *codes);
+
+/* ---- generic compiled plans (plan.hip / plan.cpp) ---- */
+
+struct PlanPredDev
+{
+	const void *col;
+	const uint8_t *nulls;
+	int width;
+	int64_t lo, hi;
+};
+
+struct PlanJoinDev
+{
+	const void *pkey;
+	const uint8_t *pnulls;
+	int width;
+	const unsigned long long *bits;	/* dense membership bitmap, or */
+	int64_t dlen;
+	const unsigned long long *hkeys;	/* hash semi-set (key^msb, 0=empty) */
+	uint64_t hslots;
+};
+
+struct PlanAggDev
+{
+	int kind;		/* gg_plan_aggkind */
+	int nf;
+	const void *col[3];
+	const uint8_t *nulls[3];
+	int width[3];
+	int8_t mod[3];		/* gg_plan_fmod */
+};
+
+struct PlanDev
+{
+	int64_t n;
+	int npreds, njoins, naggs, ngroup;
+	PlanPredDev preds[GG_PLAN_MAX_PREDS];
+	PlanJoinDev joins[GG_PLAN_MAX_JOINS];
+	const void *gcol[2];
+	const uint8_t *gnulls[2];
+	int gwidth[2];
+	unsigned long long *tkeys;	/* nslots; INT64_MIN pattern = empty */
+	unsigned long long *tvals;	/* nslots * naggs * 2 (lo, hi) */
+	uint64_t nslots;
+	unsigned long long *err;
+	PlanAggDev aggs[GG_PLAN_MAX_AGGS];
+};
+
+struct PlanBuildDev
+{
+	const void *key;
+	const uint8_t *knulls;
+	int kw;
+	int64_t n;
+	PlanPredDev preds[GG_PLAN_MAX_PREDS];
+	int npreds;
+	unsigned long long *bits;
+	int64_t dlen;
+	unsigned long long *hkeys;
+	uint64_t hslots;
+};
+
+hipError_t launch_plan_build(hipStream_t s, const PlanBuildDev &b);
+hipError_t launch_plan_scan_agg(hipStream_t s, const PlanDev &p);
+hipError_t launch_plan_compact(hipStream_t s,
+			       const unsigned long long *tkeys,
+			       const unsigned long long *tvals,
+			       uint64_t nslots, int naggs,
+			       unsigned long long *out,
+			       unsigned long long *out_n, uint64_t cap);
+
+/* plan.cpp */
+gg_status exec_plan(Pipeline *p, void *arena, size_t bytes,
+		    size_t *written);
+
+/* engine_abi.cpp helpers shared with plan.cpp */
+Table *engine_table(gg_table h);
+gg_status engine_cached_max_i64(Engine &e, Pipeline *p, const int64_t *col,
+				int64_t n, unsigned long long *ctr,
+				unsigned long long *out);
 
 #define GG_MT_MAX_ATTS 32
 

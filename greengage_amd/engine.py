@@ -41,6 +41,38 @@ class _PipelineDesc(ctypes.Structure):
                 ("regionkey", I8), ("limit_k", I64)]
 
 
+class _PlanPred(ctypes.Structure):
+    _fields_ = [("col", ctypes.c_char_p), ("lo", I64), ("hi", I64)]
+
+
+class _PlanScan(ctypes.Structure):
+    _fields_ = [("table", I32), ("preds", _PlanPred * 8),
+                ("npreds", ctypes.c_int)]
+
+
+class _PlanJoin(ctypes.Structure):
+    _fields_ = [("build", _PlanScan), ("build_key", ctypes.c_char_p),
+                ("probe_key", ctypes.c_char_p)]
+
+
+class _PlanAgg(ctypes.Structure):
+    _fields_ = [("kind", ctypes.c_int), ("nfactors", ctypes.c_int),
+                ("col", ctypes.c_char_p * 3), ("mod", ctypes.c_int8 * 3)]
+
+
+class _PlanDesc(ctypes.Structure):
+    _fields_ = [("scan", _PlanScan), ("joins", _PlanJoin * 2),
+                ("njoins", ctypes.c_int),
+                ("group_cols", ctypes.c_char_p * 2),
+                ("ngroup", ctypes.c_int), ("aggs", _PlanAgg * 8),
+                ("naggs", ctypes.c_int)]
+
+
+NEG_INF = -(1 << 63)          # one-sided range bounds
+POS_INF = (1 << 63) - 1
+PLAN_NULL_KEY = -(1 << 63) + 1
+
+
 class KernelStat(ctypes.Structure):
     _fields_ = [("name", ctypes.c_char * 48), ("launches", I64),
                 ("total_ms", ctypes.c_double), ("rows_in", I64),
@@ -101,6 +133,14 @@ def _load():
                                     ctypes.c_int, ctypes.POINTER(ctypes.c_int)]
     lib.gg_engine_comm_id.argtypes = [ctypes.c_void_p]
     lib.gg_engine_comm_init.argtypes = [ctypes.c_void_p]
+    lib.gg_engine_compile_plan.argtypes = [ctypes.POINTER(_PlanDesc),
+                                           ctypes.POINTER(I32)]
+    lib.gg_engine_table_set_nulls.argtypes = [I32, ctypes.c_char_p,
+                                              ctypes.c_void_p]
+    lib.gg_engine_hash_groupby_i64_n.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        I64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, I64,
+        ctypes.POINTER(I64)]
     lib.gg_engine_numeric_str.argtypes = [U64, I64, ctypes.c_int,
                                           ctypes.c_char_p]
     lib.gg_engine_radix_sort_u64.argtypes = [ctypes.c_void_p,
@@ -339,6 +379,117 @@ class Engine:
         _check(lib().gg_engine_compile_pipeline(ctypes.byref(d),
                                                 ctypes.byref(h)), "compile")
         return h.value
+
+    # ---- generalized pipeline descriptor (v2) ----
+    def compile_plan(self, scan_table, preds=(), joins=(), group_cols=(),
+                     aggs=()):
+        """Compile a compositional plan (engine_abi.h gg_plan_desc).
+
+        preds: [(col, lo, hi)]  (half-open; NEG_INF/POS_INF one-sided)
+        joins: [{"table": h, "build_key": c, "probe_key": c,
+                 "preds": [...]}]  (hash SEMI-join filters)
+        group_cols: 0-2 column names of the driving table
+        aggs: ["count"] | ("count", col) |
+              ("sum", [(col, "id"|"sub100"|"add100"), ...])
+        """
+        mods = {"id": 0, "sub100": 1, "add100": 2}
+
+        def fill_scan(dst, table, plist):
+            dst.table = table
+            dst.npreds = len(plist)
+            for i, (c, lo, hi) in enumerate(plist):
+                dst.preds[i].col = c.encode()
+                dst.preds[i].lo = lo
+                dst.preds[i].hi = hi
+
+        d = _PlanDesc()
+        fill_scan(d.scan, scan_table, list(preds))
+        d.njoins = len(joins)
+        for j, spec in enumerate(joins):
+            fill_scan(d.joins[j].build, spec["table"],
+                      list(spec.get("preds", ())))
+            d.joins[j].build_key = spec["build_key"].encode()
+            d.joins[j].probe_key = spec["probe_key"].encode()
+        d.ngroup = len(group_cols)
+        for g, c in enumerate(group_cols):
+            d.group_cols[g] = c.encode()
+        d.naggs = len(aggs)
+        for a, spec in enumerate(aggs):
+            if spec == "count" or spec == ("count",):
+                d.aggs[a].kind = 0
+                d.aggs[a].nfactors = 0
+            elif spec[0] == "count":
+                d.aggs[a].kind = 1
+                d.aggs[a].nfactors = 1
+                d.aggs[a].col[0] = spec[1].encode()
+                d.aggs[a].mod[0] = 0
+            else:
+                assert spec[0] == "sum", spec
+                d.aggs[a].kind = 2
+                d.aggs[a].nfactors = len(spec[1])
+                for f, (c, m) in enumerate(spec[1]):
+                    d.aggs[a].col[f] = c.encode()
+                    d.aggs[a].mod[f] = mods[m]
+        h = I32()
+        _check(lib().gg_engine_compile_plan(ctypes.byref(d),
+                                            ctypes.byref(h)), "compile_plan")
+        return h.value
+
+    def execute_plan(self, p, max_groups=1 << 16):
+        naggs_guess = 8
+        raw = self.execute_raw(
+            p, 16 + max_groups * (16 + 16 * naggs_guess))
+        ng = int.from_bytes(raw[0:8], "little", signed=True)
+        naggs = int.from_bytes(raw[8:12], "little", signed=True)
+        groups = []
+        off = 16
+        for _ in range(ng):
+            k0 = int.from_bytes(raw[off:off + 8], "little", signed=True)
+            k1 = int.from_bytes(raw[off + 8:off + 16], "little",
+                                signed=True)
+            off += 16
+            vals = []
+            for _ in range(naggs):
+                lo = int.from_bytes(raw[off:off + 8], "little")
+                hi = int.from_bytes(raw[off + 8:off + 16], "little",
+                                    signed=True)
+                vals.append((hi << 64) | lo)
+                off += 16
+            groups.append((k0, k1, vals))
+        return groups
+
+    def set_nulls(self, table, col, nulls):
+        import numpy as np
+        nl = np.ascontiguousarray(nulls, np.uint8)
+        _check(lib().gg_engine_table_set_nulls(
+            table, col.encode(), nl.ctypes.data_as(ctypes.c_void_p)),
+            "set_nulls")
+
+    @staticmethod
+    def hash_groupby_n(keys, key_nulls, vals, val_nulls):
+        import numpy as np
+        keys = np.ascontiguousarray(keys, np.int64)
+        vals = np.ascontiguousarray(vals, np.int64)
+        kn = (None if key_nulls is None
+              else np.ascontiguousarray(key_nulls, np.uint8))
+        vn = (None if val_nulls is None
+              else np.ascontiguousarray(val_nulls, np.uint8))
+        cap = len(keys) if len(keys) else 1
+        ok = np.empty(cap, np.int64)
+        os_ = np.empty(cap, np.int64)
+        oc = np.empty(cap, np.int64)
+        ng = I64()
+        _check(lib().gg_engine_hash_groupby_i64_n(
+            keys.ctypes.data_as(ctypes.c_void_p),
+            None if kn is None else kn.ctypes.data_as(ctypes.c_void_p),
+            vals.ctypes.data_as(ctypes.c_void_p),
+            None if vn is None else vn.ctypes.data_as(ctypes.c_void_p),
+            len(keys), ok.ctypes.data_as(ctypes.c_void_p),
+            os_.ctypes.data_as(ctypes.c_void_p),
+            oc.ctypes.data_as(ctypes.c_void_p), cap,
+            ctypes.byref(ng)), "hash_groupby_n")
+        n = ng.value
+        return ok[:n], os_[:n], oc[:n]
 
     def execute_raw(self, p, arena_bytes):
         arena = ctypes.create_string_buffer(arena_bytes)

@@ -1,0 +1,273 @@
+"""Generalized pipeline descriptor (VERDICT r01 missing #2 / next #2).
+
+The compositional gg_plan_desc must run queries the 4 named pipelines
+cannot — mpph6 (Q6), ad-hoc predicates, ad-hoc semi-join group-bys —
+from the descriptor alone, with ZERO query-specific kernels
+(plan.hip's generic build/scan-agg pair).  Parity:
+
+- Q6 over the reference-shaped small fixture (tests/golden/
+  small_inputs.npz) against an independent Python-Decimal recompute,
+  and over dbgen-shaped SF1 synthetic data against numpy;
+- a Q1-shaped descriptor (2 char1 group keys, the full 6-agg set with
+  exact (100±col) product expressions) bit-exact against the same CPU
+  oracle that pins the specialized Q1 kernel;
+- an ad-hoc join+group query against numpy;
+- NULL semantics (strict transitions, NULL group keys, NULL never
+  matching a join key) against explicit numpy models.
+"""
+import os
+
+import numpy as np
+import pytest
+
+import pyoracle
+from conftest import GOLDEN
+
+pytestmark = pytest.mark.gpu
+
+NEG_INF = -(1 << 63)
+POS_INF = (1 << 63) - 1
+NULL_KEY = -(1 << 63) + 1
+
+
+@pytest.fixture(scope="module")
+def eng():
+    from greengage_amd import Engine
+    e = Engine(device=0, n_segments=1, segment_id=0)
+    yield e
+    e.shutdown()
+
+
+@pytest.fixture(scope="module")
+def small():
+    return np.load(os.path.join(GOLDEN, "small_inputs.npz"))
+
+
+def li_small(eng, small):
+    cols = [
+        ("orderkey", "int64", small["li_orderkey"]),
+        ("qty", "dec64", small["li_qty_c"]),
+        ("price", "dec64", small["li_price_c"]),
+        ("disc", "dec64", small["li_disc_c"]),
+        ("tax", "dec64", small["li_tax_c"]),
+        ("shipdate", "int32", small["li_shipdate"]),
+        ("rflag", "char1", small["li_rflag"]),
+        ("lstatus", "char1", small["li_lstatus"]),
+    ]
+    return eng.register_table("li_small_plan", cols,
+                              len(small["li_orderkey"]))
+
+
+def q6_expected(shipdate, qty, price, disc, lo, hi):
+    m = ((shipdate >= lo) & (shipdate < hi) & (disc >= 5) & (disc < 8)
+         & (qty < 2400))
+    # exact scale-4 revenue: sum(price_c * disc_c), python ints
+    return (int(np.count_nonzero(m)),
+            sum(int(p) * int(d) for p, d in
+                zip(price[m].tolist(), disc[m].tolist())))
+
+
+def test_q6_small_fixture(eng, small):
+    """mpph6 (Q6) from the descriptor alone over the reference CSV
+    fixture rows (input/bb_mpph.source:199 query shape)."""
+    from greengage_amd import pgdate
+    li = li_small(eng, small)
+    lo, hi = pgdate(1994, 1, 1), pgdate(1995, 1, 1)
+    p = eng.compile_plan(
+        li,
+        preds=[("shipdate", lo, hi), ("disc", 5, 8),
+               ("qty", NEG_INF, 2400)],
+        aggs=[("sum", [("price", "id"), ("disc", "id")]), "count"])
+    groups = eng.execute_plan(p)
+    assert len(groups) == 1
+    cnt, rev = q6_expected(small["li_shipdate"], small["li_qty_c"],
+                           small["li_price_c"], small["li_disc_c"], lo, hi)
+    assert groups[0][2][0] == rev
+    assert groups[0][2][1] == cnt
+
+
+def test_q6_sf1_synth(eng):
+    from greengage_amd import pgdate
+    li = eng.register_synth("lineitem", seed=42, sf=1)
+    lo, hi = pgdate(1994, 1, 1), pgdate(1995, 1, 1)
+    p = eng.compile_plan(
+        li,
+        preds=[("shipdate", lo, hi), ("disc", 5, 8),
+               ("qty", NEG_INF, 2400)],
+        aggs=[("sum", [("price", "id"), ("disc", "id")]), "count"])
+    groups = eng.execute_plan(p)
+    g = pyoracle.gen_lineitem(42, 0, 6_000_000)
+    cnt, rev = q6_expected(g["shipdate"], g["qty"], g["price"],
+                           g["disc"], lo, hi)
+    assert groups[0][2][0] == rev
+    assert groups[0][2][1] == cnt
+    # no query-specific kernels ran: only the generic plan stats rows
+    names = {s["name"] for s in eng.stats(p)}
+    assert names <= {"plan_build", "plan_scan_agg",
+                     "path_plan_join_bitmap", "path_plan_join_hash"}, names
+
+
+def test_q1_shaped_descriptor_vs_oracle(eng):
+    """The full Q1 agg set (incl. exact scale-4/6 product expressions
+    and 2 char1 group keys) expressed as a generic descriptor must
+    reproduce the oracle bit-exactly — the shim could offload Q1 with
+    a changed predicate column tomorrow."""
+    from greengage_amd import PGDate
+    li = eng.register_synth("lineitem", seed=42, sf=1)
+    cutoff = PGDate("1998-08-15")
+    p = eng.compile_plan(
+        li,
+        preds=[("shipdate", NEG_INF, cutoff + 1)],
+        group_cols=["rflag", "lstatus"],
+        aggs=["count",
+              ("sum", [("qty", "id")]),
+              ("sum", [("price", "id")]),
+              ("sum", [("disc", "id")]),
+              ("sum", [("price", "id"), ("disc", "sub100")]),
+              ("sum", [("price", "id"), ("disc", "sub100"),
+                       ("tax", "add100")])])
+    groups = eng.execute_plan(p)
+    expect = [g for g in pyoracle.q1_synth(42, 1, cutoff) if g["count"]]
+    # oracle group order is (returnflag, linestatus) byte-ascending —
+    # identical to the plan result's key order
+    assert len(groups) == len(expect)
+    for (k0, k1, vals), exp in zip(groups, expect):
+        assert chr(k0) == exp["returnflag"]
+        assert chr(k1) == exp["linestatus"]
+        assert vals[0] == exp["count"]
+        assert vals[1] == exp["sum_qty_c"]
+        assert vals[2] == exp["sum_base_c"]
+        assert vals[3] == exp["sum_dcol_c"]
+        assert vals[4] == exp["sum_disc4"]
+        assert vals[5] == exp["sum_charge6"]
+
+
+def test_adhoc_join_groupby(eng):
+    """Non-TPC-H ad-hoc query: lineitem rows whose order is URGENT
+    (semi-join on a filtered orders build) grouped by l_linestatus —
+    a shape no named pipeline covers."""
+    from greengage_amd import PGDate
+    li = eng.register_synth("lineitem", seed=42, sf=1)
+    od = eng.register_synth("orders", seed=42, sf=1)
+    p = eng.compile_plan(
+        li,
+        preds=[("qty", 1000, POS_INF)],
+        joins=[{"table": od, "build_key": "orderkey",
+                "probe_key": "orderkey",
+                "preds": [("shippriority", 0, 1),
+                          ("orderdate", PGDate("1996-01-01"), POS_INF)]}],
+        group_cols=["lstatus"],
+        aggs=["count", ("sum", [("qty", "id"), ("tax", "add100")])])
+    groups = eng.execute_plan(p)
+
+    g = pyoracle.gen_lineitem(42, 0, 6_000_000)
+    o = pyoracle.gen_orders(42, 1, 0, 1_500_000)
+    ok = o["orderkey"][(o["shippriority"] == 0) &
+                       (o["orderdate"] >= PGDate("1996-01-01"))]
+    member = np.zeros(int(o["orderkey"].max()) + 2, bool)
+    member[ok] = True
+    m = (g["qty"] >= 1000) & member[g["orderkey"]]
+    exp = {}
+    for ls in np.unique(g["lstatus"][m]):
+        sel = m & (g["lstatus"] == ls)
+        exp[int(ls)] = (int(np.count_nonzero(sel)),
+                        sum(int(q) * (100 + int(t)) for q, t in
+                            zip(g["qty"][sel].tolist(),
+                                g["tax"][sel].tolist())))
+    assert len(groups) == len(exp)
+    for k0, _k1, vals in groups:
+        assert vals[0] == exp[k0][0]
+        assert vals[1] == exp[k0][1]
+
+
+def test_sparse_build_keys_take_hash_path(eng):
+    rng = np.random.default_rng(3)
+    n = 200_000
+    key = (rng.integers(1, 50_000, n) * 100_003).astype(np.int64)
+    val = rng.integers(0, 1000, n).astype(np.int64)
+    t = eng.register_table("plan_probe", [("k", "int64", key),
+                                          ("v", "dec64", val)], n)
+    nb = 30_000
+    bkey = (rng.integers(1, 50_000, nb) * 100_003).astype(np.int64)
+    b = eng.register_table("plan_build", [("bk", "int64", bkey)], nb)
+    p = eng.compile_plan(
+        t, joins=[{"table": b, "build_key": "bk", "probe_key": "k"}],
+        aggs=["count", ("sum", [("v", "id")])])
+    groups = eng.execute_plan(p)
+    names = {s["name"] for s in eng.stats(p)}
+    assert "path_plan_join_hash" in names, names
+    member = set(bkey.tolist())
+    m = np.array([k in member for k in key.tolist()])
+    assert groups[0][2][0] == int(np.count_nonzero(m))
+    assert groups[0][2][1] == int(val[m].sum())
+
+
+def test_nullable_plan_strict_semantics(eng):
+    """AO-style null flags through the generic path: NULL fails quals,
+    NULL join keys never match (nodeHash.c:1070), SUM/COUNT(col) skip
+    NULL inputs, COUNT(*) does not, NULL group keys group together
+    (execHHashagg.c:531)."""
+    rng = np.random.default_rng(11)
+    n = 500_000
+    k = rng.integers(0, 50, n).astype(np.int64)
+    v = rng.integers(-500, 500, n).astype(np.int64)
+    w = rng.integers(0, 100, n).astype(np.int64)
+    knull = (rng.random(n) < 0.1).astype(np.uint8)
+    vnull = (rng.random(n) < 0.2).astype(np.uint8)
+    t = eng.register_table("plan_nulls", [
+        ("k", "int64", k), ("v", "dec64", v), ("w", "dec64", w)], n)
+    eng.set_nulls(t, "k", knull)
+    eng.set_nulls(t, "v", vnull)
+
+    p = eng.compile_plan(
+        t, preds=[("w", 10, POS_INF)], group_cols=["k"],
+        aggs=["count", ("count", "v"), ("sum", [("v", "id")]),
+              ("sum", [("v", "id"), ("w", "sub100")])])
+    groups = eng.execute_plan(p)
+
+    m = w >= 10
+    exp = {}
+    for i in np.nonzero(m)[0]:
+        kk = NULL_KEY if knull[i] else int(k[i])
+        e = exp.setdefault(kk, [0, 0, 0, 0])
+        e[0] += 1
+        if not vnull[i]:
+            e[1] += 1
+            e[2] += int(v[i])
+            e[3] += int(v[i]) * (100 - int(w[i]))
+    assert len(groups) == len(exp)
+    for k0, _k1, vals in groups:
+        assert vals == exp[k0], k0
+    # NULL group sorts first (NULL_KEY < any real key here)
+    assert groups[0][0] == NULL_KEY
+
+    # NULL probe keys never match a join
+    b = eng.register_table("plan_nulls_b",
+                           [("bk", "int64",
+                             np.arange(50, dtype=np.int64))], 50)
+    p2 = eng.compile_plan(
+        t, joins=[{"table": b, "build_key": "bk", "probe_key": "k"}],
+        aggs=["count"])
+    groups2 = eng.execute_plan(p2)
+    assert groups2[0][2][0] == int(np.count_nonzero(~knull.astype(bool)))
+
+
+def test_hash_groupby_null_aware(eng):
+    rng = np.random.default_rng(5)
+    n = 300_000
+    keys = rng.integers(-1000, 1000, n).astype(np.int64)
+    vals = rng.integers(-10**6, 10**6, n).astype(np.int64)
+    kn = (rng.random(n) < 0.05).astype(np.uint8)
+    vn = (rng.random(n) < 0.3).astype(np.uint8)
+    ok, os_, oc = eng.hash_groupby_n(keys, kn, vals, vn)
+    exp = {}
+    for i in range(n):
+        kk = NULL_KEY if kn[i] else int(keys[i])
+        e = exp.setdefault(kk, [0, 0])
+        if not vn[i]:
+            e[0] += int(vals[i])
+            e[1] += 1
+    assert len(ok) == len(exp)
+    for kk, ss, cc in zip(ok.tolist(), os_.tolist(), oc.tolist()):
+        assert [ss, cc] == exp[kk], kk
+    assert np.all(np.diff(ok) > 0)  # sorted ascending
