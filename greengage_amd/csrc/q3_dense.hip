@@ -436,23 +436,24 @@ launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 
 /* Q3 lineitem probe against the dense orders map.  Round-1 PMC showed
  * this kernel at ~3.5 TB/s effective with only ~12 B/row of sequential
- * traffic (okey+shipdate): the limiter was memory-level parallelism
- * and narrow NT loads, not occupancy.  v2: 4 consecutive rows per lane
- * read with 16-B vector NT loads (3 loads per quad instead of 8) + XCD
- * chunking so each XCD's bitmap/rev slice stays in its private L2. */
-__global__ __launch_bounds__(DN_THREADS, 4)
-void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
-			 const int32_t *__restrict__ shipdate,
-			 const int64_t *__restrict__ price,
-			 const int64_t *__restrict__ disc, int64_t n,
-			 int32_t cutoff,
-			 const unsigned long long *__restrict__ pay,
-			 unsigned long long *__restrict__ rev,
-			 int64_t dense_len,
-			 const unsigned long long *__restrict__ bloom,
-			 uint64_t bwords, unsigned long long *join_rows)
+ * traffic (okey+shipdate): the limiter was memory-level parallelism,
+ * not occupancy.  Two measured layouts, both XCD-chunked so each XCD's
+ * bitmap/rev slice stays in its private L2: QUAD = 4 consecutive rows
+ * per lane via 16-B vector NT loads; strided = 4-way unrolled strided
+ * streams.  GG_Q3_PROBE_VAR selects (sweep evidence in profiles/). */
+template <int QUAD, int WAVES>
+__global__ __launch_bounds__(DN_THREADS, WAVES)
+void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
+			   const int32_t *__restrict__ shipdate,
+			   const int64_t *__restrict__ price,
+			   const int64_t *__restrict__ disc, int64_t n,
+			   int32_t cutoff,
+			   const unsigned long long *__restrict__ pay,
+			   unsigned long long *__restrict__ rev,
+			   int64_t dense_len,
+			   const unsigned long long *__restrict__ bloom,
+			   uint64_t bwords, unsigned long long *join_rows)
 {
-	DnChunk c = dn_chunk4(n);
 	unsigned long long joined = 0;
 
 	auto body = [&](int64_t i, int32_t sd, int64_t k)
@@ -467,31 +468,70 @@ void k_dn_probe_lineitem(const int64_t *__restrict__ okey,
 		atomicAdd(&rev[k],
 			  (unsigned long long) (price[i] * (100 - disc[i])));
 	};
-	const int64_t S = c.stride;
 
-	for (int64_t r = c.base + (int64_t) threadIdx.x * 4; r < c.hi;
-	     r += S)
+	if (QUAD)
 	{
-		if (r + 3 < c.hi)
-		{
-			gg_i4 sd = dn_ld4x32(&shipdate[r]);
-			gg_ll2 k01 = dn_ld2(&okey[r]);
-			gg_ll2 k23 = dn_ld2(&okey[r + 2]);
+		DnChunk c = dn_chunk4(n);
+		const int64_t S = c.stride;
 
-			body(r, sd.x, k01.x);
-			body(r + 1, sd.y, k01.y);
-			body(r + 2, sd.z, k23.x);
-			body(r + 3, sd.w, k23.y);
+		for (int64_t r = c.base + (int64_t) threadIdx.x * 4;
+		     r < c.hi; r += S)
+		{
+			if (r + 3 < c.hi)
+			{
+				gg_i4 sd = dn_ld4x32(&shipdate[r]);
+				gg_ll2 k01 = dn_ld2(&okey[r]);
+				gg_ll2 k23 = dn_ld2(&okey[r + 2]);
+
+				body(r, sd.x, k01.x);
+				body(r + 1, sd.y, k01.y);
+				body(r + 2, sd.z, k23.x);
+				body(r + 3, sd.w, k23.y);
+			}
+			else
+				for (int64_t j = r; j < c.hi; j++)
+					body(j, dn_ld32(&shipdate[j]),
+					     dn_ld64(&okey[j]));
 		}
-		else
-			for (int64_t j = r; j < c.hi; j++)
-				body(j, dn_ld32(&shipdate[j]),
-				     dn_ld64(&okey[j]));
+	}
+	else
+	{
+		DnChunk c = dn_chunk(n);
+		const int64_t S = c.stride;
+		int64_t i = c.base + threadIdx.x;
+
+		for (; i + 3 * S < c.hi; i += 4 * S)
+		{
+			int32_t sd0 = dn_ld32(&shipdate[i]);
+			int32_t sd1 = dn_ld32(&shipdate[i + S]);
+			int32_t sd2 = dn_ld32(&shipdate[i + 2 * S]);
+			int32_t sd3 = dn_ld32(&shipdate[i + 3 * S]);
+			int64_t k0 = dn_ld64(&okey[i]);
+			int64_t k1 = dn_ld64(&okey[i + S]);
+			int64_t k2 = dn_ld64(&okey[i + 2 * S]);
+			int64_t k3 = dn_ld64(&okey[i + 3 * S]);
+
+			body(i, sd0, k0);
+			body(i + S, sd1, k1);
+			body(i + 2 * S, sd2, k2);
+			body(i + 3 * S, sd3, k3);
+		}
+		for (; i < c.hi; i += S)
+			body(i, dn_ld32(&shipdate[i]), dn_ld64(&okey[i]));
 	}
 	for (int off = 32; off; off >>= 1)
 		joined += __shfl_down(joined, off, 64);
 	if ((threadIdx.x & 63) == 0 && joined)
 		atomicAdd(join_rows, joined);
+}
+
+/* sweep-selected default: strided streams at 8 blocks/CU (see
+ * profiles/r02 probe sweep) */
+static int dn_probe_var(const char *env, int dflt)
+{
+	const char *v = getenv(env);
+
+	return (v && *v) ? atoi(v) : dflt;
 }
 
 static int
@@ -522,11 +562,40 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 uint64_t bwords, unsigned long long *join_rows)
 {
 	int g = (dn_grid_env(n, "GG_Q3_PROBE_GRID") + 7) & ~7;
+	dim3 gg(g), bb(DN_THREADS);
 
-	hipLaunchKernelGGL(k_dn_probe_lineitem, dim3(g),
-			   dim3(DN_THREADS), 0, s, okey, shipdate, price,
-			   disc, n, cutoff, pay, rev, dense_len, bloom,
-			   bwords, join_rows);
+	switch (dn_probe_var("GG_Q3_PROBE_VAR", 0))
+	{
+		default:
+		case 0:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_t<0, 8>), gg,
+					   bb, 0, s, okey, shipdate, price,
+					   disc, n, cutoff, pay, rev,
+					   dense_len, bloom, bwords,
+					   join_rows);
+			break;
+		case 1:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 4>), gg,
+					   bb, 0, s, okey, shipdate, price,
+					   disc, n, cutoff, pay, rev,
+					   dense_len, bloom, bwords,
+					   join_rows);
+			break;
+		case 2:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 8>), gg,
+					   bb, 0, s, okey, shipdate, price,
+					   disc, n, cutoff, pay, rev,
+					   dense_len, bloom, bwords,
+					   join_rows);
+			break;
+		case 3:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_t<0, 4>), gg,
+					   bb, 0, s, okey, shipdate, price,
+					   disc, n, cutoff, pay, rev,
+					   dense_len, bloom, bwords,
+					   join_rows);
+			break;
+	}
 	return hipGetLastError();
 }
 
@@ -648,8 +717,9 @@ launch_dn_insert_orders_q5_u8(hipStream_t s, const int64_t *okey,
 	return hipGetLastError();
 }
 
-__global__ __launch_bounds__(DN_THREADS, 4)
-void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
+template <int QUAD, int WAVES>
+__global__ __launch_bounds__(DN_THREADS, WAVES)
+void k_dn_probe_lineitem_q5_u8_t(const int64_t *__restrict__ okey,
 			  const int64_t *__restrict__ skey,
 			  const int64_t *__restrict__ price,
 			  const int64_t *__restrict__ disc, int64_t n,
@@ -666,7 +736,6 @@ void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 		((unsigned long long *) lds)[i] = 0;
 	__syncthreads();
 
-	DnChunk c = dn_chunk4(n);
 	unsigned long long joined = 0;
 
 	/* filter order: the orders map passes ~3% of rows (date range x
@@ -700,24 +769,50 @@ void k_dn_probe_lineitem_q5_u8(const int64_t *__restrict__ okey,
 			atomicAdd(&lds[snat][1], rev4);
 		}
 	};
-	const int64_t S = c.stride;
 
-	for (int64_t r = c.base + (int64_t) threadIdx.x * 4; r < c.hi;
-	     r += S)
+	if (QUAD)
 	{
-		if (r + 3 < c.hi)
-		{
-			gg_ll2 k01 = dn_ld2(&okey[r]);
-			gg_ll2 k23 = dn_ld2(&okey[r + 2]);
+		DnChunk c = dn_chunk4(n);
+		const int64_t S = c.stride;
 
-			body(r, k01.x);
-			body(r + 1, k01.y);
-			body(r + 2, k23.x);
-			body(r + 3, k23.y);
+		for (int64_t r = c.base + (int64_t) threadIdx.x * 4;
+		     r < c.hi; r += S)
+		{
+			if (r + 3 < c.hi)
+			{
+				gg_ll2 k01 = dn_ld2(&okey[r]);
+				gg_ll2 k23 = dn_ld2(&okey[r + 2]);
+
+				body(r, k01.x);
+				body(r + 1, k01.y);
+				body(r + 2, k23.x);
+				body(r + 3, k23.y);
+			}
+			else
+				for (int64_t j = r; j < c.hi; j++)
+					body(j, dn_ld64(&okey[j]));
 		}
-		else
-			for (int64_t j = r; j < c.hi; j++)
-				body(j, dn_ld64(&okey[j]));
+	}
+	else
+	{
+		DnChunk c = dn_chunk(n);
+		const int64_t S = c.stride;
+		int64_t i = c.base + threadIdx.x;
+
+		for (; i + 3 * S < c.hi; i += 4 * S)
+		{
+			int64_t k0 = dn_ld64(&okey[i]);
+			int64_t k1 = dn_ld64(&okey[i + S]);
+			int64_t k2 = dn_ld64(&okey[i + 2 * S]);
+			int64_t k3 = dn_ld64(&okey[i + 3 * S]);
+
+			body(i, k0);
+			body(i + S, k1);
+			body(i + 2 * S, k2);
+			body(i + 3 * S, k3);
+		}
+		for (; i < c.hi; i += S)
+			body(i, dn_ld64(&okey[i]));
 	}
 	for (int off = 32; off; off >>= 1)
 		joined += __shfl_down(joined, off, 64);
@@ -750,11 +845,41 @@ launch_dn_probe_lineitem_q5_u8(hipStream_t s, const int64_t *okey,
 			       unsigned long long *acc,
 			       unsigned long long *join_rows)
 {
-	hipLaunchKernelGGL(k_dn_probe_lineitem_q5_u8,
-			   dim3((dn_grid_env(n, "GG_Q5_PROBE_GRID") + 7) & ~7),
-			   dim3(DN_THREADS), 0, s, okey, skey, price, disc,
-			   n, pay8, dense_len, supp_dense, supp_dlen, acc,
-			   join_rows);
+	dim3 gg((dn_grid_env(n, "GG_Q5_PROBE_GRID") + 7) & ~7);
+	dim3 bb(DN_THREADS);
+
+	switch (dn_probe_var("GG_Q5_PROBE_VAR", 0))
+	{
+		default:
+		case 0:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_q5_u8_t<0, 8>),
+					   gg, bb, 0, s, okey, skey, price,
+					   disc, n, pay8, dense_len,
+					   supp_dense, supp_dlen, acc,
+					   join_rows);
+			break;
+		case 1:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_q5_u8_t<1, 4>),
+					   gg, bb, 0, s, okey, skey, price,
+					   disc, n, pay8, dense_len,
+					   supp_dense, supp_dlen, acc,
+					   join_rows);
+			break;
+		case 2:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_q5_u8_t<1, 8>),
+					   gg, bb, 0, s, okey, skey, price,
+					   disc, n, pay8, dense_len,
+					   supp_dense, supp_dlen, acc,
+					   join_rows);
+			break;
+		case 3:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_q5_u8_t<0, 4>),
+					   gg, bb, 0, s, okey, skey, price,
+					   disc, n, pay8, dense_len,
+					   supp_dense, supp_dlen, acc,
+					   join_rows);
+			break;
+	}
 	return hipGetLastError();
 }
 

@@ -127,13 +127,16 @@ def test_q1_shaped_descriptor_vs_oracle(eng):
               ("sum", [("price", "id"), ("disc", "sub100"),
                        ("tax", "add100")])])
     groups = eng.execute_plan(p)
-    expect = [g for g in pyoracle.q1_synth(42, 1, cutoff) if g["count"]]
-    # oracle group order is (returnflag, linestatus) byte-ascending —
+    # oracle slot order is (returnflag, linestatus) byte-ascending —
     # identical to the plan result's key order
+    slots = [("A", "F"), ("A", "O"), ("N", "F"), ("N", "O"), ("R", "F"),
+             ("R", "O")]
+    expect = [(s, g) for s, g in zip(slots, pyoracle.q1_synth(42, 1, cutoff))
+              if g["count"]]
     assert len(groups) == len(expect)
-    for (k0, k1, vals), exp in zip(groups, expect):
-        assert chr(k0) == exp["returnflag"]
-        assert chr(k1) == exp["linestatus"]
+    for (k0, k1, vals), ((rf, ls), exp) in zip(groups, expect):
+        assert chr(k0) == rf
+        assert chr(k1) == ls
         assert vals[0] == exp["count"]
         assert vals[1] == exp["sum_qty_c"]
         assert vals[2] == exp["sum_base_c"]
