@@ -43,7 +43,7 @@ static inline int dn_grid(int64_t n)
 
 /* grid for XCD-chunked kernels: multiple of 8 so each XCD gets the
  * same number of blocks */
-static inline int dn_grid8(int64_t n)
+__attribute__((unused)) static inline int dn_grid8(int64_t n)
 {
 	int g = dn_grid(n);
 
@@ -222,8 +222,13 @@ dn_bit_test(const unsigned long long *__restrict__ bm, int64_t key)
 	return (bm[key >> 6] >> (key & 63)) & 1;
 }
 
-/* Q3 orders build: date filter + customer membership → pay store */
-__global__ __launch_bounds__(DN_THREADS, 4)
+/* Q3 orders build: date filter + customer membership → pay store.
+ * Round-2 sweep note: both a 4-way strided unroll and a quad 16-B
+ * vector-load layout measured SLOWER than this simple stride loop
+ * (1.44 ms vs 1.30 ms — the wave-aggregated bitmap OR dominates and
+ * extra unroll state just raises register pressure), so round 1's
+ * layout stands (profiles/r02c sweep). */
+__global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_build_orders(const int64_t *__restrict__ okey,
 		  const int64_t *__restrict__ ckey,
 		  const int32_t *__restrict__ odate,
@@ -238,16 +243,27 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 		  unsigned long long *__restrict__ bloom, uint64_t bwords,
 		  unsigned long long *match_count)
 {
-	DnChunk c = dn_chunk4(n);
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long matches = 0;
+	/* round the per-lane range up so every lane of a wave executes
+	 * the same number of iterations (the wave-aggregated bitmap OR
+	 * below needs a convergent wave) */
+	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	const int64_t n_up = ((n + stride - 1) / stride) * stride;
 
-	/* one fully-evaluated row; returns ok so the caller can do the
-	 * wave-aggregated bitmap OR convergently */
-	auto row = [&](int64_t i, bool live, int32_t d, int64_t ck,
-		       int64_t k) -> bool
+	for (int64_t i = i0; i < n_up; i += stride)
 	{
-		bool ok = live && d < cutoff;
+		bool ok = i < n;
+		int32_t d = 0;
+		int64_t ck = 0, k = -1;
 
+		if (ok)
+		{
+			d = dn_ld32(&odate[i]);
+			ck = dn_ld64(&ckey[i]);
+			k = dn_ld64(&okey[i]);
+			ok = d < cutoff;
+		}
 		if (ok)
 		{
 			if (cust_bits)
@@ -304,78 +320,7 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 				((unsigned long long) (uint32_t) prio[i]
 				 << 32), &pay[k]);
 		}
-		return ok;
-	};
-	const int64_t S = c.stride;
-
-	/* block-uniform outer loop keeps every wave converged for the
-	 * wave-aggregated bitmap OR (its butterfly reduction reads all
-	 * 64 lanes); per-lane/element validity is the live flag.  Each
-	 * lane owns 4 consecutive rows → 16-B vector loads, and its
-	 * quad's bitmap bits land in at most 2 words, merged lane-
-	 * locally before ONE wave aggregation per word. */
-	for (int64_t rb = c.base; rb < c.hi; rb += S)
-	{
-		int64_t r = rb + (int64_t) threadIdx.x * 4;
-		bool quad = r + 3 < c.hi;
-		int32_t d[4];
-		int64_t ck[4], k[4];
-		int nrows;
-
-		if (quad)
-		{
-			gg_i4 dd = dn_ld4x32(&odate[r]);
-			gg_ll2 c01 = dn_ld2(&ckey[r]);
-			gg_ll2 c23 = dn_ld2(&ckey[r + 2]);
-			gg_ll2 k01 = dn_ld2(&okey[r]);
-			gg_ll2 k23 = dn_ld2(&okey[r + 2]);
-
-			d[0] = dd.x; d[1] = dd.y; d[2] = dd.z; d[3] = dd.w;
-			ck[0] = c01.x; ck[1] = c01.y;
-			ck[2] = c23.x; ck[3] = c23.y;
-			k[0] = k01.x; k[1] = k01.y;
-			k[2] = k23.x; k[3] = k23.y;
-			nrows = 4;
-		}
-		else
-		{
-			nrows = (int) (r < c.hi ? c.hi - r : 0);
-			for (int j = 0; j < nrows; j++)
-			{
-				d[j] = dn_ld32(&odate[r + j]);
-				ck[j] = dn_ld64(&ckey[r + j]);
-				k[j] = dn_ld64(&okey[r + j]);
-			}
-		}
-		{
-			unsigned long long w0 = ~0ull, m0 = 0;
-			unsigned long long w1 = ~0ull, m1 = 0;
-
-			for (int j = 0; j < nrows; j++)
-				if (row(r + j, true, d[j], ck[j], k[j]))
-				{
-					unsigned long long w =
-						(unsigned long long)
-						(k[j] >> 6);
-					unsigned long long bit =
-						1ull << (k[j] & 63);
-
-					if (!m0 || w0 == w)
-					{
-						w0 = w;
-						m0 |= bit;
-					}
-					else if (!m1 || w1 == w)
-					{
-						w1 = w;
-						m1 |= bit;
-					}
-					else	/* >2 words per quad: rare */
-						dn_bit_set(bloom, k[j]);
-				}
-			dn_bit_set_wave_mask(bloom, w0, m0, m0 != 0);
-			dn_bit_set_wave_mask(bloom, w1, m1, m1 != 0);
-		}
+		dn_bit_set_wave(bloom, k, ok);
 	}
 	for (int off = 32; off; off >>= 1)
 		matches += __shfl_down(matches, off, 64);
@@ -393,7 +338,7 @@ launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 		       int64_t dense_len, unsigned long long *bloom,
 		       uint64_t bwords, unsigned long long *match_count)
 {
-	hipLaunchKernelGGL(k_dn_build_orders, dim3(dn_grid8(n)),
+	hipLaunchKernelGGL(k_dn_build_orders, dim3(dn_grid(n)),
 			   dim3(DN_THREADS), 0, s, okey, ckey, odate, prio, n,
 			   cutoff, cust.keys, cust.nslots, cust.bloom,
 			   cust.bloom_words, cust_bits, cust_dlen, pay,
@@ -494,6 +439,29 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 					     dn_ld64(&okey[j]));
 		}
 	}
+	else if (QUAD == 2)
+	{
+		/* 8-way strided unroll: 16 loads in flight per lane */
+		DnChunk c = dn_chunk(n);
+		const int64_t S = c.stride;
+		int64_t i = c.base + threadIdx.x;
+
+		for (; i + 7 * S < c.hi; i += 8 * S)
+		{
+			int32_t sd[8];
+			int64_t k[8];
+
+			for (int u = 0; u < 8; u++)
+			{
+				sd[u] = dn_ld32(&shipdate[i + u * S]);
+				k[u] = dn_ld64(&okey[i + u * S]);
+			}
+			for (int u = 0; u < 8; u++)
+				body(i + u * S, sd[u], k[u]);
+		}
+		for (; i < c.hi; i += S)
+			body(i, dn_ld32(&shipdate[i]), dn_ld64(&okey[i]));
+	}
 	else
 	{
 		DnChunk c = dn_chunk(n);
@@ -590,6 +558,20 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			break;
 		case 3:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<0, 4>), gg,
+					   bb, 0, s, okey, shipdate, price,
+					   disc, n, cutoff, pay, rev,
+					   dense_len, bloom, bwords,
+					   join_rows);
+			break;
+		case 4:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 8>), gg,
+					   bb, 0, s, okey, shipdate, price,
+					   disc, n, cutoff, pay, rev,
+					   dense_len, bloom, bwords,
+					   join_rows);
+			break;
+		case 5:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
@@ -793,6 +775,24 @@ void k_dn_probe_lineitem_q5_u8_t(const int64_t *__restrict__ okey,
 					body(j, dn_ld64(&okey[j]));
 		}
 	}
+	else if (QUAD == 2)
+	{
+		DnChunk c = dn_chunk(n);
+		const int64_t S = c.stride;
+		int64_t i = c.base + threadIdx.x;
+
+		for (; i + 7 * S < c.hi; i += 8 * S)
+		{
+			int64_t k[8];
+
+			for (int u = 0; u < 8; u++)
+				k[u] = dn_ld64(&okey[i + u * S]);
+			for (int u = 0; u < 8; u++)
+				body(i + u * S, k[u]);
+		}
+		for (; i < c.hi; i += S)
+			body(i, dn_ld64(&okey[i]));
+	}
 	else
 	{
 		DnChunk c = dn_chunk(n);
@@ -874,6 +874,20 @@ launch_dn_probe_lineitem_q5_u8(hipStream_t s, const int64_t *okey,
 			break;
 		case 3:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_q5_u8_t<0, 4>),
+					   gg, bb, 0, s, okey, skey, price,
+					   disc, n, pay8, dense_len,
+					   supp_dense, supp_dlen, acc,
+					   join_rows);
+			break;
+		case 4:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_q5_u8_t<2, 8>),
+					   gg, bb, 0, s, okey, skey, price,
+					   disc, n, pay8, dense_len,
+					   supp_dense, supp_dlen, acc,
+					   join_rows);
+			break;
+		case 5:
+			hipLaunchKernelGGL((k_dn_probe_lineitem_q5_u8_t<2, 4>),
 					   gg, bb, 0, s, okey, skey, price,
 					   disc, n, pay8, dense_len,
 					   supp_dense, supp_dlen, acc,
@@ -1112,24 +1126,28 @@ void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
 				mine[at] = (unsigned long long) i;
 		}
 	};
-	/* 4 consecutive slots per lane → two 16-B NT loads per quad */
-	const int64_t qstride = stride * 4;
+	/* 16 consecutive slots per lane → eight 16-B NT loads in flight:
+	 * PMC shows this sweep moves exactly its algorithmic 1.2 GB, so
+	 * the old 2-load version was LATENCY-bound at 2.1 TB/s, not
+	 * bandwidth-bound (r02c PMC) */
+	const int64_t qstride = stride * 16;
 
 	for (int64_t r = ((int64_t) blockIdx.x * blockDim.x +
-			  threadIdx.x) * 4;
+			  threadIdx.x) * 16;
 	     r < dense_len; r += qstride)
 	{
-		if (r + 3 < dense_len)
+		if (r + 15 < dense_len)
 		{
-			gg_ull2 v01 = __builtin_nontemporal_load(
-				(const gg_ull2 *) &rev[r]);
-			gg_ull2 v23 = __builtin_nontemporal_load(
-				(const gg_ull2 *) &rev[r + 2]);
+			gg_ull2 v[8];
 
-			body(r, v01.x);
-			body(r + 1, v01.y);
-			body(r + 2, v23.x);
-			body(r + 3, v23.y);
+			for (int q = 0; q < 8; q++)
+				v[q] = __builtin_nontemporal_load(
+					(const gg_ull2 *) &rev[r + 2 * q]);
+			for (int q = 0; q < 8; q++)
+			{
+				body(r + 2 * q, v[q].x);
+				body(r + 2 * q + 1, v[q].y);
+			}
 		}
 		else
 			for (int64_t j = r; j < dense_len; j++)

@@ -14,6 +14,20 @@ typedef uint32 LocalTransactionId;
 typedef struct varlena bytea;
 typedef struct varlena text;
 typedef int16 AttrNumber;
+#ifndef Size
+typedef size_t Size;
+#endif
+typedef int pgsocket;
+typedef struct List List;
+typedef struct Node Node;
+struct MotionConn { int32 sent_record_typmod; };
+#define ERRCODE_UNDEFINED_OBJECT 0
+#define ERRCODE_PROTOCOL_VIOLATION 0
+#define ERRCODE_GP_INTERCONNECTION_ERROR 0
+typedef uint8 bits8;
+typedef uint16 bits16;
+#include "nodes/nodes.h"
+typedef struct MemoryContextCallback MemoryContextCallback;
 
 static inline void MemoryContextReset(MemoryContext c) { (void) c; }
 
