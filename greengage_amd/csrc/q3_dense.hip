@@ -1126,28 +1126,27 @@ void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
 				mine[at] = (unsigned long long) i;
 		}
 	};
-	/* 16 consecutive slots per lane → eight 16-B NT loads in flight:
-	 * PMC shows this sweep moves exactly its algorithmic 1.2 GB, so
-	 * the old 2-load version was LATENCY-bound at 2.1 TB/s, not
-	 * bandwidth-bound (r02c PMC) */
-	const int64_t qstride = stride * 16;
+	/* 4 consecutive slots per lane, two 16-B NT loads per quad.  PMC
+	 * shows this sweep moves exactly its algorithmic 1.2 GB (r02c);
+	 * a 16-slot/8-load variant measured slightly SLOWER (r02d sweep),
+	 * so the shallow unroll stands. */
+	const int64_t qstride = stride * 4;
 
 	for (int64_t r = ((int64_t) blockIdx.x * blockDim.x +
-			  threadIdx.x) * 16;
+			  threadIdx.x) * 4;
 	     r < dense_len; r += qstride)
 	{
-		if (r + 15 < dense_len)
+		if (r + 3 < dense_len)
 		{
-			gg_ull2 v[8];
+			gg_ull2 v01 = __builtin_nontemporal_load(
+				(const gg_ull2 *) &rev[r]);
+			gg_ull2 v23 = __builtin_nontemporal_load(
+				(const gg_ull2 *) &rev[r + 2]);
 
-			for (int q = 0; q < 8; q++)
-				v[q] = __builtin_nontemporal_load(
-					(const gg_ull2 *) &rev[r + 2 * q]);
-			for (int q = 0; q < 8; q++)
-			{
-				body(r + 2 * q, v[q].x);
-				body(r + 2 * q + 1, v[q].y);
-			}
+			body(r, v01.x);
+			body(r + 1, v01.y);
+			body(r + 2, v23.x);
+			body(r + 3, v23.y);
 		}
 		else
 			for (int64_t j = r; j < dense_len; j++)

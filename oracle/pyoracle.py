@@ -220,6 +220,43 @@ def ref():
 _dsb = None
 
 
+_tupser = None
+
+
+def tupser_ref():
+    """The reference's own Motion chunk serializer (tupser.c
+    SerializeTuple compiled in place); None when the prebuilt .so is
+    missing and the reference tree is absent."""
+    global _tupser
+    if _tupser is None:
+        path = os.path.join(_DIR, "_ref", "libpg_tupserref.so")
+        if not os.path.exists(path) and os.path.isdir("/root/reference"):
+            _build()
+        if not os.path.exists(path):
+            return None
+        _tupser = ctypes.CDLL(path)
+        _tupser.ref_tupser_chunks.restype = ctypes.c_int
+        _tupser.ref_tupser_chunks.argtypes = [
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_void_p, ctypes.c_int]
+    return _tupser
+
+
+def ref_tupser_chunks(memtuple_bytes, natts, max_chunk):
+    """Chunk stream the REFERENCE SerializeTuple emits for one
+    MemTuple (b->pri=NULL → the chunked interconnect path)."""
+    T = tupser_ref()
+    assert T is not None, "reference tupser codec missing"
+    mt = np.ascontiguousarray(memtuple_bytes, np.uint8)
+    cap = len(mt) + (len(mt) // 16 + 64) * 8 + (1 << 16)
+    out = np.zeros(cap, np.uint8)
+    n = T.ref_tupser_chunks(mt.ctypes.data_as(ctypes.c_void_p), len(mt),
+                            natts, max_chunk,
+                            out.ctypes.data_as(ctypes.c_void_p), cap)
+    assert n >= 0, n
+    return out[:n].copy()
+
+
 def dsb_ref():
     """The reference's own AOCS datum-stream codec (datumstreamblock.c
     compiled in place); None when the prebuilt .so is missing and the

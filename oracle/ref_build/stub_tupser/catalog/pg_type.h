@@ -11,6 +11,7 @@ typedef struct FormData_pg_type
 } FormData_pg_type;
 typedef FormData_pg_type *Form_pg_type;
 #define TYPTYPE_BASE 'b'
+#define BPCHAROID 1042
 #define RECORDOID 2249
 #define INT4OID 23
 #define INT8OID 20
