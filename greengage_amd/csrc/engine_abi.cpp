@@ -1904,9 +1904,24 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	}
 
 	/* 3. lineitem probe + group aggregation (probe fused with the
-	 * group-by transition; group slot ≡ matched-order slot) */
-	unsigned long long njoin = 0;
+	 * group-by transition; group slot ≡ matched-order slot).  The
+	 * dense probe also appends each group's index to the survivor
+	 * list (0→nonzero rev transition) so the top-k never sweeps
+	 * the dense array. */
+	unsigned long long njoin = 0, nsurv = 0;
+	unsigned long long *dsurv = nullptr, *dnsurv = nullptr;
+	uint64_t surv_cap = 0;
 
+	if (ord_dlen)
+	{
+		surv_cap = (uint64_t) nmatch + 64;
+		dsurv = (unsigned long long *)
+			p->sget("surv", surv_cap * 8);
+		dnsurv = (unsigned long long *) p->sget("nsurv", 8);
+		if (!dsurv || !dnsurv)
+			return fail(GG_ENOMEM, "survivor scratch");
+		GG_HIP(hipMemsetAsync(dnsurv, 0, 8, e.stream));
+	}
 	GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 	{
 		Timed tm(e.stream);
@@ -1915,13 +1930,23 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			GG_HIP(launch_dn_probe_lineitem(
 				e.stream, l_ok, l_sd, l_pc, l_dc, li->nrows,
 				cutoff, ordd_pay, ordd_rev, ord_dlen,
-				ordd_bloom, ordd_bwords, ctr));
+				ordd_bloom, ordd_bwords, ctr, dsurv, dnsurv,
+				surv_cap));
 		else
 			GG_HIP(launch_probe_lineitem(e.stream, l_ok, l_sd,
 						     l_pc, l_dc, li->nrows,
 						     cutoff, ord, ctr));
 		double ms = tm.stop();
 		GG_TRY(read_counter(ctr, &njoin));
+		if (ord_dlen)
+		{
+			GG_TRY(read_counter(dnsurv, &nsurv));
+			if (nsurv > surv_cap)
+				return fail(GG_ESTATE,
+					    "survivor overflow (%llu > %llu)",
+					    nsurv,
+					    (unsigned long long) surv_cap);
+		}
 		KernelStatAcc &st = p->stat("probe_lineitem");
 
 		st.launches++;
@@ -1931,11 +1956,11 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		st.hbm_bytes += li->nrows * 28;	/* SURVEY §8(d) */
 	}
 
-	/* 4. top-k select.  Dense path: ONE fused sweep of rev[] (stats +
-	 * exponent histogram + survivor compaction) → device threshold →
-	 * finish over the ~ngroups survivors only — no host round trip
-	 * between the three kernels.  Hash-table path (and any fused-path
-	 * overflow): the original stats→hist→collect sweeps. */
+	/* 4. top-k select.  Dense path: stats+hist, threshold, collect
+	 * and sparse clear all walk the ~ngroups survivor list the probe
+	 * recorded — no sweep of the GB-sized dense array, no host round
+	 * trip between kernels.  Hash-table path: the original
+	 * stats→hist→collect sweeps. */
 	unsigned long long hstats[5] = {0, 0, 0, 0, 0};
 	unsigned long long *stats5 =
 		(unsigned long long *) p->sget("stats5", 5 * 8);
@@ -1949,67 +1974,37 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 		if (ord_dlen)
 		{
-			/* grid must match launch_dn_q3_pass1's dn_grid() */
-			int64_t pgrid = (ord_dlen + 255) / 256;
-
-			if (pgrid > 2048)
-				pgrid = 2048;
-			if (pgrid < 1)
-				pgrid = 1;
-			/* survivors (groups) <= nmatch (matched orders);
-			 * grid-stride sampling keeps per-block counts near
-			 * uniform — 2x expected + slack */
-			int64_t region =
-				2 * ((int64_t) nmatch / pgrid) + 256;
 			uint64_t cap = 4 * (uint64_t) k + 65536;
 			unsigned int *dhist =
 				(unsigned int *) p->sget("hist", 65536 * 4);
 			unsigned long long *dthr =
 				(unsigned long long *) p->sget("thr", 8);
-			unsigned long long *dsurv = (unsigned long long *)
-				p->sget("surv", (size_t) pgrid * region * 8);
-			unsigned long long *dcnts = (unsigned long long *)
-				p->sget("surv.cnts", (size_t) pgrid * 8);
-			unsigned long long *dovf =
-				(unsigned long long *) p->sget("topk.ovf", 8);
 			gg_q3_result_row *dout = (gg_q3_result_row *)
 				p->sget("cand", cap * sizeof(gg_q3_result_row));
-			unsigned long long ncand = 0, ovf = 0;
-			int pass1_grid = 0;
+			unsigned long long ncand = 0;
 
-			if (!dhist || !dthr || !dsurv || !dcnts || !dovf ||
-			    !dout)
+			if (!dhist || !dthr || !dout)
 				return fail(GG_ENOMEM, "topk scratch");
 			GG_HIP(hipMemsetAsync(stats5, 0, 5 * 8, e.stream));
 			GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4, e.stream));
-			GG_HIP(hipMemsetAsync(dovf, 0, 8, e.stream));
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-			GG_HIP(launch_dn_q3_pass1(e.stream, ordd_rev,
-						  ord_dlen, stats5, dhist,
-						  dsurv, region, dcnts,
-						  &pass1_grid));
-			if (pass1_grid != (int) pgrid)
-				return fail(GG_ESTATE,
-					    "pass1 grid mismatch %d vs %lld",
-					    pass1_grid, (long long) pgrid);
+			GG_HIP(launch_dn_q3_stats_surv(e.stream, dsurv,
+						       (int64_t) nsurv,
+						       ordd_pay, ordd_rev,
+						       stats5, dhist));
 			GG_HIP(launch_dn_q3_threshold2(e.stream, dhist, k,
 						       dthr));
-			GG_HIP(launch_dn_q3_finish(e.stream, dsurv, dcnts,
-						   region, pgrid, ordd_pay,
-						   ordd_rev, dthr, stats5,
-						   dout, ctr, cap, dovf));
+			GG_HIP(launch_dn_q3_collect_surv(
+				e.stream, dsurv, (int64_t) nsurv, ordd_pay,
+				ordd_rev, dthr, dout, ctr, cap));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_HIP(hipMemcpy(hstats, stats5, 40,
 					 hipMemcpyDeviceToHost));
 			GG_TRY(read_counter(ctr, &ncand));
-			GG_HIP(hipMemcpy(&ovf, dovf, 8,
-					 hipMemcpyDeviceToHost));
-			if (ovf)
-				need_old = true;	/* region spill */
-			else if (ncand > cap)
+			if (ncand > cap)
 			{
-				/* tie-heavy threshold bin: retry the finish
-				 * sweep alone with an exact-size buffer */
+				/* tie-heavy threshold bin: retry collect
+				 * alone with an exact-size buffer */
 				cap = hstats[0] + 1;
 				dout = (gg_q3_result_row *) p->sget(
 					"cand",
@@ -2017,40 +2012,31 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				if (!dout)
 					return fail(GG_ENOMEM, "topk retry");
 				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-				/* reset the checksum word the retry re-adds */
-				GG_HIP(hipMemsetAsync(stats5 + 3, 0, 8,
-						      e.stream));
-				GG_HIP(launch_dn_q3_finish(
-					e.stream, dsurv, dcnts, region,
-					pgrid, ordd_pay, ordd_rev, dthr,
-					stats5, dout, ctr, cap, dovf));
+				GG_HIP(launch_dn_q3_collect_surv(
+					e.stream, dsurv, (int64_t) nsurv,
+					ordd_pay, ordd_rev, dthr, dout, ctr,
+					cap));
 				GG_HIP(hipStreamSynchronize(e.stream));
-				GG_HIP(hipMemcpy(hstats, stats5, 40,
-						 hipMemcpyDeviceToHost));
 				GG_TRY(read_counter(ctr, &ncand));
 				if (ncand > cap)
 					return fail(GG_EINVAL,
 						    "top-k overflow (%llu)",
 						    ncand);
 			}
-			if (!need_old && ncand)
+			if (ncand)
 			{
 				cand.resize(ncand);
 				GG_HIP(hipMemcpy(cand.data(), dout,
 						 ncand * sizeof(gg_q3_result_row),
 						 hipMemcpyDeviceToHost));
 			}
-			if (!need_old)
-			{
-				/* zero only the touched rev[] entries so
-				 * the next execute skips the 1.2 GB
-				 * memset (runs async; later work queues
-				 * behind it on the engine stream) */
-				GG_HIP(launch_dn_q3_clear(
-					e.stream, dsurv, dcnts, region,
-					pgrid, ordd_rev));
-				p->q3_rev_dirty = false;
-			}
+			/* zero only the touched rev[] entries so the next
+			 * execute skips the dense memset (runs async;
+			 * later work queues behind it on the stream) */
+			GG_HIP(launch_dn_q3_clear_surv(e.stream, dsurv,
+						       (int64_t) nsurv,
+						       ordd_rev));
+			p->q3_rev_dirty = false;
 		}
 
 		if (need_old)

@@ -437,7 +437,10 @@ hipError_t launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 				    int64_t dense_len,
 				    unsigned long long *bloom,
 				    uint64_t bwords,
-				    unsigned long long *join_rows);
+				    unsigned long long *join_rows,
+				    unsigned long long *surv,
+				    unsigned long long *nsurv,
+				    uint64_t surv_cap);
 hipError_t launch_td_insert(hipStream_t s, const uint8_t *pool,
 			    const unsigned long long *offs,
 			    const uint32_t *lens, const uint8_t *nulls,
@@ -602,30 +605,28 @@ hipError_t launch_dn_probe_lineitem_q5_u8(hipStream_t s,
 					  int64_t supp_dlen,
 					  unsigned long long *acc,
 					  unsigned long long *join_rows);
-hipError_t launch_dn_q3_pass1(hipStream_t s, const unsigned long long *rev,
-			      int64_t dense_len, unsigned long long *out5,
-			      unsigned int *hist64k,
-			      unsigned long long *surv, int64_t region,
-			      unsigned long long *counts, int *out_grid);
+hipError_t launch_dn_q3_stats_surv(hipStream_t s,
+				   const unsigned long long *surv,
+				   int64_t nsurv,
+				   const unsigned long long *pay,
+				   const unsigned long long *rev,
+				   unsigned long long *out5,
+				   unsigned int *hist64k);
 hipError_t launch_dn_q3_threshold2(hipStream_t s,
 				   const unsigned int *hist64k, int64_t k,
 				   unsigned long long *out_thr);
-hipError_t launch_dn_q3_clear(hipStream_t s,
-			      const unsigned long long *surv,
-			      const unsigned long long *counts,
-			      int64_t region, int64_t nregions,
-			      unsigned long long *rev);
-hipError_t launch_dn_q3_finish(hipStream_t s,
-			       const unsigned long long *surv,
-			       const unsigned long long *counts,
-			       int64_t region, int64_t nregions,
-			       const unsigned long long *pay,
-			       const unsigned long long *rev,
-			       const unsigned long long *thr_ptr,
-			       unsigned long long *out5,
-			       gg_q3_result_row *out,
-			       unsigned long long *out_count, uint64_t cap,
-			       unsigned long long *overflow);
+hipError_t launch_dn_q3_collect_surv(hipStream_t s,
+				     const unsigned long long *surv,
+				     int64_t nsurv,
+				     const unsigned long long *pay,
+				     const unsigned long long *rev,
+				     const unsigned long long *thr_ptr,
+				     gg_q3_result_row *out,
+				     unsigned long long *out_count,
+				     uint64_t cap);
+hipError_t launch_dn_q3_clear_surv(hipStream_t s,
+				   const unsigned long long *surv,
+				   int64_t nsurv, unsigned long long *rev);
 hipError_t launch_dn_q3_stats(hipStream_t s, const unsigned long long *pay,
 			      const unsigned long long *rev,
 			      int64_t dense_len, unsigned long long *out5);

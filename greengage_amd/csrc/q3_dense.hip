@@ -397,7 +397,9 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 			   unsigned long long *__restrict__ rev,
 			   int64_t dense_len,
 			   const unsigned long long *__restrict__ bloom,
-			   uint64_t bwords, unsigned long long *join_rows)
+			   uint64_t bwords, unsigned long long *join_rows,
+			   unsigned long long *__restrict__ surv,
+			   unsigned long long *nsurv, uint64_t surv_cap)
 {
 	unsigned long long joined = 0;
 
@@ -410,8 +412,26 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 		if (!dn_bit_test(bloom, k))	/* exact: no false positives */
 			return;
 		joined++;
-		atomicAdd(&rev[k],
-			  (unsigned long long) (price[i] * (100 - disc[i])));
+		{
+			unsigned long long r4 = (unsigned long long)
+				(price[i] * (100 - disc[i]));
+			unsigned long long old = atomicAdd(&rev[k], r4);
+
+			/* group creation (execHHashagg find-or-create):
+			 * exactly one adder sees the 0→nonzero transition,
+			 * so the survivor list gets each group once — the
+			 * top-k then gathers ~ngroups entries instead of
+			 * sweeping the GB-sized dense array (zero-sum
+			 * groups stay invisible, as rev==0 always meant) */
+			if (old == 0 && r4 != 0)
+			{
+				unsigned long long at =
+					atomicAdd(nsurv, 1ull);
+
+				if (at < surv_cap)
+					surv[at] = (unsigned long long) k;
+			}
+		}
 	};
 
 	if (QUAD)
@@ -527,7 +547,9 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 const int64_t *disc, int64_t n, int32_t cutoff,
 			 unsigned long long *pay, unsigned long long *rev,
 			 int64_t dense_len, unsigned long long *bloom,
-			 uint64_t bwords, unsigned long long *join_rows)
+			 uint64_t bwords, unsigned long long *join_rows,
+			 unsigned long long *surv, unsigned long long *nsurv,
+			 uint64_t surv_cap)
 {
 	int g = (dn_grid_env(n, "GG_Q3_PROBE_GRID") + 7) & ~7;
 	dim3 gg(g), bb(DN_THREADS);
@@ -540,42 +562,42 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows);
+					   join_rows, surv, nsurv, surv_cap);
 			break;
 		case 1:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows);
+					   join_rows, surv, nsurv, surv_cap);
 			break;
 		case 2:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 8>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows);
+					   join_rows, surv, nsurv, surv_cap);
 			break;
 		case 3:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<0, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows);
+					   join_rows, surv, nsurv, surv_cap);
 			break;
 		case 4:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 8>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows);
+					   join_rows, surv, nsurv, surv_cap);
 			break;
 		case 5:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows);
+					   join_rows, surv, nsurv, surv_cap);
 			break;
 	}
 	return hipGetLastError();
@@ -1052,17 +1074,13 @@ launch_dn_q5_gather(hipStream_t s, const unsigned long long *comp,
 	return hipGetLastError();
 }
 
-/* ---- fused single-sweep top-k (v2) --------------------------------
- * Replaces the stats→hist→collect triple sweep of the 1.2 GB rev[]
- * array (the reference's bounded-heap switch, tuplesort.c:1360–1377)
- * with ONE sweep: k_dn_q3_pass1 computes ngroups/revsum/max, builds a
- * 16-bit monotonic exponent-mantissa histogram (no dependence on max,
- * so no pre-pass), and compacts survivor indices into per-block
- * regions (LDS counter — a single global counter saturates at ~88
- * returning atomics/µs).  k_dn_q3_threshold2 picks the revenue
- * threshold from the histogram; k_dn_q3_finish walks only the ~ngroups
- * survivors (not the dense array) to compute the group checksum and
- * collect candidates ≥ threshold. */
+/* ---- survivor-list top-k (v3) --------------------------------------
+ * The probe records each group's dense index once (0→nonzero rev
+ * transition), so the top-k (the reference's bounded-heap switch,
+ * tuplesort.c:1360–1377) never sweeps the GB-sized dense array:
+ * stats+hist, threshold, collect and clear all walk the ~ngroups
+ * survivor entries.  The histogram uses a 16-bit monotonic
+ * exponent-mantissa code, so no max pre-pass is needed. */
 
 /* monotonic 16-bit code of a u64 (r > 0): 6-bit exponent bucket + 10
  * mantissa bits; order-preserving, max code 55 295 < 65 536 */
@@ -1089,27 +1107,33 @@ __device__ inline unsigned long long dn_code16_lo(unsigned code)
 	}
 }
 
-__global__ __launch_bounds__(DN_THREADS, 4)
-void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
-		   int64_t dense_len, unsigned long long *__restrict__ out5,
-		   unsigned int *__restrict__ hist64k,
-		   unsigned long long *__restrict__ surv, int64_t region,
-		   unsigned long long *__restrict__ counts)
+/* Survivor-list top-k: the probe appended each group's dense index
+ * once (0→nonzero transition of its rev[] slot), so every stage walks
+ * ~ngroups entries (~10 MB of gathers at SF100) instead of sweeping
+ * the 1.2 GB dense array — the r02c PMC showed those sweeps were
+ * latency-bound at 2.1 TB/s. */
+
+/* gather stats + 16-bit exponent-mantissa histogram over survivors */
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
+			int64_t nsurv,
+			const unsigned long long *__restrict__ pay,
+			const unsigned long long *__restrict__ rev,
+			unsigned long long *__restrict__ out5,
+			unsigned int *__restrict__ hist64k)
 {
-	__shared__ unsigned long long lcnt;
-
-	if (threadIdx.x == 0)
-		lcnt = 0;
-	__syncthreads();
-
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
-	unsigned long long ng = 0, revsum = 0, carry = 0, mx = 0;
-	unsigned long long *mine = surv + (int64_t) blockIdx.x * region;
+	unsigned long long ng = 0, revsum = 0, carry = 0, ck = 0, mx = 0;
 
-	auto body = [&](int64_t i, unsigned long long r)
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nsurv; i += stride)
 	{
+		int64_t k = (int64_t) surv[i];
+		unsigned long long r = rev[k];
+
 		if (!r)
-			return;
+			continue;	/* zero-sum group: invisible, as the
+					 * dense-array sweep always treated it */
 		ng++;
 		mx = max(mx, r);
 		{
@@ -1120,37 +1144,13 @@ void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
 		}
 		atomicAdd(&hist64k[dn_code16(r)], 1u);
 		{
-			unsigned long long at = atomicAdd(&lcnt, 1ull);
+			unsigned long long p = pay[k];
+			int32_t date = (int32_t) (uint32_t) p;
+			int32_t prio = (int32_t) (uint32_t) (p >> 32);
 
-			if (at < (unsigned long long) region)
-				mine[at] = (unsigned long long) i;
+			ck += gg_group_hash((unsigned long long) k, r, 0,
+					    date, prio);
 		}
-	};
-	/* 4 consecutive slots per lane, two 16-B NT loads per quad.  PMC
-	 * shows this sweep moves exactly its algorithmic 1.2 GB (r02c);
-	 * a 16-slot/8-load variant measured slightly SLOWER (r02d sweep),
-	 * so the shallow unroll stands. */
-	const int64_t qstride = stride * 4;
-
-	for (int64_t r = ((int64_t) blockIdx.x * blockDim.x +
-			  threadIdx.x) * 4;
-	     r < dense_len; r += qstride)
-	{
-		if (r + 3 < dense_len)
-		{
-			gg_ull2 v01 = __builtin_nontemporal_load(
-				(const gg_ull2 *) &rev[r]);
-			gg_ull2 v23 = __builtin_nontemporal_load(
-				(const gg_ull2 *) &rev[r + 2]);
-
-			body(r, v01.x);
-			body(r + 1, v01.y);
-			body(r + 2, v23.x);
-			body(r + 3, v23.y);
-		}
-		else
-			for (int64_t j = r; j < dense_len; j++)
-				body(j, __builtin_nontemporal_load(&rev[j]));
 	}
 	for (int off = 32; off; off >>= 1)
 	{
@@ -1159,6 +1159,7 @@ void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
 		ng += __shfl_down(ng, off, 64);
 		revsum += __shfl_down(revsum, off, 64);
 		carry += __shfl_down(carry, off, 64) + (revsum < orev);
+		ck += __shfl_down(ck, off, 64);
 		mx = max(mx, __shfl_down(mx, off, 64));
 	}
 	if ((threadIdx.x & 63) == 0 && ng)
@@ -1171,25 +1172,20 @@ void k_dn_q3_pass1(const unsigned long long *__restrict__ rev,
 				atomicAdd(&out5[2], 1ull);
 			atomicAdd(&out5[2], carry);
 		}
+		atomicAdd(&out5[3], ck);
 		atomicMax(&out5[4], mx);
 	}
-	__syncthreads();
-	if (threadIdx.x == 0)
-		counts[blockIdx.x] = lcnt;
 }
 
 hipError_t
-launch_dn_q3_pass1(hipStream_t s, const unsigned long long *rev,
-		   int64_t dense_len, unsigned long long *out5,
-		   unsigned int *hist64k, unsigned long long *surv,
-		   int64_t region, unsigned long long *counts, int *out_grid)
+launch_dn_q3_stats_surv(hipStream_t s, const unsigned long long *surv,
+			int64_t nsurv, const unsigned long long *pay,
+			const unsigned long long *rev,
+			unsigned long long *out5, unsigned int *hist64k)
 {
-	int g = dn_grid(dense_len);
-
-	*out_grid = g;
-	hipLaunchKernelGGL(k_dn_q3_pass1, dim3(g), dim3(DN_THREADS), 0, s,
-			   rev, dense_len, out5, hist64k, surv, region,
-			   counts);
+	hipLaunchKernelGGL(k_dn_q3_stats_surv, dim3(dn_grid(nsurv)),
+			   dim3(DN_THREADS), 0, s, surv, nsurv, pay, rev,
+			   out5, hist64k);
 	return hipGetLastError();
 }
 
@@ -1248,122 +1244,74 @@ launch_dn_q3_threshold2(hipStream_t s, const unsigned int *hist64k, int64_t k,
 	return hipGetLastError();
 }
 
-/* walk the compacted survivors: group checksum (gathers pay only for
- * real groups) + threshold collect.  overflow is set when any region
- * spilled (host falls back to the full-sweep path). */
+/* collect candidates >= threshold from the survivor list */
 __global__ __launch_bounds__(DN_THREADS, 8)
-void k_dn_q3_finish(const unsigned long long *__restrict__ surv,
-		    const unsigned long long *__restrict__ counts,
-		    int64_t region, int64_t nregions,
-		    const unsigned long long *__restrict__ pay,
-		    const unsigned long long *__restrict__ rev,
-		    const unsigned long long *__restrict__ thr_ptr,
-		    unsigned long long *__restrict__ out5,
-		    gg_q3_result_row *__restrict__ out,
-		    unsigned long long *out_count, uint64_t cap,
-		    unsigned long long *overflow)
+void k_dn_q3_collect_surv(const unsigned long long *__restrict__ surv,
+			  int64_t nsurv,
+			  const unsigned long long *__restrict__ pay,
+			  const unsigned long long *__restrict__ rev,
+			  const unsigned long long *__restrict__ thr_ptr,
+			  gg_q3_result_row *__restrict__ out,
+			  unsigned long long *out_count, uint64_t cap)
 {
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	const unsigned long long threshold = *thr_ptr;
-	unsigned long long ck = 0;
 
-	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nsurv; i += stride)
 	{
-		unsigned long long m = counts[b];
+		int64_t k = (int64_t) surv[i];
+		unsigned long long r = rev[k];
+		bool take = (r != 0 && r >= threshold);
+		unsigned long long idx = dn_wave_append(out_count, take);
 
-		if (m > (unsigned long long) region)
+		if (!take || idx >= cap)
+			continue;
 		{
-			if (threadIdx.x == 0)
-				atomicOr(overflow, 1ull);
-			m = (unsigned long long) region;
-		}
-		const unsigned long long *seg = surv + b * region;
+			unsigned long long p = pay[k];
 
-		for (unsigned long long j = threadIdx.x; j < m;
-		     j += blockDim.x)
-		{
-			int64_t i = (int64_t) seg[j];
-			unsigned long long r = rev[i];
-			unsigned long long p = pay[i];
-			int32_t date = (int32_t) (uint32_t) p;
-			int32_t prio = (int32_t) (uint32_t) (p >> 32);
-
-			ck += gg_group_hash((unsigned long long) i, r, 0,
-					    date, prio);
-			{
-				bool take = r >= threshold;
-				unsigned long long idx =
-					dn_wave_append(out_count, take);
-
-				if (take && idx < cap)
-				{
-					out[idx].orderkey = i;
-					out[idx].rev_lo = r;
-					out[idx].rev_hi = 0;
-					out[idx].orderdate = date;
-					out[idx].shippriority = prio;
-				}
-			}
+			out[idx].orderkey = k;
+			out[idx].rev_lo = r;
+			out[idx].rev_hi = 0;
+			out[idx].orderdate = (int32_t) (uint32_t) p;
+			out[idx].shippriority = (int32_t) (uint32_t) (p >> 32);
 		}
 	}
-	for (int off = 32; off; off >>= 1)
-		ck += __shfl_down(ck, off, 64);
-	if ((threadIdx.x & 63) == 0 && ck)
-		atomicAdd(&out5[3], ck);
 }
 
 hipError_t
-launch_dn_q3_finish(hipStream_t s, const unsigned long long *surv,
-		    const unsigned long long *counts, int64_t region,
-		    int64_t nregions, const unsigned long long *pay,
-		    const unsigned long long *rev,
-		    const unsigned long long *thr_ptr,
-		    unsigned long long *out5, gg_q3_result_row *out,
-		    unsigned long long *out_count, uint64_t cap,
-		    unsigned long long *overflow)
+launch_dn_q3_collect_surv(hipStream_t s, const unsigned long long *surv,
+			  int64_t nsurv, const unsigned long long *pay,
+			  const unsigned long long *rev,
+			  const unsigned long long *thr_ptr,
+			  gg_q3_result_row *out,
+			  unsigned long long *out_count, uint64_t cap)
 {
-	int g = (int) (nregions < 1 ? 1 :
-		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
-
-	hipLaunchKernelGGL(k_dn_q3_finish, dim3(g), dim3(DN_THREADS), 0, s,
-			   surv, counts, region, nregions, pay, rev, thr_ptr,
-			   out5, out, out_count, cap, overflow);
+	hipLaunchKernelGGL(k_dn_q3_collect_surv, dim3(dn_grid(nsurv)),
+			   dim3(DN_THREADS), 0, s, surv, nsurv, pay, rev,
+			   thr_ptr, out, out_count, cap);
 	return hipGetLastError();
 }
 
-/* zero exactly the rev[] entries this pass touched (the survivor
- * list), so the next execute skips the full 1.2 GB memset */
-__global__ __launch_bounds__(DN_THREADS, 4)
-void k_dn_q3_clear(const unsigned long long *__restrict__ surv,
-		   const unsigned long long *__restrict__ counts,
-		   int64_t region, int64_t nregions,
-		   unsigned long long *__restrict__ rev)
+/* zero exactly the rev[] entries this pass touched, so the next
+ * execute skips the full dense memset */
+__global__ __launch_bounds__(DN_THREADS, 8)
+void k_dn_q3_clear_surv(const unsigned long long *__restrict__ surv,
+			int64_t nsurv, unsigned long long *__restrict__ rev)
 {
-	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
-	{
-		unsigned long long m = counts[b];
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
-		if (m > (unsigned long long) region)
-			m = (unsigned long long) region;
-		{
-			const unsigned long long *seg = surv + b * region;
-
-			for (unsigned long long j = threadIdx.x; j < m;
-			     j += blockDim.x)
-				rev[seg[j]] = 0;
-		}
-	}
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < nsurv; i += stride)
+		rev[surv[i]] = 0;
 }
 
 hipError_t
-launch_dn_q3_clear(hipStream_t s, const unsigned long long *surv,
-		   const unsigned long long *counts, int64_t region,
-		   int64_t nregions, unsigned long long *rev)
+launch_dn_q3_clear_surv(hipStream_t s, const unsigned long long *surv,
+			int64_t nsurv, unsigned long long *rev)
 {
-	int g = (int) (nregions < 1 ? 1 :
-		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
-
-	hipLaunchKernelGGL(k_dn_q3_clear, dim3(g), dim3(DN_THREADS), 0, s,
-			   surv, counts, region, nregions, rev);
+	hipLaunchKernelGGL(k_dn_q3_clear_surv, dim3(dn_grid(nsurv)),
+			   dim3(DN_THREADS), 0, s, surv, nsurv, rev);
 	return hipGetLastError();
 }
 
