@@ -688,8 +688,9 @@ ao_decode_to_device(Engine &e, const uint8_t *stream, int64_t stream_len,
  * way cdbbufferedread.c hands blocks up.  Each column goes through
  * the full storage layer (headers + CRC32C + zlib/zstd + datum-stream
  * decode) and lands DIRECTLY in a device-resident column — no host
- * round trip — then the table registers like any other.  Columns must
- * be NOT NULL (the hot path's columns are); a NULL anywhere errors.
+ * round trip — then the table registers like any other.  Columns are
+ * NOT NULL by default (a NULL anywhere errors); nullable mounts keep
+ * the AO null bitmap as a device NULL-flag array on the column.
  */
 extern "C" gg_status
 gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
@@ -842,12 +843,14 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 			break;
 
 		void *d_vals = nullptr;
+		uint8_t *d_nulls = nullptr;
 		int64_t total_rows = 0;
 
 		st = ao_decode_to_device(e, ac.stream, ac.stream_len,
 					 descs, spill, ac.dsb_version,
 					 datumlen, datumlen, &d_vals,
-					 nullptr /* NOT NULL required */ ,
+					 ac.nullable ? &d_nulls
+					 : nullptr /* else NOT NULL required */,
 					 &total_rows);
 		if (st != GG_OK)
 			break;
@@ -856,6 +859,7 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 		else if (t->nrows != total_rows)
 		{
 			(void) hipFree(d_vals);
+			(void) hipFree(d_nulls);
 			st = fail(GG_EINVAL, "column %s has %lld rows, "
 				  "table has %lld", ac.name,
 				  (long long) total_rows,
@@ -867,6 +871,7 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 
 		c.name = ac.name;
 		c.type = ac.type;
+		c.nulls = d_nulls;
 		if (ac.type == GG_COL_CHAR1 && total_rows)
 		{
 			uint8_t *d_u8 = nullptr;
@@ -883,6 +888,7 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 			if (he != hipSuccess)
 			{
 				(void) hipFree(d_u8);
+				(void) hipFree(d_nulls);
 				st = fail(GG_EGPU, "narrow: %s",
 					  hipGetErrorString(he));
 				break;
@@ -900,8 +906,12 @@ gg_engine_register_table_ao(const char *name, const gg_ao_column *cols,
 	if (st != GG_OK)
 	{
 		for (auto &c : t->cols)
+		{
 			if (c.bytes)
 				(void) hipFree(c.dev);
+			if (c.nulls)
+				(void) hipFree(c.nulls);
+		}
 		delete t;
 		return st;
 	}
@@ -1005,8 +1015,12 @@ extern "C" gg_status gg_engine_drop_table(gg_table h)
 	if (!t)
 		return fail(GG_EINVAL, "bad table handle %d", h);
 	for (auto &c : t->cols)
+	{
 		if (c.bytes)
 			(void) hipFree(c.dev);
+		if (c.nulls)
+			(void) hipFree(c.nulls);
+	}
 	delete t;
 	e.tables[h] = nullptr;
 	return GG_OK;
@@ -1905,22 +1919,28 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 	/* 3. lineitem probe + group aggregation (probe fused with the
 	 * group-by transition; group slot ≡ matched-order slot).  The
-	 * dense probe also appends each group's index to the survivor
-	 * list (0→nonzero rev transition) so the top-k never sweeps
-	 * the dense array. */
-	unsigned long long njoin = 0, nsurv = 0;
-	unsigned long long *dsurv = nullptr, *dnsurv = nullptr;
-	uint64_t surv_cap = 0;
+	 * dense probe also appends each group's index ONCE (0→nonzero
+	 * rev transition) into a per-block survivor region — LDS
+	 * counter, no contended global atomic — so the top-k never
+	 * sweeps the dense array. */
+	unsigned long long njoin = 0, sovf = 0;
+	unsigned long long *dsurv = nullptr, *dscnt = nullptr,
+		*dsovf = nullptr;
+	uint64_t region = 0;
+	int pgrid = 0;
 
 	if (ord_dlen)
 	{
-		surv_cap = (uint64_t) nmatch + 64;
+		pgrid = dn_probe_grid(li->nrows);
+		region = 2 * ((uint64_t) nmatch / (uint64_t) pgrid) + 256;
 		dsurv = (unsigned long long *)
-			p->sget("surv", surv_cap * 8);
-		dnsurv = (unsigned long long *) p->sget("nsurv", 8);
-		if (!dsurv || !dnsurv)
+			p->sget("surv", (size_t) pgrid * region * 8);
+		dscnt = (unsigned long long *)
+			p->sget("surv.cnts", (size_t) pgrid * 8);
+		dsovf = (unsigned long long *) p->sget("surv.ovf", 8);
+		if (!dsurv || !dscnt || !dsovf)
 			return fail(GG_ENOMEM, "survivor scratch");
-		GG_HIP(hipMemsetAsync(dnsurv, 0, 8, e.stream));
+		GG_HIP(hipMemsetAsync(dsovf, 0, 8, e.stream));
 	}
 	GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 	{
@@ -1930,8 +1950,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			GG_HIP(launch_dn_probe_lineitem(
 				e.stream, l_ok, l_sd, l_pc, l_dc, li->nrows,
 				cutoff, ordd_pay, ordd_rev, ord_dlen,
-				ordd_bloom, ordd_bwords, ctr, dsurv, dnsurv,
-				surv_cap));
+				ordd_bloom, ordd_bwords, ctr, dsurv, region,
+				dscnt, dsovf, pgrid));
 		else
 			GG_HIP(launch_probe_lineitem(e.stream, l_ok, l_sd,
 						     l_pc, l_dc, li->nrows,
@@ -1939,14 +1959,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		double ms = tm.stop();
 		GG_TRY(read_counter(ctr, &njoin));
 		if (ord_dlen)
-		{
-			GG_TRY(read_counter(dnsurv, &nsurv));
-			if (nsurv > surv_cap)
-				return fail(GG_ESTATE,
-					    "survivor overflow (%llu > %llu)",
-					    nsurv,
-					    (unsigned long long) surv_cap);
-		}
+			GG_TRY(read_counter(dsovf, &sovf));
 		KernelStatAcc &st = p->stat("probe_lineitem");
 
 		st.launches++;
@@ -1970,9 +1983,9 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	{
 		Timed tm(e.stream);
 		std::vector<gg_q3_result_row> cand;
-		bool need_old = !ord_dlen;
+		bool need_old = !ord_dlen || sovf != 0;
 
-		if (ord_dlen)
+		if (!need_old)
 		{
 			uint64_t cap = 4 * (uint64_t) k + 65536;
 			unsigned int *dhist =
@@ -1989,14 +2002,14 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			GG_HIP(hipMemsetAsync(dhist, 0, 65536 * 4, e.stream));
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 			GG_HIP(launch_dn_q3_stats_surv(e.stream, dsurv,
-						       (int64_t) nsurv,
+						       dscnt, region, pgrid,
 						       ordd_pay, ordd_rev,
 						       stats5, dhist));
 			GG_HIP(launch_dn_q3_threshold2(e.stream, dhist, k,
 						       dthr));
 			GG_HIP(launch_dn_q3_collect_surv(
-				e.stream, dsurv, (int64_t) nsurv, ordd_pay,
-				ordd_rev, dthr, dout, ctr, cap));
+				e.stream, dsurv, dscnt, region, pgrid,
+				ordd_pay, ordd_rev, dthr, dout, ctr, cap));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_HIP(hipMemcpy(hstats, stats5, 40,
 					 hipMemcpyDeviceToHost));
@@ -2013,9 +2026,9 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 					return fail(GG_ENOMEM, "topk retry");
 				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 				GG_HIP(launch_dn_q3_collect_surv(
-					e.stream, dsurv, (int64_t) nsurv,
-					ordd_pay, ordd_rev, dthr, dout, ctr,
-					cap));
+					e.stream, dsurv, dscnt, region,
+					pgrid, ordd_pay, ordd_rev, dthr,
+					dout, ctr, cap));
 				GG_HIP(hipStreamSynchronize(e.stream));
 				GG_TRY(read_counter(ctr, &ncand));
 				if (ncand > cap)
@@ -2034,7 +2047,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			 * execute skips the dense memset (runs async;
 			 * later work queues behind it on the stream) */
 			GG_HIP(launch_dn_q3_clear_surv(e.stream, dsurv,
-						       (int64_t) nsurv,
+						       dscnt, region, pgrid,
 						       ordd_rev));
 			p->q3_rev_dirty = false;
 		}

@@ -439,8 +439,9 @@ hipError_t launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 				    uint64_t bwords,
 				    unsigned long long *join_rows,
 				    unsigned long long *surv,
-				    unsigned long long *nsurv,
-				    uint64_t surv_cap);
+				    uint64_t region,
+				    unsigned long long *counts,
+				    unsigned long long *ovf, int grid);
 hipError_t launch_td_insert(hipStream_t s, const uint8_t *pool,
 			    const unsigned long long *offs,
 			    const uint32_t *lens, const uint8_t *nulls,
@@ -605,9 +606,11 @@ hipError_t launch_dn_probe_lineitem_q5_u8(hipStream_t s,
 					  int64_t supp_dlen,
 					  unsigned long long *acc,
 					  unsigned long long *join_rows);
+int dn_probe_grid(int64_t n);
 hipError_t launch_dn_q3_stats_surv(hipStream_t s,
 				   const unsigned long long *surv,
-				   int64_t nsurv,
+				   const unsigned long long *counts,
+				   uint64_t region, int64_t nregions,
 				   const unsigned long long *pay,
 				   const unsigned long long *rev,
 				   unsigned long long *out5,
@@ -617,7 +620,8 @@ hipError_t launch_dn_q3_threshold2(hipStream_t s,
 				   unsigned long long *out_thr);
 hipError_t launch_dn_q3_collect_surv(hipStream_t s,
 				     const unsigned long long *surv,
-				     int64_t nsurv,
+				     const unsigned long long *counts,
+				     uint64_t region, int64_t nregions,
 				     const unsigned long long *pay,
 				     const unsigned long long *rev,
 				     const unsigned long long *thr_ptr,
@@ -626,7 +630,9 @@ hipError_t launch_dn_q3_collect_surv(hipStream_t s,
 				     uint64_t cap);
 hipError_t launch_dn_q3_clear_surv(hipStream_t s,
 				   const unsigned long long *surv,
-				   int64_t nsurv, unsigned long long *rev);
+				   const unsigned long long *counts,
+				   uint64_t region, int64_t nregions,
+				   unsigned long long *rev);
 hipError_t launch_dn_q3_stats(hipStream_t s, const unsigned long long *pay,
 			      const unsigned long long *rev,
 			      int64_t dense_len, unsigned long long *out5);

@@ -399,9 +399,21 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 			   const unsigned long long *__restrict__ bloom,
 			   uint64_t bwords, unsigned long long *join_rows,
 			   unsigned long long *__restrict__ surv,
-			   unsigned long long *nsurv, uint64_t surv_cap)
+			   uint64_t region,
+			   unsigned long long *__restrict__ counts,
+			   unsigned long long *ovf)
 {
+	/* per-block survivor region: group creators append via an LDS
+	 * counter (a single global counter saturates at ~88 returning
+	 * atomics/µs — measured 13.7 ms probe when tried) */
+	__shared__ unsigned long long lsurv;
+
+	if (threadIdx.x == 0)
+		lsurv = 0;
+	__syncthreads();
+
 	unsigned long long joined = 0;
+	unsigned long long *mine = surv + (uint64_t) blockIdx.x * region;
 
 	auto body = [&](int64_t i, int32_t sd, int64_t k)
 	{
@@ -419,17 +431,17 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 
 			/* group creation (execHHashagg find-or-create):
 			 * exactly one adder sees the 0→nonzero transition,
-			 * so the survivor list gets each group once — the
+			 * so the survivor regions get each group once — the
 			 * top-k then gathers ~ngroups entries instead of
 			 * sweeping the GB-sized dense array (zero-sum
 			 * groups stay invisible, as rev==0 always meant) */
 			if (old == 0 && r4 != 0)
 			{
 				unsigned long long at =
-					atomicAdd(nsurv, 1ull);
+					atomicAdd(&lsurv, 1ull);
 
-				if (at < surv_cap)
-					surv[at] = (unsigned long long) k;
+				if (at < region)
+					mine[at] = (unsigned long long) k;
 			}
 		}
 	};
@@ -511,6 +523,13 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 		joined += __shfl_down(joined, off, 64);
 	if ((threadIdx.x & 63) == 0 && joined)
 		atomicAdd(join_rows, joined);
+	__syncthreads();
+	if (threadIdx.x == 0)
+	{
+		counts[blockIdx.x] = lsurv < region ? lsurv : region;
+		if (lsurv > region)
+			atomicOr(ovf, 1ull);
+	}
 }
 
 /* sweep-selected default: strided streams at 8 blocks/CU (see
@@ -541,6 +560,12 @@ dn_grid_env(int64_t n, const char *env)
 	return dn_grid(n);
 }
 
+/* host-visible probe grid (exec_q3 sizes the survivor regions by it) */
+int dn_probe_grid(int64_t n)
+{
+	return (dn_grid_env(n, "GG_Q3_PROBE_GRID") + 7) & ~7;
+}
+
 hipError_t
 launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 const int32_t *shipdate, const int64_t *price,
@@ -548,11 +573,11 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 unsigned long long *pay, unsigned long long *rev,
 			 int64_t dense_len, unsigned long long *bloom,
 			 uint64_t bwords, unsigned long long *join_rows,
-			 unsigned long long *surv, unsigned long long *nsurv,
-			 uint64_t surv_cap)
+			 unsigned long long *surv, uint64_t region,
+			 unsigned long long *counts, unsigned long long *ovf,
+			 int grid)
 {
-	int g = (dn_grid_env(n, "GG_Q3_PROBE_GRID") + 7) & ~7;
-	dim3 gg(g), bb(DN_THREADS);
+	dim3 gg(grid), bb(DN_THREADS);
 
 	switch (dn_probe_var("GG_Q3_PROBE_VAR", 0))
 	{
@@ -562,42 +587,48 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows, surv, nsurv, surv_cap);
+					   join_rows, surv, region, counts,
+					   ovf);
 			break;
 		case 1:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows, surv, nsurv, surv_cap);
+					   join_rows, surv, region, counts,
+					   ovf);
 			break;
 		case 2:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 8>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows, surv, nsurv, surv_cap);
+					   join_rows, surv, region, counts,
+					   ovf);
 			break;
 		case 3:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<0, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows, surv, nsurv, surv_cap);
+					   join_rows, surv, region, counts,
+					   ovf);
 			break;
 		case 4:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 8>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows, surv, nsurv, surv_cap);
+					   join_rows, surv, region, counts,
+					   ovf);
 			break;
 		case 5:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
 					   disc, n, cutoff, pay, rev,
 					   dense_len, bloom, bwords,
-					   join_rows, surv, nsurv, surv_cap);
+					   join_rows, surv, region, counts,
+					   ovf);
 			break;
 	}
 	return hipGetLastError();
@@ -1113,22 +1144,27 @@ __device__ inline unsigned long long dn_code16_lo(unsigned code)
  * the 1.2 GB dense array — the r02c PMC showed those sweeps were
  * latency-bound at 2.1 TB/s. */
 
-/* gather stats + 16-bit exponent-mantissa histogram over survivors */
+/* gather stats + 16-bit exponent-mantissa histogram over the
+ * per-block survivor regions the probe filled */
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
-			int64_t nsurv,
+			const unsigned long long *__restrict__ counts,
+			uint64_t region, int64_t nregions,
 			const unsigned long long *__restrict__ pay,
 			const unsigned long long *__restrict__ rev,
 			unsigned long long *__restrict__ out5,
 			unsigned int *__restrict__ hist64k)
 {
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long ng = 0, revsum = 0, carry = 0, ck = 0, mx = 0;
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < nsurv; i += stride)
+	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
 	{
-		int64_t k = (int64_t) surv[i];
+	const unsigned long long *seg = surv + (uint64_t) b * region;
+	unsigned long long m = counts[b];
+
+	for (unsigned long long i = threadIdx.x; i < m; i += blockDim.x)
+	{
+		int64_t k = (int64_t) seg[i];
 		unsigned long long r = rev[k];
 
 		if (!r)
@@ -1151,6 +1187,7 @@ void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
 			ck += gg_group_hash((unsigned long long) k, r, 0,
 					    date, prio);
 		}
+	}
 	}
 	for (int off = 32; off; off >>= 1)
 	{
@@ -1179,13 +1216,17 @@ void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
 
 hipError_t
 launch_dn_q3_stats_surv(hipStream_t s, const unsigned long long *surv,
-			int64_t nsurv, const unsigned long long *pay,
+			const unsigned long long *counts, uint64_t region,
+			int64_t nregions, const unsigned long long *pay,
 			const unsigned long long *rev,
 			unsigned long long *out5, unsigned int *hist64k)
 {
-	hipLaunchKernelGGL(k_dn_q3_stats_surv, dim3(dn_grid(nsurv)),
-			   dim3(DN_THREADS), 0, s, surv, nsurv, pay, rev,
-			   out5, hist64k);
+	int g = (int) (nregions < 1 ? 1 :
+		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
+
+	hipLaunchKernelGGL(k_dn_q3_stats_surv, dim3(g),
+			   dim3(DN_THREADS), 0, s, surv, counts, region,
+			   nregions, pay, rev, out5, hist64k);
 	return hipGetLastError();
 }
 
@@ -1244,52 +1285,64 @@ launch_dn_q3_threshold2(hipStream_t s, const unsigned int *hist64k, int64_t k,
 	return hipGetLastError();
 }
 
-/* collect candidates >= threshold from the survivor list */
+/* collect candidates >= threshold from the survivor regions */
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q3_collect_surv(const unsigned long long *__restrict__ surv,
-			  int64_t nsurv,
+			  const unsigned long long *__restrict__ counts,
+			  uint64_t region, int64_t nregions,
 			  const unsigned long long *__restrict__ pay,
 			  const unsigned long long *__restrict__ rev,
 			  const unsigned long long *__restrict__ thr_ptr,
 			  gg_q3_result_row *__restrict__ out,
 			  unsigned long long *out_count, uint64_t cap)
 {
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	const unsigned long long threshold = *thr_ptr;
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < nsurv; i += stride)
+	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
 	{
-		int64_t k = (int64_t) surv[i];
-		unsigned long long r = rev[k];
-		bool take = (r != 0 && r >= threshold);
-		unsigned long long idx = dn_wave_append(out_count, take);
+		const unsigned long long *seg = surv + (uint64_t) b * region;
+		unsigned long long m = counts[b];
 
-		if (!take || idx >= cap)
-			continue;
+		for (unsigned long long i = threadIdx.x; i < m;
+		     i += blockDim.x)
 		{
-			unsigned long long p = pay[k];
+			int64_t k = (int64_t) seg[i];
+			unsigned long long r = rev[k];
+			bool take = (r != 0 && r >= threshold);
+			unsigned long long idx =
+				dn_wave_append(out_count, take);
 
-			out[idx].orderkey = k;
-			out[idx].rev_lo = r;
-			out[idx].rev_hi = 0;
-			out[idx].orderdate = (int32_t) (uint32_t) p;
-			out[idx].shippriority = (int32_t) (uint32_t) (p >> 32);
+			if (!take || idx >= cap)
+				continue;
+			{
+				unsigned long long p = pay[k];
+
+				out[idx].orderkey = k;
+				out[idx].rev_lo = r;
+				out[idx].rev_hi = 0;
+				out[idx].orderdate = (int32_t) (uint32_t) p;
+				out[idx].shippriority =
+					(int32_t) (uint32_t) (p >> 32);
+			}
 		}
 	}
 }
 
 hipError_t
 launch_dn_q3_collect_surv(hipStream_t s, const unsigned long long *surv,
-			  int64_t nsurv, const unsigned long long *pay,
+			  const unsigned long long *counts, uint64_t region,
+			  int64_t nregions, const unsigned long long *pay,
 			  const unsigned long long *rev,
 			  const unsigned long long *thr_ptr,
 			  gg_q3_result_row *out,
 			  unsigned long long *out_count, uint64_t cap)
 {
-	hipLaunchKernelGGL(k_dn_q3_collect_surv, dim3(dn_grid(nsurv)),
-			   dim3(DN_THREADS), 0, s, surv, nsurv, pay, rev,
-			   thr_ptr, out, out_count, cap);
+	int g = (int) (nregions < 1 ? 1 :
+		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
+
+	hipLaunchKernelGGL(k_dn_q3_collect_surv, dim3(g),
+			   dim3(DN_THREADS), 0, s, surv, counts, region,
+			   nregions, pay, rev, thr_ptr, out, out_count, cap);
 	return hipGetLastError();
 }
 
@@ -1297,21 +1350,32 @@ launch_dn_q3_collect_surv(hipStream_t s, const unsigned long long *surv,
  * execute skips the full dense memset */
 __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q3_clear_surv(const unsigned long long *__restrict__ surv,
-			int64_t nsurv, unsigned long long *__restrict__ rev)
+			const unsigned long long *__restrict__ counts,
+			uint64_t region, int64_t nregions,
+			unsigned long long *__restrict__ rev)
 {
-	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
+	{
+		const unsigned long long *seg = surv + (uint64_t) b * region;
+		unsigned long long m = counts[b];
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < nsurv; i += stride)
-		rev[surv[i]] = 0;
+		for (unsigned long long i = threadIdx.x; i < m;
+		     i += blockDim.x)
+			rev[seg[i]] = 0;
+	}
 }
 
 hipError_t
 launch_dn_q3_clear_surv(hipStream_t s, const unsigned long long *surv,
-			int64_t nsurv, unsigned long long *rev)
+			const unsigned long long *counts, uint64_t region,
+			int64_t nregions, unsigned long long *rev)
 {
-	hipLaunchKernelGGL(k_dn_q3_clear_surv, dim3(dn_grid(nsurv)),
-			   dim3(DN_THREADS), 0, s, surv, nsurv, rev);
+	int g = (int) (nregions < 1 ? 1 :
+		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
+
+	hipLaunchKernelGGL(k_dn_q3_clear_surv, dim3(g),
+			   dim3(DN_THREADS), 0, s, surv, counts, region,
+			   nregions, rev);
 	return hipGetLastError();
 }
 

@@ -186,7 +186,8 @@ def _load():
                     ("ao_version", ctypes.c_int),
                     ("dsb_version", ctypes.c_int),
                     ("comptype", ctypes.c_int),
-                    ("text_dict", ctypes.c_int)]
+                    ("text_dict", ctypes.c_int),
+                    ("nullable", ctypes.c_int)]
     lib.gg_AOCol = _AOCol
     lib.gg_engine_table_text_dict.restype = ctypes.c_int
     lib.gg_engine_table_text_dict.argtypes = [
@@ -333,6 +334,7 @@ class Engine:
             descs[i].dsb_version = dsbv
             descs[i].comptype = ct
             descs[i].text_dict = col[7] if len(col) > 7 else 0
+            descs[i].nullable = col[8] if len(col) > 8 else 0
         h = I32()
         _check(L.gg_engine_register_table_ao(name.encode(), descs,
                                              len(cols),

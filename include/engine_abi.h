@@ -249,6 +249,13 @@ typedef struct gg_ao_column
 	 * lexicographically-sorted dictionary is retrievable with
 	 * gg_engine_table_text_dict for predicate-constant lookup. */
 	int			text_dict;
+	/* nonzero: the column may contain NULLs — the AO null bitmap
+	 * (datumstreamblock.c) lands as a device NULL-flag array on the
+	 * column, consumed by the generalized plan path under the
+	 * strict-transition/NULL-qual rules (nodeAgg.c:413,
+	 * execScan.c:185).  Zero keeps the round-1 behavior: any NULL
+	 * errors the mount. */
+	int			nullable;
 } gg_ao_column;
 
 gg_status gg_engine_register_table_ao(const char *name,
