@@ -287,7 +287,7 @@ def test_ao_mounted_nullable_column(eng):
     n = 50_000
     vals = rng.integers(-10**6, 10**6, n).astype(np.int64)
     nulls = (rng.random(n) < 0.15).astype(np.uint8)
-    framed = pyoracle.dsb_encode(vals, nulls, 8, 2, 0, 0)
+    framed, _nblocks = pyoracle.dsb_encode(vals, nulls, 8, 2, 0, 0)
     ao = pyoracle.ao_wrap(framed)
     t = eng.register_table_ao("ao_nullable", [
         ("v", "int64", ao, 1, 2, 2, 0, 0, 1)])
