@@ -69,6 +69,37 @@ pl_preds_pass(const PlanPredDev *preds, int npreds, int64_t i)
 	return true;
 }
 
+/* 4 rows at once, predicate loads UNCONDITIONAL: short-circuiting per
+ * row turns the scan into dependent sparse gathers (measured 2.3 TB/s
+ * on Q6); streaming every predicate column and ANDing the results
+ * keeps the loads wide and independent like the named pipelines. */
+__device__ static inline void
+pl_preds_pass4(const PlanPredDev *preds, int npreds, int64_t i, int64_t S,
+	       bool ok[4])
+{
+	ok[0] = ok[1] = ok[2] = ok[3] = true;
+	for (int p = 0; p < npreds; p++)
+	{
+		const PlanPredDev &P = preds[p];
+		int64_t v0 = pl_ld(P.col, P.width, i);
+		int64_t v1 = pl_ld(P.col, P.width, i + S);
+		int64_t v2 = pl_ld(P.col, P.width, i + 2 * S);
+		int64_t v3 = pl_ld(P.col, P.width, i + 3 * S);
+
+		ok[0] &= v0 >= P.lo && v0 < P.hi;
+		ok[1] &= v1 >= P.lo && v1 < P.hi;
+		ok[2] &= v2 >= P.lo && v2 < P.hi;
+		ok[3] &= v3 >= P.lo && v3 < P.hi;
+		if (P.nulls)
+		{
+			ok[0] &= !P.nulls[i];
+			ok[1] &= !P.nulls[i + S];
+			ok[2] &= !P.nulls[i + 2 * S];
+			ok[3] &= !P.nulls[i + 3 * S];
+		}
+	}
+}
+
 /* ---- semi-join build ------------------------------------------------ */
 
 __global__ __launch_bounds__(PL_THREADS, 4)
@@ -288,17 +319,17 @@ void k_plan_scan_agg(PlanDev P)
 	{
 		/* global-group fast path: register accumulators, one
 		 * wave-reduced atomic flush per wave (the Q1/sumprice
-		 * pattern — keeps e.g. Q6 at streaming rate) */
+		 * pattern — keeps e.g. Q6 at streaming rate); 4-way
+		 * strided rows with unconditional predicate streams */
 		unsigned long long alo[GG_PLAN_MAX_AGGS] = {};
 		long long ahi[GG_PLAN_MAX_AGGS] = {};
 
-		for (int64_t i = (int64_t) blockIdx.x * blockDim.x +
-		     threadIdx.x; i < P.n; i += stride)
+		auto tail = [&](int64_t i)
 		{
 			if (!pl_preds_pass(P.preds, P.npreds, i))
-				continue;
+				return;
 			if (!pl_joins_pass(P.joins, P.njoins, i))
-				continue;
+				return;
 			for (int a = 0; a < P.naggs; a++)
 			{
 				__int128 v;
@@ -315,7 +346,44 @@ void k_plan_scan_agg(PlanDev P)
 						(alo[a] < old);
 				}
 			}
+		};
+		int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+
+		for (; i + 3 * stride < P.n; i += 4 * stride)
+		{
+			bool ok[4];
+
+			pl_preds_pass4(P.preds, P.npreds, i, stride, ok);
+			for (int j = 0; j < 4; j++)
+			{
+				int64_t r = i + j * stride;
+
+				if (!ok[j])
+					continue;
+				if (!pl_joins_pass(P.joins, P.njoins, r))
+					continue;
+				for (int a = 0; a < P.naggs; a++)
+				{
+					__int128 v;
+
+					if (!pl_agg_val(P.aggs[a], r, &v))
+						continue;
+					{
+						unsigned long long vlo =
+							(unsigned long long) v;
+						unsigned long long old =
+							alo[a];
+
+						alo[a] += vlo;
+						ahi[a] += (long long)
+							(v >> 64) +
+							(alo[a] < old);
+					}
+				}
+			}
 		}
+		for (; i < P.n; i += stride)
+			tail(i);
 		for (int a = 0; a < P.naggs; a++)
 		{
 			for (int off = 32; off; off >>= 1)
@@ -341,13 +409,10 @@ void k_plan_scan_agg(PlanDev P)
 		return;
 	}
 
-	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	     i < P.n; i += stride)
+	auto grouped_row = [&](int64_t i)
 	{
-		if (!pl_preds_pass(P.preds, P.npreds, i))
-			continue;
 		if (!pl_joins_pass(P.joins, P.njoins, i))
-			continue;
+			return;
 		{
 			long long code = pl_group_code(P, i);
 			int64_t slot = pl_slot(P, code);
@@ -355,7 +420,7 @@ void k_plan_scan_agg(PlanDev P)
 			if (slot < 0)
 			{
 				atomicOr(P.err, 1ull);
-				continue;
+				return;
 			}
 			for (int a = 0; a < P.naggs; a++)
 			{
@@ -369,7 +434,21 @@ void k_plan_scan_agg(PlanDev P)
 					v);
 			}
 		}
+	};
+	int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+
+	for (; i + 3 * stride < P.n; i += 4 * stride)
+	{
+		bool ok[4];
+
+		pl_preds_pass4(P.preds, P.npreds, i, stride, ok);
+		for (int j = 0; j < 4; j++)
+			if (ok[j])
+				grouped_row(i + j * stride);
 	}
+	for (; i < P.n; i += stride)
+		if (pl_preds_pass(P.preds, P.npreds, i))
+			grouped_row(i);
 }
 
 hipError_t launch_plan_scan_agg(hipStream_t s, const PlanDev &p)

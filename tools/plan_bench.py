@@ -18,12 +18,12 @@ p = eng.compile_plan(
     li,
     preds=[("shipdate", lo, hi), ("disc", 5, 8), ("qty", NEG_INF, 2400)],
     aggs=[("sum", [("price", "id"), ("disc", "id")]), "count"])
-g0 = eng.execute_plan(p)
+g0 = eng.execute_plan(p, max_groups=8)
 before = {s["name"]: dict(s) for s in eng.stats(p)}
 t0 = time.perf_counter()
 steps = 10
 for _ in range(steps):
-    g = eng.execute_plan(p)
+    g = eng.execute_plan(p, max_groups=8)
 wall = (time.perf_counter() - t0) / steps
 assert g == g0
 after = {s["name"]: s for s in eng.stats(p)}
