@@ -2620,7 +2620,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
 			double ms0 = tm.stop();
-			KernelStatAcc &st0 = p->stat("build_orders");
+			KernelStatAcc &st0 = p->stat("build_orders_q5");
 
 			st0.launches++;
 			st0.total_ms += ms0;
@@ -2663,7 +2663,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 		GG_TRY(read_counter(ctr, &nmatch));
 		{
 			double ms = tm.stop();
-			KernelStatAcc &st = p->stat("build_orders");
+			KernelStatAcc &st = p->stat("build_orders_q5");
 
 			st.launches++;
 			st.total_ms += ms;
