@@ -70,6 +70,22 @@ def main():
                      cutoff_hi=PGDate("1998-01-01"), regionkey=1)
     out["q5"] = eng.execute_q5(p5)
 
+    # generic-descriptor plans at world>1: exercises exec_plan's
+    # partial-group allgather combine (2-stage agg)
+    from greengage_amd import pgdate
+    from greengage_amd.engine import NEG_INF
+    lo, hi = pgdate(1994, 1, 1), pgdate(1995, 1, 1)
+    pq6 = eng.compile_plan(
+        li, preds=[("shipdate", lo, hi), ("disc", 5, 8),
+                   ("qty", NEG_INF, 2400)],
+        aggs=[("sum", [("price", "id"), ("disc", "id")]), "count"])
+    out["plan_q6"] = eng.execute_plan(pq6, max_groups=8)
+    pg = eng.compile_plan(
+        li, preds=[("shipdate", NEG_INF, PGDate("1998-08-15") + 1)],
+        group_cols=["rflag", "lstatus"],
+        aggs=["count", ("sum", [("qty", "id")])])
+    out["plan_grouped"] = eng.execute_plan(pg, max_groups=64)
+
     eng.shutdown()
     with open(outfile + ".tmp", "w") as f:
         json.dump(out, f)

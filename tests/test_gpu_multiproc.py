@@ -68,6 +68,8 @@ def check_results(results, world, sf):
         assert r["q3_hdr"] == results[0]["q3_hdr"]
         assert r["q3_rows"] == results[0]["q3_rows"]
         assert r["q5"] == results[0]["q5"]
+        assert r["plan_q6"] == results[0]["plan_q6"]
+        assert r["plan_grouped"] == results[0]["plan_grouped"]
 
     exp_q1 = [g for g in pyoracle.q1_synth(42, sf, PGDate("1998-08-15"))
               if g["count"]]
@@ -92,6 +94,32 @@ def check_results(results, world, sf):
     assert [(r["nationkey"], r["count"], r["revenue4"]) for r in
             results[0]["q5"]] == \
            [(r["nationkey"], r["count"], r["revenue4"]) for r in exp_q5]
+
+    # generic-descriptor plans combined across segments
+    import numpy as np
+    from greengage_amd import pgdate
+    g = pyoracle.gen_lineitem(42, 0, 6_000_000 * sf)
+    lo, hi = pgdate(1994, 1, 1), pgdate(1995, 1, 1)
+    m = ((g["shipdate"] >= lo) & (g["shipdate"] < hi) & (g["disc"] >= 5)
+         & (g["disc"] < 8) & (g["qty"] < 2400))
+    exp_rev = sum(int(p) * int(d) for p, d in
+                  zip(g["price"][m].tolist(), g["disc"][m].tolist()))
+    q6 = results[0]["plan_q6"]
+    assert q6[0][2][0] == exp_rev
+    assert q6[0][2][1] == int(np.count_nonzero(m))
+
+    cutoff = PGDate("1998-08-15")
+    mg = g["shipdate"] <= cutoff
+    grouped = results[0]["plan_grouped"]
+    exp_g = {}
+    for rf, ls in {(int(a), int(b)) for a, b in
+                   zip(g["rflag"][mg].tolist(), g["lstatus"][mg].tolist())}:
+        sel = mg & (g["rflag"] == rf) & (g["lstatus"] == ls)
+        exp_g[(rf, ls)] = [int(np.count_nonzero(sel)),
+                           int(g["qty"][sel].sum())]
+    assert len(grouped) == len(exp_g)
+    for k0, k1, vals in grouped:
+        assert vals == exp_g[(k0, k1)]
 
 
 def test_world2_q1_q3_q5_bitexact(tmp_path):
