@@ -1178,7 +1178,16 @@ void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
 			revsum += r;
 			carry += (revsum < old);
 		}
-		atomicAdd(&hist64k[dn_code16(r)], 1u);
+		/* SUBSET-sampled histogram: every 16th survivor.  The
+		 * threshold walk still targets k, and k sampled values
+		 * >= thr imply >= k TRUE values >= thr (samples are a
+		 * subset), so the threshold is always admissible; the
+		 * candidate set just inflates ~16x k (collected exactly,
+		 * host-sorts the tail).  This cuts the global hist
+		 * atomics 16x — they dominated this kernel (537 µs vs
+		 * the 25 µs the same gathers cost in collect_surv). */
+		if ((i & 15) == 0)
+			atomicAdd(&hist64k[dn_code16(r)], 1u);
 		{
 			unsigned long long p = pay[k];
 			int32_t date = (int32_t) (uint32_t) p;
