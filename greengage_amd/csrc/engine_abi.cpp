@@ -1284,10 +1284,11 @@ static gg_status exec_sumprice(Pipeline *p, void *arena, size_t bytes,
 
 	if (!sd || !pr)
 		return fail(GG_EINVAL, "lineitem lacks shipdate/price");
-	SumPriceAcc *acc;
+	SumPriceAcc *acc = (SumPriceAcc *) p->sget("sp.acc", sizeof(*acc));
 
-	GG_HIP(hipMalloc((void **) &acc, sizeof(*acc)));
-	GG_HIP(hipMemset(acc, 0, sizeof(*acc)));
+	if (!acc)
+		return fail(GG_ENOMEM, "sumprice acc");
+	GG_HIP(hipMemsetAsync(acc, 0, sizeof(*acc), e.stream));
 	{
 		Timed tm(e.stream);
 
@@ -1304,7 +1305,6 @@ static gg_status exec_sumprice(Pipeline *p, void *arena, size_t bytes,
 	SumPriceAcc h;
 
 	GG_HIP(hipMemcpy(&h, acc, sizeof(h), hipMemcpyDeviceToHost));
-	(void) hipFree(acc);
 
 	u128 sum = h.sum_c;
 	u128 cnt = h.count;
@@ -1360,10 +1360,11 @@ static gg_status exec_q1(Pipeline *p, void *arena, size_t bytes,
 	if (!sd || !rf || !ls || !q || !pr || !d || !tx)
 		return fail(GG_EINVAL, "lineitem lacks a Q1 column");
 
-	Q1DeviceAcc *acc;
+	Q1DeviceAcc *acc = (Q1DeviceAcc *) p->sget("q1.acc", sizeof(*acc));
 
-	GG_HIP(hipMalloc((void **) &acc, sizeof(*acc)));
-	GG_HIP(hipMemset(acc, 0, sizeof(*acc)));
+	if (!acc)
+		return fail(GG_ENOMEM, "q1 acc");
+	GG_HIP(hipMemsetAsync(acc, 0, sizeof(*acc), e.stream));
 	{
 		Timed tm(e.stream);
 
@@ -1406,7 +1407,6 @@ static gg_status exec_q1(Pipeline *p, void *arena, size_t bytes,
 		GG_HIP(hipMemcpy(parts.data(), acc, 49 * 8,
 				 hipMemcpyDeviceToHost));
 	}
-	(void) hipFree(acc);
 
 	gg_q1_result *res = (gg_q1_result *) arena;
 
