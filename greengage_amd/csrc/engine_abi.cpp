@@ -197,7 +197,11 @@ extern "C" gg_status gg_engine_shutdown(void)
 	{
 		if (t)
 			for (auto &c : t->cols)
+			{
 				(void) hipFree(c.dev);
+				if (c.nulls)
+					(void) hipFree(c.nulls);
+			}
 		delete t;
 	}
 	e.tables.clear();
