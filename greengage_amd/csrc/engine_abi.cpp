@@ -1603,8 +1603,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	 * collapses the orders hash table to direct-map pay/rev arrays
 	 * (q3_dense.hip); hash table fallback otherwise. */
 	DeviceHashTable ord{};
-	unsigned long long *ordd_pay = nullptr;
-	unsigned long long *ordd_rev = nullptr;
+	unsigned long long *ordd_pr = nullptr;	/* interleaved (rev, pay) */
 	unsigned long long *ordd_bloom = nullptr;
 	uint64_t ordd_bwords = 0;
 	int64_t ord_dlen = 0;
@@ -1623,24 +1622,24 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 	{
 		/* exact membership bitmap over dense keys */
 		ordd_bwords = (uint64_t) (ord_dlen / 64 + 2);
-		ordd_pay = (unsigned long long *)
-			p->sget("ordd.pay", (size_t) ord_dlen * 8);
-		ordd_rev = (unsigned long long *)
-			p->sget("ordd.rev", (size_t) ord_dlen * 8);
+		/* (rev, pay) interleave as 16-B pairs: the top-k's cold
+		 * random gathers then touch ONE cache line per survivor */
+		ordd_pr = (unsigned long long *)
+			p->sget("ordd.pr", (size_t) ord_dlen * 16);
 		ordd_bloom = (unsigned long long *)
 			p->sget("ordd.bloom", ordd_bwords * 8);
-		if (!ordd_pay || !ordd_rev || !ordd_bloom)
+		if (!ordd_pr || !ordd_bloom)
 			return fail(GG_ENOMEM, "ord dense");
-		/* pay[] needs NO init: entries are read only behind the
+		/* pay slots need NO init: they are read only behind the
 		 * exact membership bitmap, which is rebuilt every pass,
-		 * so any read slot was stored this pass.  rev[] is fully
-		 * zeroed only on first use / after a fallback pass; the
-		 * fused top-k otherwise zeroes exactly the entries the
-		 * probe touched (k_dn_q3_clear over the survivor list). */
+		 * so any read slot was stored this pass.  rev slots are
+		 * fully zeroed only on first use / after a fallback pass;
+		 * the fused top-k otherwise zeroes exactly the entries
+		 * the probe touched (clear over the survivor list). */
 		if (!p->q3_rev_inited || p->q3_rev_dirty)
 		{
-			GG_HIP(hipMemsetAsync(ordd_rev, 0,
-					      (size_t) ord_dlen * 8,
+			GG_HIP(hipMemsetAsync(ordd_pr, 0,
+					      (size_t) ord_dlen * 16,
 					      e.stream));
 			p->q3_rev_inited = true;
 		}
@@ -1658,7 +1657,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		{
 			GG_HIP(launch_dn_build_orders(
 				e.stream, o_ok, o_ck, o_dt, o_pr, od->nrows,
-				cutoff, cust, cust_bits, cust_dlen, ordd_pay,
+				cutoff, cust, cust_bits, cust_dlen, ordd_pr,
 				ord_dlen, ordd_bloom, ordd_bwords, ctr));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_TRY(read_counter(ctr, &nmatch));
@@ -1879,7 +1878,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		{
 			GG_HIP(launch_dn_insert_orders(e.stream, r2_ok, r2_pay,
 						       (int64_t) rtotal2,
-						       ordd_pay, ord_dlen,
+						       ordd_pr, ord_dlen,
 						       ordd_bloom,
 						       ordd_bwords));
 			GG_HIP(hipStreamSynchronize(e.stream));
@@ -1953,7 +1952,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		if (ord_dlen)
 			GG_HIP(launch_dn_probe_lineitem(
 				e.stream, l_ok, l_sd, l_pc, l_dc, li->nrows,
-				cutoff, ordd_pay, ordd_rev, ord_dlen,
+				cutoff, ordd_pr, ord_dlen,
 				ordd_bloom, ordd_bwords, ctr, dsurv, region,
 				dscnt, dsovf, pgrid));
 		else
@@ -2007,13 +2006,13 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 			GG_HIP(launch_dn_q3_stats_surv(e.stream, dsurv,
 						       dscnt, region, pgrid,
-						       ordd_pay, ordd_rev,
+						       ordd_pr,
 						       stats5, dhist));
 			GG_HIP(launch_dn_q3_threshold2(e.stream, dhist, k,
 						       dthr));
 			GG_HIP(launch_dn_q3_collect_surv(
 				e.stream, dsurv, dscnt, region, pgrid,
-				ordd_pay, ordd_rev, dthr, dout, ctr, cap));
+				ordd_pr, dthr, dout, ctr, cap));
 			GG_HIP(hipStreamSynchronize(e.stream));
 			GG_HIP(hipMemcpy(hstats, stats5, 40,
 					 hipMemcpyDeviceToHost));
@@ -2031,7 +2030,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
 				GG_HIP(launch_dn_q3_collect_surv(
 					e.stream, dsurv, dscnt, region,
-					pgrid, ordd_pay, ordd_rev, dthr,
+					pgrid, ordd_pr, dthr,
 					dout, ctr, cap));
 				GG_HIP(hipStreamSynchronize(e.stream));
 				GG_TRY(read_counter(ctr, &ncand));
@@ -2052,7 +2051,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 			 * later work queues behind it on the stream) */
 			GG_HIP(launch_dn_q3_clear_surv(e.stream, dsurv,
 						       dscnt, region, pgrid,
-						       ordd_rev));
+						       ordd_pr));
 			p->q3_rev_dirty = false;
 		}
 
@@ -2060,8 +2059,8 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 		{
 			GG_HIP(hipMemsetAsync(stats5, 0, 5 * 8, e.stream));
 			if (ord_dlen)
-				GG_HIP(launch_dn_q3_stats(e.stream, ordd_pay,
-							  ordd_rev, ord_dlen,
+				GG_HIP(launch_dn_q3_stats(e.stream, ordd_pr,
+							  ord_dlen,
 							  stats5));
 			else
 				GG_HIP(launch_q3_stats(e.stream, ord, stats5));
@@ -2092,13 +2091,13 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 				if (ord_dlen)
 				{
 					GG_HIP(launch_dn_q3_hist(
-						e.stream, ordd_rev, ord_dlen,
+						e.stream, ordd_pr, ord_dlen,
 						stats5, dhist));
 					GG_HIP(launch_q3_threshold(
 						e.stream, dhist, stats5, k,
 						dthr));
 					GG_HIP(launch_dn_q3_collect(
-						e.stream, ordd_pay, ordd_rev,
+						e.stream, ordd_pr,
 						ord_dlen, dthr, dout, ctr,
 						cap));
 				}

@@ -418,13 +418,13 @@ hipError_t launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 				  const int32_t *prio, int64_t n,
 				  int32_t cutoff, DeviceHashTable cust,
 				  const unsigned long long *cust_bits,
-				  int64_t cust_dlen, unsigned long long *pay,
+				  int64_t cust_dlen, unsigned long long *pr,
 				  int64_t dense_len, unsigned long long *bloom,
 				  uint64_t bwords,
 				  unsigned long long *match_count);
 hipError_t launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 				   const int64_t *rowpay, int64_t n,
-				   unsigned long long *pay,
+				   unsigned long long *pr,
 				   int64_t dense_len,
 				   unsigned long long *bloom,
 				   uint64_t bwords);
@@ -432,8 +432,7 @@ hipError_t launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 				    const int32_t *shipdate,
 				    const int64_t *price,
 				    const int64_t *disc, int64_t n,
-				    int32_t cutoff, unsigned long long *pay,
-				    unsigned long long *rev,
+				    int32_t cutoff, unsigned long long *pr,
 				    int64_t dense_len,
 				    unsigned long long *bloom,
 				    uint64_t bwords,
@@ -611,8 +610,7 @@ hipError_t launch_dn_q3_stats_surv(hipStream_t s,
 				   const unsigned long long *surv,
 				   const unsigned long long *counts,
 				   uint64_t region, int64_t nregions,
-				   const unsigned long long *pay,
-				   const unsigned long long *rev,
+				   const unsigned long long *pr,
 				   unsigned long long *out5,
 				   unsigned int *hist64k);
 hipError_t launch_dn_q3_threshold2(hipStream_t s,
@@ -622,8 +620,7 @@ hipError_t launch_dn_q3_collect_surv(hipStream_t s,
 				     const unsigned long long *surv,
 				     const unsigned long long *counts,
 				     uint64_t region, int64_t nregions,
-				     const unsigned long long *pay,
-				     const unsigned long long *rev,
+				     const unsigned long long *pr,
 				     const unsigned long long *thr_ptr,
 				     gg_q3_result_row *out,
 				     unsigned long long *out_count,
@@ -632,17 +629,15 @@ hipError_t launch_dn_q3_clear_surv(hipStream_t s,
 				   const unsigned long long *surv,
 				   const unsigned long long *counts,
 				   uint64_t region, int64_t nregions,
-				   unsigned long long *rev);
-hipError_t launch_dn_q3_stats(hipStream_t s, const unsigned long long *pay,
-			      const unsigned long long *rev,
+				   unsigned long long *pr);
+hipError_t launch_dn_q3_stats(hipStream_t s, const unsigned long long *pr,
 			      int64_t dense_len, unsigned long long *out5);
-hipError_t launch_dn_q3_hist(hipStream_t s, const unsigned long long *rev,
+hipError_t launch_dn_q3_hist(hipStream_t s, const unsigned long long *pr,
 			     int64_t dense_len,
 			     const unsigned long long *stats5,
 			     unsigned int *hist64k);
 hipError_t launch_dn_q3_collect(hipStream_t s,
-				const unsigned long long *pay,
-				const unsigned long long *rev,
+				const unsigned long long *pr,
 				int64_t dense_len,
 				const unsigned long long *thr_ptr,
 				gg_q3_result_row *out,

@@ -239,7 +239,7 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 		  uint64_t cust_bwords,
 		  const unsigned long long *__restrict__ cust_bits,
 		  int64_t cust_dlen,
-		  unsigned long long *__restrict__ pay, int64_t dense_len,
+		  unsigned long long *__restrict__ pr, int64_t dense_len,
 		  unsigned long long *__restrict__ bloom, uint64_t bwords,
 		  unsigned long long *match_count)
 {
@@ -318,7 +318,7 @@ void k_dn_build_orders(const int64_t *__restrict__ okey,
 			__builtin_nontemporal_store(
 				(unsigned long long) (uint32_t) d |
 				((unsigned long long) (uint32_t) prio[i]
-				 << 32), &pay[k]);
+				 << 32), &pr[2 * k + 1]);
 		}
 		dn_bit_set_wave(bloom, k, ok);
 	}
@@ -334,14 +334,14 @@ launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 		       const int32_t *prio, int64_t n, int32_t cutoff,
 		       DeviceHashTable cust,
 		       const unsigned long long *cust_bits,
-		       int64_t cust_dlen, unsigned long long *pay,
+		       int64_t cust_dlen, unsigned long long *pr,
 		       int64_t dense_len, unsigned long long *bloom,
 		       uint64_t bwords, unsigned long long *match_count)
 {
 	hipLaunchKernelGGL(k_dn_build_orders, dim3(dn_grid(n)),
 			   dim3(DN_THREADS), 0, s, okey, ckey, odate, prio, n,
 			   cutoff, cust.keys, cust.nslots, cust.bloom,
-			   cust.bloom_words, cust_bits, cust_dlen, pay,
+			   cust.bloom_words, cust_bits, cust_dlen, pr,
 			   dense_len, bloom, bwords, match_count);
 	return hipGetLastError();
 }
@@ -350,7 +350,7 @@ launch_dn_build_orders(hipStream_t s, const int64_t *okey,
 __global__ void
 k_dn_insert_orders(const int64_t *__restrict__ okey,
 		   const int64_t *__restrict__ rowpay, int64_t n,
-		   unsigned long long *__restrict__ pay, int64_t dense_len,
+		   unsigned long long *__restrict__ pr, int64_t dense_len,
 		   unsigned long long *__restrict__ bloom, uint64_t bwords)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
@@ -362,7 +362,7 @@ k_dn_insert_orders(const int64_t *__restrict__ okey,
 
 		if (k < 0 || k >= dense_len)
 			continue;
-		pay[k] = (unsigned long long) rowpay[i];
+		pr[2 * k + 1] = (unsigned long long) rowpay[i];
 		dn_bit_set(bloom, k);
 	}
 }
@@ -370,11 +370,11 @@ k_dn_insert_orders(const int64_t *__restrict__ okey,
 hipError_t
 launch_dn_insert_orders(hipStream_t s, const int64_t *okey,
 			const int64_t *rowpay, int64_t n,
-			unsigned long long *pay, int64_t dense_len,
+			unsigned long long *pr, int64_t dense_len,
 			unsigned long long *bloom, uint64_t bwords)
 {
 	hipLaunchKernelGGL(k_dn_insert_orders, dim3(dn_grid(n)),
-			   dim3(DN_THREADS), 0, s, okey, rowpay, n, pay,
+			   dim3(DN_THREADS), 0, s, okey, rowpay, n, pr,
 			   dense_len, bloom, bwords);
 	return hipGetLastError();
 }
@@ -393,8 +393,7 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 			   const int64_t *__restrict__ price,
 			   const int64_t *__restrict__ disc, int64_t n,
 			   int32_t cutoff,
-			   const unsigned long long *__restrict__ pay,
-			   unsigned long long *__restrict__ rev,
+			   unsigned long long *__restrict__ pr,
 			   int64_t dense_len,
 			   const unsigned long long *__restrict__ bloom,
 			   uint64_t bwords, unsigned long long *join_rows,
@@ -427,7 +426,7 @@ void k_dn_probe_lineitem_t(const int64_t *__restrict__ okey,
 		{
 			unsigned long long r4 = (unsigned long long)
 				(price[i] * (100 - disc[i]));
-			unsigned long long old = atomicAdd(&rev[k], r4);
+			unsigned long long old = atomicAdd(&pr[2 * k], r4);
 
 			/* group creation (execHHashagg find-or-create):
 			 * exactly one adder sees the 0→nonzero transition,
@@ -570,7 +569,7 @@ hipError_t
 launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 			 const int32_t *shipdate, const int64_t *price,
 			 const int64_t *disc, int64_t n, int32_t cutoff,
-			 unsigned long long *pay, unsigned long long *rev,
+			 unsigned long long *pr,
 			 int64_t dense_len, unsigned long long *bloom,
 			 uint64_t bwords, unsigned long long *join_rows,
 			 unsigned long long *surv, uint64_t region,
@@ -585,7 +584,7 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 		case 0:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<0, 8>), gg,
 					   bb, 0, s, okey, shipdate, price,
-					   disc, n, cutoff, pay, rev,
+					   disc, n, cutoff, pr,
 					   dense_len, bloom, bwords,
 					   join_rows, surv, region, counts,
 					   ovf);
@@ -593,7 +592,7 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 		case 1:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
-					   disc, n, cutoff, pay, rev,
+					   disc, n, cutoff, pr,
 					   dense_len, bloom, bwords,
 					   join_rows, surv, region, counts,
 					   ovf);
@@ -601,7 +600,7 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 		case 2:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<1, 8>), gg,
 					   bb, 0, s, okey, shipdate, price,
-					   disc, n, cutoff, pay, rev,
+					   disc, n, cutoff, pr,
 					   dense_len, bloom, bwords,
 					   join_rows, surv, region, counts,
 					   ovf);
@@ -609,7 +608,7 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 		case 3:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<0, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
-					   disc, n, cutoff, pay, rev,
+					   disc, n, cutoff, pr,
 					   dense_len, bloom, bwords,
 					   join_rows, surv, region, counts,
 					   ovf);
@@ -617,7 +616,7 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 		case 4:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 8>), gg,
 					   bb, 0, s, okey, shipdate, price,
-					   disc, n, cutoff, pay, rev,
+					   disc, n, cutoff, pr,
 					   dense_len, bloom, bwords,
 					   join_rows, surv, region, counts,
 					   ovf);
@@ -625,7 +624,7 @@ launch_dn_probe_lineitem(hipStream_t s, const int64_t *okey,
 		case 5:
 			hipLaunchKernelGGL((k_dn_probe_lineitem_t<2, 4>), gg,
 					   bb, 0, s, okey, shipdate, price,
-					   disc, n, cutoff, pay, rev,
+					   disc, n, cutoff, pr,
 					   dense_len, bloom, bwords,
 					   join_rows, surv, region, counts,
 					   ovf);
@@ -1150,8 +1149,7 @@ __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
 			const unsigned long long *__restrict__ counts,
 			uint64_t region, int64_t nregions,
-			const unsigned long long *__restrict__ pay,
-			const unsigned long long *__restrict__ rev,
+			const unsigned long long *__restrict__ pr,
 			unsigned long long *__restrict__ out5,
 			unsigned int *__restrict__ hist64k)
 {
@@ -1165,7 +1163,11 @@ void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
 	for (unsigned long long i = threadIdx.x; i < m; i += blockDim.x)
 	{
 		int64_t k = (int64_t) seg[i];
-		unsigned long long r = rev[k];
+		/* rev and pay interleave as a 16-B pair: one cache line
+		 * per survivor instead of two (this kernel is cold-
+		 * random-gather bound) */
+		gg_ull2 rp = *(const gg_ull2 *) &pr[2 * k];
+		unsigned long long r = rp.x;
 
 		if (!r)
 			continue;	/* zero-sum group: invisible, as the
@@ -1189,7 +1191,7 @@ void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
 		if ((i & 15) == 0)
 			atomicAdd(&hist64k[dn_code16(r)], 1u);
 		{
-			unsigned long long p = pay[k];
+			unsigned long long p = rp.y;
 			int32_t date = (int32_t) (uint32_t) p;
 			int32_t prio = (int32_t) (uint32_t) (p >> 32);
 
@@ -1226,8 +1228,7 @@ void k_dn_q3_stats_surv(const unsigned long long *__restrict__ surv,
 hipError_t
 launch_dn_q3_stats_surv(hipStream_t s, const unsigned long long *surv,
 			const unsigned long long *counts, uint64_t region,
-			int64_t nregions, const unsigned long long *pay,
-			const unsigned long long *rev,
+			int64_t nregions, const unsigned long long *pr,
 			unsigned long long *out5, unsigned int *hist64k)
 {
 	int g = (int) (nregions < 1 ? 1 :
@@ -1235,7 +1236,7 @@ launch_dn_q3_stats_surv(hipStream_t s, const unsigned long long *surv,
 
 	hipLaunchKernelGGL(k_dn_q3_stats_surv, dim3(g),
 			   dim3(DN_THREADS), 0, s, surv, counts, region,
-			   nregions, pay, rev, out5, hist64k);
+			   nregions, pr, out5, hist64k);
 	return hipGetLastError();
 }
 
@@ -1299,8 +1300,7 @@ __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q3_collect_surv(const unsigned long long *__restrict__ surv,
 			  const unsigned long long *__restrict__ counts,
 			  uint64_t region, int64_t nregions,
-			  const unsigned long long *__restrict__ pay,
-			  const unsigned long long *__restrict__ rev,
+			  const unsigned long long *__restrict__ pr,
 			  const unsigned long long *__restrict__ thr_ptr,
 			  gg_q3_result_row *__restrict__ out,
 			  unsigned long long *out_count, uint64_t cap)
@@ -1316,7 +1316,8 @@ void k_dn_q3_collect_surv(const unsigned long long *__restrict__ surv,
 		     i += blockDim.x)
 		{
 			int64_t k = (int64_t) seg[i];
-			unsigned long long r = rev[k];
+			gg_ull2 rp = *(const gg_ull2 *) &pr[2 * k];
+			unsigned long long r = rp.x;
 			bool take = (r != 0 && r >= threshold);
 			unsigned long long idx =
 				dn_wave_append(out_count, take);
@@ -1324,7 +1325,7 @@ void k_dn_q3_collect_surv(const unsigned long long *__restrict__ surv,
 			if (!take || idx >= cap)
 				continue;
 			{
-				unsigned long long p = pay[k];
+				unsigned long long p = rp.y;
 
 				out[idx].orderkey = k;
 				out[idx].rev_lo = r;
@@ -1340,8 +1341,7 @@ void k_dn_q3_collect_surv(const unsigned long long *__restrict__ surv,
 hipError_t
 launch_dn_q3_collect_surv(hipStream_t s, const unsigned long long *surv,
 			  const unsigned long long *counts, uint64_t region,
-			  int64_t nregions, const unsigned long long *pay,
-			  const unsigned long long *rev,
+			  int64_t nregions, const unsigned long long *pr,
 			  const unsigned long long *thr_ptr,
 			  gg_q3_result_row *out,
 			  unsigned long long *out_count, uint64_t cap)
@@ -1351,7 +1351,7 @@ launch_dn_q3_collect_surv(hipStream_t s, const unsigned long long *surv,
 
 	hipLaunchKernelGGL(k_dn_q3_collect_surv, dim3(g),
 			   dim3(DN_THREADS), 0, s, surv, counts, region,
-			   nregions, pay, rev, thr_ptr, out, out_count, cap);
+			   nregions, pr, thr_ptr, out, out_count, cap);
 	return hipGetLastError();
 }
 
@@ -1361,7 +1361,7 @@ __global__ __launch_bounds__(DN_THREADS, 8)
 void k_dn_q3_clear_surv(const unsigned long long *__restrict__ surv,
 			const unsigned long long *__restrict__ counts,
 			uint64_t region, int64_t nregions,
-			unsigned long long *__restrict__ rev)
+			unsigned long long *__restrict__ pr)
 {
 	for (int64_t b = blockIdx.x; b < nregions; b += gridDim.x)
 	{
@@ -1370,37 +1370,35 @@ void k_dn_q3_clear_surv(const unsigned long long *__restrict__ surv,
 
 		for (unsigned long long i = threadIdx.x; i < m;
 		     i += blockDim.x)
-			rev[seg[i]] = 0;
+			pr[2 * seg[i]] = 0;
 	}
 }
 
 hipError_t
 launch_dn_q3_clear_surv(hipStream_t s, const unsigned long long *surv,
 			const unsigned long long *counts, uint64_t region,
-			int64_t nregions, unsigned long long *rev)
+			int64_t nregions, unsigned long long *pr)
 {
 	int g = (int) (nregions < 1 ? 1 :
 		       (nregions > DN_MAX_BLOCKS ? DN_MAX_BLOCKS : nregions));
 
 	hipLaunchKernelGGL(k_dn_q3_clear_surv, dim3(g),
 			   dim3(DN_THREADS), 0, s, surv, counts, region,
-			   nregions, rev);
+			   nregions, pr);
 	return hipGetLastError();
 }
 
 /* stats (+max) over the dense group arrays (orderkey = index) */
 __global__ __launch_bounds__(DN_THREADS, 8)
-void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
-	      const unsigned long long *__restrict__ rev, int64_t dense_len,
-	      unsigned long long *out5)
+void k_dn_q3_stats(const unsigned long long *__restrict__ pr,
+	      int64_t dense_len, unsigned long long *out5)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	unsigned long long ng = 0, revsum = 0, carry = 0, ck = 0, mx = 0;
 
-	/* 4-way unrolled: a single-stream scan needs explicit
-	 * memory-level parallelism to hide NT-load latency (one load per
-	 * iteration measured 1.9 TB/s; Q1's five streams hit 6.4) */
-	auto body = [&](int64_t i, unsigned long long r)
+	/* fallback full sweep over the interleaved (rev, pay) pairs —
+	 * 16-B NT loads, 2 per iteration for memory-level parallelism */
+	auto body = [&](int64_t i, unsigned long long r, unsigned long long p)
 	{
 		if (!r)
 			return;
@@ -1413,7 +1411,6 @@ void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
 			carry += (revsum < old);
 		}
 		{
-			unsigned long long p = pay[i];
 			int32_t date = (int32_t) (uint32_t) p;
 			int32_t prio = (int32_t) (uint32_t) (p >> 32);
 
@@ -1423,23 +1420,18 @@ void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
 	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	int64_t i = i0;
 
-	for (; i + 3 * stride < dense_len; i += 4 * stride)
+	for (; i + stride < dense_len; i += 2 * stride)
 	{
-		unsigned long long r0 = __builtin_nontemporal_load(&rev[i]);
-		unsigned long long r1 =
-			__builtin_nontemporal_load(&rev[i + stride]);
-		unsigned long long r2 =
-			__builtin_nontemporal_load(&rev[i + 2 * stride]);
-		unsigned long long r3 =
-			__builtin_nontemporal_load(&rev[i + 3 * stride]);
+		gg_ull2 a = __builtin_nontemporal_load(
+			(const gg_ull2 *) &pr[2 * i]);
+		gg_ull2 b = __builtin_nontemporal_load(
+			(const gg_ull2 *) &pr[2 * (i + stride)]);
 
-		body(i, r0);
-		body(i + stride, r1);
-		body(i + 2 * stride, r2);
-		body(i + 3 * stride, r3);
+		body(i, a.x, a.y);
+		body(i + stride, b.x, b.y);
 	}
 	for (; i < dense_len; i += stride)
-		body(i, __builtin_nontemporal_load(&rev[i]));
+		body(i, pr[2 * i], pr[2 * i + 1]);
 	for (int off = 32; off; off >>= 1)
 	{
 		unsigned long long orev = revsum;
@@ -1466,17 +1458,16 @@ void k_dn_q3_stats(const unsigned long long *__restrict__ pay,
 }
 
 hipError_t
-launch_dn_q3_stats(hipStream_t s, const unsigned long long *pay,
-		   const unsigned long long *rev, int64_t dense_len,
-		   unsigned long long *out5)
+launch_dn_q3_stats(hipStream_t s, const unsigned long long *pr,
+		   int64_t dense_len, unsigned long long *out5)
 {
 	hipLaunchKernelGGL(k_dn_q3_stats, dim3(dn_grid(dense_len)),
-			   dim3(DN_THREADS), 0, s, pay, rev, dense_len, out5);
+			   dim3(DN_THREADS), 0, s, pr, dense_len, out5);
 	return hipGetLastError();
 }
 
 __global__ void
-k_dn_q3_hist(const unsigned long long *__restrict__ rev, int64_t dense_len,
+k_dn_q3_hist(const unsigned long long *__restrict__ pr, int64_t dense_len,
 	     const unsigned long long *__restrict__ stats5,
 	     unsigned int *__restrict__ hist64k)
 {
@@ -1489,7 +1480,7 @@ k_dn_q3_hist(const unsigned long long *__restrict__ rev, int64_t dense_len,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < dense_len; i += stride)
 	{
-		unsigned long long r = rev[i];
+		unsigned long long r = pr[2 * i];
 
 		if (!r)
 			continue;
@@ -1504,19 +1495,18 @@ k_dn_q3_hist(const unsigned long long *__restrict__ rev, int64_t dense_len,
 }
 
 hipError_t
-launch_dn_q3_hist(hipStream_t s, const unsigned long long *rev,
+launch_dn_q3_hist(hipStream_t s, const unsigned long long *pr,
 		  int64_t dense_len, const unsigned long long *stats5,
 		  unsigned int *hist64k)
 {
 	hipLaunchKernelGGL(k_dn_q3_hist, dim3(dn_grid(dense_len)),
-			   dim3(DN_THREADS), 0, s, rev, dense_len, stats5,
+			   dim3(DN_THREADS), 0, s, pr, dense_len, stats5,
 			   hist64k);
 	return hipGetLastError();
 }
 
 __global__ void
-k_dn_q3_collect(const unsigned long long *__restrict__ pay,
-		const unsigned long long *__restrict__ rev,
+k_dn_q3_collect(const unsigned long long *__restrict__ pr,
 		int64_t dense_len,
 		const unsigned long long *__restrict__ thr_ptr,
 		gg_q3_result_row *__restrict__ out,
@@ -1528,14 +1518,14 @@ k_dn_q3_collect(const unsigned long long *__restrict__ pay,
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < dense_len; i += stride)
 	{
-		unsigned long long r = rev[i];
+		unsigned long long r = pr[2 * i];
 		bool take = (r != 0 && r >= threshold);
 		unsigned long long idx = dn_wave_append(out_count, take);
 
 		if (!take || idx >= cap)
 			continue;
 		{
-			unsigned long long p = pay[i];
+			unsigned long long p = pr[2 * i + 1];
 
 			out[idx].orderkey = i;
 			out[idx].rev_lo = r;
@@ -1547,13 +1537,13 @@ k_dn_q3_collect(const unsigned long long *__restrict__ pay,
 }
 
 hipError_t
-launch_dn_q3_collect(hipStream_t s, const unsigned long long *pay,
-		     const unsigned long long *rev, int64_t dense_len,
+launch_dn_q3_collect(hipStream_t s, const unsigned long long *pr,
+		     int64_t dense_len,
 		     const unsigned long long *thr_ptr, gg_q3_result_row *out,
 		     unsigned long long *out_count, uint64_t cap)
 {
 	hipLaunchKernelGGL(k_dn_q3_collect, dim3(dn_grid(dense_len)),
-			   dim3(DN_THREADS), 0, s, pay, rev, dense_len,
+			   dim3(DN_THREADS), 0, s, pr, dense_len,
 			   thr_ptr, out, out_count, cap);
 	return hipGetLastError();
 }
