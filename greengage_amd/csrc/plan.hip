@@ -409,17 +409,85 @@ void k_plan_scan_agg(PlanDev P)
 		return;
 	}
 
+	/* Per-block LDS group cache: low-cardinality group-bys (Q1's 6
+	 * groups, nation's 25) otherwise hammer a handful of global
+	 * accumulator words with per-row atomics — the ~88 atomics/µs
+	 * hot-word wall.  64-slot LDS open addressing absorbs the per-row
+	 * transitions; overflow rows (cardinality > ~64 per block) fall
+	 * back to the global table, and the LDS slots flush once per
+	 * block (execHHashagg.c:456 find-or-create, two-level). */
+	constexpr int LSLOTS = 64;
+	constexpr long long LEMPTY = (long long) 0x8000000000000000ull;
+	__shared__ long long lkeys[LSLOTS];
+	__shared__ unsigned long long lvals[LSLOTS][2 * GG_PLAN_MAX_AGGS];
+
+	for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x)
+	{
+		lkeys[s] = LEMPTY;
+		for (int a = 0; a < 2 * GG_PLAN_MAX_AGGS; a++)
+			lvals[s][a] = 0;
+	}
+	__syncthreads();
+
+	auto global_row = [&](int64_t i, long long code)
+	{
+		int64_t slot = pl_slot(P, code);
+
+		if (slot < 0)
+		{
+			atomicOr(P.err, 1ull);
+			return;
+		}
+		for (int a = 0; a < P.naggs; a++)
+		{
+			__int128 v;
+
+			if (!pl_agg_val(P.aggs[a], i, &v))
+				continue;
+			pl_atomic_add128(
+				&P.tvals[(slot * P.naggs + a) * 2],
+				&P.tvals[(slot * P.naggs + a) * 2 + 1], v);
+		}
+	};
 	auto grouped_row = [&](int64_t i)
 	{
 		if (!pl_joins_pass(P.joins, P.njoins, i))
 			return;
 		{
 			long long code = pl_group_code(P, i);
-			int64_t slot = pl_slot(P, code);
+			int ls = -1;
+			uint32_t pos = (uint32_t) gg_hashint8(code) &
+				(LSLOTS - 1);
 
-			if (slot < 0)
+			for (int probe = 0; probe < 8; probe++)
 			{
-				atomicOr(P.err, 1ull);
+				long long cur = lkeys[pos];
+
+				if (cur == code)
+				{
+					ls = (int) pos;
+					break;
+				}
+				if (cur == LEMPTY)
+				{
+					long long prev = atomicCAS(
+						(unsigned long long *)
+						&lkeys[pos],
+						(unsigned long long) LEMPTY,
+						(unsigned long long) code);
+
+					if (prev == LEMPTY || prev == code)
+					{
+						ls = (int) pos;
+						break;
+					}
+					continue;
+				}
+				pos = (pos + 1) & (LSLOTS - 1);
+			}
+			if (ls < 0)
+			{
+				global_row(i, code);	/* cache full */
 				return;
 			}
 			for (int a = 0; a < P.naggs; a++)
@@ -428,10 +496,20 @@ void k_plan_scan_agg(PlanDev P)
 
 				if (!pl_agg_val(P.aggs[a], i, &v))
 					continue;
-				pl_atomic_add128(
-					&P.tvals[(slot * P.naggs + a) * 2],
-					&P.tvals[(slot * P.naggs + a) * 2 + 1],
-					v);
+				{
+					unsigned long long vlo =
+						(unsigned long long) v;
+					unsigned long long vhi =
+						(unsigned long long) (v >> 64);
+					unsigned long long old = atomicAdd(
+						&lvals[ls][2 * a], vlo);
+
+					if (old + vlo < old)
+						vhi++;
+					if (vhi)
+						atomicAdd(&lvals[ls][2 * a + 1],
+							  vhi);
+				}
 			}
 		}
 	};
@@ -449,6 +527,44 @@ void k_plan_scan_agg(PlanDev P)
 	for (; i < P.n; i += stride)
 		if (pl_preds_pass(P.preds, P.npreds, i))
 			grouped_row(i);
+
+	/* flush the block's LDS groups into the global table */
+	__syncthreads();
+	for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x)
+	{
+		if (lkeys[s] == LEMPTY)
+			continue;
+		{
+			int64_t slot = pl_slot(P, lkeys[s]);
+
+			if (slot < 0)
+			{
+				atomicOr(P.err, 1ull);
+				continue;
+			}
+			for (int a = 0; a < P.naggs; a++)
+			{
+				unsigned long long lo = lvals[s][2 * a];
+				unsigned long long hi = lvals[s][2 * a + 1];
+
+				if (!lo && !hi)
+					continue;
+				{
+					unsigned long long old = atomicAdd(
+						&P.tvals[(slot * P.naggs + a)
+							 * 2], lo);
+
+					if (old + lo < old)
+						hi++;
+					if (hi)
+						atomicAdd(
+							&P.tvals[(slot *
+								  P.naggs + a)
+								 * 2 + 1], hi);
+				}
+			}
+		}
+	}
 }
 
 hipError_t launch_plan_scan_agg(hipStream_t s, const PlanDev &p)
