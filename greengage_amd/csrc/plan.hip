@@ -416,18 +416,23 @@ void k_plan_scan_agg(PlanDev P)
 	 * transitions; overflow rows (cardinality > ~64 per block) fall
 	 * back to the global table, and the LDS slots flush once per
 	 * block (execHHashagg.c:456 find-or-create, two-level). */
-	constexpr int LSLOTS = 64;
+	constexpr int LSLOTS = 32;
+	constexpr int LREPL = 8;	/* replicas split the per-word LDS
+					 * atomic serialization 8 ways (the
+					 * k_q1_agg replica pattern) */
 	constexpr long long LEMPTY = (long long) 0x8000000000000000ull;
 	__shared__ long long lkeys[LSLOTS];
-	__shared__ unsigned long long lvals[LSLOTS][2 * GG_PLAN_MAX_AGGS];
+	__shared__ unsigned long long
+		lvals[LREPL][LSLOTS][2 * GG_PLAN_MAX_AGGS];
 
 	for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x)
-	{
 		lkeys[s] = LEMPTY;
-		for (int a = 0; a < 2 * GG_PLAN_MAX_AGGS; a++)
-			lvals[s][a] = 0;
-	}
+	for (int s = threadIdx.x;
+	     s < LREPL * LSLOTS * 2 * GG_PLAN_MAX_AGGS; s += blockDim.x)
+		((unsigned long long *) lvals)[s] = 0;
 	__syncthreads();
+
+	const int lrep = (int) (threadIdx.x & (LREPL - 1));
 
 	auto global_row = [&](int64_t i, long long code)
 	{
@@ -502,13 +507,13 @@ void k_plan_scan_agg(PlanDev P)
 					unsigned long long vhi =
 						(unsigned long long) (v >> 64);
 					unsigned long long old = atomicAdd(
-						&lvals[ls][2 * a], vlo);
+						&lvals[lrep][ls][2 * a], vlo);
 
 					if (old + vlo < old)
 						vhi++;
 					if (vhi)
-						atomicAdd(&lvals[ls][2 * a + 1],
-							  vhi);
+						atomicAdd(&lvals[lrep][ls]
+							  [2 * a + 1], vhi);
 				}
 			}
 		}
@@ -544,8 +549,18 @@ void k_plan_scan_agg(PlanDev P)
 			}
 			for (int a = 0; a < P.naggs; a++)
 			{
-				unsigned long long lo = lvals[s][2 * a];
-				unsigned long long hi = lvals[s][2 * a + 1];
+				unsigned long long lo = 0, hi = 0;
+
+				for (int rr = 0; rr < LREPL; rr++)
+				{
+					unsigned long long rl =
+						lvals[rr][s][2 * a];
+					unsigned long long o = lo;
+
+					lo += rl;
+					hi += lvals[rr][s][2 * a + 1] +
+						(lo < o);
+				}
 
 				if (!lo && !hi)
 					continue;
