@@ -60,3 +60,27 @@ print(f"q1-shaped grouped plan_scan_agg: {d*1000:.0f} us/launch "
       f"({600e6*38/(d/1e3)/1e12:.2f} TB/s algorithmic 38 B/row); "
       f"wall {wall*1e3:.2f} ms/exec")
 print("PLAN_BENCH2_OK")
+
+# NT-load variant of the generic kernel (GG_PLAN_NT=1)
+os.environ["GG_PLAN_NT"] = "1"
+gnt = eng.execute_plan(p, max_groups=8)
+assert gnt == g0
+before = {s["name"]: dict(s) for s in eng.stats(p)}
+t0 = time.perf_counter()
+for _ in range(10):
+    gnt = eng.execute_plan(p, max_groups=8)
+after = {s["name"]: s for s in eng.stats(p)}
+d = ((after["plan_scan_agg"]["total_ms"] - before["plan_scan_agg"]["total_ms"])
+     / (after["plan_scan_agg"]["launches"] - before["plan_scan_agg"]["launches"]))
+print(f"q6 NT plan_scan_agg: {d*1000:.0f} us/launch")
+gq = eng.execute_plan(pq1, max_groups=16)
+assert gq == gq0
+before = {s["name"]: dict(s) for s in eng.stats(pq1)}
+for _ in range(5):
+    gq = eng.execute_plan(pq1, max_groups=16)
+after = {s["name"]: s for s in eng.stats(pq1)}
+d = ((after["plan_scan_agg"]["total_ms"] - before["plan_scan_agg"]["total_ms"])
+     / (after["plan_scan_agg"]["launches"] - before["plan_scan_agg"]["launches"]))
+print(f"q1-shaped NT grouped: {d*1000:.0f} us/launch")
+os.environ.pop("GG_PLAN_NT")
+print("PLAN_BENCH3_OK")
