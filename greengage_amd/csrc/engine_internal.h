@@ -525,6 +525,12 @@ hipError_t launch_plan_compact(hipStream_t s,
 gg_status exec_plan(Pipeline *p, void *arena, size_t bytes,
 		    size_t *written);
 
+/* plan_rtc.cpp — per-plan kernel specialization via hipRTC */
+gg_status plan_rtc_compile(const PlanDev &D, bool has_gnull0,
+			   bool has_gnull1, std::shared_ptr<void> *out);
+gg_status plan_rtc_launch(hipStream_t s, const std::shared_ptr<void> &h,
+			  PlanDev P, int grid, int block);
+
 /* engine_abi.cpp helpers shared with plan.cpp */
 Table *engine_table(gg_table h);
 gg_status engine_cached_max_i64(Engine &e, Pipeline *p, const int64_t *col,

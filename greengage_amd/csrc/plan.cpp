@@ -45,6 +45,11 @@ struct PlanResolved
 	std::vector<BJoin> joins;
 	int64_t scan_rows = 0;
 	int64_t pred_bytes_per_row = 0;
+	/* hipRTC-specialized kernel (nullptr = interpreted fallback);
+	 * compiled lazily on first execute, when the join structures'
+	 * shape (bitmap vs hash set) is known */
+	std::shared_ptr<void> rtc;
+	bool rtc_tried = false;
 };
 
 static int coltype_width(gg_coltype t)
@@ -334,7 +339,31 @@ exec_plan(Pipeline *p, void *arena, size_t bytes, size_t *written)
 		R->dev.tvals = tvals;
 		R->dev.nslots = nslots;
 		R->dev.err = derr;
-		GG_HIP(launch_plan_scan_agg(e.stream, R->dev));
+		if (!R->rtc_tried)
+		{
+			/* specialize now: bitmap-vs-hash join shape is
+			 * resolved; NULL presence and widths are baked */
+			gg_status rs = plan_rtc_compile(
+				R->dev, R->dev.gnulls[0] != nullptr,
+				R->dev.gnulls[1] != nullptr, &R->rtc);
+
+			R->rtc_tried = true;
+			p->stat(rs == GG_OK ? "path_plan_rtc"
+				: "path_plan_interp").launches++;
+		}
+		if (R->rtc)
+		{
+			int64_t g = (R->dev.n + 255) / 256;
+
+			if (g > 2048)
+				g = 2048;
+			if (g < 1)
+				g = 1;
+			GG_TRY(plan_rtc_launch(e.stream, R->rtc, R->dev,
+					       (int) g, 256));
+		}
+		else
+			GG_HIP(launch_plan_scan_agg(e.stream, R->dev));
 		{
 			double ms = tm.stop();
 			KernelStatAcc &st = p->stat("plan_scan_agg");

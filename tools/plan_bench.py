@@ -84,3 +84,36 @@ d = ((after["plan_scan_agg"]["total_ms"] - before["plan_scan_agg"]["total_ms"])
 print(f"q1-shaped NT grouped: {d*1000:.0f} us/launch")
 os.environ.pop("GG_PLAN_NT")
 print("PLAN_BENCH3_OK")
+
+# interpreted-vs-RTC comparison (fresh pipelines; RTC decision is per
+# pipeline at first execute)
+os.environ["GG_PLAN_RTC"] = "0"
+p_i = eng.compile_plan(
+    li, preds=[("shipdate", lo, hi), ("disc", 5, 8), ("qty", NEG_INF, 2400)],
+    aggs=[("sum", [("price", "id"), ("disc", "id")]), "count"])
+gi = eng.execute_plan(p_i, max_groups=8)
+assert gi == g0, "interp/rtc parity"
+pq_i = eng.compile_plan(
+    li, preds=[("shipdate", NEG_INF, cutoff + 1)],
+    group_cols=["rflag", "lstatus"],
+    aggs=["count", ("sum", [("qty", "id")]), ("sum", [("price", "id")]),
+          ("sum", [("disc", "id")]),
+          ("sum", [("price", "id"), ("disc", "sub100")]),
+          ("sum", [("price", "id"), ("disc", "sub100"), ("tax", "add100")])])
+gqi = eng.execute_plan(pq_i, max_groups=16)
+assert gqi == gq0, "interp/rtc grouped parity"
+os.environ.pop("GG_PLAN_RTC")
+for name, pp in (("q6", p_i), ("q1-shaped", pq_i)):
+    before = {s["name"]: dict(s) for s in eng.stats(pp)}
+    for _ in range(5):
+        eng.execute_plan(pp, max_groups=16)
+    after = {s["name"]: s for s in eng.stats(pp)}
+    d = ((after["plan_scan_agg"]["total_ms"]
+          - before["plan_scan_agg"]["total_ms"])
+         / (after["plan_scan_agg"]["launches"]
+            - before["plan_scan_agg"]["launches"]))
+    print(f"{name} interp plan_scan_agg: {d*1000:.0f} us/launch")
+for pp, name in ((p, "q6"), (pq1, "q1-shaped")):
+    paths = [s["name"] for s in eng.stats(pp) if s["name"].startswith("path")]
+    print(f"{name} default paths: {paths}")
+print("PLAN_BENCH4_OK")
