@@ -102,9 +102,11 @@ def test_q6_sf1_synth(eng):
     assert groups[0][2][0] == rev
     assert groups[0][2][1] == cnt
     # no query-specific kernels ran: only the generic plan stats rows
+    # (path_plan_rtc = the hipRTC-specialized form of the SAME plan)
     names = {s["name"] for s in eng.stats(p)}
-    assert names <= {"plan_build", "plan_scan_agg",
-                     "path_plan_join_bitmap", "path_plan_join_hash"}, names
+    assert names <= {"plan_build", "plan_scan_agg", "path_plan_rtc",
+                     "path_plan_interp", "path_plan_join_bitmap",
+                     "path_plan_join_hash"}, names
 
 
 def test_q1_shaped_descriptor_vs_oracle(eng):
