@@ -151,11 +151,20 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 	std::string s = STRUCT_DEFS;
 	char buf[512];
 
+	int lrepl = 8;
+	const char *lr = getenv("GG_PLAN_LREPL");
+
+	if (lr && atoi(lr) >= 1 && atoi(lr) <= 32)
+		lrepl = atoi(lr);
+	/* LDS budget: LREPL x 32 slots x 2*naggs u64 must stay within
+	 * the per-block share; shrink replicas for wide agg lists */
+	while (lrepl > 1 && lrepl * 32 * 2 * D.naggs * 8 > 64 * 1024)
+		lrepl /= 2;
 	std::snprintf(buf, sizeof(buf),
 		      "static_assert(sizeof(PlanDev) == %zu, \"layout\");\n"
-		      "#define LSLOTS 32\n#define LREPL 8\n"
+		      "#define LSLOTS 32\n#define LREPL %d\n"
 		      "#define LEMPTY ((long long) 0x8000000000000000ull)\n",
-		      sizeof(PlanDev));
+		      sizeof(PlanDev), lrepl);
 	s += buf;
 
 	s += "extern \"C\" __global__ "

@@ -117,3 +117,28 @@ for pp, name in ((p, "q6"), (pq1, "q1-shaped")):
     paths = [s["name"] for s in eng.stats(pp) if s["name"].startswith("path")]
     print(f"{name} default paths: {paths}")
 print("PLAN_BENCH4_OK")
+
+# LDS replica sweep for the RTC grouped kernel
+for repl in (4, 8, 16):
+    os.environ["GG_PLAN_LREPL"] = str(repl)
+    pr = eng.compile_plan(
+        li, preds=[("shipdate", NEG_INF, cutoff + 1)],
+        group_cols=["rflag", "lstatus"],
+        aggs=["count", ("sum", [("qty", "id")]), ("sum", [("price", "id")]),
+              ("sum", [("disc", "id")]),
+              ("sum", [("price", "id"), ("disc", "sub100")]),
+              ("sum", [("price", "id"), ("disc", "sub100"),
+                       ("tax", "add100")])])
+    gr = eng.execute_plan(pr, max_groups=16)
+    assert gr == gq0
+    before = {s["name"]: dict(s) for s in eng.stats(pr)}
+    for _ in range(5):
+        eng.execute_plan(pr, max_groups=16)
+    after = {s["name"]: s for s in eng.stats(pr)}
+    d = ((after["plan_scan_agg"]["total_ms"]
+          - before["plan_scan_agg"]["total_ms"])
+         / (after["plan_scan_agg"]["launches"]
+            - before["plan_scan_agg"]["launches"]))
+    print(f"grouped RTC LREPL={repl}: {d*1000:.0f} us/launch")
+os.environ.pop("GG_PLAN_LREPL")
+print("PLAN_BENCH5_OK")
