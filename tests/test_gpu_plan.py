@@ -308,3 +308,74 @@ def test_ao_mounted_nullable_column(eng):
     with pytest.raises(EngineError):
         eng.register_table_ao("ao_notnull", [
             ("v", "int64", ao, 1, 2, 2, 0)])
+
+
+def test_rtc_and_interpreted_paths_agree(eng):
+    """The hipRTC-specialized kernel and the interpreted generic
+    kernel are the same plan in two forms: identical results, path
+    reported in stats (GG_PLAN_RTC=0 forces the fallback)."""
+    rng = np.random.default_rng(41)
+    n = 300_000
+    k = rng.integers(0, 30, n).astype(np.int64)
+    a = rng.integers(-100, 100, n).astype(np.int64)
+    b = rng.integers(0, 50, n).astype(np.int64)
+    t = eng.register_table("rtc_cmp", [("k", "int64", k),
+                                       ("a", "dec64", a),
+                                       ("b", "dec64", b)], n)
+
+    def run(env):
+        if env is None:
+            os.environ.pop("GG_PLAN_RTC", None)
+        else:
+            os.environ["GG_PLAN_RTC"] = env
+        p = eng.compile_plan(
+            t, preds=[("b", 5, 45)], group_cols=["k"],
+            aggs=["count", ("sum", [("a", "id"), ("b", "sub100")])])
+        g = eng.execute_plan(p)
+        paths = {s["name"] for s in eng.stats(p)
+                 if s["name"].startswith("path_plan_")}
+        return g, paths
+
+    g_rtc, p_rtc = run(None)
+    g_int, p_int = run("0")
+    os.environ.pop("GG_PLAN_RTC", None)
+    assert g_rtc == g_int
+    assert "path_plan_interp" in p_int
+    # RTC may legitimately fall back on exotic boxes; if it compiled,
+    # the row says so
+    assert p_rtc & {"path_plan_rtc", "path_plan_interp"}
+
+    m = (b >= 5) & (b < 45)
+    exp = {}
+    for kk in np.unique(k[m]):
+        sel = m & (k == kk)
+        exp[int(kk)] = [int(np.count_nonzero(sel)),
+                        sum(int(x) * (100 - int(y)) for x, y in
+                            zip(a[sel].tolist(), b[sel].tolist()))]
+    assert len(g_rtc) == len(exp)
+    for k0, _k1, vals in g_rtc:
+        assert vals == exp[k0]
+
+
+def test_ao_nullable_int32_dsb_v1(eng):
+    """Nullable AO mount, int32 column, Dense (v1) blocks with RLE."""
+    if pyoracle.dsb_ref() is None:
+        pytest.skip("reference dsb codec not built")
+    rng = np.random.default_rng(29)
+    n = 30_000
+    vals = rng.integers(-1000, 1000, n).astype(np.int64)
+    vals[rng.random(n) < 0.3] = 77          # RLE-friendly runs
+    nulls = (rng.random(n) < 0.2).astype(np.uint8)
+    framed, _nb = pyoracle.dsb_encode(vals, nulls, 4, 1, 1, 0)
+    ao = pyoracle.ao_wrap(framed)
+    t = eng.register_table_ao("ao_null_i32", [
+        ("v", "int32", ao, 1, 2, 1, 0, 0, 1)])
+    p = eng.compile_plan(t, preds=[("v", 0, 1 << 31)],
+                         aggs=["count", ("count", "v"),
+                               ("sum", [("v", "id")])])
+    g = eng.execute_plan(p, max_groups=8)
+    m = (nulls == 0) & (vals >= 0)
+    # NULL fails the qual (execScan.c:185), so COUNT(*) == COUNT(v) here
+    assert g[0][2][0] == int(np.count_nonzero(m))
+    assert g[0][2][1] == int(np.count_nonzero(m))
+    assert g[0][2][2] == int(vals[m].sum())
