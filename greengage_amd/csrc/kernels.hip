@@ -506,6 +506,36 @@ wave_append(unsigned long long *ctr, bool take)
 						    ((1ull << lane) - 1));
 }
 
+
+/* Block-claimed compaction: pass 1 counts the block's survivors
+ * (registers + one wave/LDS reduction), thread 0 claims ONE
+ * contiguous region from the global counter, pass 2 appends through
+ * an LDS cursor.  Even the wave-aggregated global append serializes
+ * at the ~88 returning-atomics/µs hot-word wall (~10 ms per 57M-row
+ * compaction); this makes it 2048 global atomics total.  Output
+ * order within the region is block-local — all consumers on this
+ * path are order-insensitive (aggregates / unique-key stores). */
+__device__ inline unsigned long long
+block_claim(unsigned long long *lds4, unsigned long long *bbase,
+	    unsigned long long my, unsigned long long *global_ctr)
+{
+	for (int off = 32; off; off >>= 1)
+		my += __shfl_down(my, off, 64);
+	if ((threadIdx.x & 63) == 0)
+		lds4[threadIdx.x >> 6] = my;
+	__syncthreads();
+	if (threadIdx.x == 0)
+	{
+		unsigned long long t = 0;
+
+		for (int i = 0; i < (int) (blockDim.x >> 6); i++)
+			t += lds4[i];
+		*bbase = t ? atomicAdd(global_ctr, t) : 0;
+	}
+	__syncthreads();
+	return *bbase;
+}
+
 /* ------------------------------------------------------------------ */
 /* hash join build/probe (open addressing, key 0 = empty)              */
 /* ------------------------------------------------------------------ */
@@ -1093,18 +1123,33 @@ __global__ void
 k_part_count(const int64_t *__restrict__ key, int64_t n, int nseg,
 	     unsigned long long *__restrict__ counts)
 {
+	/* per-row global atomics on <= nseg words serialize at the
+	 * ~88 atomics/µs hot-word wall (650 ms measured at 57M rows,
+	 * nseg=1): count in LDS, flush once per block per dest */
+	__shared__ unsigned long long lcnt[64];
+
+	for (int d = threadIdx.x; d < nseg; d += blockDim.x)
+		lcnt[d] = 0;
+	__syncthreads();
+
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
-		atomicAdd(&counts[gg_cdbhash_segment_int8(key[i], nseg)],
+		atomicAdd(&lcnt[gg_cdbhash_segment_int8(key[i], nseg)],
 			  1ull);
+	__syncthreads();
+	for (int d = threadIdx.x; d < nseg; d += blockDim.x)
+		if (lcnt[d])
+			atomicAdd(&counts[d], lcnt[d]);
 }
 
 hipError_t
 launch_part_count(hipStream_t s, const int64_t *key, int64_t n, int nseg,
 		  unsigned long long *counts)
 {
+	if (nseg > 64)
+		return hipErrorInvalidValue;	/* LDS count array bound */
 	hipLaunchKernelGGL(k_part_count, dim3(grid_for(n)), dim3(THREADS), 0,
 			   s, key, n, nseg, counts);
 	return hipGetLastError();
@@ -1118,13 +1163,34 @@ k_part_scatter3(const int64_t *__restrict__ key, int64_t n, int nseg,
 		int64_t *__restrict__ oa, int64_t *__restrict__ ob,
 		int64_t *__restrict__ oc)
 {
+	/* two passes: count the block's rows per dest in LDS, claim one
+	 * contiguous region per (block, dest) with a single global
+	 * atomic each, then append through LDS cursors — per-row
+	 * returning atomics on nseg hot words measured 650 ms at 57M
+	 * rows (the ~88 atomics/µs wall) */
+	__shared__ unsigned long long lcnt[64], lbase[64];
+
+	for (int d = threadIdx.x; d < nseg; d += blockDim.x)
+		lcnt[d] = 0;
+	__syncthreads();
+
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
+		atomicAdd(&lcnt[gg_cdbhash_segment_int8(key[i], nseg)], 1ull);
+	__syncthreads();
+	for (int d = threadIdx.x; d < nseg; d += blockDim.x)
+	{
+		lbase[d] = lcnt[d] ? atomicAdd(&offsets[d], lcnt[d]) : 0;
+		lcnt[d] = 0;
+	}
+	__syncthreads();
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
 	{
 		int d = gg_cdbhash_segment_int8(key[i], nseg);
-		unsigned long long idx = atomicAdd(&offsets[d], 1ull);
+		unsigned long long idx = lbase[d] + atomicAdd(&lcnt[d], 1ull);
 
 		oa[idx] = a[i];
 		if (b)
@@ -1140,6 +1206,8 @@ launch_part_scatter3(hipStream_t s, const int64_t *key, int64_t n, int nseg,
 		     unsigned long long *offsets, int64_t *oa, int64_t *ob,
 		     int64_t *oc)
 {
+	if (nseg > 64)
+		return hipErrorInvalidValue;	/* LDS cursor array bound */
 	hipLaunchKernelGGL(k_part_scatter3, dim3(grid_for(n)), dim3(THREADS),
 			   0, s, key, n, nseg, a, b, c, offsets, oa, ob, oc);
 	return hipGetLastError();
@@ -1161,14 +1229,28 @@ k_orders_filter_compact(const int64_t *__restrict__ okey,
 			int64_t *__restrict__ out_pay,
 			unsigned long long *out_count)
 {
+	__shared__ unsigned long long lds4[4];
+	__shared__ unsigned long long bbase, bcur;
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long my = 0;
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
 		int32_t d = nt_ld32(&odate[i]);
+
+		my += (d >= date_lo && d < date_hi);
+	}
+	block_claim(lds4, &bbase, my, out_count);
+	if (threadIdx.x == 0)
+		bcur = 0;
+	__syncthreads();
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int32_t d = nt_ld32(&odate[i]);
 		bool take = (d >= date_lo && d < date_hi);
-		unsigned long long idx = wave_append(out_count, take);
+		unsigned long long idx = bbase + wave_append(&bcur, take);
 
 		if (!take)
 			continue;
@@ -1212,6 +1294,25 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
+	__shared__ unsigned long long lds4[4];
+	__shared__ unsigned long long bbase, bcur;
+	unsigned long long my = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t ck = ckey[i];
+
+		my += cust_bits
+			? (ck >= 0 && ck < cust_dlen &&
+			   ((cust_bits[ck >> 6] >> (ck & 63)) & 1))
+			: ht_contains_b(cust_keys, cust_slots, cust_bloom,
+					cust_bwords, ck);
+	}
+	block_claim(lds4, &bbase, my, out_count);
+	if (threadIdx.x == 0)
+		bcur = 0;
+	__syncthreads();
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
@@ -1221,7 +1322,7 @@ k_probe_cust_compact(const int64_t *__restrict__ ckey,
 			   ((cust_bits[ck >> 6] >> (ck & 63)) & 1))
 			: ht_contains_b(cust_keys, cust_slots, cust_bloom,
 					cust_bwords, ck);
-		unsigned long long idx = wave_append(out_count, take);
+		unsigned long long idx = bbase + wave_append(&bcur, take);
 
 		if (!take)
 			continue;
@@ -1442,13 +1543,27 @@ k_supp_filter_compact(const int64_t *__restrict__ suppkey,
 		      unsigned long long *out_count)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	__shared__ unsigned long long lds4[4];
+	__shared__ unsigned long long bbase, bcur;
+	unsigned long long my = 0;
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
 		uint8_t sn = snation[i];
+
+		my += (sn < 25 && region_of[sn] == regionkey);
+	}
+	block_claim(lds4, &bbase, my, out_count);
+	if (threadIdx.x == 0)
+		bcur = 0;
+	__syncthreads();
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		uint8_t sn = snation[i];
 		bool take = (sn < 25 && region_of[sn] == regionkey);
-		unsigned long long idx = wave_append(out_count, take);
+		unsigned long long idx = bbase + wave_append(&bcur, take);
 
 		if (!take)
 			continue;
@@ -1639,26 +1754,43 @@ k_probe_cust_map_compact(const int64_t *__restrict__ ckey,
 			 unsigned long long *out_count)
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	__shared__ unsigned long long lds4[4];
+	__shared__ unsigned long long bbase, bcur;
+	unsigned long long my = 0;
+
+	auto look = [&](int64_t i, unsigned long long *nat) -> bool
+	{
+		if (cust_dense)
+		{
+			int64_t ck = ckey[i];
+			bool t = (ck >= 0 && ck < cust_dlen &&
+				  cust_dense[ck] != 255);
+
+			if (t)
+				*nat = cust_dense[ck];
+			return t;
+		}
+		return ht_lookup_b(cust_keys, cust_pay, cust_slots, nullptr,
+				   0, ckey[i], nat);
+	};
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
 	{
 		unsigned long long nat = 0;
-		bool take;
 
-		if (cust_dense)
-		{
-			int64_t ck = ckey[i];
-
-			take = (ck >= 0 && ck < cust_dlen &&
-				cust_dense[ck] != 255);
-			if (take)
-				nat = cust_dense[ck];
-		}
-		else
-			take = ht_lookup_b(cust_keys, cust_pay, cust_slots,
-					   nullptr, 0, ckey[i], &nat);
-		unsigned long long idx = wave_append(out_count, take);
+		my += look(i, &nat);
+	}
+	block_claim(lds4, &bbase, my, out_count);
+	if (threadIdx.x == 0)
+		bcur = 0;
+	__syncthreads();
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		unsigned long long nat = 0;
+		bool take = look(i, &nat);
+		unsigned long long idx = bbase + wave_append(&bcur, take);
 
 		if (!take)
 			continue;
