@@ -2059,41 +2059,62 @@ k_sj_probe(const int64_t *__restrict__ keys,
 {
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 	const int64_t i0 = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
-	const int64_t n_up = ((n + stride - 1) / stride) * stride;
+	__shared__ unsigned long long lds4[4];
+	__shared__ unsigned long long bbase, bcur;
+	unsigned long long my = 0;
 
-	for (int64_t i = i0; i < n_up; i += stride)
+	auto probe = [&](int64_t i, unsigned long long *v) -> bool
 	{
-		bool hit = false;
-		unsigned long long v = 0;
-		int64_t kk = 0;
+		int64_t kk = keys[i];
+		uint64_t pos = (uint64_t) gg_hashint8(kk) & (nslots - 1);
 
-		if (i < n)
+		for (;;)
 		{
-			kk = keys[i];
-			uint64_t pos = (uint64_t) gg_hashint8(kk) &
-				(nslots - 1);
+			unsigned long long cur = tkeys[pos];
 
-			for (;;)
+			if (cur == (unsigned long long) kk)
 			{
-				unsigned long long cur = tkeys[pos];
-
-				if (cur == (unsigned long long) kk)
-				{
-					v = tvals[pos];
-					hit = true;
-					break;
-				}
-				if (cur == 0)
-					break;
-				pos = (pos + 1) & (nslots - 1);
+				*v = tvals[pos];
+				return true;
 			}
+			if (cur == 0)
+				return false;
+			pos = (pos + 1) & (nslots - 1);
 		}
-		unsigned long long at = wave_append(out_count, hit);
+	};
 
-		if (hit)
+	/* block-claimed append (the per-wave global counter serializes
+	 * at the ~88 atomics/µs wall): count, claim once, write */
+	for (int64_t i = i0; i < n; i += stride)
+	{
+		unsigned long long v;
+
+		my += probe(i, &v);
+	}
+	block_claim(lds4, &bbase, my, out_count);
+	if (threadIdx.x == 0)
+		bcur = 0;
+	__syncthreads();
+	{
+		const int64_t n_up = ((n + stride - 1) / stride) * stride;
+
+		for (int64_t i = i0; i < n_up; i += stride)
 		{
-			out_idx[at] = idxs[i];
-			out_val[at] = (int64_t) v;
+			bool hit = false;
+			unsigned long long v = 0;
+
+			if (i < n)
+				hit = probe(i, &v);
+			{
+				unsigned long long at = bbase +
+					wave_append(&bcur, hit);
+
+				if (hit)
+				{
+					out_idx[at] = idxs[i];
+					out_val[at] = (int64_t) v;
+				}
+			}
 		}
 	}
 }
@@ -2144,32 +2165,91 @@ k_gb_part_scatter_idx(const int64_t *__restrict__ keys, int64_t n,
 	}
 }
 
-hipError_t
-launch_gb_part_scatter_idx(hipStream_t s, const int64_t *keys, int64_t n,
-			   int64_t base, int shift,
-			   unsigned long long *cursors, int64_t *out_k,
-			   int64_t *out_v)
-{
-	hipLaunchKernelGGL(k_gb_part_scatter_idx, dim3(grid_for(n)),
-			   dim3(THREADS), 0, s, keys, n, base, shift,
-			   cursors, out_k, out_v);
-	return hipGetLastError();
-}
 
 /* hash-range partitioning for the spill tier (execHHashagg.c:1350
  * spill_hash_table semantics): partition id = TOP bits of the key
  * hash, decoupled from the group table's slot index (low bits) */
+/* spill-tier partitions are <= 4096 but almost always <= 1024:
+ * the LDS-staged form covers that; the launchers fall back to the
+ * direct-atomic form above it (suffix _direct) otherwise */
+#define GB_LDS_PARTS 1024
+
+template <int LDS>
 __global__ void
-k_gb_part_count(const int64_t *__restrict__ keys, int64_t n, int shift,
-		unsigned long long *__restrict__ counts)
+k_gb_part_count_t(const int64_t *__restrict__ keys, int64_t n, int shift,
+		  unsigned long long *__restrict__ counts)
 {
+	__shared__ unsigned long long lcnt[LDS ? GB_LDS_PARTS : 1];
+	const int nparts = (int) ((uint64_t) 0xffffffffu >> shift) + 1;
+
+	if (LDS)
+	{
+		for (int d = threadIdx.x; d < nparts; d += blockDim.x)
+			lcnt[d] = 0;
+		__syncthreads();
+	}
+
 	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
 	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
 	     i < n; i += stride)
+	{
 		/* 64-bit shift: shift==32 (single partition) must yield 0 */
-		atomicAdd(&counts[(uint64_t) gg_hashint8(keys[i]) >> shift],
+		uint64_t p = (uint64_t) gg_hashint8(keys[i]) >> shift;
+
+		if (LDS)
+			atomicAdd(&lcnt[p], 1ull);
+		else
+			atomicAdd(&counts[p], 1ull);
+	}
+	if (LDS)
+	{
+		__syncthreads();
+		for (int d = threadIdx.x; d < nparts; d += blockDim.x)
+			if (lcnt[d])
+				atomicAdd(&counts[d], lcnt[d]);
+	}
+}
+
+/* block-claimed scatter: count per block in LDS, claim one region per
+ * (block, partition), append through LDS cursors (VAL==1: payload =
+ * vals[i]; VAL==0: payload = base + i, the index-carrying variant) */
+template <int VAL>
+__global__ void
+k_gb_part_scatter_t(const int64_t *__restrict__ keys,
+		    const int64_t *__restrict__ vals, int64_t n,
+		    int64_t base, int shift,
+		    unsigned long long *__restrict__ cursors,
+		    int64_t *__restrict__ out_k, int64_t *__restrict__ out_v)
+{
+	__shared__ unsigned long long lcnt[GB_LDS_PARTS];
+	__shared__ unsigned long long lbase[GB_LDS_PARTS];
+	const int nparts = (int) ((uint64_t) 0xffffffffu >> shift) + 1;
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int d = threadIdx.x; d < nparts; d += blockDim.x)
+		lcnt[d] = 0;
+	__syncthreads();
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+		atomicAdd(&lcnt[(uint64_t) gg_hashint8(keys[i]) >> shift],
 			  1ull);
+	__syncthreads();
+	for (int d = threadIdx.x; d < nparts; d += blockDim.x)
+	{
+		lbase[d] = lcnt[d] ? atomicAdd(&cursors[d], lcnt[d]) : 0;
+		lcnt[d] = 0;
+	}
+	__syncthreads();
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		uint64_t p = (uint64_t) gg_hashint8(keys[i]) >> shift;
+		unsigned long long at = lbase[p] + atomicAdd(&lcnt[p], 1ull);
+
+		out_k[at] = keys[i];
+		out_v[at] = VAL ? vals[i] : base + i;
+	}
 }
 
 __global__ void
@@ -2191,12 +2271,42 @@ k_gb_part_scatter(const int64_t *__restrict__ keys,
 	}
 }
 
+static inline int gb_nparts(int shift)
+{
+	return (int) ((uint64_t) 0xffffffffu >> shift) + 1;
+}
+
+hipError_t
+launch_gb_part_scatter_idx(hipStream_t s, const int64_t *keys, int64_t n,
+			   int64_t base, int shift,
+			   unsigned long long *cursors, int64_t *out_k,
+			   int64_t *out_v)
+{
+	if (gb_nparts(shift) <= GB_LDS_PARTS)
+		hipLaunchKernelGGL((k_gb_part_scatter_t<0>),
+				   dim3(grid_for(n)), dim3(THREADS), 0, s,
+				   keys, nullptr, n, base, shift, cursors,
+				   out_k, out_v);
+	else
+		hipLaunchKernelGGL(k_gb_part_scatter_idx, dim3(grid_for(n)),
+				   dim3(THREADS), 0, s, keys, n, base, shift,
+				   cursors, out_k, out_v);
+	return hipGetLastError();
+}
+
+
 hipError_t
 launch_gb_part_count(hipStream_t s, const int64_t *keys, int64_t n,
 		     int shift, unsigned long long *counts)
 {
-	hipLaunchKernelGGL(k_gb_part_count, dim3(grid_for(n)), dim3(THREADS),
-			   0, s, keys, n, shift, counts);
+	if (gb_nparts(shift) <= GB_LDS_PARTS)
+		hipLaunchKernelGGL((k_gb_part_count_t<1>), dim3(grid_for(n)),
+				   dim3(THREADS), 0, s, keys, n, shift,
+				   counts);
+	else
+		hipLaunchKernelGGL((k_gb_part_count_t<0>), dim3(grid_for(n)),
+				   dim3(THREADS), 0, s, keys, n, shift,
+				   counts);
 	return hipGetLastError();
 }
 
@@ -2206,9 +2316,15 @@ launch_gb_part_scatter(hipStream_t s, const int64_t *keys,
 		       unsigned long long *cursors, int64_t *out_k,
 		       int64_t *out_v)
 {
-	hipLaunchKernelGGL(k_gb_part_scatter, dim3(grid_for(n)),
-			   dim3(THREADS), 0, s, keys, vals, n, shift,
-			   cursors, out_k, out_v);
+	if (gb_nparts(shift) <= GB_LDS_PARTS)
+		hipLaunchKernelGGL((k_gb_part_scatter_t<1>),
+				   dim3(grid_for(n)), dim3(THREADS), 0, s,
+				   keys, vals, n, 0, shift, cursors, out_k,
+				   out_v);
+	else
+		hipLaunchKernelGGL(k_gb_part_scatter, dim3(grid_for(n)),
+				   dim3(THREADS), 0, s, keys, vals, n, shift,
+				   cursors, out_k, out_v);
 	return hipGetLastError();
 }
 
