@@ -234,7 +234,10 @@ void k_sumprice(const int32_t *__restrict__ shipdate,
 			sum += (unsigned long long) nt_ld64(&price[i]);
 			cnt++;
 		}
-	/* wave reduction then one atomic per wave */
+	/* block-level reduction: one atomic pair per block (per-wave
+	 * flushes serialize at the ~88 atomics/µs hot-word wall) */
+	__shared__ unsigned long long ls[4][2];
+
 	for (int off = 32; off; off >>= 1)
 	{
 		sum += __shfl_down(sum, off, 64);
@@ -242,8 +245,24 @@ void k_sumprice(const int32_t *__restrict__ shipdate,
 	}
 	if ((threadIdx.x & 63) == 0)
 	{
-		atomicAdd(&acc->sum_c, sum);
-		atomicAdd(&acc->count, cnt);
+		ls[threadIdx.x >> 6][0] = sum;
+		ls[threadIdx.x >> 6][1] = cnt;
+	}
+	__syncthreads();
+	if (threadIdx.x == 0)
+	{
+		unsigned long long s = 0, c = 0;
+
+		for (int i = 0; i < (int) (blockDim.x >> 6); i++)
+		{
+			s += ls[i][0];
+			c += ls[i][1];
+		}
+		if (s || c)
+		{
+			atomicAdd(&acc->sum_c, s);
+			atomicAdd(&acc->count, c);
+		}
 	}
 }
 
