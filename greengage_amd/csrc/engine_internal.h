@@ -211,7 +211,16 @@ struct Timed
 		(void) hipEventElapsedTime(&ms, a, b);
 		(void) hipEventDestroy(a);
 		(void) hipEventDestroy(b);
+		a = b = nullptr;
 		return ms;
+	}
+	/* error paths return without stop(): don't leak the events */
+	~Timed()
+	{
+		if (a)
+			(void) hipEventDestroy(a);
+		if (b)
+			(void) hipEventDestroy(b);
 	}
 };
 
