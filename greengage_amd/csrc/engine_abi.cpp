@@ -1536,7 +1536,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 		GG_TRY(cached_max_i64(e, p, c_ck, cu->nrows, ctr, &maxk));
 		if (cu->nrows > 0 && maxk > 0 &&
-		    (int64_t) maxk <= 8 * cu->nrows + 16)
+		    maxk <= (unsigned long long) (8 * cu->nrows + 16))
 			cust_dlen = (int64_t) maxk + 1;
 	}
 	/* record which build-side layout ran (nodeHash.c:450 sizing
@@ -1614,7 +1614,7 @@ static gg_status exec_q3(Pipeline *p, void *arena, size_t bytes,
 
 		GG_TRY(cached_max_i64(e, p, o_ok, od->nrows, ctr, &maxk));
 		if (od->nrows > 0 && maxk > 0 &&
-		    (int64_t) maxk <= 8 * od->nrows + 16)
+		    maxk <= (unsigned long long) (8 * od->nrows + 16))
 			ord_dlen = (int64_t) maxk + 1;
 	}
 	p->stat(ord_dlen ? "path_orders_dense" : "path_orders_hash").launches++;
@@ -2323,7 +2323,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 
 		GG_TRY(cached_max_i64(e, p, c_ck, cu->nrows, ctr, &maxk));
 		if (cu->nrows > 0 && maxk > 0 &&
-		    (int64_t) maxk <= 8 * cu->nrows + 16)
+		    maxk <= (unsigned long long) (8 * cu->nrows + 16))
 			cust_dlen = (int64_t) maxk + 1;
 	}
 	p->stat(cust_dlen ? "path_cust_dense" : "path_cust_hash").launches++;
@@ -2381,7 +2381,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 
 		GG_TRY(cached_max_i64(e, p, s_sk, su->nrows, ctr, &maxk));
 		if (su->nrows > 0 && maxk > 0 &&
-		    (int64_t) maxk <= 8 * su->nrows + 16)
+		    maxk <= (unsigned long long) (8 * su->nrows + 16))
 			supp_dense_len = (int64_t) maxk + 1;
 	}
 	{
@@ -2593,7 +2593,7 @@ static gg_status exec_q5(Pipeline *p, void *arena, size_t bytes,
 
 		GG_TRY(cached_max_i64(e, p, o_ok, od->nrows, ctr, &maxk));
 		if (od->nrows > 0 && maxk > 0 &&
-		    (int64_t) maxk <= 8 * od->nrows + 16)
+		    maxk <= (unsigned long long) (8 * od->nrows + 16))
 			ord_dlen = (int64_t) maxk + 1;
 	}
 	p->stat(ord_dlen ? "path_orders_dense" : "path_orders_hash").launches++;

@@ -2475,7 +2475,14 @@ k_max_i64(const int64_t *__restrict__ v, int64_t n,
 	{
 		int64_t x = nt_ld64(&v[i]);
 
-		if (x > 0 && (unsigned long long) x > m)
+		/* a negative key poisons the result to ~0 so every
+		 * dense-path sizing guard (unsigned compare against
+		 * 8*rows) rejects the column: the dense bitmap/array
+		 * builds index by key and would silently drop negative
+		 * keys, diverging from the reference's hash join */
+		if (x < 0)
+			m = ~0ull;
+		else if ((unsigned long long) x > m)
 			m = (unsigned long long) x;
 	}
 	for (int off = 32; off; off >>= 1)

@@ -263,7 +263,7 @@ exec_plan(Pipeline *p, void *arena, size_t bytes, size_t *written)
 				e, p, (const int64_t *) B.bd.key, B.bd.n,
 				ctr, &maxk));
 			if (B.bd.n > 0 && maxk > 0 &&
-			    (int64_t) maxk <= 8 * B.bd.n + 16)
+			    maxk <= (unsigned long long) (8 * B.bd.n + 16))
 				B.dlen = (int64_t) maxk + 1;
 		}
 		if (!B.dlen && !B.hslots)
