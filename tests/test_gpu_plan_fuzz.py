@@ -34,6 +34,7 @@ def eng():
     from greengage_amd import Engine
     e = Engine()
     yield e
+    e.shutdown()
 
 
 TYPES = [("int64", np.int64), ("int32", np.int32),
@@ -137,6 +138,12 @@ def _eval_plan(n, data, nulls, preds, joins, group_cols, aggs):
         return tuple(out)
 
     groups = {}
+    if not group_cols:
+        # plain aggregate: exactly one output row even over zero
+        # input rows (nodeAgg emits the initial transvalues; the
+        # arena represents empty SUMs as 0 with COUNT=0 available to
+        # the finalizer to distinguish NULL)
+        groups[(0, 0)] = [0] * len(aggs)
     for i in idx.tolist():
         key = gkey(i)
         acc = groups.setdefault(key, [0] * len(aggs))
