@@ -577,9 +577,15 @@ hipError_t launch_plan_compact(hipStream_t s,
 gg_status exec_plan(Pipeline *p, void *arena, size_t bytes,
 		    size_t *written);
 
-/* plan_rtc.cpp — per-plan kernel specialization via hipRTC */
+/* plan_rtc.cpp — per-plan kernel specialization via hipRTC.
+ * bake/nbake: optional observed group codes (from a prior execute of
+ * the same plan over the same immutable tables); when given, the
+ * generated kernel indexes a direct LDS accumulator per baked code
+ * (compare chain, no key probing) and routes unseen codes through
+ * the global table. */
 gg_status plan_rtc_compile(const PlanDev &D, bool has_gnull0,
-			   bool has_gnull1, std::shared_ptr<void> *out);
+			   bool has_gnull1, std::shared_ptr<void> *out,
+			   const long long *bake = nullptr, int nbake = 0);
 gg_status plan_rtc_launch(hipStream_t s, const std::shared_ptr<void> &h,
 			  PlanDev P, int grid, int block);
 

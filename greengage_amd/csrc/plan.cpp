@@ -50,6 +50,11 @@ struct PlanResolved
 	 * shape (bitmap vs hash set) is known */
 	std::shared_ptr<void> rtc;
 	bool rtc_tried = false;
+	/* second-stage specialization: after the first execute the
+	 * (small) group set is known exactly — tables are immutable —
+	 * and gets baked into a direct-indexed kernel */
+	std::shared_ptr<void> rtc_baked;
+	bool bake_tried = false;
 };
 
 static int coltype_width(gg_coltype t)
@@ -359,8 +364,10 @@ exec_plan(Pipeline *p, void *arena, size_t bytes, size_t *written)
 				g = 2048;
 			if (g < 1)
 				g = 1;
-			GG_TRY(plan_rtc_launch(e.stream, R->rtc, R->dev,
-					       (int) g, 256));
+			GG_TRY(plan_rtc_launch(
+				e.stream,
+				R->rtc_baked ? R->rtc_baked : R->rtc,
+				R->dev, (int) g, 256));
 		}
 		else
 			GG_HIP(launch_plan_scan_agg(e.stream, R->dev));
@@ -553,6 +560,36 @@ exec_plan(Pipeline *p, void *arena, size_t bytes, size_t *written)
 		}
 	}
 	*written = need;
+
+	/* second-stage bake: with the group set observed (and exact —
+	 * registered tables are immutable), recompile the RTC kernel
+	 * with the codes as a constexpr compare chain so later
+	 * executes of this plan skip key probing entirely.  Gated on a
+	 * small group count (register/LDS budget) and on the first
+	 * RTC compile having succeeded. */
+	{
+		int bake_max = 8;
+		const char *bm = getenv("GG_PLAN_BAKE");
+
+		if (bm && bm[0] == '0')
+			bake_max = 0;
+		if (!R->bake_tried && R->rtc && R->dev.ngroup > 0 &&
+		    ng > 0 && (int64_t) ng <= bake_max)
+		{
+			std::vector<long long> codes(ng);
+
+			for (size_t i = 0; i < ng; i++)
+				codes[i] = (long long) rows[i * rowsz];
+			gg_status rs = plan_rtc_compile(
+				R->dev, R->dev.gnulls[0] != nullptr,
+				R->dev.gnulls[1] != nullptr, &R->rtc_baked,
+				codes.data(), (int) ng);
+
+			R->bake_tried = true;
+			if (rs == GG_OK)
+				p->stat("path_plan_rtc_baked").launches++;
+		}
+	}
 	return GG_OK;
 }
 

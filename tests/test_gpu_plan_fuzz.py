@@ -67,7 +67,11 @@ def _mk_table(eng, rng, name, nrows, want_char1=0):
             a = rng.integers(-120, 120, nrows).astype(np.int64)
         cols.append((f"v{i}", tname, a))
     for i in range(want_char1):
-        a = rng.integers(0, 6, nrows).astype(np.uint8)
+        # small random alphabet: 2x2 .. 6x6 group grids, so repeated
+        # executes sometimes cross the <=8-group bake threshold and
+        # sometimes stay on the generic grouped kernel
+        a = rng.integers(0, int(rng.integers(2, 7)),
+                         nrows).astype(np.uint8)
         cols.append((f"g{i}", "char1", a))
 
     t = eng.register_table(name, cols, nrows)
@@ -236,6 +240,11 @@ def _run_one(eng, rng, it, force_interp=False):
         assert g[0] == e[0] and g[1] == e[1], f"it={it}: key {g} vs {e}"
         assert g[2] == e[2], f"it={it}: vals {g} vs {e}"
 
+    # second execute: plans with <=8 observed groups re-run on the
+    # baked-codes kernel — results must be identical
+    got2 = sorted(eng.execute_plan(p, max_groups=1 << 17))
+    assert got2 == got_t, f"it={it}: repeat/baked execute diverged"
+
 
 def test_plan_fuzz_rtc(eng):
     rng = np.random.default_rng(20260915)
@@ -247,6 +256,41 @@ def test_plan_fuzz_interpreted(eng):
     rng = np.random.default_rng(777)
     for it in range(8):
         _run_one(eng, rng, 100 + it, force_interp=True)
+
+
+def test_baked_group_codes_kernel(eng):
+    """Q1-shaped repeat execution: after the first execute the 2x2
+    char1 group set is baked into a direct-indexed RTC kernel
+    (path_plan_rtc_baked stat row); executes 2..4 must stay
+    bit-identical to the exact python evaluation."""
+    rng = np.random.default_rng(99)
+    n = 400_000
+    g0 = rng.choice(np.frombuffer(b"AN", np.uint8), n)
+    g1 = rng.choice(np.frombuffer(b"FO", np.uint8), n)
+    q = rng.integers(1, 51, n).astype(np.int64)
+    pr = rng.integers(900, 105000, n).astype(np.int64)
+    d = rng.integers(0, 11, n).astype(np.int64)
+    sd = rng.integers(0, 2526, n).astype(np.int64)
+    t = eng.register_table("bake_q1", [
+        ("g0", "char1", g0), ("g1", "char1", g1),
+        ("q", "dec64", q), ("pr", "dec64", pr), ("d", "dec64", d),
+        ("sd", "int64", sd)], n)
+    data = {c: (a.astype(object), ty) for (c, ty, a) in [
+        ("g0", "char1", g0), ("g1", "char1", g1), ("q", "dec64", q),
+        ("pr", "dec64", pr), ("d", "dec64", d), ("sd", "int64", sd)]}
+    nulls = {c: np.zeros(n, bool) for c in data}
+    preds = [("sd", NEG_INF, 2400)]
+    aggs = ["count", ("sum", [("q", "id")]), ("sum", [("pr", "id")]),
+            ("sum", [("pr", "id"), ("d", "sub100")])]
+    p = eng.compile_plan(t, preds=preds, group_cols=["g0", "g1"],
+                         aggs=aggs)
+    exp = _eval_plan(n, data, nulls, preds, [], ["g0", "g1"], aggs)
+    runs = [sorted(eng.execute_plan(p)) for _ in range(4)]
+    for r in runs:
+        assert r == exp
+    paths = {s["name"] for s in eng.stats(p)}
+    assert "path_plan_rtc_baked" in paths or \
+        "path_plan_interp" in paths  # interp-only boxes: no bake
 
 
 def test_plan_fuzz_group1_with_nulls(eng):
