@@ -141,6 +141,46 @@ engine_cached_max_i64(Engine &e, Pipeline *p, const int64_t *col, int64_t n,
 	return cached_max_i64(e, p, col, n, ctr, out);
 }
 
+/* signed min/max of an immutable registered column, cached on the
+ * engine; one scan per column ever */
+gg_status
+engine_col_minmax(Engine &e, const void *dcol, int width, int64_t n,
+		  long long *mn, long long *mx)
+{
+	auto it = e.mm_cache.find(dcol);
+
+	if (it != e.mm_cache.end())
+	{
+		*mn = it->second.first;
+		*mx = it->second.second;
+		return GG_OK;
+	}
+	{
+		unsigned long long seed[2] = {~0ull, 0ull};
+		unsigned long long *d = nullptr;
+
+		GG_HIP(hipMalloc((void **) &d, 16));
+		GG_HIP(hipMemcpyAsync(d, seed, 16, hipMemcpyHostToDevice,
+				      e.stream));
+		hipError_t he = launch_minmax_i64(e.stream, dcol, width, n, d);
+
+		if (he != hipSuccess)
+		{
+			(void) hipFree(d);
+			return fail(GG_EGPU, "minmax launch: %s",
+				    hipGetErrorString(he));
+		}
+		GG_HIP(hipMemcpyAsync(seed, d, 16, hipMemcpyDeviceToHost,
+				      e.stream));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		(void) hipFree(d);
+		*mn = (long long) (seed[0] ^ 0x8000000000000000ull);
+		*mx = (long long) (seed[1] ^ 0x8000000000000000ull);
+		e.mm_cache[dcol] = {*mn, *mx};
+	}
+	return GG_OK;
+}
+
 /* ---------------- int128 host helpers (combine/finalize) ---------------- */
 
 typedef __int128 i128;
@@ -205,6 +245,7 @@ extern "C" gg_status gg_engine_shutdown(void)
 		delete t;
 	}
 	e.tables.clear();
+	e.mm_cache.clear();
 	for (auto *p : e.pipelines)
 		delete p;
 	e.pipelines.clear();
@@ -1020,6 +1061,10 @@ extern "C" gg_status gg_engine_drop_table(gg_table h)
 		return fail(GG_EINVAL, "bad table handle %d", h);
 	for (auto &c : t->cols)
 	{
+		/* the min/max cache is keyed by device pointer: a freed
+		 * buffer's address can be handed out again by a later
+		 * hipMalloc, so stale entries must not survive the free */
+		e.mm_cache.erase(c.dev);
 		if (c.bytes)
 			(void) hipFree(c.dev);
 		if (c.nulls)

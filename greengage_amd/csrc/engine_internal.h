@@ -8,6 +8,7 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <map>
 #include <memory>
 #include <string>
 #include <vector>
@@ -187,6 +188,9 @@ struct Engine
 	std::vector<Table *> tables;
 	std::vector<Pipeline *> pipelines;
 	void *comm = nullptr;	/* ncclComm_t when comm_init'ed */
+	/* per-column min/max cache (registered columns are immutable);
+	 * keyed by device pointer */
+	std::map<const void *, std::pair<long long, long long>> mm_cache;
 };
 
 Engine &engine();
@@ -585,7 +589,8 @@ gg_status exec_plan(Pipeline *p, void *arena, size_t bytes,
  * the global table. */
 gg_status plan_rtc_compile(const PlanDev &D, bool has_gnull0,
 			   bool has_gnull1, std::shared_ptr<void> *out,
-			   const long long *bake = nullptr, int nbake = 0);
+			   const long long *bake = nullptr, int nbake = 0,
+			   bool fast = false);
 gg_status plan_rtc_launch(hipStream_t s, const std::shared_ptr<void> &h,
 			  PlanDev P, int grid, int block);
 
@@ -594,6 +599,12 @@ Table *engine_table(gg_table h);
 gg_status engine_cached_max_i64(Engine &e, Pipeline *p, const int64_t *col,
 				int64_t n, unsigned long long *ctr,
 				unsigned long long *out);
+/* signed min/max of a registered column (width 1/4/8), cached on the
+ * engine (columns are immutable) */
+gg_status engine_col_minmax(Engine &e, const void *dcol, int width,
+			    int64_t n, long long *mn, long long *mx);
+hipError_t launch_minmax_i64(hipStream_t s, const void *v, int width,
+			     int64_t n, unsigned long long *out2);
 
 #define GG_MT_MAX_ATTS 32
 

@@ -2505,6 +2505,64 @@ launch_max_i64(hipStream_t s, const int64_t *v, int64_t n,
 	return hipGetLastError();
 }
 
+/* signed min/max of a column (width 1/4/8), order-encoded into u64
+ * (x ^ sign bit) so LDS-free atomicMin/Max work; host seeds
+ * out2 = {~0, 0} and decodes.  Feeds the aggregate overflow-budget
+ * proof for the baked plan kernel (plan.cpp). */
+template <int W>
+__global__ void
+k_minmax_i64(const void *__restrict__ vp, int64_t n,
+	     unsigned long long *out2)
+{
+	const int64_t stride = (int64_t) gridDim.x * blockDim.x;
+	unsigned long long mn = ~0ull, mx = 0;
+
+	for (int64_t i = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     i < n; i += stride)
+	{
+		int64_t x = W == 1 ? (int64_t) ((const uint8_t *) vp)[i]
+			: W == 4 ? (int64_t) ((const int32_t *) vp)[i]
+			: ((const int64_t *) vp)[i];
+		unsigned long long e = (unsigned long long) x
+			^ 0x8000000000000000ull;
+
+		if (e < mn)
+			mn = e;
+		if (e > mx)
+			mx = e;
+	}
+	for (int off = 32; off; off >>= 1)
+	{
+		unsigned long long a = __shfl_down(mn, off, 64);
+		unsigned long long b = __shfl_down(mx, off, 64);
+
+		if (a < mn)
+			mn = a;
+		if (b > mx)
+			mx = b;
+	}
+	if ((threadIdx.x & 63) == 0)
+	{
+		atomicMin(&out2[0], mn);
+		atomicMax(&out2[1], mx);
+	}
+}
+
+hipError_t
+launch_minmax_i64(hipStream_t s, const void *v, int width, int64_t n,
+		  unsigned long long *out2)
+{
+	dim3 g(grid_for(n)), b(THREADS);
+
+	if (width == 1)
+		hipLaunchKernelGGL(k_minmax_i64<1>, g, b, 0, s, v, n, out2);
+	else if (width == 4)
+		hipLaunchKernelGGL(k_minmax_i64<4>, g, b, 0, s, v, n, out2);
+	else
+		hipLaunchKernelGGL(k_minmax_i64<8>, g, b, 0, s, v, n, out2);
+	return hipGetLastError();
+}
+
 __global__ void
 k_supp_dense_fill(const int64_t *__restrict__ suppkey,
 		  const uint8_t *__restrict__ snation, int64_t n,

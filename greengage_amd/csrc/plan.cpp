@@ -580,14 +580,96 @@ exec_plan(Pipeline *p, void *arena, size_t bytes, size_t *written)
 
 			for (size_t i = 0; i < ng; i++)
 				codes[i] = (long long) rows[i * rowsz];
+
+			/* overflow-budget proof for the fire-and-forget
+			 * LDS tier: interval-bound every aggregate value
+			 * from cached column min/max (immutable data).
+			 * All values must be non-negative and
+			 * bound * rows_per_block must clear 2^61 with a
+			 * 4x margin (long double keeps int64 exact; the
+			 * margin absorbs product rounding). */
+			bool fastok = true;
+			int64_t grid = (R->dev.n + 255) / 256;
+
+			if (grid > 2048)
+				grid = 2048;
+			if (grid < 1)
+				grid = 1;
+			{
+				long double rows_pb = (long double)
+					R->dev.n / (long double) grid + 256.0L;
+
+				for (int a = 0;
+				     fastok && a < R->dev.naggs; a++)
+				{
+					const PlanAggDev &A = R->dev.aggs[a];
+					long double lo = 1.0L, hi = 1.0L;
+
+					if (A.kind != 2)
+						continue;	/* count: [1,1] */
+					for (int f = 0;
+					     fastok && f < A.nf; f++)
+					{
+						long long mn, mx;
+
+						if (engine_col_minmax(
+							e, A.col[f],
+							A.width[f],
+							R->dev.n, &mn,
+							&mx) != GG_OK)
+						{
+							fastok = false;
+							break;
+						}
+						long double a0 = (long double) mn;
+						long double a1 = (long double) mx;
+
+						if (A.mod[f] == 1)
+						{
+							long double t0 = 100.0L - a1;
+							long double t1 = 100.0L - a0;
+
+							a0 = t0;
+							a1 = t1;
+						}
+						else if (A.mod[f] == 2)
+						{
+							a0 = 100.0L + a0;
+							a1 = 100.0L + a1;
+						}
+						{
+							long double c[4] = {
+								lo * a0, lo * a1,
+								hi * a0, hi * a1};
+							long double nl = c[0],
+								nh = c[0];
+
+							for (int q = 1; q < 4; q++)
+							{
+								if (c[q] < nl)
+									nl = c[q];
+								if (c[q] > nh)
+									nh = c[q];
+							}
+							lo = nl;
+							hi = nh;
+						}
+					}
+					if (lo < 0.0L ||
+					    hi * rows_pb >=
+					    2305843009213693952.0L) /* 2^61 */
+						fastok = false;
+				}
+			}
 			gg_status rs = plan_rtc_compile(
 				R->dev, R->dev.gnulls[0] != nullptr,
 				R->dev.gnulls[1] != nullptr, &R->rtc_baked,
-				codes.data(), (int) ng);
+				codes.data(), (int) ng, fastok);
 
 			R->bake_tried = true;
 			if (rs == GG_OK)
-				p->stat("path_plan_rtc_baked").launches++;
+				p->stat(fastok ? "path_plan_rtc_baked_fast"
+					: "path_plan_rtc_baked").launches++;
 		}
 	}
 	return GG_OK;

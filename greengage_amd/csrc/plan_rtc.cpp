@@ -142,7 +142,7 @@ static void emit_ld(std::string &s, const char *dst, const char *ptr,
 gg_status
 plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 		 std::shared_ptr<void> *out, const long long *bake,
-		 int nbake)
+		 int nbake, bool fast)
 {
 	const char *dis = getenv("GG_PLAN_RTC");
 
@@ -417,11 +417,22 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 				s += buf;
 			}
 			s += "};\n";
+			/* LW = u64 words per accumulator.  fast (=proven
+			 * by plan.cpp's interval bound: every agg value
+			 * non-negative and bound*rows_per_block < 2^61)
+			 * uses ONE word and fire-and-forget LDS adds —
+			 * the returned-value carry check otherwise makes
+			 * every add a RETURNING atomic, which serializes
+			 * at the ~88 returning-atomics/us hot-word wall
+			 * (measured: baked 2-word = generic 5.5 ms;
+			 * k_q1_agg's budget-proven 1-word form = 3.6 ms
+			 * on the same shape). */
+			s += fast ? "#define LW 1\n" : "#define LW 2\n";
 			s += R"GG(
-	__shared__ unsigned long long lvals[LREPL][NG][2 * NA];
+	__shared__ unsigned long long lvals[LREPL][NG][LW * NA];
 	__shared__ unsigned int btouch;
 
-	for (int q = threadIdx.x; q < LREPL * NG * 2 * NA; q += blockDim.x)
+	for (int q = threadIdx.x; q < LREPL * NG * LW * NA; q += blockDim.x)
 		((unsigned long long *) lvals)[q] = 0;
 	if (threadIdx.x == 0)
 		btouch = 0;
@@ -479,6 +490,10 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 			{
 				if (!aok[a])
 					continue;
+#if LW == 1
+				atomicAdd(&lvals[lrep][g][a],
+					  (unsigned long long) av[a]);
+#else
 				unsigned long long vlo =
 					(unsigned long long) av[a];
 				unsigned long long vhi =
@@ -491,6 +506,7 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 				if (vhi)
 					atomicAdd(&lvals[lrep][g][2 * a + 1],
 						  vhi);
+#endif
 			}
 		}
 		else
@@ -529,6 +545,12 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 		if (slot < 0) { atomicOr(P.err, 1ull); continue; }
 		for (int a = 0; a < NA; a++)
 		{
+#if LW == 1
+			unsigned long long lo = 0, hi = 0;
+
+			for (int rr = 0; rr < LREPL; rr++)
+				lo += lvals[rr][q][a];
+#else
 			unsigned long long lo = 0, hi = 0;
 
 			for (int rr = 0; rr < LREPL; rr++)
@@ -539,6 +561,7 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 				lo += rl;
 				hi += lvals[rr][q][2 * a + 1] + (lo < o);
 			}
+#endif
 			if (!lo && !hi)
 				continue;
 			unsigned long long old =
