@@ -148,6 +148,8 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 
 	if (dis && dis[0] == '0')
 		return fail(GG_ENOTSUP, "rtc disabled");
+	if (nbake <= 0)
+		fast = false;	/* the u64 fast tier exists only baked */
 
 	std::string s = STRUCT_DEFS;
 	char buf[512];
@@ -175,6 +177,12 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 		      "#define LEMPTY ((long long) 0x8000000000000000ull)\n",
 		      sizeof(PlanDev), lrepl);
 	s += buf;
+	/* fast (bounds-proven) aggregate values are exact in modular u64
+	 * arithmetic — the true per-row value is non-negative < 2^61, so
+	 * wrapping intermediates preserve the result and the int128
+	 * multiplies (3-4 u64 muls each) collapse to one mul per factor */
+	s += fast ? "#define VAT unsigned long long\n"
+		  : "#define VAT __int128\n";
 
 	s += "extern \"C\" __global__ "
 	     "__launch_bounds__(256, 4)\nvoid plan_kernel(PlanDev P)\n{\n"
@@ -242,7 +250,7 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 	s += rowfn;
 
 	/* agg input values: emits "__int128 aY; bool aY_ok;" per agg */
-	std::string aggfn = "\tauto agg_vals = [&](int64_t i, __int128 *av, bool *aok)\n\t{\n";
+	std::string aggfn = "\tauto agg_vals = [&](int64_t i, VAT *av, bool *aok)\n\t{\n";
 	for (int a = 0; a < D.naggs; a++)
 	{
 		if (D.aggs[a].kind == 0)
@@ -268,7 +276,7 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 			aggfn += buf;
 			continue;
 		}
-		std::snprintf(buf, sizeof(buf), "\t\t{ __int128 v = 1;\n");
+		std::snprintf(buf, sizeof(buf), "\t\t{ VAT v = 1;\n");
 		aggfn += buf;
 		for (int f = 0; f < D.aggs[a].nf; f++)
 		{
@@ -290,8 +298,8 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 					      "\t\t%s = 100 + %s;\n", d, d);
 				aggfn += buf;
 			}
-			std::snprintf(buf, sizeof(buf), "\t\tv *= x%d_%d;\n",
-				      a, f);
+			std::snprintf(buf, sizeof(buf),
+				      "\t\tv *= (VAT) x%d_%d;\n", a, f);
 			aggfn += buf;
 		}
 		std::snprintf(buf, sizeof(buf), "\t\tav[%d] = v; }\n", a);
@@ -428,12 +436,15 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 			 * k_q1_agg's budget-proven 1-word form = 3.6 ms
 			 * on the same shape). */
 			s += fast ? "#define LW 1\n" : "#define LW 2\n";
+			/* odd per-replica stride: spreads replicas across
+			 * LDS banks (k_q1_agg's Q1_STRIDE trick) */
+			s += "#define LPAD ((NG * LW * NA) | 1)\n";
 			s += R"GG(
-	__shared__ unsigned long long lvals[LREPL][NG][LW * NA];
+	__shared__ unsigned long long lvals[LREPL * LPAD];
 	__shared__ unsigned int btouch;
 
-	for (int q = threadIdx.x; q < LREPL * NG * LW * NA; q += blockDim.x)
-		((unsigned long long *) lvals)[q] = 0;
+	for (int q = threadIdx.x; q < LREPL * LPAD; q += blockDim.x)
+		lvals[q] = 0;
 	if (threadIdx.x == 0)
 		btouch = 0;
 	__syncthreads();
@@ -474,7 +485,7 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 )GG";
 			s += gcode;
 			s += R"GG(
-		__int128 av[NA];
+		VAT av[NA];
 		bool aok[NA];
 
 		agg_vals(i, av, aok);
@@ -485,13 +496,15 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 				g = q;
 		if (g >= 0)
 		{
+			unsigned long long *mine = &lvals[lrep * LPAD];
+
 			tmask |= 1u << g;
 			for (int a = 0; a < NA; a++)
 			{
 				if (!aok[a])
 					continue;
 #if LW == 1
-				atomicAdd(&lvals[lrep][g][a],
+				atomicAdd(&mine[g * NA + a],
 					  (unsigned long long) av[a]);
 #else
 				unsigned long long vlo =
@@ -499,12 +512,12 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 				unsigned long long vhi =
 					(unsigned long long) (av[a] >> 64);
 				unsigned long long old = atomicAdd(
-					&lvals[lrep][g][2 * a], vlo);
+					&mine[(g * NA + a) * 2], vlo);
 
 				if (old + vlo < old)
 					vhi++;
 				if (vhi)
-					atomicAdd(&lvals[lrep][g][2 * a + 1],
+					atomicAdd(&mine[(g * NA + a) * 2 + 1],
 						  vhi);
 #endif
 			}
@@ -520,8 +533,12 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 					continue;
 				unsigned long long vlo =
 					(unsigned long long) av[a];
+#if LW == 1
+				unsigned long long vhi = 0;
+#else
 				unsigned long long vhi =
 					(unsigned long long) (av[a] >> 64);
+#endif
 				unsigned long long old = atomicAdd(
 					&P.tvals[(slot * NA + a) * 2], vlo);
 
@@ -549,17 +566,19 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 			unsigned long long lo = 0, hi = 0;
 
 			for (int rr = 0; rr < LREPL; rr++)
-				lo += lvals[rr][q][a];
+				lo += lvals[rr * LPAD + q * NA + a];
 #else
 			unsigned long long lo = 0, hi = 0;
 
 			for (int rr = 0; rr < LREPL; rr++)
 			{
-				unsigned long long rl = lvals[rr][q][2 * a];
+				unsigned long long rl =
+					lvals[rr * LPAD + (q * NA + a) * 2];
 				unsigned long long o = lo;
 
 				lo += rl;
-				hi += lvals[rr][q][2 * a + 1] + (lo < o);
+				hi += lvals[rr * LPAD +
+					    (q * NA + a) * 2 + 1] + (lo < o);
 			}
 #endif
 			if (!lo && !hi)

@@ -119,7 +119,7 @@ for pp, name in ((p, "q6"), (pq1, "q1-shaped")):
 print("PLAN_BENCH4_OK")
 
 # LDS replica sweep for the RTC grouped kernel
-for repl in (4, 8, 16):
+for repl in (4, 8, 16, 32):
     os.environ["GG_PLAN_LREPL"] = str(repl)
     pr = eng.compile_plan(
         li, preds=[("shipdate", NEG_INF, cutoff + 1)],
