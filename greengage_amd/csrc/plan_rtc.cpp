@@ -123,17 +123,25 @@ __device__ inline uint32_t gg_hashint8(int64_t v)
 }
 )GG";
 
-/* emit a typed, width-baked load */
+/* emit a typed, width-baked load; nt = nontemporal streaming load
+ * (k_q1_agg measured +11% from NT on its scan columns — the data is
+ * touched once per pass, so L2 retention only evicts useful lines) */
 static void emit_ld(std::string &s, const char *dst, const char *ptr,
-		    int width, const char *idx)
+		    int width, const char *idx, bool nt = false)
 {
 	char buf[256];
 	const char *ty = width == 1 ? "const uint8_t *"
 		: width == 4 ? "const int32_t *" : "const int64_t *";
 
-	std::snprintf(buf, sizeof(buf),
-		      "\t\tint64_t %s = (int64_t) ((%s) %s)[%s];\n",
-		      dst, ty, ptr, idx);
+	if (nt && width != 1)
+		std::snprintf(buf, sizeof(buf),
+			      "\t\tint64_t %s = (int64_t) "
+			      "__builtin_nontemporal_load(&((%s) %s)[%s]);\n",
+			      dst, ty, ptr, idx);
+	else
+		std::snprintf(buf, sizeof(buf),
+			      "\t\tint64_t %s = (int64_t) ((%s) %s)[%s];\n",
+			      dst, ty, ptr, idx);
 	s += buf;
 }
 
@@ -150,6 +158,9 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 		return fail(GG_ENOTSUP, "rtc disabled");
 	if (nbake <= 0)
 		fast = false;	/* the u64 fast tier exists only baked */
+
+	const char *ntv = getenv("GG_PLAN_RTC_NT");
+	bool nt = nbake > 0 && !(ntv && ntv[0] == '0');
 
 	std::string s = STRUCT_DEFS;
 	char buf[512];
@@ -198,7 +209,7 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 
 		std::snprintf(d, sizeof(d), "v%d", p);
 		std::snprintf(ptr, sizeof(ptr), "P.preds[%d].col", p);
-		emit_ld(rowfn, d, ptr, D.preds[p].width, "i");
+		emit_ld(rowfn, d, ptr, D.preds[p].width, "i", nt);
 		if (D.preds[p].nulls)
 		{
 			std::snprintf(buf, sizeof(buf),
@@ -285,7 +296,7 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 			std::snprintf(d, sizeof(d), "x%d_%d", a, f);
 			std::snprintf(ptr, sizeof(ptr), "P.aggs[%d].col[%d]",
 				      a, f);
-			emit_ld(aggfn, d, ptr, D.aggs[a].width[f], "i");
+			emit_ld(aggfn, d, ptr, D.aggs[a].width[f], "i", nt);
 			if (D.aggs[a].mod[f] == 1)
 			{
 				std::snprintf(buf, sizeof(buf),
