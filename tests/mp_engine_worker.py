@@ -85,6 +85,10 @@ def main():
         group_cols=["rflag", "lstatus"],
         aggs=["count", ("sum", [("qty", "id")])])
     out["plan_grouped"] = eng.execute_plan(pg, max_groups=64)
+    # repeat execute: the second run takes the baked-codes kernel
+    # (codes observed from the MERGED cross-segment group set) and
+    # must still combine to the identical result
+    out["plan_grouped_repeat"] = eng.execute_plan(pg, max_groups=64)
 
     eng.shutdown()
     with open(outfile + ".tmp", "w") as f:

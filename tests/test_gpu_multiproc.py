@@ -70,6 +70,7 @@ def check_results(results, world, sf):
         assert r["q5"] == results[0]["q5"]
         assert r["plan_q6"] == results[0]["plan_q6"]
         assert r["plan_grouped"] == results[0]["plan_grouped"]
+        assert r["plan_grouped_repeat"] == results[0]["plan_grouped"]
 
     exp_q1 = [g for g in pyoracle.q1_synth(42, sf, PGDate("1998-08-15"))
               if g["count"]]
@@ -120,6 +121,8 @@ def check_results(results, world, sf):
     assert len(grouped) == len(exp_g)
     for k0, k1, vals in grouped:
         assert vals == exp_g[(k0, k1)]
+    # baked-kernel repeat through the cross-segment combine
+    assert results[0]["plan_grouped_repeat"] == grouped
 
 
 def test_world2_q1_q3_q5_bitexact(tmp_path):

@@ -248,13 +248,13 @@ def _run_one(eng, rng, it, force_interp=False):
 
 def test_plan_fuzz_rtc(eng):
     rng = np.random.default_rng(20260915)
-    for it in range(24):
+    for it in range(32):
         _run_one(eng, rng, it)
 
 
 def test_plan_fuzz_interpreted(eng):
     rng = np.random.default_rng(777)
-    for it in range(8):
+    for it in range(10):
         _run_one(eng, rng, 100 + it, force_interp=True)
 
 
