@@ -4722,85 +4722,185 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 	for (uint64_t t = P; t > 1; t >>= 1)
 		shift--;
 
-	/* chunked partition pass: GPU hashes + scatters, host stages */
-	int64_t chunk = budget_bytes / (8 * 4);	/* in+out pairs resident */
+	/* chunked partition pass, double-buffered: the round-1 serial
+	 * form (pageable hipMemcpy both ways, a mid-chunk sync for the
+	 * host prefix, and up to P separate D2H copies per chunk) was
+	 * measured PCIe-staging-bound at 62 M rows/s.  Now: pinned
+	 * staging both directions, the cursor prefix computed on
+	 * device (k_gb_prefix2_u64 — no mid-chunk sync), ONE bulk D2H
+	 * per chunk, and two streams so chunk i+1's H2D and kernels
+	 * run while chunk i's scatter drains back and the host
+	 * distributes it (OpenMP memcpys) into the partition vectors.
+	 * Two buffer sets split the same device budget (chunk halved),
+	 * so peak GPU memory is unchanged. */
+	int64_t chunk = budget_bytes / (8 * 4 * 2);
 
 	if (chunk < 1024)
 		chunk = 1024;
+	if (chunk > n)
+		chunk = n;
 	std::vector<std::vector<int64_t>> part_k(P), part_v(P);
-	int64_t *dk = nullptr, *dv = nullptr, *sk = nullptr, *sv = nullptr;
-	unsigned long long *dcnt = nullptr;
 	gg_status st = GG_OK;
 
-	GG_HIP(hipMalloc((void **) &dk, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &dv, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &sk, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &sv, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &dcnt, (P + 1) * 8));
+	struct GbsBuf
+	{
+		int64_t *dk = nullptr, *dv = nullptr;
+		int64_t *sk = nullptr, *sv = nullptr;
+		unsigned long long *dcnt = nullptr;	/* [P cnt][P work][P+1 pristine] */
+		int64_t *hk = nullptr, *hv = nullptr;	/* pinned in */
+		int64_t *ok = nullptr, *ov = nullptr;	/* pinned out */
+		unsigned long long *hcur = nullptr;	/* pinned P+1 */
+		hipStream_t stm = nullptr;
+		hipEvent_t done = nullptr;
+		int64_t m = 0;
+		bool busy = false;
+	} B[2];
+	bool alloc_ok = true;
 
-	std::vector<unsigned long long> counts(P), curs(P + 1);
+	for (int b = 0; b < 2 && alloc_ok; b++)
+	{
+		GbsBuf &q = B[b];
 
-	/* no-leak variant: errors fall through to the frees below */
+		alloc_ok =
+			hipMalloc((void **) &q.dk, (size_t) chunk * 8) == hipSuccess &&
+			hipMalloc((void **) &q.dv, (size_t) chunk * 8) == hipSuccess &&
+			hipMalloc((void **) &q.sk, (size_t) chunk * 8) == hipSuccess &&
+			hipMalloc((void **) &q.sv, (size_t) chunk * 8) == hipSuccess &&
+			hipMalloc((void **) &q.dcnt, (3 * P + 1) * 8) == hipSuccess &&
+			hipHostMalloc((void **) &q.hk, (size_t) chunk * 8) == hipSuccess &&
+			hipHostMalloc((void **) &q.hv, (size_t) chunk * 8) == hipSuccess &&
+			hipHostMalloc((void **) &q.ok, (size_t) chunk * 8) == hipSuccess &&
+			hipHostMalloc((void **) &q.ov, (size_t) chunk * 8) == hipSuccess &&
+			hipHostMalloc((void **) &q.hcur, (P + 1) * 8) == hipSuccess &&
+			hipStreamCreate(&q.stm) == hipSuccess &&
+			hipEventCreate(&q.done) == hipSuccess;
+	}
+	if (!alloc_ok)
+		st = fail(GG_ENOMEM, "groupby_spill staging");
+
+	/* error policy: record first failure, stop issuing, drain */
 #define GG_HIP_GBS(x) \
 	{ hipError_t e_ = (x); \
 	  if (st == GG_OK && e_ != hipSuccess) \
 		st = fail(GG_EGPU, "groupby_spill: %s", \
-			  hipGetErrorString(e_)); \
-	  if (st != GG_OK) break; }
-	for (int64_t base = 0; st == GG_OK && base < n; base += chunk)
+			  hipGetErrorString(e_)); }
+
+	auto drain = [&](GbsBuf &q)
 	{
-		int64_t m = (n - base < chunk) ? n - base : chunk;
+		if (!q.busy)
+			return;
+		GG_HIP_GBS(hipEventSynchronize(q.done));
+		q.busy = false;
+		if (st != GG_OK)
+			return;
+		/* append staged rows to their partitions; resizes are
+		 * serial, the copies partition-parallel */
+		std::vector<size_t> old(P);
 
-		GG_HIP_GBS(hipMemcpy(dk, keys + base, (size_t) m * 8,
-				     hipMemcpyHostToDevice));
-		GG_HIP_GBS(hipMemcpy(dv, vals + base, (size_t) m * 8,
-				     hipMemcpyHostToDevice));
-		GG_HIP_GBS(hipMemsetAsync(dcnt, 0, P * 8, e.stream));
-		GG_HIP_GBS(launch_gb_part_count(e.stream, dk, m, shift,
-						dcnt));
-		GG_HIP_GBS(hipStreamSynchronize(e.stream));
-		GG_HIP_GBS(hipMemcpy(counts.data(), dcnt, P * 8,
-				     hipMemcpyDeviceToHost));
-		curs[0] = 0;
 		for (uint64_t p2 = 0; p2 < P; p2++)
-			curs[p2 + 1] = curs[p2] + counts[p2];
-		GG_HIP_GBS(hipMemcpy(dcnt, curs.data(), P * 8,
-				     hipMemcpyHostToDevice));
-		GG_HIP_GBS(launch_gb_part_scatter(e.stream, dk, dv, m,
-						  shift, dcnt, sk, sv));
-		GG_HIP_GBS(hipStreamSynchronize(e.stream));
-		bool bad = false;
-
-		for (uint64_t p2 = 0; p2 < P && !bad; p2++)
 		{
-			if (!counts[p2])
-				continue;
-			size_t old = part_k[p2].size();
+			size_t c = (size_t) (q.hcur[p2 + 1] - q.hcur[p2]);
 
-			part_k[p2].resize(old + counts[p2]);
-			part_v[p2].resize(old + counts[p2]);
-			hipError_t e1 = hipMemcpy(part_k[p2].data() + old,
-						  sk + curs[p2],
-						  counts[p2] * 8,
-						  hipMemcpyDeviceToHost);
-			hipError_t e2 = hipMemcpy(part_v[p2].data() + old,
-						  sv + curs[p2],
-						  counts[p2] * 8,
-						  hipMemcpyDeviceToHost);
-
-			if (e1 != hipSuccess || e2 != hipSuccess)
+			old[p2] = part_k[p2].size();
+			if (c)
 			{
-				st = fail(GG_EGPU, "groupby_spill stage");
-				bad = true;
+				part_k[p2].resize(old[p2] + c);
+				part_v[p2].resize(old[p2] + c);
 			}
 		}
+		ao_parallel_for((int64_t) P, [&](int64_t p2)
+		{
+			size_t c = (size_t) (q.hcur[p2 + 1] - q.hcur[p2]);
+
+			if (!c)
+				return;
+			std::memcpy(part_k[p2].data() + old[p2],
+				    q.ok + q.hcur[p2], c * 8);
+			std::memcpy(part_v[p2].data() + old[p2],
+				    q.ov + q.hcur[p2], c * 8);
+		});
+	};
+
+	for (int64_t base = 0, it = 0; st == GG_OK && base < n;
+	     base += chunk, it ^= 1)
+	{
+		GbsBuf &q = B[it];
+		int64_t m = (n - base < chunk) ? n - base : chunk;
+
+		drain(q);	/* reclaim this buffer's previous chunk */
+		if (st != GG_OK)
+			break;
+		/* pageable -> pinned bounce (threaded memcpy;
+		 * hipMemcpyAsync from pageable memory would serialize
+		 * in the driver's staging path) */
+		{
+			int64_t nblk = (m + (1 << 20) - 1) >> 20;
+
+			ao_parallel_for(nblk, [&](int64_t blk)
+			{
+				int64_t o = blk << 20;
+				int64_t c = (m - o < (1 << 20))
+					? m - o : (1 << 20);
+
+				std::memcpy(q.hk + o, keys + base + o,
+					    c * 8);
+				std::memcpy(q.hv + o, vals + base + o,
+					    c * 8);
+			});
+		}
+		GG_HIP_GBS(hipMemcpyAsync(q.dk, q.hk, (size_t) m * 8,
+					  hipMemcpyHostToDevice, q.stm));
+		GG_HIP_GBS(hipMemcpyAsync(q.dv, q.hv, (size_t) m * 8,
+					  hipMemcpyHostToDevice, q.stm));
+		GG_HIP_GBS(hipMemsetAsync(q.dcnt, 0, P * 8, q.stm));
+		GG_HIP_GBS(launch_gb_part_count(q.stm, q.dk, m, shift,
+						q.dcnt));
+		GG_HIP_GBS(launch_gb_prefix2_u64(q.stm, q.dcnt, (int) P,
+						 q.dcnt + P,
+						 q.dcnt + 2 * P));
+		GG_HIP_GBS(launch_gb_part_scatter(q.stm, q.dk, q.dv, m,
+						  shift, q.dcnt + P,
+						  q.sk, q.sv));
+		GG_HIP_GBS(hipMemcpyAsync(q.ok, q.sk, (size_t) m * 8,
+					  hipMemcpyDeviceToHost, q.stm));
+		GG_HIP_GBS(hipMemcpyAsync(q.ov, q.sv, (size_t) m * 8,
+					  hipMemcpyDeviceToHost, q.stm));
+		GG_HIP_GBS(hipMemcpyAsync(q.hcur, q.dcnt + 2 * P,
+					  (P + 1) * 8,
+					  hipMemcpyDeviceToHost, q.stm));
+		GG_HIP_GBS(hipEventRecord(q.done, q.stm));
+		q.m = m;
+		q.busy = true;
 	}
+	drain(B[0]);
+	drain(B[1]);
 #undef GG_HIP_GBS
-	(void) hipFree(dk);
-	(void) hipFree(dv);
-	(void) hipFree(sk);
-	(void) hipFree(sv);
-	(void) hipFree(dcnt);
+	for (int b = 0; b < 2; b++)
+	{
+		GbsBuf &q = B[b];
+
+		if (q.stm)
+			(void) hipStreamSynchronize(q.stm);
+		(void) hipFree(q.dk);
+		(void) hipFree(q.dv);
+		(void) hipFree(q.sk);
+		(void) hipFree(q.sv);
+		(void) hipFree(q.dcnt);
+		if (q.hk)
+			(void) hipHostFree(q.hk);
+		if (q.hv)
+			(void) hipHostFree(q.hv);
+		if (q.ok)
+			(void) hipHostFree(q.ok);
+		if (q.ov)
+			(void) hipHostFree(q.ov);
+		if (q.hcur)
+			(void) hipHostFree(q.hcur);
+		if (q.done)
+			(void) hipEventDestroy(q.done);
+		if (q.stm)
+			(void) hipStreamDestroy(q.stm);
+	}
 	if (st != GG_OK)
 		return st;
 

@@ -449,6 +449,10 @@ hipError_t launch_gb_part_scatter_idx(hipStream_t s, const int64_t *keys,
 				      int64_t n, int64_t base, int shift,
 				      unsigned long long *cursors,
 				      int64_t *out_k, int64_t *out_v);
+hipError_t launch_gb_prefix2_u64(hipStream_t s,
+				 const unsigned long long *cnt, int n,
+				 unsigned long long *work,
+				 unsigned long long *pristine);
 hipError_t launch_gb_part_count(hipStream_t s, const int64_t *keys,
 				int64_t n, int shift,
 				unsigned long long *counts);

@@ -2290,6 +2290,39 @@ k_gb_part_scatter(const int64_t *__restrict__ keys,
 	}
 }
 
+/* tiny on-device exclusive prefix over the partition counts: writes a
+ * WORKING cursor copy (consumed by the scatter's atomics) and a
+ * PRISTINE copy (+ total) for the host's staging distribution — keeps
+ * the spill partition pipeline free of mid-chunk host syncs.  P <=
+ * 4096: one thread's serial scan is ~us and fully overlapped. */
+__global__ void
+k_gb_prefix2_u64(const unsigned long long *__restrict__ cnt, int n,
+		 unsigned long long *__restrict__ work,
+		 unsigned long long *__restrict__ pristine)
+{
+	if (blockIdx.x == 0 && threadIdx.x == 0)
+	{
+		unsigned long long s = 0;
+
+		for (int i = 0; i < n; i++)
+		{
+			work[i] = s;
+			pristine[i] = s;
+			s += cnt[i];
+		}
+		pristine[n] = s;
+	}
+}
+
+hipError_t
+launch_gb_prefix2_u64(hipStream_t s, const unsigned long long *cnt, int n,
+		      unsigned long long *work, unsigned long long *pristine)
+{
+	hipLaunchKernelGGL(k_gb_prefix2_u64, dim3(1), dim3(64), 0, s,
+			   cnt, n, work, pristine);
+	return hipGetLastError();
+}
+
 static inline int gb_nparts(int shift)
 {
 	return (int) ((uint64_t) 0xffffffffu >> shift) + 1;
