@@ -4904,46 +4904,257 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 	if (st != GG_OK)
 		return st;
 
-	/* reload + aggregate partition by partition */
+	/* reload + aggregate, pipelined like the partition pass: pinned
+	 * double-buffered uploads; ONE shared hash table (partitions
+	 * serialize on it via an inter-stream event — H2D of partition
+	 * p+1 overlaps the kernels of p, which is where the time was);
+	 * the compact kernel writes its (few) result rows directly to
+	 * pinned host memory; each partition's run is host-sorted
+	 * while the next runs on the GPU, and the sorted runs k-way
+	 * merge at the end (the old single global std::sort of every
+	 * group was serial tail time). */
 	int64_t ng_total = 0;
+	int64_t maxm = 0;
 
 	for (uint64_t p2 = 0; p2 < P; p2++)
+		maxm = std::max(maxm, (int64_t) part_k[p2].size());
+	if (maxm == 0)
 	{
-		if (part_k[p2].empty())
-			continue;
-		int64_t ng = 0;
+		*out_ngroups = 0;
+		*out_npartitions = (int32_t) P;
+		return GG_OK;
+	}
+	{
+		uint64_t rslots = next_pow2(2 * (uint64_t) maxm);
+		unsigned long long *tk = nullptr, *ts = nullptr,
+			*tc = nullptr;
+		hipEvent_t tbl_free = nullptr;
+		struct RlBuf
+		{
+			int64_t *dk = nullptr, *dv = nullptr;
+			int64_t *hk = nullptr, *hv = nullptr;	/* pinned in */
+			int64_t *pok = nullptr, *pos = nullptr,
+				*poc = nullptr;			/* pinned out */
+			unsigned long long *dctr = nullptr;
+			unsigned long long *hng = nullptr;	/* pinned */
+			hipStream_t stm = nullptr;
+			hipEvent_t done = nullptr;
+			bool busy = false;
+		} R[2];
+		std::vector<std::pair<int64_t, int64_t>> runs;	/* off,len */
+		bool aok =
+			hipMalloc((void **) &tk, rslots * 8) == hipSuccess &&
+			hipMalloc((void **) &ts, rslots * 8) == hipSuccess &&
+			hipMalloc((void **) &tc, rslots * 8) == hipSuccess &&
+			hipEventCreate(&tbl_free) == hipSuccess;
 
-		st = gg_engine_hash_groupby_i64(
-			part_k[p2].data(), part_v[p2].data(),
-			(int64_t) part_k[p2].size(), out_keys + ng_total,
-			out_sums + ng_total, out_counts + ng_total,
-			cap - ng_total, &ng);
+		for (int b = 0; b < 2 && aok; b++)
+		{
+			RlBuf &q = R[b];
+
+			aok =
+				hipMalloc((void **) &q.dk, (size_t) maxm * 8) == hipSuccess &&
+				hipMalloc((void **) &q.dv, (size_t) maxm * 8) == hipSuccess &&
+				hipMalloc((void **) &q.dctr, 8) == hipSuccess &&
+				hipHostMalloc((void **) &q.hk, (size_t) maxm * 8) == hipSuccess &&
+				hipHostMalloc((void **) &q.hv, (size_t) maxm * 8) == hipSuccess &&
+				hipHostMalloc((void **) &q.pok, (size_t) maxm * 8) == hipSuccess &&
+				hipHostMalloc((void **) &q.pos, (size_t) maxm * 8) == hipSuccess &&
+				hipHostMalloc((void **) &q.poc, (size_t) maxm * 8) == hipSuccess &&
+				hipHostMalloc((void **) &q.hng, 8) == hipSuccess &&
+				hipStreamCreate(&q.stm) == hipSuccess &&
+				hipEventCreate(&q.done) == hipSuccess;
+		}
+		if (!aok && st == GG_OK)
+			st = fail(GG_ENOMEM, "groupby_spill reload staging");
+
+#define GG_HIP_GBR(x) \
+	{ hipError_t e_ = (x); \
+	  if (st == GG_OK && e_ != hipSuccess) \
+		st = fail(GG_EGPU, "groupby_spill reload: %s", \
+			  hipGetErrorString(e_)); }
+
+		auto rl_finish = [&](RlBuf &q)
+		{
+			if (!q.busy)
+				return;
+			GG_HIP_GBR(hipEventSynchronize(q.done));
+			q.busy = false;
+			if (st != GG_OK)
+				return;
+			int64_t ng = (int64_t) *q.hng;
+
+			if (ng_total + ng > cap)
+			{
+				st = fail(GG_EINVAL,
+					  "groupby cap %lld exceeded",
+					  (long long) cap);
+				return;
+			}
+			/* sort this run by key while the other buffer's
+			 * partition executes on the GPU */
+			std::vector<size_t> idx((size_t) ng);
+
+			for (int64_t i = 0; i < ng; i++)
+				idx[i] = (size_t) i;
+			std::sort(idx.begin(), idx.end(),
+				  [&](size_t a2, size_t b2)
+				  { return q.pok[a2] < q.pok[b2]; });
+			for (int64_t i = 0; i < ng; i++)
+			{
+				out_keys[ng_total + i] = q.pok[idx[i]];
+				out_sums[ng_total + i] = q.pos[idx[i]];
+				out_counts[ng_total + i] = q.poc[idx[i]];
+			}
+			runs.push_back({ng_total, ng});
+			ng_total += ng;
+		};
+
+		for (uint64_t p2 = 0, it = 0; st == GG_OK && p2 < P; p2++)
+		{
+			if (part_k[p2].empty())
+				continue;
+			RlBuf &q = R[it & 1];
+			int64_t m = (int64_t) part_k[p2].size();
+
+			it++;
+			rl_finish(q);
+			if (st != GG_OK)
+				break;
+			{
+				int64_t nblk = (m + (1 << 20) - 1) >> 20;
+
+				ao_parallel_for(nblk, [&](int64_t blk)
+				{
+					int64_t o = blk << 20;
+					int64_t c = (m - o < (1 << 20))
+						? m - o : (1 << 20);
+
+					std::memcpy(q.hk + o,
+						    part_k[p2].data() + o,
+						    c * 8);
+					std::memcpy(q.hv + o,
+						    part_v[p2].data() + o,
+						    c * 8);
+				});
+			}
+			part_k[p2].clear();
+			part_k[p2].shrink_to_fit();
+			part_v[p2].clear();
+			part_v[p2].shrink_to_fit();
+			GG_HIP_GBR(hipMemcpyAsync(q.dk, q.hk,
+						  (size_t) m * 8,
+						  hipMemcpyHostToDevice,
+						  q.stm));
+			GG_HIP_GBR(hipMemcpyAsync(q.dv, q.hv,
+						  (size_t) m * 8,
+						  hipMemcpyHostToDevice,
+						  q.stm));
+			/* the shared table frees when the PREVIOUS
+			 * partition's compact is done */
+			if (it > 1)
+				GG_HIP_GBR(hipStreamWaitEvent(q.stm,
+							      tbl_free, 0));
+			GG_HIP_GBR(launch_fill_u64(q.stm, tk, rslots,
+						   0x8000000000000000ull));
+			GG_HIP_GBR(hipMemsetAsync(ts, 0, rslots * 8,
+						  q.stm));
+			GG_HIP_GBR(hipMemsetAsync(tc, 0, rslots * 8,
+						  q.stm));
+			GG_HIP_GBR(hipMemsetAsync(q.dctr, 0, 8, q.stm));
+			GG_HIP_GBR(launch_groupby_build(q.stm, q.dk, q.dv,
+							m, tk, ts, tc,
+							rslots));
+			GG_HIP_GBR(launch_groupby_compact(
+				q.stm, tk, ts, tc, rslots, q.pok, q.pos,
+				q.poc, q.dctr, (uint64_t) maxm));
+			GG_HIP_GBR(hipEventRecord(tbl_free, q.stm));
+			GG_HIP_GBR(hipMemcpyAsync(q.hng, q.dctr, 8,
+						  hipMemcpyDeviceToHost,
+						  q.stm));
+			GG_HIP_GBR(hipEventRecord(q.done, q.stm));
+			q.busy = true;
+		}
+		rl_finish(R[0]);
+		rl_finish(R[1]);
+#undef GG_HIP_GBR
+		for (int b = 0; b < 2; b++)
+		{
+			RlBuf &q = R[b];
+
+			if (q.stm)
+				(void) hipStreamSynchronize(q.stm);
+			(void) hipFree(q.dk);
+			(void) hipFree(q.dv);
+			(void) hipFree(q.dctr);
+			if (q.hk)
+				(void) hipHostFree(q.hk);
+			if (q.hv)
+				(void) hipHostFree(q.hv);
+			if (q.pok)
+				(void) hipHostFree(q.pok);
+			if (q.pos)
+				(void) hipHostFree(q.pos);
+			if (q.poc)
+				(void) hipHostFree(q.poc);
+			if (q.hng)
+				(void) hipHostFree(q.hng);
+			if (q.done)
+				(void) hipEventDestroy(q.done);
+			if (q.stm)
+				(void) hipStreamDestroy(q.stm);
+		}
+		(void) hipFree(tk);
+		(void) hipFree(ts);
+		(void) hipFree(tc);
+		if (tbl_free)
+			(void) hipEventDestroy(tbl_free);
 		if (st != GG_OK)
 			return st;
-		ng_total += ng;
-		part_k[p2].clear();
-		part_k[p2].shrink_to_fit();
-		part_v[p2].clear();
-		part_v[p2].shrink_to_fit();
-	}
 
-	/* per-partition results are key-sorted; merge to global order */
-	{
-		std::vector<size_t> idx(ng_total);
-		std::vector<int64_t> mk(out_keys, out_keys + ng_total);
-		std::vector<int64_t> ms(out_sums, out_sums + ng_total);
-		std::vector<int64_t> mc(out_counts, out_counts + ng_total);
-
-		for (int64_t i = 0; i < ng_total; i++)
-			idx[i] = (size_t) i;
-		std::sort(idx.begin(), idx.end(),
-			  [&](size_t a2, size_t b2)
-			  { return mk[a2] < mk[b2]; });
-		for (int64_t i = 0; i < ng_total; i++)
+		/* k-way merge of the sorted runs into global key order
+		 * (hash-range partitions interleave in keyspace) */
+		if (runs.size() > 1)
 		{
-			out_keys[i] = mk[idx[i]];
-			out_sums[i] = ms[idx[i]];
-			out_counts[i] = mc[idx[i]];
+			std::vector<int64_t> mk(out_keys,
+						out_keys + ng_total);
+			std::vector<int64_t> ms(out_sums,
+						out_sums + ng_total);
+			std::vector<int64_t> mc(out_counts,
+						out_counts + ng_total);
+			/* min-heap of (key, run); pop -> emit -> push next */
+			std::vector<std::pair<int64_t, size_t>> heap;
+			std::vector<int64_t> pos(runs.size());
+
+			auto cmp = [](const std::pair<int64_t, size_t> &a2,
+				      const std::pair<int64_t, size_t> &b2)
+			{ return a2.first > b2.first; };
+			for (size_t r = 0; r < runs.size(); r++)
+			{
+				pos[r] = runs[r].first;
+				if (runs[r].second)
+					heap.push_back({mk[pos[r]], r});
+			}
+			std::make_heap(heap.begin(), heap.end(), cmp);
+			for (int64_t o = 0; !heap.empty(); o++)
+			{
+				std::pop_heap(heap.begin(), heap.end(),
+					      cmp);
+				size_t r = heap.back().second;
+
+				heap.pop_back();
+				out_keys[o] = mk[pos[r]];
+				out_sums[o] = ms[pos[r]];
+				out_counts[o] = mc[pos[r]];
+				pos[r]++;
+				if (pos[r] <
+				    runs[r].first + runs[r].second)
+				{
+					heap.push_back({mk[pos[r]], r});
+					std::push_heap(heap.begin(),
+						       heap.end(), cmp);
+				}
+			}
 		}
 	}
 	*out_ngroups = ng_total;
