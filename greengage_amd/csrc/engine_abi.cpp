@@ -326,16 +326,18 @@ gg_engine_register_table(const char *name, const gg_column_desc *cols,
  * and codecs run OpenMP-parallel across blocks — the sequential
  * single-thread walk only reads headers. ---- */
 
-/* simple static-partition parallel-for over host threads */
+/* simple dynamic parallel-for over host threads; grain = items per
+ * grab (use 1 for few/large items — e.g. per-partition memcpys — or
+ * the work degenerates to a couple of threads) */
 template <typename F>
 static void
-ao_parallel_for(int64_t n, F fn)
+par_for_grain(int64_t n, int64_t grain, F fn)
 {
 	unsigned nt = std::thread::hardware_concurrency();
 
 	if (nt > 32)
 		nt = 32;
-	if (nt < 2 || n < 64)
+	if (nt < 2 || n < grain || n < 2)
 	{
 		for (int64_t i = 0; i < n; i++)
 			fn(i);
@@ -349,11 +351,11 @@ ao_parallel_for(int64_t n, F fn)
 		{
 			for (;;)
 			{
-				int64_t i = next.fetch_add(64);
+				int64_t i = next.fetch_add(grain);
 
 				if (i >= n)
 					return;
-				int64_t hi = i + 64 < n ? i + 64 : n;
+				int64_t hi = i + grain < n ? i + grain : n;
 
 				for (; i < hi; i++)
 					fn(i);
@@ -361,6 +363,14 @@ ao_parallel_for(int64_t n, F fn)
 		});
 	for (auto &th : ths)
 		th.join();
+}
+
+/* legacy grain (many small items: AO block jobs) */
+template <typename F>
+static void
+ao_parallel_for(int64_t n, F fn)
+{
+	par_for_grain(n, 64, fn);
 }
 
 struct AoDesc
@@ -4808,7 +4818,7 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 				part_v[p2].resize(old[p2] + c);
 			}
 		}
-		ao_parallel_for((int64_t) P, [&](int64_t p2)
+		par_for_grain((int64_t) P, 1, [&](int64_t p2)
 		{
 			size_t c = (size_t) (q.hcur[p2 + 1] - q.hcur[p2]);
 
@@ -4834,13 +4844,13 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 		 * hipMemcpyAsync from pageable memory would serialize
 		 * in the driver's staging path) */
 		{
-			int64_t nblk = (m + (1 << 20) - 1) >> 20;
+			int64_t nblk = (m + (1 << 17) - 1) >> 17;
 
-			ao_parallel_for(nblk, [&](int64_t blk)
+			par_for_grain(nblk, 1, [&](int64_t blk)
 			{
-				int64_t o = blk << 20;
-				int64_t c = (m - o < (1 << 20))
-					? m - o : (1 << 20);
+				int64_t o = blk << 17;
+				int64_t c = (m - o < (1 << 17))
+					? m - o : (1 << 17);
 
 				std::memcpy(q.hk + o, keys + base + o,
 					    c * 8);
@@ -5022,13 +5032,13 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 			if (st != GG_OK)
 				break;
 			{
-				int64_t nblk = (m + (1 << 20) - 1) >> 20;
+				int64_t nblk = (m + (1 << 17) - 1) >> 17;
 
-				ao_parallel_for(nblk, [&](int64_t blk)
+				par_for_grain(nblk, 1, [&](int64_t blk)
 				{
-					int64_t o = blk << 20;
-					int64_t c = (m - o < (1 << 20))
-						? m - o : (1 << 20);
+					int64_t o = blk << 17;
+					int64_t c = (m - o < (1 << 17))
+						? m - o : (1 << 17);
 
 					std::memcpy(q.hk + o,
 						    part_k[p2].data() + o,
@@ -5194,15 +5204,126 @@ gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,
 	GG_HIP(hipMalloc((void **) &os, (size_t) n * 8));
 	GG_HIP(hipMalloc((void **) &oc, (size_t) n * 8));
 	GG_HIP(hipMalloc((void **) &ctr, 8));
-	GG_HIP(hipMemcpy(dk, keys, (size_t) n * 8, hipMemcpyHostToDevice));
-	GG_HIP(hipMemcpy(dv, vals, (size_t) n * 8, hipMemcpyHostToDevice));
 	/* the empty sentinel (INT64_MIN) is not a memset byte pattern —
 	 * fill with a kernel */
 	GG_HIP(launch_fill_u64(e.stream, tk, nslots, 0x8000000000000000ull));
 	GG_HIP(hipMemsetAsync(ts, 0, nslots * 8, e.stream));
 	GG_HIP(hipMemsetAsync(tc, 0, nslots * 8, e.stream));
 	GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
-	GG_HIP(launch_groupby_build(e.stream, dk, dv, n, tk, ts, tc, nslots));
+	/* chunked pinned upload overlapped with the build: the build
+	 * kernel is per-row independent (CAS find-or-create + atomic
+	 * transitions), so it runs on each chunk as it lands — a
+	 * single pageable hipMemcpy of the whole input was measured
+	 * bound at ~6 GB/s and serialized ahead of the build */
+	{
+		const int64_t ch = 1 << 24;	/* 16M rows = 128 MB */
+		int64_t *bk_ = nullptr, *bv_ = nullptr;
+		gg_status st2 = GG_OK;
+
+		if (n > (1 << 22) &&
+		    hipHostMalloc((void **) &bk_, 2 * ch * 8) == hipSuccess &&
+		    hipHostMalloc((void **) &bv_, 2 * ch * 8) == hipSuccess)
+		{
+			hipEvent_t up[2] = {};
+			hipStream_t cstream = nullptr;
+
+			(void) hipEventCreate(&up[0]);
+			(void) hipEventCreate(&up[1]);
+			(void) hipStreamCreate(&cstream);
+			for (int64_t base = 0, it = 0; base < n;
+			     base += ch, it ^= 1)
+			{
+				int64_t m = (n - base < ch) ? n - base : ch;
+				int64_t *pk_ = bk_ + it * ch;
+				int64_t *pv_ = bv_ + it * ch;
+
+				/* buffer free once its H2D completed */
+				if (base >= 2 * ch)
+					(void) hipEventSynchronize(up[it]);
+				{
+					int64_t nblk =
+						(m + (1 << 17) - 1) >> 17;
+
+					par_for_grain(nblk, 1,
+						      [&](int64_t blk)
+					{
+						int64_t o = blk << 17;
+						int64_t c =
+							(m - o < (1 << 17))
+							? m - o : (1 << 17);
+
+						std::memcpy(pk_ + o,
+							    keys + base + o,
+							    c * 8);
+						std::memcpy(pv_ + o,
+							    vals + base + o,
+							    c * 8);
+					});
+				}
+				/* copies on their own stream: chunk
+				 * i+1's upload overlaps chunk i's
+				 * build kernel */
+				hipError_t e1 = hipMemcpyAsync(
+					dk + base, pk_, (size_t) m * 8,
+					hipMemcpyHostToDevice, cstream);
+				hipError_t e2 = hipMemcpyAsync(
+					dv + base, pv_, (size_t) m * 8,
+					hipMemcpyHostToDevice, cstream);
+
+				if (e1 != hipSuccess || e2 != hipSuccess)
+				{
+					st2 = fail(GG_EGPU,
+						   "groupby upload");
+					break;
+				}
+				(void) hipEventRecord(up[it], cstream);
+				(void) hipStreamWaitEvent(e.stream,
+							  up[it], 0);
+				if (launch_groupby_build(
+					e.stream, dk + base, dv + base, m,
+					tk, ts, tc, nslots) != hipSuccess)
+				{
+					st2 = fail(GG_EGPU,
+						   "groupby build");
+					break;
+				}
+			}
+			(void) hipStreamSynchronize(cstream);
+			(void) hipStreamSynchronize(e.stream);
+			(void) hipEventDestroy(up[0]);
+			(void) hipEventDestroy(up[1]);
+			(void) hipStreamDestroy(cstream);
+			(void) hipHostFree(bk_);
+			(void) hipHostFree(bv_);
+		}
+		else
+		{
+			if (bk_)
+				(void) hipHostFree(bk_);
+			if (bv_)
+				(void) hipHostFree(bv_);
+			hipError_t e1 = hipMemcpy(dk, keys, (size_t) n * 8,
+						  hipMemcpyHostToDevice);
+			hipError_t e2 = hipMemcpy(dv, vals, (size_t) n * 8,
+						  hipMemcpyHostToDevice);
+
+			if (e1 != hipSuccess || e2 != hipSuccess)
+				st2 = fail(GG_EGPU, "groupby H2D");
+			else if (launch_groupby_build(e.stream, dk, dv, n,
+						      tk, ts, tc, nslots)
+				 != hipSuccess)
+				st2 = fail(GG_EGPU, "groupby build");
+		}
+		if (st2 != GG_OK)
+		{
+			(void) hipFree(dk); (void) hipFree(dv);
+			(void) hipFree(tk); (void) hipFree(ts);
+			(void) hipFree(tc); (void) hipFree(ok);
+			(void) hipFree(os); (void) hipFree(oc);
+			(void) hipFree(ctr);
+			return st2;
+		}
+	}
 	GG_HIP(launch_groupby_compact(e.stream, tk, ts, tc, nslots, ok, os,
 				      oc, ctr, (uint64_t) n));
 	GG_HIP(hipStreamSynchronize(e.stream));
