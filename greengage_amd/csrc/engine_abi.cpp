@@ -12,6 +12,7 @@
 #include <thread>
 #include <cstdarg>
 #include <cstdio>
+#include <chrono>
 #include <cstring>
 #include <vector>
 
@@ -246,6 +247,9 @@ extern "C" gg_status gg_engine_shutdown(void)
 	}
 	e.tables.clear();
 	e.mm_cache.clear();
+	for (auto &kv : e.escratch)
+		(void) hipFree(kv.second.first);
+	e.escratch.clear();
 	for (auto *p : e.pipelines)
 		delete p;
 	e.pipelines.clear();
@@ -4751,6 +4755,18 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 		chunk = n;
 	std::vector<std::vector<int64_t>> part_k(P), part_v(P);
 	gg_status st = GG_OK;
+	const bool spdbg = getenv("GG_SPILL_DEBUG") != nullptr;
+	auto spt0 = std::chrono::steady_clock::now();
+	auto spmark = [&](const char *what)
+	{
+		if (!spdbg)
+			return;
+		auto t1 = std::chrono::steady_clock::now();
+
+		fprintf(stderr, "[spill] %s: %.2fs\n", what,
+			std::chrono::duration<double>(t1 - spt0).count());
+		spt0 = t1;
+	};
 
 	struct GbsBuf
 	{
@@ -4884,6 +4900,7 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 	}
 	drain(B[0]);
 	drain(B[1]);
+	spmark("partition pass");
 #undef GG_HIP_GBS
 	for (int b = 0; b < 2; b++)
 	{
@@ -5087,6 +5104,7 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 		}
 		rl_finish(R[0]);
 		rl_finish(R[1]);
+		spmark("reload");
 #undef GG_HIP_GBR
 		for (int b = 0; b < 2; b++)
 		{
@@ -5167,6 +5185,7 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 			}
 		}
 	}
+	spmark("merge");
 	*out_ngroups = ng_total;
 	*out_npartitions = (int32_t) P;
 	return GG_OK;
@@ -5190,20 +5209,26 @@ gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,
 		return GG_OK;
 
 	uint64_t nslots = next_pow2(2 * (uint64_t) n);
-	int64_t *dk = nullptr, *dv = nullptr;
-	unsigned long long *tk = nullptr, *ts = nullptr, *tc = nullptr;
-	int64_t *ok = nullptr, *os = nullptr, *oc = nullptr;
-	unsigned long long *ctr = nullptr;
+	/* engine scratch pool: repeated calls at these sizes re-map
+	 * ~6x n x 8 + 3 x nslots x 8 bytes of HBM per call otherwise
+	 * (tens of GB of hipMalloc/hipFree at 1B rows — measured as
+	 * the dominant in-memory cost, not the copies) */
+	int64_t *dk = (int64_t *) e.esget("gb.dk", (size_t) n * 8);
+	int64_t *dv = (int64_t *) e.esget("gb.dv", (size_t) n * 8);
+	unsigned long long *tk = (unsigned long long *)
+		e.esget("gb.tk", nslots * 8);
+	unsigned long long *ts = (unsigned long long *)
+		e.esget("gb.ts", nslots * 8);
+	unsigned long long *tc = (unsigned long long *)
+		e.esget("gb.tc", nslots * 8);
+	int64_t *ok = (int64_t *) e.esget("gb.ok", (size_t) n * 8);
+	int64_t *os = (int64_t *) e.esget("gb.os", (size_t) n * 8);
+	int64_t *oc = (int64_t *) e.esget("gb.oc", (size_t) n * 8);
+	unsigned long long *ctr = (unsigned long long *)
+		e.esget("gb.ctr", 8);
 
-	GG_HIP(hipMalloc((void **) &dk, (size_t) n * 8));
-	GG_HIP(hipMalloc((void **) &dv, (size_t) n * 8));
-	GG_HIP(hipMalloc((void **) &tk, nslots * 8));
-	GG_HIP(hipMalloc((void **) &ts, nslots * 8));
-	GG_HIP(hipMalloc((void **) &tc, nslots * 8));
-	GG_HIP(hipMalloc((void **) &ok, (size_t) n * 8));
-	GG_HIP(hipMalloc((void **) &os, (size_t) n * 8));
-	GG_HIP(hipMalloc((void **) &oc, (size_t) n * 8));
-	GG_HIP(hipMalloc((void **) &ctr, 8));
+	if (!dk || !dv || !tk || !ts || !tc || !ok || !os || !oc || !ctr)
+		return fail(GG_ENOMEM, "groupby buffers");
 	/* the empty sentinel (INT64_MIN) is not a memset byte pattern —
 	 * fill with a kernel */
 	GG_HIP(launch_fill_u64(e.stream, tk, nslots, 0x8000000000000000ull));
@@ -5315,14 +5340,7 @@ gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,
 				st2 = fail(GG_EGPU, "groupby build");
 		}
 		if (st2 != GG_OK)
-		{
-			(void) hipFree(dk); (void) hipFree(dv);
-			(void) hipFree(tk); (void) hipFree(ts);
-			(void) hipFree(tc); (void) hipFree(ok);
-			(void) hipFree(os); (void) hipFree(oc);
-			(void) hipFree(ctr);
-			return st2;
-		}
+			return st2;	/* pooled buffers stay owned */
 	}
 	GG_HIP(launch_groupby_compact(e.stream, tk, ts, tc, nslots, ok, os,
 				      oc, ctr, (uint64_t) n));
@@ -5357,15 +5375,6 @@ gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,
 		}
 		*out_ngroups = (int64_t) ng;
 	}
-	(void) hipFree(dk);
-	(void) hipFree(dv);
-	(void) hipFree(tk);
-	(void) hipFree(ts);
-	(void) hipFree(tc);
-	(void) hipFree(ok);
-	(void) hipFree(os);
-	(void) hipFree(oc);
-	(void) hipFree(ctr);
 	return st;
 }
 

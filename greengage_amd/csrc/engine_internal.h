@@ -191,6 +191,33 @@ struct Engine
 	/* per-column min/max cache (registered columns are immutable);
 	 * keyed by device pointer */
 	std::map<const void *, std::pair<long long, long long>> mm_cache;
+	/* engine-level grow-only device scratch (standalone ABI calls
+	 * like hash_groupby_i64: repeated calls reuse instead of
+	 * re-mapping tens of GB per call); freed at shutdown */
+	std::vector<std::pair<std::string, std::pair<void *, size_t>>> escratch;
+
+	void *esget(const char *name, size_t bytes)
+	{
+		for (auto &kv : escratch)
+			if (kv.first == name)
+			{
+				if (kv.second.second >= bytes)
+					return kv.second.first;
+				(void) hipFree(kv.second.first);
+				kv.second.first = nullptr;
+				if (hipMalloc(&kv.second.first, bytes)
+				    != hipSuccess)
+					return nullptr;
+				kv.second.second = bytes;
+				return kv.second.first;
+			}
+		void *p = nullptr;
+
+		if (hipMalloc(&p, bytes ? bytes : 1) != hipSuccess)
+			return nullptr;
+		escratch.push_back({name, {p, bytes}});
+		return p;
+	}
 };
 
 Engine &engine();

@@ -1,6 +1,6 @@
 #!/bin/bash
 cd /root/repo
-timeout 800 python - <<'PY' > gpurun_out/spill_bench.log 2>&1
+GG_SPILL_DEBUG=1 timeout 800 python - <<'PY' > gpurun_out/spill_bench.log 2>&1
 import sys, time
 sys.path.insert(0, ".")
 import numpy as np
@@ -14,7 +14,8 @@ keys = rng.integers(0, 5_000_000, n).astype(np.int64)
 vals = rng.integers(-100, 100, n).astype(np.int64)
 print(f"input {n:,} rows = {n*16/1e9:.0f} GB of pairs")
 for budget, label in ((1 << 30, "1 GiB budget (spill)"),
-                      (1 << 39, "in-memory")):
+                      (1 << 39, "in-memory (cold pool)"),
+                      (1 << 39, "in-memory (warm pool)")):
     t0 = time.time()
     k, s, c, nparts = E.hash_groupby_spill(keys, vals, budget)
     t1 = time.time()
@@ -27,4 +28,4 @@ assert np.array_equal(k1, k2) and np.array_equal(s1, s2) and np.array_equal(c1, 
 print("spill == in-memory on 50M subset: OK")
 eng.shutdown()
 PY
-tail -6 gpurun_out/spill_bench.log
+tail -12 gpurun_out/spill_bench.log
