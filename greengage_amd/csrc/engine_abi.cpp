@@ -250,6 +250,9 @@ extern "C" gg_status gg_engine_shutdown(void)
 	for (auto &kv : e.escratch)
 		(void) hipFree(kv.second.first);
 	e.escratch.clear();
+	for (auto &kv : e.ehscratch)
+		(void) hipHostFree(kv.second.first);
+	e.ehscratch.clear();
 	for (auto *p : e.pipelines)
 		delete p;
 	e.pipelines.clear();
@@ -4786,20 +4789,23 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 	for (int b = 0; b < 2 && alloc_ok; b++)
 	{
 		GbsBuf &q = B[b];
+		char nm[32];
 
+#define GBS_POOL(fld, kind, sz) 		(std::snprintf(nm, sizeof(nm), "gbs.%s%d", #fld, b), 		 (q.fld = (decltype(q.fld)) e.kind(nm, (sz))) != nullptr)
 		alloc_ok =
-			hipMalloc((void **) &q.dk, (size_t) chunk * 8) == hipSuccess &&
-			hipMalloc((void **) &q.dv, (size_t) chunk * 8) == hipSuccess &&
-			hipMalloc((void **) &q.sk, (size_t) chunk * 8) == hipSuccess &&
-			hipMalloc((void **) &q.sv, (size_t) chunk * 8) == hipSuccess &&
-			hipMalloc((void **) &q.dcnt, (3 * P + 1) * 8) == hipSuccess &&
-			hipHostMalloc((void **) &q.hk, (size_t) chunk * 8) == hipSuccess &&
-			hipHostMalloc((void **) &q.hv, (size_t) chunk * 8) == hipSuccess &&
-			hipHostMalloc((void **) &q.ok, (size_t) chunk * 8) == hipSuccess &&
-			hipHostMalloc((void **) &q.ov, (size_t) chunk * 8) == hipSuccess &&
-			hipHostMalloc((void **) &q.hcur, (P + 1) * 8) == hipSuccess &&
+			GBS_POOL(dk, esget, (size_t) chunk * 8) &&
+			GBS_POOL(dv, esget, (size_t) chunk * 8) &&
+			GBS_POOL(sk, esget, (size_t) chunk * 8) &&
+			GBS_POOL(sv, esget, (size_t) chunk * 8) &&
+			GBS_POOL(dcnt, esget, (3 * P + 1) * 8) &&
+			GBS_POOL(hk, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(hv, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(ok, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(ov, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(hcur, ehget, (P + 1) * 8) &&
 			hipStreamCreate(&q.stm) == hipSuccess &&
 			hipEventCreate(&q.done) == hipSuccess;
+#undef GBS_POOL
 	}
 	if (!alloc_ok)
 		st = fail(GG_ENOMEM, "groupby_spill staging");
@@ -4830,8 +4836,30 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 			old[p2] = part_k[p2].size();
 			if (c)
 			{
-				part_k[p2].resize(old[p2] + c);
-				part_v[p2].resize(old[p2] + c);
+				size_t need = old[p2] + c;
+
+				/* resize() grows to EXACTLY the request
+				 * in libstdc++ — per-chunk exact growth
+				 * re-copied every partition every chunk
+				 * (hundreds of GB of hidden memcpy at
+				 * 1B rows, measured 7.8s).  Reserve the
+				 * uniform-hash expectation up front and
+				 * grow geometrically past it. */
+				if (part_k[p2].capacity() < need)
+				{
+					size_t want = std::max(
+						need,
+						std::max(part_k[p2]
+							 .capacity() * 2,
+							 (size_t) (n / P +
+								   n / P / 8 +
+								   1024)));
+
+					part_k[p2].reserve(want);
+					part_v[p2].reserve(want);
+				}
+				part_k[p2].resize(need);
+				part_v[p2].resize(need);
 			}
 		}
 		par_for_grain((int64_t) P, 1, [&](int64_t p2)
@@ -4908,21 +4936,6 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 
 		if (q.stm)
 			(void) hipStreamSynchronize(q.stm);
-		(void) hipFree(q.dk);
-		(void) hipFree(q.dv);
-		(void) hipFree(q.sk);
-		(void) hipFree(q.sv);
-		(void) hipFree(q.dcnt);
-		if (q.hk)
-			(void) hipHostFree(q.hk);
-		if (q.hv)
-			(void) hipHostFree(q.hv);
-		if (q.ok)
-			(void) hipHostFree(q.ok);
-		if (q.ov)
-			(void) hipHostFree(q.ov);
-		if (q.hcur)
-			(void) hipHostFree(q.hcur);
 		if (q.done)
 			(void) hipEventDestroy(q.done);
 		if (q.stm)
@@ -4970,27 +4983,33 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 		} R[2];
 		std::vector<std::pair<int64_t, int64_t>> runs;	/* off,len */
 		bool aok =
-			hipMalloc((void **) &tk, rslots * 8) == hipSuccess &&
-			hipMalloc((void **) &ts, rslots * 8) == hipSuccess &&
-			hipMalloc((void **) &tc, rslots * 8) == hipSuccess &&
+			(tk = (unsigned long long *)
+			 e.esget("rl.tk", rslots * 8)) != nullptr &&
+			(ts = (unsigned long long *)
+			 e.esget("rl.ts", rslots * 8)) != nullptr &&
+			(tc = (unsigned long long *)
+			 e.esget("rl.tc", rslots * 8)) != nullptr &&
 			hipEventCreate(&tbl_free) == hipSuccess;
 
 		for (int b = 0; b < 2 && aok; b++)
 		{
 			RlBuf &q = R[b];
+			char nm[32];
 
+#define RL_POOL(fld, kind, sz) 			(std::snprintf(nm, sizeof(nm), "rl.%s%d", #fld, b), 			 (q.fld = (decltype(q.fld)) e.kind(nm, (sz))) != nullptr)
 			aok =
-				hipMalloc((void **) &q.dk, (size_t) maxm * 8) == hipSuccess &&
-				hipMalloc((void **) &q.dv, (size_t) maxm * 8) == hipSuccess &&
-				hipMalloc((void **) &q.dctr, 8) == hipSuccess &&
-				hipHostMalloc((void **) &q.hk, (size_t) maxm * 8) == hipSuccess &&
-				hipHostMalloc((void **) &q.hv, (size_t) maxm * 8) == hipSuccess &&
-				hipHostMalloc((void **) &q.pok, (size_t) maxm * 8) == hipSuccess &&
-				hipHostMalloc((void **) &q.pos, (size_t) maxm * 8) == hipSuccess &&
-				hipHostMalloc((void **) &q.poc, (size_t) maxm * 8) == hipSuccess &&
-				hipHostMalloc((void **) &q.hng, 8) == hipSuccess &&
+				RL_POOL(dk, esget, (size_t) maxm * 8) &&
+				RL_POOL(dv, esget, (size_t) maxm * 8) &&
+				RL_POOL(dctr, esget, 8) &&
+				RL_POOL(hk, ehget, (size_t) maxm * 8) &&
+				RL_POOL(hv, ehget, (size_t) maxm * 8) &&
+				RL_POOL(pok, ehget, (size_t) maxm * 8) &&
+				RL_POOL(pos, ehget, (size_t) maxm * 8) &&
+				RL_POOL(poc, ehget, (size_t) maxm * 8) &&
+				RL_POOL(hng, ehget, 8) &&
 				hipStreamCreate(&q.stm) == hipSuccess &&
 				hipEventCreate(&q.done) == hipSuccess;
+#undef RL_POOL
 		}
 		if (!aok && st == GG_OK)
 			st = fail(GG_ENOMEM, "groupby_spill reload staging");
@@ -5112,29 +5131,11 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 
 			if (q.stm)
 				(void) hipStreamSynchronize(q.stm);
-			(void) hipFree(q.dk);
-			(void) hipFree(q.dv);
-			(void) hipFree(q.dctr);
-			if (q.hk)
-				(void) hipHostFree(q.hk);
-			if (q.hv)
-				(void) hipHostFree(q.hv);
-			if (q.pok)
-				(void) hipHostFree(q.pok);
-			if (q.pos)
-				(void) hipHostFree(q.pos);
-			if (q.poc)
-				(void) hipHostFree(q.poc);
-			if (q.hng)
-				(void) hipHostFree(q.hng);
 			if (q.done)
 				(void) hipEventDestroy(q.done);
 			if (q.stm)
 				(void) hipStreamDestroy(q.stm);
 		}
-		(void) hipFree(tk);
-		(void) hipFree(ts);
-		(void) hipFree(tc);
 		if (tbl_free)
 			(void) hipEventDestroy(tbl_free);
 		if (st != GG_OK)
@@ -5246,8 +5247,10 @@ gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,
 		gg_status st2 = GG_OK;
 
 		if (n > (1 << 22) &&
-		    hipHostMalloc((void **) &bk_, 2 * ch * 8) == hipSuccess &&
-		    hipHostMalloc((void **) &bv_, 2 * ch * 8) == hipSuccess)
+		    (bk_ = (int64_t *) e.ehget("gb.hbk", 2 * ch * 8))
+		    != nullptr &&
+		    (bv_ = (int64_t *) e.ehget("gb.hbv", 2 * ch * 8))
+		    != nullptr)
 		{
 			hipEvent_t up[2] = {};
 			hipStream_t cstream = nullptr;
@@ -5318,15 +5321,9 @@ gg_engine_hash_groupby_i64(const int64_t *keys, const int64_t *vals,
 			(void) hipEventDestroy(up[0]);
 			(void) hipEventDestroy(up[1]);
 			(void) hipStreamDestroy(cstream);
-			(void) hipHostFree(bk_);
-			(void) hipHostFree(bv_);
 		}
 		else
 		{
-			if (bk_)
-				(void) hipHostFree(bk_);
-			if (bv_)
-				(void) hipHostFree(bv_);
 			hipError_t e1 = hipMemcpy(dk, keys, (size_t) n * 8,
 						  hipMemcpyHostToDevice);
 			hipError_t e2 = hipMemcpy(dv, vals, (size_t) n * 8,

@@ -218,6 +218,33 @@ struct Engine
 		escratch.push_back({name, {p, bytes}});
 		return p;
 	}
+
+	/* same, for pinned host memory (hipHostMalloc is ~1 GB/s of
+	 * page-locking — reuse across calls) */
+	std::vector<std::pair<std::string, std::pair<void *, size_t>>> ehscratch;
+
+	void *ehget(const char *name, size_t bytes)
+	{
+		for (auto &kv : ehscratch)
+			if (kv.first == name)
+			{
+				if (kv.second.second >= bytes)
+					return kv.second.first;
+				(void) hipHostFree(kv.second.first);
+				kv.second.first = nullptr;
+				if (hipHostMalloc(&kv.second.first, bytes)
+				    != hipSuccess)
+					return nullptr;
+				kv.second.second = bytes;
+				return kv.second.first;
+			}
+		void *p = nullptr;
+
+		if (hipHostMalloc(&p, bytes ? bytes : 1) != hipSuccess)
+			return nullptr;
+		ehscratch.push_back({name, {p, bytes}});
+		return p;
+	}
 };
 
 Engine &engine();
