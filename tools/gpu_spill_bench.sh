@@ -13,7 +13,8 @@ rng = np.random.default_rng(11)
 keys = rng.integers(0, 5_000_000, n).astype(np.int64)
 vals = rng.integers(-100, 100, n).astype(np.int64)
 print(f"input {n:,} rows = {n*16/1e9:.0f} GB of pairs")
-for budget, label in ((1 << 30, "1 GiB budget (spill)"),
+for budget, label in ((1 << 30, "1 GiB budget (spill, cold pool)"),
+                      (1 << 30, "1 GiB budget (spill, warm pool)"),
                       (1 << 39, "in-memory (cold pool)"),
                       (1 << 39, "in-memory (warm pool)")):
     t0 = time.time()
@@ -28,4 +29,4 @@ assert np.array_equal(k1, k2) and np.array_equal(s1, s2) and np.array_equal(c1, 
 print("spill == in-memory on 50M subset: OK")
 eng.shutdown()
 PY
-tail -12 gpurun_out/spill_bench.log
+tail -16 gpurun_out/spill_bench.log
