@@ -4470,6 +4470,213 @@ gg_engine_motion_dechunkify(const uint8_t *chunks, int64_t chunks_len,
  * unique (PK-side build, as in the engine's pipeline joins); output
  * is (probe_row_index, build_val) for every probe row with a match.
  */
+/* pipelined hash-range partition stage shared by the spill tiers:
+ * pinned double-buffered staging both directions, cursor prefix on
+ * device (k_gb_prefix2_u64 — no mid-chunk host sync), one bulk D2H
+ * per chunk, host distribution overlapped with the other buffer's
+ * GPU work.  vals == nullptr selects the index side (out_v receives
+ * GLOBAL row indices via the scatter_idx kernel).  All staging comes
+ * from the engine pools. */
+static gg_status
+spill_stage_partitions(Engine &e, const int64_t *keys, const int64_t *vals,
+		       int64_t n, int64_t budget_bytes, uint64_t P,
+		       int shift,
+		       std::vector<std::vector<int64_t>> &out_k,
+		       std::vector<std::vector<int64_t>> &out_v)
+{
+	int64_t chunk = budget_bytes / (8 * 4 * 2);
+	gg_status st = GG_OK;
+
+	if (chunk < 1024)
+		chunk = 1024;
+	if (chunk > n)
+		chunk = n;
+	if (n <= 0)
+		return GG_OK;
+
+	struct GbsBuf
+	{
+		int64_t *dk = nullptr, *dv = nullptr;
+		int64_t *sk = nullptr, *sv = nullptr;
+		unsigned long long *dcnt = nullptr;	/* [P cnt][P work][P+1 pristine] */
+		int64_t *hk = nullptr, *hv = nullptr;	/* pinned in */
+		int64_t *ok = nullptr, *ov = nullptr;	/* pinned out */
+		unsigned long long *hcur = nullptr;	/* pinned P+1 */
+		hipStream_t stm = nullptr;
+		hipEvent_t done = nullptr;
+		bool busy = false;
+	} B[2];
+	bool alloc_ok = true;
+
+	for (int b = 0; b < 2 && alloc_ok; b++)
+	{
+		GbsBuf &q = B[b];
+		char nm[32];
+
+#define GBS_POOL(fld, kind, sz) (std::snprintf(nm, sizeof(nm), "gbs.%s%d", #fld, b), (q.fld = (decltype(q.fld)) e.kind(nm, (sz))) != nullptr)
+		alloc_ok =
+			GBS_POOL(dk, esget, (size_t) chunk * 8) &&
+			GBS_POOL(dv, esget, (size_t) chunk * 8) &&
+			GBS_POOL(sk, esget, (size_t) chunk * 8) &&
+			GBS_POOL(sv, esget, (size_t) chunk * 8) &&
+			GBS_POOL(dcnt, esget, (3 * P + 1) * 8) &&
+			GBS_POOL(hk, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(hv, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(ok, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(ov, ehget, (size_t) chunk * 8) &&
+			GBS_POOL(hcur, ehget, (P + 1) * 8) &&
+			hipStreamCreate(&q.stm) == hipSuccess &&
+			hipEventCreate(&q.done) == hipSuccess;
+#undef GBS_POOL
+	}
+	if (!alloc_ok)
+		st = fail(GG_ENOMEM, "spill staging");
+
+	/* error policy: record first failure, stop issuing, drain */
+#define GG_HIP_GBS(x) \
+	{ hipError_t e_ = (x); \
+	  if (st == GG_OK && e_ != hipSuccess) \
+		st = fail(GG_EGPU, "spill stage: %s", \
+			  hipGetErrorString(e_)); }
+
+	auto drain = [&](GbsBuf &q)
+	{
+		if (!q.busy)
+			return;
+		GG_HIP_GBS(hipEventSynchronize(q.done));
+		q.busy = false;
+		if (st != GG_OK)
+			return;
+		/* append staged rows to their partitions; resizes are
+		 * serial, the copies partition-parallel */
+		std::vector<size_t> old(P);
+
+		for (uint64_t p2 = 0; p2 < P; p2++)
+		{
+			size_t c = (size_t) (q.hcur[p2 + 1] - q.hcur[p2]);
+
+			old[p2] = out_k[p2].size();
+			if (c)
+			{
+				size_t need = old[p2] + c;
+
+				/* resize() grows to EXACTLY the request
+				 * in libstdc++ — per-chunk exact growth
+				 * re-copied every partition every chunk
+				 * (hundreds of GB of hidden memcpy at
+				 * 1B rows, measured 7.8s).  Reserve the
+				 * uniform-hash expectation up front and
+				 * grow geometrically past it. */
+				if (out_k[p2].capacity() < need)
+				{
+					size_t want = std::max(
+						need,
+						std::max(out_k[p2]
+							 .capacity() * 2,
+							 (size_t) (n / P +
+								   n / P / 8 +
+								   1024)));
+
+					out_k[p2].reserve(want);
+					out_v[p2].reserve(want);
+				}
+				out_k[p2].resize(need);
+				out_v[p2].resize(need);
+			}
+		}
+		par_for_grain((int64_t) P, 1, [&](int64_t p2)
+		{
+			size_t c = (size_t) (q.hcur[p2 + 1] - q.hcur[p2]);
+
+			if (!c)
+				return;
+			std::memcpy(out_k[p2].data() + old[p2],
+				    q.ok + q.hcur[p2], c * 8);
+			std::memcpy(out_v[p2].data() + old[p2],
+				    q.ov + q.hcur[p2], c * 8);
+		});
+	};
+
+	for (int64_t base = 0, it = 0; st == GG_OK && base < n;
+	     base += chunk, it ^= 1)
+	{
+		GbsBuf &q = B[it];
+		int64_t m = (n - base < chunk) ? n - base : chunk;
+
+		drain(q);	/* reclaim this buffer's previous chunk */
+		if (st != GG_OK)
+			break;
+		/* pageable -> pinned bounce (threaded memcpy;
+		 * hipMemcpyAsync from pageable memory would serialize
+		 * in the driver's staging path) */
+		{
+			int64_t nblk = (m + (1 << 17) - 1) >> 17;
+
+			par_for_grain(nblk, 1, [&](int64_t blk)
+			{
+				int64_t o = blk << 17;
+				int64_t c = (m - o < (1 << 17))
+					? m - o : (1 << 17);
+
+				std::memcpy(q.hk + o, keys + base + o,
+					    c * 8);
+				if (vals)
+					std::memcpy(q.hv + o,
+						    vals + base + o, c * 8);
+			});
+		}
+		GG_HIP_GBS(hipMemcpyAsync(q.dk, q.hk, (size_t) m * 8,
+					  hipMemcpyHostToDevice, q.stm));
+		if (vals)
+			GG_HIP_GBS(hipMemcpyAsync(q.dv, q.hv,
+						  (size_t) m * 8,
+						  hipMemcpyHostToDevice,
+						  q.stm));
+		GG_HIP_GBS(hipMemsetAsync(q.dcnt, 0, P * 8, q.stm));
+		GG_HIP_GBS(launch_gb_part_count(q.stm, q.dk, m, shift,
+						q.dcnt));
+		GG_HIP_GBS(launch_gb_prefix2_u64(q.stm, q.dcnt, (int) P,
+						 q.dcnt + P,
+						 q.dcnt + 2 * P));
+		if (vals)
+		{
+			GG_HIP_GBS(launch_gb_part_scatter(
+				q.stm, q.dk, q.dv, m, shift, q.dcnt + P,
+				q.sk, q.sv));
+		}
+		else
+		{
+			GG_HIP_GBS(launch_gb_part_scatter_idx(
+				q.stm, q.dk, m, base, shift, q.dcnt + P,
+				q.sk, q.sv));
+		}
+		GG_HIP_GBS(hipMemcpyAsync(q.ok, q.sk, (size_t) m * 8,
+					  hipMemcpyDeviceToHost, q.stm));
+		GG_HIP_GBS(hipMemcpyAsync(q.ov, q.sv, (size_t) m * 8,
+					  hipMemcpyDeviceToHost, q.stm));
+		GG_HIP_GBS(hipMemcpyAsync(q.hcur, q.dcnt + 2 * P,
+					  (P + 1) * 8,
+					  hipMemcpyDeviceToHost, q.stm));
+		GG_HIP_GBS(hipEventRecord(q.done, q.stm));
+		q.busy = true;
+	}
+	drain(B[0]);
+	drain(B[1]);
+#undef GG_HIP_GBS
+	for (int b = 0; b < 2; b++)
+	{
+		GbsBuf &q = B[b];
+
+		if (q.stm)
+			(void) hipStreamSynchronize(q.stm);
+		if (q.done)
+			(void) hipEventDestroy(q.done);
+		if (q.stm)
+			(void) hipStreamDestroy(q.stm);
+	}
+	return st;
+}
+
 extern "C" gg_status
 gg_engine_hash_join_i64_spill(const int64_t *build_keys,
 			      const int64_t *build_vals, int64_t nb,
@@ -4506,90 +4713,16 @@ gg_engine_hash_join_i64_spill(const int64_t *build_keys,
 	*out_npartitions = (int32_t) P;
 
 	/* stage both sides into host partitions (P==1: single "partition"
-	 * passthrough still goes through the same code path) */
-	int64_t chunk = budget_bytes / (8 * 4);
-	int64_t biggest = nb > np ? nb : np;
-
-	if (chunk > biggest)
-		chunk = biggest;	/* never allocate beyond the input */
-	if (chunk < 1024)
-		chunk = 1024;
+	 * passthrough still goes through the same code path), on the
+	 * shared pipelined stage */
 	std::vector<std::vector<int64_t>> bk(P), bv(P), pk(P), pi(P);
-	int64_t *dk = nullptr, *dv = nullptr, *sk = nullptr, *sv = nullptr;
-	unsigned long long *dcnt = nullptr;
-	gg_status st = GG_OK;
+	gg_status st = spill_stage_partitions(e, build_keys, build_vals,
+					      nb, budget_bytes, P, shift,
+					      bk, bv);
 
-	GG_HIP(hipMalloc((void **) &dk, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &dv, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &sk, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &sv, (size_t) chunk * 8));
-	GG_HIP(hipMalloc((void **) &dcnt, (P + 1) * 8));
-
-	std::vector<unsigned long long> counts(P), curs(P + 1);
-	auto stage = [&](const int64_t *keys, const int64_t *vals,
-			 int64_t n, bool idx_side,
-			 std::vector<std::vector<int64_t>> &out_k,
-			 std::vector<std::vector<int64_t>> &out_v)
-		-> gg_status
-	{
-		for (int64_t base = 0; base < n; base += chunk)
-		{
-			int64_t m = (n - base < chunk) ? n - base : chunk;
-
-			GG_HIP(hipMemcpy(dk, keys + base, (size_t) m * 8,
-					 hipMemcpyHostToDevice));
-			if (!idx_side)
-				GG_HIP(hipMemcpy(dv, vals + base,
-						 (size_t) m * 8,
-						 hipMemcpyHostToDevice));
-			GG_HIP(hipMemsetAsync(dcnt, 0, P * 8, e.stream));
-			GG_HIP(launch_gb_part_count(e.stream, dk, m, shift,
-						    dcnt));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			GG_HIP(hipMemcpy(counts.data(), dcnt, P * 8,
-					 hipMemcpyDeviceToHost));
-			curs[0] = 0;
-			for (uint64_t p2 = 0; p2 < P; p2++)
-				curs[p2 + 1] = curs[p2] + counts[p2];
-			GG_HIP(hipMemcpy(dcnt, curs.data(), P * 8,
-					 hipMemcpyHostToDevice));
-			if (idx_side)
-				GG_HIP(launch_gb_part_scatter_idx(
-					e.stream, dk, m, base, shift, dcnt,
-					sk, sv));
-			else
-				GG_HIP(launch_gb_part_scatter(
-					e.stream, dk, dv, m, shift, dcnt,
-					sk, sv));
-			GG_HIP(hipStreamSynchronize(e.stream));
-			for (uint64_t p2 = 0; p2 < P; p2++)
-			{
-				if (!counts[p2])
-					continue;
-				size_t old = out_k[p2].size();
-
-				out_k[p2].resize(old + counts[p2]);
-				out_v[p2].resize(old + counts[p2]);
-				GG_HIP(hipMemcpy(out_k[p2].data() + old,
-						 sk + curs[p2],
-						 counts[p2] * 8,
-						 hipMemcpyDeviceToHost));
-				GG_HIP(hipMemcpy(out_v[p2].data() + old,
-						 sv + curs[p2],
-						 counts[p2] * 8,
-						 hipMemcpyDeviceToHost));
-			}
-		}
-		return GG_OK;
-	};
-	st = stage(build_keys, build_vals, nb, false, bk, bv);
 	if (st == GG_OK)
-		st = stage(probe_keys, nullptr, np, true, pk, pi);
-	(void) hipFree(dk);
-	(void) hipFree(dv);
-	(void) hipFree(sk);
-	(void) hipFree(sv);
-	(void) hipFree(dcnt);
+		st = spill_stage_partitions(e, probe_keys, nullptr, np,
+					    budget_bytes, P, shift, pk, pi);
 	if (st != GG_OK)
 		return st;
 
@@ -4604,41 +4737,57 @@ gg_engine_hash_join_i64_spill(const int64_t *build_keys,
 		if (b_n == 0 || p_n == 0)
 			continue;
 		uint64_t nslots = next_pow2(2 * (uint64_t) b_n);
-		int64_t *d_bk = nullptr, *d_bv = nullptr, *d_pk = nullptr,
-			*d_pi = nullptr, *d_oi = nullptr, *d_ov = nullptr;
-		unsigned long long *tk = nullptr, *tv = nullptr,
-			*ctr = nullptr;
+		/* engine pools: no per-partition hipMalloc/free; pinned
+		 * bounce for the uploads (pageable hipMemcpy was the
+		 * serial cost here, like the group-by reload) */
+		int64_t *d_bk = (int64_t *) e.esget("sj.bk", (size_t) b_n * 8);
+		int64_t *d_bv = (int64_t *) e.esget("sj.bv", (size_t) b_n * 8);
+		int64_t *d_pk = (int64_t *) e.esget("sj.pk", (size_t) p_n * 8);
+		int64_t *d_pi = (int64_t *) e.esget("sj.pi", (size_t) p_n * 8);
+		int64_t *d_oi = (int64_t *) e.esget("sj.oi", (size_t) p_n * 8);
+		int64_t *d_ov = (int64_t *) e.esget("sj.ov", (size_t) p_n * 8);
+		unsigned long long *tk = (unsigned long long *)
+			e.esget("sj.tk", nslots * 8);
+		unsigned long long *tv = (unsigned long long *)
+			e.esget("sj.tv", nslots * 8);
+		unsigned long long *ctr = (unsigned long long *)
+			e.esget("sj.ctr", 8);
+		int64_t *hb = (int64_t *)
+			e.ehget("sj.hb", (size_t) (b_n > p_n ? b_n : p_n) * 8);
 
-		/* no-leak variant of GG_HIP: errors fall through to the
-		 * frees below instead of returning */
+		if (!d_bk || !d_bv || !d_pk || !d_pi || !d_oi || !d_ov ||
+		    !tk || !tv || !ctr || !hb)
+			st = fail(GG_ENOMEM, "join_spill buffers");
 #define GG_HIP_BRK(x) \
 		{ hipError_t e_ = (x); \
 		  if (st == GG_OK && e_ != hipSuccess) \
 			st = fail(GG_EGPU, "join_spill: %s", \
 				  hipGetErrorString(e_)); }
-		GG_HIP_BRK(hipMalloc((void **) &d_bk, (size_t) b_n * 8));
-		GG_HIP_BRK(hipMalloc((void **) &d_bv, (size_t) b_n * 8));
-		GG_HIP_BRK(hipMalloc((void **) &d_pk, (size_t) p_n * 8));
-		GG_HIP_BRK(hipMalloc((void **) &d_pi, (size_t) p_n * 8));
-		GG_HIP_BRK(hipMalloc((void **) &d_oi, (size_t) p_n * 8));
-		GG_HIP_BRK(hipMalloc((void **) &d_ov, (size_t) p_n * 8));
-		GG_HIP_BRK(hipMalloc((void **) &tk, nslots * 8));
-		GG_HIP_BRK(hipMalloc((void **) &tv, nslots * 8));
-		GG_HIP_BRK(hipMalloc((void **) &ctr, 8));
+		auto up = [&](int64_t *dst, const int64_t *src, int64_t nn)
+		{
+			if (st != GG_OK)
+				return;
+			int64_t nblk = (nn + (1 << 17) - 1) >> 17;
+
+			par_for_grain(nblk, 1, [&](int64_t blk)
+			{
+				int64_t o = blk << 17;
+				int64_t c = (nn - o < (1 << 17))
+					? nn - o : (1 << 17);
+
+				std::memcpy(hb + o, src + o, c * 8);
+			});
+			GG_HIP_BRK(hipMemcpyAsync(dst, hb, (size_t) nn * 8,
+						  hipMemcpyHostToDevice,
+						  e.stream));
+			GG_HIP_BRK(hipStreamSynchronize(e.stream));
+		};
 		if (st == GG_OK)
 		{
-			GG_HIP_BRK(hipMemcpy(d_bk, bk[p2].data(),
-					     (size_t) b_n * 8,
-					     hipMemcpyHostToDevice));
-			GG_HIP_BRK(hipMemcpy(d_bv, bv[p2].data(),
-					     (size_t) b_n * 8,
-					     hipMemcpyHostToDevice));
-			GG_HIP_BRK(hipMemcpy(d_pk, pk[p2].data(),
-					     (size_t) p_n * 8,
-					     hipMemcpyHostToDevice));
-			GG_HIP_BRK(hipMemcpy(d_pi, pi[p2].data(),
-					     (size_t) p_n * 8,
-					     hipMemcpyHostToDevice));
+			up(d_bk, bk[p2].data(), b_n);
+			up(d_bv, bv[p2].data(), b_n);
+			up(d_pk, pk[p2].data(), p_n);
+			up(d_pi, pi[p2].data(), p_n);
 			GG_HIP_BRK(hipMemsetAsync(tk, 0, nslots * 8,
 						  e.stream));
 			GG_HIP_BRK(hipMemsetAsync(ctr, 0, 8, e.stream));
@@ -4670,15 +4819,6 @@ gg_engine_hash_join_i64_spill(const int64_t *build_keys,
 			total += (int64_t) nm;
 		}
 #undef GG_HIP_BRK
-		(void) hipFree(d_bk);
-		(void) hipFree(d_bv);
-		(void) hipFree(d_pk);
-		(void) hipFree(d_pi);
-		(void) hipFree(d_oi);
-		(void) hipFree(d_ov);
-		(void) hipFree(tk);
-		(void) hipFree(tv);
-		(void) hipFree(ctr);
 		bk[p2].clear(); bk[p2].shrink_to_fit();
 		bv[p2].clear(); bv[p2].shrink_to_fit();
 		pk[p2].clear(); pk[p2].shrink_to_fit();
@@ -4739,25 +4879,9 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 	for (uint64_t t = P; t > 1; t >>= 1)
 		shift--;
 
-	/* chunked partition pass, double-buffered: the round-1 serial
-	 * form (pageable hipMemcpy both ways, a mid-chunk sync for the
-	 * host prefix, and up to P separate D2H copies per chunk) was
-	 * measured PCIe-staging-bound at 62 M rows/s.  Now: pinned
-	 * staging both directions, the cursor prefix computed on
-	 * device (k_gb_prefix2_u64 — no mid-chunk sync), ONE bulk D2H
-	 * per chunk, and two streams so chunk i+1's H2D and kernels
-	 * run while chunk i's scatter drains back and the host
-	 * distributes it (OpenMP memcpys) into the partition vectors.
-	 * Two buffer sets split the same device budget (chunk halved),
-	 * so peak GPU memory is unchanged. */
-	int64_t chunk = budget_bytes / (8 * 4 * 2);
-
-	if (chunk < 1024)
-		chunk = 1024;
-	if (chunk > n)
-		chunk = n;
+	/* partition pass: the shared pipelined stage (see
+	 * spill_stage_partitions above for the measured story) */
 	std::vector<std::vector<int64_t>> part_k(P), part_v(P);
-	gg_status st = GG_OK;
 	const bool spdbg = getenv("GG_SPILL_DEBUG") != nullptr;
 	auto spt0 = std::chrono::steady_clock::now();
 	auto spmark = [&](const char *what)
@@ -4770,177 +4894,11 @@ gg_engine_hash_groupby_i64_spill(const int64_t *keys, const int64_t *vals,
 			std::chrono::duration<double>(t1 - spt0).count());
 		spt0 = t1;
 	};
+	gg_status st = spill_stage_partitions(e, keys, vals, n,
+					      budget_bytes, P, shift,
+					      part_k, part_v);
 
-	struct GbsBuf
-	{
-		int64_t *dk = nullptr, *dv = nullptr;
-		int64_t *sk = nullptr, *sv = nullptr;
-		unsigned long long *dcnt = nullptr;	/* [P cnt][P work][P+1 pristine] */
-		int64_t *hk = nullptr, *hv = nullptr;	/* pinned in */
-		int64_t *ok = nullptr, *ov = nullptr;	/* pinned out */
-		unsigned long long *hcur = nullptr;	/* pinned P+1 */
-		hipStream_t stm = nullptr;
-		hipEvent_t done = nullptr;
-		int64_t m = 0;
-		bool busy = false;
-	} B[2];
-	bool alloc_ok = true;
-
-	for (int b = 0; b < 2 && alloc_ok; b++)
-	{
-		GbsBuf &q = B[b];
-		char nm[32];
-
-#define GBS_POOL(fld, kind, sz) 		(std::snprintf(nm, sizeof(nm), "gbs.%s%d", #fld, b), 		 (q.fld = (decltype(q.fld)) e.kind(nm, (sz))) != nullptr)
-		alloc_ok =
-			GBS_POOL(dk, esget, (size_t) chunk * 8) &&
-			GBS_POOL(dv, esget, (size_t) chunk * 8) &&
-			GBS_POOL(sk, esget, (size_t) chunk * 8) &&
-			GBS_POOL(sv, esget, (size_t) chunk * 8) &&
-			GBS_POOL(dcnt, esget, (3 * P + 1) * 8) &&
-			GBS_POOL(hk, ehget, (size_t) chunk * 8) &&
-			GBS_POOL(hv, ehget, (size_t) chunk * 8) &&
-			GBS_POOL(ok, ehget, (size_t) chunk * 8) &&
-			GBS_POOL(ov, ehget, (size_t) chunk * 8) &&
-			GBS_POOL(hcur, ehget, (P + 1) * 8) &&
-			hipStreamCreate(&q.stm) == hipSuccess &&
-			hipEventCreate(&q.done) == hipSuccess;
-#undef GBS_POOL
-	}
-	if (!alloc_ok)
-		st = fail(GG_ENOMEM, "groupby_spill staging");
-
-	/* error policy: record first failure, stop issuing, drain */
-#define GG_HIP_GBS(x) \
-	{ hipError_t e_ = (x); \
-	  if (st == GG_OK && e_ != hipSuccess) \
-		st = fail(GG_EGPU, "groupby_spill: %s", \
-			  hipGetErrorString(e_)); }
-
-	auto drain = [&](GbsBuf &q)
-	{
-		if (!q.busy)
-			return;
-		GG_HIP_GBS(hipEventSynchronize(q.done));
-		q.busy = false;
-		if (st != GG_OK)
-			return;
-		/* append staged rows to their partitions; resizes are
-		 * serial, the copies partition-parallel */
-		std::vector<size_t> old(P);
-
-		for (uint64_t p2 = 0; p2 < P; p2++)
-		{
-			size_t c = (size_t) (q.hcur[p2 + 1] - q.hcur[p2]);
-
-			old[p2] = part_k[p2].size();
-			if (c)
-			{
-				size_t need = old[p2] + c;
-
-				/* resize() grows to EXACTLY the request
-				 * in libstdc++ — per-chunk exact growth
-				 * re-copied every partition every chunk
-				 * (hundreds of GB of hidden memcpy at
-				 * 1B rows, measured 7.8s).  Reserve the
-				 * uniform-hash expectation up front and
-				 * grow geometrically past it. */
-				if (part_k[p2].capacity() < need)
-				{
-					size_t want = std::max(
-						need,
-						std::max(part_k[p2]
-							 .capacity() * 2,
-							 (size_t) (n / P +
-								   n / P / 8 +
-								   1024)));
-
-					part_k[p2].reserve(want);
-					part_v[p2].reserve(want);
-				}
-				part_k[p2].resize(need);
-				part_v[p2].resize(need);
-			}
-		}
-		par_for_grain((int64_t) P, 1, [&](int64_t p2)
-		{
-			size_t c = (size_t) (q.hcur[p2 + 1] - q.hcur[p2]);
-
-			if (!c)
-				return;
-			std::memcpy(part_k[p2].data() + old[p2],
-				    q.ok + q.hcur[p2], c * 8);
-			std::memcpy(part_v[p2].data() + old[p2],
-				    q.ov + q.hcur[p2], c * 8);
-		});
-	};
-
-	for (int64_t base = 0, it = 0; st == GG_OK && base < n;
-	     base += chunk, it ^= 1)
-	{
-		GbsBuf &q = B[it];
-		int64_t m = (n - base < chunk) ? n - base : chunk;
-
-		drain(q);	/* reclaim this buffer's previous chunk */
-		if (st != GG_OK)
-			break;
-		/* pageable -> pinned bounce (threaded memcpy;
-		 * hipMemcpyAsync from pageable memory would serialize
-		 * in the driver's staging path) */
-		{
-			int64_t nblk = (m + (1 << 17) - 1) >> 17;
-
-			par_for_grain(nblk, 1, [&](int64_t blk)
-			{
-				int64_t o = blk << 17;
-				int64_t c = (m - o < (1 << 17))
-					? m - o : (1 << 17);
-
-				std::memcpy(q.hk + o, keys + base + o,
-					    c * 8);
-				std::memcpy(q.hv + o, vals + base + o,
-					    c * 8);
-			});
-		}
-		GG_HIP_GBS(hipMemcpyAsync(q.dk, q.hk, (size_t) m * 8,
-					  hipMemcpyHostToDevice, q.stm));
-		GG_HIP_GBS(hipMemcpyAsync(q.dv, q.hv, (size_t) m * 8,
-					  hipMemcpyHostToDevice, q.stm));
-		GG_HIP_GBS(hipMemsetAsync(q.dcnt, 0, P * 8, q.stm));
-		GG_HIP_GBS(launch_gb_part_count(q.stm, q.dk, m, shift,
-						q.dcnt));
-		GG_HIP_GBS(launch_gb_prefix2_u64(q.stm, q.dcnt, (int) P,
-						 q.dcnt + P,
-						 q.dcnt + 2 * P));
-		GG_HIP_GBS(launch_gb_part_scatter(q.stm, q.dk, q.dv, m,
-						  shift, q.dcnt + P,
-						  q.sk, q.sv));
-		GG_HIP_GBS(hipMemcpyAsync(q.ok, q.sk, (size_t) m * 8,
-					  hipMemcpyDeviceToHost, q.stm));
-		GG_HIP_GBS(hipMemcpyAsync(q.ov, q.sv, (size_t) m * 8,
-					  hipMemcpyDeviceToHost, q.stm));
-		GG_HIP_GBS(hipMemcpyAsync(q.hcur, q.dcnt + 2 * P,
-					  (P + 1) * 8,
-					  hipMemcpyDeviceToHost, q.stm));
-		GG_HIP_GBS(hipEventRecord(q.done, q.stm));
-		q.m = m;
-		q.busy = true;
-	}
-	drain(B[0]);
-	drain(B[1]);
 	spmark("partition pass");
-#undef GG_HIP_GBS
-	for (int b = 0; b < 2; b++)
-	{
-		GbsBuf &q = B[b];
-
-		if (q.stm)
-			(void) hipStreamSynchronize(q.stm);
-		if (q.done)
-			(void) hipEventDestroy(q.done);
-		if (q.stm)
-			(void) hipStreamDestroy(q.stm);
-	}
 	if (st != GG_OK)
 		return st;
 
