@@ -359,7 +359,9 @@ __device__ int
 decode_one_block_text(const uint8_t *buf, int32_t size, int32_t rowcount,
 		      int version, uint8_t *pool, int64_t pool_base,
 		      unsigned long long *out_offs, uint32_t *out_lens,
-		      uint8_t *out_nulls, int64_t out_base)
+		      uint8_t *out_nulls, int64_t out_base,
+		      unsigned long long *src_offs,
+		      unsigned long long src_bias)
 {
 	const uint8_t *p = buf;
 	int16_t flags;
@@ -456,6 +458,8 @@ decode_one_block_text(const uint8_t *buf, int32_t size, int32_t rowcount,
 				out_nulls[out_base + nth] = 0;
 				out_offs[out_base + nth] = cur_off;
 				out_lens[out_base + nth] = cur_len;
+				if (src_offs)
+					src_offs[out_base + nth] = ~0ull;
 				continue;
 			}
 			if (has_null)
@@ -466,6 +470,9 @@ decode_one_block_text(const uint8_t *buf, int32_t size, int32_t rowcount,
 					out_nulls[out_base + nth] = 1;
 					out_offs[out_base + nth] = 0;
 					out_lens[out_base + nth] = 0;
+					if (src_offs)
+						src_offs[out_base + nth] =
+							~0ull;
 					continue;
 				}
 			}
@@ -514,9 +521,18 @@ decode_one_block_text(const uint8_t *buf, int32_t size, int32_t rowcount,
 				}
 				if (datump + ppos + vlen > datum_after)
 					return 3;
-				for (uint32_t z = 0; z < paylen; z++)
-					pool[pool_base + wpos + z] =
-						datump[payoff + z];
+				if (src_offs)
+					/* two-phase: the parallel copy
+					 * kernel moves the bytes */
+					src_offs[out_base + nth] =
+						src_bias +
+						(unsigned long long)
+						(datump - buf) +
+						(unsigned long long) payoff;
+				else
+					for (uint32_t z = 0; z < paylen; z++)
+						pool[pool_base + wpos + z] =
+							datump[payoff + z];
 				cur_off = (unsigned long long)
 					(pool_base + wpos);
 				cur_len = paylen;
@@ -543,7 +559,8 @@ k_dsb_decode_text(const uint8_t *__restrict__ stream,
 		  unsigned long long *__restrict__ out_offs,
 		  uint32_t *__restrict__ out_lens,
 		  uint8_t *__restrict__ out_nulls,
-		  unsigned long long *__restrict__ err)
+		  unsigned long long *__restrict__ err,
+		  unsigned long long *__restrict__ src_offs)
 {
 	int64_t stride = (int64_t) gridDim.x * blockDim.x;
 
@@ -553,14 +570,61 @@ k_dsb_decode_text(const uint8_t *__restrict__ stream,
 		int64_t off = offsets[b];
 		const uint8_t *src = off >= 0 ? stream + off
 			: spill + (-off - 1);
+		unsigned long long bias = off >= 0
+			? (unsigned long long) off
+			: (1ull << 63) | (unsigned long long) (-off - 1);
 		int rc = decode_one_block_text(src, sizes[b],
 					       rowcounts[b], version, pool,
 					       pool_offsets[b], out_offs,
 					       out_lens, out_nulls,
-					       out_offsets[b]);
+					       out_offsets[b], src_offs,
+					       bias);
 
 		if (rc)
 			atomicOr(err, 1ull << rc);
+	}
+}
+
+/* phase 2 of the split text decode: one thread per ROW copies its
+ * payload (parse emitted pool offset + encoded source offset).  The
+ * single-pass form copied bytes serially inside the per-BLOCK parse
+ * thread — ~5k threads moving 165 MB one byte at a time was the
+ * measured 600 M rows/s ceiling (round-3 candidate list). */
+__global__ void
+k_dsb_text_copy(const uint8_t *__restrict__ stream,
+		const uint8_t *__restrict__ spill,
+		const unsigned long long *__restrict__ src_offs,
+		const unsigned long long *__restrict__ out_offs,
+		const uint32_t *__restrict__ out_lens, int64_t nrows,
+		uint8_t *__restrict__ pool)
+{
+	int64_t stride = (int64_t) gridDim.x * blockDim.x;
+
+	for (int64_t r = (int64_t) blockIdx.x * blockDim.x + threadIdx.x;
+	     r < nrows; r += stride)
+	{
+		unsigned long long so = src_offs[r];
+
+		if (so == ~0ull)
+			continue;
+		{
+			const uint8_t *sp = (so >> 63)
+				? spill + (so & ((1ull << 63) - 1))
+				: stream + so;
+			uint8_t *dp = pool + out_offs[r];
+			uint32_t len = out_lens[r];
+			uint32_t z = 0;
+
+			for (; z + 4 <= len; z += 4)
+			{
+				uint32_t w;
+
+				memcpy(&w, sp + z, 4);
+				memcpy(dp + z, &w, 4);
+			}
+			for (; z < len; z++)
+				dp[z] = sp[z];
+		}
 	}
 }
 
@@ -641,7 +705,8 @@ launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
 		       const int64_t *pool_offsets, int32_t nblocks,
 		       int version, uint8_t *pool,
 		       unsigned long long *out_offs, uint32_t *out_lens,
-		       uint8_t *out_nulls, unsigned long long *err)
+		       uint8_t *out_nulls, unsigned long long *err,
+		       unsigned long long *src_offs)
 {
 	int blocks = (nblocks + 255) / 256;
 
@@ -652,7 +717,26 @@ launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
 	hipLaunchKernelGGL(k_dsb_decode_text, dim3(blocks), dim3(256), 0, s,
 			   stream, spill, offsets, sizes, rowcounts,
 			   out_offsets, pool_offsets, nblocks, version, pool,
-			   out_offs, out_lens, out_nulls, err);
+			   out_offs, out_lens, out_nulls, err, src_offs);
+	return hipGetLastError();
+}
+
+hipError_t
+launch_dsb_text_copy(hipStream_t s, const uint8_t *stream,
+		     const uint8_t *spill,
+		     const unsigned long long *src_offs,
+		     const unsigned long long *out_offs,
+		     const uint32_t *out_lens, int64_t nrows, uint8_t *pool)
+{
+	int blocks = (int) ((nrows + 255) / 256);
+
+	if (blocks > 4096)
+		blocks = 4096;
+	if (blocks < 1)
+		blocks = 1;
+	hipLaunchKernelGGL(k_dsb_text_copy, dim3(blocks), dim3(256), 0, s,
+			   stream, spill, src_offs, out_offs, out_lens,
+			   nrows, pool);
 	return hipGetLastError();
 }
 

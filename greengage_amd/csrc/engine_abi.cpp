@@ -3476,6 +3476,7 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 	unsigned long long *d_out_offs = nullptr;
 	uint32_t *d_lens = nullptr;
 	unsigned long long *d_err = nullptr;
+	unsigned long long *d_srco = nullptr;
 
 #define GG_HIP_AT(x) \
 	{ hipError_t e_ = (x); \
@@ -3497,6 +3498,7 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 	GG_HIP_AT(hipMalloc((void **) &d_lens, (size_t) total_rows * 4));
 	GG_HIP_AT(hipMalloc((void **) &d_nulls, (size_t) total_rows));
 	GG_HIP_AT(hipMalloc((void **) &d_err, 8));
+	GG_HIP_AT(hipMalloc((void **) &d_srco, (size_t) total_rows * 8));
 	if (st == GG_OK)
 	{
 		if (stream_len)
@@ -3520,10 +3522,19 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 		GG_HIP_AT(hipMemcpyAsync(d_rows, rows.data(), nb * 4,
 					 hipMemcpyHostToDevice, e.stream));
 		GG_HIP_AT(hipMemsetAsync(d_err, 0, 8, e.stream));
+		/* 0xFF = the ~0 "no payload" sentinel: rows of a block
+		 * that errors out mid-parse stay skipped by the copy */
+		GG_HIP_AT(hipMemsetAsync(d_srco, 0xFF,
+					 (size_t) total_rows * 8,
+					 e.stream));
 		GG_HIP_AT(launch_dsb_decode_text(
 			e.stream, d_stream, d_spill, d_offs, d_sizes,
 			d_rows, d_oo, d_po, (int32_t) nb, dsb_version,
-			d_pool, d_out_offs, d_lens, d_nulls, d_err));
+			d_pool, d_out_offs, d_lens, d_nulls, d_err,
+			d_srco));
+		GG_HIP_AT(launch_dsb_text_copy(
+			e.stream, d_stream, d_spill, d_srco, d_out_offs,
+			d_lens, total_rows, d_pool));
 		GG_HIP_AT(hipStreamSynchronize(e.stream));
 	}
 	if (st == GG_OK)
@@ -3560,6 +3571,8 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 	(void) hipFree(d_po);
 	(void) hipFree(d_sizes);
 	(void) hipFree(d_rows);
+	(void) hipFree(d_srco);
+	(void) hipFree(d_srco);
 	(void) hipFree(d_out_offs);
 	(void) hipFree(d_lens);
 	(void) hipFree(d_nulls);
@@ -3645,6 +3658,9 @@ gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
 	GG_HIP(hipMalloc((void **) &d_lens, (size_t) total_rows * 4));
 	GG_HIP(hipMalloc((void **) &d_nulls, (size_t) total_rows));
 	GG_HIP(hipMalloc((void **) &d_err, 8));
+	unsigned long long *d_srco = nullptr;
+
+	GG_HIP(hipMalloc((void **) &d_srco, (size_t) total_rows * 8));
 	GG_HIP(hipMemcpy(d_stream, stream, (size_t) stream_len,
 			 hipMemcpyHostToDevice));
 	GG_HIP(hipMemcpy(d_offs, offs.data(), nb * 8,
@@ -3658,12 +3674,18 @@ gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
 	GG_HIP(hipMemcpy(d_rows, rows.data(), nb * 4,
 			 hipMemcpyHostToDevice));
 	GG_HIP(hipMemset(d_err, 0, 8));
+	GG_HIP(hipMemsetAsync(d_srco, 0xFF, (size_t) total_rows * 8,
+			      e.stream));
 	{
 		hipError_t he = launch_dsb_decode_text(
 			e.stream, d_stream, nullptr, d_offs, d_sizes,
 			d_rows, d_oo, d_po, (int32_t) nb, version, d_pool,
-			d_out_offs, d_lens, d_nulls, d_err);
+			d_out_offs, d_lens, d_nulls, d_err, d_srco);
 
+		if (he == hipSuccess)
+			he = launch_dsb_text_copy(
+				e.stream, d_stream, nullptr, d_srco,
+				d_out_offs, d_lens, total_rows, d_pool);
 		if (he != hipSuccess)
 			st = fail(GG_EGPU, "dsb_decode_text: %s",
 				  hipGetErrorString(he));

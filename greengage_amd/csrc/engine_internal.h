@@ -787,7 +787,14 @@ hipError_t launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
 				  uint8_t *pool,
 				  unsigned long long *out_offs,
 				  uint32_t *out_lens, uint8_t *out_nulls,
-				  unsigned long long *err);
+				  unsigned long long *err,
+				  unsigned long long *src_offs);
+hipError_t launch_dsb_text_copy(hipStream_t s, const uint8_t *stream,
+				const uint8_t *spill,
+				const unsigned long long *src_offs,
+				const unsigned long long *out_offs,
+				const uint32_t *out_lens, int64_t nrows,
+				uint8_t *pool);
 hipError_t launch_dsb_decode2(hipStream_t s, const uint8_t *stream,
 			      const uint8_t *spill,
 			      const int64_t *offsets,
