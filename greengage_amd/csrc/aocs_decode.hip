@@ -714,6 +714,8 @@ launch_dsb_decode_text(hipStream_t s, const uint8_t *stream,
 		blocks = 2048;
 	if (blocks < 1)
 		blocks = 1;
+	(void) hipGetLastError();	/* clear stale state: we return
+					 * the launch's own status */
 	hipLaunchKernelGGL(k_dsb_decode_text, dim3(blocks), dim3(256), 0, s,
 			   stream, spill, offsets, sizes, rowcounts,
 			   out_offsets, pool_offsets, nblocks, version, pool,

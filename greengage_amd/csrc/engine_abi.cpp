@@ -3481,7 +3481,7 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 #define GG_HIP_AT(x) \
 	{ hipError_t e_ = (x); \
 	  if (st == GG_OK && e_ != hipSuccess) \
-		st = fail(GG_EGPU, "decode_ao_text: %s", \
+		st = fail(GG_EGPU, "decode_ao_text:%d: %s", __LINE__, \
 			  hipGetErrorString(e_)); }
 	GG_HIP_AT(hipMalloc((void **) &d_stream,
 			    stream_len ? (size_t) stream_len : 1));
@@ -3498,7 +3498,11 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 	GG_HIP_AT(hipMalloc((void **) &d_lens, (size_t) total_rows * 4));
 	GG_HIP_AT(hipMalloc((void **) &d_nulls, (size_t) total_rows));
 	GG_HIP_AT(hipMalloc((void **) &d_err, 8));
-	GG_HIP_AT(hipMalloc((void **) &d_srco, (size_t) total_rows * 8));
+	{ hipError_t e_ = hipMalloc((void **) &d_srco,
+				    (size_t) total_rows * 8);
+	  if (st == GG_OK && e_ != hipSuccess)
+		st = fail(GG_EGPU, "decode_ao_text srco-alloc: %s",
+			  hipGetErrorString(e_)); }
 	if (st == GG_OK)
 	{
 		if (stream_len)
@@ -3524,17 +3528,35 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 		GG_HIP_AT(hipMemsetAsync(d_err, 0, 8, e.stream));
 		/* 0xFF = the ~0 "no payload" sentinel: rows of a block
 		 * that errors out mid-parse stay skipped by the copy */
-		GG_HIP_AT(hipMemsetAsync(d_srco, 0xFF,
-					 (size_t) total_rows * 8,
-					 e.stream));
-		GG_HIP_AT(launch_dsb_decode_text(
-			e.stream, d_stream, d_spill, d_offs, d_sizes,
-			d_rows, d_oo, d_po, (int32_t) nb, dsb_version,
-			d_pool, d_out_offs, d_lens, d_nulls, d_err,
-			d_srco));
-		GG_HIP_AT(launch_dsb_text_copy(
-			e.stream, d_stream, d_spill, d_srco, d_out_offs,
-			d_lens, total_rows, d_pool));
+		{ hipError_t e_ = hipMemsetAsync(d_srco, 0xFF,
+						 (size_t) total_rows * 8,
+						 e.stream);
+		  if (st == GG_OK && e_ != hipSuccess)
+			st = fail(GG_EGPU, "decode_ao_text srco-set: %s",
+				  hipGetErrorString(e_)); }
+		{
+			const char *sp_ = getenv("GG_TEXT_SPLIT");
+			bool split_ = !(sp_ && sp_[0] == '0');
+
+			GG_HIP_AT(launch_dsb_decode_text(
+				e.stream, d_stream, d_spill, d_offs,
+				d_sizes, d_rows, d_oo, d_po, (int32_t) nb,
+				dsb_version, d_pool, d_out_offs, d_lens,
+				d_nulls, d_err,
+				split_ ? d_srco : nullptr));
+			if (split_)
+			{
+				hipError_t e_ = launch_dsb_text_copy(
+					e.stream, d_stream, d_spill,
+					d_srco, d_out_offs, d_lens,
+					total_rows, d_pool);
+
+				if (st == GG_OK && e_ != hipSuccess)
+					st = fail(GG_EGPU,
+						  "decode_ao_text copy: %s",
+						  hipGetErrorString(e_));
+			}
+		}
 		GG_HIP_AT(hipStreamSynchronize(e.stream));
 	}
 	if (st == GG_OK)
@@ -3571,7 +3593,6 @@ gg_engine_aocs_decode_ao_text(const uint8_t *stream, int64_t stream_len,
 	(void) hipFree(d_po);
 	(void) hipFree(d_sizes);
 	(void) hipFree(d_rows);
-	(void) hipFree(d_srco);
 	(void) hipFree(d_srco);
 	(void) hipFree(d_out_offs);
 	(void) hipFree(d_lens);
@@ -3677,12 +3698,15 @@ gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
 	GG_HIP(hipMemsetAsync(d_srco, 0xFF, (size_t) total_rows * 8,
 			      e.stream));
 	{
+		const char *sp_ = getenv("GG_TEXT_SPLIT");
+		bool split_ = !(sp_ && sp_[0] == '0');
 		hipError_t he = launch_dsb_decode_text(
 			e.stream, d_stream, nullptr, d_offs, d_sizes,
 			d_rows, d_oo, d_po, (int32_t) nb, version, d_pool,
-			d_out_offs, d_lens, d_nulls, d_err, d_srco);
+			d_out_offs, d_lens, d_nulls, d_err,
+			split_ ? d_srco : nullptr);
 
-		if (he == hipSuccess)
+		if (split_ && he == hipSuccess)
 			he = launch_dsb_text_copy(
 				e.stream, d_stream, nullptr, d_srco,
 				d_out_offs, d_lens, total_rows, d_pool);
@@ -3726,6 +3750,7 @@ gg_engine_aocs_decode_text(const uint8_t *stream, int64_t stream_len,
 	(void) hipFree(d_lens);
 	(void) hipFree(d_nulls);
 	(void) hipFree(d_err);
+	(void) hipFree(d_srco);
 	return st;
 }
 
