@@ -8,6 +8,8 @@
   > gpurun_out/r02z_planbench.log 2>&1
 { timeout 200 python -c "import __graft_entry__ as g; g.smoke()"; \
   echo "SM_RC=$?"; } > gpurun_out/r02z_smoke.log 2>&1
+bash tools/gpu_spill_bench.sh
+cp gpurun_out/spill_bench.log gpurun_out/r02z_spill.log
 export TMPDIR=/tmp
 cd /tmp
 R=/root/repo
@@ -26,3 +28,4 @@ cd $R
 grep -E "passed|failed|RC=" gpurun_out/r02z_pytest.log gpurun_out/r02z_prof.log
 grep -oE "\"value\": [0-9.]+|\"ms_per_step\": [0-9.]+|\"q3_ms_per_step\": [0-9.]+|\"q5_ms_per_step\": [0-9.]+|\"frac\": [0-9.]+" gpurun_out/r02z_bench.log | head -5
 tail -5 gpurun_out/r02z_planbench.log; tail -1 gpurun_out/r02z_smoke.log
+tail -6 gpurun_out/r02z_spill.log
