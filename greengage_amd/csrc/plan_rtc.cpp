@@ -198,9 +198,23 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 	s += fast ? "#define VAT unsigned long long\n"
 		  : "#define VAT __int128\n";
 
-	s += "extern \"C\" __global__ "
-	     "__launch_bounds__(256, 4)\nvoid plan_kernel(PlanDev P)\n{\n"
-	     "\tconst int64_t stride = (int64_t) gridDim.x * blockDim.x;\n";
+	{
+		/* waves/SIMD occupancy hint; 4 matches the hand kernels,
+		 * sweepable (GG_PLAN_WAVES) — fewer waves = more VGPRs
+		 * per wave for the register-heavy baked tier */
+		int waves = 4;
+		const char *wv = getenv("GG_PLAN_WAVES");
+
+		if (wv && atoi(wv) >= 1 && atoi(wv) <= 8)
+			waves = atoi(wv);
+		std::snprintf(buf, sizeof(buf),
+			      "extern \"C\" __global__ "
+			      "__launch_bounds__(256, %d)\n"
+			      "void plan_kernel(PlanDev P)\n{\n"
+			      "\tconst int64_t stride = "
+			      "(int64_t) gridDim.x * blockDim.x;\n", waves);
+		s += buf;
+	}
 
 	/* ---- per-row evaluation as a macro-free inline sequence ---- */
 	std::string rowfn;

@@ -142,3 +142,28 @@ for repl in (4, 8, 16, 32):
     print(f"grouped RTC LREPL={repl}: {d*1000:.0f} us/launch")
 os.environ.pop("GG_PLAN_LREPL")
 print("PLAN_BENCH5_OK")
+
+# waves/SIMD sweep for the baked grouped kernel (GG_PLAN_WAVES)
+for wv in (2, 3, 4):
+    os.environ["GG_PLAN_WAVES"] = str(wv)
+    pw = eng.compile_plan(
+        li, preds=[("shipdate", NEG_INF, cutoff + 1)],
+        group_cols=["rflag", "lstatus"],
+        aggs=["count", ("sum", [("qty", "id")]), ("sum", [("price", "id")]),
+              ("sum", [("disc", "id")]),
+              ("sum", [("price", "id"), ("disc", "sub100")]),
+              ("sum", [("price", "id"), ("disc", "sub100"),
+                       ("tax", "add100")])])
+    gw = eng.execute_plan(pw, max_groups=16)   # first: generic + bake
+    assert gw == gq0
+    before = {s["name"]: dict(s) for s in eng.stats(pw)}
+    for _ in range(5):
+        gw = eng.execute_plan(pw, max_groups=16)
+    after = {s["name"]: s for s in eng.stats(pw)}
+    d = ((after["plan_scan_agg"]["total_ms"]
+          - before["plan_scan_agg"]["total_ms"])
+         / (after["plan_scan_agg"]["launches"]
+            - before["plan_scan_agg"]["launches"]))
+    print(f"grouped baked WAVES={wv}: {d*1000:.0f} us/launch")
+os.environ.pop("GG_PLAN_WAVES", None)
+print("PLAN_BENCH6_OK")
