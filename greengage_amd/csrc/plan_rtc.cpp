@@ -277,8 +277,42 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 	rowfn += "\t\treturn true;\n\t};\n";
 	s += rowfn;
 
-	/* agg input values: emits "__int128 aY; bool aY_ok;" per agg */
+	/* agg input values.  Factor-column loads are DEDUPLICATED at
+	 * generation time: the same device pointer can appear in many
+	 * aggregates (Q1 reads price in three sums) and clang cannot
+	 * prove P.aggs[i].col[f] aliases across aggs, so without this
+	 * the kernel issued 13 per-row VMEM loads where the
+	 * hand-written k_q1_agg issues 7 (measured via offline ISA
+	 * dump, tools/rtc_dump_local.cpp). */
 	std::string aggfn = "\tauto agg_vals = [&](int64_t i, VAT *av, bool *aok)\n\t{\n";
+	std::vector<std::pair<const void *, int>> uniq;
+	auto var_of = [&](int a, int f) -> int
+	{
+		const void *c = D.aggs[a].col[f];
+		int w = D.aggs[a].width[f];
+
+		for (size_t u = 0; u < uniq.size(); u++)
+			if (uniq[u].first == c && uniq[u].second == w)
+				return (int) u;
+		uniq.push_back({c, w});
+		{
+			char d[32], ptr[64];
+
+			std::snprintf(d, sizeof(d), "xu%d",
+				      (int) uniq.size() - 1);
+			std::snprintf(ptr, sizeof(ptr),
+				      "P.aggs[%d].col[%d]", a, f);
+			emit_ld(aggfn, d, ptr, w, "i", nt);
+		}
+		return (int) uniq.size() - 1;
+	};
+
+	/* pass 1: one load per distinct (pointer, width) */
+	for (int a = 0; a < D.naggs; a++)
+		if (D.aggs[a].kind == 2)
+			for (int f = 0; f < D.aggs[a].nf; f++)
+				(void) var_of(a, f);
+	/* pass 2: per-agg strict flags + products over the shared vars */
 	for (int a = 0; a < D.naggs; a++)
 	{
 		if (D.aggs[a].kind == 0)
@@ -308,26 +342,17 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 		aggfn += buf;
 		for (int f = 0; f < D.aggs[a].nf; f++)
 		{
-			char d[32], ptr[64];
+			int u = var_of(a, f);
 
-			std::snprintf(d, sizeof(d), "x%d_%d", a, f);
-			std::snprintf(ptr, sizeof(ptr), "P.aggs[%d].col[%d]",
-				      a, f);
-			emit_ld(aggfn, d, ptr, D.aggs[a].width[f], "i", nt);
 			if (D.aggs[a].mod[f] == 1)
-			{
 				std::snprintf(buf, sizeof(buf),
-					      "\t\t%s = 100 - %s;\n", d, d);
-				aggfn += buf;
-			}
+					      "\t\tv *= (VAT) (100 - xu%d);\n", u);
 			else if (D.aggs[a].mod[f] == 2)
-			{
 				std::snprintf(buf, sizeof(buf),
-					      "\t\t%s = 100 + %s;\n", d, d);
-				aggfn += buf;
-			}
-			std::snprintf(buf, sizeof(buf),
-				      "\t\tv *= (VAT) x%d_%d;\n", a, f);
+					      "\t\tv *= (VAT) (100 + xu%d);\n", u);
+			else
+				std::snprintf(buf, sizeof(buf),
+					      "\t\tv *= (VAT) xu%d;\n", u);
 			aggfn += buf;
 		}
 		std::snprintf(buf, sizeof(buf), "\t\tav[%d] = v; }\n", a);
