@@ -397,6 +397,36 @@ exec_plan(Pipeline *p, void *arena, size_t bytes, size_t *written)
 				   dout, ctr, cap));
 	GG_HIP(hipStreamSynchronize(e.stream));
 	GG_HIP(hipMemcpy(&herr, derr, 8, hipMemcpyDeviceToHost));
+	if (herr == 2ull && R->rtc_baked)
+	{
+		/* the baked kernel saw a group code outside its baked
+		 * set (impossible over immutable tables; defensive) —
+		 * drop the baked binary and redo this execute on the
+		 * generic kernel */
+		R->rtc_baked.reset();
+		p->stat("path_plan_bake_miss").launches++;
+		GG_HIP(launch_fill_u64(e.stream, tkeys, nslots,
+				       PL_EMPTY_HOST));
+		GG_HIP(hipMemsetAsync(tvals, 0,
+				      nslots * (size_t) naggs * 16,
+				      e.stream));
+		GG_HIP(hipMemsetAsync(derr, 0, 8, e.stream));
+		{
+			int64_t g = (R->dev.n + 255) / 256;
+
+			if (g > 2048)
+				g = 2048;
+			if (g < 1)
+				g = 1;
+			GG_TRY(plan_rtc_launch(e.stream, R->rtc, R->dev,
+					       (int) g, 256));
+		}
+		GG_HIP(hipMemsetAsync(ctr, 0, 8, e.stream));
+		GG_HIP(launch_plan_compact(e.stream, tkeys, tvals, nslots,
+					   naggs, dout, ctr, cap));
+		GG_HIP(hipStreamSynchronize(e.stream));
+		GG_HIP(hipMemcpy(&herr, derr, 8, hipMemcpyDeviceToHost));
+	}
 	if (herr)
 		return fail(GG_EINVAL,
 			    "plan: group cardinality exceeds %llu "
