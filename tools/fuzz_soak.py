@@ -18,7 +18,8 @@ from greengage_amd import Engine
 
 eng = Engine(device=0, n_segments=1, segment_id=0)
 n_iter = int(sys.argv[1]) if len(sys.argv) > 1 else 150
-rng = np.random.default_rng(987654321)
+seed = int(sys.argv[2]) if len(sys.argv) > 2 else 987654321
+rng = np.random.default_rng(seed)
 fails = 0
 for it in range(n_iter):
     try:
