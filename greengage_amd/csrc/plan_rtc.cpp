@@ -159,11 +159,12 @@ plan_rtc_compile(const PlanDev &D, bool has_gnull0, bool has_gnull1,
 	if (nbake <= 0)
 		fast = false;	/* the u64 fast tier exists only baked */
 
-	/* NT measured flat on the baked kernel (4.90 vs 4.96 ms same box,
-	 * r02k) unlike k_q1_agg's +11%: the generated kernel's extra ALU
-	 * already covers the L2 latency.  Off by default, sweepable. */
+	/* NT loads: measured FLAT while the row loop carried the dead
+	 * fallback (4.90 vs 4.96 ms, r02k) but +12% once the loop got
+	 * lean (4.20 -> 3.76 ms, r02nt2) — same physics as k_q1_agg's
+	 * +11%.  Default ON for baked kernels; GG_PLAN_RTC_NT=0 reverts. */
 	const char *ntv = getenv("GG_PLAN_RTC_NT");
-	bool nt = nbake > 0 && ntv && ntv[0] == '1';
+	bool nt = nbake > 0 && !(ntv && ntv[0] == '0');
 
 	std::string s = STRUCT_DEFS;
 	char buf[512];
